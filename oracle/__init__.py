@@ -1,0 +1,314 @@
+"""CPU oracle bindings — TEST INFRASTRUCTURE ONLY.
+
+ctypes bindings to liboracle.so (the C restatement of the reference's
+vmselect rollup path; see vm_oracle.h). Only tests/, __graft_entry__.smoke()
+and bench.py's cpu_baseline leg may import this package. The product path
+(victoriametrics_amd/) must never import it.
+"""
+import ctypes
+import os
+import subprocess
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_DIR, "liboracle.so")
+
+# Func-name → enum mapping, mirroring the reference's rollupFuncs map keys
+# (app/vmselect/promql/rollup.go:24-108).  Aliases point at the shared
+# implementation exactly as the reference map does.
+FUNC_IDS = {
+    "rate": 0,
+    "increase": 1,
+    "increase_pure": 2,
+    "delta": 3,
+    "delta_prometheus": 4,
+    "rate_prometheus": 5,
+    "irate": 6,
+    "ideriv": 7,
+    "idelta": 8,
+    "deriv_fast": 9,
+    "avg_over_time": 10,
+    "min_over_time": 11,
+    "max_over_time": 12,
+    "sum_over_time": 13,
+    "sum2_over_time": 14,
+    "count_over_time": 15,
+    "first_over_time": 16,
+    "last_over_time": 17,
+    "quantile_over_time": 18,
+    "median_over_time": 19,
+    "stddev_over_time": 20,
+    "stdvar_over_time": 21,
+    "changes": 22,
+    "changes_prometheus": 23,
+    "resets": 24,
+    "lag": 25,
+    "lifetime": 26,
+    "scrape_interval": 27,
+    "rate_over_sum": 28,
+    "range_over_time": 29,
+    "tfirst_over_time": 30,
+    "tlast_over_time": 31,
+    "tmin_over_time": 32,
+    "tmax_over_time": 33,
+    "tlast_change_over_time": 34,
+    "geomean_over_time": 35,
+    "present_over_time": 36,
+    "absent_over_time": 37,
+    "stale_samples_over_time": 38,
+    "count_le_over_time": 39,
+    "count_gt_over_time": 40,
+    "count_eq_over_time": 41,
+    "count_ne_over_time": 42,
+    "share_le_over_time": 43,
+    "share_gt_over_time": 44,
+    "share_eq_over_time": 45,
+    "sum_le_over_time": 46,
+    "sum_gt_over_time": 47,
+    "sum_eq_over_time": 48,
+    "deriv": 49,
+    "predict_linear": 50,
+    "ascent_over_time": 51,
+    "descent_over_time": 52,
+    "zscore_over_time": 53,
+    "integrate": 54,
+    "distinct_over_time": 55,
+    "increases_over_time": 56,
+    "decreases_over_time": 57,
+    "mad_over_time": 58,
+    "default_rollup": 59,
+    "mode_over_time": 60,
+    "duration_over_time": 61,
+    "outlier_iqr_over_time": 62,
+    # aliases (same implementations as in the reference's map)
+    "increase_prometheus": 4,   # rollupDeltaPrometheus
+    "timestamp": 31,            # rollupTlast
+    "timestamp_with_name": 31,  # rollupTlast
+}
+
+# rollupFuncsRemoveCounterResets (rollup.go:223-232)
+REMOVE_COUNTER_RESETS_FUNCS = {
+    "increase", "increase_prometheus", "increase_pure", "irate", "rate",
+    "rate_prometheus", "rollup_increase", "rollup_rate",
+}
+
+# rollupFuncsSamplesScannedPerCall (rollup.go:238-263)
+SAMPLES_SCANNED_PER_CALL = {
+    "absent_over_time": 1, "count_over_time": 1, "default_rollup": 1,
+    "delta": 2, "delta_prometheus": 2, "deriv_fast": 2, "first_over_time": 1,
+    "idelta": 2, "ideriv": 2, "increase": 2, "increase_prometheus": 2,
+    "increase_pure": 2, "irate": 2, "lag": 1, "last_over_time": 1,
+    "lifetime": 2, "present_over_time": 1, "rate": 2, "rate_prometheus": 2,
+    "scrape_interval": 2, "tfirst_over_time": 1, "timestamp": 1,
+    "timestamp_with_name": 1, "tlast_over_time": 1,
+}
+
+# rollupFuncsCanAdjustWindow (rollup.go:204-219)
+CAN_ADJUST_WINDOW_FUNCS = {
+    "default_rollup", "deriv", "deriv_fast", "ideriv", "irate", "rate",
+    "rate_over_sum", "rollup", "rollup_candlestick", "rollup_deriv",
+    "rollup_rate", "rollup_scrape_interval", "scrape_interval", "timestamp",
+}
+
+AGGR_IDS = {
+    "none": 0, "sum": 1, "min": 2, "max": 3, "avg": 4,
+    "count": 5, "sum2": 6, "geomean": 7, "group": 8,
+}
+
+
+class RollupConfigC(ctypes.Structure):
+    _fields_ = [
+        ("func", ctypes.c_int32),
+        ("may_adjust_window", ctypes.c_int32),
+        ("start", ctypes.c_int64),
+        ("end", ctypes.c_int64),
+        ("step", ctypes.c_int64),
+        ("window", ctypes.c_int64),
+        ("lookback_delta", ctypes.c_int64),
+        ("min_staleness_interval", ctypes.c_int64),
+        ("is_default_rollup", ctypes.c_int32),
+        ("samples_scanned_per_call", ctypes.c_int32),
+        ("arg", ctypes.c_double),
+    ]
+
+
+def build(force=False):
+    """Compile liboracle.so with the committed Makefile (gcc)."""
+    if force or not os.path.exists(_LIB_PATH):
+        subprocess.run(["make", "-C", _DIR], check=True, capture_output=True)
+    return _LIB_PATH
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        build()
+        _lib = ctypes.CDLL(_LIB_PATH)
+        _lib.vm_grid_points.restype = ctypes.c_int64
+        _lib.vm_grid_points.argtypes = [ctypes.c_int64] * 3
+        _lib.vm_quantile.restype = ctypes.c_double
+        _lib.vm_quantile_sorted.restype = ctypes.c_double
+        _lib.vm_call_rollup_fn.restype = ctypes.c_double
+        _lib.vm_rollup_do.restype = ctypes.c_uint64
+        _lib.vm_stale_nan.restype = ctypes.c_double
+        _lib.vm_is_stale_nan.restype = ctypes.c_int
+        _lib.vm_is_stale_nan.argtypes = [ctypes.c_double]
+        _lib.vm_get_scrape_interval.restype = ctypes.c_int64
+        _lib.vm_get_max_prev_interval.restype = ctypes.c_int64
+        _lib.vm_get_max_prev_interval.argtypes = [ctypes.c_int64]
+        _lib.vm_drop_stale_nans.restype = ctypes.c_int64
+        _lib.vm_rollup_eval_batch.restype = ctypes.c_int
+    return _lib
+
+
+def _f64(a):
+    return np.ascontiguousarray(a, dtype=np.float64)
+
+
+def _i64(a):
+    return np.ascontiguousarray(a, dtype=np.int64)
+
+
+def _ptr(a, typ):
+    return a.ctypes.data_as(ctypes.POINTER(typ))
+
+
+def stale_nan():
+    return lib().vm_stale_nan()
+
+
+def grid_points(start, end, step):
+    return lib().vm_grid_points(start, end, step)
+
+
+def get_timestamps(start, end, step):
+    n = grid_points(start, end, step)
+    out = np.empty(n, dtype=np.int64)
+    lib().vm_get_timestamps(ctypes.c_int64(start), ctypes.c_int64(end),
+                            ctypes.c_int64(step), _ptr(out, ctypes.c_int64))
+    return out
+
+
+def remove_counter_resets(values, timestamps, msi=0):
+    v = _f64(values).copy()
+    t = _i64(timestamps)
+    lib().vm_remove_counter_resets(_ptr(v, ctypes.c_double), _ptr(t, ctypes.c_int64),
+                                   ctypes.c_int64(len(v)), ctypes.c_int64(msi))
+    return v
+
+
+def delta_values(values):
+    v = _f64(values).copy()
+    lib().vm_delta_values(_ptr(v, ctypes.c_double), ctypes.c_int64(len(v)))
+    return v
+
+
+def deriv_values(values, timestamps):
+    v = _f64(values).copy()
+    t = _i64(timestamps)
+    lib().vm_deriv_values(_ptr(v, ctypes.c_double), _ptr(t, ctypes.c_int64),
+                          ctypes.c_int64(len(v)))
+    return v
+
+
+def drop_stale_nans(values, timestamps):
+    v = _f64(values).copy()
+    t = _i64(timestamps).copy()
+    n = lib().vm_drop_stale_nans(_ptr(v, ctypes.c_double), _ptr(t, ctypes.c_int64),
+                                 ctypes.c_int64(len(v)))
+    return v[:n], t[:n]
+
+
+def quantile(phi, values):
+    v = _f64(values)
+    return lib().vm_quantile(ctypes.c_double(phi), _ptr(v, ctypes.c_double),
+                             ctypes.c_int64(len(v)))
+
+
+def call_rollup_fn(func_name, values, timestamps, prev_value=float("nan"),
+                   prev_timestamp=0, real_prev_value=float("nan"),
+                   real_next_value=float("nan"), curr_timestamp=0, idx=0,
+                   window=0, arg=0.0):
+    v = _f64(values)
+    t = _i64(timestamps)
+    return lib().vm_call_rollup_fn(
+        ctypes.c_int32(FUNC_IDS[func_name]), ctypes.c_double(prev_value),
+        ctypes.c_int64(prev_timestamp), _ptr(v, ctypes.c_double),
+        _ptr(t, ctypes.c_int64), ctypes.c_int64(len(v)),
+        ctypes.c_double(real_prev_value), ctypes.c_double(real_next_value),
+        ctypes.c_int64(curr_timestamp), ctypes.c_int64(idx),
+        ctypes.c_int64(window), ctypes.c_double(arg))
+
+
+def make_config(func_name, start, end, step, window=0, lookback_delta=0,
+                min_staleness_interval=0, arg=0.0, may_adjust_window=None,
+                samples_scanned_per_call=None):
+    """Build a RollupConfigC the way getRollupConfigs does (rollup.go:374-516)
+    when may_adjust_window/samples_scanned_per_call are None; pass explicit
+    values (e.g. 0) to mirror the reference's direct rollupConfig{} literals in
+    unit tests."""
+    if may_adjust_window is None:
+        may_adjust_window = func_name in CAN_ADJUST_WINDOW_FUNCS
+    if samples_scanned_per_call is None:
+        samples_scanned_per_call = SAMPLES_SCANNED_PER_CALL.get(func_name, 0)
+    return RollupConfigC(
+        func=FUNC_IDS[func_name],
+        may_adjust_window=1 if may_adjust_window else 0,
+        start=start, end=end, step=step, window=window,
+        lookback_delta=lookback_delta,
+        min_staleness_interval=min_staleness_interval,
+        is_default_rollup=1 if func_name == "default_rollup" else 0,
+        samples_scanned_per_call=samples_scanned_per_call,
+        arg=arg)
+
+
+def rollup_do(rc, values, timestamps):
+    v = _f64(values)
+    t = _i64(timestamps)
+    n_grid = grid_points(rc.start, rc.end, rc.step)
+    dst = np.empty(n_grid, dtype=np.float64)
+    scanned = lib().vm_rollup_do(ctypes.byref(rc), _ptr(v, ctypes.c_double),
+                                 _ptr(t, ctypes.c_int64), ctypes.c_int64(len(v)),
+                                 _ptr(dst, ctypes.c_double))
+    return dst, scanned
+
+
+def rollup_eval_batch(rc, ts, vals, offsets, group_ids=None, n_groups=0,
+                      aggr="none", remove_counter_resets=False,
+                      max_staleness_interval=0, drop_stale_nans=False,
+                      n_threads=1):
+    """Batch CSR evaluation — mirrors the product C-ABI shape."""
+    t = _i64(ts)
+    v = _f64(vals)
+    off = np.ascontiguousarray(offsets, dtype=np.uint64)
+    n_series = len(off) - 1
+    n_grid = grid_points(rc.start, rc.end, rc.step)
+    aggr_id = AGGR_IDS[aggr]
+    if group_ids is not None and aggr_id != 0:
+        gids = np.ascontiguousarray(group_ids, dtype=np.int32)
+        out = np.empty((n_groups, n_grid), dtype=np.float64)
+        counts = np.empty((n_groups, n_grid), dtype=np.float64)
+        gptr = _ptr(gids, ctypes.c_int32)
+        cptr = _ptr(counts, ctypes.c_double)
+    else:
+        out = np.empty((n_series, n_grid), dtype=np.float64)
+        counts = None
+        gptr = None
+        cptr = None
+    scanned = ctypes.c_uint64(0)
+    rcode = lib().vm_rollup_eval_batch(
+        ctypes.byref(rc), ctypes.c_int32(1 if remove_counter_resets else 0),
+        ctypes.c_int64(max_staleness_interval),
+        ctypes.c_int32(1 if drop_stale_nans else 0),
+        _ptr(t, ctypes.c_int64), _ptr(v, ctypes.c_double),
+        _ptr(off, ctypes.c_uint64), ctypes.c_uint32(n_series),
+        gptr, ctypes.c_uint32(n_groups), ctypes.c_int32(aggr_id),
+        _ptr(out, ctypes.c_double), cptr, ctypes.byref(scanned),
+        ctypes.c_int(n_threads))
+    if rcode != 0:
+        raise RuntimeError(f"vm_rollup_eval_batch failed: {rcode}")
+    return out, counts, scanned.value
