@@ -1,0 +1,135 @@
+/* vmgpu.h — C-ABI of the MI355X-native vmselect rollup/aggregation engine.
+ *
+ * This is the drop-in boundary described in SURVEY.md §8b: it replaces the
+ * worker fan-out seam inside the reference's evalRollupFuncNoCache
+ * (app/vmselect/promql/eval.go:1899-1904 → evalRollupWithIncrementalAggregate
+ * eval.go:1927 / evalRollupNoIncrementalAggregate eval.go:1968).  Everything
+ * above the seam (parse, plan, SearchQuery, rollupResultCache probe, memory
+ * limiter) and below it (TSDB block scan) stays in the host process; the host
+ * hands decoded columnar series (CSR (timestamps[], values[]) batches, the
+ * callback input of netstorage.Results.RunParallel, netstorage.go:219) to
+ * this library and receives the rollup grid / aggregated group matrix back.
+ *
+ * The intended host binding is Go cgo (see INTEGRATION.md for the stub a
+ * vmselect maintainer would add); the in-repo host mirror is Python ctypes
+ * (victoriametrics_amd/engine.py) because this image carries no Go toolchain.
+ *
+ * Plain C types only: no HIP or torch types cross this boundary.
+ */
+#ifndef VMGPU_H
+#define VMGPU_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* Rollup function ids — numerically identical to the oracle's vm_func_id
+ * (oracle/vm_oracle.h) and named after the reference's rollupFuncs map keys
+ * (app/vmselect/promql/rollup.go:24-108). */
+typedef int32_t vmgpu_func_id;
+
+/* Cross-series incremental aggregate op ids — same numbering as the oracle's
+ * vm_aggr_op (aggr_incremental.go:18-66 callback table). */
+enum {
+  VMGPU_AGGR_NONE = 0,
+  VMGPU_AGGR_SUM = 1,
+  VMGPU_AGGR_MIN = 2,
+  VMGPU_AGGR_MAX = 3,
+  VMGPU_AGGR_AVG = 4,
+  VMGPU_AGGR_COUNT = 5,
+  VMGPU_AGGR_SUM2 = 6,
+  VMGPU_AGGR_GEOMEAN = 7,
+  VMGPU_AGGR_GROUP = 8,
+};
+
+/* Evaluation plan — the fields of promql.rollupConfig (rollup.go:574-606)
+ * that shape the computation, plus the preFunc toggles that getRollupConfigs
+ * (rollup.go:374-516) would install and the aggregate op. */
+typedef struct vmgpu_plan {
+  int32_t func;                   /* vmgpu_func_id */
+  int32_t aggr;                   /* VMGPU_AGGR_* (NONE => per-series output) */
+  int64_t start, end, step;       /* grid [start:end:step], ms (eval.go:234) */
+  int64_t window;                 /* lookbehind window ms; 0 => auto-adjust */
+  int64_t lookback_delta;         /* rc.LookbackDelta */
+  int64_t min_staleness_interval; /* -search.minStalenessInterval */
+  int64_t max_staleness_interval; /* staleness arg of removeCounterResets
+                                   * (= lookback_delta + window when
+                                   *  lookback_delta != 0; rollup.go:380-387) */
+  int32_t may_adjust_window;      /* rollupFuncsCanAdjustWindow[func] */
+  int32_t is_default_rollup;
+  int32_t remove_counter_resets;  /* rollupFuncsRemoveCounterResets[func] */
+  int32_t drop_stale_nans;        /* dropStaleNaNs unless -search.noStaleMarkers
+                                   * or func==default_rollup (eval.go:2108) */
+  int32_t samples_scanned_per_call; /* rollupFuncsSamplesScannedPerCall */
+  int32_t skip_finalize;          /* leave (values,counts) un-finalized so the
+                                   * caller can all-reduce across shards first
+                                   *  */
+  double  arg;                    /* phi / le / gt / eq / secs for arg funcs */
+} vmgpu_plan;
+
+/* Select the device and create the library context.  One process drives one
+ * GPU (one process per GPU over RCCL is the scaling model); n_devices must
+ * currently be 1.  Returns 0 on success. */
+int vmgpu_init(const int* device_ids, int n_devices);
+int vmgpu_shutdown(void);
+
+/* Upload a decoded-series batch (CSR layout: series s occupies
+ * [offsets[s], offsets[s+1]) in ts/vals; offsets has n_series+1 entries;
+ * group_ids may be NULL) to the device.  Returns a handle via *out_handle. */
+int vmgpu_batch_create(const int64_t* ts, const double* vals,
+                       const uint64_t* offsets, uint32_t n_series,
+                       const int32_t* group_ids, uint32_t n_groups,
+                       uint64_t* out_handle,
+                       char* errbuf, size_t errbuf_len);
+int vmgpu_batch_destroy(uint64_t handle);
+
+/* Evaluate the plan over an uploaded batch.
+ * aggr==NONE: out is [n_series x n_grid] (row-major, n_grid =
+ *   1 + (end-start)/step, eval.go:247); out_counts unused.
+ * aggr!=NONE: out and out_counts are [n_groups x n_grid]; with
+ *   plan->skip_finalize they hold the raw partial matrices (identity-filled
+ *   where empty) ready for a cross-shard all-reduce.
+ * out/out_counts may be NULL to leave results on the device (fetch with
+ * vmgpu_batch_fetch_out) — that is how benches time the kernel without PCIe.
+ * Returns 0 on success; nonzero + errbuf otherwise. */
+int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
+                      double* out, double* out_counts,
+                      uint64_t* out_samples_scanned,
+                      char* errbuf, size_t errbuf_len);
+
+/* Download the last exec's outputs for this batch (sizes in elements). */
+int vmgpu_batch_fetch_out(uint64_t handle, double* dst, size_t n_elems,
+                          double* dst_counts, size_t n_count_elems);
+
+/* Apply the aggregate finalize step (finalizeAggrCommon/Avg/Count/Group/
+ * Geomean, aggr_incremental.go:141-168 + 189-512) to host matrices — used
+ * after the cross-shard all-reduce when exec ran with skip_finalize. */
+void vmgpu_aggr_finalize_host(int32_t aggr, double* values, double* counts,
+                              uint64_t n_elems);
+
+/* One-shot convenience wrapper (create + exec + destroy), the literal shape
+ * a cgo shim binds. */
+int vmgpu_rollup_eval(const vmgpu_plan* plan,
+                      const int64_t* ts, const double* vals,
+                      const uint64_t* offsets, uint32_t n_series,
+                      const int32_t* group_ids, uint32_t n_groups,
+                      double* out, double* out_counts,
+                      uint64_t* out_samples_scanned,
+                      char* errbuf, size_t errbuf_len);
+
+/* Wall time of the rollup kernels inside the last vmgpu_rollup_exec on this
+ * thread's context, measured with hipEvents on the launch stream (for the
+ * bench's roofline accounting). */
+int vmgpu_last_kernel_ms(double* out_ms);
+
+/* Device properties the bench needs for roofline context. */
+int vmgpu_device_info(char* name, size_t name_len, double* hbm_gib,
+                      int* cu_count);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* VMGPU_H */

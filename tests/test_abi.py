@@ -1,0 +1,113 @@
+"""CPU-side checks of the product C-ABI library: it builds, loads, exports
+every symbol include/vmgpu.h declares, and fails LOUDLY without a GPU
+(no silent CPU fallback).  No compute calls here (no GPU in CI)."""
+import ctypes
+import os
+import re
+import subprocess
+
+import numpy as np
+import pytest
+
+from conftest import REPO_ROOT
+
+LIB = os.path.join(REPO_ROOT, "victoriametrics_amd", "libvmgpu.so")
+HEADER = os.path.join(REPO_ROOT, "include", "vmgpu.h")
+
+
+def _ensure_built():
+    if not os.path.exists(LIB):
+        subprocess.run(
+            ["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
+             "-shared", "vmgpu.hip", "-o", "../libvmgpu.so"],
+            cwd=os.path.join(REPO_ROOT, "victoriametrics_amd", "csrc"),
+            check=True)
+    return LIB
+
+
+def test_lib_exports_all_header_symbols():
+    lib = ctypes.CDLL(_ensure_built())
+    header = open(HEADER).read()
+    declared = re.findall(r"^(?:int|void)\s+(vmgpu_\w+)\(", header, re.M)
+    assert len(declared) >= 9, f"header should declare the ABI; found {declared}"
+    for sym in declared:
+        assert hasattr(lib, sym), f"libvmgpu.so missing exported symbol {sym}"
+
+
+def test_host_finalize_matches_oracle():
+    """vmgpu_aggr_finalize_host is host-side product code (the post-allreduce
+    finalize) — check against the oracle finalize on CPU."""
+    import oracle
+    _ensure_built()
+    lib = ctypes.CDLL(LIB)
+    rng = np.random.default_rng(7)
+    for op_name, op_id in [("sum", 1), ("min", 2), ("max", 3), ("avg", 4),
+                           ("count", 5), ("sum2", 6), ("geomean", 7),
+                           ("group", 8)]:
+        v = rng.random(64) * 10 + 0.1
+        c = (rng.random(64) * 3).astype(np.int64).astype(np.float64)
+        v_ref = v.copy()
+        c_ref = c.copy()
+        if op_name in ("count", "group"):
+            # count/group track everything in the values row
+            v[c == 0] = 0.0
+            v_ref = v.copy()
+        oracle.lib().vm_aggr_finalize(
+            ctypes.c_int(op_id),
+            v_ref.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+            c_ref.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+            ctypes.c_int64(64))
+        lib.vmgpu_aggr_finalize_host(
+            ctypes.c_int32(op_id),
+            v.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+            c.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+            ctypes.c_uint64(64))
+        assert np.array_equal(v, v_ref, equal_nan=True), op_name
+
+
+def test_engine_fails_loudly_without_gpu():
+    import torch
+    from victoriametrics_amd import engine
+    _ensure_built()
+    if torch.cuda.is_available():
+        pytest.skip("GPU present; loud-failure path not applicable")
+    with pytest.raises(engine.VmGpuError):
+        engine.init()
+
+
+def test_plan_validation():
+    from victoriametrics_amd import engine
+    with pytest.raises(engine.VmGpuError):
+        engine.RollupPlan("not_a_func", 0, 100, 10)
+    with pytest.raises(engine.VmGpuError):
+        engine.RollupPlan("rate", 0, 100, 0)
+    with pytest.raises(engine.VmGpuError):
+        engine.RollupPlan("rate", 100, 0, 10)
+    p = engine.RollupPlan("rate", 0, 90, 10, window=300)
+    assert p.n_grid == 10
+    assert list(p.timestamps()) == list(range(0, 100, 10))
+    # preFunc wiring (getRollupConfigs, rollup.go:374-516)
+    assert p._c.remove_counter_resets == 1
+    assert p._c.may_adjust_window == 1
+    assert p._c.samples_scanned_per_call == 2
+    assert p._c.drop_stale_nans == 1
+    assert p._c.max_staleness_interval == 0  # lookback_delta == 0
+    p2 = engine.RollupPlan("rate", 0, 90, 10, window=300, lookback_delta=50)
+    assert p2._c.max_staleness_interval == 350  # lookback + window
+    p3 = engine.RollupPlan("default_rollup", 0, 90, 10)
+    assert p3._c.drop_stale_nans == 0
+    assert p3._c.is_default_rollup == 1
+    p4 = engine.RollupPlan("sum_over_time", 0, 90, 10)
+    assert p4._c.remove_counter_resets == 0
+    assert p4._c.may_adjust_window == 0
+    assert p4._c.samples_scanned_per_call == 0
+
+
+def test_func_id_tables_match_oracle():
+    import oracle
+    from victoriametrics_amd import engine
+    assert engine.FUNC_IDS == oracle.FUNC_IDS
+    assert engine.AGGR_IDS == oracle.AGGR_IDS
+    assert engine.REMOVE_COUNTER_RESETS_FUNCS == oracle.REMOVE_COUNTER_RESETS_FUNCS
+    assert engine.SAMPLES_SCANNED_PER_CALL == oracle.SAMPLES_SCANNED_PER_CALL
+    assert engine.CAN_ADJUST_WINDOW_FUNCS == oracle.CAN_ADJUST_WINDOW_FUNCS

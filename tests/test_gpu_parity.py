@@ -1,0 +1,260 @@
+"""GPU parity tests: the HIP engine against the CPU oracle on identical
+seeded inputs.  The oracle is pinned against the reference's own golden
+vectors in test_oracle_golden.py, so green here means parity with the
+reference Go path.
+
+Bar: bit-exact for arithmetic-only functions (each grid point
+is one lane running the same left-to-right loops), rtol<=1e-12 for functions
+with non-correctly-rounded transcendentals (pow), rtol<=1e-9 for cross-series
+float sums whose atomic ordering differs.
+"""
+import math
+
+import numpy as np
+import pytest
+
+import oracle
+from seriesgen import ragged_batch, assert_parity
+
+pytestmark = pytest.mark.gpu
+
+START = 1_000_000_000_000
+STEP = 15_000
+
+# funcs whose device implementation uses pow (not correctly rounded): compare
+# with rtol instead of bitwise.
+POW_FUNCS = {"geomean_over_time"}
+
+ALL_FUNCS = sorted(set(oracle.FUNC_IDS) - {"increase_prometheus", "timestamp",
+                                           "timestamp_with_name"})
+ARG_FUNCS = {
+    "quantile_over_time": 0.9, "count_le_over_time": 5.0,
+    "count_gt_over_time": 5.0, "count_eq_over_time": 1.0,
+    "count_ne_over_time": 1.0, "share_le_over_time": 5.0,
+    "share_gt_over_time": 5.0, "share_eq_over_time": 1.0,
+    "sum_le_over_time": 5.0, "sum_gt_over_time": 5.0,
+    "sum_eq_over_time": 1.0, "predict_linear": 120.0,
+    "duration_over_time": 60.0,
+}
+
+
+@pytest.fixture(scope="module")
+def engine():
+    from victoriametrics_amd import engine as e
+    e.init()
+    return e
+
+
+@pytest.fixture(scope="module")
+def counter_small():
+    from victoriametrics_amd import synth
+    return synth.counter_batch(2048, 240, START, seed=8428)
+
+
+def _oracle_batch(engine_plan, ts, vals, offsets, group_ids=None, n_groups=0,
+                  aggr="none", n_threads=4):
+    rc = oracle.RollupConfigC(
+        func=engine_plan._c.func,
+        may_adjust_window=engine_plan._c.may_adjust_window,
+        start=engine_plan._c.start, end=engine_plan._c.end,
+        step=engine_plan._c.step, window=engine_plan._c.window,
+        lookback_delta=engine_plan._c.lookback_delta,
+        min_staleness_interval=engine_plan._c.min_staleness_interval,
+        is_default_rollup=engine_plan._c.is_default_rollup,
+        samples_scanned_per_call=engine_plan._c.samples_scanned_per_call,
+        arg=engine_plan._c.arg)
+    return oracle.rollup_eval_batch(
+        rc, ts, vals, offsets, group_ids=group_ids, n_groups=n_groups,
+        aggr=aggr,
+        remove_counter_resets=bool(engine_plan._c.remove_counter_resets),
+        max_staleness_interval=engine_plan._c.max_staleness_interval,
+        drop_stale_nans=bool(engine_plan._c.drop_stale_nans),
+        n_threads=n_threads)
+
+
+def test_rate_counter_batch_bitexact(engine, counter_small):
+    ts, vals, offsets = counter_small
+    end = START + 239 * STEP
+    plan = engine.RollupPlan("rate", START, end, STEP, window=300_000)
+    out, _, scanned = engine.rollup_eval(plan, ts, vals, offsets)
+    ref, _, ref_scanned = _oracle_batch(plan, ts, vals, offsets)
+    assert scanned == ref_scanned
+    assert_parity(out, ref, exact=True, context="rate")
+
+
+@pytest.mark.parametrize("func", ALL_FUNCS)
+def test_all_funcs_ragged(engine, func):
+    """Every rollup function over a ragged batch with empty series, duplicate
+    timestamps and varying lengths, window explicitly set."""
+    ts, vals, offsets = ragged_batch(300, 260, START, seed=99, dup_p=0.02)
+    end = START + 100 * STEP
+    plan = engine.RollupPlan(func, START, end, STEP, window=200_000,
+                             arg=ARG_FUNCS.get(func, 0.0))
+    out, _, scanned = engine.rollup_eval(plan, ts, vals, offsets)
+    ref, _, ref_scanned = _oracle_batch(plan, ts, vals, offsets)
+    assert scanned == ref_scanned, f"{func}: scanned {scanned} != {ref_scanned}"
+    assert_parity(out, ref, exact=func not in POW_FUNCS, context=func)
+
+
+@pytest.mark.parametrize("func", ["rate", "delta", "avg_over_time",
+                                  "default_rollup", "deriv_fast"])
+def test_window_autoadjust(engine, func):
+    """window=0 paths: per-series scrape-interval estimate + window adjust
+    (rollup.go:719-756)."""
+    ts, vals, offsets = ragged_batch(200, 120, START, seed=7)
+    end = START + 50 * STEP
+    for lbd in (0, 40_000):
+        plan = engine.RollupPlan(func, START, end, STEP, window=0,
+                                 lookback_delta=lbd)
+        out, _, scanned = engine.rollup_eval(plan, ts, vals, offsets)
+        ref, _, ref_scanned = _oracle_batch(plan, ts, vals, offsets)
+        assert scanned == ref_scanned, f"{func} lbd={lbd}"
+        assert_parity(out, ref, exact=True, context=f"{func} lbd={lbd}")
+
+
+def test_instant_query(engine):
+    """start == end (instant query): maxPrevInterval = step directly."""
+    ts, vals, offsets = ragged_batch(100, 100, START, seed=3)
+    plan = engine.RollupPlan("rate", START, START, STEP, window=300_000)
+    out, _, scanned = engine.rollup_eval(plan, ts, vals, offsets)
+    ref, _, ref_scanned = _oracle_batch(plan, ts, vals, offsets)
+    assert scanned == ref_scanned
+    assert_parity(out, ref, exact=True, context="instant rate")
+
+
+def test_stale_nans_dropped(engine):
+    ts, vals, offsets = ragged_batch(150, 150, START, seed=11, stale_p=0.05)
+    end = START + 60 * STEP
+    for func in ("rate", "avg_over_time", "default_rollup",
+                 "stale_samples_over_time"):
+        plan = engine.RollupPlan(func, START, end, STEP, window=120_000)
+        out, _, scanned = engine.rollup_eval(plan, ts, vals, offsets)
+        ref, _, ref_scanned = _oracle_batch(plan, ts, vals, offsets)
+        assert scanned == ref_scanned, func
+        assert_parity(out, ref, exact=True, context=f"stale {func}")
+
+
+def test_staleness_gap_resets(engine):
+    """removeCounterResets with max_staleness_interval (issue 8072 path):
+    engine sets it when lookback_delta != 0."""
+    ts, vals, offsets = ragged_batch(150, 150, START, seed=13)
+    end = START + 60 * STEP
+    plan = engine.RollupPlan("increase", START, end, STEP, window=120_000,
+                             lookback_delta=20_000)
+    out, _, scanned = engine.rollup_eval(plan, ts, vals, offsets)
+    ref, _, ref_scanned = _oracle_batch(plan, ts, vals, offsets)
+    assert scanned == ref_scanned
+    assert_parity(out, ref, exact=True, context="staleness gap")
+
+
+@pytest.mark.parametrize("max_len,name", [(520, "wave_edge"),
+                                          (3000, "block"),
+                                          (4100, "block_edge"),
+                                          (9000, "huge")])
+def test_long_series_kernels(engine, max_len, name):
+    """Exercises the block (512 < n <= 4064) and huge (n > 4064) kernels,
+    including series lengths straddling each boundary."""
+    ts, vals, offsets = ragged_batch(48, max_len, START, seed=max_len)
+    end = START + 100 * STEP
+    plan = engine.RollupPlan("rate", START, end, STEP, window=400_000)
+    out, _, scanned = engine.rollup_eval(plan, ts, vals, offsets)
+    ref, _, ref_scanned = _oracle_batch(plan, ts, vals, offsets)
+    assert scanned == ref_scanned, name
+    assert_parity(out, ref, exact=True, context=name)
+    plan2 = engine.RollupPlan("avg_over_time", START, end, STEP, window=400_000)
+    out2, _, _ = engine.rollup_eval(plan2, ts, vals, offsets)
+    ref2, _, _ = _oracle_batch(plan2, ts, vals, offsets)
+    assert_parity(out2, ref2, exact=True, context=name + "_avg")
+
+
+@pytest.mark.parametrize("aggr", ["sum", "min", "max", "avg", "count",
+                                  "sum2", "group", "geomean"])
+def test_grouped_aggregation(engine, counter_small, aggr):
+    """sum/min/max/avg/... by(pod) — grouped matrices vs the oracle.
+    min/max/count/group bit-exact; sums/avg at 1e-9 (atomic order)."""
+    ts, vals, offsets = counter_small
+    n_series = len(offsets) - 1
+    n_groups = 37
+    rng = np.random.default_rng(5)
+    gids = rng.integers(0, n_groups, n_series).astype(np.int32)
+    end = START + 239 * STEP
+    plan = engine.RollupPlan("rate", START, end, STEP, window=300_000,
+                             aggr=aggr)
+    out, counts, scanned = engine.rollup_eval(plan, ts, vals, offsets,
+                                              group_ids=gids,
+                                              n_groups=n_groups)
+    ref, ref_counts, ref_scanned = _oracle_batch(plan, ts, vals, offsets,
+                                                 group_ids=gids,
+                                                 n_groups=n_groups, aggr=aggr)
+    assert scanned == ref_scanned
+    exact = aggr in ("min", "max", "count", "group")
+    assert_parity(out, ref, exact=exact, rtol=1e-9, context=f"aggr {aggr}")
+
+
+def test_grouped_skip_finalize_allreduce_cut(engine, counter_small):
+    """skip_finalize exposes the pre-all-reduce matrices; finalizing on the
+    host must match the fused finalize."""
+    ts, vals, offsets = counter_small
+    n_series = len(offsets) - 1
+    n_groups = 16
+    gids = (np.arange(n_series) % n_groups).astype(np.int32)
+    end = START + 239 * STEP
+    plan_raw = engine.RollupPlan("rate", START, end, STEP, window=300_000,
+                                 aggr="avg", skip_finalize=True)
+    raw, counts, _ = engine.rollup_eval(plan_raw, ts, vals, offsets,
+                                        group_ids=gids, n_groups=n_groups)
+    host_fin = engine.aggr_finalize("avg", raw.copy(), counts)
+    plan_fin = engine.RollupPlan("rate", START, end, STEP, window=300_000,
+                                 aggr="avg")
+    fin, _, _ = engine.rollup_eval(plan_fin, ts, vals, offsets,
+                                   group_ids=gids, n_groups=n_groups)
+    assert_parity(host_fin.reshape(fin.shape), fin, exact=True,
+                  context="skip_finalize")
+
+
+def test_negative_group_id_skipped(engine):
+    """group_id == -1 means 'not grouped' (series skipped), mirroring the
+    aggregate limit skip (aggr_incremental.go:117-121)."""
+    ts, vals, offsets = ragged_batch(64, 100, START, seed=21)
+    n_series = len(offsets) - 1
+    gids = np.full(n_series, -1, dtype=np.int32)
+    gids[::2] = 0
+    end = START + 30 * STEP
+    plan = engine.RollupPlan("rate", START, end, STEP, window=120_000,
+                             aggr="sum")
+    out, _, _ = engine.rollup_eval(plan, ts, vals, offsets, group_ids=gids,
+                                   n_groups=1)
+    ref, _, _ = _oracle_batch(plan, ts, vals, offsets, group_ids=gids,
+                              n_groups=1, aggr="sum")
+    assert_parity(out, ref, exact=False, rtol=1e-9, context="gid -1")
+
+
+def test_full_config2_checksums(engine):
+    """Full BASELINE config-2 size (1M x 240) — size-independent properties
+    plus full comparison against the multithreaded oracle (count/min/max of
+    the rate grid; bit-exact everywhere since rollup is per-series)."""
+    from victoriametrics_amd import synth
+    n_series, n_samples = 1_000_000, 240
+    ts, vals, offsets = synth.counter_batch(n_series, n_samples, START)
+    end = START + 239 * STEP
+    plan = engine.RollupPlan("rate", START, end, STEP, window=300_000)
+    with engine.SeriesBatch(ts, vals, offsets) as b:
+        out, _, scanned = b.exec(plan)
+    # oracle on a deterministic 1% subsample of series, bit-exact
+    sel = np.arange(0, n_series, 97)
+    sub_off = [0]
+    sub_ts, sub_vals = [], []
+    for s in sel:
+        lo, hi = int(offsets[s]), int(offsets[s + 1])
+        sub_ts.append(ts[lo:hi])
+        sub_vals.append(vals[lo:hi])
+        sub_off.append(sub_off[-1] + hi - lo)
+    ref, _, _ = _oracle_batch(
+        plan, np.concatenate(sub_ts), np.concatenate(sub_vals),
+        np.asarray(sub_off, dtype=np.uint64))
+    assert_parity(out[sel], ref, exact=True, context="config2 subsample")
+    # whole-grid properties
+    assert scanned == n_series * n_samples + n_series * 2 * plan.n_grid
+    nan_frac = np.isnan(out).mean()
+    assert nan_frac < 0.02, f"unexpected NaN fraction {nan_frac}"
+    assert np.nanmin(out) >= 0.0  # rates of counters are non-negative

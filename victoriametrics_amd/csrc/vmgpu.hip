@@ -1,0 +1,989 @@
+/* vmgpu.hip — MI355X-native (gfx950/CDNA4) rollup + aggregation engine.
+ *
+ * Replaces the reference's per-series worker fan-out
+ * (netstorage.Results.RunParallel, netstorage.go:219 +
+ * evalRollup{With,No}IncrementalAggregate, eval.go:1927/1968) with three HIP
+ * kernels picked by series length, all sharing the device rollup set in
+ * rollup_device.h:
+ *
+ *   rollup_wave_kernel  — series with <= CHUNK_WAVE samples: one 64-lane
+ *       wavefront per series.  The wave stages the series' (ts, vals)
+ *       columns into LDS with coalesced loads, optionally drops Prometheus
+ *       stale NaNs (eval.go:2108) and applies removeCounterResets
+ *       (rollup.go:921) as an EXACT-order fused scan (sparse sequential
+ *       correction walk over ballot'd reset/gap events + segmented prefix
+ *       max for the monotonic clamp), then evaluates the [start:end:step]
+ *       grid with one lane per grid point (binary-search window seek in LDS,
+ *       left-to-right serial window reductions => bit-exact vs the
+ *       sequential reference semantics).
+ *   rollup_block_kernel — series up to CHUNK_BLOCK samples: one 256-thread
+ *       workgroup per series, series staged in up to 128 KiB of LDS.
+ *   rollup_huge_kernel  — longer series: per-series preprocessing into a
+ *       global scratch column, grid evaluated from L2-cached global memory.
+ *
+ * Cross-series by-label aggregation (aggr_incremental.go) is fused into the
+ * grid loop: identity-initialized [n_groups x n_grid] value/count matrices
+ * updated with device-scope f64 atomics (CAS loops for min/max/product), so
+ * the 1M-series rollup never materializes per-series grids in HBM.  The
+ * multi-GPU merge all-reduces those matrices BEFORE the
+ * finalize step; plan->skip_finalize exposes exactly that cut.
+ */
+#include <hip/hip_runtime.h>
+#include <algorithm>
+#include <atomic>
+#include <cstring>
+#include <cstdio>
+#include <map>
+#include <mutex>
+#include <vector>
+
+#include "rollup_device.h"
+#include "../../include/vmgpu.h"
+
+#define WAVE 64
+#define BLOCK_THREADS 256
+#define WAVES_PER_BLOCK (BLOCK_THREADS / WAVE)
+#define CHUNK_WAVE 512
+/* block-kernel series cap: 16 B/sample of LDS + 16 B control word must stay
+ * within the 64 KiB dynamic-LDS launch limit (the 160 KiB/CU budget is a
+ * static-declaration property on gfx950) */
+#define CHUNK_BLOCK 4064
+#define MAX_WAVE_BLOCKS 4096
+
+/* ------------------------------------------------------------------ */
+/* device helpers                                                     */
+/* ------------------------------------------------------------------ */
+
+static VM_DEV void wave_lds_sync() {
+  /* All DS and VMEM ops of this wave complete + compiler barrier: safe
+   * cross-lane visibility within one wavefront for LDS *and* for the huge
+   * kernel's global-scratch variant (no workgroup barrier needed).  Used
+   * between phases only, never inside hot loops. */
+  asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+}
+
+static VM_DEV int vm_upper_bound(const int64_t* ts, int n, int64_t seek) {
+  /* first index with ts[idx] > seek (seekFirstTimestampIdxAfter semantics,
+   * rollup.go:825-855 — the reference's ±2 hint only narrows the range). */
+  int i = 0, j = n;
+  while (i < j) {
+    int h = (i + j) >> 1;
+    if (ts[h] <= seek) i = h + 1;
+    else j = h;
+  }
+  return i;
+}
+
+static VM_DEV void vm_atomic_min_f64(double* addr, double val) {
+  unsigned long long* p = (unsigned long long*)addr;
+  unsigned long long old = __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  while (val < __longlong_as_double((long long)old)) {
+    unsigned long long assumed = old;
+    old = atomicCAS(p, assumed, (unsigned long long)__double_as_longlong(val));
+    if (old == assumed) break;
+  }
+}
+
+static VM_DEV void vm_atomic_max_f64(double* addr, double val) {
+  unsigned long long* p = (unsigned long long*)addr;
+  unsigned long long old = __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  while (val > __longlong_as_double((long long)old)) {
+    unsigned long long assumed = old;
+    old = atomicCAS(p, assumed, (unsigned long long)__double_as_longlong(val));
+    if (old == assumed) break;
+  }
+}
+
+static VM_DEV void vm_atomic_mul_f64(double* addr, double val) {
+  unsigned long long* p = (unsigned long long*)addr;
+  unsigned long long old = __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  while (true) {
+    unsigned long long assumed = old;
+    double cur = __longlong_as_double((long long)assumed);
+    old = atomicCAS(p, assumed, (unsigned long long)__double_as_longlong(cur * val));
+    if (old == assumed) break;
+  }
+}
+
+/* Kernel-side plan (host plan flattened; all wave-uniform). */
+struct KPlan {
+  int64_t start, end, step;
+  int64_t window;          /* plan window (0 => auto-adjust per series) */
+  int64_t lookback_delta;
+  int64_t min_staleness;
+  int64_t max_staleness;   /* removeCounterResets staleness interval */
+  int32_t n_grid;
+  int32_t func;
+  int32_t aggr;
+  int32_t sspc;            /* samplesScannedPerCall */
+  int32_t may_adjust;
+  int32_t is_default;
+  int32_t rcr;
+  int32_t drop_stale;
+  double arg;
+};
+
+struct KIO {
+  const int64_t* ts = nullptr;
+  const double* vals = nullptr;
+  const uint64_t* offsets = nullptr;
+  const uint32_t* series_sel = nullptr; /* list of series ids for this kernel */
+  uint32_t n_sel = 0;
+  const int32_t* group_ids = nullptr;   /* may be null */
+  double* out = nullptr;                /* per-series [n_series x n_grid] or group values */
+  double* out_counts = nullptr;         /* group counts */
+  unsigned long long* samples_scanned = nullptr;
+  /* huge-kernel scratch */
+  int64_t* scr_ts = nullptr;
+  double* scr_vals = nullptr;
+  const uint64_t* scr_offsets = nullptr; /* per huge-series scratch base */
+};
+
+/* --- phase A: compacted load (executed by ONE wave) ----------------- */
+/* Copies [g_ts,g_vals)[0..n) to dst, dropping stale NaNs if requested.
+ * Returns new count. dst may be LDS or global. */
+static __device__ int load_compact_wave(const int64_t* g_ts, const double* g_vals,
+                                        int64_t n, int64_t* d_ts, double* d_vals,
+                                        bool drop_stale, int lane) {
+  int count = 0;
+  for (int64_t src = 0; src < n; src += WAVE) {
+    int64_t k = src + lane;
+    bool active = k < n;
+    double v = 0.0;
+    int64_t t = 0;
+    if (active) {
+      v = g_vals[k];
+      t = g_ts[k];
+    }
+    bool keep = active && !(drop_stale && vm_is_stale_nan(v));
+    uint64_t m = __ballot(keep);
+    if (keep) {
+      int dst = count + __popcll(m & ((lane == 63) ? 0x7fffffffffffffffULL
+                                                   : ((1ULL << lane) - 1)));
+      d_ts[dst] = t;
+      d_vals[dst] = v;
+    }
+    count += __popcll(m);
+  }
+  return count;
+}
+
+/* --- phase B: removeCounterResets over the dense column -------------- */
+/* EXACT restatement of rollup.go:921-958 as a wave scan; see file header. */
+static __device__ void rcr_scan_wave(int64_t* d_ts, double* d_vals, int count,
+                                     int64_t msi, int lane) {
+  double corr = 0.0;       /* running correction */
+  double prev_raw = 0.0;   /* raw value of previous element */
+  int64_t prev_ts = 0;
+  double prev_fin = 0.0;   /* clamped output of previous element */
+  for (int base = 0; base < count; base += WAVE) {
+    int k = base + lane;
+    bool active = k < count;
+    double v = active ? d_vals[k] : 0.0;
+    int64_t t = active ? d_ts[k] : 0;
+    wave_lds_sync();
+    double pv = __shfl_up(v, 1);
+    int64_t pt = __shfl_up(t, 1);
+    if (lane == 0) { pv = prev_raw; pt = prev_ts; }
+    bool isfirst = (k == 0);
+    double d = v - pv;
+    double inc = 0.0;
+    if (!isfirst && d < 0) inc = ((-d * 8) < pv) ? (pv - v) : pv;
+    bool gap = (!isfirst && msi > 0 && (t - pt) > msi);
+    /* corrections in exact sequential order: only reset/gap events change the
+     * running correction, and f64 `c + 0.0` is an identity, so walking the
+     * (rare) event lanes reproduces the serial left-to-right sum bitwise. */
+    uint64_t em = __ballot(active && (gap || inc != 0.0));
+    double c = corr;
+    double mycorr = corr;
+    while (em) {
+      int b = __ffsll((unsigned long long)em) - 1;
+      em &= em - 1;
+      double ib = __shfl(inc, b);
+      int gb = __shfl((int)gap, b);
+      double cn = gb ? 0.0 : (c + ib);
+      if (lane >= b) mycorr = cn;
+      c = cn;
+    }
+    double fin = v + mycorr;
+    /* monotonic clamp (rollup.go:952-956) = segmented prefix max; a gap
+     * element keeps its raw value and is exempt (the Go `continue`). */
+    bool bnd = isfirst || gap;
+    double x = fin;
+    int f = bnd ? 1 : 0;
+    if (lane == 0 && !bnd) x = fmax(x, prev_fin);
+    for (int dlt = 1; dlt < WAVE; dlt <<= 1) {
+      double xo = __shfl_up(x, dlt);
+      int fo = __shfl_up(f, dlt);
+      if (lane >= dlt) {
+        if (!f) x = fmax(x, xo);
+        f = f | fo;
+      }
+    }
+    if (active) d_vals[k] = x;
+    int last = count - base - 1;
+    if (last > 63) last = 63;
+    corr = c;
+    prev_raw = __shfl(v, last);
+    prev_ts = __shfl(t, last);
+    prev_fin = __shfl(x, last);
+    wave_lds_sync();
+  }
+}
+
+/* --- per-series window parameters (doInternal preamble) -------------- */
+/* getScrapeInterval (rollup.go:871-897): 0.6 quantile of the last <=20
+ * sample gaps, computed wave-cooperatively with a rank-based selection
+ * (identical result to sort + quantileSorted). */
+static __device__ int64_t scrape_interval_wave(const int64_t* d_ts, int count,
+                                               int64_t default_interval, int lane) {
+  if (count < 2) return default_interval;
+  int cnt = count - 1;
+  if (cnt > 20) cnt = 20;
+  bool active = lane < cnt;
+  double gap = 0.0;
+  if (active) gap = (double)(d_ts[count - 1 - lane] - d_ts[count - 2 - lane]);
+  wave_lds_sync();
+  /* rank of this lane's gap among the cnt gaps (ties broken by lane) */
+  int rank = 0;
+  for (int i = 0; i < cnt; i++) {
+    double gi = __shfl(gap, i);
+    if (active && (gi < gap || (gi == gap && i < lane))) rank++;
+  }
+  double nn = (double)cnt;
+  double q_rank = 0.6 * (nn - 1);
+  int li = (int)fmax(0.0, floor(q_rank));
+  int ui = (int)fmin(nn - 1, (double)(li + 1));
+  double w = q_rank - floor(q_rank);
+  uint64_t mlo = __ballot(active && rank == li);
+  uint64_t mhi = __ballot(active && rank == ui);
+  int lane_lo = __ffsll((unsigned long long)mlo) - 1;
+  int lane_hi = __ffsll((unsigned long long)mhi) - 1;
+  double lo = __shfl(gap, lane_lo);
+  double hi = __shfl(gap, lane_hi);
+  double q = lo * (1 - w) + hi * w;
+  int64_t si = (int64_t)q;
+  if (si <= 0) return default_interval;
+  return si;
+}
+
+static VM_DEV int64_t max_prev_interval_tiers(int64_t si) {
+  /* getMaxPrevInterval (rollup.go:899-919) */
+  if (si <= 2000) return si + 4 * si;
+  if (si <= 4000) return si + 2 * si;
+  if (si <= 8000) return si + si;
+  if (si <= 16000) return si + si / 2;
+  if (si <= 32000) return si + si / 4;
+  return si + si / 8;
+}
+
+struct SeriesWindow {
+  int64_t window;
+  int64_t max_prev_interval;
+};
+
+static VM_DEV SeriesWindow series_window(const KPlan& p, int64_t scrape_interval_est) {
+  /* doInternal window setup (rollup.go:719-756) */
+  SeriesWindow sw;
+  int64_t mpi = p.step;
+  if (p.start < p.end) mpi = max_prev_interval_tiers(scrape_interval_est);
+  if (p.lookback_delta > 0 && mpi > p.lookback_delta) mpi = p.lookback_delta;
+  if (p.min_staleness > 0 && mpi < p.min_staleness) mpi = p.min_staleness;
+  int64_t window = p.window;
+  if (window <= 0) {
+    window = p.step;
+    if (p.may_adjust && window < mpi) window = mpi;
+    if (p.is_default && p.lookback_delta > 0 && window > p.lookback_delta)
+      window = p.lookback_delta;
+  }
+  sw.window = window;
+  sw.max_prev_interval = mpi;
+  return sw;
+}
+
+/* --- grid-point evaluation (one lane, one grid point) ----------------- */
+static VM_DEV uint64_t eval_grid_point(const KPlan& p, const SeriesWindow& sw,
+                                       const int64_t* ts, const double* vals,
+                                       int count, int g, uint32_t s,
+                                       const KIO& io) {
+  int64_t t_end = p.start + (int64_t)g * p.step;
+  int64_t t_start = t_end - sw.window;
+  int i = vm_upper_bound(ts, count, t_start);
+  int j = vm_upper_bound(ts, count, t_end);
+  VmRfa r;
+  r.window = sw.window;
+  r.arg = p.arg;
+  r.prev_value = vm_dnan();
+  r.prev_timestamp = t_start - sw.max_prev_interval;
+  if (i < count && i > 0 && ts[i - 1] > r.prev_timestamp) {
+    r.prev_value = vals[i - 1];
+    r.prev_timestamp = ts[i - 1];
+  }
+  r.values = vals + i;
+  r.timestamps = ts + i;
+  r.n = j - i;
+  r.real_prev_value = vm_dnan();
+  if (i > 0) {
+    int64_t curr = (r.n > 0) ? ts[i] : t_start;
+    if (p.lookback_delta == 0 || (curr - ts[i - 1]) < p.lookback_delta)
+      r.real_prev_value = vals[i - 1];
+  }
+  r.real_next_value = (j < count) ? vals[j] : vm_dnan();
+  r.curr_timestamp = t_end;
+  double v = vm_eval_rollup_fn(p.func, &r);
+
+  if (p.aggr == VMGPU_AGGR_NONE) {
+    io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] = v;
+  } else {
+    int grp = io.group_ids ? io.group_ids[s] : -1;
+    if (grp >= 0 && !vm_isnan(v)) {
+      double* gv = io.out + (size_t)grp * (size_t)p.n_grid + (size_t)g;
+      double* gc = io.out_counts + (size_t)grp * (size_t)p.n_grid + (size_t)g;
+      switch (p.aggr) {
+        case VMGPU_AGGR_SUM:
+        case VMGPU_AGGR_AVG: atomicAdd(gv, v); atomicAdd(gc, 1.0); break;
+        case VMGPU_AGGR_MIN: vm_atomic_min_f64(gv, v); atomicAdd(gc, 1.0); break;
+        case VMGPU_AGGR_MAX: vm_atomic_max_f64(gv, v); atomicAdd(gc, 1.0); break;
+        case VMGPU_AGGR_COUNT:
+        case VMGPU_AGGR_GROUP: atomicAdd(gv, 1.0); atomicAdd(gc, 1.0); break;
+        case VMGPU_AGGR_SUM2: atomicAdd(gv, v * v); atomicAdd(gc, 1.0); break;
+        case VMGPU_AGGR_GEOMEAN: vm_atomic_mul_f64(gv, v); atomicAdd(gc, 1.0); break;
+        default: break;
+      }
+    }
+  }
+  return (p.sspc > 0) ? (uint64_t)p.sspc : (uint64_t)(j - i);
+}
+
+/* ------------------------------------------------------------------ */
+/* kernel 1: one wave per series (n <= CHUNK_WAVE)                    */
+/* ------------------------------------------------------------------ */
+
+__global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO io) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int wave_in_block = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  int64_t* lts = (int64_t*)(smem + (size_t)wave_in_block * CHUNK_WAVE * 16);
+  double* lvs = (double*)(smem + (size_t)wave_in_block * CHUNK_WAVE * 16 +
+                          CHUNK_WAVE * 8);
+  uint64_t scanned = 0;
+  const uint32_t wave_id = blockIdx.x * WAVES_PER_BLOCK + wave_in_block;
+  const uint32_t wave_stride = gridDim.x * WAVES_PER_BLOCK;
+
+  for (uint32_t ws = wave_id; ws < io.n_sel; ws += wave_stride) {
+    uint32_t s = io.series_sel ? io.series_sel[ws] : ws;
+    uint64_t lo = io.offsets[s];
+    int64_t n = (int64_t)(io.offsets[s + 1] - lo);
+
+    int count = load_compact_wave(io.ts + lo, io.vals + lo, n, lts, lvs,
+                                  p.drop_stale != 0, lane);
+    wave_lds_sync();
+    if (p.rcr) rcr_scan_wave(lts, lvs, count, p.max_staleness, lane);
+    wave_lds_sync();
+
+    int64_t si = p.step;
+    if (p.start < p.end) si = scrape_interval_wave(lts, count, p.step, lane);
+    SeriesWindow sw = series_window(p, si);
+
+    if (lane == 0) scanned += (uint64_t)count;
+    for (int g0 = 0; g0 < p.n_grid; g0 += WAVE) {
+      int g = g0 + lane;
+      if (g < p.n_grid) scanned += eval_grid_point(p, sw, lts, lvs, count, g, s, io);
+    }
+    wave_lds_sync();
+  }
+  /* reduce samplesScanned: wave shuffle + one atomic per wave */
+  for (int d = 32; d > 0; d >>= 1) scanned += __shfl_down((unsigned long long)scanned, d);
+  if (lane == 0 && scanned) atomicAdd(io.samples_scanned, (unsigned long long)scanned);
+}
+
+/* ------------------------------------------------------------------ */
+/* kernel 2: one 256-thread block per series (n <= CHUNK_BLOCK)       */
+/* ------------------------------------------------------------------ */
+
+__global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KIO io) {
+  /* ALL LDS carved from one dynamic region (no static __shared__ in front —
+   * guide §6 G17: statics shift the 16-B-aligned dynamic base).
+   * layout: [0,8) count (int) + si (packed), [16, 16+CB*8) ts, then vals. */
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  int* sh_count = (int*)smem;
+  int64_t* sh_si = (int64_t*)(smem + 8);
+  unsigned long long* sh_sum = (unsigned long long*)(smem + 16);
+  int64_t* lts = (int64_t*)(smem + 32);
+  double* lvs = (double*)(smem + 32 + CHUNK_BLOCK * 8);
+  const int tid = threadIdx.x;
+  const int lane = tid % WAVE;
+  const int wave = tid / WAVE;
+  uint64_t scanned = 0;
+
+  for (uint32_t bs = blockIdx.x; bs < io.n_sel; bs += gridDim.x) {
+    uint32_t s = io.series_sel[bs];
+    uint64_t lo = io.offsets[s];
+    int64_t n = (int64_t)(io.offsets[s + 1] - lo);
+
+    if (!p.drop_stale) {
+      /* plain strided copy by the whole block */
+      for (int64_t k = tid; k < n; k += BLOCK_THREADS) {
+        lts[k] = io.ts[lo + k];
+        lvs[k] = io.vals[lo + k];
+      }
+      if (tid == 0) *sh_count = (int)n;
+    } else if (wave == 0) {
+      int c = load_compact_wave(io.ts + lo, io.vals + lo, n, lts, lvs, true, lane);
+      if (lane == 0) *sh_count = c;
+    }
+    __syncthreads();
+    int count = *sh_count;
+    if (p.rcr && wave == 0) rcr_scan_wave(lts, lvs, count, p.max_staleness, lane);
+    __syncthreads();
+    if (wave == 0) {
+      int64_t si = p.step;
+      if (p.start < p.end) si = scrape_interval_wave(lts, count, p.step, lane);
+      if (lane == 0) *sh_si = si;
+    }
+    __syncthreads();
+    SeriesWindow sw = series_window(p, *sh_si);
+
+    if (tid == 0) scanned += (uint64_t)count;
+    for (int g0 = 0; g0 < p.n_grid; g0 += BLOCK_THREADS) {
+      int g = g0 + tid;
+      if (g < p.n_grid) scanned += eval_grid_point(p, sw, lts, lvs, count, g, s, io);
+    }
+    __syncthreads();
+  }
+  for (int d = 32; d > 0; d >>= 1) scanned += __shfl_down((unsigned long long)scanned, d);
+  if (tid == 0) *sh_sum = 0;
+  __syncthreads();
+  if (lane == 0 && scanned) atomicAdd(sh_sum, (unsigned long long)scanned);
+  __syncthreads();
+  if (tid == 0 && *sh_sum) atomicAdd(io.samples_scanned, *sh_sum);
+}
+
+/* ------------------------------------------------------------------ */
+/* kernel 3: huge series via global scratch                           */
+/* ------------------------------------------------------------------ */
+
+__global__ __launch_bounds__(BLOCK_THREADS) void rollup_huge_kernel(KPlan p, KIO io) {
+  __shared__ int sh_count;
+  __shared__ int64_t sh_si;
+  const int tid = threadIdx.x;
+  const int lane = tid % WAVE;
+  const int wave = tid / WAVE;
+  uint64_t scanned = 0;
+
+  for (uint32_t bs = blockIdx.x; bs < io.n_sel; bs += gridDim.x) {
+    uint32_t s = io.series_sel[bs];
+    uint64_t lo = io.offsets[s];
+    int64_t n = (int64_t)(io.offsets[s + 1] - lo);
+    int64_t* dts = io.scr_ts + io.scr_offsets[bs];
+    double* dvs = io.scr_vals + io.scr_offsets[bs];
+
+    const int64_t* uts;
+    const double* uvs;
+    if (p.drop_stale || p.rcr) {
+      if (wave == 0) {
+        int c = load_compact_wave(io.ts + lo, io.vals + lo, n, dts, dvs,
+                                  p.drop_stale != 0, lane);
+        if (p.rcr) rcr_scan_wave(dts, dvs, c, p.max_staleness, lane);
+        if (lane == 0) sh_count = c;
+      }
+      __syncthreads();
+      /* wave 0's global-scratch stores must be visible to the whole block:
+       * same-CU reads, but L1 is per-CU so plain loads after syncthreads are
+       * coherent within the block's CU. */
+      uts = dts;
+      uvs = dvs;
+    } else {
+      if (tid == 0) sh_count = (int)n;
+      __syncthreads();
+      uts = io.ts + lo;
+      uvs = io.vals + lo;
+    }
+    int count = sh_count;
+    if (wave == 0) {
+      int64_t si = p.step;
+      if (p.start < p.end) {
+        /* scrape interval needs the tail of the (possibly compacted) column */
+        si = scrape_interval_wave(uts, count, p.step, lane);
+      }
+      if (lane == 0) sh_si = si;
+    }
+    __syncthreads();
+    SeriesWindow sw = series_window(p, sh_si);
+
+    if (tid == 0) scanned += (uint64_t)count;
+    for (int g0 = 0; g0 < p.n_grid; g0 += BLOCK_THREADS) {
+      int g = g0 + tid;
+      if (g < p.n_grid) scanned += eval_grid_point(p, sw, uts, uvs, count, g, s, io);
+    }
+    __syncthreads();
+  }
+  for (int d = 32; d > 0; d >>= 1) scanned += __shfl_down((unsigned long long)scanned, d);
+  __shared__ unsigned long long block_sum;
+  if (tid == 0) block_sum = 0;
+  __syncthreads();
+  if (lane == 0 && scanned) atomicAdd(&block_sum, (unsigned long long)scanned);
+  __syncthreads();
+  if (tid == 0 && block_sum) atomicAdd(io.samples_scanned, block_sum);
+}
+
+/* ------------------------------------------------------------------ */
+/* aggregate identity init + finalize                                 */
+/* ------------------------------------------------------------------ */
+
+__global__ void aggr_init_kernel(double* values, double* counts, uint64_t n, int32_t aggr) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  double ident = 0.0;
+  if (aggr == VMGPU_AGGR_MIN) ident = vm_dinf();
+  else if (aggr == VMGPU_AGGR_MAX) ident = -vm_dinf();
+  else if (aggr == VMGPU_AGGR_GEOMEAN) ident = 1.0;
+  for (; i < n; i += stride) {
+    values[i] = ident;
+    counts[i] = 0.0;
+  }
+}
+
+__global__ void aggr_finalize_kernel(double* values, double* counts, uint64_t n, int32_t aggr) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    double c = counts[i];
+    switch (aggr) {
+      case VMGPU_AGGR_SUM:
+      case VMGPU_AGGR_MIN:
+      case VMGPU_AGGR_MAX:
+      case VMGPU_AGGR_SUM2:
+        if (c == 0) values[i] = vm_dnan();
+        break;
+      case VMGPU_AGGR_AVG:
+        values[i] = (c == 0) ? vm_dnan() : values[i] / c;
+        break;
+      case VMGPU_AGGR_COUNT:
+        if (values[i] == 0) values[i] = vm_dnan();
+        break;
+      case VMGPU_AGGR_GROUP:
+        values[i] = (values[i] == 0) ? vm_dnan() : 1.0;
+        break;
+      case VMGPU_AGGR_GEOMEAN:
+        values[i] = (c == 0) ? vm_dnan() : pow(values[i], 1.0 / c);
+        break;
+      default:
+        break;
+    }
+  }
+}
+
+/* ------------------------------------------------------------------ */
+/* host side                                                          */
+/* ------------------------------------------------------------------ */
+
+namespace {
+
+struct Batch {
+  int64_t* d_ts = nullptr;
+  double* d_vals = nullptr;
+  uint64_t* d_offsets = nullptr;
+  int32_t* d_group_ids = nullptr;
+  uint32_t n_series = 0;
+  uint32_t n_groups = 0;
+  uint64_t n_samples = 0;
+  /* series partition by length */
+  uint32_t* d_wave_list = nullptr;
+  uint32_t* d_block_list = nullptr;
+  uint32_t* d_huge_list = nullptr;
+  uint32_t n_wave = 0, n_block = 0, n_huge = 0;
+  bool wave_is_identity = false; /* all series small: skip the list */
+  uint64_t* d_huge_scr_offsets = nullptr;
+  uint64_t huge_scratch_elems = 0;
+  int64_t* d_scr_ts = nullptr;
+  double* d_scr_vals = nullptr;
+  /* outputs of the last exec */
+  double* d_out = nullptr;
+  size_t out_elems = 0;
+  double* d_counts = nullptr;
+  size_t count_elems = 0;
+  unsigned long long* d_scanned = nullptr;
+};
+
+struct Ctx {
+  std::mutex mu;
+  bool inited = false;
+  int device = 0;
+  hipStream_t stream = nullptr;
+  hipEvent_t ev_start = nullptr, ev_stop = nullptr;
+  double last_kernel_ms = 0.0;
+  std::map<uint64_t, Batch> batches;
+  uint64_t next_handle = 1;
+};
+
+Ctx g_ctx;
+
+int set_err(char* errbuf, size_t len, const char* msg) {
+  if (errbuf && len) {
+    snprintf(errbuf, len, "%s", msg);
+  }
+  return 1;
+}
+
+int hip_err(char* errbuf, size_t len, const char* what, hipError_t e) {
+  if (errbuf && len) {
+    snprintf(errbuf, len, "%s: %s", what, hipGetErrorString(e));
+  }
+  return 2;
+}
+
+#define HIP_TRY(expr, what)                              \
+  do {                                                   \
+    hipError_t _e = (expr);                              \
+    if (_e != hipSuccess) return hip_err(errbuf, errbuf_len, what, _e); \
+  } while (0)
+
+void free_batch(Batch& b) {
+  (void)hipFree(b.d_ts);
+  (void)hipFree(b.d_vals);
+  (void)hipFree(b.d_offsets);
+  (void)hipFree(b.d_group_ids);
+  (void)hipFree(b.d_wave_list);
+  (void)hipFree(b.d_block_list);
+  (void)hipFree(b.d_huge_list);
+  (void)hipFree(b.d_huge_scr_offsets);
+  (void)hipFree(b.d_scr_ts);
+  (void)hipFree(b.d_scr_vals);
+  (void)hipFree(b.d_out);
+  (void)hipFree(b.d_counts);
+  (void)hipFree(b.d_scanned);
+  b = Batch();
+}
+
+}  // namespace
+
+extern "C" {
+
+int vmgpu_init(const int* device_ids, int n_devices) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  if (g_ctx.inited) return 0;
+  if (n_devices != 1) return 1; /* one process per GPU (RCCL scaling model) */
+  int dev = device_ids ? device_ids[0] : 0;
+  if (hipSetDevice(dev) != hipSuccess) return 2;
+  if (hipStreamCreate(&g_ctx.stream) != hipSuccess) return 3;
+  if (hipEventCreate(&g_ctx.ev_start) != hipSuccess) return 4;
+  if (hipEventCreate(&g_ctx.ev_stop) != hipSuccess) return 5;
+  g_ctx.device = dev;
+  g_ctx.inited = true;
+  return 0;
+}
+
+int vmgpu_shutdown(void) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  if (!g_ctx.inited) return 0;
+  for (auto& kv : g_ctx.batches) free_batch(kv.second);
+  g_ctx.batches.clear();
+  (void)hipEventDestroy(g_ctx.ev_start);
+  (void)hipEventDestroy(g_ctx.ev_stop);
+  (void)hipStreamDestroy(g_ctx.stream);
+  g_ctx.stream = nullptr;
+  g_ctx.inited = false;
+  return 0;
+}
+
+int vmgpu_batch_create(const int64_t* ts, const double* vals,
+                       const uint64_t* offsets, uint32_t n_series,
+                       const int32_t* group_ids, uint32_t n_groups,
+                       uint64_t* out_handle,
+                       char* errbuf, size_t errbuf_len) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  if (!g_ctx.inited) return set_err(errbuf, errbuf_len, "vmgpu: not initialized");
+  if (!ts || !vals || !offsets || !out_handle || n_series == 0)
+    return set_err(errbuf, errbuf_len, "vmgpu: bad args");
+  Batch b;
+  b.n_series = n_series;
+  b.n_groups = n_groups;
+  b.n_samples = offsets[n_series];
+
+  /* partition series by length (host pass over offsets) */
+  std::vector<uint32_t> wave_list, block_list, huge_list;
+  std::vector<uint64_t> huge_scr_off;
+  uint64_t huge_total = 0;
+  for (uint32_t s = 0; s < n_series; s++) {
+    uint64_t n = offsets[s + 1] - offsets[s];
+    if (n <= CHUNK_WAVE) {
+      wave_list.push_back(s);
+    } else if (n <= CHUNK_BLOCK) {
+      block_list.push_back(s);
+    } else {
+      huge_list.push_back(s);
+      huge_scr_off.push_back(huge_total);
+      huge_total += n;
+    }
+  }
+  b.n_wave = (uint32_t)wave_list.size();
+  b.n_block = (uint32_t)block_list.size();
+  b.n_huge = (uint32_t)huge_list.size();
+  b.wave_is_identity = (b.n_wave == n_series);
+  b.huge_scratch_elems = huge_total;
+
+  HIP_TRY(hipMalloc(&b.d_ts, b.n_samples * sizeof(int64_t)), "alloc ts");
+  HIP_TRY(hipMalloc(&b.d_vals, b.n_samples * sizeof(double)), "alloc vals");
+  HIP_TRY(hipMalloc(&b.d_offsets, (n_series + 1) * sizeof(uint64_t)), "alloc offsets");
+  HIP_TRY(hipMemcpy(b.d_ts, ts, b.n_samples * sizeof(int64_t), hipMemcpyHostToDevice), "upload ts");
+  HIP_TRY(hipMemcpy(b.d_vals, vals, b.n_samples * sizeof(double), hipMemcpyHostToDevice), "upload vals");
+  HIP_TRY(hipMemcpy(b.d_offsets, offsets, (n_series + 1) * sizeof(uint64_t), hipMemcpyHostToDevice), "upload offsets");
+  if (group_ids) {
+    HIP_TRY(hipMalloc(&b.d_group_ids, n_series * sizeof(int32_t)), "alloc gids");
+    HIP_TRY(hipMemcpy(b.d_group_ids, group_ids, n_series * sizeof(int32_t), hipMemcpyHostToDevice), "upload gids");
+  }
+  if (!b.wave_is_identity && b.n_wave) {
+    HIP_TRY(hipMalloc(&b.d_wave_list, b.n_wave * 4), "alloc wave list");
+    HIP_TRY(hipMemcpy(b.d_wave_list, wave_list.data(), b.n_wave * 4, hipMemcpyHostToDevice), "upload wave list");
+  }
+  if (b.n_block) {
+    HIP_TRY(hipMalloc(&b.d_block_list, b.n_block * 4), "alloc block list");
+    HIP_TRY(hipMemcpy(b.d_block_list, block_list.data(), b.n_block * 4, hipMemcpyHostToDevice), "upload block list");
+  }
+  if (b.n_huge) {
+    HIP_TRY(hipMalloc(&b.d_huge_list, b.n_huge * 4), "alloc huge list");
+    HIP_TRY(hipMemcpy(b.d_huge_list, huge_list.data(), b.n_huge * 4, hipMemcpyHostToDevice), "upload huge list");
+    HIP_TRY(hipMalloc(&b.d_huge_scr_offsets, b.n_huge * 8), "alloc huge offsets");
+    HIP_TRY(hipMemcpy(b.d_huge_scr_offsets, huge_scr_off.data(), b.n_huge * 8, hipMemcpyHostToDevice), "upload huge offsets");
+    HIP_TRY(hipMalloc(&b.d_scr_ts, huge_total * sizeof(int64_t)), "alloc scratch ts");
+    HIP_TRY(hipMalloc(&b.d_scr_vals, huge_total * sizeof(double)), "alloc scratch vals");
+  }
+  HIP_TRY(hipMalloc(&b.d_scanned, sizeof(unsigned long long)), "alloc scanned");
+
+  uint64_t h = g_ctx.next_handle++;
+  g_ctx.batches[h] = b;
+  *out_handle = h;
+  return 0;
+}
+
+int vmgpu_batch_destroy(uint64_t handle) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  auto it = g_ctx.batches.find(handle);
+  if (it == g_ctx.batches.end()) return 1;
+  free_batch(it->second);
+  g_ctx.batches.erase(it);
+  return 0;
+}
+
+int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
+                      double* out, double* out_counts,
+                      uint64_t* out_samples_scanned,
+                      char* errbuf, size_t errbuf_len) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  if (!g_ctx.inited) return set_err(errbuf, errbuf_len, "vmgpu: not initialized");
+  auto it = g_ctx.batches.find(handle);
+  if (it == g_ctx.batches.end()) return set_err(errbuf, errbuf_len, "vmgpu: bad handle");
+  Batch& b = it->second;
+  if (plan->step <= 0 || plan->start > plan->end)
+    return set_err(errbuf, errbuf_len, "vmgpu: bad grid");
+  int64_t n_grid64 = 1 + (plan->end - plan->start) / plan->step;
+  if (n_grid64 > (int64_t)1 << 30)
+    return set_err(errbuf, errbuf_len, "vmgpu: grid too large");
+  int32_t n_grid = (int32_t)n_grid64;
+  bool grouped = plan->aggr != VMGPU_AGGR_NONE;
+  if (grouped && (!b.d_group_ids || b.n_groups == 0))
+    return set_err(errbuf, errbuf_len, "vmgpu: aggr plan needs group_ids");
+
+  size_t out_elems = grouped ? (size_t)b.n_groups * n_grid
+                             : (size_t)b.n_series * n_grid;
+  size_t count_elems = grouped ? out_elems : 0;
+  if (b.out_elems < out_elems) {
+    (void)hipFree(b.d_out);
+    b.d_out = nullptr;
+    HIP_TRY(hipMalloc(&b.d_out, out_elems * sizeof(double)), "alloc out");
+    b.out_elems = out_elems;
+  }
+  if (count_elems && b.count_elems < count_elems) {
+    (void)hipFree(b.d_counts);
+    b.d_counts = nullptr;
+    HIP_TRY(hipMalloc(&b.d_counts, count_elems * sizeof(double)), "alloc counts");
+    b.count_elems = count_elems;
+  }
+  HIP_TRY(hipMemsetAsync(b.d_scanned, 0, 8, g_ctx.stream), "zero scanned");
+
+  KPlan p;
+  p.start = plan->start;
+  p.end = plan->end;
+  p.step = plan->step;
+  p.window = plan->window;
+  p.lookback_delta = plan->lookback_delta;
+  p.min_staleness = plan->min_staleness_interval;
+  p.max_staleness = plan->max_staleness_interval;
+  p.n_grid = n_grid;
+  p.func = plan->func;
+  p.aggr = plan->aggr;
+  p.sspc = plan->samples_scanned_per_call;
+  p.may_adjust = plan->may_adjust_window;
+  p.is_default = plan->is_default_rollup;
+  p.rcr = plan->remove_counter_resets;
+  p.drop_stale = plan->drop_stale_nans;
+  p.arg = plan->arg;
+
+  KIO io;
+  io.ts = b.d_ts;
+  io.vals = b.d_vals;
+  io.offsets = b.d_offsets;
+  io.group_ids = b.d_group_ids;
+  io.out = b.d_out;
+  io.out_counts = b.d_counts;
+  io.samples_scanned = b.d_scanned;
+  io.scr_ts = b.d_scr_ts;
+  io.scr_vals = b.d_scr_vals;
+  io.scr_offsets = b.d_huge_scr_offsets;
+
+  if (grouped) {
+    uint64_t n = out_elems;
+    int blocks = (int)std::min<uint64_t>((n + 255) / 256, 2048);
+    hipLaunchKernelGGL(aggr_init_kernel, dim3(blocks), dim3(256), 0, g_ctx.stream,
+                       b.d_out, b.d_counts, n, plan->aggr);
+  }
+
+  HIP_TRY(hipEventRecord(g_ctx.ev_start, g_ctx.stream), "event start");
+
+  if (b.n_wave) {
+    KIO w = io;
+    w.series_sel = b.wave_is_identity ? nullptr : b.d_wave_list;
+    w.n_sel = b.n_wave;
+    uint32_t blocks = std::min<uint32_t>((b.n_wave + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK,
+                                         MAX_WAVE_BLOCKS);
+    size_t lds = (size_t)WAVES_PER_BLOCK * CHUNK_WAVE * 16;
+    hipLaunchKernelGGL(rollup_wave_kernel, dim3(blocks), dim3(BLOCK_THREADS), lds,
+                       g_ctx.stream, p, w);
+  }
+  if (b.n_block) {
+    KIO w = io;
+    w.series_sel = b.d_block_list;
+    w.n_sel = b.n_block;
+    uint32_t blocks = std::min<uint32_t>(b.n_block, 2048);
+    size_t lds = 32 + (size_t)CHUNK_BLOCK * 16;
+    hipLaunchKernelGGL(rollup_block_kernel, dim3(blocks), dim3(BLOCK_THREADS), lds,
+                       g_ctx.stream, p, w);
+  }
+  if (b.n_huge) {
+    KIO w = io;
+    w.series_sel = b.d_huge_list;
+    w.n_sel = b.n_huge;
+    uint32_t blocks = std::min<uint32_t>(b.n_huge, 2048);
+    hipLaunchKernelGGL(rollup_huge_kernel, dim3(blocks), dim3(BLOCK_THREADS), 0,
+                       g_ctx.stream, p, w);
+  }
+
+  HIP_TRY(hipEventRecord(g_ctx.ev_stop, g_ctx.stream), "event stop");
+
+  if (grouped && !plan->skip_finalize) {
+    uint64_t n = out_elems;
+    int blocks = (int)std::min<uint64_t>((n + 255) / 256, 2048);
+    hipLaunchKernelGGL(aggr_finalize_kernel, dim3(blocks), dim3(256), 0, g_ctx.stream,
+                       b.d_out, b.d_counts, n, plan->aggr);
+  }
+
+  if (out) {
+    HIP_TRY(hipMemcpyAsync(out, b.d_out, out_elems * sizeof(double),
+                           hipMemcpyDeviceToHost, g_ctx.stream), "download out");
+  }
+  if (out_counts && count_elems) {
+    HIP_TRY(hipMemcpyAsync(out_counts, b.d_counts, count_elems * sizeof(double),
+                           hipMemcpyDeviceToHost, g_ctx.stream), "download counts");
+  }
+  unsigned long long scanned_h = 0;
+  HIP_TRY(hipMemcpyAsync(&scanned_h, b.d_scanned, 8, hipMemcpyDeviceToHost, g_ctx.stream),
+          "download scanned");
+  HIP_TRY(hipStreamSynchronize(g_ctx.stream), "sync");
+  hipError_t kerr = hipGetLastError();
+  if (kerr != hipSuccess) return hip_err(errbuf, errbuf_len, "kernel", kerr);
+
+  float ms = 0;
+  HIP_TRY(hipEventElapsedTime(&ms, g_ctx.ev_start, g_ctx.ev_stop), "event elapsed");
+  g_ctx.last_kernel_ms = (double)ms;
+
+  if (out_samples_scanned) *out_samples_scanned = (uint64_t)scanned_h;
+  return 0;
+}
+
+int vmgpu_batch_fetch_out(uint64_t handle, double* dst, size_t n_elems,
+                          double* dst_counts, size_t n_count_elems) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  auto it = g_ctx.batches.find(handle);
+  if (it == g_ctx.batches.end()) return 1;
+  Batch& b = it->second;
+  if (dst && n_elems) {
+    if (n_elems > b.out_elems) return 2;
+    if (hipMemcpy(dst, b.d_out, n_elems * sizeof(double), hipMemcpyDeviceToHost) != hipSuccess)
+      return 3;
+  }
+  if (dst_counts && n_count_elems) {
+    if (n_count_elems > b.count_elems) return 4;
+    if (hipMemcpy(dst_counts, b.d_counts, n_count_elems * sizeof(double),
+                  hipMemcpyDeviceToHost) != hipSuccess)
+      return 5;
+  }
+  return 0;
+}
+
+void vmgpu_aggr_finalize_host(int32_t aggr, double* values, double* counts,
+                              uint64_t n_elems) {
+  const double dnan = __builtin_nan("");
+  for (uint64_t i = 0; i < n_elems; i++) {
+    double c = counts ? counts[i] : 0.0;
+    switch (aggr) {
+      case VMGPU_AGGR_SUM:
+      case VMGPU_AGGR_MIN:
+      case VMGPU_AGGR_MAX:
+      case VMGPU_AGGR_SUM2:
+        if (c == 0) values[i] = dnan;
+        break;
+      case VMGPU_AGGR_AVG:
+        values[i] = (c == 0) ? dnan : values[i] / c;
+        break;
+      case VMGPU_AGGR_COUNT:
+        if (values[i] == 0) values[i] = dnan;
+        break;
+      case VMGPU_AGGR_GROUP:
+        values[i] = (values[i] == 0) ? dnan : 1.0;
+        break;
+      case VMGPU_AGGR_GEOMEAN:
+        values[i] = (c == 0) ? dnan : pow(values[i], 1.0 / c);
+        break;
+      default:
+        break;
+    }
+  }
+}
+
+int vmgpu_rollup_eval(const vmgpu_plan* plan,
+                      const int64_t* ts, const double* vals,
+                      const uint64_t* offsets, uint32_t n_series,
+                      const int32_t* group_ids, uint32_t n_groups,
+                      double* out, double* out_counts,
+                      uint64_t* out_samples_scanned,
+                      char* errbuf, size_t errbuf_len) {
+  uint64_t h = 0;
+  int rc = vmgpu_batch_create(ts, vals, offsets, n_series, group_ids, n_groups,
+                              &h, errbuf, errbuf_len);
+  if (rc) return rc;
+  rc = vmgpu_rollup_exec(plan, h, out, out_counts, out_samples_scanned,
+                         errbuf, errbuf_len);
+  vmgpu_batch_destroy(h);
+  return rc;
+}
+
+int vmgpu_last_kernel_ms(double* out_ms) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  if (!g_ctx.inited) return 1;
+  *out_ms = g_ctx.last_kernel_ms;
+  return 0;
+}
+
+int vmgpu_device_info(char* name, size_t name_len, double* hbm_gib, int* cu_count) {
+  hipDeviceProp_t prop;
+  int dev = 0;
+  (void)hipGetDevice(&dev);
+  if (hipGetDeviceProperties(&prop, dev) != hipSuccess) return 1;
+  if (name && name_len) snprintf(name, name_len, "%s", prop.name);
+  if (hbm_gib) *hbm_gib = (double)prop.totalGlobalMem / (1024.0 * 1024.0 * 1024.0);
+  if (cu_count) *cu_count = prop.multiProcessorCount;
+  return 0;
+}
+
+}  /* extern "C" */

@@ -1,0 +1,335 @@
+"""Host-side mirror of the reference's rollup operator interface.
+
+This module is the product path: it mirrors the seam of
+app/vmselect/promql/eval.go:1899-1904 (getRollupConfigs + the RunParallel
+worker fan-out) over the C-ABI in include/vmgpu.h.  Function names, argument
+meaning and error behavior follow the reference's promql package; the labels →
+dense-group-id assignment that replaces marshalMetricNameSorted map keys
+(aggr_incremental.go:98-139) is the caller's (see group_ids argument).
+
+The HIP extension is REQUIRED: importing this module on a machine with a GPU
+and calling any evaluation entry point without victoriametrics_amd/libvmgpu.so
+raises immediately — there is no CPU fallback in the product path (the CPU
+restatement under oracle/ is test infrastructure only and is never imported
+here).
+"""
+import ctypes
+import os
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_DIR, "libvmgpu.so")
+
+# rollupFuncs map keys -> func ids (rollup.go:24-108).  Ids are shared with
+# include/vmgpu.h / oracle/vm_oracle.h.
+FUNC_IDS = {
+    "rate": 0, "increase": 1, "increase_pure": 2, "delta": 3,
+    "delta_prometheus": 4, "rate_prometheus": 5, "irate": 6, "ideriv": 7,
+    "idelta": 8, "deriv_fast": 9, "avg_over_time": 10, "min_over_time": 11,
+    "max_over_time": 12, "sum_over_time": 13, "sum2_over_time": 14,
+    "count_over_time": 15, "first_over_time": 16, "last_over_time": 17,
+    "quantile_over_time": 18, "median_over_time": 19, "stddev_over_time": 20,
+    "stdvar_over_time": 21, "changes": 22, "changes_prometheus": 23,
+    "resets": 24, "lag": 25, "lifetime": 26, "scrape_interval": 27,
+    "rate_over_sum": 28, "range_over_time": 29, "tfirst_over_time": 30,
+    "tlast_over_time": 31, "tmin_over_time": 32, "tmax_over_time": 33,
+    "tlast_change_over_time": 34, "geomean_over_time": 35,
+    "present_over_time": 36, "absent_over_time": 37,
+    "stale_samples_over_time": 38, "count_le_over_time": 39,
+    "count_gt_over_time": 40, "count_eq_over_time": 41,
+    "count_ne_over_time": 42, "share_le_over_time": 43,
+    "share_gt_over_time": 44, "share_eq_over_time": 45,
+    "sum_le_over_time": 46, "sum_gt_over_time": 47, "sum_eq_over_time": 48,
+    "deriv": 49, "predict_linear": 50, "ascent_over_time": 51,
+    "descent_over_time": 52, "zscore_over_time": 53, "integrate": 54,
+    "distinct_over_time": 55, "increases_over_time": 56,
+    "decreases_over_time": 57, "mad_over_time": 58, "default_rollup": 59,
+    "mode_over_time": 60, "duration_over_time": 61,
+    "outlier_iqr_over_time": 62,
+    # aliases to shared implementations, as in the reference map
+    "increase_prometheus": 4, "timestamp": 31, "timestamp_with_name": 31,
+}
+
+AGGR_IDS = {
+    "none": 0, "sum": 1, "min": 2, "max": 3, "avg": 4,
+    "count": 5, "sum2": 6, "geomean": 7, "group": 8,
+}
+
+# rollupFuncsRemoveCounterResets (rollup.go:223-232)
+REMOVE_COUNTER_RESETS_FUNCS = {
+    "increase", "increase_prometheus", "increase_pure", "irate", "rate",
+    "rate_prometheus", "rollup_increase", "rollup_rate",
+}
+
+# rollupFuncsCanAdjustWindow (rollup.go:204-219)
+CAN_ADJUST_WINDOW_FUNCS = {
+    "default_rollup", "deriv", "deriv_fast", "ideriv", "irate", "rate",
+    "rate_over_sum", "rollup", "rollup_candlestick", "rollup_deriv",
+    "rollup_rate", "rollup_scrape_interval", "scrape_interval", "timestamp",
+}
+
+# rollupFuncsSamplesScannedPerCall (rollup.go:238-263)
+SAMPLES_SCANNED_PER_CALL = {
+    "absent_over_time": 1, "count_over_time": 1, "default_rollup": 1,
+    "delta": 2, "delta_prometheus": 2, "deriv_fast": 2, "first_over_time": 1,
+    "idelta": 2, "ideriv": 2, "increase": 2, "increase_prometheus": 2,
+    "increase_pure": 2, "irate": 2, "lag": 1, "last_over_time": 1,
+    "lifetime": 2, "present_over_time": 1, "rate": 2, "rate_prometheus": 2,
+    "scrape_interval": 2, "tfirst_over_time": 1, "timestamp": 1,
+    "timestamp_with_name": 1, "tlast_over_time": 1,
+}
+
+# funcs whose stale NaNs must be kept (dropStaleNaNs, eval.go:2108-2115)
+KEEP_STALE_NANS_FUNCS = {"default_rollup", "stale_samples_over_time"}
+
+
+class VmGpuError(RuntimeError):
+    pass
+
+
+class _PlanC(ctypes.Structure):
+    _fields_ = [
+        ("func", ctypes.c_int32),
+        ("aggr", ctypes.c_int32),
+        ("start", ctypes.c_int64),
+        ("end", ctypes.c_int64),
+        ("step", ctypes.c_int64),
+        ("window", ctypes.c_int64),
+        ("lookback_delta", ctypes.c_int64),
+        ("min_staleness_interval", ctypes.c_int64),
+        ("max_staleness_interval", ctypes.c_int64),
+        ("may_adjust_window", ctypes.c_int32),
+        ("is_default_rollup", ctypes.c_int32),
+        ("remove_counter_resets", ctypes.c_int32),
+        ("drop_stale_nans", ctypes.c_int32),
+        ("samples_scanned_per_call", ctypes.c_int32),
+        ("skip_finalize", ctypes.c_int32),
+        ("arg", ctypes.c_double),
+    ]
+
+
+_lib = None
+
+
+def _load_lib():
+    global _lib
+    if _lib is None:
+        if not os.path.exists(_LIB_PATH):
+            raise VmGpuError(
+                f"HIP extension not built: {_LIB_PATH} is missing. "
+                "Run __graft_entry__.build() (hipcc --offload-arch=gfx950). "
+                "There is no CPU fallback in the product path.")
+        _lib = ctypes.CDLL(_LIB_PATH)
+        _lib.vmgpu_init.argtypes = [ctypes.POINTER(ctypes.c_int), ctypes.c_int]
+        _lib.vmgpu_last_kernel_ms.argtypes = [ctypes.POINTER(ctypes.c_double)]
+    return _lib
+
+
+_inited = False
+
+
+def init(device=0):
+    """Select the GPU (one process per GPU).  Raises if no usable device."""
+    global _inited
+    lib = _load_lib()
+    if _inited:
+        return
+    dev = ctypes.c_int(int(device))
+    rc = lib.vmgpu_init(ctypes.byref(dev), 1)
+    if rc != 0:
+        raise VmGpuError(
+            f"vmgpu_init(device={device}) failed with code {rc}: no usable "
+            "HIP device. The engine requires an MI355X-class GPU; there is "
+            "no CPU fallback.")
+    _inited = True
+
+
+def shutdown():
+    global _inited
+    if _inited:
+        _load_lib().vmgpu_shutdown()
+        _inited = False
+
+
+class RollupPlan:
+    """Mirror of rollupConfig construction in getRollupConfigs
+    (rollup.go:374-516) + the EvalConfig fields that reach it."""
+
+    def __init__(self, func, start, end, step, window=0, lookback_delta=0,
+                 min_staleness_interval=0, arg=0.0, aggr="none",
+                 skip_finalize=False, keep_stale_nans=False):
+        if func not in FUNC_IDS:
+            raise VmGpuError(f"unsupported rollup function {func!r}")
+        if step <= 0:
+            raise VmGpuError(f"step must be positive; got {step}")
+        if start > end:
+            raise VmGpuError(f"start {start} exceeds end {end}")
+        self.func = func
+        self.start = int(start)
+        self.end = int(end)
+        self.step = int(step)
+        self.window = int(window)
+        self.lookback_delta = int(lookback_delta)
+        self.min_staleness_interval = int(min_staleness_interval)
+        self.arg = float(arg)
+        self.aggr = aggr
+        self.skip_finalize = skip_finalize
+        self.keep_stale_nans = keep_stale_nans
+
+        rcr = func in REMOVE_COUNTER_RESETS_FUNCS
+        # stalenessInterval = lookbackDelta (+window when set), rollup.go:380-387
+        staleness = self.lookback_delta
+        if staleness != 0:
+            staleness += self.window
+        self._c = _PlanC(
+            func=FUNC_IDS[func],
+            aggr=AGGR_IDS[aggr],
+            start=self.start, end=self.end, step=self.step,
+            window=self.window,
+            lookback_delta=self.lookback_delta,
+            min_staleness_interval=self.min_staleness_interval,
+            max_staleness_interval=staleness if rcr else 0,
+            may_adjust_window=1 if func in CAN_ADJUST_WINDOW_FUNCS else 0,
+            is_default_rollup=1 if func == "default_rollup" else 0,
+            remove_counter_resets=1 if rcr else 0,
+            drop_stale_nans=0 if (keep_stale_nans or func in KEEP_STALE_NANS_FUNCS) else 1,
+            samples_scanned_per_call=SAMPLES_SCANNED_PER_CALL.get(func, 0),
+            skip_finalize=1 if skip_finalize else 0,
+            arg=self.arg)
+
+    @property
+    def n_grid(self):
+        return 1 + (self.end - self.start) // self.step
+
+    def timestamps(self):
+        """getTimestamps (eval.go:234-254)."""
+        return np.arange(self.start, self.end + 1, self.step, dtype=np.int64)
+
+
+class SeriesBatch:
+    """A decoded-series batch resident on the GPU (the RunParallel callback
+    input, batched: CSR (timestamps[], values[]) + optional dense group ids)."""
+
+    def __init__(self, ts, vals, offsets, group_ids=None, n_groups=0):
+        init()
+        lib = _load_lib()
+        self.ts = np.ascontiguousarray(ts, dtype=np.int64)
+        self.vals = np.ascontiguousarray(vals, dtype=np.float64)
+        self.offsets = np.ascontiguousarray(offsets, dtype=np.uint64)
+        self.n_series = len(self.offsets) - 1
+        self.n_groups = int(n_groups)
+        if group_ids is not None:
+            self.group_ids = np.ascontiguousarray(group_ids, dtype=np.int32)
+            gptr = self.group_ids.ctypes.data_as(ctypes.POINTER(ctypes.c_int32))
+        else:
+            self.group_ids = None
+            gptr = None
+        handle = ctypes.c_uint64(0)
+        errbuf = ctypes.create_string_buffer(256)
+        rc = lib.vmgpu_batch_create(
+            self.ts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+            self.vals.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+            self.offsets.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            ctypes.c_uint32(self.n_series), gptr,
+            ctypes.c_uint32(self.n_groups), ctypes.byref(handle),
+            errbuf, ctypes.c_size_t(len(errbuf)))
+        if rc != 0:
+            raise VmGpuError(f"vmgpu_batch_create failed ({rc}): "
+                             f"{errbuf.value.decode()}")
+        self.handle = handle.value
+
+    def close(self):
+        if self.handle:
+            _load_lib().vmgpu_batch_destroy(ctypes.c_uint64(self.handle))
+            self.handle = 0
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.close()
+
+    def exec(self, plan, download=True):
+        """Evaluate plan over this batch.  Returns (out, counts,
+        samples_scanned); out is [n_series x n_grid] or [n_groups x n_grid].
+        With download=False the results stay on the device (use fetch_out)."""
+        lib = _load_lib()
+        n_grid = plan.n_grid
+        grouped = plan.aggr != "none"
+        if grouped and self.group_ids is None:
+            raise VmGpuError("aggregate plan requires a batch with group_ids")
+        out = counts = None
+        optr = cptr = None
+        if download:
+            rows = self.n_groups if grouped else self.n_series
+            out = np.empty((rows, n_grid), dtype=np.float64)
+            optr = out.ctypes.data_as(ctypes.POINTER(ctypes.c_double))
+            if grouped:
+                counts = np.empty((rows, n_grid), dtype=np.float64)
+                cptr = counts.ctypes.data_as(ctypes.POINTER(ctypes.c_double))
+        scanned = ctypes.c_uint64(0)
+        errbuf = ctypes.create_string_buffer(256)
+        rc = lib.vmgpu_rollup_exec(ctypes.byref(plan._c),
+                                   ctypes.c_uint64(self.handle), optr, cptr,
+                                   ctypes.byref(scanned), errbuf,
+                                   ctypes.c_size_t(len(errbuf)))
+        if rc != 0:
+            raise VmGpuError(f"vmgpu_rollup_exec failed ({rc}): "
+                             f"{errbuf.value.decode()}")
+        return out, counts, scanned.value
+
+    def fetch_out(self, rows, n_grid, with_counts=False):
+        lib = _load_lib()
+        out = np.empty((rows, n_grid), dtype=np.float64)
+        counts = np.empty((rows, n_grid), dtype=np.float64) if with_counts else None
+        rc = lib.vmgpu_batch_fetch_out(
+            ctypes.c_uint64(self.handle),
+            out.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+            ctypes.c_size_t(out.size),
+            counts.ctypes.data_as(ctypes.POINTER(ctypes.c_double)) if with_counts else None,
+            ctypes.c_size_t(counts.size if with_counts else 0))
+        if rc != 0:
+            raise VmGpuError(f"vmgpu_batch_fetch_out failed ({rc})")
+        return out, counts
+
+
+def rollup_eval(plan, ts, vals, offsets, group_ids=None, n_groups=0):
+    """One-shot evaluation (upload + exec + free) — the cgo-shim shape."""
+    with SeriesBatch(ts, vals, offsets, group_ids, n_groups) as b:
+        return b.exec(plan)
+
+
+def aggr_finalize(aggr, values, counts):
+    """finalizeTimeseries tail (aggr_incremental.go:141-168): applied on the
+    host after the cross-shard all-reduce of skip_finalize outputs."""
+    lib = _load_lib()
+    v = np.ascontiguousarray(values, dtype=np.float64)
+    c = np.ascontiguousarray(counts, dtype=np.float64) if counts is not None else None
+    lib.vmgpu_aggr_finalize_host(
+        ctypes.c_int32(AGGR_IDS[aggr]),
+        v.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        c.ctypes.data_as(ctypes.POINTER(ctypes.c_double)) if c is not None else None,
+        ctypes.c_uint64(v.size))
+    return v
+
+
+def last_kernel_ms():
+    lib = _load_lib()
+    ms = ctypes.c_double(0)
+    rc = lib.vmgpu_last_kernel_ms(ctypes.byref(ms))
+    if rc != 0:
+        raise VmGpuError("vmgpu_last_kernel_ms failed")
+    return ms.value
+
+
+def device_info():
+    lib = _load_lib()
+    name = ctypes.create_string_buffer(128)
+    hbm = ctypes.c_double(0)
+    cus = ctypes.c_int(0)
+    rc = lib.vmgpu_device_info(name, ctypes.c_size_t(len(name)),
+                               ctypes.byref(hbm), ctypes.byref(cus))
+    if rc != 0:
+        raise VmGpuError("vmgpu_device_info failed")
+    return {"name": name.value.decode(), "hbm_gib": hbm.value, "cus": cus.value}
