@@ -1,0 +1,241 @@
+#!/usr/bin/env python3
+"""bench.py — flagship benchmark: vmselect rollup hot path on MI355X.
+
+Workload (BASELINE.json configs[1]): rate(metric[5m])[1h:15s] over 1M
+synthetic counter series x 240 samples, f64, per GPU.  A "step" is one pass
+of the hot path over one resident batch: removeCounterResets + rollup grid
+evaluation over all series (the RunParallel seam, eval.go:1927/1968).
+
+  python bench.py --gpus N --steps K --warmup W [--series S] [--aggr sum]
+
+N>1 runs under torchrun (one rank per GPU over RCCL); series are sharded by
+seriesID with a full-size shard per rank (weak scaling); the
+ungrouped headline path has no data-path collective.  --aggr sum switches to
+configs[2] (sum by(pod), 10k groups) whose only exchange is one all-reduce of
+the [groups x grid] partial matrices.
+
+Prints ONE JSON line from rank 0 (driver contract), including:
+  roofline: achieved = algorithmic bytes/launch / HIP-event kernel time
+            (16 B/sample read + 8 B/grid-point write)
+  cpu_baseline: the CPU oracle (kind "port") on this box's host cores over a
+            bounded subsample of the same workload
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+STEP_MS = 15_000
+WINDOW_MS = 300_000
+START_TS = 1_600_000_000_000
+HBM_PEAK_GBS = 8000.0  # gfx950 spec peak (MI355X_MICROARCH.md)
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--series", type=int, default=1_000_000)
+    ap.add_argument("--samples", type=int, default=240)
+    ap.add_argument("--aggr", type=str, default="none",
+                    choices=["none", "sum", "min", "max", "avg", "count"])
+    ap.add_argument("--groups", type=int, default=10_000)
+    ap.add_argument("--cpu-baseline-target-s", type=float, default=15.0)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    return ap.parse_args()
+
+
+def cpu_baseline(args, plan_c, ts, vals, offsets, n_grid):
+    """Time the CPU oracle (the reference algorithm restated in C, OpenMP)
+    on a bounded subsample; reported, not the target."""
+    import oracle
+    cores = os.cpu_count() or 1
+    n_series = len(offsets) - 1
+    # calibrate on 4k series, then size the sample for the target seconds
+    probe = min(4096, n_series)
+    rc = oracle.RollupConfigC(
+        func=plan_c.func, may_adjust_window=plan_c.may_adjust_window,
+        start=plan_c.start, end=plan_c.end, step=plan_c.step,
+        window=plan_c.window, lookback_delta=plan_c.lookback_delta,
+        min_staleness_interval=plan_c.min_staleness_interval,
+        is_default_rollup=plan_c.is_default_rollup,
+        samples_scanned_per_call=plan_c.samples_scanned_per_call,
+        arg=plan_c.arg)
+
+    def run(n):
+        off = offsets[:n + 1]
+        t0 = time.perf_counter()
+        oracle.rollup_eval_batch(
+            rc, ts[:int(off[-1])], vals[:int(off[-1])], off,
+            remove_counter_resets=bool(plan_c.remove_counter_resets),
+            drop_stale_nans=bool(plan_c.drop_stale_nans),
+            n_threads=cores)
+        return time.perf_counter() - t0
+
+    t_probe = run(probe)
+    per_series = t_probe / probe
+    n_target = int(min(n_series, max(probe, args.cpu_baseline_target_s / per_series)))
+    t = run(n_target)
+    sps = (n_target * args.samples) / t
+    return {
+        "value": sps,
+        "unit": "samples/s",
+        "cores": cores,
+        "kind": "port",
+        "sample": f"{n_target} of {n_series} series x {args.samples} samples, "
+                  f"{t:.2f}s wall",
+    }
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    distributed = world > 1
+
+    import torch
+    if distributed:
+        import torch.distributed as dist
+        dist.init_process_group("nccl")
+        torch.cuda.set_device(local_rank)
+
+    from victoriametrics_amd import engine, synth
+    engine.init(local_rank)
+
+    n_grid = args.samples  # grid has one point per scrape slot (1h @ 15s)
+    end = START_TS + (n_grid - 1) * STEP_MS
+    grouped = args.aggr != "none"
+
+    ts, vals, offsets = synth.counter_batch(
+        args.series, args.samples, START_TS, step=STEP_MS,
+        seed=8428 + rank)
+    gids = None
+    if grouped:
+        gids = (np.arange(args.series) % args.groups).astype(np.int32)
+
+    plan = engine.RollupPlan("rate", START_TS, end, STEP_MS, window=WINDOW_MS,
+                             aggr=args.aggr,
+                             skip_finalize=grouped and distributed)
+    batch = engine.SeriesBatch(ts, vals, offsets, group_ids=gids,
+                               n_groups=args.groups if grouped else 0)
+
+    def one_step():
+        out, counts, _ = batch.exec(plan, download=grouped)
+        if grouped and distributed:
+            import torch.distributed as dist
+            tv = torch.from_numpy(out).cuda(local_rank)
+            tc = torch.from_numpy(counts).cuda(local_rank)
+            dist.all_reduce(tv)  # SUM (sum/avg/count); one 19 MB exchange
+            dist.all_reduce(tc)
+            engine.aggr_finalize(args.aggr, tv.cpu().numpy(), tc.cpu().numpy())
+        return engine.last_kernel_ms()
+
+    # warmup
+    for _ in range(args.warmup):
+        one_step()
+
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    step_walls = []
+    kernel_ms_acc = 0.0
+    for _ in range(args.steps):
+        s0 = time.perf_counter()
+        kernel_ms_acc += one_step()
+        step_walls.append(time.perf_counter() - s0)
+    if distributed:
+        import torch.distributed as dist
+        torch.cuda.synchronize()
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    if distributed:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64).cuda(local_rank)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.cpu().item())
+
+    samples_per_step = args.series * args.samples
+    total_samples = world * samples_per_step * args.steps
+    value = total_samples / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+    p50_ms = float(np.median(step_walls) * 1e3)
+
+    # roofline: dominant kernel = rollup_wave_kernel (HIP-event time per exec)
+    avg_kernel_s = kernel_ms_acc / args.steps / 1e3
+    # algorithmic traffic per launch: 16 B/sample read +
+    # 8 B/grid-point write for the ungrouped rollup; grouped output is
+    # negligible (groups x grid).
+    read_bytes = samples_per_step * 16
+    write_bytes = (args.series * n_grid * 8) if not grouped else (args.groups * n_grid * 16)
+    algo_bytes = read_bytes + write_bytes
+    achieved_gbs = algo_bytes / avg_kernel_s / 1e9 if avg_kernel_s > 0 else 0.0
+    traffic = None
+    if os.environ.get("VMGPU_TRAFFIC_BYTES"):
+        traffic = float(os.environ["VMGPU_TRAFFIC_BYTES"])
+
+    result = None
+    if rank == 0:
+        cb = None
+        if not args.skip_cpu_baseline and world == 1:
+            cb = cpu_baseline(args, plan._c, ts, vals, offsets, n_grid)
+        workload = ("rate(metric[5m])[1h:15s] over 1M series x 240 samples"
+                    if not grouped else
+                    "sum by(pod)(rate(metric[5m])) over 1M series / 10k groups")
+        result = {
+            "metric": "rollup samples/sec",
+            "value": value,
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "p50_query_latency_ms": p50_ms,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "f64",
+            "data": "synthetic (seed 8428: counters, rate~Poisson(10/s), "
+                    "15s scrape +/-500ms jitter, reset p=0.01/sample)",
+            "config": {
+                "workload": workload,
+                "series_per_gpu": args.series,
+                "samples_per_series": args.samples,
+                "grid_points": n_grid,
+                "window_ms": WINDOW_MS,
+                "step_ms": STEP_MS,
+                "aggr": args.aggr,
+                "parallelism": f"series-sharded dp{world}",
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved_gbs,
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": achieved_gbs / HBM_PEAK_GBS,
+                "traffic": traffic,
+                "kernel_ms": avg_kernel_s * 1e3,
+                "algorithmic_bytes": algo_bytes,
+            },
+            "cpu_baseline": cb,
+        }
+        print(json.dumps(result), flush=True)
+
+    batch.close()
+    if distributed:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+    return result
+
+
+if __name__ == "__main__":
+    main()
