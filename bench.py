@@ -78,6 +78,7 @@ def cpu_baseline(args, plan_c, ts, vals, offsets, n_grid):
             n_threads=cores)
         return time.perf_counter() - t0
 
+    run(probe)  # warm (OpenMP spawn + page faults)
     t_probe = run(probe)
     per_series = t_probe / probe
     n_target = int(min(n_series, max(probe, args.cpu_baseline_target_s / per_series)))

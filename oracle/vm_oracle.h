@@ -8,7 +8,7 @@
  * __graft_entry__.smoke() and bench.py's cpu_baseline leg may use it.
  *
  * Pinned against the reference's own in-repo golden vectors (see
- * tests/golden/*.json, transcribed from app/vmselect/promql/rollup_test.go and
+ * tests/golden/ JSON files, transcribed from app/vmselect/promql/rollup_test.go and
  * friends). Each function cites the reference file:line it restates.
  *
  * Reference: /root/reference (VictoriaMetrics/VictoriaMetrics, 2026-08-21).

@@ -120,6 +120,7 @@ struct KPlan {
   int32_t is_default;
   int32_t rcr;
   int32_t drop_stale;
+  int32_t chunk_wave;      /* LDS samples per wave (wave kernel), 64-aligned */
   double arg;
 };
 
@@ -363,9 +364,9 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int wave_in_block = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  int64_t* lts = (int64_t*)(smem + (size_t)wave_in_block * CHUNK_WAVE * 16);
-  double* lvs = (double*)(smem + (size_t)wave_in_block * CHUNK_WAVE * 16 +
-                          CHUNK_WAVE * 8);
+  int64_t* lts = (int64_t*)(smem + (size_t)wave_in_block * p.chunk_wave * 16);
+  double* lvs = (double*)(smem + (size_t)wave_in_block * p.chunk_wave * 16 +
+                          (size_t)p.chunk_wave * 8);
   uint64_t scanned = 0;
   const uint32_t wave_id = blockIdx.x * WAVES_PER_BLOCK + wave_in_block;
   const uint32_t wave_stride = gridDim.x * WAVES_PER_BLOCK;
@@ -593,6 +594,7 @@ struct Batch {
   uint32_t* d_block_list = nullptr;
   uint32_t* d_huge_list = nullptr;
   uint32_t n_wave = 0, n_block = 0, n_huge = 0;
+  uint32_t max_wave_len = 0;
   bool wave_is_identity = false; /* all series small: skip the list */
   uint64_t* d_huge_scr_offsets = nullptr;
   uint64_t huge_scratch_elems = 0;
@@ -709,6 +711,7 @@ int vmgpu_batch_create(const int64_t* ts, const double* vals,
     uint64_t n = offsets[s + 1] - offsets[s];
     if (n <= CHUNK_WAVE) {
       wave_list.push_back(s);
+      if ((uint32_t)n > b.max_wave_len) b.max_wave_len = (uint32_t)n;
     } else if (n <= CHUNK_BLOCK) {
       block_list.push_back(s);
     } else {
@@ -818,6 +821,8 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
   p.is_default = plan->is_default_rollup;
   p.rcr = plan->remove_counter_resets;
   p.drop_stale = plan->drop_stale_nans;
+  p.chunk_wave = (int32_t)std::min<uint32_t>(
+      CHUNK_WAVE, std::max<uint32_t>(64, (b.max_wave_len + 63) & ~63u));
   p.arg = plan->arg;
 
   KIO io;
@@ -847,7 +852,7 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
     w.n_sel = b.n_wave;
     uint32_t blocks = std::min<uint32_t>((b.n_wave + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK,
                                          MAX_WAVE_BLOCKS);
-    size_t lds = (size_t)WAVES_PER_BLOCK * CHUNK_WAVE * 16;
+    size_t lds = (size_t)WAVES_PER_BLOCK * (size_t)p.chunk_wave * 16;
     hipLaunchKernelGGL(rollup_wave_kernel, dim3(blocks), dim3(BLOCK_THREADS), lds,
                        g_ctx.stream, p, w);
   }
