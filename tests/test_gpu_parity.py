@@ -208,7 +208,9 @@ def test_grouped_skip_finalize_allreduce_cut(engine, counter_small):
                                  aggr="avg")
     fin, _, _ = engine.rollup_eval(plan_fin, ts, vals, offsets,
                                    group_ids=gids, n_groups=n_groups)
-    assert_parity(host_fin.reshape(fin.shape), fin, exact=True,
+    # two independent execs accumulate f64 atomics in different orders, so
+    # the comparison bar is the float-sum bar (1e-9), not bit equality
+    assert_parity(host_fin.reshape(fin.shape), fin, exact=False, rtol=1e-9,
                   context="skip_finalize")
 
 

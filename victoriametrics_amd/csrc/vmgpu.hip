@@ -302,7 +302,11 @@ static VM_DEV SeriesWindow series_window(const KPlan& p, int64_t scrape_interval
   return sw;
 }
 
-/* --- grid-point evaluation (one lane, one grid point) ----------------- */
+/* --- grid-point evaluation (one lane, one grid point) -----------------
+ * FUNC_CT >= 0 folds the rollup-function dispatch at compile time (the hot
+ * functions get specialized kernels with small register footprints);
+ * FUNC_CT == -1 is the generic runtime-dispatch fallback. */
+template <int FUNC_CT>
 static VM_DEV uint64_t eval_grid_point(const KPlan& p, const SeriesWindow& sw,
                                        const int64_t* ts, const double* vals,
                                        int count, int g, uint32_t s,
@@ -331,7 +335,7 @@ static VM_DEV uint64_t eval_grid_point(const KPlan& p, const SeriesWindow& sw,
   }
   r.real_next_value = (j < count) ? vals[j] : vm_dnan();
   r.curr_timestamp = t_end;
-  double v = vm_eval_rollup_fn(p.func, &r);
+  double v = vm_eval_rollup_fn(FUNC_CT >= 0 ? FUNC_CT : p.func, &r);
 
   if (p.aggr == VMGPU_AGGR_NONE) {
     io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] = v;
@@ -360,6 +364,7 @@ static VM_DEV uint64_t eval_grid_point(const KPlan& p, const SeriesWindow& sw,
 /* kernel 1: one wave per series (n <= CHUNK_WAVE)                    */
 /* ------------------------------------------------------------------ */
 
+template <int FUNC_CT>
 __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO io) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int wave_in_block = threadIdx.x / WAVE;
@@ -389,7 +394,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
     if (lane == 0) scanned += (uint64_t)count;
     for (int g0 = 0; g0 < p.n_grid; g0 += WAVE) {
       int g = g0 + lane;
-      if (g < p.n_grid) scanned += eval_grid_point(p, sw, lts, lvs, count, g, s, io);
+      if (g < p.n_grid) scanned += eval_grid_point<FUNC_CT>(p, sw, lts, lvs, count, g, s, io);
     }
     wave_lds_sync();
   }
@@ -402,6 +407,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
 /* kernel 2: one 256-thread block per series (n <= CHUNK_BLOCK)       */
 /* ------------------------------------------------------------------ */
 
+template <int FUNC_CT>
 __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KIO io) {
   /* ALL LDS carved from one dynamic region (no static __shared__ in front —
    * guide §6 G17: statics shift the 16-B-aligned dynamic base).
@@ -448,7 +454,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KI
     if (tid == 0) scanned += (uint64_t)count;
     for (int g0 = 0; g0 < p.n_grid; g0 += BLOCK_THREADS) {
       int g = g0 + tid;
-      if (g < p.n_grid) scanned += eval_grid_point(p, sw, lts, lvs, count, g, s, io);
+      if (g < p.n_grid) scanned += eval_grid_point<FUNC_CT>(p, sw, lts, lvs, count, g, s, io);
     }
     __syncthreads();
   }
@@ -464,6 +470,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KI
 /* kernel 3: huge series via global scratch                           */
 /* ------------------------------------------------------------------ */
 
+template <int FUNC_CT>
 __global__ __launch_bounds__(BLOCK_THREADS) void rollup_huge_kernel(KPlan p, KIO io) {
   __shared__ int sh_count;
   __shared__ int64_t sh_si;
@@ -515,7 +522,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_huge_kernel(KPlan p, KIO
     if (tid == 0) scanned += (uint64_t)count;
     for (int g0 = 0; g0 < p.n_grid; g0 += BLOCK_THREADS) {
       int g = g0 + tid;
-      if (g < p.n_grid) scanned += eval_grid_point(p, sw, uts, uvs, count, g, s, io);
+      if (g < p.n_grid) scanned += eval_grid_point<FUNC_CT>(p, sw, uts, uvs, count, g, s, io);
     }
     __syncthreads();
   }
@@ -576,6 +583,47 @@ __global__ void aggr_finalize_kernel(double* values, double* counts, uint64_t n,
 }
 
 /* ------------------------------------------------------------------ */
+/* launch dispatch: compile-time specialization for hot funcs         */
+/* ------------------------------------------------------------------ */
+
+template <int FUNC_CT>
+static void launch_rollup_t(int which, uint32_t blocks, size_t lds,
+                            const KPlan& p, const KIO& w, hipStream_t stream) {
+  if (which == 0) {
+    hipLaunchKernelGGL(rollup_wave_kernel<FUNC_CT>, dim3(blocks),
+                       dim3(BLOCK_THREADS), lds, stream, p, w);
+  } else if (which == 1) {
+    hipLaunchKernelGGL(rollup_block_kernel<FUNC_CT>, dim3(blocks),
+                       dim3(BLOCK_THREADS), lds, stream, p, w);
+  } else {
+    hipLaunchKernelGGL(rollup_huge_kernel<FUNC_CT>, dim3(blocks),
+                       dim3(BLOCK_THREADS), lds, stream, p, w);
+  }
+}
+
+static hipStream_t launch_stream();
+
+static void launch_rollup(int which, uint32_t blocks, size_t lds,
+                          const KPlan& p, const KIO& w) {
+  hipStream_t s = launch_stream();
+  switch (p.func) {
+    case VMF_RATE: launch_rollup_t<VMF_RATE>(which, blocks, lds, p, w, s); break;
+    case VMF_INCREASE: launch_rollup_t<VMF_INCREASE>(which, blocks, lds, p, w, s); break;
+    case VMF_INCREASE_PURE: launch_rollup_t<VMF_INCREASE_PURE>(which, blocks, lds, p, w, s); break;
+    case VMF_DELTA: launch_rollup_t<VMF_DELTA>(which, blocks, lds, p, w, s); break;
+    case VMF_AVG: launch_rollup_t<VMF_AVG>(which, blocks, lds, p, w, s); break;
+    case VMF_MIN: launch_rollup_t<VMF_MIN>(which, blocks, lds, p, w, s); break;
+    case VMF_MAX: launch_rollup_t<VMF_MAX>(which, blocks, lds, p, w, s); break;
+    case VMF_SUM: launch_rollup_t<VMF_SUM>(which, blocks, lds, p, w, s); break;
+    case VMF_COUNT: launch_rollup_t<VMF_COUNT>(which, blocks, lds, p, w, s); break;
+    case VMF_LAST: launch_rollup_t<VMF_LAST>(which, blocks, lds, p, w, s); break;
+    case VMF_DEFAULT_ROLLUP: launch_rollup_t<VMF_DEFAULT_ROLLUP>(which, blocks, lds, p, w, s); break;
+    case VMF_QUANTILE: launch_rollup_t<VMF_QUANTILE>(which, blocks, lds, p, w, s); break;
+    default: launch_rollup_t<-1>(which, blocks, lds, p, w, s); break;
+  }
+}
+
+/* ------------------------------------------------------------------ */
 /* host side                                                          */
 /* ------------------------------------------------------------------ */
 
@@ -620,6 +668,11 @@ struct Ctx {
 };
 
 Ctx g_ctx;
+}  // namespace
+
+static hipStream_t launch_stream() { return g_ctx.stream; }
+
+namespace {
 
 int set_err(char* errbuf, size_t len, const char* msg) {
   if (errbuf && len) {
@@ -846,6 +899,7 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
 
   HIP_TRY(hipEventRecord(g_ctx.ev_start, g_ctx.stream), "event start");
 
+  /* compile-time specializations for the hot functions (see launch_rollup) */
   if (b.n_wave) {
     KIO w = io;
     w.series_sel = b.wave_is_identity ? nullptr : b.d_wave_list;
@@ -853,8 +907,7 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
     uint32_t blocks = std::min<uint32_t>((b.n_wave + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK,
                                          MAX_WAVE_BLOCKS);
     size_t lds = (size_t)WAVES_PER_BLOCK * (size_t)p.chunk_wave * 16;
-    hipLaunchKernelGGL(rollup_wave_kernel, dim3(blocks), dim3(BLOCK_THREADS), lds,
-                       g_ctx.stream, p, w);
+    launch_rollup(0, blocks, lds, p, w);
   }
   if (b.n_block) {
     KIO w = io;
@@ -862,16 +915,14 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
     w.n_sel = b.n_block;
     uint32_t blocks = std::min<uint32_t>(b.n_block, 2048);
     size_t lds = 32 + (size_t)CHUNK_BLOCK * 16;
-    hipLaunchKernelGGL(rollup_block_kernel, dim3(blocks), dim3(BLOCK_THREADS), lds,
-                       g_ctx.stream, p, w);
+    launch_rollup(1, blocks, lds, p, w);
   }
   if (b.n_huge) {
     KIO w = io;
     w.series_sel = b.d_huge_list;
     w.n_sel = b.n_huge;
     uint32_t blocks = std::min<uint32_t>(b.n_huge, 2048);
-    hipLaunchKernelGGL(rollup_huge_kernel, dim3(blocks), dim3(BLOCK_THREADS), 0,
-                       g_ctx.stream, p, w);
+    launch_rollup(2, blocks, 0, p, w);
   }
 
   HIP_TRY(hipEventRecord(g_ctx.ev_stop, g_ctx.stream), "event stop");
