@@ -43,6 +43,7 @@ def parse_args():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--series", type=int, default=1_000_000)
     ap.add_argument("--samples", type=int, default=240)
+    ap.add_argument("--func", type=str, default="rate")
     ap.add_argument("--aggr", type=str, default="none",
                     choices=["none", "sum", "min", "max", "avg", "count"])
     ap.add_argument("--groups", type=int, default=10_000)
@@ -121,7 +122,7 @@ def main():
     if grouped:
         gids = (np.arange(args.series) % args.groups).astype(np.int32)
 
-    plan = engine.RollupPlan("rate", START_TS, end, STEP_MS, window=WINDOW_MS,
+    plan = engine.RollupPlan(args.func, START_TS, end, STEP_MS, window=WINDOW_MS,
                              aggr=args.aggr,
                              skip_finalize=grouped and distributed)
     batch = engine.SeriesBatch(ts, vals, offsets, group_ids=gids,
@@ -189,7 +190,7 @@ def main():
         cb = None
         if not args.skip_cpu_baseline and world == 1:
             cb = cpu_baseline(args, plan._c, ts, vals, offsets, n_grid)
-        workload = ("rate(metric[5m])[1h:15s] over 1M series x 240 samples"
+        workload = (f"{args.func}(metric[5m])[1h:15s] over 1M series x 240 samples"
                     if not grouped else
                     "sum by(pod)(rate(metric[5m])) over 1M series / 10k groups")
         result = {

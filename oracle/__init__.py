@@ -312,3 +312,190 @@ def rollup_eval_batch(rc, ts, vals, offsets, group_ids=None, n_groups=0,
     if rcode != 0:
         raise RuntimeError(f"vm_rollup_eval_batch failed: {rcode}")
     return out, counts, scanned.value
+
+
+# ---------------------------------------------------------------------------
+# lib/decimal + lib/encoding + merge/dedup oracle (vm_decimal.h)
+# ---------------------------------------------------------------------------
+
+MT_ZSTD_NEAREST_DELTA2 = 1
+MT_DELTA_CONST = 2
+MT_CONST = 3
+MT_ZSTD_NEAREST_DELTA = 4
+MT_NEAREST_DELTA2 = 5
+MT_NEAREST_DELTA = 6
+
+
+def _codec_lib():
+    l = lib()
+    if not hasattr(l.vm_decimal_to_float, "_typed"):
+        l.vm_decimal_to_float.restype = ctypes.c_double
+        l.vm_decimal_to_float.argtypes = [ctypes.c_int64, ctypes.c_int16]
+        l.vm_decimal_round_to_decimal_digits.restype = ctypes.c_double
+        l.vm_decimal_round_to_decimal_digits.argtypes = [ctypes.c_double, ctypes.c_int]
+        l.vm_decimal_round_to_significant_figures.restype = ctypes.c_double
+        l.vm_decimal_round_to_significant_figures.argtypes = [ctypes.c_double, ctypes.c_int]
+        l.vm_decimal_max_up_exponent.restype = ctypes.c_int16
+        l.vm_decimal_max_up_exponent.argtypes = [ctypes.c_int64]
+        l.vm_decimal_calibrate_scale.restype = ctypes.c_int16
+        l.vm_marshal_varint64s.restype = ctypes.c_size_t
+        l.vm_unmarshal_varint64s.restype = ctypes.c_int64
+        l.vm_marshal_nearest_delta.restype = ctypes.c_size_t
+        l.vm_marshal_nearest_delta2.restype = ctypes.c_size_t
+        l.vm_unmarshal_nearest_delta.restype = ctypes.c_int
+        l.vm_unmarshal_nearest_delta2.restype = ctypes.c_int
+        l.vm_marshal_int64_array.restype = ctypes.c_int64
+        l.vm_unmarshal_int64_array.restype = ctypes.c_int
+        l.vm_deduplicate_samples.restype = ctypes.c_int64
+        l.vm_merge_sort_blocks.restype = ctypes.c_int64
+        l.vm_decimal_to_float._typed = True
+    return l
+
+
+def decimal_from_float(f):
+    l = _codec_lib()
+    v = ctypes.c_int64(0)
+    e = ctypes.c_int16(0)
+    l.vm_decimal_from_float(ctypes.c_double(f), ctypes.byref(v), ctypes.byref(e))
+    return v.value, e.value
+
+
+def positive_float_to_decimal(f):
+    l = _codec_lib()
+    v = ctypes.c_int64(0)
+    e = ctypes.c_int16(0)
+    l.vm_decimal_positive_float_to_decimal(ctypes.c_double(f), ctypes.byref(v),
+                                           ctypes.byref(e))
+    return v.value, e.value
+
+
+def decimal_to_float(v, e):
+    return _codec_lib().vm_decimal_to_float(ctypes.c_int64(v), ctypes.c_int16(e))
+
+
+def decimal_append_to_float(va, e):
+    l = _codec_lib()
+    a = np.ascontiguousarray(va, dtype=np.int64)
+    out = np.empty(len(a), dtype=np.float64)
+    l.vm_decimal_append_to_float(_ptr(out, ctypes.c_double), _ptr(a, ctypes.c_int64),
+                                 ctypes.c_int64(len(a)), ctypes.c_int16(e))
+    return out
+
+
+def float_to_decimal(src):
+    l = _codec_lib()
+    s = np.ascontiguousarray(src, dtype=np.float64)
+    va = np.empty(len(s), dtype=np.int64)
+    e = ctypes.c_int16(0)
+    l.vm_decimal_append_float_to_decimal(_ptr(s, ctypes.c_double),
+                                         ctypes.c_int64(len(s)),
+                                         _ptr(va, ctypes.c_int64), ctypes.byref(e))
+    return va, e.value
+
+
+def marshal_varint64s(vs):
+    l = _codec_lib()
+    a = np.ascontiguousarray(vs, dtype=np.int64)
+    dst = np.empty(len(a) * 10 + 16, dtype=np.uint8)
+    n = l.vm_marshal_varint64s(_ptr(dst, ctypes.c_uint8), _ptr(a, ctypes.c_int64),
+                               ctypes.c_int64(len(a)))
+    return bytes(dst[:n])
+
+
+def unmarshal_varint64s(data, n):
+    l = _codec_lib()
+    src = np.frombuffer(bytes(data), dtype=np.uint8)
+    dst = np.empty(n, dtype=np.int64)
+    used = l.vm_unmarshal_varint64s(_ptr(dst, ctypes.c_int64), ctypes.c_int64(n),
+                                    _ptr(src, ctypes.c_uint8),
+                                    ctypes.c_size_t(len(src)))
+    if used < 0:
+        raise ValueError(f"varint decode error {used}")
+    return dst, used
+
+
+def marshal_nearest_delta(src, precision_bits, delta2=False):
+    l = _codec_lib()
+    a = np.ascontiguousarray(src, dtype=np.int64)
+    dst = np.empty(len(a) * 10 + 16, dtype=np.uint8)
+    first = ctypes.c_int64(0)
+    fn = l.vm_marshal_nearest_delta2 if delta2 else l.vm_marshal_nearest_delta
+    n = fn(_ptr(dst, ctypes.c_uint8), _ptr(a, ctypes.c_int64),
+           ctypes.c_int64(len(a)), ctypes.c_uint8(precision_bits),
+           ctypes.byref(first))
+    return bytes(dst[:n]), first.value
+
+
+def unmarshal_nearest_delta(data, first_value, items, delta2=False):
+    l = _codec_lib()
+    src = np.frombuffer(bytes(data), dtype=np.uint8)
+    dst = np.empty(items, dtype=np.int64)
+    fn = l.vm_unmarshal_nearest_delta2 if delta2 else l.vm_unmarshal_nearest_delta
+    rc = fn(_ptr(dst, ctypes.c_int64),
+            _ptr(src, ctypes.c_uint8) if len(src) else None,
+            ctypes.c_size_t(len(src)), ctypes.c_int64(first_value),
+            ctypes.c_int64(items))
+    if rc != 0:
+        raise ValueError(f"nearest-delta decode error {rc}")
+    return dst
+
+
+def marshal_int64_array(a, precision_bits=64):
+    l = _codec_lib()
+    arr = np.ascontiguousarray(a, dtype=np.int64)
+    dst = np.empty(len(arr) * 10 + 64, dtype=np.uint8)
+    mt = ctypes.c_uint8(0)
+    first = ctypes.c_int64(0)
+    n = l.vm_marshal_int64_array(_ptr(dst, ctypes.c_uint8), _ptr(arr, ctypes.c_int64),
+                                 ctypes.c_int64(len(arr)),
+                                 ctypes.c_uint8(precision_bits),
+                                 ctypes.byref(mt), ctypes.byref(first))
+    if n < 0:
+        raise ValueError(f"marshal error {n}")
+    return bytes(dst[:n]), mt.value, first.value
+
+
+def unmarshal_int64_array(data, items, mt, first_value):
+    l = _codec_lib()
+    src = np.frombuffer(bytes(data), dtype=np.uint8)
+    dst = np.empty(items, dtype=np.int64)
+    rc = l.vm_unmarshal_int64_array(_ptr(dst, ctypes.c_int64), ctypes.c_int64(items),
+                                    _ptr(src, ctypes.c_uint8) if len(src) else None,
+                                    ctypes.c_size_t(len(src)),
+                                    ctypes.c_uint8(mt), ctypes.c_int64(first_value))
+    if rc != 0:
+        raise ValueError(f"unmarshal error {rc}")
+    return dst
+
+
+def deduplicate_samples(ts, vals, dedup_interval):
+    l = _codec_lib()
+    t = np.ascontiguousarray(ts, dtype=np.int64).copy()
+    v = np.ascontiguousarray(vals, dtype=np.float64).copy()
+    n = l.vm_deduplicate_samples(_ptr(t, ctypes.c_int64), _ptr(v, ctypes.c_double),
+                                 ctypes.c_int64(len(t)),
+                                 ctypes.c_int64(dedup_interval))
+    return t[:n], v[:n]
+
+
+def merge_sort_blocks(blocks, dedup_interval=0):
+    """blocks: list of (ts array, vals array)."""
+    l = _codec_lib()
+    offs = [0]
+    for t, v in blocks:
+        offs.append(offs[-1] + len(t))
+    ts = np.concatenate([np.asarray(t, dtype=np.int64) for t, _ in blocks]) \
+        if blocks and offs[-1] else np.empty(0, np.int64)
+    vals = np.concatenate([np.asarray(v, dtype=np.float64) for _, v in blocks]) \
+        if blocks and offs[-1] else np.empty(0, np.float64)
+    offsets = np.asarray(offs, dtype=np.uint64)
+    dst_t = np.empty(max(offs[-1], 1), dtype=np.int64)
+    dst_v = np.empty(max(offs[-1], 1), dtype=np.float64)
+    n = l.vm_merge_sort_blocks(_ptr(ts, ctypes.c_int64) if len(ts) else None,
+                               _ptr(vals, ctypes.c_double) if len(vals) else None,
+                               _ptr(offsets, ctypes.c_uint64),
+                               ctypes.c_int32(len(blocks)),
+                               ctypes.c_int64(dedup_interval),
+                               _ptr(dst_t, ctypes.c_int64),
+                               _ptr(dst_v, ctypes.c_double))
+    return dst_t[:n], dst_v[:n]

@@ -7,8 +7,15 @@
  *   RoundToSignificantFigures 341-370, ToFloat 376-392, FromFloat 437-457,
  *   positiveFloatToDecimal 467-550, special values 403-431.
  */
+#define _USE_MATH_DEFINES
 #include "vm_decimal.h"
 #include <math.h>
+#ifndef M_LN2
+#define M_LN2 0.693147180559945309417232121458176568
+#endif
+#ifndef M_LN10
+#define M_LN10 2.30258509299404568401799145468436421
+#endif
 #include <string.h>
 
 #define V_INF_POS ((int64_t)0x7fffffffffffffffLL)        /* 1<<63 - 1 */

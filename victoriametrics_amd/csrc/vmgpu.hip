@@ -619,6 +619,8 @@ static void launch_rollup(int which, uint32_t blocks, size_t lds,
     case VMF_LAST: launch_rollup_t<VMF_LAST>(which, blocks, lds, p, w, s); break;
     case VMF_DEFAULT_ROLLUP: launch_rollup_t<VMF_DEFAULT_ROLLUP>(which, blocks, lds, p, w, s); break;
     case VMF_QUANTILE: launch_rollup_t<VMF_QUANTILE>(which, blocks, lds, p, w, s); break;
+    case VMF_DERIV_FAST: launch_rollup_t<VMF_DERIV_FAST>(which, blocks, lds, p, w, s); break;
+    case VMF_FIRST: launch_rollup_t<VMF_FIRST>(which, blocks, lds, p, w, s); break;
     default: launch_rollup_t<-1>(which, blocks, lds, p, w, s); break;
   }
 }
