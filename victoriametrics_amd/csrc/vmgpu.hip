@@ -74,6 +74,45 @@ static VM_DEV int vm_upper_bound(const int64_t* ts, int n, int64_t seek) {
   return i;
 }
 
+/* Wave-cooperative cursor advance: returns the upper bound of `seek` in
+ * ts[0..n) given a wave-uniform starting cursor cur <= result.  All 64 lanes
+ * probe cur+lane per round; since ts is sorted the actives form a prefix. */
+static VM_DEV int vm_ub_advance(const int64_t* ts, int n, int64_t seek, int cur) {
+  const int lane = threadIdx.x % WAVE;
+  for (;;) {
+    int k = cur + lane;
+    bool le = (k < n) && (ts[k] <= seek);
+    uint64_t m = __ballot(le);
+    int adv = __popcll(m);
+    cur += adv;
+    if (adv < WAVE || cur >= n) return cur;
+  }
+}
+
+/* Per-lane upper bound with an interpolated guess inside a known bracket
+ * [lo, hi] (ts[lo-1] <= seek and (hi==n or ts[hi] > seek)).  One round of two
+ * independent LDS probes resolves the regular-sampling case; otherwise
+ * narrow and fall back to binary search.  Result identical to
+ * vm_upper_bound. */
+static VM_DEV int vm_ub_hint(const int64_t* ts, int n, int64_t seek,
+                             int lo, int hi, int guess) {
+  int g = guess < lo ? lo : (guess > hi ? hi : guess);
+  /* probe ts[g-1] and ts[g]: independent loads, one latency */
+  int64_t below = (g > 0) ? ts[g - 1] : 0;
+  int64_t at = (g < n) ? ts[g] : 0;
+  bool ok_lo = (g == 0) || (below <= seek);
+  bool ok_hi = (g >= n) || (at > seek);
+  if (ok_lo && ok_hi) return g;
+  if (!ok_lo) hi = g - 1;  /* result < g */
+  else lo = g + 1;         /* result > g */
+  while (lo < hi) {
+    int h = (lo + hi) >> 1;
+    if (ts[h] <= seek) lo = h + 1;
+    else hi = h;
+  }
+  return lo;
+}
+
 static VM_DEV void vm_atomic_min_f64(double* addr, double val) {
   unsigned long long* p = (unsigned long long*)addr;
   unsigned long long old = __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
@@ -182,7 +221,6 @@ static __device__ void rcr_scan_wave(int64_t* d_ts, double* d_vals, int count,
     bool active = k < count;
     double v = active ? d_vals[k] : 0.0;
     int64_t t = active ? d_ts[k] : 0;
-    wave_lds_sync();
     double pv = __shfl_up(v, 1);
     int64_t pt = __shfl_up(t, 1);
     if (lane == 0) { pv = prev_raw; pt = prev_ts; }
@@ -228,7 +266,6 @@ static __device__ void rcr_scan_wave(int64_t* d_ts, double* d_vals, int count,
     prev_raw = __shfl(v, last);
     prev_ts = __shfl(t, last);
     prev_fin = __shfl(x, last);
-    wave_lds_sync();
   }
 }
 
@@ -307,14 +344,12 @@ static VM_DEV SeriesWindow series_window(const KPlan& p, int64_t scrape_interval
  * functions get specialized kernels with small register footprints);
  * FUNC_CT == -1 is the generic runtime-dispatch fallback. */
 template <int FUNC_CT>
-static VM_DEV uint64_t eval_grid_point(const KPlan& p, const SeriesWindow& sw,
-                                       const int64_t* ts, const double* vals,
-                                       int count, int g, uint32_t s,
-                                       const KIO& io) {
+static VM_DEV uint64_t eval_grid_point_ij(const KPlan& p, const SeriesWindow& sw,
+                                          const int64_t* ts, const double* vals,
+                                          int count, int g, uint32_t s,
+                                          const KIO& io, int i, int j) {
   int64_t t_end = p.start + (int64_t)g * p.step;
   int64_t t_start = t_end - sw.window;
-  int i = vm_upper_bound(ts, count, t_start);
-  int j = vm_upper_bound(ts, count, t_end);
   VmRfa r;
   r.window = sw.window;
   r.arg = p.arg;
@@ -336,6 +371,7 @@ static VM_DEV uint64_t eval_grid_point(const KPlan& p, const SeriesWindow& sw,
   r.real_next_value = (j < count) ? vals[j] : vm_dnan();
   r.curr_timestamp = t_end;
   double v = vm_eval_rollup_fn(FUNC_CT >= 0 ? FUNC_CT : p.func, &r);
+  (void)t_start;
 
   if (p.aggr == VMGPU_AGGR_NONE) {
     io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] = v;
@@ -358,6 +394,18 @@ static VM_DEV uint64_t eval_grid_point(const KPlan& p, const SeriesWindow& sw,
     }
   }
   return (p.sspc > 0) ? (uint64_t)p.sspc : (uint64_t)(j - i);
+}
+
+template <int FUNC_CT>
+static VM_DEV uint64_t eval_grid_point(const KPlan& p, const SeriesWindow& sw,
+                                       const int64_t* ts, const double* vals,
+                                       int count, int g, uint32_t s,
+                                       const KIO& io) {
+  int64_t t_end = p.start + (int64_t)g * p.step;
+  int64_t t_start = t_end - sw.window;
+  int i = vm_upper_bound(ts, count, t_start);
+  int j = vm_upper_bound(ts, count, t_end);
+  return eval_grid_point_ij<FUNC_CT>(p, sw, ts, vals, count, g, s, io, i, j);
 }
 
 /* ------------------------------------------------------------------ */
@@ -392,9 +440,31 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
     SeriesWindow sw = series_window(p, si);
 
     if (lane == 0) scanned += (uint64_t)count;
+    /* window seek via wave-uniform cursors + per-lane interpolated probes:
+     * i(g) and j(g) are monotone in g, so each 64-point grid block is
+     * bracketed by i(g0)..i(gLast) (resp. j), advanced cooperatively. */
+    int ci = 0, cj = 0;
     for (int g0 = 0; g0 < p.n_grid; g0 += WAVE) {
+      int g_last = g0 + WAVE - 1;
+      if (g_last >= p.n_grid) g_last = p.n_grid - 1;
+      int64_t t_end0 = p.start + (int64_t)g0 * p.step;
+      int64_t t_end_last = p.start + (int64_t)g_last * p.step;
+      int ci_lo = vm_ub_advance(lts, count, t_end0 - sw.window, ci);
+      int ci_hi = vm_ub_advance(lts, count, t_end_last - sw.window, ci_lo);
+      int cj_lo = vm_ub_advance(lts, count, t_end0, cj);
+      int cj_hi = vm_ub_advance(lts, count, t_end_last, cj_lo);
+      ci = ci_lo;
+      cj = cj_lo;
       int g = g0 + lane;
-      if (g < p.n_grid) scanned += eval_grid_point<FUNC_CT>(p, sw, lts, lvs, count, g, s, io);
+      if (g < p.n_grid) {
+        int span = g_last - g0;
+        int gi = ci_lo + (span ? (int)(((long)(ci_hi - ci_lo) * lane) / span) : 0);
+        int gj = cj_lo + (span ? (int)(((long)(cj_hi - cj_lo) * lane) / span) : 0);
+        int64_t t_end = p.start + (int64_t)g * p.step;
+        int i = vm_ub_hint(lts, count, t_end - sw.window, ci_lo, ci_hi, gi);
+        int j = vm_ub_hint(lts, count, t_end, cj_lo, cj_hi, gj);
+        scanned += eval_grid_point_ij<FUNC_CT>(p, sw, lts, lvs, count, g, s, io, i, j);
+      }
     }
     wave_lds_sync();
   }
