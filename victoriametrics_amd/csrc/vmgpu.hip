@@ -74,21 +74,6 @@ static VM_DEV int vm_upper_bound(const int64_t* ts, int n, int64_t seek) {
   return i;
 }
 
-/* Wave-cooperative cursor advance: returns the upper bound of `seek` in
- * ts[0..n) given a wave-uniform starting cursor cur <= result.  All 64 lanes
- * probe cur+lane per round; since ts is sorted the actives form a prefix. */
-static VM_DEV int vm_ub_advance(const int64_t* ts, int n, int64_t seek, int cur) {
-  const int lane = threadIdx.x % WAVE;
-  for (;;) {
-    int k = cur + lane;
-    bool le = (k < n) && (ts[k] <= seek);
-    uint64_t m = __ballot(le);
-    int adv = __popcll(m);
-    cur += adv;
-    if (adv < WAVE || cur >= n) return cur;
-  }
-}
-
 /* Per-lane upper bound with an interpolated guess inside a known bracket
  * [lo, hi] (ts[lo-1] <= seek and (hi==n or ts[hi] > seek)).  One round of two
  * independent LDS probes resolves the regular-sampling case; otherwise
@@ -440,21 +425,25 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
     SeriesWindow sw = series_window(p, si);
 
     if (lane == 0) scanned += (uint64_t)count;
-    /* window seek via wave-uniform cursors + per-lane interpolated probes:
-     * i(g) and j(g) are monotone in g, so each 64-point grid block is
-     * bracketed by i(g0)..i(gLast) (resp. j), advanced cooperatively. */
-    int ci = 0, cj = 0;
+    /* window seek: i(g) and j(g) are monotone in g, so each 64-point grid
+     * block is bracketed by i(g0)..i(gLast) (resp. j).  One SIMT binary
+     * search computes all four bracket endpoints at once (lanes 0-3 carry
+     * the four seek keys; identical trip counts keep the wave in lockstep),
+     * then each lane resolves its own boundary with an interpolated probe. */
     for (int g0 = 0; g0 < p.n_grid; g0 += WAVE) {
       int g_last = g0 + WAVE - 1;
       if (g_last >= p.n_grid) g_last = p.n_grid - 1;
       int64_t t_end0 = p.start + (int64_t)g0 * p.step;
       int64_t t_end_last = p.start + (int64_t)g_last * p.step;
-      int ci_lo = vm_ub_advance(lts, count, t_end0 - sw.window, ci);
-      int ci_hi = vm_ub_advance(lts, count, t_end_last - sw.window, ci_lo);
-      int cj_lo = vm_ub_advance(lts, count, t_end0, cj);
-      int cj_hi = vm_ub_advance(lts, count, t_end_last, cj_lo);
-      ci = ci_lo;
-      cj = cj_lo;
+      int64_t sk = (lane == 0)   ? (t_end0 - sw.window)
+                   : (lane == 1) ? (t_end_last - sw.window)
+                   : (lane == 2) ? t_end0
+                                 : t_end_last;
+      int rbr = vm_upper_bound(lts, count, sk);
+      int ci_lo = __shfl(rbr, 0);
+      int ci_hi = __shfl(rbr, 1);
+      int cj_lo = __shfl(rbr, 2);
+      int cj_hi = __shfl(rbr, 3);
       int g = g0 + lane;
       if (g < p.n_grid) {
         int span = g_last - g0;
