@@ -74,28 +74,19 @@ static VM_DEV int vm_upper_bound(const int64_t* ts, int n, int64_t seek) {
   return i;
 }
 
-/* Per-lane upper bound with an interpolated guess inside a known bracket
- * [lo, hi] (ts[lo-1] <= seek and (hi==n or ts[hi] > seek)).  One round of two
- * independent LDS probes resolves the regular-sampling case; otherwise
- * narrow and fall back to binary search.  Result identical to
- * vm_upper_bound. */
-static VM_DEV int vm_ub_hint(const int64_t* ts, int n, int64_t seek,
-                             int lo, int hi, int guess) {
-  int g = guess < lo ? lo : (guess > hi ? hi : guess);
-  /* probe ts[g-1] and ts[g]: independent loads, one latency */
-  int64_t below = (g > 0) ? ts[g - 1] : 0;
-  int64_t at = (g < n) ? ts[g] : 0;
-  bool ok_lo = (g == 0) || (below <= seek);
-  bool ok_hi = (g >= n) || (at > seek);
-  if (ok_lo && ok_hi) return g;
-  if (!ok_lo) hi = g - 1;  /* result < g */
-  else lo = g + 1;         /* result > g */
-  while (lo < hi) {
-    int h = (lo + hi) >> 1;
-    if (ts[h] <= seek) lo = h + 1;
-    else hi = h;
+/* Per-lane upper bound from a density-interpolated guess: a bounded walk
+ * resolves near-uniform sampling in 1-2 probes; anything irregular falls
+ * back to a plain binary search.  Result identical to vm_upper_bound. */
+static VM_DEV int vm_ub_hint(const int64_t* ts, int n, int64_t seek, int g) {
+  if (g < 0) g = 0;
+  if (g > n) g = n;
+  for (int steps = 0; steps < 4; steps++) {
+    bool gt_here = (g >= n) || (ts[g] > seek);
+    if (!gt_here) { g++; continue; }       /* result is above g */
+    if (g == 0 || ts[g - 1] <= seek) return g;
+    g--;                                   /* result is below g */
   }
-  return lo;
+  return vm_upper_bound(ts, n, seek);
 }
 
 static VM_DEV void vm_atomic_min_f64(double* addr, double val) {
@@ -425,33 +416,25 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
     SeriesWindow sw = series_window(p, si);
 
     if (lane == 0) scanned += (uint64_t)count;
-    /* window seek: i(g) and j(g) are monotone in g, so each 64-point grid
-     * block is bracketed by i(g0)..i(gLast) (resp. j).  One SIMT binary
-     * search computes all four bracket endpoints at once (lanes 0-3 carry
-     * the four seek keys; identical trip counts keep the wave in lockstep),
-     * then each lane resolves its own boundary with an interpolated probe. */
+    /* window seek: a per-series linear time->index map gives each lane a
+     * density-interpolated starting guess; vm_ub_hint's bounded walk
+     * resolves it (binary-search fallback keeps irregular series exact). */
+    double idx_per_ms = 0.0;
+    int64_t ts0 = 0;
+    if (count > 1) {
+      ts0 = lts[0];
+      int64_t span_ms = lts[count - 1] - ts0;
+      idx_per_ms = span_ms > 0 ? (double)(count - 1) / (double)span_ms : 0.0;
+    }
     for (int g0 = 0; g0 < p.n_grid; g0 += WAVE) {
-      int g_last = g0 + WAVE - 1;
-      if (g_last >= p.n_grid) g_last = p.n_grid - 1;
-      int64_t t_end0 = p.start + (int64_t)g0 * p.step;
-      int64_t t_end_last = p.start + (int64_t)g_last * p.step;
-      int64_t sk = (lane == 0)   ? (t_end0 - sw.window)
-                   : (lane == 1) ? (t_end_last - sw.window)
-                   : (lane == 2) ? t_end0
-                                 : t_end_last;
-      int rbr = vm_upper_bound(lts, count, sk);
-      int ci_lo = __shfl(rbr, 0);
-      int ci_hi = __shfl(rbr, 1);
-      int cj_lo = __shfl(rbr, 2);
-      int cj_hi = __shfl(rbr, 3);
       int g = g0 + lane;
       if (g < p.n_grid) {
-        int span = g_last - g0;
-        int gi = ci_lo + (span ? (int)(((long)(ci_hi - ci_lo) * lane) / span) : 0);
-        int gj = cj_lo + (span ? (int)(((long)(cj_hi - cj_lo) * lane) / span) : 0);
         int64_t t_end = p.start + (int64_t)g * p.step;
-        int i = vm_ub_hint(lts, count, t_end - sw.window, ci_lo, ci_hi, gi);
-        int j = vm_ub_hint(lts, count, t_end, cj_lo, cj_hi, gj);
+        int64_t t_start = t_end - sw.window;
+        int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
+        int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
+        int i = vm_ub_hint(lts, count, t_start, gi);
+        int j = vm_ub_hint(lts, count, t_end, gj);
         scanned += eval_grid_point_ij<FUNC_CT>(p, sw, lts, lvs, count, g, s, io, i, j);
       }
     }
