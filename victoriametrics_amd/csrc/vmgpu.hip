@@ -405,14 +405,27 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
     uint64_t lo = io.offsets[s];
     int64_t n = (int64_t)(io.offsets[s + 1] - lo);
 
+#ifdef VMGPU_ABL_RAW_COPY
+    int count = (int)n;
+    for (int64_t k = lane; k < n; k += WAVE) {
+      lts[k] = io.ts[lo + k];
+      lvs[k] = io.vals[lo + k];
+    }
+    wave_lds_sync();
+#else
     int count = load_compact_wave(io.ts + lo, io.vals + lo, n, lts, lvs,
                                   p.drop_stale != 0, lane);
     wave_lds_sync();
+#endif
+#ifndef VMGPU_ABL_NO_RCR
     if (p.rcr) rcr_scan_wave(lts, lvs, count, p.max_staleness, lane);
     wave_lds_sync();
+#endif
 
     int64_t si = p.step;
+#ifndef VMGPU_ABL_NO_SCRAPE
     if (p.start < p.end) si = scrape_interval_wave(lts, count, p.step, lane);
+#endif
     SeriesWindow sw = series_window(p, si);
 
     if (lane == 0) scanned += (uint64_t)count;
@@ -433,9 +446,20 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
         int64_t t_start = t_end - sw.window;
         int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
         int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
+#ifdef VMGPU_ABL_NO_SEEK
+        int i = gi < 0 ? 0 : (gi > count ? count : gi);
+        int j = gj < 0 ? 0 : (gj > count ? count : gj);
+        if (j < i) j = i;
+#else
         int i = vm_ub_hint(lts, count, t_start, gi);
         int j = vm_ub_hint(lts, count, t_end, gj);
+#endif
+#ifdef VMGPU_ABL_NO_EVAL
+        io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] = (j > 0 && j <= count) ? lvs[j - 1] : 0.0;
+        scanned += 2;
+#else
         scanned += eval_grid_point_ij<FUNC_CT>(p, sw, lts, lvs, count, g, s, io, i, j);
+#endif
       }
     }
     wave_lds_sync();

@@ -19,7 +19,7 @@ import os
 import numpy as np
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
-_LIB_PATH = os.path.join(_DIR, "libvmgpu.so")
+_LIB_PATH = os.environ.get("VMGPU_LIB", os.path.join(_DIR, "libvmgpu.so"))
 
 # rollupFuncs map keys -> func ids (rollup.go:24-108).  Ids are shared with
 # include/vmgpu.h / oracle/vm_oracle.h.
