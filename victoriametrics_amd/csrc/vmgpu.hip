@@ -209,6 +209,20 @@ static __device__ void rcr_scan_wave(int64_t* d_ts, double* d_vals, int count,
      * running correction, and f64 `c + 0.0` is an identity, so walking the
      * (rare) event lanes reproduces the serial left-to-right sum bitwise. */
     uint64_t em = __ballot(active && (gap || inc != 0.0));
+    /* fast path: no events, no negative deltas, and the chunk starts at or
+     * above the carried clamp level -> corrections are uniform and the
+     * monotonic clamp is a no-op. */
+    uint64_t dm = __ballot(active && !isfirst && d < 0);
+    int last_fast = count - base - 1;
+    if (last_fast > 63) last_fast = 63;
+    if (em == 0 && dm == 0 &&
+        (base == 0 || __shfl(v, 0) + corr >= prev_fin)) {
+      if (corr != 0.0 && active) d_vals[k] = v + corr;
+      prev_raw = __shfl(v, last_fast);
+      prev_ts = __shfl(t, last_fast);
+      prev_fin = prev_raw + corr;
+      continue;
+    }
     double c = corr;
     double mycorr = corr;
     while (em) {
@@ -250,18 +264,22 @@ static __device__ void rcr_scan_wave(int64_t* d_ts, double* d_vals, int count,
  * sample gaps, computed wave-cooperatively with a rank-based selection
  * (identical result to sort + quantileSorted). */
 static __device__ int64_t scrape_interval_wave(const int64_t* d_ts, int count,
-                                               int64_t default_interval, int lane) {
+                                               int64_t default_interval, int lane,
+                                               double* scratch) {
   if (count < 2) return default_interval;
   int cnt = count - 1;
   if (cnt > 20) cnt = 20;
   bool active = lane < cnt;
   double gap = 0.0;
   if (active) gap = (double)(d_ts[count - 1 - lane] - d_ts[count - 2 - lane]);
+  if (active) scratch[lane] = gap;
   wave_lds_sync();
-  /* rank of this lane's gap among the cnt gaps (ties broken by lane) */
+  /* rank of this lane's gap among the cnt gaps (ties broken by lane);
+   * scratch broadcast keeps the cnt probes independent (one LDS round trip)
+   * instead of cnt serial shuffles */
   int rank = 0;
   for (int i = 0; i < cnt; i++) {
-    double gi = __shfl(gap, i);
+    double gi = scratch[i];
     if (active && (gi < gap || (gi == gap && i < lane))) rank++;
   }
   double nn = (double)cnt;
@@ -393,9 +411,12 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int wave_in_block = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  int64_t* lts = (int64_t*)(smem + (size_t)wave_in_block * p.chunk_wave * 16);
-  double* lvs = (double*)(smem + (size_t)wave_in_block * p.chunk_wave * 16 +
+  const size_t wave_bytes = (size_t)p.chunk_wave * 16 + 256;
+  int64_t* lts = (int64_t*)(smem + (size_t)wave_in_block * wave_bytes);
+  double* lvs = (double*)(smem + (size_t)wave_in_block * wave_bytes +
                           (size_t)p.chunk_wave * 8);
+  double* lscratch = (double*)(smem + (size_t)wave_in_block * wave_bytes +
+                               (size_t)p.chunk_wave * 16);
   uint64_t scanned = 0;
   const uint32_t wave_id = blockIdx.x * WAVES_PER_BLOCK + wave_in_block;
   const uint32_t wave_stride = gridDim.x * WAVES_PER_BLOCK;
@@ -424,7 +445,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
 
     int64_t si = p.step;
 #ifndef VMGPU_ABL_NO_SCRAPE
-    if (p.start < p.end) si = scrape_interval_wave(lts, count, p.step, lane);
+    if (p.start < p.end) si = scrape_interval_wave(lts, count, p.step, lane, lscratch);
 #endif
     SeriesWindow sw = series_window(p, si);
 
@@ -439,27 +460,33 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
       int64_t span_ms = lts[count - 1] - ts0;
       idx_per_ms = span_ms > 0 ? (double)(count - 1) / (double)span_ms : 0.0;
     }
-    for (int g0 = 0; g0 < p.n_grid; g0 += WAVE) {
-      int g = g0 + lane;
-      if (g < p.n_grid) {
-        int64_t t_end = p.start + (int64_t)g * p.step;
-        int64_t t_start = t_end - sw.window;
-        int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
-        int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
+    for (int g0 = 0; g0 < p.n_grid; g0 += 4 * WAVE) {
+      /* four grid points per lane per outer iteration: independent seek and
+       * eval chains that the scheduler interleaves (the phases are LDS-
+       * latency chains, not bandwidth-bound) */
+#pragma unroll
+      for (int u = 0; u < 4; u++) {
+        int g = g0 + u * WAVE + lane;
+        if (g < p.n_grid) {
+          int64_t t_end = p.start + (int64_t)g * p.step;
+          int64_t t_start = t_end - sw.window;
+          int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
+          int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
 #ifdef VMGPU_ABL_NO_SEEK
-        int i = gi < 0 ? 0 : (gi > count ? count : gi);
-        int j = gj < 0 ? 0 : (gj > count ? count : gj);
-        if (j < i) j = i;
+          int i = gi < 0 ? 0 : (gi > count ? count : gi);
+          int j = gj < 0 ? 0 : (gj > count ? count : gj);
+          if (j < i) j = i;
 #else
-        int i = vm_ub_hint(lts, count, t_start, gi);
-        int j = vm_ub_hint(lts, count, t_end, gj);
+          int i = vm_ub_hint(lts, count, t_start, gi);
+          int j = vm_ub_hint(lts, count, t_end, gj);
 #endif
 #ifdef VMGPU_ABL_NO_EVAL
-        io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] = (j > 0 && j <= count) ? lvs[j - 1] : 0.0;
-        scanned += 2;
+          io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] = (j > 0 && j <= count) ? lvs[j - 1] : 0.0;
+          scanned += 2;
 #else
-        scanned += eval_grid_point_ij<FUNC_CT>(p, sw, lts, lvs, count, g, s, io, i, j);
+          scanned += eval_grid_point_ij<FUNC_CT>(p, sw, lts, lvs, count, g, s, io, i, j);
 #endif
+        }
       }
     }
     wave_lds_sync();
@@ -484,6 +511,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KI
   unsigned long long* sh_sum = (unsigned long long*)(smem + 16);
   int64_t* lts = (int64_t*)(smem + 32);
   double* lvs = (double*)(smem + 32 + CHUNK_BLOCK * 8);
+  double* lscratch = (double*)(smem + 32 + (size_t)CHUNK_BLOCK * 16);
   const int tid = threadIdx.x;
   const int lane = tid % WAVE;
   const int wave = tid / WAVE;
@@ -511,7 +539,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KI
     __syncthreads();
     if (wave == 0) {
       int64_t si = p.step;
-      if (p.start < p.end) si = scrape_interval_wave(lts, count, p.step, lane);
+      if (p.start < p.end) si = scrape_interval_wave(lts, count, p.step, lane, lscratch);
       if (lane == 0) *sh_si = si;
     }
     __syncthreads();
@@ -540,6 +568,7 @@ template <int FUNC_CT>
 __global__ __launch_bounds__(BLOCK_THREADS) void rollup_huge_kernel(KPlan p, KIO io) {
   __shared__ int sh_count;
   __shared__ int64_t sh_si;
+  __shared__ double sh_scratch[32];
   const int tid = threadIdx.x;
   const int lane = tid % WAVE;
   const int wave = tid / WAVE;
@@ -578,7 +607,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_huge_kernel(KPlan p, KIO
       int64_t si = p.step;
       if (p.start < p.end) {
         /* scrape interval needs the tail of the (possibly compacted) column */
-        si = scrape_interval_wave(uts, count, p.step, lane);
+        si = scrape_interval_wave(uts, count, p.step, lane, sh_scratch);
       }
       if (lane == 0) sh_si = si;
     }
@@ -974,7 +1003,7 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
     w.n_sel = b.n_wave;
     uint32_t blocks = std::min<uint32_t>((b.n_wave + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK,
                                          MAX_WAVE_BLOCKS);
-    size_t lds = (size_t)WAVES_PER_BLOCK * (size_t)p.chunk_wave * 16;
+    size_t lds = (size_t)WAVES_PER_BLOCK * ((size_t)p.chunk_wave * 16 + 256);
     launch_rollup(0, blocks, lds, p, w);
   }
   if (b.n_block) {
@@ -982,7 +1011,7 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
     w.series_sel = b.d_block_list;
     w.n_sel = b.n_block;
     uint32_t blocks = std::min<uint32_t>(b.n_block, 2048);
-    size_t lds = 32 + (size_t)CHUNK_BLOCK * 16;
+    size_t lds = 32 + (size_t)CHUNK_BLOCK * 16 + 256;
     launch_rollup(1, blocks, lds, p, w);
   }
   if (b.n_huge) {
