@@ -499,3 +499,56 @@ def merge_sort_blocks(blocks, dedup_interval=0):
                                _ptr(dst_t, ctypes.c_int64),
                                _ptr(dst_v, ctypes.c_double))
     return dst_t[:n], dst_v[:n]
+
+
+# ---------------------------------------------------------------------------
+# topk family + histogram_quantile oracle (oracle/topk.c)
+# ---------------------------------------------------------------------------
+
+TOPK_SUMMARY_OPS = {"avg": 0, "min": 1, "max": 2, "median": 3, "last": 4}
+
+
+def topk_pointwise(values, k, reverse=False):
+    """newAggrFuncTopK: per-point top-k NaN-fill; returns a new matrix."""
+    l = lib()
+    v = np.ascontiguousarray(values, dtype=np.float64).copy()
+    n_series, n_grid = v.shape
+    l.vm_topk_pointwise(_ptr(v, ctypes.c_double), ctypes.c_int64(n_series),
+                        ctypes.c_int64(n_grid), ctypes.c_double(k),
+                        ctypes.c_int32(1 if reverse else 0))
+    return v
+
+
+def topk_range(values, k, summary="avg", reverse=False, remaining=False):
+    l = lib()
+    l.vm_topk_range.restype = ctypes.c_int64
+    v = np.ascontiguousarray(values, dtype=np.float64)
+    n_series, n_grid = v.shape
+    sel = np.empty(max(int(k), 1), dtype=np.int64)
+    rem = np.empty(n_grid, dtype=np.float64) if remaining else None
+    m = l.vm_topk_range(_ptr(v, ctypes.c_double), ctypes.c_int64(n_series),
+                        ctypes.c_int64(n_grid), ctypes.c_double(k),
+                        ctypes.c_int32(TOPK_SUMMARY_OPS[summary]),
+                        ctypes.c_int32(1 if reverse else 0),
+                        _ptr(sel, ctypes.c_int64),
+                        _ptr(rem, ctypes.c_double) if remaining else None)
+    return sel[:m], rem
+
+
+def histogram_quantile(phi, bucket_values, les, group_offsets, bounds=False):
+    l = lib()
+    bv = np.ascontiguousarray(bucket_values, dtype=np.float64)
+    le = np.ascontiguousarray(les, dtype=np.float64)
+    off = np.ascontiguousarray(group_offsets, dtype=np.uint64)
+    n_groups = len(off) - 1
+    n_grid = bv.shape[1]
+    out = np.empty((n_groups, n_grid), dtype=np.float64)
+    lo = np.empty((n_groups, n_grid), dtype=np.float64) if bounds else None
+    hi = np.empty((n_groups, n_grid), dtype=np.float64) if bounds else None
+    l.vm_histogram_quantile(ctypes.c_double(phi), _ptr(bv, ctypes.c_double),
+                            _ptr(le, ctypes.c_double), _ptr(off, ctypes.c_uint64),
+                            ctypes.c_int64(n_groups), ctypes.c_int64(n_grid),
+                            _ptr(out, ctypes.c_double),
+                            _ptr(lo, ctypes.c_double) if bounds else None,
+                            _ptr(hi, ctypes.c_double) if bounds else None)
+    return (out, lo, hi) if bounds else (out, None, None)

@@ -191,6 +191,19 @@ int vm_rollup_eval_batch(const vm_rollup_config* rc,
                          uint64_t* out_samples_scanned,
                          int n_threads);
 
+/* ---- topk family (aggr.go:646-741) + histogram_quantile (transform.go:992)
+ * — dense-matrix restatements; see oracle/topk.c ---- */
+void vm_topk_pointwise(double* values, int64_t n_series, int64_t n_grid,
+                       double k, int32_t reverse);
+double vm_topk_summary(int32_t op, const double* v, int64_t n);
+int64_t vm_topk_range(const double* values, int64_t n_series, int64_t n_grid,
+                      double k, int32_t summary_op, int32_t reverse,
+                      int64_t* out_sel, double* out_remaining);
+void vm_histogram_quantile(double phi, const double* bucket_values,
+                           const double* les, const uint64_t* group_offsets,
+                           int64_t n_groups, int64_t n_grid,
+                           double* out, double* out_lower, double* out_upper);
+
 #ifdef __cplusplus
 }
 #endif

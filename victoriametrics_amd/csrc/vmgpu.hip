@@ -89,6 +89,38 @@ static VM_DEV int vm_ub_hint(const int64_t* ts, int n, int64_t seek, int g) {
   return vm_upper_bound(ts, n, seek);
 }
 
+/* Branch-free variant: four independent probes decide among {g-1, g, g+1}
+ * with pure selects; lanes outside the +-1 window (rare on real scrape
+ * cadences) take a wave-coordinated binary-search fallback.  Result
+ * identical to vm_upper_bound. */
+static VM_DEV int vm_ub_hint_fast(const int64_t* ts, int n, int64_t seek, int g) {
+  if (g < 0) g = 0;
+  if (g > n) g = n;
+  int i_m2 = g - 2 < 0 ? 0 : g - 2;
+  int i_m1 = g - 1 < 0 ? 0 : g - 1;
+  int i_0 = g < n - 1 ? g : (n - 1 < 0 ? 0 : n - 1);
+  int i_p1 = g + 1 < n - 1 ? g + 1 : (n - 1 < 0 ? 0 : n - 1);
+  int64_t t_m2 = ts[i_m2];
+  int64_t t_m1 = ts[i_m1];
+  int64_t t_0 = ts[i_0];
+  int64_t t_p1 = ts[i_p1];
+  bool le_m2 = (g - 2 < 0) || (t_m2 <= seek);  /* ts[g-2] <= seek (vacuous at edge) */
+  bool le_m1 = (g - 1 < 0) || (g - 1 >= n) || (t_m1 <= seek);
+  bool gt_m1 = (g - 1 >= 0) && (g - 1 < n) && (t_m1 > seek);
+  bool le_0 = (g < n) && (t_0 <= seek);
+  bool gt_0 = (g >= n) || (t_0 > seek);
+  bool gt_p1 = (g + 1 >= n) || (t_p1 > seek);
+  /* ok(x): (x==0 or ts[x-1]<=seek) and (x==n or ts[x]>seek) */
+  bool ok_g = le_m1 && gt_0;
+  bool ok_p1 = (g + 1 <= n) && le_0 && gt_p1;
+  bool ok_m1 = (g - 1 >= 0) && le_m2 && gt_m1;
+  int r = ok_g ? g : (ok_p1 ? g + 1 : (ok_m1 ? g - 1 : -1));
+  if (__any(r < 0)) {
+    if (r < 0) r = vm_upper_bound(ts, n, seek);
+  }
+  return r;
+}
+
 static VM_DEV void vm_atomic_min_f64(double* addr, double val) {
   unsigned long long* p = (unsigned long long*)addr;
   unsigned long long old = __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
@@ -337,6 +369,32 @@ static VM_DEV SeriesWindow series_window(const KPlan& p, int64_t scrape_interval
  * FUNC_CT >= 0 folds the rollup-function dispatch at compile time (the hot
  * functions get specialized kernels with small register footprints);
  * FUNC_CT == -1 is the generic runtime-dispatch fallback. */
+/* Fused branch-free rate/deriv_fast evaluator (rollupDerivFast,
+ * rollup.go:1954-1989 + the rfa construction of doInternal:779-810 reduced
+ * to the fields rate reads): six independent LDS loads + selects. */
+static VM_DEV double eval_rate_fused(const KPlan& p, const SeriesWindow& sw,
+                                     const int64_t* ts, const double* vals,
+                                     int count, int i, int j, int64_t t_start) {
+  int im1 = i - 1 < 0 ? 0 : i - 1;
+  int ii = i < count - 1 ? i : (count - 1 < 0 ? 0 : count - 1);
+  int jm1 = j - 1 < 0 ? 0 : j - 1;
+  double v_prevc = vals[im1];
+  int64_t t_prevc = ts[im1];
+  double v_first = vals[ii];
+  int64_t t_first = ts[ii];
+  double v_end = vals[jm1];
+  int64_t t_end_s = ts[jm1];
+  int n = j - i;
+  bool has_prev = (i < count) && (i > 0) &&
+                  (t_prevc > t_start - sw.max_prev_interval);
+  double pv = has_prev ? v_prevc : v_first;
+  int64_t ptm = has_prev ? t_prevc : t_first;
+  double slope = (v_end - pv) / ((double)(t_end_s - ptm) / 1e3);
+  double res_prev = (n == 0) ? 0.0 : slope;
+  double res_nop = (n <= 1) ? vm_dnan() : slope;
+  return has_prev ? res_prev : res_nop;
+}
+
 template <int FUNC_CT>
 static VM_DEV uint64_t eval_grid_point_ij(const KPlan& p, const SeriesWindow& sw,
                                           const int64_t* ts, const double* vals,
@@ -477,14 +535,26 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
           int j = gj < 0 ? 0 : (gj > count ? count : gj);
           if (j < i) j = i;
 #else
-          int i = vm_ub_hint(lts, count, t_start, gi);
-          int j = vm_ub_hint(lts, count, t_end, gj);
+          int i, j;
+          if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
+            i = vm_ub_hint_fast(lts, count, t_start, gi);
+            j = vm_ub_hint_fast(lts, count, t_end, gj);
+          } else {
+            i = vm_ub_hint(lts, count, t_start, gi);
+            j = vm_ub_hint(lts, count, t_end, gj);
+          }
 #endif
 #ifdef VMGPU_ABL_NO_EVAL
           io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] = (j > 0 && j <= count) ? lvs[j - 1] : 0.0;
           scanned += 2;
 #else
-          scanned += eval_grid_point_ij<FUNC_CT>(p, sw, lts, lvs, count, g, s, io, i, j);
+          if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
+            io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] =
+                eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start);
+            scanned += 2; /* samplesScannedPerCall for rate/deriv_fast */
+          } else {
+            scanned += eval_grid_point_ij<FUNC_CT>(p, sw, lts, lvs, count, g, s, io, i, j);
+          }
 #endif
         }
       }
