@@ -369,6 +369,33 @@ static VM_DEV SeriesWindow series_window(const KPlan& p, int64_t scrape_interval
  * FUNC_CT >= 0 folds the rollup-function dispatch at compile time (the hot
  * functions get specialized kernels with small register footprints);
  * FUNC_CT == -1 is the generic runtime-dispatch fallback. */
+/* Route one grid value to the per-series matrix or the grouped aggregate
+ * (the eval seam of evalRollupNoIncrementalAggregate vs
+ * evalRollupWithIncrementalAggregate). */
+static VM_DEV void vm_emit_value(const KPlan& p, const KIO& io, uint32_t s,
+                                 int g, double v) {
+  if (p.aggr == VMGPU_AGGR_NONE) {
+    io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] = v;
+    return;
+  }
+  int grp = io.group_ids ? io.group_ids[s] : -1;
+  if (grp >= 0 && !vm_isnan(v)) {
+    double* gv = io.out + (size_t)grp * (size_t)p.n_grid + (size_t)g;
+    double* gc = io.out_counts + (size_t)grp * (size_t)p.n_grid + (size_t)g;
+    switch (p.aggr) {
+      case VMGPU_AGGR_SUM:
+      case VMGPU_AGGR_AVG: atomicAdd(gv, v); atomicAdd(gc, 1.0); break;
+      case VMGPU_AGGR_MIN: vm_atomic_min_f64(gv, v); atomicAdd(gc, 1.0); break;
+      case VMGPU_AGGR_MAX: vm_atomic_max_f64(gv, v); atomicAdd(gc, 1.0); break;
+      case VMGPU_AGGR_COUNT:
+      case VMGPU_AGGR_GROUP: atomicAdd(gv, 1.0); atomicAdd(gc, 1.0); break;
+      case VMGPU_AGGR_SUM2: atomicAdd(gv, v * v); atomicAdd(gc, 1.0); break;
+      case VMGPU_AGGR_GEOMEAN: vm_atomic_mul_f64(gv, v); atomicAdd(gc, 1.0); break;
+      default: break;
+    }
+  }
+}
+
 /* Fused branch-free rate/deriv_fast evaluator (rollupDerivFast,
  * rollup.go:1954-1989 + the rfa construction of doInternal:779-810 reduced
  * to the fields rate reads): six independent LDS loads + selects. */
@@ -424,27 +451,7 @@ static VM_DEV uint64_t eval_grid_point_ij(const KPlan& p, const SeriesWindow& sw
   r.curr_timestamp = t_end;
   double v = vm_eval_rollup_fn(FUNC_CT >= 0 ? FUNC_CT : p.func, &r);
   (void)t_start;
-
-  if (p.aggr == VMGPU_AGGR_NONE) {
-    io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] = v;
-  } else {
-    int grp = io.group_ids ? io.group_ids[s] : -1;
-    if (grp >= 0 && !vm_isnan(v)) {
-      double* gv = io.out + (size_t)grp * (size_t)p.n_grid + (size_t)g;
-      double* gc = io.out_counts + (size_t)grp * (size_t)p.n_grid + (size_t)g;
-      switch (p.aggr) {
-        case VMGPU_AGGR_SUM:
-        case VMGPU_AGGR_AVG: atomicAdd(gv, v); atomicAdd(gc, 1.0); break;
-        case VMGPU_AGGR_MIN: vm_atomic_min_f64(gv, v); atomicAdd(gc, 1.0); break;
-        case VMGPU_AGGR_MAX: vm_atomic_max_f64(gv, v); atomicAdd(gc, 1.0); break;
-        case VMGPU_AGGR_COUNT:
-        case VMGPU_AGGR_GROUP: atomicAdd(gv, 1.0); atomicAdd(gc, 1.0); break;
-        case VMGPU_AGGR_SUM2: atomicAdd(gv, v * v); atomicAdd(gc, 1.0); break;
-        case VMGPU_AGGR_GEOMEAN: vm_atomic_mul_f64(gv, v); atomicAdd(gc, 1.0); break;
-        default: break;
-      }
-    }
-  }
+  vm_emit_value(p, io, s, g, v);
   return (p.sspc > 0) ? (uint64_t)p.sspc : (uint64_t)(j - i);
 }
 
@@ -549,8 +556,8 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
           scanned += 2;
 #else
           if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
-            io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] =
-                eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start);
+            vm_emit_value(p, io, s, g,
+                          eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
             scanned += 2; /* samplesScannedPerCall for rate/deriv_fast */
           } else {
             scanned += eval_grid_point_ij<FUNC_CT>(p, sw, lts, lvs, count, g, s, io, i, j);
