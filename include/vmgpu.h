@@ -120,6 +120,29 @@ int vmgpu_rollup_eval(const vmgpu_plan* plan,
                       uint64_t* out_samples_scanned,
                       char* errbuf, size_t errbuf_len);
 
+/* topk family (newAggrFuncTopK / getRangeTopKTimeseries, aggr.go:646-741)
+ * over the last evaluated output of a batch.
+ * vmgpu_topk_range: ranks rows by a whole-range summary (summary_op:
+ *   0=avg 1=min 2=max 3=median 4=last), returns up to k row ids in the
+ *   reference's output order and optionally the per-point remaining sum.
+ * vmgpu_topk_pointwise: per-grid-point top/bottom-k across rows; all other
+ *   values become NaN (fillNaNsAtIdx); out may be NULL to keep on device. */
+int vmgpu_topk_range(uint64_t handle, double k, int32_t summary_op,
+                     int32_t reverse, int64_t* out_sel, int64_t* out_n_sel,
+                     double* out_remaining, char* errbuf, size_t errbuf_len);
+int vmgpu_topk_pointwise(uint64_t handle, double k, int32_t reverse,
+                         double* out, char* errbuf, size_t errbuf_len);
+
+/* histogram_quantile (transform.go:992-1137) over grouped le-bucket rows
+ * (rows sorted by (group, le), same-le rows pre-merged by the host;
+ * group_offsets has n_groups+1 entries).  bucket_values is
+ * [rows x n_grid] host memory; outputs are [n_groups x n_grid]. */
+int vmgpu_histogram_quantile(double phi, const double* bucket_values,
+                             const double* les, const uint64_t* group_offsets,
+                             uint32_t n_groups, int32_t n_grid,
+                             double* out, double* out_lower, double* out_upper,
+                             char* errbuf, size_t errbuf_len);
+
 /* Wall time of the rollup kernels inside the last vmgpu_rollup_exec on this
  * thread's context, measured with hipEvents on the launch stream (for the
  * bench's roofline accounting). */

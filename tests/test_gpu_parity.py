@@ -260,3 +260,99 @@ def test_full_config2_checksums(engine):
     nan_frac = np.isnan(out).mean()
     assert nan_frac < 0.02, f"unexpected NaN fraction {nan_frac}"
     assert np.nanmin(out) >= 0.0  # rates of counters are non-negative
+
+
+# ---- topk family + histogram_quantile (configs 4-5 rows) ----
+
+def test_topk_range_gpu(engine, counter_small):
+    ts, vals, offsets = counter_small
+    end = START + 239 * STEP
+    plan = engine.RollupPlan("avg_over_time", START, end, STEP, window=300_000)
+    with engine.SeriesBatch(ts, vals, offsets) as b:
+        host_out, _, _ = b.exec(plan)
+        for summary in ("avg", "min", "max", "last"):
+            for rev in (False, True):
+                sel, rem = engine.topk_range(b, 100, summary=summary,
+                                             reverse=rev, remaining=True)
+                ref_sel, ref_rem = oracle.topk_range(host_out, 100, summary,
+                                                     reverse=rev,
+                                                     remaining=True)
+                assert set(sel) == set(ref_sel), f"{summary} rev={rev}"
+                assert list(sel) == list(ref_sel), \
+                    f"{summary} rev={rev}: order differs"
+                gn, rn = np.isnan(rem), np.isnan(ref_rem)
+                assert (gn == rn).all()
+                assert np.allclose(rem[~gn], ref_rem[~rn], rtol=1e-9, atol=0), \
+                    f"{summary} rev={rev} remaining"
+
+
+def test_topk_range_k_edge_cases(engine, counter_small):
+    ts, vals, offsets = counter_small
+    end = START + 239 * STEP
+    plan = engine.RollupPlan("avg_over_time", START, end, STEP, window=300_000)
+    with engine.SeriesBatch(ts, vals, offsets) as b:
+        b.exec(plan, download=False)
+        sel, _ = engine.topk_range(b, 0)
+        assert len(sel) == 0
+        sel, _ = engine.topk_range(b, math.nan)
+        assert len(sel) == 0
+        sel, _ = engine.topk_range(b, 10**9)
+        assert len(sel) == len(offsets) - 1
+
+
+def test_topk_pointwise_gpu(engine):
+    from victoriametrics_amd import synth
+    ts, vals, offsets = synth.gauge_batch(3000, 120, START, seed=77)
+    end = START + 119 * STEP
+    plan = engine.RollupPlan("avg_over_time", START, end, STEP, window=45_000)
+    with engine.SeriesBatch(ts, vals, offsets) as b:
+        host_out, _, _ = b.exec(plan)
+        got = engine.topk_pointwise(b, 10)
+    ref = oracle.topk_pointwise(host_out, 10)
+    gn, rn = np.isnan(got), np.isnan(ref)
+    assert (gn == rn).all(), \
+        f"NaN placement differs at {np.argwhere(gn != rn)[:5]}"
+    assert np.array_equal(got[~gn], ref[~rn])
+
+
+def test_bottomk_pointwise_gpu(engine):
+    from victoriametrics_amd import synth
+    ts, vals, offsets = synth.gauge_batch(1000, 60, START, seed=78)
+    end = START + 59 * STEP
+    plan = engine.RollupPlan("last_over_time", START, end, STEP, window=45_000)
+    with engine.SeriesBatch(ts, vals, offsets) as b:
+        host_out, _, _ = b.exec(plan)
+        got = engine.topk_pointwise(b, 7, reverse=True)
+    ref = oracle.topk_pointwise(host_out, 7, reverse=True)
+    gn, rn = np.isnan(got), np.isnan(ref)
+    assert (gn == rn).all()
+    assert np.array_equal(got[~gn], ref[~rn])
+
+
+def test_histogram_quantile_gpu(engine):
+    rng = np.random.default_rng(12)
+    n_groups, n_les, n_grid = 50, 20, 100
+    les = np.concatenate([np.sort(rng.random(n_les - 1) * 100),
+                          [np.inf]])
+    rows = []
+    le_col = []
+    offs = [0]
+    for gid in range(n_groups):
+        counts = np.cumsum(rng.random((n_les, n_grid)) * 10, axis=0)
+        # inject broken buckets + NaNs + zero columns
+        counts[rng.integers(0, n_les), rng.integers(0, n_grid, 5)] = np.nan
+        if gid % 7 == 0:
+            counts[:, 0] = 0.0
+        rows.append(counts)
+        le_col.extend(les)
+        offs.append(offs[-1] + n_les)
+    bv = np.vstack(rows)
+    for phi in (0.5, 0.99, 0.0, 1.0, -0.5, 1.5, math.nan):
+        got, glo, ghi = engine.histogram_quantile(phi, bv, le_col, offs,
+                                                  bounds=True)
+        ref, rlo, rhi = oracle.histogram_quantile(phi, bv, le_col, offs,
+                                                  bounds=True)
+        for g, r, name in ((got, ref, "q"), (glo, rlo, "lo"), (ghi, rhi, "hi")):
+            gn, rn = np.isnan(g), np.isnan(r)
+            assert (gn == rn).all(), f"phi={phi} {name}"
+            assert np.array_equal(g[~gn], r[~rn]), f"phi={phi} {name}"

@@ -754,6 +754,360 @@ __global__ void aggr_finalize_kernel(double* values, double* counts, uint64_t n,
   }
 }
 
+
+/* ------------------------------------------------------------------ */
+/* topk family (aggr.go:646-741) + histogram_quantile (transform.go)  */
+/* ------------------------------------------------------------------ */
+
+/* Order-preserving u64 key for f64 under lessWithNaNs (aggr.go:1259):
+ * NaN sorts below every number for topk; reverse flips the order and puts
+ * NaN on top (greaterWithNaNs), so selection is always "k largest keys". */
+static VM_DEV unsigned long long vm_topk_key(double v, int reverse) {
+  if (vm_isnan(v)) return reverse ? ~0ULL : 0ULL;
+  unsigned long long b = (unsigned long long)__double_as_longlong(v);
+  unsigned long long ord = (b >> 63) ? ~b : (b | 0x8000000000000000ULL);
+  /* compress into (0, ~0) so the NaN sentinels stay exclusive */
+  ord = (ord >> 1) | 0x2000000000000000ULL;
+  return reverse ? ~ord : ord;
+}
+
+/* per-series range summaries (aggr.go:804-858), one wave per series row */
+__global__ __launch_bounds__(BLOCK_THREADS) void topk_summary_kernel(
+    const double* values, uint32_t n_series, int32_t n_grid, int32_t op,
+    int32_t reverse, unsigned long long* keys) {
+  const int wave_in_block = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  uint32_t wid = blockIdx.x * WAVES_PER_BLOCK + wave_in_block;
+  uint32_t stride = gridDim.x * WAVES_PER_BLOCK;
+  for (uint32_t s = wid; s < n_series; s += stride) {
+    const double* row = values + (size_t)s * n_grid;
+    double acc = vm_dnan();
+    if (op == 0 || op == 1 || op == 2) { /* avg / min / max */
+      double sum = 0, cnt = 0;
+      double mn = vm_dnan(), mx = vm_dnan();
+      for (int g = lane; g < n_grid; g += WAVE) {
+        double v = row[g];
+        if (vm_isnan(v)) continue;
+        cnt += 1;
+        sum += v;
+        if (vm_isnan(mn) || v < mn) mn = v;
+        if (vm_isnan(mx) || v > mx) mx = v;
+      }
+      for (int d = 32; d > 0; d >>= 1) {
+        double so = __shfl_down(sum, d);
+        double co = __shfl_down(cnt, d);
+        double mno = __shfl_down(mn, d);
+        double mxo = __shfl_down(mx, d);
+        sum += so;
+        cnt += co;
+        if (vm_isnan(mn) || (!vm_isnan(mno) && mno < mn)) mn = mno;
+        if (vm_isnan(mx) || (!vm_isnan(mxo) && mxo > mx)) mx = mxo;
+      }
+      if (op == 0) acc = (cnt == 0) ? vm_dnan() : sum / cnt;
+      else if (op == 1) acc = mn;
+      else acc = mx;
+    } else if (op == 4) { /* last non-NaN */
+      int base = ((n_grid + WAVE - 1) / WAVE - 1) * WAVE;
+      for (; base >= 0; base -= WAVE) {
+        int g = base + lane;
+        double v = (g < n_grid) ? row[g] : vm_dnan();
+        uint64_t m = __ballot(!vm_isnan(v));
+        if (m) {
+          int hi = 63 - __clzll((unsigned long long)m);
+          acc = __shfl(v, hi);
+          break;
+        }
+      }
+    }
+    if (lane == 0) keys[s] = vm_topk_key(acc, reverse);
+  }
+}
+
+__global__ void topk_hist_kernel(const unsigned long long* keys, uint32_t n,
+                                 unsigned long long prefix, int shift,
+                                 uint32_t* hist) {
+  /* 16-bit histogram of keys whose bits above shift+16 equal prefix */
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    unsigned long long k = keys[i];
+    if (shift < 48 && (k >> (shift + 16)) != prefix) continue;
+    atomicAdd(&hist[(k >> shift) & 0xffff], 1u);
+  }
+}
+
+__global__ void topk_collect_kernel(const unsigned long long* keys, uint32_t n,
+                                    unsigned long long kstar, uint32_t ties_quota,
+                                    uint32_t* counters, uint32_t* sel,
+                                    uint32_t sel_cap) {
+  uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+  uint32_t stride = gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    unsigned long long k = keys[i];
+    if (k > kstar) {
+      uint32_t slot = atomicAdd(&counters[0], 1u);
+      if (slot < sel_cap) sel[slot] = i;
+    } else if (k == kstar) {
+      uint32_t t = atomicAdd(&counters[1], 1u);
+      if (t < ties_quota) {
+        uint32_t slot = atomicAdd(&counters[0], 1u);
+        if (slot < sel_cap) sel[slot] = i;
+      }
+    }
+  }
+}
+
+__global__ void topk_remaining_kernel(const double* values, uint32_t n_series,
+                                      int32_t n_grid, const uint8_t* selected,
+                                      double* rem_sum, double* rem_cnt) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t total = (size_t)n_series * n_grid;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    uint32_t s = (uint32_t)(i / n_grid);
+    int32_t g = (int32_t)(i % n_grid);
+    if (selected[s]) continue;
+    double v = values[i];
+    if (vm_isnan(v)) continue;
+    atomicAdd(&rem_sum[g], v);
+    atomicAdd(&rem_cnt[g], 1.0);
+  }
+}
+
+/* ---- pointwise topk over [n_series x n_grid] ---- */
+
+__global__ void topk_col_hist_kernel(const double* values, uint32_t n_series,
+                                     int32_t n_grid, int32_t reverse,
+                                     uint32_t* hists) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t total = (size_t)n_series * n_grid;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    int32_t g = (int32_t)(i % n_grid);
+    unsigned long long k = vm_topk_key(values[i], reverse);
+    atomicAdd(&hists[(size_t)g * 65536 + (k >> 48)], 1u);
+  }
+}
+
+__global__ void topk_col_threshold_kernel(const uint32_t* hists, int32_t n_grid,
+                                          uint32_t k, uint32_t* bin_of_col,
+                                          uint32_t* above_of_col) {
+  int g = blockIdx.x * blockDim.x + threadIdx.x;
+  if (g >= n_grid) return;
+  const uint32_t* h = hists + (size_t)g * 65536;
+  uint32_t cum = 0;
+  int bin = 0;
+  for (int b = 65535; b >= 0; b--) {
+    uint32_t c = h[b];
+    if (cum + c >= k) {
+      bin = b;
+      break;
+    }
+    cum += c;
+  }
+  bin_of_col[g] = (uint32_t)bin;
+  above_of_col[g] = cum;
+}
+
+__global__ void topk_col_candidates_kernel(const double* values, uint32_t n_series,
+                                           int32_t n_grid, int32_t reverse,
+                                           const uint32_t* bin_of_col,
+                                           unsigned long long* cand,
+                                           uint32_t* cand_n, uint32_t cap,
+                                           uint32_t* overflow) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t total = (size_t)n_series * n_grid;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    int32_t g = (int32_t)(i % n_grid);
+    unsigned long long k = vm_topk_key(values[i], reverse);
+    if ((uint32_t)(k >> 48) != bin_of_col[g]) continue;
+    uint32_t slot = atomicAdd(&cand_n[g], 1u);
+    if (slot < cap) cand[(size_t)g * cap + slot] = k;
+    else atomicAdd(overflow, 1u);
+  }
+}
+
+__global__ __launch_bounds__(256) void topk_col_kstar_kernel(
+    const unsigned long long* cand, const uint32_t* cand_n, uint32_t cap,
+    const uint32_t* above_of_col, uint32_t k, int32_t n_grid,
+    unsigned long long* kstar_of_col, uint32_t* ties_of_col) {
+  int g = blockIdx.x;
+  if (g >= n_grid) return;
+  uint32_t n = cand_n[g];
+  if (n > cap) n = cap;
+  uint32_t need = k > above_of_col[g] ? k - above_of_col[g] : 0;
+  const unsigned long long* c = cand + (size_t)g * cap;
+  __shared__ unsigned long long sh_lo, sh_hi, sh_mid;
+  __shared__ uint32_t sh_cnt;
+  if (threadIdx.x == 0) {
+    sh_lo = 0;
+    sh_hi = ~0ULL;
+  }
+  __syncthreads();
+  if (need == 0 || n == 0) {
+    if (threadIdx.x == 0) {
+      kstar_of_col[g] = ~0ULL;
+      ties_of_col[g] = 0;
+    }
+    return;
+  }
+  for (int it = 0; it < 64; it++) {
+    if (threadIdx.x == 0) {
+      sh_mid = sh_lo + ((sh_hi - sh_lo) >> 1);
+      sh_cnt = 0;
+    }
+    __syncthreads();
+    unsigned long long mid = sh_mid;
+    uint32_t local = 0;
+    for (uint32_t t = threadIdx.x; t < n; t += blockDim.x)
+      if (c[t] >= mid) local++;
+    atomicAdd(&sh_cnt, local);
+    __syncthreads();
+    bool done = false;
+    if (threadIdx.x == 0) {
+      if (sh_cnt >= need) sh_lo = sh_mid + 1;
+      else sh_hi = sh_mid;
+    }
+    __syncthreads();
+    if (sh_lo >= sh_hi) { done = true; }
+    if (done) break;
+  }
+  unsigned long long kstar = sh_lo - 1;
+  if (threadIdx.x == 0) sh_cnt = 0;
+  __syncthreads();
+  uint32_t local = 0;
+  for (uint32_t t = threadIdx.x; t < n; t += blockDim.x)
+    if (c[t] > kstar) local++;
+  atomicAdd(&sh_cnt, local);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    kstar_of_col[g] = kstar;
+    ties_of_col[g] = need - sh_cnt;
+  }
+}
+
+__global__ void topk_col_fill_kernel(double* values, uint32_t n_series,
+                                     int32_t n_grid, int32_t reverse, uint32_t k,
+                                     const uint32_t* bin_of_col,
+                                     const unsigned long long* kstar_of_col,
+                                     uint32_t* ties_taken,
+                                     const uint32_t* ties_of_col) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t total = (size_t)n_series * n_grid;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    int32_t g = (int32_t)(i % n_grid);
+    unsigned long long key = vm_topk_key(values[i], reverse);
+    uint32_t bin = bin_of_col[g];
+    uint32_t kb = (uint32_t)(key >> 48);
+    bool keep;
+    if (kb > bin) {
+      keep = true;
+    } else if (kb < bin) {
+      keep = false;
+    } else {
+      unsigned long long kstar = kstar_of_col[g];
+      if (key > kstar) keep = true;
+      else if (key == kstar) {
+        uint32_t t = atomicAdd(&ties_taken[g], 1u);
+        keep = t < ties_of_col[g];
+      } else {
+        keep = false;
+      }
+    }
+    if (!keep) values[i] = vm_dnan();
+  }
+}
+
+/* histogram_quantile (transform.go:1028-1074 + fixBrokenBuckets 1140):
+ * one thread per (group, grid point) */
+__global__ void hq_kernel(double phi, const double* bv, const double* les,
+                          const uint64_t* goff, int64_t n_groups, int32_t n_grid,
+                          double* out, double* out_lo, double* out_hi) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t total = (size_t)n_groups * n_grid;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    int64_t grp = (int64_t)(i / n_grid);
+    int32_t g = (int32_t)(i % n_grid);
+    int64_t lo = (int64_t)goff[grp];
+    int64_t n_les = (int64_t)(goff[grp + 1] - lo);
+    double q = vm_dnan(), lb = vm_dnan(), ub = vm_dnan();
+    if (!vm_isnan(phi)) {
+      double fix_prev = 0;
+      double v_last = 0;
+      for (int64_t j = 0; j < n_les; j++) {
+        double v = bv[(size_t)(lo + j) * n_grid + g];
+        if (j == 0) fix_prev = vm_isnan(v) ? 0 : v;
+        else if (!(vm_isnan(v) || fix_prev > v)) fix_prev = v;
+        v_last = fix_prev;
+      }
+      if (v_last != 0) {
+        if (phi < 0) {
+          q = -vm_dinf();
+          lb = -vm_dinf();
+          double v0 = bv[(size_t)lo * n_grid + g];
+          ub = vm_isnan(v0) ? 0 : v0;
+        } else if (phi > 1) {
+          q = vm_dinf();
+          lb = v_last;
+          ub = vm_dinf();
+        } else {
+          double v_req = v_last * phi;
+          double vp = 0, lep = 0;
+          fix_prev = 0;
+          bool done = false;
+          for (int64_t j = 0; j < n_les && !done; j++) {
+            double raw = bv[(size_t)(lo + j) * n_grid + g];
+            double v;
+            if (j == 0) v = vm_isnan(raw) ? 0 : raw;
+            else v = (vm_isnan(raw) || fix_prev > raw) ? fix_prev : raw;
+            fix_prev = v;
+            double le = les[lo + j];
+            if (v <= 0) {
+              lep = le;
+              continue;
+            }
+            if (v < v_req) {
+              vp = v;
+              lep = le;
+              continue;
+            }
+            if (isinf(le)) break;
+            if (v == vp) {
+              q = lep;
+              lb = lep;
+              ub = v;
+              done = true;
+              break;
+            }
+            q = lep + (le - lep) * (v_req - vp) / (v - vp);
+            lb = lep;
+            ub = le;
+            done = true;
+          }
+          if (!done) {
+            double vv = vm_dnan();
+            for (int64_t j = n_les - 1; j >= 0; j--) {
+              if (!isinf(les[lo + j])) {
+                vv = les[lo + j];
+                break;
+              }
+            }
+            q = vv;
+            lb = vv;
+            ub = vm_dinf();
+          }
+        }
+      }
+    }
+    out[i] = q;
+    if (out_lo) out_lo[i] = lb;
+    if (out_hi) out_hi[i] = ub;
+  }
+}
+
 /* ------------------------------------------------------------------ */
 /* launch dispatch: compile-time specialization for hot funcs         */
 /* ------------------------------------------------------------------ */
@@ -828,6 +1182,9 @@ struct Batch {
   double* d_counts = nullptr;
   size_t count_elems = 0;
   unsigned long long* d_scanned = nullptr;
+  /* shape of the last exec's output */
+  uint32_t last_rows = 0;
+  int32_t last_grid = 0;
 };
 
 struct Ctx {
@@ -1128,6 +1485,8 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
   g_ctx.last_kernel_ms = (double)ms;
 
   if (out_samples_scanned) *out_samples_scanned = (uint64_t)scanned_h;
+  b.last_rows = grouped ? b.n_groups : b.n_series;
+  b.last_grid = n_grid;
   return 0;
 }
 
@@ -1196,6 +1555,235 @@ int vmgpu_rollup_eval(const vmgpu_plan* plan,
                          errbuf, errbuf_len);
   vmgpu_batch_destroy(h);
   return rc;
+}
+
+
+int vmgpu_topk_range(uint64_t handle, double k, int32_t summary_op,
+                     int32_t reverse, int64_t* out_sel, int64_t* out_n_sel,
+                     double* out_remaining, char* errbuf, size_t errbuf_len) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  if (!g_ctx.inited) return set_err(errbuf, errbuf_len, "vmgpu: not initialized");
+  auto it = g_ctx.batches.find(handle);
+  if (it == g_ctx.batches.end()) return set_err(errbuf, errbuf_len, "vmgpu: bad handle");
+  Batch& b = it->second;
+  if (!b.d_out || b.last_rows == 0)
+    return set_err(errbuf, errbuf_len, "vmgpu: no evaluated output on this batch");
+  if (summary_op == 3)
+    return set_err(errbuf, errbuf_len, "vmgpu: topk_median not implemented on device yet");
+  uint32_t n = b.last_rows;
+  int32_t n_grid = b.last_grid;
+  uint32_t kk = 0;
+  if (k == k && k > 0) kk = (k > (double)n) ? n : (uint32_t)k;
+  if (out_n_sel) *out_n_sel = 0;
+  if (kk == 0) return 0;
+
+  hipStream_t st = g_ctx.stream;
+  unsigned long long* d_keys = nullptr;
+  uint32_t* d_hist = nullptr;
+  uint32_t* d_misc = nullptr; /* counters */
+  uint32_t* d_sel = nullptr;
+  HIP_TRY(hipMalloc(&d_keys, (size_t)n * 8), "alloc keys");
+  HIP_TRY(hipMalloc(&d_hist, 65536 * 4), "alloc hist");
+  HIP_TRY(hipMalloc(&d_misc, 8), "alloc counters");
+  HIP_TRY(hipMalloc(&d_sel, (size_t)kk * 4), "alloc sel");
+  uint32_t blocks = std::min<uint32_t>((n + BLOCK_THREADS - 1) / BLOCK_THREADS * WAVES_PER_BLOCK, 2048);
+  hipLaunchKernelGGL(topk_summary_kernel, dim3(blocks), dim3(BLOCK_THREADS), 0, st,
+                     b.d_out, n, n_grid, summary_op, reverse, d_keys);
+  /* refine the k-th largest key threshold 16 bits at a time */
+  unsigned long long prefix = 0;
+  uint32_t need_above = kk; /* how many keys >= threshold so far required */
+  std::vector<uint32_t> hist(65536);
+  uint32_t above_total = 0;
+  for (int shift = 48; shift >= 0; shift -= 16) {
+    HIP_TRY(hipMemsetAsync(d_hist, 0, 65536 * 4, st), "zero hist");
+    uint32_t hblocks = std::min<uint32_t>((n + 255) / 256, 2048);
+    hipLaunchKernelGGL(topk_hist_kernel, dim3(hblocks), dim3(256), 0, st,
+                       d_keys, n, prefix, shift, d_hist);
+    HIP_TRY(hipMemcpyAsync(hist.data(), d_hist, 65536 * 4, hipMemcpyDeviceToHost, st), "dl hist");
+    HIP_TRY(hipStreamSynchronize(st), "sync hist");
+    uint32_t cum = 0;
+    int bin = 0;
+    for (int bb = 65535; bb >= 0; bb--) {
+      uint32_t c = hist[bb];
+      if (cum + c >= need_above - above_total) { bin = bb; break; }
+      cum += c;
+    }
+    above_total += cum;
+    prefix = (prefix << 16) | (unsigned long long)bin;
+  }
+  unsigned long long kstar = prefix;
+  uint32_t ties_quota = need_above - above_total;
+  HIP_TRY(hipMemsetAsync(d_misc, 0, 8, st), "zero counters");
+  uint32_t cblocks = std::min<uint32_t>((n + 255) / 256, 2048);
+  hipLaunchKernelGGL(topk_collect_kernel, dim3(cblocks), dim3(256), 0, st,
+                     d_keys, n, kstar, ties_quota, d_misc, d_sel, kk);
+  std::vector<uint32_t> sel(kk);
+  uint32_t nsel_h[2];
+  HIP_TRY(hipMemcpyAsync(sel.data(), d_sel, (size_t)kk * 4, hipMemcpyDeviceToHost, st), "dl sel");
+  HIP_TRY(hipMemcpyAsync(nsel_h, d_misc, 8, hipMemcpyDeviceToHost, st), "dl counters");
+  HIP_TRY(hipStreamSynchronize(st), "sync sel");
+  uint32_t m = std::min<uint32_t>(nsel_h[0], kk);
+  /* order the selection by key descending (reference output order after
+   * reverseSeries): download keys of the selected rows */
+  std::vector<unsigned long long> skeys(m);
+  for (uint32_t x = 0; x < m; x++) {
+    HIP_TRY(hipMemcpyAsync(&skeys[x], d_keys + sel[x], 8, hipMemcpyDeviceToHost, st), "dl key");
+  }
+  HIP_TRY(hipStreamSynchronize(st), "sync keys");
+  std::vector<uint32_t> order(m);
+  for (uint32_t x = 0; x < m; x++) order[x] = x;
+  std::sort(order.begin(), order.end(), [&](uint32_t a, uint32_t bx) {
+    if (skeys[a] != skeys[bx]) return skeys[a] > skeys[bx];
+    return sel[a] < sel[bx];
+  });
+  for (uint32_t x = 0; x < m; x++) out_sel[x] = (int64_t)sel[order[x]];
+  if (out_n_sel) *out_n_sel = (int64_t)m;
+
+  if (out_remaining) {
+    uint8_t* d_mask = nullptr;
+    double* d_rem = nullptr;
+    HIP_TRY(hipMalloc(&d_mask, n), "alloc mask");
+    HIP_TRY(hipMalloc(&d_rem, (size_t)n_grid * 16), "alloc rem");
+    HIP_TRY(hipMemsetAsync(d_mask, 0, n, st), "zero mask");
+    HIP_TRY(hipMemsetAsync(d_rem, 0, (size_t)n_grid * 16, st), "zero rem");
+    std::vector<uint8_t> mask(n, 0);
+    for (uint32_t x = 0; x < m; x++) mask[sel[x]] = 1;
+    HIP_TRY(hipMemcpyAsync(d_mask, mask.data(), n, hipMemcpyHostToDevice, st), "ul mask");
+    uint32_t rblocks = std::min<uint32_t>(
+        (uint32_t)(((size_t)n * n_grid + 255) / 256), 4096u);
+    hipLaunchKernelGGL(topk_remaining_kernel, dim3(rblocks), dim3(256), 0, st,
+                       b.d_out, n, n_grid, d_mask, d_rem, d_rem + n_grid);
+    std::vector<double> rem(2 * (size_t)n_grid);
+    HIP_TRY(hipMemcpyAsync(rem.data(), d_rem, (size_t)n_grid * 16, hipMemcpyDeviceToHost, st), "dl rem");
+    HIP_TRY(hipStreamSynchronize(st), "sync rem");
+    for (int32_t g = 0; g < n_grid; g++)
+      out_remaining[g] = (rem[n_grid + g] == 0) ? __builtin_nan("") : rem[g];
+    (void)hipFree(d_mask);
+    (void)hipFree(d_rem);
+  }
+  (void)hipFree(d_keys);
+  (void)hipFree(d_hist);
+  (void)hipFree(d_misc);
+  (void)hipFree(d_sel);
+  return 0;
+}
+
+int vmgpu_topk_pointwise(uint64_t handle, double k, int32_t reverse,
+                         double* out, char* errbuf, size_t errbuf_len) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  if (!g_ctx.inited) return set_err(errbuf, errbuf_len, "vmgpu: not initialized");
+  auto it = g_ctx.batches.find(handle);
+  if (it == g_ctx.batches.end()) return set_err(errbuf, errbuf_len, "vmgpu: bad handle");
+  Batch& b = it->second;
+  if (!b.d_out || b.last_rows == 0)
+    return set_err(errbuf, errbuf_len, "vmgpu: no evaluated output on this batch");
+  uint32_t n = b.last_rows;
+  int32_t n_grid = b.last_grid;
+  uint32_t kk = 0;
+  if (k == k && k > 0) kk = (k > (double)n) ? n : (uint32_t)k;
+  hipStream_t st = g_ctx.stream;
+  if (kk >= n || n == 0) {
+    if (out) {
+      HIP_TRY(hipMemcpyAsync(out, b.d_out, (size_t)n * n_grid * 8,
+                             hipMemcpyDeviceToHost, st), "dl out");
+      HIP_TRY(hipStreamSynchronize(st), "sync");
+    }
+    return 0;
+  }
+  const uint32_t CAND_CAP = 8192;
+  uint32_t* d_hists = nullptr;
+  uint32_t* d_bins = nullptr;
+  unsigned long long* d_cand = nullptr;
+  unsigned long long* d_kstar = nullptr;
+  uint32_t* d_small = nullptr; /* cand_n[n_grid], ties[n_grid], taken[n_grid], overflow */
+  HIP_TRY(hipMalloc(&d_hists, (size_t)n_grid * 65536 * 4), "alloc col hists");
+  HIP_TRY(hipMalloc(&d_bins, (size_t)n_grid * 8), "alloc bins");
+  HIP_TRY(hipMalloc(&d_cand, (size_t)n_grid * CAND_CAP * 8), "alloc cands");
+  HIP_TRY(hipMalloc(&d_kstar, (size_t)n_grid * 8), "alloc kstar");
+  HIP_TRY(hipMalloc(&d_small, (size_t)n_grid * 12 + 4), "alloc small");
+  uint32_t* d_cand_n = d_small;
+  uint32_t* d_ties = d_small + n_grid;
+  uint32_t* d_taken = d_small + 2 * (size_t)n_grid;
+  uint32_t* d_overflow = d_small + 3 * (size_t)n_grid;
+  HIP_TRY(hipMemsetAsync(d_hists, 0, (size_t)n_grid * 65536 * 4, st), "zero hists");
+  HIP_TRY(hipMemsetAsync(d_small, 0, (size_t)n_grid * 12 + 4, st), "zero small");
+  size_t total = (size_t)n * n_grid;
+  uint32_t eblocks = (uint32_t)std::min<size_t>((total + 255) / 256, 4096);
+  hipLaunchKernelGGL(topk_col_hist_kernel, dim3(eblocks), dim3(256), 0, st,
+                     b.d_out, n, n_grid, reverse, d_hists);
+  hipLaunchKernelGGL(topk_col_threshold_kernel,
+                     dim3((n_grid + 255) / 256), dim3(256), 0, st,
+                     d_hists, n_grid, kk, d_bins, d_bins + n_grid);
+  hipLaunchKernelGGL(topk_col_candidates_kernel, dim3(eblocks), dim3(256), 0, st,
+                     b.d_out, n, n_grid, reverse, d_bins, d_cand, d_cand_n,
+                     CAND_CAP, d_overflow);
+  uint32_t overflow_h = 0;
+  HIP_TRY(hipMemcpyAsync(&overflow_h, d_overflow, 4, hipMemcpyDeviceToHost, st), "dl ovf");
+  HIP_TRY(hipStreamSynchronize(st), "sync ovf");
+  if (overflow_h) {
+    (void)hipFree(d_hists); (void)hipFree(d_bins); (void)hipFree(d_cand);
+    (void)hipFree(d_kstar); (void)hipFree(d_small);
+    return set_err(errbuf, errbuf_len,
+                   "vmgpu: topk candidate overflow (massive ties); unsupported");
+  }
+  hipLaunchKernelGGL(topk_col_kstar_kernel, dim3(n_grid), dim3(256), 0, st,
+                     d_cand, d_cand_n, CAND_CAP, d_bins + n_grid, kk, n_grid,
+                     d_kstar, d_ties);
+  hipLaunchKernelGGL(topk_col_fill_kernel, dim3(eblocks), dim3(256), 0, st,
+                     b.d_out, n, n_grid, reverse, kk, d_bins, d_kstar, d_taken,
+                     d_ties);
+  if (out) {
+    HIP_TRY(hipMemcpyAsync(out, b.d_out, total * 8, hipMemcpyDeviceToHost, st), "dl out");
+  }
+  HIP_TRY(hipStreamSynchronize(st), "sync fill");
+  hipError_t kerr = hipGetLastError();
+  (void)hipFree(d_hists); (void)hipFree(d_bins); (void)hipFree(d_cand);
+  (void)hipFree(d_kstar); (void)hipFree(d_small);
+  if (kerr != hipSuccess) return hip_err(errbuf, errbuf_len, "topk kernel", kerr);
+  return 0;
+}
+
+int vmgpu_histogram_quantile(double phi, const double* bucket_values,
+                             const double* les, const uint64_t* group_offsets,
+                             uint32_t n_groups, int32_t n_grid,
+                             double* out, double* out_lower, double* out_upper,
+                             char* errbuf, size_t errbuf_len) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  if (!g_ctx.inited) return set_err(errbuf, errbuf_len, "vmgpu: not initialized");
+  if (!bucket_values || !les || !group_offsets || !out || n_groups == 0 || n_grid <= 0)
+    return set_err(errbuf, errbuf_len, "vmgpu: bad args");
+  hipStream_t st = g_ctx.stream;
+  uint64_t n_rows = group_offsets[n_groups];
+  double* d_bv = nullptr;
+  double* d_les = nullptr;
+  uint64_t* d_off = nullptr;
+  double* d_out = nullptr;
+  double* d_lo = nullptr;
+  double* d_hi = nullptr;
+  size_t out_elems = (size_t)n_groups * n_grid;
+  HIP_TRY(hipMalloc(&d_bv, (size_t)n_rows * n_grid * 8), "alloc bv");
+  HIP_TRY(hipMalloc(&d_les, (size_t)n_rows * 8), "alloc les");
+  HIP_TRY(hipMalloc(&d_off, (size_t)(n_groups + 1) * 8), "alloc off");
+  HIP_TRY(hipMalloc(&d_out, out_elems * 8), "alloc out");
+  if (out_lower) HIP_TRY(hipMalloc(&d_lo, out_elems * 8), "alloc lo");
+  if (out_upper) HIP_TRY(hipMalloc(&d_hi, out_elems * 8), "alloc hi");
+  HIP_TRY(hipMemcpyAsync(d_bv, bucket_values, (size_t)n_rows * n_grid * 8,
+                         hipMemcpyHostToDevice, st), "ul bv");
+  HIP_TRY(hipMemcpyAsync(d_les, les, (size_t)n_rows * 8, hipMemcpyHostToDevice, st), "ul les");
+  HIP_TRY(hipMemcpyAsync(d_off, group_offsets, (size_t)(n_groups + 1) * 8,
+                         hipMemcpyHostToDevice, st), "ul off");
+  uint32_t blocks = (uint32_t)std::min<size_t>((out_elems + 255) / 256, 4096);
+  hipLaunchKernelGGL(hq_kernel, dim3(blocks), dim3(256), 0, st, phi, d_bv, d_les,
+                     d_off, (int64_t)n_groups, n_grid, d_out, d_lo, d_hi);
+  HIP_TRY(hipMemcpyAsync(out, d_out, out_elems * 8, hipMemcpyDeviceToHost, st), "dl out");
+  if (out_lower) HIP_TRY(hipMemcpyAsync(out_lower, d_lo, out_elems * 8, hipMemcpyDeviceToHost, st), "dl lo");
+  if (out_upper) HIP_TRY(hipMemcpyAsync(out_upper, d_hi, out_elems * 8, hipMemcpyDeviceToHost, st), "dl hi");
+  HIP_TRY(hipStreamSynchronize(st), "sync hq");
+  hipError_t kerr = hipGetLastError();
+  (void)hipFree(d_bv); (void)hipFree(d_les); (void)hipFree(d_off);
+  (void)hipFree(d_out); (void)hipFree(d_lo); (void)hipFree(d_hi);
+  if (kerr != hipSuccess) return hip_err(errbuf, errbuf_len, "hq kernel", kerr);
+  return 0;
 }
 
 int vmgpu_last_kernel_ms(double* out_ms) {

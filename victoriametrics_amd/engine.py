@@ -277,6 +277,8 @@ class SeriesBatch:
         if rc != 0:
             raise VmGpuError(f"vmgpu_rollup_exec failed ({rc}): "
                              f"{errbuf.value.decode()}")
+        self._last_rows = self.n_groups if grouped else self.n_series
+        self._last_n_grid = n_grid
         return out, counts, scanned.value
 
     def fetch_out(self, rows, n_grid, with_counts=False):
@@ -312,6 +314,86 @@ def aggr_finalize(aggr, values, counts):
         c.ctypes.data_as(ctypes.POINTER(ctypes.c_double)) if c is not None else None,
         ctypes.c_uint64(v.size))
     return v
+
+
+TOPK_SUMMARY_OPS = {"avg": 0, "min": 1, "max": 2, "median": 3, "last": 4}
+
+
+def topk_range(batch, k, summary="avg", reverse=False, remaining=False):
+    """getRangeTopKTimeseries over the batch's last evaluated output:
+    returns (selected row ids in output order, remaining-sum row or None)."""
+    lib = _load_lib()
+    kk = max(int(k) if k == k and k > 0 else 0, 0)
+    sel = np.empty(max(kk, 1), dtype=np.int64)
+    n_sel = ctypes.c_int64(0)
+    rows = batch.n_groups if batch.group_ids is not None else batch.n_series
+    del rows
+    rem = None
+    rptr = None
+    if remaining:
+        n_grid = batch._last_n_grid
+        rem = np.empty(n_grid, dtype=np.float64)
+        rptr = rem.ctypes.data_as(ctypes.POINTER(ctypes.c_double))
+    errbuf = ctypes.create_string_buffer(256)
+    rc = lib.vmgpu_topk_range(
+        ctypes.c_uint64(batch.handle), ctypes.c_double(k),
+        ctypes.c_int32(TOPK_SUMMARY_OPS[summary]),
+        ctypes.c_int32(1 if reverse else 0),
+        sel.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+        ctypes.byref(n_sel), rptr, errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise VmGpuError(f"vmgpu_topk_range failed ({rc}): {errbuf.value.decode()}")
+    return sel[:n_sel.value], rem
+
+
+def topk_pointwise(batch, k, reverse=False):
+    """newAggrFuncTopK over the batch's last evaluated output: returns the
+    [rows x n_grid] matrix with all but the per-point top k NaN-filled.
+    NOTE: mutates the device-resident output."""
+    lib = _load_lib()
+    rows = batch._last_rows
+    n_grid = batch._last_n_grid
+    out = np.empty((rows, n_grid), dtype=np.float64)
+    errbuf = ctypes.create_string_buffer(256)
+    rc = lib.vmgpu_topk_pointwise(
+        ctypes.c_uint64(batch.handle), ctypes.c_double(k),
+        ctypes.c_int32(1 if reverse else 0),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise VmGpuError(f"vmgpu_topk_pointwise failed ({rc}): {errbuf.value.decode()}")
+    return out
+
+
+def histogram_quantile(phi, bucket_values, les, group_offsets, bounds=False):
+    """transformHistogramQuantile over grouped le-bucket rows (rows sorted by
+    (group, le), same-le pre-merged; the host label plumbing — vmrange->le
+    conversion and group-key assignment — happens above this call)."""
+    init()
+    lib = _load_lib()
+    bv = np.ascontiguousarray(bucket_values, dtype=np.float64)
+    le = np.ascontiguousarray(les, dtype=np.float64)
+    off = np.ascontiguousarray(group_offsets, dtype=np.uint64)
+    n_groups = len(off) - 1
+    n_grid = bv.shape[1]
+    out = np.empty((n_groups, n_grid), dtype=np.float64)
+    lo = np.empty((n_groups, n_grid), dtype=np.float64) if bounds else None
+    hi = np.empty((n_groups, n_grid), dtype=np.float64) if bounds else None
+    errbuf = ctypes.create_string_buffer(256)
+    rc = lib.vmgpu_histogram_quantile(
+        ctypes.c_double(phi),
+        bv.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        le.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        off.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        ctypes.c_uint32(n_groups), ctypes.c_int32(n_grid),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        lo.ctypes.data_as(ctypes.POINTER(ctypes.c_double)) if bounds else None,
+        hi.ctypes.data_as(ctypes.POINTER(ctypes.c_double)) if bounds else None,
+        errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise VmGpuError(f"vmgpu_histogram_quantile failed ({rc}): "
+                         f"{errbuf.value.decode()}")
+    return (out, lo, hi) if bounds else (out, None, None)
 
 
 def last_kernel_ms():
