@@ -552,3 +552,11 @@ def histogram_quantile(phi, bucket_values, les, group_offsets, bounds=False):
                             _ptr(lo, ctypes.c_double) if bounds else None,
                             _ptr(hi, ctypes.c_double) if bounds else None)
     return (out, lo, hi) if bounds else (out, None, None)
+
+
+def topk_summary(op, row):
+    l = lib()
+    l.vm_topk_summary.restype = ctypes.c_double
+    r = _f64(row)
+    return l.vm_topk_summary(ctypes.c_int32(TOPK_SUMMARY_OPS[op]),
+                             _ptr(r, ctypes.c_double), ctypes.c_int64(len(r)))

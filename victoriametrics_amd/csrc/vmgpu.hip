@@ -766,8 +766,8 @@ static VM_DEV unsigned long long vm_topk_key(double v, int reverse) {
   if (vm_isnan(v)) return reverse ? ~0ULL : 0ULL;
   unsigned long long b = (unsigned long long)__double_as_longlong(v);
   unsigned long long ord = (b >> 63) ? ~b : (b | 0x8000000000000000ULL);
-  /* compress into (0, ~0) so the NaN sentinels stay exclusive */
-  ord = (ord >> 1) | 0x2000000000000000ULL;
+  /* the NaN sentinels 0 and ~0 are unreachable for non-NaN doubles: their
+   * preimages under the order map are NaN bit patterns */
   return reverse ? ~ord : ord;
 }
 
@@ -1634,7 +1634,9 @@ int vmgpu_topk_range(uint64_t handle, double k, int32_t summary_op,
   for (uint32_t x = 0; x < m; x++) order[x] = x;
   std::sort(order.begin(), order.end(), [&](uint32_t a, uint32_t bx) {
     if (skeys[a] != skeys[bx]) return skeys[a] > skeys[bx];
-    return sel[a] < sel[bx];
+    /* ties: descending row id, matching the oracle's reversed ascending
+     * sort (the reference's own sort is unstable: tie order unspecified) */
+    return sel[a] > sel[bx];
   });
   for (uint32_t x = 0; x < m; x++) out_sel[x] = (int64_t)sel[order[x]];
   if (out_n_sel) *out_n_sel = (int64_t)m;
