@@ -277,16 +277,26 @@ def test_topk_range_gpu(engine, counter_small):
                 ref_sel, ref_rem = oracle.topk_range(host_out, 100, summary,
                                                      reverse=rev,
                                                      remaining=True)
-                assert set(sel) == set(ref_sel), f"{summary} rev={rev}"
-                # order must match by summary value; tie ids are a
-                # convention (the reference's sort is unstable on ties)
+                # the selected summary-value sequences must be identical;
+                # WHICH rows carry a boundary-tied value is unspecified in
+                # the reference too (unstable sort), so compare ids only
+                # away from the boundary value.
                 sv = [oracle.topk_summary(summary, host_out[i]) for i in sel]
                 rv = [oracle.topk_summary(summary, host_out[i]) for i in ref_sel]
                 assert sv == rv, f"{summary} rev={rev}: value order differs"
-                gn, rn = np.isnan(rem), np.isnan(ref_rem)
-                assert (gn == rn).all()
-                assert np.allclose(rem[~gn], ref_rem[~rn], rtol=1e-9, atol=0), \
-                    f"{summary} rev={rev} remaining"
+                boundary = sv[-1] if sv else None
+                ids_g = {i for i, v in zip(sel, sv) if v != boundary}
+                ids_r = {i for i, v in zip(ref_sel, rv) if v != boundary}
+                assert ids_g == ids_r, f"{summary} rev={rev}"
+                tied = sv.count(boundary) if sv else 0
+                all_vals = [oracle.topk_summary(summary, host_out[i])
+                            for i in range(host_out.shape[0])]
+                if tied == sum(1 for v in all_vals if v == boundary):
+                    # no boundary tie ambiguity: remaining sums comparable
+                    gn, rn = np.isnan(rem), np.isnan(ref_rem)
+                    assert (gn == rn).all()
+                    assert np.allclose(rem[~gn], ref_rem[~rn], rtol=1e-9,
+                                       atol=0), f"{summary} rev={rev} remaining"
 
 
 def test_topk_range_k_edge_cases(engine, counter_small):
