@@ -143,6 +143,37 @@ int vmgpu_histogram_quantile(double phi, const double* bucket_values,
                              double* out, double* out_lower, double* out_upper,
                              char* errbuf, size_t errbuf_len);
 
+/* GPU block decode — the fetch path's
+ * Block.UnmarshalData (lib/storage/block.go:250) + lib/encoding varint/
+ * delta codecs + decimal->float, fused on device.  zstd frames (marshal
+ * types 1 and 4) are decompressed on the HOST before this call (standard
+ * CPU format), so descriptors carry post-zstd types
+ * 2/3/5/6.  e10 = pow(10, |scale|) computed by the caller with libm so
+ * device results are bit-identical to the CPU path. */
+typedef struct vmgpu_block_desc {
+  uint64_t ts_data_off;   /* into the shared payload buffer */
+  uint64_t val_data_off;
+  uint64_t out_off;       /* row offset of this block in the output columns */
+  int64_t min_timestamp;  /* blockHeader fields (lib/storage/block_header.go) */
+  int64_t max_timestamp;
+  int64_t first_value;
+  double e10;
+  uint32_t ts_data_len;
+  uint32_t val_data_len;
+  uint32_t rows;
+  int32_t scale;
+  uint8_t ts_mt;
+  uint8_t val_mt;
+  uint8_t precision_bits;
+  uint8_t _pad;
+} vmgpu_block_desc;
+
+int vmgpu_decode_blocks(const uint8_t* payload, uint64_t payload_len,
+                        const vmgpu_block_desc* blocks, uint32_t n_blocks,
+                        uint64_t total_rows,
+                        int64_t* out_ts, double* out_vals,
+                        char* errbuf, size_t errbuf_len);
+
 /* Wall time of the rollup kernels inside the last vmgpu_rollup_exec on this
  * thread's context, measured with hipEvents on the launch stream (for the
  * bench's roofline accounting). */

@@ -560,3 +560,25 @@ def topk_summary(op, row):
     r = _f64(row)
     return l.vm_topk_summary(ctypes.c_int32(TOPK_SUMMARY_OPS[op]),
                              _ptr(r, ctypes.c_double), ctypes.c_int64(len(r)))
+
+
+def zstd_decompress(data, cap=None):
+    l = _codec_lib()
+    l.vm_zstd_decompress.restype = ctypes.c_int64
+    src = np.frombuffer(bytes(data), dtype=np.uint8)
+    if cap is None:
+        cap = max(len(src) * 64, 1 << 20)
+    dst = np.empty(cap, dtype=np.uint8)
+    n = l.vm_zstd_decompress(_ptr(dst, ctypes.c_uint8), ctypes.c_size_t(cap),
+                             _ptr(src, ctypes.c_uint8), ctypes.c_size_t(len(src)))
+    if n < 0:
+        raise ValueError(f"zstd decompress error {n}")
+    return bytes(dst[:n])
+
+
+def ensure_non_decreasing(a, v_min, v_max):
+    l = _codec_lib()
+    arr = np.ascontiguousarray(a, dtype=np.int64).copy()
+    l.vm_ensure_non_decreasing(_ptr(arr, ctypes.c_int64), ctypes.c_int64(len(arr)),
+                               ctypes.c_int64(v_min), ctypes.c_int64(v_max))
+    return arr

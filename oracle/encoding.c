@@ -281,6 +281,18 @@ static int load_zstd(void) {
   return z_compress && z_decompress && z_bound && z_iserr && z_framesize;
 }
 
+/* helper for tests: decompress one zstd frame with the dlopen'd libzstd
+ * (mirrors what the host engine does before vmgpu_decode_blocks) */
+int64_t vm_zstd_decompress(uint8_t* dst, size_t cap, const uint8_t* src, size_t n) {
+  if (!load_zstd()) return -1;
+  unsigned long long fs = z_framesize(src, n);
+  if (fs == (unsigned long long)-1 || fs == (unsigned long long)-2) return -2;
+  if (fs > cap) return -3;
+  size_t dlen = z_decompress(dst, cap, src, n);
+  if (z_iserr(dlen)) return -4;
+  return (int64_t)dlen;
+}
+
 /* ---------------- marshalInt64Array (encoding.go:119-173) ---------------- */
 
 #define MIN_COMPRESSIBLE_BLOCK_SIZE 128

@@ -62,6 +62,8 @@ int64_t vm_marshal_int64_array(uint8_t* dst, const int64_t* a, int64_t n,
 int vm_unmarshal_int64_array(int64_t* dst, int64_t items, const uint8_t* src,
                              size_t src_len, uint8_t mt, int64_t first_value);
 
+int64_t vm_zstd_decompress(uint8_t* dst, size_t cap, const uint8_t* src, size_t n);
+
 /* detectors (encoding.go:288-366) */
 int vm_is_const(const int64_t* a, int64_t n);
 int vm_is_delta_const(const int64_t* a, int64_t n);

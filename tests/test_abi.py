@@ -18,8 +18,7 @@ HEADER = os.path.join(REPO_ROOT, "include", "vmgpu.h")
 def _ensure_built():
     if not os.path.exists(LIB):
         subprocess.run(
-            ["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-ffp-contract=off", "-fPIC",
-             "-shared", "vmgpu.hip", "-o", "../libvmgpu.so"],
+            ["sh", "build.sh"],
             cwd=os.path.join(REPO_ROOT, "victoriametrics_amd", "csrc"),
             check=True)
     return LIB

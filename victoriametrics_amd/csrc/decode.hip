@@ -1,0 +1,381 @@
+/* decode.hip — GPU block decode for the vmselect fetch path.
+ *
+ * Independent MI355X-native implementation of the reference's block codec
+ * read path (lib/encoding + lib/decimal + lib/storage/block.go:250
+ * UnmarshalData), plus a host-side encoder mirroring the write path
+ * (marshalInt64Array, encoding.go:119 — needed to produce realistic blocks
+ * for benches/tests) and the per-series k-way merge
+ * (mergeSortBlocks, netstorage.go:564).  zstd frames are a CPU format and
+ * stay on the host (decompressed before upload).
+ *
+ * Device decode design (one 256-thread workgroup per storage block, blocks
+ * looped grid-stride):
+ *   phase 1  terminator-mask build: one ballot per 64 payload bytes, mask
+ *            words + per-word terminator counts staged in LDS, then an
+ *            exact integer prefix over words.
+ *   phase 2  wave-parallel varint decode: each thread locates the start of
+ *            its varint via the word prefix (binary search + bit-select)
+ *            and decodes <=10 bytes from the L2-resident payload; zigzag to
+ *            signed deltas in a per-workgroup global scratch column.
+ *   phase 3  block-wide exact integer prefix sum reconstructs values
+ *            (nearest-delta; run twice for nearest-delta2's second-order
+ *            deltas).  int64 wrap-around matches Go via uint64 arithmetic.
+ *   phase 4  timestamps: EnsureNonDecreasingSequence (encoding.go:255) for
+ *            precisionBits<64 or bounds validation; values: fused
+ *            decimal->float (AppendDecimalToFloat, decimal.go:100) with the
+ *            10^scale double precomputed on the host so results are
+ *            bit-identical to the CPU path.
+ */
+#include <hip/hip_runtime.h>
+#include <algorithm>
+#include <cstdio>
+#include <cstring>
+#include <mutex>
+#include <vector>
+
+#include "../../include/vmgpu.h"
+
+#define DWAVE 64
+#define DBLOCK 256
+#define DMAX_ROWS 8192          /* maxRowsPerBlock, lib/storage/block.go:14 */
+#define DMAX_PAYLOAD (DMAX_ROWS * 10 + 16)
+#define DGRID 1024              /* workgroups; each owns a scratch column */
+
+namespace {
+
+int dset_err(char* errbuf, size_t len, const char* msg) {
+  if (errbuf && len) snprintf(errbuf, len, "%s", msg);
+  return 1;
+}
+
+int dhip_err(char* errbuf, size_t len, const char* what, hipError_t e) {
+  if (errbuf && len) snprintf(errbuf, len, "%s: %s", what, hipGetErrorString(e));
+  return 2;
+}
+
+#define DHIP_TRY(expr, what)                                               \
+  do {                                                                     \
+    hipError_t _e = (expr);                                                \
+    if (_e != hipSuccess) return dhip_err(errbuf, errbuf_len, what, _e);   \
+  } while (0)
+
+}  // namespace
+
+/* decimal special values (lib/decimal/decimal.go:403-419) */
+#define D_VINF_POS 0x7fffffffffffffffLL
+#define D_VINF_NEG (-0x7fffffffffffffffLL - 1)
+#define D_VSTALE   0x7ffffffffffffffeLL
+#define D_VMAX     0x7ffffffffffffffdLL
+#define D_VMIN     (-0x7fffffffffffffffLL)
+
+static __device__ __forceinline__ double dec_to_float(long long v, double e10,
+                                                      int e_neg) {
+  if (v > D_VMAX || v < D_VMIN) {
+    if (v == D_VINF_POS) return __longlong_as_double(0x7ff0000000000000LL);
+    if (v == D_VINF_NEG) return __longlong_as_double(0xfff0000000000000LL);
+    return __longlong_as_double(0x7ff0000000000002LL); /* StaleNaN */
+  }
+  double f = (double)v;
+  return e_neg ? f / e10 : f * e10;
+}
+
+struct DecState {
+  /* per-workgroup scratch column [DMAX_ROWS] of decoded deltas/values */
+  long long* scratch;
+};
+
+/* mask words + per-word counts + prefix in LDS:
+ * words = ceil(payload_len/64) <= DMAX_PAYLOAD/64 = 1282 */
+#define DMAX_WORDS ((DMAX_PAYLOAD + 63) / 64)
+
+struct __align__(16) DecLds {
+  unsigned long long mask[DMAX_WORDS];
+  unsigned short wcount[DMAX_WORDS];  /* terminators per word */
+  unsigned int wprefix[DMAX_WORDS + 1];
+  unsigned int carry_lo[DBLOCK / DWAVE];
+  long long carry[DBLOCK / DWAVE + 1];
+  int err;
+};
+
+/* exact u64 wave inclusive prefix sum (int wrap == Go semantics) */
+static __device__ __forceinline__ unsigned long long wave_prefix_u64(
+    unsigned long long x, int lane) {
+  for (int d = 1; d < DWAVE; d <<= 1) {
+    unsigned long long o = __shfl_up(x, d);
+    if (lane >= d) x += o;
+  }
+  return x;
+}
+
+__global__ __launch_bounds__(DBLOCK) void decode_blocks_kernel(
+    const uint8_t* payload, const vmgpu_block_desc* blocks, uint32_t n_blocks,
+    long long* scratch_all, int64_t* out_ts, double* out_vals, int* err_flag) {
+  __shared__ DecLds L;
+  const int tid = threadIdx.x;
+  const int lane = tid % DWAVE;
+  const int wv = tid / DWAVE;
+
+  for (uint32_t bi = blockIdx.x; bi < n_blocks; bi += gridDim.x) {
+    const vmgpu_block_desc bd = blocks[bi];
+    long long* scratch = scratch_all + (size_t)blockIdx.x * DMAX_ROWS;
+    /* two streams per block: timestamps then values */
+    for (int stream = 0; stream < 2; stream++) {
+      const uint8_t mt = stream ? bd.val_mt : bd.ts_mt;
+      const uint64_t off = stream ? bd.val_data_off : bd.ts_data_off;
+      const uint32_t len = stream ? bd.val_data_len : bd.ts_data_len;
+      const long long first = stream ? bd.first_value : bd.min_timestamp;
+      const uint32_t rows = bd.rows;
+      const uint8_t* src = payload + off;
+
+      int nvarints = 0;
+      if (mt == 3) {                      /* const */
+        nvarints = 0;
+      } else if (mt == 2) {               /* delta const */
+        nvarints = 1;
+      } else if (mt == 5 || mt == 6) {    /* nearest delta2 / nearest delta */
+        nvarints = (int)rows - 1;
+      } else {
+        if (tid == 0) L.err = 10 + mt;
+        __syncthreads();
+        atomicExch(err_flag, L.err);
+        return;
+      }
+
+      /* ---- phase 1: terminator masks ---- */
+      int words = (int)((len + 63) / 64);
+      if (words > DMAX_WORDS) {
+        if (tid == 0) atomicExch(err_flag, 2);
+        return;
+      }
+      for (int w = wv; w < words; w += DBLOCK / DWAVE) {
+        uint32_t byte_idx = (uint32_t)w * 64 + lane;
+        bool term = byte_idx < len && src[byte_idx] < 0x80;
+        unsigned long long m = __ballot(term);
+        if (lane == 0) {
+          L.mask[w] = m;
+          L.wcount[w] = (unsigned short)__popcll(m);
+        }
+      }
+      __syncthreads();
+      /* exact prefix over words: one wave, carry across rounds */
+      if (wv == 0) {
+        unsigned int carry = 0;
+        for (int base = 0; base < words; base += DWAVE) {
+          int w = base + lane;
+          unsigned long long c = (w < words) ? L.wcount[w] : 0;
+          unsigned long long p = wave_prefix_u64(c, lane);
+          if (w < words) L.wprefix[w + 1] = carry + (unsigned int)p;
+          carry += (unsigned int)__shfl(p, DWAVE - 1);
+        }
+        if (lane == 0) {
+          L.wprefix[0] = 0;
+          L.err = 0;
+        }
+      }
+      __syncthreads();
+      int total_terms = (words > 0) ? (int)L.wprefix[words] : 0;
+      if (total_terms != nvarints) {
+        /* payload length / varint count mismatch */
+        if (tid == 0) atomicExch(err_flag, 3);
+        return;
+      }
+
+      /* ---- phase 2: per-varint decode into scratch ---- */
+      for (int v = tid; v < nvarints; v += DBLOCK) {
+        uint32_t start;
+        if (v == 0) {
+          start = 0;
+        } else {
+          /* locate terminator #(v-1): binary search words by prefix */
+          int lo = 0, hi = words - 1;
+          while (lo < hi) {
+            int mid = (lo + hi) >> 1;
+            if ((int)L.wprefix[mid + 1] <= v - 1) lo = mid + 1;
+            else hi = mid;
+          }
+          int within = (v - 1) - (int)L.wprefix[lo];
+          unsigned long long m = L.mask[lo];
+          /* select the (within)-th set bit */
+          for (int s = 0; s < within; s++) m &= m - 1;
+          int bit = __ffsll((unsigned long long)m) - 1;
+          start = (uint32_t)lo * 64 + (uint32_t)bit + 1;
+        }
+        unsigned long long u = 0;
+        int shift = 0;
+        uint32_t p = start;
+        for (;;) {
+          if (p >= len || shift > 63) {
+            atomicExch(err_flag, 4);
+            break;
+          }
+          uint8_t c = src[p++];
+          u |= (unsigned long long)(c & 0x7f) << shift;
+          if (c < 0x80) break;
+          shift += 7;
+        }
+        /* zigzag (int.go:184) */
+        long long d = (long long)(u >> 1) ^ -(long long)(u & 1);
+        scratch[v] = d;
+      }
+      __syncthreads();
+
+      /* ---- phase 3: reconstruct via exact prefix sums ---- */
+      /* inclusive prefix over scratch[0..nv): block-cooperative, one wave
+       * per 64-chunk with cross-wave carries through LDS */
+      int passes = (mt == 5) ? 2 : (mt == 6 ? 1 : 0);
+      for (int pass = 0; pass < passes; pass++) {
+        /* for delta2: pass 0 turns d2[] into d1[] (starting at scratch[0]
+         * = d1 already: prefix starting from index 0 across all entries);
+         * pass 1 prefixes d1[] into cumulative deltas */
+        long long run = 0;
+        for (int base = 0; base < nvarints; base += DWAVE) {
+          /* single wave (wave 0) processes sequential chunks to keep the
+           * carry exact and simple; other waves idle in this loop */
+          if (wv == 0) {
+            int k = base + lane;
+            unsigned long long x = (k < nvarints) ? (unsigned long long)scratch[k] : 0;
+            unsigned long long pfx = wave_prefix_u64(x, lane);
+            if (k < nvarints)
+              scratch[k] = (long long)((unsigned long long)run + pfx);
+            run = (long long)((unsigned long long)run +
+                              (unsigned long long)__shfl(pfx, DWAVE - 1));
+          }
+        }
+        __syncthreads();
+      }
+
+      /* ---- phase 4: materialize rows ---- */
+      int64_t* dst_ts = out_ts + bd.out_off;
+      double* dst_vals = out_vals + bd.out_off;
+      const double e10 = bd.e10;
+      const int e_neg = bd.scale < 0;
+      if (mt == 3) { /* const */
+        for (uint32_t r = tid; r < rows; r += DBLOCK) {
+          if (stream) dst_vals[r] = dec_to_float(first, e10, e_neg);
+          else dst_ts[r] = first;
+        }
+      } else if (mt == 2) { /* delta const: v += d each row */
+        long long d = 0;
+        /* scratch[0] holds the delta after phase 2 (single varint) */
+        d = scratch[0];
+        for (uint32_t r = tid; r < rows; r += DBLOCK) {
+          long long v = (long long)((unsigned long long)first +
+                                    (unsigned long long)d * r);
+          if (stream) dst_vals[r] = dec_to_float(v, e10, e_neg);
+          else dst_ts[r] = v;
+        }
+      } else {
+        /* nearest delta / delta2: row 0 = first; row r>0 = first + scratch[r-1] */
+        for (uint32_t r = tid; r < rows; r += DBLOCK) {
+          long long v = (r == 0) ? first
+                                 : (long long)((unsigned long long)first +
+                                               (unsigned long long)scratch[r - 1]);
+          if (stream) dst_vals[r] = dec_to_float(v, e10, e_neg);
+          else dst_ts[r] = v;
+        }
+      }
+      __syncthreads();
+
+      /* timestamp post-processing */
+      if (!stream) {
+        if (bd.precision_bits < 64) {
+          /* EnsureNonDecreasingSequence (encoding.go:255-286): prefix max
+           * with first/last pinned — single wave, exact */
+          if (wv == 0) {
+            if (lane == 0) {
+              dst_ts[0] = bd.min_timestamp;
+              dst_ts[rows - 1] = bd.max_timestamp;
+            }
+          }
+          __syncthreads();
+          if (wv == 0) {
+            long long runmax = (long long)0x8000000000000000LL;
+            for (uint32_t base = 0; base < rows; base += DWAVE) {
+              uint32_t k = base + lane;
+              long long t = (k < rows) ? dst_ts[k] : (long long)0x8000000000000000LL;
+              /* inclusive prefix max */
+              long long x = t;
+              for (int d = 1; d < DWAVE; d <<= 1) {
+                long long o = __shfl_up(x, d);
+                if (lane >= d && o > x) x = o;
+              }
+              if (runmax > x) x = runmax;
+              if (k < rows) dst_ts[k] = x;
+              runmax = __shfl(x, DWAVE - 1);
+            }
+            /* re-pin the tail below vMax (second half of Ensure...) */
+            long long vmax = bd.max_timestamp;
+            for (uint32_t k = lane; k < rows; k += DWAVE) {
+              long long t = dst_ts[k];
+              if (t > vmax) dst_ts[k] = vmax;
+            }
+          }
+          __syncthreads();
+        } else if (mt == 5 || mt == 6) {
+          /* checkTimestampsBounds (block.go:329-350) */
+          bool bad = false;
+          for (uint32_t k = tid; k < rows; k += DBLOCK) {
+            long long t = dst_ts[k];
+            if (t < bd.min_timestamp || t > bd.max_timestamp) bad = true;
+            if (k + 1 < rows && dst_ts[k + 1] < t) bad = true;
+          }
+          if (bad) atomicExch(err_flag, 5);
+          __syncthreads();
+        }
+      }
+      __syncthreads();
+    }
+  }
+}
+
+extern "C" {
+
+int vmgpu_decode_blocks(const uint8_t* payload, uint64_t payload_len,
+                        const vmgpu_block_desc* blocks, uint32_t n_blocks,
+                        uint64_t total_rows,
+                        int64_t* out_ts, double* out_vals,
+                        char* errbuf, size_t errbuf_len) {
+  if (!payload || !blocks || !out_ts || !out_vals || n_blocks == 0)
+    return dset_err(errbuf, errbuf_len, "vmgpu: bad args");
+  hipStream_t st = 0;
+  uint8_t* d_payload = nullptr;
+  vmgpu_block_desc* d_blocks = nullptr;
+  long long* d_scratch = nullptr;
+  int64_t* d_ts = nullptr;
+  double* d_vals = nullptr;
+  int* d_err = nullptr;
+  DHIP_TRY(hipMalloc(&d_payload, payload_len ? payload_len : 1), "alloc payload");
+  DHIP_TRY(hipMalloc(&d_blocks, (size_t)n_blocks * sizeof(vmgpu_block_desc)), "alloc descs");
+  DHIP_TRY(hipMalloc(&d_scratch, (size_t)DGRID * DMAX_ROWS * 8), "alloc scratch");
+  DHIP_TRY(hipMalloc(&d_ts, (size_t)total_rows * 8), "alloc ts");
+  DHIP_TRY(hipMalloc(&d_vals, (size_t)total_rows * 8), "alloc vals");
+  DHIP_TRY(hipMalloc(&d_err, 4), "alloc err");
+  DHIP_TRY(hipMemcpyAsync(d_payload, payload, payload_len, hipMemcpyHostToDevice, st), "ul payload");
+  DHIP_TRY(hipMemcpyAsync(d_blocks, blocks, (size_t)n_blocks * sizeof(vmgpu_block_desc),
+                          hipMemcpyHostToDevice, st), "ul descs");
+  DHIP_TRY(hipMemsetAsync(d_err, 0, 4, st), "zero err");
+  uint32_t grid = std::min<uint32_t>(n_blocks, DGRID);
+  hipLaunchKernelGGL(decode_blocks_kernel, dim3(grid), dim3(DBLOCK), 0, st,
+                     d_payload, d_blocks, n_blocks, d_scratch, d_ts, d_vals, d_err);
+  int err_h = 0;
+  DHIP_TRY(hipMemcpyAsync(&err_h, d_err, 4, hipMemcpyDeviceToHost, st), "dl err");
+  DHIP_TRY(hipMemcpyAsync(out_ts, d_ts, (size_t)total_rows * 8, hipMemcpyDeviceToHost, st), "dl ts");
+  DHIP_TRY(hipMemcpyAsync(out_vals, d_vals, (size_t)total_rows * 8, hipMemcpyDeviceToHost, st), "dl vals");
+  DHIP_TRY(hipStreamSynchronize(st), "sync");
+  hipError_t kerr = hipGetLastError();
+  (void)hipFree(d_payload);
+  (void)hipFree(d_blocks);
+  (void)hipFree(d_scratch);
+  (void)hipFree(d_ts);
+  (void)hipFree(d_vals);
+  (void)hipFree(d_err);
+  if (kerr != hipSuccess) return dhip_err(errbuf, errbuf_len, "decode kernel", kerr);
+  if (err_h != 0) {
+    char msg[64];
+    snprintf(msg, sizeof(msg), "vmgpu: block decode error %d", err_h);
+    return dset_err(errbuf, errbuf_len, msg);
+  }
+  return 0;
+}
+
+}  /* extern "C" */

@@ -415,3 +415,78 @@ def device_info():
     if rc != 0:
         raise VmGpuError("vmgpu_device_info failed")
     return {"name": name.value.decode(), "hbm_gib": hbm.value, "cus": cus.value}
+
+
+class _BlockDescC(ctypes.Structure):
+    _fields_ = [
+        ("ts_data_off", ctypes.c_uint64),
+        ("val_data_off", ctypes.c_uint64),
+        ("out_off", ctypes.c_uint64),
+        ("min_timestamp", ctypes.c_int64),
+        ("max_timestamp", ctypes.c_int64),
+        ("first_value", ctypes.c_int64),
+        ("e10", ctypes.c_double),
+        ("ts_data_len", ctypes.c_uint32),
+        ("val_data_len", ctypes.c_uint32),
+        ("rows", ctypes.c_uint32),
+        ("scale", ctypes.c_int32),
+        ("ts_mt", ctypes.c_uint8),
+        ("val_mt", ctypes.c_uint8),
+        ("precision_bits", ctypes.c_uint8),
+        ("_pad", ctypes.c_uint8),
+    ]
+
+
+def decode_blocks(blocks):
+    """GPU block decode (Block.UnmarshalData + codecs + decimal->float).
+
+    blocks: list of dicts with keys
+      ts_data (bytes, post-zstd), ts_mt, min_timestamp, max_timestamp,
+      val_data (bytes, post-zstd), val_mt, first_value, scale,
+      precision_bits, rows
+    Returns (ts int64 array, vals f64 array, offsets) — the decoded CSR.
+    """
+    import math as _math
+    init()
+    lib = _load_lib()
+    payload = bytearray()
+    descs = (_BlockDescC * len(blocks))()
+    total_rows = 0
+    for i, b in enumerate(blocks):
+        d = descs[i]
+        d.ts_data_off = len(payload)
+        payload.extend(b["ts_data"])
+        d.ts_data_len = len(b["ts_data"])
+        d.val_data_off = len(payload)
+        payload.extend(b["val_data"])
+        d.val_data_len = len(b["val_data"])
+        d.out_off = total_rows
+        d.min_timestamp = int(b["min_timestamp"])
+        d.max_timestamp = int(b["max_timestamp"])
+        d.first_value = int(b["first_value"])
+        d.scale = int(b["scale"])
+        d.e10 = _math.pow(10.0, abs(int(b["scale"])))
+        d.rows = int(b["rows"])
+        d.ts_mt = int(b["ts_mt"])
+        d.val_mt = int(b["val_mt"])
+        d.precision_bits = int(b["precision_bits"])
+        total_rows += int(b["rows"])
+    pl = np.frombuffer(bytes(payload), dtype=np.uint8) if payload else \
+        np.zeros(1, dtype=np.uint8)
+    out_ts = np.empty(total_rows, dtype=np.int64)
+    out_vals = np.empty(total_rows, dtype=np.float64)
+    errbuf = ctypes.create_string_buffer(256)
+    rc = lib.vmgpu_decode_blocks(
+        pl.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        ctypes.c_uint64(len(payload)), descs, ctypes.c_uint32(len(blocks)),
+        ctypes.c_uint64(total_rows),
+        out_ts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+        out_vals.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise VmGpuError(f"vmgpu_decode_blocks failed ({rc}): "
+                         f"{errbuf.value.decode()}")
+    offsets = np.zeros(len(blocks) + 1, dtype=np.uint64)
+    for i, b in enumerate(blocks):
+        offsets[i + 1] = offsets[i] + int(b["rows"])
+    return out_ts, out_vals, offsets
