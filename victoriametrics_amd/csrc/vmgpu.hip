@@ -382,15 +382,19 @@ static VM_DEV void vm_emit_value(const KPlan& p, const KIO& io, uint32_t s,
   if (grp >= 0 && !vm_isnan(v)) {
     double* gv = io.out + (size_t)grp * (size_t)p.n_grid + (size_t)g;
     double* gc = io.out_counts + (size_t)grp * (size_t)p.n_grid + (size_t)g;
+    /* counts are true contribution counts only for avg (finalize divides);
+     * every other op uses them as a presence gate, where an idempotent
+     * plain store of 1.0 replaces the second atomic (cross-shard merge
+     * still works: the all-reduce SUM of flags stays nonzero) */
     switch (p.aggr) {
-      case VMGPU_AGGR_SUM:
+      case VMGPU_AGGR_SUM: atomicAdd(gv, v); *gc = 1.0; break;
       case VMGPU_AGGR_AVG: atomicAdd(gv, v); atomicAdd(gc, 1.0); break;
-      case VMGPU_AGGR_MIN: vm_atomic_min_f64(gv, v); atomicAdd(gc, 1.0); break;
-      case VMGPU_AGGR_MAX: vm_atomic_max_f64(gv, v); atomicAdd(gc, 1.0); break;
+      case VMGPU_AGGR_MIN: vm_atomic_min_f64(gv, v); *gc = 1.0; break;
+      case VMGPU_AGGR_MAX: vm_atomic_max_f64(gv, v); *gc = 1.0; break;
       case VMGPU_AGGR_COUNT:
-      case VMGPU_AGGR_GROUP: atomicAdd(gv, 1.0); atomicAdd(gc, 1.0); break;
-      case VMGPU_AGGR_SUM2: atomicAdd(gv, v * v); atomicAdd(gc, 1.0); break;
-      case VMGPU_AGGR_GEOMEAN: vm_atomic_mul_f64(gv, v); atomicAdd(gc, 1.0); break;
+      case VMGPU_AGGR_GROUP: atomicAdd(gv, 1.0); *gc = 1.0; break;
+      case VMGPU_AGGR_SUM2: atomicAdd(gv, v * v); *gc = 1.0; break;
+      case VMGPU_AGGR_GEOMEAN: vm_atomic_mul_f64(gv, v); *gc = 1.0; break;
       default: break;
     }
   }
