@@ -68,6 +68,7 @@ typedef struct vmgpu_plan {
                                    * caller can all-reduce across shards first
                                    *  */
   double  arg;                    /* phi / le / gt / eq / secs for arg funcs */
+  double  arg2;                   /* second scalar (holt_winters tf) */
 } vmgpu_plan;
 
 /* Select the device and create the library context.  One process drives one

@@ -81,6 +81,13 @@ FUNC_IDS = {
     "mode_over_time": 60,
     "duration_over_time": 61,
     "outlier_iqr_over_time": 62,
+    "rollup_open": 63,
+    "rollup_close": 64,
+    "rollup_low": 65,
+    "rollup_high": 66,
+    "holt_winters": 67,
+    "hoeffding_bound_lower": 68,
+    "hoeffding_bound_upper": 69,
     # aliases (same implementations as in the reference's map)
     "increase_prometheus": 4,   # rollupDeltaPrometheus
     "timestamp": 31,            # rollupTlast
@@ -109,6 +116,9 @@ CAN_ADJUST_WINDOW_FUNCS = {
     "default_rollup", "deriv", "deriv_fast", "ideriv", "irate", "rate",
     "rate_over_sum", "rollup", "rollup_candlestick", "rollup_deriv",
     "rollup_rate", "rollup_scrape_interval", "scrape_interval", "timestamp",
+    # the four per-config funcs getRollupConfigs derives from
+    # rollup_candlestick (rollup.go:454-475) inherit its MayAdjustWindow
+    "rollup_open", "rollup_close", "rollup_low", "rollup_high",
 }
 
 AGGR_IDS = {
@@ -130,6 +140,7 @@ class RollupConfigC(ctypes.Structure):
         ("is_default_rollup", ctypes.c_int32),
         ("samples_scanned_per_call", ctypes.c_int32),
         ("arg", ctypes.c_double),
+        ("arg2", ctypes.c_double),
     ]
 
 
@@ -232,7 +243,7 @@ def quantile(phi, values):
 def call_rollup_fn(func_name, values, timestamps, prev_value=float("nan"),
                    prev_timestamp=0, real_prev_value=float("nan"),
                    real_next_value=float("nan"), curr_timestamp=0, idx=0,
-                   window=0, arg=0.0):
+                   window=0, arg=0.0, arg2=0.0):
     v = _f64(values)
     t = _i64(timestamps)
     return lib().vm_call_rollup_fn(
@@ -241,12 +252,12 @@ def call_rollup_fn(func_name, values, timestamps, prev_value=float("nan"),
         _ptr(t, ctypes.c_int64), ctypes.c_int64(len(v)),
         ctypes.c_double(real_prev_value), ctypes.c_double(real_next_value),
         ctypes.c_int64(curr_timestamp), ctypes.c_int64(idx),
-        ctypes.c_int64(window), ctypes.c_double(arg))
+        ctypes.c_int64(window), ctypes.c_double(arg), ctypes.c_double(arg2))
 
 
 def make_config(func_name, start, end, step, window=0, lookback_delta=0,
-                min_staleness_interval=0, arg=0.0, may_adjust_window=None,
-                samples_scanned_per_call=None):
+                min_staleness_interval=0, arg=0.0, arg2=0.0,
+                may_adjust_window=None, samples_scanned_per_call=None):
     """Build a RollupConfigC the way getRollupConfigs does (rollup.go:374-516)
     when may_adjust_window/samples_scanned_per_call are None; pass explicit
     values (e.g. 0) to mirror the reference's direct rollupConfig{} literals in
@@ -263,7 +274,7 @@ def make_config(func_name, start, end, step, window=0, lookback_delta=0,
         min_staleness_interval=min_staleness_interval,
         is_default_rollup=1 if func_name == "default_rollup" else 0,
         samples_scanned_per_call=samples_scanned_per_call,
-        arg=arg)
+        arg=arg, arg2=arg2)
 
 
 def rollup_do(rc, values, timestamps):

@@ -220,7 +220,8 @@ typedef struct {
   int64_t curr_timestamp;
   int64_t idx;
   int64_t window;
-  double arg; /* phi / le / gt / eq / secs for parameterized funcs */
+  double arg;  /* phi / le / gt / eq / secs for parameterized funcs */
+  double arg2; /* holt_winters tf */
 } vm_rfa;
 
 /* ---------- individual rollup funcs ---------- */
@@ -925,6 +926,113 @@ static double fn_outlier_iqr(const vm_rfa* rfa) {
   return NAN_V;
 }
 
+/* candlestick helpers (rollup.go:2206-2291): the window is
+ * (currTimestamp-window, currTimestamp) EXCLUSIVE of currTimestamp, and the
+ * pre-window sample participates when it falls inside the window. */
+static int64_t candlestick_len(const vm_rfa* rfa) {
+  int64_t n = rfa->nvalues;
+  while (n > 0 && rfa->timestamps[n - 1] >= rfa->curr_timestamp) n--;
+  return n;
+}
+static double candlestick_first(const vm_rfa* rfa) {
+  if (rfa->prev_timestamp + rfa->window >= rfa->curr_timestamp) return rfa->prev_value;
+  return NAN_V;
+}
+static double fn_open(const vm_rfa* rfa) {
+  double v = candlestick_first(rfa);
+  if (!isnan(v)) return v;
+  int64_t n = candlestick_len(rfa);
+  if (n == 0) return NAN_V;
+  return rfa->values[0];
+}
+static double fn_close(const vm_rfa* rfa) {
+  int64_t n = candlestick_len(rfa);
+  if (n == 0) return candlestick_first(rfa);
+  return rfa->values[n - 1];
+}
+static double fn_high(const vm_rfa* rfa) {
+  int64_t n = candlestick_len(rfa);
+  const double* values = rfa->values;
+  double maxv = candlestick_first(rfa);
+  int64_t i = 0;
+  if (isnan(maxv)) {
+    if (n == 0) return NAN_V;
+    maxv = values[0];
+    i = 1;
+  }
+  for (; i < n; i++)
+    if (values[i] > maxv) maxv = values[i];
+  return maxv;
+}
+static double fn_low(const vm_rfa* rfa) {
+  int64_t n = candlestick_len(rfa);
+  const double* values = rfa->values;
+  double minv = candlestick_first(rfa);
+  int64_t i = 0;
+  if (isnan(minv)) {
+    if (n == 0) return NAN_V;
+    minv = values[0];
+    i = 1;
+  }
+  for (; i < n; i++)
+    if (values[i] < minv) minv = values[i];
+  return minv;
+}
+
+/* newRollupHoltWinters (rollup.go:1030-1077); arg=sf, arg2=tf */
+static double fn_holt_winters(const vm_rfa* rfa) {
+  const double* values = rfa->values;
+  int64_t n = rfa->nvalues;
+  if (n == 0) return NAN_V;
+  double sf = rfa->arg;
+  if (sf < 0 || sf > 1) return NAN_V;
+  double tf = rfa->arg2;
+  if (tf < 0 || tf > 1) return NAN_V;
+  double s0 = rfa->prev_value;
+  if (isnan(s0)) {
+    s0 = values[0];
+    values++;
+    n--;
+    if (n == 0) return s0;
+  }
+  double b0 = values[0] - s0;
+  for (int64_t i = 0; i < n; i++) {
+    double v = values[i];
+    double s1 = sf * v + (1 - sf) * (s0 + b0);
+    double b1 = tf * (s1 - s0) + (1 - tf) * b0;
+    s0 = s1;
+    b0 = b1;
+  }
+  return s0;
+}
+
+/* rollupHoeffdingBoundInternal (rollup.go:1353-1381); arg=phi */
+static void hoeffding_internal(const vm_rfa* rfa, double* out_bound, double* out_avg) {
+  int64_t n = rfa->nvalues;
+  if (n == 0) { *out_bound = NAN_V; *out_avg = NAN_V; return; }
+  if (n == 1) { *out_bound = 0; *out_avg = rfa->values[0]; return; }
+  double v_max = fn_max(rfa);
+  double v_min = fn_min(rfa);
+  double v_avg = fn_avg(rfa);
+  double v_range = v_max - v_min;
+  if (v_range <= 0) { *out_bound = 0; *out_avg = v_avg; return; }
+  double phi = rfa->arg;
+  if (phi >= 1) { *out_bound = INFINITY; *out_avg = v_avg; return; }
+  if (phi <= 0) { *out_bound = 0; *out_avg = v_avg; return; }
+  *out_bound = v_range * sqrt(log(1 / (1 - phi)) / (2 * (double)n));
+  *out_avg = v_avg;
+}
+static double fn_hoeffding_lower(const vm_rfa* rfa) {
+  double b, a;
+  hoeffding_internal(rfa, &b, &a);
+  return a - b;
+}
+static double fn_hoeffding_upper(const vm_rfa* rfa) {
+  double b, a;
+  hoeffding_internal(rfa, &b, &a);
+  return a + b;
+}
+
 typedef double (*vm_rollup_fn)(const vm_rfa*);
 
 static vm_rollup_fn fn_table(int32_t func) {
@@ -992,6 +1100,13 @@ static vm_rollup_fn fn_table(int32_t func) {
     case VM_FN_MODE: return fn_mode;
     case VM_FN_DURATION: return fn_duration;
     case VM_FN_OUTLIER_IQR: return fn_outlier_iqr;
+    case VM_FN_OPEN: return fn_open;
+    case VM_FN_CLOSE: return fn_close;
+    case VM_FN_LOW: return fn_low;
+    case VM_FN_HIGH: return fn_high;
+    case VM_FN_HOLT_WINTERS: return fn_holt_winters;
+    case VM_FN_HOEFFDING_LOWER: return fn_hoeffding_lower;
+    case VM_FN_HOEFFDING_UPPER: return fn_hoeffding_upper;
     default: return NULL;
   }
 }
@@ -999,7 +1114,8 @@ static vm_rollup_fn fn_table(int32_t func) {
 double vm_call_rollup_fn(int32_t func, double prev_value, int64_t prev_timestamp,
                          const double* values, const int64_t* timestamps, int64_t n,
                          double real_prev_value, double real_next_value,
-                         int64_t curr_timestamp, int64_t idx, int64_t window, double arg) {
+                         int64_t curr_timestamp, int64_t idx, int64_t window, double arg,
+                         double arg2) {
   vm_rollup_fn f = fn_table(func);
   if (!f) return NAN_V;
   vm_rfa rfa;
@@ -1014,6 +1130,7 @@ double vm_call_rollup_fn(int32_t func, double prev_value, int64_t prev_timestamp
   rfa.idx = idx;
   rfa.window = window;
   rfa.arg = arg;
+  rfa.arg2 = arg2;
   return f(&rfa);
 }
 
@@ -1061,6 +1178,7 @@ uint64_t vm_rollup_do(const vm_rollup_config* rc, const double* values,
   vm_rfa rfa;
   rfa.window = window;
   rfa.arg = rc->arg;
+  rfa.arg2 = rc->arg2;
   vm_rollup_fn f = fn_table(rc->func);
   if (!f) return 0;
 

@@ -89,6 +89,13 @@ typedef enum {
   VM_FN_MODE,                /* rollupModeOverTime, rollup.go:2293 + modeNoNaNs aggr.go:541 */
   VM_FN_DURATION,            /* newRollupDurationOverTime, rollup.go:1151 (arg=dMax secs) */
   VM_FN_OUTLIER_IQR,         /* rollupOutlierIQR, rollup.go:1427 */
+  VM_FN_OPEN,                /* rollupOpen, rollup.go:2227 (candlestick) */
+  VM_FN_CLOSE,               /* rollupClose, rollup.go:2238 */
+  VM_FN_LOW,                 /* rollupLow, rollup.go:2262 */
+  VM_FN_HIGH,                /* rollupHigh, rollup.go:2245 */
+  VM_FN_HOLT_WINTERS,        /* newRollupHoltWinters, rollup.go:1030 (arg=sf, arg2=tf) */
+  VM_FN_HOEFFDING_LOWER,     /* newRollupHoeffdingBoundLower, rollup.go:1323 (arg=phi) */
+  VM_FN_HOEFFDING_UPPER,     /* newRollupHoeffdingBoundUpper, rollup.go:1338 */
   VM_FN__COUNT
 } vm_func_id;
 
@@ -119,6 +126,7 @@ typedef struct {
   int32_t is_default_rollup;
   int32_t samples_scanned_per_call; /* rollupFuncsSamplesScannedPerCall */
   double  arg;                   /* phi / le / gt / eq / secs for arg funcs */
+  double  arg2;                  /* second scalar (holt_winters tf) */
 } vm_rollup_config;
 
 /* getTimestamps, eval.go:234-254. Returns number of grid points. */
@@ -153,7 +161,8 @@ uint64_t vm_rollup_do(const vm_rollup_config* rc, const double* values,
 double vm_call_rollup_fn(int32_t func, double prev_value, int64_t prev_timestamp,
                          const double* values, const int64_t* timestamps, int64_t n,
                          double real_prev_value, double real_next_value,
-                         int64_t curr_timestamp, int64_t idx, int64_t window, double arg);
+                         int64_t curr_timestamp, int64_t idx, int64_t window, double arg,
+                         double arg2);
 
 /* quantile over unsorted values with NaN filtering (aggr.go:870-876) and
  * quantileSorted (aggr.go:922-940). */

@@ -34,8 +34,11 @@ ARG_FUNCS = {
     "share_gt_over_time": 5.0, "share_eq_over_time": 1.0,
     "sum_le_over_time": 5.0, "sum_gt_over_time": 5.0,
     "sum_eq_over_time": 1.0, "predict_linear": 120.0,
-    "duration_over_time": 60.0,
+    "duration_over_time": 60.0, "quantile_over_time": 0.9,
+    "hoeffding_bound_lower": 0.9, "hoeffding_bound_upper": 0.9,
+    "holt_winters": 0.3,
 }
+ARG2_FUNCS = {"holt_winters": 0.4}
 
 
 @pytest.fixture(scope="module")
@@ -89,7 +92,8 @@ def test_all_funcs_ragged(engine, func):
     ts, vals, offsets = ragged_batch(300, 260, START, seed=99, dup_p=0.02)
     end = START + 100 * STEP
     plan = engine.RollupPlan(func, START, end, STEP, window=200_000,
-                             arg=ARG_FUNCS.get(func, 0.0))
+                             arg=ARG_FUNCS.get(func, 0.0),
+                             arg2=ARG2_FUNCS.get(func, 0.0))
     out, _, scanned = engine.rollup_eval(plan, ts, vals, offsets)
     ref, _, ref_scanned = _oracle_batch(plan, ts, vals, offsets)
     assert scanned == ref_scanned, f"{func}: scanned {scanned} != {ref_scanned}"

@@ -78,7 +78,7 @@ def test_func_scalar(case):
         prev_value=math.nan, prev_timestamp=0,
         real_prev_value=math.nan, real_next_value=0.0,
         curr_timestamp=0, idx=0, window=ts[-1] - ts[0],
-        arg=float(case.get("arg", 0.0)))
+        arg=float(case.get("arg", 0.0)), arg2=float(case.get("arg2", 0.0)))
     expected = decode_float(case["expected"])
     assert_values_equal([got], [expected], rel=1e-13, context=func)
 

@@ -40,7 +40,8 @@ enum {
   VMF_SUM_LE, VMF_SUM_GT, VMF_SUM_EQ, VMF_DERIV, VMF_PREDICT_LINEAR,
   VMF_ASCENT, VMF_DESCENT, VMF_ZSCORE, VMF_INTEGRATE, VMF_DISTINCT,
   VMF_INCREASES, VMF_DECREASES, VMF_MAD, VMF_DEFAULT_ROLLUP, VMF_MODE,
-  VMF_DURATION, VMF_OUTLIER_IQR,
+  VMF_DURATION, VMF_OUTLIER_IQR, VMF_OPEN, VMF_CLOSE, VMF_LOW, VMF_HIGH,
+  VMF_HOLT_WINTERS, VMF_HOEFFDING_LOWER, VMF_HOEFFDING_UPPER,
 };
 
 /* rollupFuncArg (rollup.go:523-556) restricted to what the funcs read. */
@@ -55,6 +56,7 @@ struct VmRfa {
   int64_t curr_timestamp;
   int64_t window;
   double arg;
+  double arg2;
 };
 
 /* ---- order-statistic helpers (count-based selection, no sort) ---- */
@@ -753,6 +755,100 @@ static VM_DEV double vmf_outlier_iqr(const VmRfa* r) {
   return vm_dnan();
 }
 
+/* candlestick family (rollup.go:2206-2291): window excludes currTimestamp;
+ * the pre-window sample opens the candle when within the window. */
+static VM_DEV int vmf_candlestick_len(const VmRfa* r) {
+  int n = r->n;
+  while (n > 0 && r->timestamps[n - 1] >= r->curr_timestamp) n--;
+  return n;
+}
+static VM_DEV double vmf_candlestick_first(const VmRfa* r) {
+  if (r->prev_timestamp + r->window >= r->curr_timestamp) return r->prev_value;
+  return vm_dnan();
+}
+static VM_DEV double vmf_open(const VmRfa* r) {
+  double v = vmf_candlestick_first(r);
+  if (!vm_isnan(v)) return v;
+  int n = vmf_candlestick_len(r);
+  if (n == 0) return vm_dnan();
+  return r->values[0];
+}
+static VM_DEV double vmf_close(const VmRfa* r) {
+  int n = vmf_candlestick_len(r);
+  if (n == 0) return vmf_candlestick_first(r);
+  return r->values[n - 1];
+}
+static VM_DEV double vmf_high(const VmRfa* r) {
+  int n = vmf_candlestick_len(r);
+  double m = vmf_candlestick_first(r);
+  int i = 0;
+  if (vm_isnan(m)) {
+    if (n == 0) return vm_dnan();
+    m = r->values[0];
+    i = 1;
+  }
+  for (; i < n; i++)
+    if (r->values[i] > m) m = r->values[i];
+  return m;
+}
+static VM_DEV double vmf_low(const VmRfa* r) {
+  int n = vmf_candlestick_len(r);
+  double m = vmf_candlestick_first(r);
+  int i = 0;
+  if (vm_isnan(m)) {
+    if (n == 0) return vm_dnan();
+    m = r->values[0];
+    i = 1;
+  }
+  for (; i < n; i++)
+    if (r->values[i] < m) m = r->values[i];
+  return m;
+}
+
+/* holt_winters (rollup.go:1030-1077); arg=sf, arg2=tf */
+static VM_DEV double vmf_holt_winters(const VmRfa* r) {
+  const double* values = r->values;
+  int n = r->n;
+  if (n == 0) return vm_dnan();
+  double sf = r->arg;
+  if (sf < 0 || sf > 1) return vm_dnan();
+  double tf = r->arg2;
+  if (tf < 0 || tf > 1) return vm_dnan();
+  double s0 = r->prev_value;
+  if (vm_isnan(s0)) {
+    s0 = values[0];
+    values++;
+    n--;
+    if (n == 0) return s0;
+  }
+  double b0 = values[0] - s0;
+  for (int i = 0; i < n; i++) {
+    double v = values[i];
+    double s1 = sf * v + (1 - sf) * (s0 + b0);
+    double b1 = tf * (s1 - s0) + (1 - tf) * b0;
+    s0 = s1;
+    b0 = b1;
+  }
+  return s0;
+}
+
+/* hoeffding bounds (rollup.go:1323-1381); arg=phi */
+static VM_DEV double vmf_hoeffding(const VmRfa* r, int upper) {
+  int n = r->n;
+  if (n == 0) return vm_dnan();
+  if (n == 1) return r->values[0];
+  double v_max = vmf_max(r);
+  double v_min = vmf_min(r);
+  double v_avg = vmf_avg(r);
+  double v_range = v_max - v_min;
+  double bound;
+  if (v_range <= 0) bound = 0;
+  else if (r->arg >= 1) bound = vm_dinf();
+  else if (r->arg <= 0) bound = 0;
+  else bound = v_range * sqrt(log(1 / (1 - r->arg)) / (2 * (double)n));
+  return upper ? v_avg + bound : v_avg - bound;
+}
+
 static VM_DEV double vm_eval_rollup_fn(int32_t func, const VmRfa* r) {
   switch (func) {
     case VMF_RATE: case VMF_DERIV_FAST: return vmf_deriv_fast(r);
@@ -813,6 +909,13 @@ static VM_DEV double vm_eval_rollup_fn(int32_t func, const VmRfa* r) {
     case VMF_MODE: return vmf_mode(r);
     case VMF_DURATION: return vmf_duration(r);
     case VMF_OUTLIER_IQR: return vmf_outlier_iqr(r);
+    case VMF_OPEN: return vmf_open(r);
+    case VMF_CLOSE: return vmf_close(r);
+    case VMF_LOW: return vmf_low(r);
+    case VMF_HIGH: return vmf_high(r);
+    case VMF_HOLT_WINTERS: return vmf_holt_winters(r);
+    case VMF_HOEFFDING_LOWER: return vmf_hoeffding(r, 0);
+    case VMF_HOEFFDING_UPPER: return vmf_hoeffding(r, 1);
     default: return vm_dnan();
   }
 }

@@ -169,6 +169,7 @@ struct KPlan {
   int32_t drop_stale;
   int32_t chunk_wave;      /* LDS samples per wave (wave kernel), 64-aligned */
   double arg;
+  double arg2;
 };
 
 struct KIO {
@@ -436,6 +437,7 @@ static VM_DEV uint64_t eval_grid_point_ij(const KPlan& p, const SeriesWindow& sw
   VmRfa r;
   r.window = sw.window;
   r.arg = p.arg;
+  r.arg2 = p.arg2;
   r.prev_value = vm_dnan();
   r.prev_timestamp = t_start - sw.max_prev_interval;
   if (i < count && i > 0 && ts[i - 1] > r.prev_timestamp) {
@@ -1412,6 +1414,7 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
   p.chunk_wave = (int32_t)std::min<uint32_t>(
       CHUNK_WAVE, std::max<uint32_t>(64, (b.max_wave_len + 63) & ~63u));
   p.arg = plan->arg;
+  p.arg2 = plan->arg2;
 
   KIO io;
   io.ts = b.d_ts;

@@ -47,6 +47,9 @@ FUNC_IDS = {
     "decreases_over_time": 57, "mad_over_time": 58, "default_rollup": 59,
     "mode_over_time": 60, "duration_over_time": 61,
     "outlier_iqr_over_time": 62,
+    "rollup_open": 63, "rollup_close": 64, "rollup_low": 65,
+    "rollup_high": 66, "holt_winters": 67, "hoeffding_bound_lower": 68,
+    "hoeffding_bound_upper": 69,
     # aliases to shared implementations, as in the reference map
     "increase_prometheus": 4, "timestamp": 31, "timestamp_with_name": 31,
 }
@@ -67,6 +70,7 @@ CAN_ADJUST_WINDOW_FUNCS = {
     "default_rollup", "deriv", "deriv_fast", "ideriv", "irate", "rate",
     "rate_over_sum", "rollup", "rollup_candlestick", "rollup_deriv",
     "rollup_rate", "rollup_scrape_interval", "scrape_interval", "timestamp",
+    "rollup_open", "rollup_close", "rollup_low", "rollup_high",
 }
 
 # rollupFuncsSamplesScannedPerCall (rollup.go:238-263)
@@ -106,6 +110,7 @@ class _PlanC(ctypes.Structure):
         ("samples_scanned_per_call", ctypes.c_int32),
         ("skip_finalize", ctypes.c_int32),
         ("arg", ctypes.c_double),
+        ("arg2", ctypes.c_double),
     ]
 
 
@@ -157,7 +162,7 @@ class RollupPlan:
     (rollup.go:374-516) + the EvalConfig fields that reach it."""
 
     def __init__(self, func, start, end, step, window=0, lookback_delta=0,
-                 min_staleness_interval=0, arg=0.0, aggr="none",
+                 min_staleness_interval=0, arg=0.0, arg2=0.0, aggr="none",
                  skip_finalize=False, keep_stale_nans=False):
         if func not in FUNC_IDS:
             raise VmGpuError(f"unsupported rollup function {func!r}")
@@ -173,6 +178,7 @@ class RollupPlan:
         self.lookback_delta = int(lookback_delta)
         self.min_staleness_interval = int(min_staleness_interval)
         self.arg = float(arg)
+        self.arg2 = float(arg2)
         self.aggr = aggr
         self.skip_finalize = skip_finalize
         self.keep_stale_nans = keep_stale_nans
@@ -196,7 +202,7 @@ class RollupPlan:
             drop_stale_nans=0 if (keep_stale_nans or func in KEEP_STALE_NANS_FUNCS) else 1,
             samples_scanned_per_call=SAMPLES_SCANNED_PER_CALL.get(func, 0),
             skip_finalize=1 if skip_finalize else 0,
-            arg=self.arg)
+            arg=self.arg, arg2=self.arg2)
 
     @property
     def n_grid(self):
