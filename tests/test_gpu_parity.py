@@ -373,3 +373,40 @@ def test_histogram_quantile_gpu(engine):
             gn, rn = np.isnan(g), np.isnan(r)
             assert (gn == rn).all(), f"phi={phi} {name}"
             assert np.array_equal(g[~gn], r[~rn]), f"phi={phi} {name}"
+
+
+def test_subquery_rollup(engine):
+    """rate(m[5m])[1h:15s]-style subquery (§3e): the same kernels applied to
+    a resident inner grid, vs the oracle doing the identical orchestration."""
+    rng = np.random.default_rng(31)
+    n_series = 300
+
+    def inner_eval(sq_start, sq_end, sq_step):
+        m = (sq_end - sq_start) // sq_step + 1
+        vals = np.cumsum(rng.random((n_series, m)) * 10, axis=1)
+        vals[rng.random((n_series, m)) < 0.05] = np.nan  # holes
+        inner_eval.cache = vals
+        return vals
+
+    start = START
+    end = START + 60 * STEP
+    out, _, scanned = engine.rollup_subquery(
+        "rate", start, end, STEP, window=300_000, sq_step=STEP,
+        inner_eval=inner_eval)
+
+    # oracle side: same removeNanValues + preFunc + doInternal
+    vals = inner_eval.cache
+    sq_start = start - (300_000 + STEP + engine.MAX_SILENCE_INTERVAL_MS)
+    sq_start, sq_end = engine.align_start_end(sq_start, end + STEP, STEP)
+    sq_ts = np.arange(sq_start, sq_end + 1, STEP, dtype=np.int64)
+    keep = ~np.isnan(vals)
+    offsets = np.zeros(n_series + 1, dtype=np.uint64)
+    np.cumsum(keep.sum(axis=1), out=offsets[1:])
+    cvals = vals[keep]
+    cts = np.broadcast_to(sq_ts, vals.shape)[keep].astype(np.int64)
+    rc = oracle.make_config("rate", start, end, STEP, window=300_000)
+    ref, _, ref_scanned = oracle.rollup_eval_batch(
+        rc, cts, cvals, offsets, remove_counter_resets=True,
+        drop_stale_nans=False)
+    assert scanned == ref_scanned
+    assert_parity(out, ref, exact=True, context="subquery")

@@ -496,3 +496,51 @@ def decode_blocks(blocks):
     for i, b in enumerate(blocks):
         offsets[i + 1] = offsets[i] + int(b["rows"])
     return out_ts, out_vals, offsets
+
+
+def align_start_end(start, end, step):
+    """alignStartEnd (eval.go): floor start, ceil end to step multiples."""
+    start = start - start % step
+    adjust = end % step
+    if adjust > 0:
+        end += step - adjust
+    return start, end
+
+
+MAX_SILENCE_INTERVAL_MS = 5 * 60 * 1000  # maxSilenceInterval (eval.go:1920)
+
+
+def rollup_subquery(outer_func, start, end, step, window, sq_step,
+                    inner_eval, lookback_delta=0, arg=0.0, arg2=0.0,
+                    aggr="none", group_ids=None, n_groups=0):
+    """evalRollupFuncWithSubquery (eval.go:1033-1100): evaluate the inner
+    expression on a finer, extended grid via `inner_eval(sq_start, sq_end,
+    sq_step) -> [n_series x m] f64 grid values`, strip NaN points per series
+    (removeNanValues, eval.go:1150), then run the SAME rollup machinery over
+    the resident grid — no separate kernel.
+
+    Returns (out, counts, samples_scanned) like rollup_eval."""
+    if sq_step == 0:
+        sq_step = step
+    sq_start = start - (window + sq_step + MAX_SILENCE_INTERVAL_MS)
+    sq_end = end + sq_step
+    sq_start, sq_end = align_start_end(sq_start, sq_end, sq_step)
+    inner_vals = np.asarray(inner_eval(sq_start, sq_end, sq_step),
+                            dtype=np.float64)
+    sq_ts = np.arange(sq_start, sq_end + 1, sq_step, dtype=np.int64)
+    n_series, m = inner_vals.shape
+    assert m == len(sq_ts), "inner grid shape mismatch"
+    # removeNanValues per series -> CSR
+    keep = ~np.isnan(inner_vals)
+    counts_per_series = keep.sum(axis=1)
+    offsets = np.zeros(n_series + 1, dtype=np.uint64)
+    np.cumsum(counts_per_series, out=offsets[1:])
+    vals = inner_vals[keep]
+    ts = np.broadcast_to(sq_ts, inner_vals.shape)[keep].astype(np.int64)
+    plan = RollupPlan(outer_func, start, end, step, window=window,
+                      lookback_delta=lookback_delta, arg=arg, arg2=arg2,
+                      aggr=aggr,
+                      # subquery input is already stale-free grid data
+                      keep_stale_nans=True)
+    return rollup_eval(plan, ts, vals, offsets, group_ids=group_ids,
+                       n_groups=n_groups)
