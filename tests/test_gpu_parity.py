@@ -65,7 +65,7 @@ def _oracle_batch(engine_plan, ts, vals, offsets, group_ids=None, n_groups=0,
         min_staleness_interval=engine_plan._c.min_staleness_interval,
         is_default_rollup=engine_plan._c.is_default_rollup,
         samples_scanned_per_call=engine_plan._c.samples_scanned_per_call,
-        arg=engine_plan._c.arg)
+        arg=engine_plan._c.arg, arg2=engine_plan._c.arg2)
     return oracle.rollup_eval_batch(
         rc, ts, vals, offsets, group_ids=group_ids, n_groups=n_groups,
         aggr=aggr,

@@ -395,7 +395,8 @@ static VM_DEV void vm_emit_value(const KPlan& p, const KIO& io, uint32_t s,
       case VMGPU_AGGR_COUNT:
       case VMGPU_AGGR_GROUP: atomicAdd(gv, 1.0); *gc = 1.0; break;
       case VMGPU_AGGR_SUM2: atomicAdd(gv, v * v); *gc = 1.0; break;
-      case VMGPU_AGGR_GEOMEAN: vm_atomic_mul_f64(gv, v); *gc = 1.0; break;
+      /* geomean finalize takes pow(product, 1/count): true counts */
+      case VMGPU_AGGR_GEOMEAN: vm_atomic_mul_f64(gv, v); atomicAdd(gc, 1.0); break;
       default: break;
     }
   }
