@@ -175,6 +175,24 @@ int vmgpu_decode_blocks(const uint8_t* payload, uint64_t payload_len,
                         int64_t* out_ts, double* out_vals,
                         char* errbuf, size_t errbuf_len);
 
+/* Per-series k-way merge of decoded blocks + optional dedup — the GPU
+ * equivalent of app/vmselect/netstorage/netstorage.go:564 mergeSortBlocks
+ * and lib/storage/dedup.go:29 DeduplicateSamples.  Input: total_rows decoded
+ * samples in ts/vals, partitioned into n_blocks sorted blocks by
+ * block_offsets[n_blocks+1]; series s owns blocks
+ * [series_block_start[s], series_block_start[s+1]).  Output: merged (and
+ * deduped when dedup_interval>0) samples, densely packed; out_offsets has
+ * n_series+1 entries, out_counts[s] = merged length of series s.  Blocks
+ * with disjoint time ranges (the common LSM-part case) take a wave-parallel
+ * concatenation path; overlapping blocks take the exact heap-merge path. */
+int vmgpu_merge_blocks(const int64_t* ts, const double* vals,
+                       const uint64_t* block_offsets, uint32_t n_blocks,
+                       const uint32_t* series_block_start, uint32_t n_series,
+                       int64_t dedup_interval,
+                       int64_t* out_ts, double* out_vals,
+                       uint64_t* out_offsets, uint64_t* out_counts,
+                       char* errbuf, size_t errbuf_len);
+
 /* Wall time of the rollup kernels inside the last vmgpu_rollup_exec on this
  * thread's context, measured with hipEvents on the launch stream (for the
  * bench's roofline accounting). */

@@ -379,3 +379,342 @@ int vmgpu_decode_blocks(const uint8_t* payload, uint64_t payload_len,
 }
 
 }  /* extern "C" */
+
+
+/* ------------------------------------------------------------------ */
+/* per-series k-way merge + dedup (netstorage.go:564 mergeSortBlocks, */
+/* lib/storage/dedup.go:29 DeduplicateSamples)                        */
+/* ------------------------------------------------------------------ */
+
+/* One wave per series.  Fast path: when the series' blocks are disjoint in
+ * time (the overwhelmingly common case: LSM parts hold distinct time
+ * ranges), the merge is a concatenation of the blocks in min-timestamp
+ * order — wave-parallel coalesced copies.  Overlapping blocks fall back to
+ * the exact serial heap merge on lane 0 (matching the reference's
+ * container/heap order for duplicate timestamps).  Dedup (when enabled)
+ * runs as a lane-cooperative pass identical to DeduplicateSamples. */
+
+#define MAX_BLOCKS_PER_SERIES 64
+
+static __device__ __forceinline__ void d_wave_sync() {
+  asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+}
+
+static __device__ int d_is_stale(double v) {
+  return __double_as_longlong(v) == 0x7ff0000000000002LL;
+}
+
+__global__ __launch_bounds__(DBLOCK) void merge_blocks_kernel(
+    const int64_t* ts, const double* vals,
+    const uint64_t* block_offsets,      /* [n_blocks+1] into ts/vals */
+    const uint32_t* series_block_start, /* [n_series+1]: block range per series */
+    const uint64_t* out_offsets,        /* [n_series+1] capacity offsets */
+    uint32_t n_series, int64_t dedup_interval,
+    int64_t* out_ts, double* out_vals,
+    uint64_t* out_counts,               /* [n_series] merged lengths */
+    int* err_flag) {
+  const int lane = threadIdx.x % DWAVE;
+  const int wv = threadIdx.x / DWAVE;
+  uint32_t wid = blockIdx.x * (DBLOCK / DWAVE) + wv;
+  uint32_t stride = gridDim.x * (DBLOCK / DWAVE);
+  for (uint32_t s = wid; s < n_series; s += stride) {
+    uint32_t b0 = series_block_start[s];
+    uint32_t b1 = series_block_start[s + 1];
+    uint32_t nb = b1 - b0;
+    int64_t* dts = out_ts + out_offsets[s];
+    double* dvs = out_vals + out_offsets[s];
+    if (nb == 0) {
+      if (lane == 0) out_counts[s] = 0;
+      continue;
+    }
+    if (nb > MAX_BLOCKS_PER_SERIES) {
+      if (lane == 0) atomicExch(err_flag, 20);
+      continue;
+    }
+    /* order blocks by first timestamp (insertion order on lane 0, tiny nb),
+     * detect overlap */
+    uint32_t order[MAX_BLOCKS_PER_SERIES];
+    uint32_t nb_nonempty = 0;
+    bool overlap = false;
+    if (lane == 0) {
+      for (uint32_t b = 0; b < nb; b++) {
+        uint64_t lo = block_offsets[b0 + b], hi = block_offsets[b0 + b + 1];
+        if (hi == lo) continue; /* skip empty blocks (mergeSortBlocks:570) */
+        uint32_t idx = nb_nonempty++;
+        order[idx] = b;
+        while (idx > 0) {
+          int64_t cur_first = ts[block_offsets[b0 + order[idx]]];
+          int64_t prev_first = ts[block_offsets[b0 + order[idx - 1]]];
+          if (prev_first <= cur_first) break;
+          uint32_t t = order[idx];
+          order[idx] = order[idx - 1];
+          order[idx - 1] = t;
+          idx--;
+        }
+      }
+      for (uint32_t i = 1; i < nb_nonempty; i++) {
+        uint64_t prev_end = block_offsets[b0 + order[i - 1] + 1];
+        uint64_t cur_lo = block_offsets[b0 + order[i]];
+        if (ts[prev_end - 1] >= ts[cur_lo]) overlap = true;
+      }
+    }
+    nb_nonempty = __shfl(nb_nonempty, 0);
+    overlap = __shfl((int)overlap, 0);
+    for (uint32_t i = 0; i < nb_nonempty; i++) order[i] = __shfl(order[i], 0);
+
+    uint64_t n_merged = 0;
+    if (!overlap) {
+      /* concatenate in order: wave-parallel coalesced copy */
+      for (uint32_t i = 0; i < nb_nonempty; i++) {
+        uint64_t lo = block_offsets[b0 + order[i]];
+        uint64_t hi = block_offsets[b0 + order[i] + 1];
+        for (uint64_t k = lane; k < hi - lo; k += DWAVE) {
+          dts[n_merged + k] = ts[lo + k];
+          dvs[n_merged + k] = vals[lo + k];
+        }
+        n_merged += hi - lo;
+      }
+      d_wave_sync();
+    } else if (lane == 0) {
+      /* exact serial heap merge (mergeSortBlocks + getNextBlock +
+       * equalSamplesPrefix semantics) — Go container/heap order */
+      uint64_t next[MAX_BLOCKS_PER_SERIES];
+      uint64_t lim[MAX_BLOCKS_PER_SERIES];
+      uint32_t heap[MAX_BLOCKS_PER_SERIES];
+      uint32_t hn = 0;
+      /* heap seeded in INPUT order (skipping empties), exactly as the
+       * reference builds sbs before heap.Init — initial array order decides
+       * tie-breaking for equal timestamps */
+      for (uint32_t b = 0; b < nb; b++) {
+        uint64_t lo = block_offsets[b0 + b], hi = block_offsets[b0 + b + 1];
+        if (hi == lo) continue;
+        next[b] = lo;
+        lim[b] = hi;
+        heap[hn++] = b;
+      }
+      /* heap.Init: siftDown from hn/2-1 */
+      for (int i = (int)hn / 2 - 1; i >= 0; i--) {
+        uint32_t j = (uint32_t)i;
+        for (;;) {
+          uint32_t c = 2 * j + 1;
+          if (c >= hn) break;
+          if (c + 1 < hn && ts[next[heap[c + 1]]] < ts[next[heap[c]]]) c++;
+          if (ts[next[heap[c]]] >= ts[next[heap[j]]]) break;
+          uint32_t t = heap[j]; heap[j] = heap[c]; heap[c] = t;
+          j = c;
+        }
+      }
+      while (hn > 0) {
+        uint32_t top = heap[0];
+        if (hn == 1) {
+          for (uint64_t k = next[top]; k < lim[top]; k++) {
+            dts[n_merged] = ts[k];
+            dvs[n_merged] = vals[k];
+            n_merged++;
+          }
+          break;
+        }
+        uint32_t nxt = (hn < 3) ? heap[1]
+                       : (ts[next[heap[2]]] < ts[next[heap[1]]] ? heap[2] : heap[1]);
+        int64_t ts_next = ts[next[nxt]];
+        /* equalSamplesPrefix: identical (ts, value-bits) prefixes are
+         * replicated samples; skip them at top when dedup is enabled */
+        uint64_t neq = 0;
+        {
+          uint64_t a = next[top], b = next[nxt];
+          while (a + neq < lim[top] && b + neq < lim[nxt] &&
+                 ts[a + neq] == ts[b + neq] &&
+                 __double_as_longlong(vals[a + neq]) ==
+                     __double_as_longlong(vals[b + neq]))
+            neq++;
+        }
+        if (neq > 0 && dedup_interval > 0) {
+          next[top] += neq;
+        } else {
+          while (next[top] < lim[top] && ts[next[top]] <= ts_next) {
+            dts[n_merged] = ts[next[top]];
+            dvs[n_merged] = vals[next[top]];
+            n_merged++;
+            next[top]++;
+          }
+        }
+        if (next[top] < lim[top]) {
+          /* heap.Fix(0) = siftDown then siftUp (top only needs down) */
+          uint32_t j = 0;
+          for (;;) {
+            uint32_t c = 2 * j + 1;
+            if (c >= hn) break;
+            if (c + 1 < hn && ts[next[heap[c + 1]]] < ts[next[heap[c]]]) c++;
+            if (ts[next[heap[c]]] >= ts[next[heap[j]]]) break;
+            uint32_t t = heap[j]; heap[j] = heap[c]; heap[c] = t;
+            j = c;
+          }
+        } else {
+          /* heap.Pop */
+          heap[0] = heap[hn - 1];
+          hn--;
+          uint32_t j = 0;
+          for (;;) {
+            uint32_t c = 2 * j + 1;
+            if (c >= hn) break;
+            if (c + 1 < hn && ts[next[heap[c + 1]]] < ts[next[heap[c]]]) c++;
+            if (ts[next[heap[c]]] >= ts[next[heap[j]]]) break;
+            uint32_t t = heap[j]; heap[j] = heap[c]; heap[c] = t;
+            j = c;
+          }
+        }
+      }
+    }
+    n_merged = __shfl((unsigned long long)n_merged, 0);
+    d_wave_sync();
+
+    /* DeduplicateSamples (dedup.go:29-92), serial exact on lane 0 (rare:
+     * needsDedup gate runs wave-parallel first) */
+    if (dedup_interval > 0 && n_merged >= 2) {
+      bool needs = false;
+      /* needsDedup (dedup.go:149): sequential bucket walk — cheap serial */
+      if (lane == 0) {
+        int64_t ts_next_b = dts[0] + dedup_interval - 1;
+        ts_next_b -= ts_next_b % dedup_interval;
+        for (uint64_t i = 1; i < n_merged; i++) {
+          int64_t t = dts[i];
+          if (t <= ts_next_b) { needs = true; break; }
+          ts_next_b += dedup_interval;
+          if (ts_next_b < t) {
+            ts_next_b = t + dedup_interval - 1;
+            ts_next_b -= ts_next_b % dedup_interval;
+          }
+        }
+      }
+      needs = __shfl((int)needs, 0);
+      if (needs && lane == 0) {
+        int64_t ts_next_b = dts[0] + dedup_interval - 1;
+        ts_next_b -= ts_next_b % dedup_interval;
+        uint64_t k = 0;
+        for (uint64_t i = 1; i < n_merged; i++) {
+          int64_t t = dts[i];
+          if (t <= ts_next_b) continue;
+          uint64_t j = i - 1;
+          int64_t tp = dts[j];
+          double vp = dvs[j];
+          while (j > 0 && dts[j - 1] == tp) {
+            j--;
+            if (d_is_stale(dvs[j])) continue;
+            if (d_is_stale(vp)) { vp = dvs[j]; continue; }
+            if (dvs[j] > vp) vp = dvs[j];
+          }
+          dts[k] = tp;
+          dvs[k] = vp;
+          k++;
+          ts_next_b += dedup_interval;
+          if (ts_next_b < t) {
+            ts_next_b = t + dedup_interval - 1;
+            ts_next_b -= ts_next_b % dedup_interval;
+          }
+        }
+        uint64_t j = n_merged - 1;
+        int64_t tp = dts[j];
+        double vp = dvs[j];
+        while (j > 0 && dts[j - 1] == tp) {
+          j--;
+          if (d_is_stale(dvs[j])) continue;
+          if (d_is_stale(vp)) { vp = dvs[j]; continue; }
+          if (dvs[j] > vp) vp = dvs[j];
+        }
+        dts[k] = tp;
+        dvs[k] = vp;
+        k++;
+        n_merged = k;
+      }
+      n_merged = __shfl((unsigned long long)n_merged, 0);
+    }
+    if (lane == 0) out_counts[s] = n_merged;
+    d_wave_sync();
+  }
+}
+
+extern "C" {
+
+int vmgpu_merge_blocks(const int64_t* ts, const double* vals,
+                       const uint64_t* block_offsets, uint32_t n_blocks,
+                       const uint32_t* series_block_start, uint32_t n_series,
+                       int64_t dedup_interval,
+                       int64_t* out_ts, double* out_vals,
+                       uint64_t* out_offsets, uint64_t* out_counts,
+                       char* errbuf, size_t errbuf_len) {
+  if (!ts || !vals || !block_offsets || !series_block_start || !out_ts ||
+      !out_vals || !out_offsets || !out_counts || n_series == 0)
+    return dset_err(errbuf, errbuf_len, "vmgpu: bad args");
+  hipStream_t st = 0;
+  uint64_t total = block_offsets[n_blocks];
+  /* capacity offsets = pre-merge block extents per series */
+  std::vector<uint64_t> cap_off(n_series + 1);
+  cap_off[0] = 0;
+  for (uint32_t s = 0; s < n_series; s++) {
+    uint64_t lo = block_offsets[series_block_start[s]];
+    uint64_t hi = block_offsets[series_block_start[s + 1]];
+    cap_off[s + 1] = cap_off[s] + (hi - lo);
+  }
+  int64_t* d_ts = nullptr;
+  double* d_vals = nullptr;
+  uint64_t* d_boff = nullptr;
+  uint32_t* d_sbs = nullptr;
+  uint64_t* d_ooff = nullptr;
+  int64_t* d_ots = nullptr;
+  double* d_ovals = nullptr;
+  uint64_t* d_ocnt = nullptr;
+  int* d_err = nullptr;
+  DHIP_TRY(hipMalloc(&d_ts, total * 8), "alloc ts");
+  DHIP_TRY(hipMalloc(&d_vals, total * 8), "alloc vals");
+  DHIP_TRY(hipMalloc(&d_boff, (size_t)(n_blocks + 1) * 8), "alloc boff");
+  DHIP_TRY(hipMalloc(&d_sbs, (size_t)(n_series + 1) * 4), "alloc sbs");
+  DHIP_TRY(hipMalloc(&d_ooff, (size_t)(n_series + 1) * 8), "alloc ooff");
+  DHIP_TRY(hipMalloc(&d_ots, cap_off[n_series] * 8), "alloc out ts");
+  DHIP_TRY(hipMalloc(&d_ovals, cap_off[n_series] * 8), "alloc out vals");
+  DHIP_TRY(hipMalloc(&d_ocnt, (size_t)n_series * 8), "alloc out counts");
+  DHIP_TRY(hipMalloc(&d_err, 4), "alloc err");
+  DHIP_TRY(hipMemcpyAsync(d_ts, ts, total * 8, hipMemcpyHostToDevice, st), "ul ts");
+  DHIP_TRY(hipMemcpyAsync(d_vals, vals, total * 8, hipMemcpyHostToDevice, st), "ul vals");
+  DHIP_TRY(hipMemcpyAsync(d_boff, block_offsets, (size_t)(n_blocks + 1) * 8, hipMemcpyHostToDevice, st), "ul boff");
+  DHIP_TRY(hipMemcpyAsync(d_sbs, series_block_start, (size_t)(n_series + 1) * 4, hipMemcpyHostToDevice, st), "ul sbs");
+  DHIP_TRY(hipMemcpyAsync(d_ooff, cap_off.data(), (size_t)(n_series + 1) * 8, hipMemcpyHostToDevice, st), "ul ooff");
+  DHIP_TRY(hipMemsetAsync(d_err, 0, 4, st), "zero err");
+  uint32_t grid = std::min<uint32_t>((n_series + 3) / 4, 2048);
+  hipLaunchKernelGGL(merge_blocks_kernel, dim3(grid), dim3(DBLOCK), 0, st,
+                     d_ts, d_vals, d_boff, d_sbs, d_ooff, n_series,
+                     dedup_interval, d_ots, d_ovals, d_ocnt, d_err);
+  std::vector<uint64_t> counts(n_series);
+  int err_h = 0;
+  DHIP_TRY(hipMemcpyAsync(counts.data(), d_ocnt, (size_t)n_series * 8, hipMemcpyDeviceToHost, st), "dl counts");
+  DHIP_TRY(hipMemcpyAsync(&err_h, d_err, 4, hipMemcpyDeviceToHost, st), "dl err");
+  /* compact download: per series, copy the merged prefix */
+  DHIP_TRY(hipStreamSynchronize(st), "sync");
+  hipError_t kerr = hipGetLastError();
+  if (kerr == hipSuccess && err_h == 0) {
+    uint64_t w = 0;
+    out_offsets[0] = 0;
+    for (uint32_t s = 0; s < n_series; s++) {
+      uint64_t n = counts[s];
+      DHIP_TRY(hipMemcpyAsync(out_ts + w, d_ots + cap_off[s], n * 8,
+                              hipMemcpyDeviceToHost, st), "dl merged ts");
+      DHIP_TRY(hipMemcpyAsync(out_vals + w, d_ovals + cap_off[s], n * 8,
+                              hipMemcpyDeviceToHost, st), "dl merged vals");
+      w += n;
+      out_offsets[s + 1] = w;
+      out_counts[s] = n;
+    }
+    DHIP_TRY(hipStreamSynchronize(st), "sync dl");
+  }
+  (void)hipFree(d_ts); (void)hipFree(d_vals); (void)hipFree(d_boff);
+  (void)hipFree(d_sbs); (void)hipFree(d_ooff); (void)hipFree(d_ots);
+  (void)hipFree(d_ovals); (void)hipFree(d_ocnt); (void)hipFree(d_err);
+  if (kerr != hipSuccess) return dhip_err(errbuf, errbuf_len, "merge kernel", kerr);
+  if (err_h != 0) {
+    char msg[64];
+    snprintf(msg, sizeof(msg), "vmgpu: merge error %d", err_h);
+    return dset_err(errbuf, errbuf_len, msg);
+  }
+  return 0;
+}
+
+}  /* extern "C" */

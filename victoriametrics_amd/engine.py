@@ -498,6 +498,49 @@ def decode_blocks(blocks):
     return out_ts, out_vals, offsets
 
 
+def merge_blocks(ts, vals, block_offsets, series_block_start,
+                 dedup_interval=0):
+    """GPU per-series k-way merge of decoded blocks + dedup
+    (netstorage.go:564 mergeSortBlocks + dedup.go:29 DeduplicateSamples).
+
+    ts/vals: all decoded samples, block b = [block_offsets[b],
+    block_offsets[b+1]); series s owns blocks [series_block_start[s],
+    series_block_start[s+1]).  Returns (out_ts, out_vals, out_offsets):
+    merged samples packed densely per series with n_series+1 offsets.
+    """
+    init()
+    lib = _load_lib()
+    ts = np.ascontiguousarray(ts, dtype=np.int64)
+    vals = np.ascontiguousarray(vals, dtype=np.float64)
+    boff = np.ascontiguousarray(block_offsets, dtype=np.uint64)
+    sbs = np.ascontiguousarray(series_block_start, dtype=np.uint32)
+    n_blocks = len(boff) - 1
+    n_series = len(sbs) - 1
+    cap = int(boff[-1])
+    out_ts = np.empty(max(cap, 1), dtype=np.int64)
+    out_vals = np.empty(max(cap, 1), dtype=np.float64)
+    out_off = np.zeros(n_series + 1, dtype=np.uint64)
+    out_cnt = np.zeros(max(n_series, 1), dtype=np.uint64)
+    errbuf = ctypes.create_string_buffer(256)
+    rc = lib.vmgpu_merge_blocks(
+        ts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+        vals.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        boff.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        ctypes.c_uint32(n_blocks),
+        sbs.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        ctypes.c_uint32(n_series), ctypes.c_int64(int(dedup_interval)),
+        out_ts.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+        out_vals.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        out_off.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        out_cnt.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise VmGpuError(f"vmgpu_merge_blocks failed ({rc}): "
+                         f"{errbuf.value.decode()}")
+    total = int(out_off[n_series])
+    return out_ts[:total], out_vals[:total], out_off
+
+
 def align_start_end(start, end, step):
     """alignStartEnd (eval.go): floor start, ceil end to step multiples."""
     start = start - start % step
