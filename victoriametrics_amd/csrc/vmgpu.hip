@@ -168,6 +168,8 @@ struct KPlan {
   int32_t rcr;
   int32_t drop_stale;
   int32_t chunk_wave;      /* LDS samples per wave (wave kernel), 64-aligned */
+  int32_t jbuf_elems;      /* LDS window-end index cache [n_grid] u16 for the
+                              shared-boundary rate path (0 = disabled) */
   double arg;
   double arg2;
 };
@@ -191,11 +193,40 @@ struct KIO {
 /* --- phase A: compacted load (executed by ONE wave) ----------------- */
 /* Copies [g_ts,g_vals)[0..n) to dst, dropping stale NaNs if requested.
  * Returns new count. dst may be LDS or global. */
+struct __align__(16) vm_i64x2 { int64_t x, y; };
+
 static __device__ int load_compact_wave(const int64_t* g_ts, const double* g_vals,
                                         int64_t n, int64_t* d_ts, double* d_vals,
                                         bool drop_stale, int lane) {
   int count = 0;
-  for (int64_t src = 0; src < n; src += WAVE) {
+  int64_t src = 0;
+  /* vector fast path: 128-element tiles via 16-B loads/stores (the staging
+   * loop is instruction-issue-bound, not bandwidth-bound — see profiles/).
+   * Falls to the scalar tile path when stale NaNs must be compacted out or
+   * the running count is odd (LDS b128 stores need 16-B alignment). */
+  if ((((uintptr_t)g_ts | (uintptr_t)g_vals) & 15) == 0) {
+    const uint64_t lt_mask = (lane == 63) ? 0x7fffffffffffffffULL
+                                          : ((1ULL << lane) - 1);
+    for (; src + 128 <= n; src += 128) {
+      double2 v = *(const double2*)(g_vals + src + 2 * lane);
+      vm_i64x2 t = *(const vm_i64x2*)(g_ts + src + 2 * lane);
+      bool k0 = !(drop_stale && vm_is_stale_nan(v.x));
+      bool k1 = !(drop_stale && vm_is_stale_nan(v.y));
+      uint64_t m0 = __ballot(k0), m1 = __ballot(k1);
+      if ((m0 & m1) == ~0ULL && (count & 1) == 0) {
+        *(double2*)(d_vals + count + 2 * lane) = v;
+        *(vm_i64x2*)(d_ts + count + 2 * lane) = t;
+        count += 128;
+      } else {
+        int below = __popcll(m0 & lt_mask) + __popcll(m1 & lt_mask);
+        int dst0 = count + below;
+        if (k0) { d_ts[dst0] = t.x; d_vals[dst0] = v.x; }
+        if (k1) { int d1 = dst0 + (k0 ? 1 : 0); d_ts[d1] = t.y; d_vals[d1] = v.y; }
+        count += __popcll(m0) + __popcll(m1);
+      }
+    }
+  }
+  for (; src < n; src += WAVE) {
     int64_t k = src + lane;
     bool active = k < n;
     double v = 0.0;
@@ -483,12 +514,15 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int wave_in_block = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  const size_t wave_bytes = (size_t)p.chunk_wave * 16 + 256;
+  const size_t jbuf_bytes = ((size_t)p.jbuf_elems * 2 + 15) & ~(size_t)15;
+  const size_t wave_bytes = (size_t)p.chunk_wave * 16 + 256 + jbuf_bytes;
   int64_t* lts = (int64_t*)(smem + (size_t)wave_in_block * wave_bytes);
   double* lvs = (double*)(smem + (size_t)wave_in_block * wave_bytes +
                           (size_t)p.chunk_wave * 8);
   double* lscratch = (double*)(smem + (size_t)wave_in_block * wave_bytes +
                                (size_t)p.chunk_wave * 16);
+  uint16_t* jbuf = (uint16_t*)(smem + (size_t)wave_in_block * wave_bytes +
+                               (size_t)p.chunk_wave * 16 + 256);
   uint64_t scanned = 0;
   const uint32_t wave_id = blockIdx.x * WAVES_PER_BLOCK + wave_in_block;
   const uint32_t wave_stride = gridDim.x * WAVES_PER_BLOCK;
@@ -532,6 +566,47 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
       int64_t span_ms = lts[count - 1] - ts0;
       idx_per_ms = span_ms > 0 ? (double)(count - 1) / (double)span_ms : 0.0;
     }
+#if !defined(VMGPU_ABL_NO_SEEK) && !defined(VMGPU_ABL_NO_EVAL)
+    if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
+      /* shared-boundary path: when window is a step multiple (the standard
+       * rate(m[5m]) @ 15s grid), the window START of point g is the window
+       * END of point g-dg, so i(g) = j(g-dg) — one seek per point.  Phase 1
+       * caches every j in LDS; phase 2 evaluates with two u16 reads. */
+      int dg64 = (sw.window > 0 && p.step > 0 && sw.window % p.step == 0)
+                     ? (int)(sw.window / p.step) : 0;
+      if (p.jbuf_elems >= p.n_grid && dg64 > 0 && count <= 65535) {
+        for (int g = lane; g < p.n_grid; g += WAVE) {
+          int64_t t_end = p.start + (int64_t)g * p.step;
+          int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
+          jbuf[g] = (uint16_t)vm_ub_hint_fast(lts, count, t_end, gj);
+        }
+        wave_lds_sync();
+        for (int g0 = 0; g0 < p.n_grid; g0 += 4 * WAVE) {
+#pragma unroll
+          for (int u = 0; u < 4; u++) {
+            int g = g0 + u * WAVE + lane;
+            if (g < p.n_grid) {
+              int64_t t_end = p.start + (int64_t)g * p.step;
+              int64_t t_start = t_end - sw.window;
+              int j = jbuf[g];
+              int i;
+              if (g >= dg64) {
+                i = jbuf[g - dg64];
+              } else {
+                int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
+                i = vm_ub_hint_fast(lts, count, t_start, gi);
+              }
+              vm_emit_value(p, io, s, g,
+                            eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
+              scanned += 2;
+            }
+          }
+        }
+        wave_lds_sync();
+        continue;
+      }
+    }
+#endif
     for (int g0 = 0; g0 < p.n_grid; g0 += 4 * WAVE) {
       /* four grid points per lane per outer iteration: independent seek and
        * eval chains that the scheduler interleaves (the phases are LDS-
@@ -1414,6 +1489,13 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
   p.drop_stale = plan->drop_stale_nans;
   p.chunk_wave = (int32_t)std::min<uint32_t>(
       CHUNK_WAVE, std::max<uint32_t>(64, (b.max_wave_len + 63) & ~63u));
+  /* shared-boundary seek (i(g) = j(g - window/step) when window is a step
+   * multiple): give the wave kernel an LDS u16 j-cache when the fused rate
+   * path can use it.  Bounded so LDS stays within occupancy budget. */
+  p.jbuf_elems = 0;
+  if ((plan->func == VMF_RATE || plan->func == VMF_DERIV_FAST) &&
+      n_grid > 1 && n_grid <= 4096)
+    p.jbuf_elems = n_grid;
   p.arg = plan->arg;
   p.arg2 = plan->arg2;
 
@@ -1445,7 +1527,9 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
     w.n_sel = b.n_wave;
     uint32_t blocks = std::min<uint32_t>((b.n_wave + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK,
                                          MAX_WAVE_BLOCKS);
-    size_t lds = (size_t)WAVES_PER_BLOCK * ((size_t)p.chunk_wave * 16 + 256);
+    size_t lds = (size_t)WAVES_PER_BLOCK *
+                 ((size_t)p.chunk_wave * 16 + 256 +
+                  (((size_t)p.jbuf_elems * 2 + 15) & ~(size_t)15));
     launch_rollup(0, blocks, lds, p, w);
   }
   if (b.n_block) {
