@@ -185,6 +185,68 @@ int vmgpu_decode_blocks(const uint8_t* payload, uint64_t payload_len,
  * n_series+1 entries, out_counts[s] = merged length of series s.  Blocks
  * with disjoint time ranges (the common LSM-part case) take a wave-parallel
  * concatenation path; overlapping blocks take the exact heap-merge path. */
+/* Binary operators on the result grid (binary_op.go + metricsql/binaryop).
+ * Label matching (adjustBinaryOpTags, binary_op.go:271) is host metadata
+ * work; these entry points do the per-point math over matched pair lists.
+ * Op ids follow binaryOpFuncs (binary_op.go:15-43). */
+#define VMGPU_BINOP_PLUS 0
+#define VMGPU_BINOP_MINUS 1
+#define VMGPU_BINOP_MUL 2
+#define VMGPU_BINOP_DIV 3
+#define VMGPU_BINOP_MOD 4
+#define VMGPU_BINOP_POW 5
+#define VMGPU_BINOP_ATAN2 6
+#define VMGPU_BINOP_EQ 7
+#define VMGPU_BINOP_NEQ 8
+#define VMGPU_BINOP_GT 9
+#define VMGPU_BINOP_LT 10
+#define VMGPU_BINOP_GTE 11
+#define VMGPU_BINOP_LTE 12
+#define VMGPU_BINOP_DEFAULT 13
+#define VMGPU_BINOP_IF 14
+#define VMGPU_BINOP_IFNOT 15
+#define VMGPU_BINOP_AND 16
+#define VMGPU_BINOP_OR 17
+
+/* newBinaryOpFunc's value loop (binary_op.go:162-236): pair p reads row
+ * left_idx[p] of left_vals (NULL idx = p) and right_idx[p] of right_vals;
+ * fill_left/fill_right mirror the fill()/fill_left()/fill_right()
+ * modifiers; drop_nan_right the vector-comparison NaN rule (:199). */
+int vmgpu_binop_eval(int32_t op, int32_t is_bool, int32_t drop_nan_right,
+                     const double* left_vals, uint32_t n_left_rows,
+                     const uint32_t* left_idx,
+                     const double* right_vals, uint32_t n_right_rows,
+                     const uint32_t* right_idx,
+                     uint32_t n_pairs, uint32_t n_grid,
+                     int32_t has_fill_left, double fill_left,
+                     int32_t has_fill_right, double fill_right,
+                     double* out, char* errbuf, size_t errbuf_len);
+
+/* Set-op masking: mode 0 = and/if (addRightNaNsToLeft, binary_op.go:549),
+ * 1 = unless/ifnot (addLeftNaNsIfNoRightNaNs, :729), 2 = default fill
+ * (fillLeftNaNsWithRightValues, :622).  left row l is masked against the
+ * right rows of its key group (CSR: group_offsets/group_rows); left_vals
+ * modified in place. */
+int vmgpu_binop_mask(int32_t mode, double* left_vals, uint32_t n_left,
+                     const uint32_t* left_group,
+                     const double* right_vals, uint32_t n_right,
+                     const uint32_t* group_offsets, uint32_t n_groups,
+                     const uint32_t* group_rows,
+                     uint32_t n_grid, char* errbuf, size_t errbuf_len);
+
+/* `or` merge walk (fillLeftNaNsWithRightValuesOrMerge, binary_op.go:645):
+ * per key group, the exact left-outer/right-inner consume-and-fill order;
+ * can_merge[(l,r)] (row-major per group, bases in merge_offsets) is
+ * precomputed on the host from marshaled metric names.  Both value
+ * matrices are modified in place. */
+int vmgpu_binop_or(double* left_vals, uint32_t n_left,
+                   double* right_vals, uint32_t n_right,
+                   const uint32_t* lgroup_offsets, const uint32_t* lgroup_rows,
+                   const uint32_t* rgroup_offsets, const uint32_t* rgroup_rows,
+                   const uint8_t* can_merge, const uint64_t* merge_offsets,
+                   uint64_t merge_len, uint32_t n_groups, uint32_t n_grid,
+                   char* errbuf, size_t errbuf_len);
+
 int vmgpu_merge_blocks(const int64_t* ts, const double* vals,
                        const uint64_t* block_offsets, uint32_t n_blocks,
                        const uint32_t* series_block_start, uint32_t n_series,

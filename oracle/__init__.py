@@ -593,3 +593,53 @@ def ensure_non_decreasing(a, v_min, v_max):
     l.vm_ensure_non_decreasing(_ptr(arr, ctypes.c_int64), ctypes.c_int64(len(arr)),
                                ctypes.c_int64(v_min), ctypes.c_int64(v_max))
     return arr
+
+
+# ---------------------------------------------------------------------------
+# binary operator oracle (oracle/binop.c)
+# ---------------------------------------------------------------------------
+
+BINOP_IDS = {
+    "+": 0, "-": 1, "*": 2, "/": 3, "%": 4, "^": 5, "atan2": 6,
+    "==": 7, "!=": 8, ">": 9, "<": 10, ">=": 11, "<=": 12,
+    "default": 13, "if": 14, "ifnot": 15, "and": 16, "or": 17,
+}
+
+
+def _binop_lib():
+    l = lib()
+    if not hasattr(l.vm_binop_scalar, "_typed"):
+        l.vm_binop_scalar.restype = ctypes.c_double
+        l.vm_binop_scalar.argtypes = [ctypes.c_int32, ctypes.c_int32,
+                                      ctypes.c_double, ctypes.c_double]
+        l.vm_binop_apply.restype = None
+        l.vm_binop_scalar._typed = True
+    return l
+
+
+def binop_scalar(op, a, b, is_bool=False):
+    return _binop_lib().vm_binop_scalar(
+        ctypes.c_int32(BINOP_IDS[op] if isinstance(op, str) else op),
+        ctypes.c_int32(1 if is_bool else 0),
+        ctypes.c_double(a), ctypes.c_double(b))
+
+
+def binop_apply(op, a, b, is_bool=False, drop_nan_right=False,
+                fill_left=None, fill_right=None):
+    """newBinaryOpFunc value loop (binary_op.go:162-236) over one pair."""
+    l = _binop_lib()
+    a = np.ascontiguousarray(a, dtype=np.float64)
+    b = np.ascontiguousarray(b, dtype=np.float64)
+    out = np.empty(len(a), dtype=np.float64)
+    l.vm_binop_apply(
+        ctypes.c_int32(BINOP_IDS[op] if isinstance(op, str) else op),
+        ctypes.c_int32(1 if is_bool else 0),
+        ctypes.c_int32(1 if drop_nan_right else 0),
+        _ptr(a, ctypes.c_double), _ptr(b, ctypes.c_double),
+        ctypes.c_int64(len(a)),
+        ctypes.c_int32(0 if fill_left is None else 1),
+        ctypes.c_double(fill_left if fill_left is not None else 0.0),
+        ctypes.c_int32(0 if fill_right is None else 1),
+        ctypes.c_double(fill_right if fill_right is not None else 0.0),
+        _ptr(out, ctypes.c_double))
+    return out

@@ -248,6 +248,84 @@ static __device__ int load_compact_wave(const int64_t* g_ts, const double* g_val
   return count;
 }
 
+/* --- fused phase A+B: staging + removeCounterResets in one pass ------ */
+/* When no sample is dropped (no stale NaNs — the overwhelmingly common
+ * case), the counter-reset scan can run on the staging registers, saving
+ * the whole LDS re-read pass.  Returns the kept count on success, or -1 to
+ * signal "drops present, redo via load_compact_wave + rcr_scan_wave"
+ * (wave-uniform).  Scan semantics identical to rcr_scan_wave below. */
+static __device__ int load_rcr_fused_wave(const int64_t* g_ts, const double* g_vals,
+                                          int64_t n, int64_t* d_ts, double* d_vals,
+                                          bool drop_stale, int64_t msi, int lane) {
+  double corr = 0.0;
+  double prev_raw = 0.0;
+  int64_t prev_ts = 0;
+  double prev_fin = 0.0;
+  for (int64_t base = 0; base < n; base += WAVE) {
+    int64_t k = base + lane;
+    bool active = k < n;
+    double v = 0.0;
+    int64_t t = 0;
+    if (active) {
+      v = g_vals[k];
+      t = g_ts[k];
+    }
+    if (drop_stale && __ballot(active && vm_is_stale_nan(v)) != 0) return -1;
+    if (active) d_ts[k] = t;
+    /* scan round (identical to rcr_scan_wave, values from registers) */
+    double pv = __shfl_up(v, 1);
+    int64_t pt = __shfl_up(t, 1);
+    if (lane == 0) { pv = prev_raw; pt = prev_ts; }
+    bool isfirst = (k == 0);
+    double d = v - pv;
+    double inc = 0.0;
+    if (!isfirst && d < 0) inc = ((-d * 8) < pv) ? (pv - v) : pv;
+    bool gap = (!isfirst && msi > 0 && (t - pt) > msi);
+    uint64_t em = __ballot(active && (gap || inc != 0.0));
+    uint64_t dm = __ballot(active && !isfirst && d < 0);
+    int last = (int)(n - base - 1);
+    if (last > 63) last = 63;
+    if (em == 0 && dm == 0 &&
+        (base == 0 || __shfl(v, 0) + corr >= prev_fin)) {
+      if (active) d_vals[k] = v + corr;
+      prev_raw = __shfl(v, last);
+      prev_ts = __shfl(t, last);
+      prev_fin = prev_raw + corr;
+      continue;
+    }
+    double c = corr;
+    double mycorr = corr;
+    while (em) {
+      int b = __ffsll((unsigned long long)em) - 1;
+      em &= em - 1;
+      double ib = __shfl(inc, b);
+      int gb = __shfl((int)gap, b);
+      double cn = gb ? 0.0 : (c + ib);
+      if (lane >= b) mycorr = cn;
+      c = cn;
+    }
+    double fin = v + mycorr;
+    bool bnd = isfirst || gap;
+    double x = fin;
+    int f = bnd ? 1 : 0;
+    if (lane == 0 && !bnd) x = fmax(x, prev_fin);
+    for (int dlt = 1; dlt < WAVE; dlt <<= 1) {
+      double xo = __shfl_up(x, dlt);
+      int fo = __shfl_up(f, dlt);
+      if (lane >= dlt) {
+        if (!f) x = fmax(x, xo);
+        f = f | fo;
+      }
+    }
+    if (active) d_vals[k] = x;
+    corr = c;
+    prev_raw = __shfl(v, last);
+    prev_ts = __shfl(t, last);
+    prev_fin = __shfl(x, last);
+  }
+  return (int)n;
+}
+
 /* --- phase B: removeCounterResets over the dense column -------------- */
 /* EXACT restatement of rollup.go:921-958 as a wave scan; see file header. */
 static __device__ void rcr_scan_wave(int64_t* d_ts, double* d_vals, int count,
@@ -539,13 +617,26 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
       lvs[k] = io.vals[lo + k];
     }
     wave_lds_sync();
-#else
-    int count = load_compact_wave(io.ts + lo, io.vals + lo, n, lts, lvs,
-                                  p.drop_stale != 0, lane);
-    wave_lds_sync();
-#endif
 #ifndef VMGPU_ABL_NO_RCR
     if (p.rcr) rcr_scan_wave(lts, lvs, count, p.max_staleness, lane);
+    wave_lds_sync();
+#endif
+#else
+    int count = -1;
+#ifndef VMGPU_ABL_NO_RCR
+    if (p.rcr)
+      count = load_rcr_fused_wave(io.ts + lo, io.vals + lo, n, lts, lvs,
+                                  p.drop_stale != 0, p.max_staleness, lane);
+#endif
+    if (count < 0) {
+      /* no rcr, or stale NaNs present: compact first, then scan */
+      count = load_compact_wave(io.ts + lo, io.vals + lo, n, lts, lvs,
+                                p.drop_stale != 0, lane);
+      wave_lds_sync();
+#ifndef VMGPU_ABL_NO_RCR
+      if (p.rcr) rcr_scan_wave(lts, lvs, count, p.max_staleness, lane);
+#endif
+    }
     wave_lds_sync();
 #endif
 

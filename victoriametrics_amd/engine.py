@@ -498,6 +498,111 @@ def decode_blocks(blocks):
     return out_ts, out_vals, offsets
 
 
+def binop_eval(op, is_bool, drop_nan_right, left_mat, left_idx, right_mat,
+               right_idx, fill_left=None, fill_right=None):
+    """Pairwise binary-op values loop (binary_op.go:162-236) on device.
+
+    left_mat/right_mat: [rows x n_grid] f64; pair p = (left_idx[p],
+    right_idx[p]).  Returns [n_pairs x n_grid]."""
+    init()
+    lib = _load_lib()
+    left_mat = np.ascontiguousarray(left_mat, dtype=np.float64)
+    right_mat = np.ascontiguousarray(right_mat, dtype=np.float64)
+    li = np.ascontiguousarray(left_idx, dtype=np.uint32)
+    ri = np.ascontiguousarray(right_idx, dtype=np.uint32)
+    n_pairs = len(li)
+    n_grid = left_mat.shape[1]
+    out = np.empty((n_pairs, n_grid), dtype=np.float64)
+    errbuf = ctypes.create_string_buffer(256)
+    rc = lib.vmgpu_binop_eval(
+        ctypes.c_int32(int(op)), ctypes.c_int32(1 if is_bool else 0),
+        ctypes.c_int32(1 if drop_nan_right else 0),
+        left_mat.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        ctypes.c_uint32(left_mat.shape[0]),
+        li.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        right_mat.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        ctypes.c_uint32(right_mat.shape[0]),
+        ri.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        ctypes.c_uint32(n_pairs), ctypes.c_uint32(n_grid),
+        ctypes.c_int32(0 if fill_left is None else 1),
+        ctypes.c_double(fill_left if fill_left is not None else 0.0),
+        ctypes.c_int32(0 if fill_right is None else 1),
+        ctypes.c_double(fill_right if fill_right is not None else 0.0),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise VmGpuError(f"vmgpu_binop_eval failed ({rc}): "
+                         f"{errbuf.value.decode()}")
+    return out
+
+
+def binop_mask(mode, left_mat, left_group, right_mat, group_offsets):
+    """Set-op masking (and/if, unless/ifnot, default fill) on device;
+    left_mat modified in place."""
+    init()
+    lib = _load_lib()
+    left_mat_c = np.ascontiguousarray(left_mat, dtype=np.float64)
+    right_mat = np.ascontiguousarray(right_mat, dtype=np.float64)
+    lg = np.ascontiguousarray(left_group, dtype=np.uint32)
+    go = np.ascontiguousarray(group_offsets, dtype=np.uint32)
+    rows = np.arange(right_mat.shape[0], dtype=np.uint32)
+    errbuf = ctypes.create_string_buffer(256)
+    rc = lib.vmgpu_binop_mask(
+        ctypes.c_int32(int(mode)),
+        left_mat_c.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        ctypes.c_uint32(left_mat_c.shape[0]),
+        lg.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        right_mat.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        ctypes.c_uint32(right_mat.shape[0]),
+        go.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        ctypes.c_uint32(len(go) - 1),
+        rows.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        ctypes.c_uint32(left_mat_c.shape[1]),
+        errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise VmGpuError(f"vmgpu_binop_mask failed ({rc}): "
+                         f"{errbuf.value.decode()}")
+    left_mat[:] = left_mat_c
+    return left_mat
+
+
+def binop_or(left_mat, right_mat, lgroup_offsets, lgroup_rows,
+             rgroup_offsets, rgroup_rows, can_merge, merge_offsets):
+    """`or` merge walk (binary_op.go:645) on device; both matrices
+    modified in place."""
+    init()
+    lib = _load_lib()
+    lm = np.ascontiguousarray(left_mat, dtype=np.float64)
+    rm = np.ascontiguousarray(right_mat, dtype=np.float64)
+    lo = np.ascontiguousarray(lgroup_offsets, dtype=np.uint32)
+    lr = np.ascontiguousarray(lgroup_rows, dtype=np.uint32)
+    ro = np.ascontiguousarray(rgroup_offsets, dtype=np.uint32)
+    rr = np.ascontiguousarray(rgroup_rows, dtype=np.uint32)
+    cm = np.ascontiguousarray(can_merge, dtype=np.uint8)
+    mo = np.ascontiguousarray(merge_offsets, dtype=np.uint64)
+    errbuf = ctypes.create_string_buffer(256)
+    rc = lib.vmgpu_binop_or(
+        lm.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        ctypes.c_uint32(lm.shape[0]),
+        rm.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        ctypes.c_uint32(rm.shape[0]),
+        lo.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        lr.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        ro.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        rr.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        cm.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        mo.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        ctypes.c_uint64(cm.size), ctypes.c_uint32(len(mo)),
+        ctypes.c_uint32(lm.shape[1]),
+        errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise VmGpuError(f"vmgpu_binop_or failed ({rc}): "
+                         f"{errbuf.value.decode()}")
+    left_mat[:] = lm
+    right_mat[:] = rm
+    return left_mat, right_mat
+
+
 def merge_blocks(ts, vals, block_offsets, series_block_start,
                  dedup_interval=0):
     """GPU per-series k-way merge of decoded blocks + dedup
