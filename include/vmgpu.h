@@ -247,6 +247,44 @@ int vmgpu_binop_or(double* left_vals, uint32_t n_left,
                    uint64_t merge_len, uint32_t n_groups, uint32_t n_grid,
                    char* errbuf, size_t errbuf_len);
 
+/* Transform functions on the result grid (transform.go).  Elementwise ids
+ * below VMGPU_TF_SERIES_BASE run flat over the matrix; per-series ids walk
+ * columns sequentially (one thread per series).  arg1/arg2 are per-grid
+ * scalar-arg rows (getScalar semantics); scalar_arg the phi/k/z scalars. */
+enum {
+  VMGPU_TF_ABS = 0, VMGPU_TF_CEIL, VMGPU_TF_FLOOR, VMGPU_TF_EXP,
+  VMGPU_TF_LN, VMGPU_TF_LOG2, VMGPU_TF_LOG10, VMGPU_TF_SQRT,
+  VMGPU_TF_SIN, VMGPU_TF_COS, VMGPU_TF_TAN, VMGPU_TF_ASIN, VMGPU_TF_ACOS,
+  VMGPU_TF_ATAN, VMGPU_TF_SINH, VMGPU_TF_COSH, VMGPU_TF_TANH,
+  VMGPU_TF_ASINH, VMGPU_TF_ACOSH, VMGPU_TF_ATANH, VMGPU_TF_DEG,
+  VMGPU_TF_RAD, VMGPU_TF_SGN, VMGPU_TF_CLAMP, VMGPU_TF_CLAMP_MIN,
+  VMGPU_TF_CLAMP_MAX, VMGPU_TF_ROUND, VMGPU_TF_BITMAP_AND,
+  VMGPU_TF_BITMAP_OR, VMGPU_TF_BITMAP_XOR, VMGPU_TF_DAY_OF_MONTH,
+  VMGPU_TF_DAY_OF_WEEK, VMGPU_TF_DAY_OF_YEAR, VMGPU_TF_DAYS_IN_MONTH,
+  VMGPU_TF_HOUR, VMGPU_TF_MINUTE, VMGPU_TF_MONTH, VMGPU_TF_YEAR,
+
+  VMGPU_TF_SERIES_BASE = 100,
+  VMGPU_TF_KEEP_LAST_VALUE = 100, VMGPU_TF_KEEP_NEXT_VALUE,
+  VMGPU_TF_INTERPOLATE, VMGPU_TF_RUNNING_SUM, VMGPU_TF_RUNNING_MIN,
+  VMGPU_TF_RUNNING_MAX, VMGPU_TF_RUNNING_AVG, VMGPU_TF_RANGE_SUM,
+  VMGPU_TF_RANGE_MIN, VMGPU_TF_RANGE_MAX, VMGPU_TF_RANGE_AVG,
+  VMGPU_TF_RANGE_FIRST, VMGPU_TF_RANGE_LAST, VMGPU_TF_RANGE_NORMALIZE,
+  VMGPU_TF_RANGE_ZSCORE, VMGPU_TF_RANGE_TRIM_ZSCORE, VMGPU_TF_RANGE_STDDEV,
+  VMGPU_TF_RANGE_STDVAR, VMGPU_TF_RANGE_LINREG, VMGPU_TF_RANGE_MAD,
+  VMGPU_TF_RANGE_TRIM_OUTLIERS, VMGPU_TF_RANGE_TRIM_SPIKES,
+  VMGPU_TF_RANGE_QUANTILE, VMGPU_TF_SMOOTH_EXPONENTIAL,
+  VMGPU_TF_REMOVE_RESETS
+};
+
+/* values: [n_series x n_grid] f64, transformed in place.  ts: the shared
+ * grid timestamps (range_linear_regression).  keep_flags (nullable,
+ * [n_series]): 0 = series dropped from the result (range_normalize on an
+ * all-NaN series, transform.go:1383). */
+int vmgpu_transform(int32_t func, double* values, uint32_t n_series,
+                    uint32_t n_grid, const int64_t* ts, const double* arg1,
+                    const double* arg2, double scalar_arg,
+                    uint8_t* keep_flags, char* errbuf, size_t errbuf_len);
+
 int vmgpu_merge_blocks(const int64_t* ts, const double* vals,
                        const uint64_t* block_offsets, uint32_t n_blocks,
                        const uint32_t* series_block_start, uint32_t n_series,

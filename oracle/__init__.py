@@ -643,3 +643,27 @@ def binop_apply(op, a, b, is_bool=False, drop_nan_right=False,
         ctypes.c_double(fill_right if fill_right is not None else 0.0),
         _ptr(out, ctypes.c_double))
     return out
+
+
+# ---------------------------------------------------------------------------
+# transform oracle (oracle/transform.c); func ids = include/vmgpu.h VMGPU_TF_*
+# ---------------------------------------------------------------------------
+
+def tf_apply(func_id, values, ts=None, arg1=None, arg2=None, scalar=0.0):
+    """Apply a transform func to [n_series x n_grid] values in place;
+    returns (values, keep_flags)."""
+    l = lib()
+    l.vm_tf_apply.restype = None
+    v = np.ascontiguousarray(values, dtype=np.float64)
+    ns, ng = v.shape
+    t = np.ascontiguousarray(ts, dtype=np.int64) if ts is not None else None
+    a1 = np.ascontiguousarray(arg1, dtype=np.float64) if arg1 is not None else None
+    a2 = np.ascontiguousarray(arg2, dtype=np.float64) if arg2 is not None else None
+    keep = np.ones(ns, dtype=np.uint8)
+    l.vm_tf_apply(ctypes.c_int32(func_id), _ptr(v, ctypes.c_double),
+                  ctypes.c_int64(ns), ctypes.c_int64(ng),
+                  _ptr(t, ctypes.c_int64) if t is not None else None,
+                  _ptr(a1, ctypes.c_double) if a1 is not None else None,
+                  _ptr(a2, ctypes.c_double) if a2 is not None else None,
+                  ctypes.c_double(scalar), _ptr(keep, ctypes.c_uint8))
+    return v, keep

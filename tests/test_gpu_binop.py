@@ -25,6 +25,24 @@ def _rand(rng, n, nan_frac=0.2):
     return v
 
 
+# pow/atan2 are software transcendentals: the device (OCML) and host
+# (glibc, and Go's own FDLIBM ports) implementations legitimately differ in
+# the last ulps.  The reference's own tests compare these with tolerance
+# (exec_test.go timeseriesEqual); everything else is bit-exact.
+ULP_OPS = {"^", "atan2"}
+
+
+def _assert_op_equal(op, got, exp, msg):
+    if op in ULP_OPS:
+        both_nan = np.isnan(got) & np.isnan(exp)
+        np.testing.assert_allclose(got[~both_nan], exp[~both_nan],
+                                   rtol=1e-12, atol=0, err_msg=msg)
+        assert np.array_equal(np.isnan(got), np.isnan(exp)), msg
+    else:
+        np.testing.assert_array_equal(got.view(np.int64), exp.view(np.int64),
+                                      err_msg=msg)
+
+
 @pytest.mark.parametrize("op", ALL_OPS)
 @pytest.mark.parametrize("is_bool", [False, True])
 def test_pairs_kernel_all_ops(op, is_bool):
@@ -40,9 +58,7 @@ def test_pairs_kernel_all_ops(op, is_bool):
                             np.arange(n_pairs, dtype=np.uint32))
     for p in range(n_pairs):
         exp = oracle.binop_apply(op, a[p], b[p], is_bool=is_bool)
-        np.testing.assert_array_equal(got[p].view(np.int64),
-                                      exp.view(np.int64),
-                                      err_msg=f"{op} bool={is_bool} row {p}")
+        _assert_op_equal(op, got[p], exp, f"{op} bool={is_bool} row {p}")
 
 
 def test_pairs_kernel_fills_and_dropnan():
@@ -199,7 +215,8 @@ def test_histogram_quantile_pipeline_on_device():
     from victoriametrics_amd import synth
     from victoriametrics_amd.engine import RollupPlan, SeriesBatch
     n_series, n_samples = 256, 120
-    ts, vals, offsets = synth.counter_batch(n_series, n_samples)
+    ts, vals, offsets = synth.counter_batch(n_series, n_samples,
+                                            1_000_000_000_000)
     start = int(ts[0]) + 60_000
     plan = RollupPlan("rate", start, start + 30 * 15_000, 15_000,
                       window=300_000)

@@ -1,0 +1,239 @@
+"""Transform layer, CPU side: oracle value-semantics pins (hand-derived
+from transform.go) + the host label/metadata funcs + the decimal exponent
+port transformRound needs."""
+import math
+
+import numpy as np
+import pytest
+
+import oracle
+from victoriametrics_amd import transform as tf
+from victoriametrics_amd.binary_op import Series
+from victoriametrics_amd.metric_name import MetricName
+
+NAN = math.nan
+
+
+def row(*vals):
+    return np.asarray([list(vals)], dtype=np.float64)
+
+
+def S(name, tags, values):
+    return Series(MetricName(name, tags), np.asarray(values, np.float64))
+
+
+# ---------------------------------------------------------------------------
+# oracle value semantics
+# ---------------------------------------------------------------------------
+
+def test_keep_last_value():
+    out, _ = oracle.tf_apply(100, row(NAN, 1, NAN, NAN, 4, NAN))
+    # leading NaN keeps values[0] (= NaN seed), gaps carry last value
+    assert math.isnan(out[0, 0])
+    assert list(out[0, 1:]) == [1, 1, 1, 4, 4]
+
+
+def test_keep_next_value():
+    out, _ = oracle.tf_apply(101, row(NAN, 1, NAN, 4, NAN))
+    assert out[0, 0] == 1 and out[0, 2] == 4 and math.isnan(out[0, 4])
+
+
+def test_interpolate():
+    out, _ = oracle.tf_apply(102, row(NAN, 1.0, NAN, NAN, 4.0, NAN))
+    # leading/trailing NaNs untouched; interior linear
+    assert math.isnan(out[0, 0]) and math.isnan(out[0, 5])
+    np.testing.assert_allclose(out[0, 1:5], [1, 2, 3, 4])
+
+
+def test_running_and_range_sum():
+    out, _ = oracle.tf_apply(103, row(1, 2, NAN, 3))
+    assert list(out[0]) == [1, 3, 3, 6]
+    out2, _ = oracle.tf_apply(107, row(1, 2, NAN, 3))
+    assert list(out2[0]) == [6, 6, 6, 6]
+
+
+def test_running_avg():
+    out, _ = oracle.tf_apply(106, row(2, 4, 6))
+    np.testing.assert_allclose(out[0], [2, 3, 4])
+
+
+def test_range_first_last():
+    out, _ = oracle.tf_apply(111, row(NAN, 5, 7, NAN))
+    assert list(out[0]) == [5, 5, 5, 5]
+    out2, _ = oracle.tf_apply(112, row(NAN, 5, 7, NAN))
+    assert list(out2[0]) == [7, 7, 7, 7]
+
+
+def test_range_normalize_keep_flag():
+    v = np.asarray([[1.0, 3.0, 2.0], [NAN, NAN, NAN]])
+    out, keep = oracle.tf_apply(113, v)
+    np.testing.assert_allclose(out[0], [0, 1, 0.5])
+    assert keep[0] == 1 and keep[1] == 0
+
+
+def test_range_quantile_and_median():
+    out, _ = oracle.tf_apply(122, row(4, 1, 3, 2), scalar=0.5)
+    np.testing.assert_allclose(out[0], [2.5] * 4)
+
+
+def test_range_trim_spikes():
+    vals = list(range(10)) + [1000.0, -1000.0]
+    out, _ = oracle.tf_apply(121, row(*vals), scalar=0.2)
+    assert math.isnan(out[0, 10]) and math.isnan(out[0, 11])
+    assert not math.isnan(out[0, 5])
+
+
+def test_range_zscore_properties():
+    rng = np.random.default_rng(5)
+    v = rng.standard_normal((1, 50)) * 7 + 3
+    out, _ = oracle.tf_apply(114, v.copy())
+    assert abs(float(np.mean(out))) < 1e-12
+    assert abs(float(np.std(out)) - 1.0) < 1e-12
+
+
+def test_remove_resets_matches_rollup_semantics():
+    # removeCounterResetsMaybeNaNs: NaNs skipped, partial-reset heuristic
+    out, _ = oracle.tf_apply(124, row(10, 12, NAN, 2, 15))
+    assert list(out[0])[:2] == [10, 12]
+    assert math.isnan(out[0, 2])
+    assert list(out[0])[3:] == [14, 27]
+
+
+def test_smooth_exponential():
+    sfs = [0.5] * 4
+    out, _ = oracle.tf_apply(123, row(10.0, 20.0, 30.0, NAN), arg1=sfs)
+    assert out[0, 0] == 10 and out[0, 1] == 15 and out[0, 2] == 22.5
+    assert math.isnan(out[0, 3])
+
+
+def test_elementwise_specials():
+    out, _ = oracle.tf_apply(22, row(-3.5, 0.0, 2.0, NAN))  # sgn
+    # transformSgn has no NaN check: NaN compares false both ways -> 0
+    assert list(out[0]) == [-1, 0, 1, 0]
+    out2, _ = oracle.tf_apply(23, row(1.0, 5.0, 9.0), arg1=[2, 2, 2],
+                              arg2=[8, 8, 8])  # clamp
+    assert list(out2[0]) == [2, 5, 8]
+
+
+def test_bitmap_go_semantics():
+    out, _ = oracle.tf_apply(27, row(6.0, 255.0, NAN), arg1=[3.0, 3.0, 3.0])
+    assert list(out[0])[:2] == [2.0, 3.0] and math.isnan(out[0, 2])
+    out2, _ = oracle.tf_apply(29, row(6.0), arg1=[3.0])  # xor
+    assert out2[0, 0] == 5.0
+
+
+def test_datetime_funcs():
+    # 2021-03-14 15:09:26 UTC = 1615734566
+    t = 1615734566.0
+    cases = {30: 14, 31: 0, 32: 73, 33: 31, 34: 15, 35: 9, 36: 3, 37: 2021}
+    for fid, exp in cases.items():
+        out, _ = oracle.tf_apply(fid, row(t))
+        assert out[0, 0] == exp, (fid, out[0, 0], exp)
+    # leap year Feb
+    t2 = 1582934400.0  # 2020-02-29
+    out, _ = oracle.tf_apply(33, row(t2))
+    assert out[0, 0] == 29
+
+
+def test_range_linear_regression():
+    ts = np.arange(4, dtype=np.int64) * 1000
+    out, _ = oracle.tf_apply(118, row(1.0, 3.0, 5.0, 7.0), ts=ts)
+    np.testing.assert_allclose(out[0], [1, 3, 5, 7], rtol=1e-12)
+    # const fast path
+    out2, _ = oracle.tf_apply(118, row(4.0, 4.0, 4.0, 4.0), ts=ts)
+    assert list(out2[0]) == [4, 4, 4, 4]
+
+
+# ---------------------------------------------------------------------------
+# transformRound's decimal exponent port
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("nearest,expected_p10", [
+    (1.0, 1.0), (0.1, 10.0), (0.01, 100.0), (0.5, 10.0), (10.0, 0.1),
+    (100.0, 0.01), (0.25, 100.0), (2.0, 1.0), (1000.0, 0.001),
+])
+def test_decimal_exponent(nearest, expected_p10):
+    e = tf.decimal_from_float_exponent(nearest)
+    assert math.pow(10.0, -e) == expected_p10, (nearest, e)
+
+
+def test_round_host_path_matches_oracle():
+    vals = np.asarray([[1.234, 5.678, -2.345, 0.05]])
+    nearest = np.asarray([0.1, 0.1, 0.1, 0.1])
+    p10 = np.asarray([10.0] * 4)
+    exp, _ = oracle.tf_apply(26, vals.copy(), arg1=nearest, arg2=p10)
+    np.testing.assert_allclose(exp[0], [1.2, 5.7, -2.3, 0.1])
+
+
+# ---------------------------------------------------------------------------
+# host label funcs
+# ---------------------------------------------------------------------------
+
+def test_label_set_del_keep():
+    s = S("m", [("a", "1"), ("b", "2")], [1])
+    tf.label_set([s], [("c", "3"), ("a", "9")])
+    assert s.mn.get_tag_value("c") == b"3" and s.mn.get_tag_value("a") == b"9"
+    tf.label_del([s], ["b"])
+    assert s.mn.get_tag_value("b") is None
+    tf.label_keep([s], ["a"])
+    assert s.mn.get_tag_value("c") is None
+    assert s.mn.metric_group == b""  # __name__ not kept
+
+
+def test_label_copy_move():
+    s = S("m", [("src", "x")], [1])
+    tf.label_copy([s], [("src", "dst")])
+    assert s.mn.get_tag_value("dst") == b"x"
+    assert s.mn.get_tag_value("src") == b"x"
+    s2 = S("m", [("src", "y")], [1])
+    tf.label_move([s2], [("src", "dst")])
+    assert s2.mn.get_tag_value("dst") == b"y"
+    assert s2.mn.get_tag_value("src") is None
+
+
+def test_label_join_replace():
+    s = S("m", [("a", "x"), ("b", "y")], [1])
+    tf.label_join([s], "joined", "-", ["a", "b"])
+    assert s.mn.get_tag_value("joined") == b"x-y"
+    tf.label_replace([s], "r", "$1!", "joined", "(x)-y")
+    assert s.mn.get_tag_value("r") == b"x!"
+    # non-matching regex leaves things alone
+    tf.label_replace([s], "r2", "$1", "joined", "zzz")
+    assert s.mn.get_tag_value("r2") is None
+
+
+def test_label_match_mismatch():
+    a = S("m", [("env", "prod")], [1])
+    b = S("m", [("env", "dev")], [1])
+    assert tf.label_match([a, b], "env", "prod") == [a]
+    assert tf.label_match([a, b], "env", "prod", negate=True) == [b]
+
+
+def test_drop_common_labels():
+    a = S("m", [("common", "c"), ("x", "1")], [1])
+    b = S("m", [("common", "c"), ("x", "2")], [1])
+    tf.drop_common_labels([a, b])
+    assert a.mn.get_tag_value("common") is None
+    assert a.mn.get_tag_value("x") == b"1"
+    assert a.mn.metric_group == b""  # __name__ common -> dropped
+
+
+def test_sort_series():
+    a = S("a", [], [1, 5])
+    b = S("b", [], [9, NAN])   # last non-NaN = 9
+    c = S("c", [], [NAN, NAN])
+    out = tf.sort_series([a, b, c])
+    assert out == [c, a, b]
+    out_desc = tf.sort_series([a, b, c], desc=True)
+    assert out_desc == [b, a, c]
+
+
+def test_sort_by_label():
+    a = S("m", [("k", "2")], [1])
+    b = S("m", [("k", "1")], [1])
+    assert tf.sort_by_label([a, b], ["k"]) == [b, a]
+
+
+def test_limit_offset():
+    ss = [S("m", [("i", str(i))], [1]) for i in range(5)]
+    assert tf.limit_offset(ss, 2, 1) == ss[1:3]

@@ -55,16 +55,40 @@ static __device__ __forceinline__ double b_nan() {
 
 /* metricsql/binaryop/funcs.go, applied with the bool-modifier wrapper from
  * newBinaryOpCmpFunc (binary_op.go:136-153). */
+/* x86-64 (and Go-on-amd64) SSE NaN propagation: the FIRST NaN operand's
+ * payload, quieted.  CDNA VALU ops instead propagate through the negated/
+ * canonicalized encoding (a-b lowers to add(a,-b), flipping a NaN b's sign
+ * bit), so arith ops route NaNs explicitly to stay bit-identical with the
+ * host path. */
+static __device__ __forceinline__ double b_quiet(double x) {
+  return __longlong_as_double(__double_as_longlong(x) |
+                              0x0008000000000000LL);
+}
+
+#define B_ARITH(expr)                                      \
+  do {                                                     \
+    if (isnan(a)) return b_quiet(a);                       \
+    if (isnan(b)) return b_quiet(b);                       \
+    return (expr);                                         \
+  } while (0)
+
 static __device__ double binop_apply(int op, int is_bool, double a, double b) {
   bool cmp_hit = false;
   switch (op) {
-    case VMGPU_BINOP_PLUS:  return a + b;
-    case VMGPU_BINOP_MINUS: return a - b;
-    case VMGPU_BINOP_MUL:   return a * b;
-    case VMGPU_BINOP_DIV:   return a / b;
-    case VMGPU_BINOP_MOD:   return fmod(a, b);
-    case VMGPU_BINOP_POW:   return isnan(a) ? b_nan() : pow(a, b);
-    case VMGPU_BINOP_ATAN2: return atan2(a, b);
+    case VMGPU_BINOP_PLUS:  B_ARITH(a + b);
+    case VMGPU_BINOP_MINUS: B_ARITH(a - b);
+    case VMGPU_BINOP_MUL:   B_ARITH(a * b);
+    case VMGPU_BINOP_DIV:   B_ARITH(a / b);
+    case VMGPU_BINOP_MOD:   B_ARITH(fmod(a, b));
+    case VMGPU_BINOP_POW:
+      /* binaryop.Pow: NaN^any = NaN; then C99 pow edge cases */
+      if (isnan(a)) return b_nan();
+      if (isnan(b)) return (a == 1.0) ? 1.0 : b_quiet(b);
+      return pow(a, b);
+    case VMGPU_BINOP_ATAN2:
+      if (isnan(a)) return b_quiet(a);
+      if (isnan(b)) return b_quiet(b);
+      return atan2(a, b);
     case VMGPU_BINOP_EQ:    cmp_hit = isnan(a) ? isnan(b) : (a == b); break;
     case VMGPU_BINOP_NEQ:
       cmp_hit = isnan(a) ? !isnan(b) : (isnan(b) ? true : (a != b));

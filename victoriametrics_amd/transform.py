@@ -1,0 +1,368 @@
+"""Transform-function layer — host mirror of transform.go's dispatch.
+
+Value math runs on the GPU (vmgpu_transform, csrc/transform.hip); label
+functions (label_*, sort_by_label, drop_common_labels, limit_offset) are
+host metadata work mirrored here on MetricName objects, exactly as the
+reference keeps them on the Go side of the C-ABI seam.
+
+Value funcs operate on a [n_series x n_grid] matrix in place and take
+scalar args as per-grid rows (getScalar semantics, eval.go): `transform(
+"clamp", values, args=[min_row, max_row])`.
+"""
+import ctypes
+import math
+import re
+
+import numpy as np
+
+from . import engine
+from .metric_name import MetricName
+
+# func name -> (device id, n scalar-arg rows, scalar_arg index or None)
+_ELEMENTWISE = {
+    "abs": 0, "ceil": 1, "floor": 2, "exp": 3, "ln": 4, "log2": 5,
+    "log10": 6, "sqrt": 7, "sin": 8, "cos": 9, "tan": 10, "asin": 11,
+    "acos": 12, "atan": 13, "sinh": 14, "cosh": 15, "tanh": 16,
+    "asinh": 17, "acosh": 18, "atanh": 19, "deg": 20, "rad": 21, "sgn": 22,
+}
+_CLAMP = {"clamp": 23, "clamp_min": 24, "clamp_max": 25}
+_ROUND = 26
+_BITMAP = {"bitmap_and": 27, "bitmap_or": 28, "bitmap_xor": 29}
+_DATETIME = {
+    "day_of_month": 30, "day_of_week": 31, "day_of_year": 32,
+    "days_in_month": 33, "hour": 34, "minute": 35, "month": 36, "year": 37,
+}
+_SERIES = {
+    "keep_last_value": 100, "keep_next_value": 101, "interpolate": 102,
+    "running_sum": 103, "running_min": 104, "running_max": 105,
+    "running_avg": 106, "range_sum": 107, "range_min": 108,
+    "range_max": 109, "range_avg": 110, "range_first": 111,
+    "range_last": 112, "range_normalize": 113, "range_zscore": 114,
+    "range_trim_zscore": 115, "range_stddev": 116, "range_stdvar": 117,
+    "range_linear_regression": 118, "range_mad": 119,
+    "range_trim_outliers": 120, "range_trim_spikes": 121,
+    "range_quantile": 122, "smooth_exponential": 123, "remove_resets": 124,
+}
+# funcs whose MetricName group must be reset (transform.go:142
+# transformFuncsKeepMetricName lists the keepers; everything else resets)
+KEEP_METRIC_NAME_FUNCS = {
+    "ceil", "clamp", "clamp_min", "clamp_max", "floor", "interpolate",
+    "keep_last_value", "keep_next_value", "range_avg", "range_first",
+    "range_last", "range_linear_regression", "range_max", "range_median",
+    "range_min", "range_normalize", "range_quantile", "range_stddev",
+    "range_stdvar", "range_sum", "range_trim_outliers", "range_trim_spikes",
+    "range_trim_zscore", "range_zscore", "round", "running_avg",
+    "running_max", "running_min", "running_sum", "smooth_exponential",
+}
+
+
+def decimal_from_float_exponent(f):
+    """FromFloat's decimal exponent (lib/decimal/decimal.go:437 +
+    positiveFloatToDecimal :467) — the host-side piece transformRound
+    needs for its p10 (transform.go:2357-2361)."""
+    if f == 0 or math.isnan(f) or math.isinf(f):
+        return 0
+    f = abs(f)
+    u = int(f)
+    if float(u) == f:
+        if u < (1 << 55) and u % 10 != 0:
+            return 0
+        # getDecimalAndScale (decimal.go:480)
+        scale = 0
+        v = u
+        if v < (1 << 55) and v % 10 != 0:
+            return scale
+        while v >= (1 << 55):
+            v //= 10
+            scale += 1
+        if v % 10 != 0:
+            return scale
+        v //= 10
+        scale += 1
+        while v != 0 and v % 10 == 0:
+            v //= 10
+            scale += 1
+        return scale
+    # positiveFloatToDecimalSlow (decimal.go:502)
+    scale = 0
+    prec = 1e12
+    if f > 1e6 or f < 1e-6:
+        if f > 1e6:
+            prec = 1e15
+        _, exp = math.frexp(f)
+        exp = max(-1022, min(1023, exp))
+        scale = int(exp * (math.log(2) / math.log(10)))
+        f *= math.pow(10.0, -scale)
+    while f < prec:
+        frac, x = math.modf(f)
+        if frac * prec < x:
+            f = x
+            break
+        if (1 - frac) * prec < x:
+            f = x + 1
+            break
+        f *= 100
+        scale -= 2
+    u = int(f)
+    if u % 10 != 0:
+        return scale
+    return scale + 1
+
+
+def _call(func_id, values, ts=None, arg1=None, arg2=None, scalar_arg=0.0,
+          want_keep=False):
+    engine.init()
+    lib = engine._load_lib()
+    v = np.ascontiguousarray(values, dtype=np.float64)
+    n_series, n_grid = v.shape
+    a1 = np.ascontiguousarray(arg1, dtype=np.float64) if arg1 is not None else None
+    a2 = np.ascontiguousarray(arg2, dtype=np.float64) if arg2 is not None else None
+    t = np.ascontiguousarray(ts, dtype=np.int64) if ts is not None else None
+    keep = np.ones(n_series, dtype=np.uint8) if want_keep else None
+    errbuf = ctypes.create_string_buffer(256)
+    dp = ctypes.POINTER(ctypes.c_double)
+    rc = lib.vmgpu_transform(
+        ctypes.c_int32(func_id), v.ctypes.data_as(dp),
+        ctypes.c_uint32(n_series), ctypes.c_uint32(n_grid),
+        t.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)) if t is not None else None,
+        a1.ctypes.data_as(dp) if a1 is not None else None,
+        a2.ctypes.data_as(dp) if a2 is not None else None,
+        ctypes.c_double(scalar_arg),
+        keep.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)) if keep is not None else None,
+        errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise engine.VmGpuError(f"vmgpu_transform failed ({rc}): "
+                                f"{errbuf.value.decode()}")
+    values[:] = v
+    return (values, keep) if want_keep else values
+
+
+def transform(name, values, ts=None, args=(), scalar=0.0):
+    """Apply one transform func to the [n_series x n_grid] matrix in place.
+
+    args: scalar-arg rows [n_grid] in the reference's argument order
+    (clamp: min,max; round: nearest; bitmap_*: mask; smooth_exponential:
+    smoothing factors).  scalar: phi/k/z for the range_trim/quantile funcs.
+    Returns values (and for range_normalize a keep mask)."""
+    name = name.lower()
+    if name in _ELEMENTWISE:
+        return _call(_ELEMENTWISE[name], values)
+    if name in _CLAMP:
+        return _call(_CLAMP[name], values,
+                     arg1=args[0] if len(args) > 0 else None,
+                     arg2=args[1] if len(args) > 1 else None)
+    if name == "round":
+        nearest = np.ascontiguousarray(
+            args[0] if args else np.ones(values.shape[1]), dtype=np.float64)
+        # p10 per point, hoisting the reference's nPrev cache host-side
+        p10 = np.empty_like(nearest)
+        prev_n, prev_p = None, 1.0
+        for i, nv in enumerate(nearest):
+            if nv != prev_n:
+                prev_n = nv
+                # p10 = Pow10(-e) of FromFloat(nearest) (transform.go:2360)
+                prev_p = math.pow(10.0, -decimal_from_float_exponent(nv))
+            p10[i] = prev_p
+        return _call(_ROUND, values, arg1=nearest, arg2=p10)
+    if name in _BITMAP:
+        return _call(_BITMAP[name], values, arg1=args[0])
+    if name in _DATETIME:
+        return _call(_DATETIME[name], values)
+    if name in _SERIES:
+        fid = _SERIES[name]
+        if name == "range_normalize":
+            return _call(fid, values, want_keep=True)
+        if name == "smooth_exponential":
+            return _call(fid, values, arg1=args[0] if args else None)
+        if name == "range_linear_regression":
+            return _call(fid, values, ts=ts)
+        return _call(fid, values, scalar_arg=scalar)
+    raise ValueError(f"unknown transform func {name!r}")
+
+
+# ---------------------------------------------------------------------------
+# label funcs — host metadata work (transform.go label_* family)
+# ---------------------------------------------------------------------------
+
+def reset_metric_group_if_needed(name, series_list):
+    """doTransformValues resets MetricGroup unless the func keeps the
+    physical meaning (transform.go:142-209) or keep_metric_names is set."""
+    if name.lower() in KEEP_METRIC_NAME_FUNCS:
+        return
+    for s in series_list:
+        s.mn.reset_metric_group()
+
+
+def label_set(series_list, pairs):
+    # transformLabelSet (transform.go): (label, value) literal pairs
+    for s in series_list:
+        for k, v in pairs:
+            if k == "__name__":
+                s.mn.metric_group = MetricName._b(v)
+            elif v == "":
+                s.mn.remove_tag(k)
+            else:
+                s.mn.set_tag(k, v)
+    return series_list
+
+
+def label_del(series_list, labels):
+    for s in series_list:
+        for k in labels:
+            s.mn.remove_tag(k)
+    return series_list
+
+
+def label_keep(series_list, labels):
+    keep = {MetricName._b(k) for k in labels}
+    for s in series_list:
+        if b"__name__" not in keep:
+            s.mn.reset_metric_group()
+        s.mn.tags = [(k, v) for k, v in s.mn.tags if k in keep]
+    return series_list
+
+
+def label_copy(series_list, pairs, remove_src=False):
+    # transformLabelCopy / transformLabelMove (src, dst pairs)
+    for s in series_list:
+        for src, dst in pairs:
+            v = s.mn.get_tag_value(src)
+            if v is None or len(v) == 0:
+                continue
+            if dst == "__name__":
+                s.mn.metric_group = v
+            else:
+                s.mn.set_tag(dst, v)
+            if remove_src and src != dst:
+                s.mn.remove_tag(src)
+    return series_list
+
+
+def label_move(series_list, pairs):
+    return label_copy(series_list, pairs, remove_src=True)
+
+
+def label_join(series_list, dst_label, separator, src_labels):
+    # transformLabelJoin
+    sep = MetricName._b(separator)
+    for s in series_list:
+        parts = []
+        for k in src_labels:
+            v = s.mn.get_tag_value(k)
+            parts.append(v if v is not None else b"")
+        joined = sep.join(parts)
+        if dst_label == "__name__":
+            s.mn.metric_group = joined
+        else:
+            s.mn.set_tag(dst_label, joined)
+    return series_list
+
+
+def label_replace(series_list, dst_label, replacement, src_label, regex):
+    # transformLabelReplace: RE2 full-anchored match semantics
+    pat = re.compile("^(?:" + regex + ")$")
+    for s in series_list:
+        v = s.mn.get_tag_value(src_label)
+        src = (v or b"").decode("utf-8", "surrogateescape")
+        m = pat.match(src)
+        if m is None:
+            continue
+        val = m.expand(re.sub(r"\$(\d+|\{\w+\})",
+                              lambda g: "\\" + g.group(1).strip("{}"),
+                              replacement))
+        if dst_label == "__name__":
+            s.mn.metric_group = val.encode()
+        elif val == "":
+            s.mn.remove_tag(dst_label)
+        else:
+            s.mn.set_tag(dst_label, val)
+    return series_list
+
+
+def label_uppercase(series_list, labels):
+    for s in series_list:
+        for k in labels:
+            v = s.mn.get_tag_value(k)
+            if v is not None:
+                target_up = v.decode("utf-8", "surrogateescape").upper()
+                if k == "__name__":
+                    s.mn.metric_group = target_up.encode()
+                else:
+                    s.mn.set_tag(k, target_up)
+    return series_list
+
+
+def label_lowercase(series_list, labels):
+    for s in series_list:
+        for k in labels:
+            v = s.mn.get_tag_value(k)
+            if v is not None:
+                low = v.decode("utf-8", "surrogateescape").lower()
+                if k == "__name__":
+                    s.mn.metric_group = low.encode()
+                else:
+                    s.mn.set_tag(k, low)
+    return series_list
+
+
+def label_match(series_list, label, regex, negate=False):
+    # transformLabelMatch / transformLabelMismatch
+    pat = re.compile("^(?:" + regex + ")$")
+    out = []
+    for s in series_list:
+        v = (s.mn.get_tag_value(label) or b"").decode("utf-8",
+                                                      "surrogateescape")
+        hit = pat.match(v) is not None
+        if hit != negate:
+            out.append(s)
+    return out
+
+
+def drop_common_labels(series_list):
+    # transformDropCommonLabels: remove (k,v) present on every series;
+    # __name__ participates as a label
+    if not series_list:
+        return series_list
+    from collections import Counter
+    counts = Counter()
+    for s in series_list:
+        seen = {(b"__name__", s.mn.metric_group)} if s.mn.metric_group else set()
+        for kv in s.mn.tags:
+            seen.add(kv)
+        for kv in seen:
+            counts[kv] += 1
+    n = len(series_list)
+    common = {kv for kv, c in counts.items() if c == n}
+    for s in series_list:
+        if (b"__name__", s.mn.metric_group) in common:
+            s.mn.reset_metric_group()
+        s.mn.tags = [kv for kv in s.mn.tags if kv not in common]
+    return series_list
+
+
+def drop_empty_series(series_list):
+    return [s for s in series_list if not np.all(np.isnan(s.values))]
+
+
+def limit_offset(series_list, limit, offset):
+    # transformLimitOffset (transform.go)
+    return series_list[offset:offset + limit]
+
+
+def sort_by_label(series_list, labels, desc=False):
+    # newTransformFuncSortByLabel: stable multi-key sort by tag values
+    def key(s):
+        return tuple((s.mn.get_tag_value(k) or b"") for k in labels)
+    return sorted(series_list, key=key, reverse=desc)
+
+
+def sort_series(series_list, desc=False):
+    """sort/sort_desc (newTransformFuncSort, transform.go): order by the
+    LAST non-NaN value per series; NaN-only series sort first (asc)."""
+    def key(s):
+        v = s.values
+        for i in range(len(v) - 1, -1, -1):
+            if not math.isnan(v[i]):
+                return v[i]
+        return -math.inf
+    return sorted(series_list, key=key, reverse=desc)
