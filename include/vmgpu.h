@@ -293,6 +293,20 @@ int vmgpu_merge_blocks(const int64_t* ts, const double* vals,
                        uint64_t* out_offsets, uint64_t* out_counts,
                        char* errbuf, size_t errbuf_len);
 
+/* Fused cold-cache fetch: compressed-block payload -> decode -> per-series
+ * merge+dedup -> device-resident rollup batch, decoded columns never
+ * crossing PCIe.  blocks[b].out_off must be the running
+ * row offset (CSR over blocks); series s owns blocks
+ * [series_block_start[s], series_block_start[s+1]).  out_offsets
+ * ([n_series+1], caller-alloc) receives the merged per-series CSR. */
+int vmgpu_batch_create_from_blocks(
+    const uint8_t* payload, uint64_t payload_len,
+    const vmgpu_block_desc* blocks, uint32_t n_blocks, uint64_t total_rows,
+    const uint32_t* series_block_start, uint32_t n_series,
+    int64_t dedup_interval, const int32_t* group_ids, uint32_t n_groups,
+    uint64_t* out_handle, uint64_t* out_offsets,
+    char* errbuf, size_t errbuf_len);
+
 /* Wall time of the rollup kernels inside the last vmgpu_rollup_exec on this
  * thread's context, measured with hipEvents on the launch stream (for the
  * bench's roofline accounting). */

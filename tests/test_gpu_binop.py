@@ -221,7 +221,7 @@ def test_histogram_quantile_pipeline_on_device():
     plan = RollupPlan("rate", start, start + 30 * 15_000, 15_000,
                       window=300_000)
     batch = SeriesBatch(ts, vals, offsets)
-    grid = batch.rollup_eval(plan)
+    grid, _, _ = batch.exec(plan)
     n_grid = grid.shape[1]
     # a / (a + 1) ratio via the binop kernel
     from victoriametrics_amd.binary_op import OP_IDS

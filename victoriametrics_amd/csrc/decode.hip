@@ -718,3 +718,158 @@ int vmgpu_merge_blocks(const int64_t* ts, const double* vals,
 }
 
 }  /* extern "C" */
+
+/* ------------------------------------------------------------------ */
+/* fused decode -> merge -> resident batch (cold-cache query path)    */
+/* ------------------------------------------------------------------ */
+
+/* gather each series' merged prefix [cap_off[s], cap_off[s]+counts[s])
+ * into the dense final CSR at final_off[s] — wave-parallel per series */
+__global__ __launch_bounds__(DBLOCK) void compact_series_kernel(
+    const int64_t* src_ts, const double* src_vals, const uint64_t* cap_off,
+    const uint64_t* final_off, uint32_t n_series,
+    int64_t* dst_ts, double* dst_vals) {
+  const int lane = threadIdx.x % DWAVE;
+  const int wv = threadIdx.x / DWAVE;
+  uint32_t wid = blockIdx.x * (DBLOCK / DWAVE) + wv;
+  uint32_t stride = gridDim.x * (DBLOCK / DWAVE);
+  for (uint32_t s = wid; s < n_series; s += stride) {
+    uint64_t src = cap_off[s];
+    uint64_t dst = final_off[s];
+    uint64_t n = final_off[s + 1] - dst;
+    for (uint64_t k = lane; k < n; k += DWAVE) {
+      dst_ts[dst + k] = src_ts[src + k];
+      dst_vals[dst + k] = src_vals[src + k];
+    }
+  }
+}
+
+/* Internal (cross-TU) seam used by vmgpu_batch_create_from_blocks:
+ * decode all blocks, merge+dedup per series, compact to a dense CSR —
+ * everything device-resident.  On success *out_d_ts / *out_d_vals are
+ * device buffers owned by the caller and h_final_offsets[n_series+1]
+ * holds the CSR offsets. */
+extern "C" int vmdec_decode_merge_device(
+    const uint8_t* payload, uint64_t payload_len,
+    const vmgpu_block_desc* blocks, uint32_t n_blocks, uint64_t total_rows,
+    const uint32_t* series_block_start, uint32_t n_series,
+    int64_t dedup_interval, hipStream_t st,
+    int64_t** out_d_ts, double** out_d_vals, uint64_t* h_final_offsets,
+    char* errbuf, size_t errbuf_len) {
+  if (!payload || !blocks || !series_block_start || !out_d_ts || !out_d_vals ||
+      !h_final_offsets || n_blocks == 0 || n_series == 0)
+    return dset_err(errbuf, errbuf_len, "vmgpu: bad fused decode args");
+  /* host-side offsets for the merge: block b rows at [out_off, out_off+rows) */
+  std::vector<uint64_t> boff(n_blocks + 1);
+  for (uint32_t b = 0; b < n_blocks; b++) boff[b] = blocks[b].out_off;
+  boff[n_blocks] = total_rows;
+  std::vector<uint64_t> cap_off(n_series + 1);
+  cap_off[0] = 0;
+  for (uint32_t s = 0; s < n_series; s++) {
+    uint64_t lo = boff[series_block_start[s]];
+    uint64_t hi = boff[series_block_start[s + 1]];
+    cap_off[s + 1] = cap_off[s] + (hi - lo);
+  }
+
+  uint8_t* d_payload = nullptr;
+  vmgpu_block_desc* d_blocks = nullptr;
+  long long* d_scratch = nullptr;
+  int64_t* d_raw_ts = nullptr;
+  double* d_raw_vals = nullptr;
+  uint64_t* d_boff = nullptr;
+  uint32_t* d_sbs = nullptr;
+  uint64_t* d_capoff = nullptr;
+  int64_t* d_mts = nullptr;
+  double* d_mvals = nullptr;
+  uint64_t* d_cnt = nullptr;
+  uint64_t* d_finaloff = nullptr;
+  int* d_err = nullptr;
+  int64_t* d_fts = nullptr;
+  double* d_fvals = nullptr;
+  int rc = 0;
+  hipError_t kerr = hipSuccess;
+  std::vector<uint64_t> counts(n_series);
+  int err_h = 0;
+  uint32_t grid = 0, mgrid = 0;
+
+#define FDM_TRY(expr, what)                                                  \
+  do {                                                                       \
+    hipError_t _e = (expr);                                                  \
+    if (_e != hipSuccess) { rc = dhip_err(errbuf, errbuf_len, what, _e);     \
+      goto cleanup; }                                                        \
+  } while (0)
+
+  FDM_TRY(hipMalloc(&d_payload, payload_len ? payload_len : 1), "alloc payload");
+  FDM_TRY(hipMalloc(&d_blocks, (size_t)n_blocks * sizeof(vmgpu_block_desc)), "alloc descs");
+  FDM_TRY(hipMalloc(&d_scratch, (size_t)DGRID * DMAX_ROWS * 8), "alloc scratch");
+  FDM_TRY(hipMalloc(&d_raw_ts, (size_t)total_rows * 8), "alloc raw ts");
+  FDM_TRY(hipMalloc(&d_raw_vals, (size_t)total_rows * 8), "alloc raw vals");
+  FDM_TRY(hipMalloc(&d_err, 4), "alloc err");
+  FDM_TRY(hipMemcpyAsync(d_payload, payload, payload_len ? payload_len : 1,
+                         hipMemcpyHostToDevice, st), "ul payload");
+  FDM_TRY(hipMemcpyAsync(d_blocks, blocks, (size_t)n_blocks * sizeof(vmgpu_block_desc),
+                         hipMemcpyHostToDevice, st), "ul descs");
+  FDM_TRY(hipMemsetAsync(d_err, 0, 4, st), "zero err");
+  grid = std::min<uint32_t>(n_blocks, DGRID);
+  hipLaunchKernelGGL(decode_blocks_kernel, dim3(grid), dim3(DBLOCK), 0, st,
+                     d_payload, d_blocks, n_blocks, d_scratch,
+                     d_raw_ts, d_raw_vals, d_err);
+
+  FDM_TRY(hipMalloc(&d_boff, (size_t)(n_blocks + 1) * 8), "alloc boff");
+  FDM_TRY(hipMalloc(&d_sbs, (size_t)(n_series + 1) * 4), "alloc sbs");
+  FDM_TRY(hipMalloc(&d_capoff, (size_t)(n_series + 1) * 8), "alloc capoff");
+  FDM_TRY(hipMalloc(&d_mts, (cap_off[n_series] ? cap_off[n_series] : 1) * 8), "alloc merged ts");
+  FDM_TRY(hipMalloc(&d_mvals, (cap_off[n_series] ? cap_off[n_series] : 1) * 8), "alloc merged vals");
+  FDM_TRY(hipMalloc(&d_cnt, (size_t)n_series * 8), "alloc counts");
+  FDM_TRY(hipMemcpyAsync(d_boff, boff.data(), (size_t)(n_blocks + 1) * 8,
+                         hipMemcpyHostToDevice, st), "ul boff");
+  FDM_TRY(hipMemcpyAsync(d_sbs, series_block_start, (size_t)(n_series + 1) * 4,
+                         hipMemcpyHostToDevice, st), "ul sbs");
+  FDM_TRY(hipMemcpyAsync(d_capoff, cap_off.data(), (size_t)(n_series + 1) * 8,
+                         hipMemcpyHostToDevice, st), "ul capoff");
+  mgrid = std::min<uint32_t>((n_series + 3) / 4, 2048);
+  hipLaunchKernelGGL(merge_blocks_kernel, dim3(mgrid), dim3(DBLOCK), 0, st,
+                     d_raw_ts, d_raw_vals, d_boff, d_sbs, d_capoff, n_series,
+                     dedup_interval, d_mts, d_mvals, d_cnt, d_err);
+  FDM_TRY(hipMemcpyAsync(counts.data(), d_cnt, (size_t)n_series * 8,
+                         hipMemcpyDeviceToHost, st), "dl counts");
+  FDM_TRY(hipMemcpyAsync(&err_h, d_err, 4, hipMemcpyDeviceToHost, st), "dl err");
+  FDM_TRY(hipStreamSynchronize(st), "sync fused");
+  kerr = hipGetLastError();
+  if (kerr != hipSuccess) { rc = dhip_err(errbuf, errbuf_len, "fused kernels", kerr); goto cleanup; }
+  if (err_h != 0) {
+    char msg[64];
+    snprintf(msg, sizeof(msg), "vmgpu: fused decode/merge error %d", err_h);
+    rc = dset_err(errbuf, errbuf_len, msg);
+    goto cleanup;
+  }
+
+  h_final_offsets[0] = 0;
+  for (uint32_t s = 0; s < n_series; s++)
+    h_final_offsets[s + 1] = h_final_offsets[s] + counts[s];
+  FDM_TRY(hipMalloc(&d_finaloff, (size_t)(n_series + 1) * 8), "alloc finaloff");
+  FDM_TRY(hipMemcpyAsync(d_finaloff, h_final_offsets, (size_t)(n_series + 1) * 8,
+                         hipMemcpyHostToDevice, st), "ul finaloff");
+  FDM_TRY(hipMalloc(&d_fts, (h_final_offsets[n_series] ? h_final_offsets[n_series] : 1) * 8), "alloc final ts");
+  FDM_TRY(hipMalloc(&d_fvals, (h_final_offsets[n_series] ? h_final_offsets[n_series] : 1) * 8), "alloc final vals");
+  hipLaunchKernelGGL(compact_series_kernel, dim3(mgrid), dim3(DBLOCK), 0, st,
+                     d_mts, d_mvals, d_capoff, d_finaloff, n_series,
+                     d_fts, d_fvals);
+  FDM_TRY(hipStreamSynchronize(st), "sync compact");
+  kerr = hipGetLastError();
+  if (kerr != hipSuccess) { rc = dhip_err(errbuf, errbuf_len, "compact kernel", kerr); goto cleanup; }
+
+  *out_d_ts = d_fts;
+  *out_d_vals = d_fvals;
+  d_fts = nullptr;
+  d_fvals = nullptr;
+
+cleanup:
+  (void)hipFree(d_payload); (void)hipFree(d_blocks); (void)hipFree(d_scratch);
+  (void)hipFree(d_raw_ts); (void)hipFree(d_raw_vals); (void)hipFree(d_boff);
+  (void)hipFree(d_sbs); (void)hipFree(d_capoff); (void)hipFree(d_mts);
+  (void)hipFree(d_mvals); (void)hipFree(d_cnt); (void)hipFree(d_finaloff);
+  (void)hipFree(d_err); (void)hipFree(d_fts); (void)hipFree(d_fvals);
+#undef FDM_TRY
+  return rc;
+}

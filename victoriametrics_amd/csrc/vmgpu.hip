@@ -1517,6 +1517,112 @@ int vmgpu_batch_create(const int64_t* ts, const double* vals,
   return 0;
 }
 
+/* decode.hip seam: decode blocks, merge+dedup, compact — device-resident */
+int vmdec_decode_merge_device(
+    const uint8_t* payload, uint64_t payload_len,
+    const vmgpu_block_desc* blocks, uint32_t n_blocks, uint64_t total_rows,
+    const uint32_t* series_block_start, uint32_t n_series,
+    int64_t dedup_interval, hipStream_t st,
+    int64_t** out_d_ts, double** out_d_vals, uint64_t* h_final_offsets,
+    char* errbuf, size_t errbuf_len);
+
+/* Cold-cache fetch path fused on device (SURVEY.md §8f(1)+(2) motivation):
+ * compressed-block payload in, resident rollup batch out — the decoded
+ * columns never cross PCIe.  out_offsets[n_series+1] reports the merged
+ * CSR so the host can interpret per-series results. */
+int vmgpu_batch_create_from_blocks(
+    const uint8_t* payload, uint64_t payload_len,
+    const vmgpu_block_desc* blocks, uint32_t n_blocks, uint64_t total_rows,
+    const uint32_t* series_block_start, uint32_t n_series,
+    int64_t dedup_interval, const int32_t* group_ids, uint32_t n_groups,
+    uint64_t* out_handle, uint64_t* out_offsets,
+    char* errbuf, size_t errbuf_len) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  if (!g_ctx.inited) return set_err(errbuf, errbuf_len, "vmgpu: not initialized");
+  if (!out_handle || !out_offsets || n_series == 0)
+    return set_err(errbuf, errbuf_len, "vmgpu: bad args");
+  int64_t* d_ts = nullptr;
+  double* d_vals = nullptr;
+  int rc = vmdec_decode_merge_device(payload, payload_len, blocks, n_blocks,
+                                     total_rows, series_block_start, n_series,
+                                     dedup_interval, g_ctx.stream,
+                                     &d_ts, &d_vals, out_offsets,
+                                     errbuf, errbuf_len);
+  if (rc != 0) return rc;
+
+  Batch b;
+  b.n_series = n_series;
+  b.n_groups = n_groups;
+  b.n_samples = out_offsets[n_series];
+  b.d_ts = d_ts;
+  b.d_vals = d_vals;
+
+  std::vector<uint32_t> wave_list, block_list, huge_list;
+  std::vector<uint64_t> huge_scr_off;
+  uint64_t huge_total = 0;
+  for (uint32_t s = 0; s < n_series; s++) {
+    uint64_t n = out_offsets[s + 1] - out_offsets[s];
+    if (n <= CHUNK_WAVE) {
+      wave_list.push_back(s);
+      if ((uint32_t)n > b.max_wave_len) b.max_wave_len = (uint32_t)n;
+    } else if (n <= CHUNK_BLOCK) {
+      block_list.push_back(s);
+    } else {
+      huge_list.push_back(s);
+      huge_scr_off.push_back(huge_total);
+      huge_total += n;
+    }
+  }
+  b.n_wave = (uint32_t)wave_list.size();
+  b.n_block = (uint32_t)block_list.size();
+  b.n_huge = (uint32_t)huge_list.size();
+  b.wave_is_identity = (b.n_wave == n_series);
+  b.huge_scratch_elems = huge_total;
+
+#define FBB_TRY(expr, what)                                                  \
+  do {                                                                       \
+    hipError_t _e = (expr);                                                  \
+    if (_e != hipSuccess) { free_batch(b);                                   \
+      return hip_err(errbuf, errbuf_len, what, _e); }                        \
+  } while (0)
+
+  FBB_TRY(hipMalloc(&b.d_offsets, (n_series + 1) * sizeof(uint64_t)), "alloc offsets");
+  FBB_TRY(hipMemcpy(b.d_offsets, out_offsets, (n_series + 1) * sizeof(uint64_t),
+                    hipMemcpyHostToDevice), "upload offsets");
+  if (group_ids) {
+    FBB_TRY(hipMalloc(&b.d_group_ids, n_series * sizeof(int32_t)), "alloc gids");
+    FBB_TRY(hipMemcpy(b.d_group_ids, group_ids, n_series * sizeof(int32_t),
+                      hipMemcpyHostToDevice), "upload gids");
+  }
+  if (!b.wave_is_identity && b.n_wave) {
+    FBB_TRY(hipMalloc(&b.d_wave_list, b.n_wave * 4), "alloc wave list");
+    FBB_TRY(hipMemcpy(b.d_wave_list, wave_list.data(), b.n_wave * 4,
+                      hipMemcpyHostToDevice), "upload wave list");
+  }
+  if (b.n_block) {
+    FBB_TRY(hipMalloc(&b.d_block_list, b.n_block * 4), "alloc block list");
+    FBB_TRY(hipMemcpy(b.d_block_list, block_list.data(), b.n_block * 4,
+                      hipMemcpyHostToDevice), "upload block list");
+  }
+  if (b.n_huge) {
+    FBB_TRY(hipMalloc(&b.d_huge_list, b.n_huge * 4), "alloc huge list");
+    FBB_TRY(hipMemcpy(b.d_huge_list, huge_list.data(), b.n_huge * 4,
+                      hipMemcpyHostToDevice), "upload huge list");
+    FBB_TRY(hipMalloc(&b.d_huge_scr_offsets, b.n_huge * 8), "alloc huge offsets");
+    FBB_TRY(hipMemcpy(b.d_huge_scr_offsets, huge_scr_off.data(), b.n_huge * 8,
+                      hipMemcpyHostToDevice), "upload huge offsets");
+    FBB_TRY(hipMalloc(&b.d_scr_ts, huge_total * sizeof(int64_t)), "alloc scratch ts");
+    FBB_TRY(hipMalloc(&b.d_scr_vals, huge_total * sizeof(double)), "alloc scratch vals");
+  }
+  FBB_TRY(hipMalloc(&b.d_scanned, sizeof(unsigned long long)), "alloc scanned");
+#undef FBB_TRY
+
+  uint64_t h = g_ctx.next_handle++;
+  g_ctx.batches[h] = b;
+  *out_handle = h;
+  return 0;
+}
+
 int vmgpu_batch_destroy(uint64_t handle) {
   std::lock_guard<std::mutex> lock(g_ctx.mu);
   auto it = g_ctx.batches.find(handle);

@@ -245,6 +245,78 @@ class SeriesBatch:
                              f"{errbuf.value.decode()}")
         self.handle = handle.value
 
+    @classmethod
+    def from_blocks(cls, blocks, series_block_start, dedup_interval=0,
+                    group_ids=None, n_groups=0):
+        """Fused cold-cache fetch (§8f(1)): compressed-block payload ->
+        device decode -> per-series merge+dedup -> resident batch.  The
+        decoded columns never cross PCIe.
+
+        blocks: list of dicts as for decode_blocks(); series i owns blocks
+        [series_block_start[i], series_block_start[i+1]).
+        """
+        import math as _math
+        init()
+        lib = _load_lib()
+        payload = bytearray()
+        descs = (_BlockDescC * len(blocks))()
+        total_rows = 0
+        for i, b in enumerate(blocks):
+            d = descs[i]
+            d.ts_data_off = len(payload)
+            payload.extend(b["ts_data"])
+            d.ts_data_len = len(b["ts_data"])
+            d.val_data_off = len(payload)
+            payload.extend(b["val_data"])
+            d.val_data_len = len(b["val_data"])
+            d.out_off = total_rows
+            d.min_timestamp = int(b["min_timestamp"])
+            d.max_timestamp = int(b["max_timestamp"])
+            d.first_value = int(b["first_value"])
+            d.scale = int(b["scale"])
+            d.e10 = _math.pow(10.0, abs(int(b["scale"])))
+            d.rows = int(b["rows"])
+            d.ts_mt = int(b["ts_mt"])
+            d.val_mt = int(b["val_mt"])
+            d.precision_bits = int(b["precision_bits"])
+            total_rows += int(b["rows"])
+        pl = np.frombuffer(bytes(payload), dtype=np.uint8) if payload else \
+            np.zeros(1, dtype=np.uint8)
+        sbs = np.ascontiguousarray(series_block_start, dtype=np.uint32)
+        n_series = len(sbs) - 1
+        out_offsets = np.zeros(n_series + 1, dtype=np.uint64)
+        if group_ids is not None:
+            gids = np.ascontiguousarray(group_ids, dtype=np.int32)
+            gptr = gids.ctypes.data_as(ctypes.POINTER(ctypes.c_int32))
+        else:
+            gids, gptr = None, None
+        handle = ctypes.c_uint64(0)
+        errbuf = ctypes.create_string_buffer(256)
+        rc = lib.vmgpu_batch_create_from_blocks(
+            pl.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+            ctypes.c_uint64(len(payload)), descs,
+            ctypes.c_uint32(len(blocks)), ctypes.c_uint64(total_rows),
+            sbs.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+            ctypes.c_uint32(n_series), ctypes.c_int64(int(dedup_interval)),
+            gptr, ctypes.c_uint32(int(n_groups)),
+            ctypes.byref(handle),
+            out_offsets.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            errbuf, ctypes.c_size_t(256))
+        if rc != 0:
+            raise VmGpuError(f"vmgpu_batch_create_from_blocks failed ({rc}):"
+                             f" {errbuf.value.decode()}")
+        self = cls.__new__(cls)
+        self.ts = None
+        self.vals = None
+        self.offsets = out_offsets
+        self.n_series = n_series
+        self.n_groups = int(n_groups)
+        self.group_ids = gids
+        self.handle = handle.value
+        self._last_rows = 0
+        self._last_n_grid = 0
+        return self
+
     def close(self):
         if self.handle:
             _load_lib().vmgpu_batch_destroy(ctypes.c_uint64(self.handle))
