@@ -168,11 +168,23 @@ struct KPlan {
   int32_t rcr;
   int32_t drop_stale;
   int32_t chunk_wave;      /* LDS samples per wave (wave kernel), 64-aligned */
-  int32_t jbuf_elems;      /* LDS window-end index cache [n_grid] u16 for the
-                              shared-boundary rate path (0 = disabled) */
+  int32_t jbuf_elems;      /* LDS boundary-cache elements for the rate paths */
+  int32_t jbuf_mode;       /* 0 = none; 1 = u16 j-cache (shared-boundary);
+                              2 = sample-scatter Et/Ev/J boundary map */
   double arg;
   double arg2;
 };
+
+/* LDS bytes for the rate boundary cache (host sizing and kernel carve must
+ * agree) */
+static __host__ __device__ inline size_t vm_jbuf_bytes(int32_t mode,
+                                                       int32_t elems) {
+  size_t a8 = ((size_t)elems * 8 + 15) & ~(size_t)15;
+  size_t a2 = ((size_t)elems * 2 + 15) & ~(size_t)15;
+  if (mode == 2) return 2 * a8 + a2;   /* Et i64 + Ev f64 + J u16 */
+  if (mode == 1) return a2;            /* J u16 */
+  return 0;
+}
 
 struct KIO {
   const int64_t* ts = nullptr;
@@ -592,15 +604,18 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int wave_in_block = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  const size_t jbuf_bytes = ((size_t)p.jbuf_elems * 2 + 15) & ~(size_t)15;
+  const size_t jbuf_bytes = vm_jbuf_bytes(p.jbuf_mode, p.jbuf_elems);
   const size_t wave_bytes = (size_t)p.chunk_wave * 16 + 256 + jbuf_bytes;
   int64_t* lts = (int64_t*)(smem + (size_t)wave_in_block * wave_bytes);
   double* lvs = (double*)(smem + (size_t)wave_in_block * wave_bytes +
                           (size_t)p.chunk_wave * 8);
   double* lscratch = (double*)(smem + (size_t)wave_in_block * wave_bytes +
                                (size_t)p.chunk_wave * 16);
-  uint16_t* jbuf = (uint16_t*)(smem + (size_t)wave_in_block * wave_bytes +
-                               (size_t)p.chunk_wave * 16 + 256);
+  char* jbuf_region = smem + (size_t)wave_in_block * wave_bytes +
+                      (size_t)p.chunk_wave * 16 + 256;
+  /* mode-1 layout (u16 j-cache); in mode 2 the same base is re-carved by
+   * the scatter path below */
+  uint16_t* jbuf = (uint16_t*)jbuf_region;
   uint64_t scanned = 0;
   const uint32_t wave_id = blockIdx.x * WAVES_PER_BLOCK + wave_in_block;
   const uint32_t wave_stride = gridDim.x * WAVES_PER_BLOCK;
@@ -659,13 +674,92 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
     }
 #if !defined(VMGPU_ABL_NO_SEEK) && !defined(VMGPU_ABL_NO_EVAL)
     if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
-      /* shared-boundary path: when window is a step multiple (the standard
-       * rate(m[5m]) @ 15s grid), the window START of point g is the window
-       * END of point g-dg, so i(g) = j(g-dg) — one seek per point.  Phase 1
-       * caches every j in LDS; phase 2 evaluates with two u16 reads. */
+      /* Both fast paths exploit i(g) = j(g-dg) when the window is a step
+       * multiple (the standard rate(m[5m]) @ 15s grid): the window START
+       * boundary of point g is the window END boundary of point g-dg. */
       int dg64 = (sw.window > 0 && p.step > 0 && sw.window % p.step == 0)
                      ? (int)(sw.window / p.step) : 0;
-      if (p.jbuf_elems >= p.n_grid && dg64 > 0 && count <= 65535) {
+      if (p.jbuf_mode == 2 && dg64 > 0 && count <= 65534 &&
+          p.n_grid + dg64 <= p.jbuf_elems) {
+        /* sample-scatter boundary map: each sample computes the grid range
+         * where it is the last sample <= t_end (one boundary per sample —
+         * t_hi(k) = t_lo(k+1)) and scatters (ts, corrected value, index+1)
+         * into Et/Ev/J over the extended range [-dg, n_grid).  Eval is then
+         * pure elementwise over coalesced LDS rows: no per-point seek, no
+         * gather chains.  Exactly ub() semantics: J[g] = #samples <= t_end(g). */
+        size_t a8 = (((size_t)p.jbuf_elems * 8 + 15) & ~(size_t)15);
+        int64_t* Et = (int64_t*)jbuf_region;
+        double* Ev = (double*)(jbuf_region + a8);
+        uint16_t* Jb = (uint16_t*)(jbuf_region + 2 * a8);
+        const int ext = p.n_grid + dg64;
+        for (int e = lane; e < ext; e += WAVE) Jb[e] = 0;
+        wave_lds_sync();
+        const double inv_step = 1.0 / (double)p.step;
+        for (int base = 0; base < count; base += WAVE) {
+          int k = base + lane;
+          bool active = k < count;
+          int64_t t_k = active ? lts[k] : 0;
+          int64_t t_n = (active && k + 1 < count) ? lts[k + 1] : 0;
+          /* g_lo = ceil((t_k - start)/step): float estimate, exact int fixup */
+          int g_lo = (int)floor((double)(t_k - p.start) * inv_step) - 1;
+          g_lo += (p.start + (int64_t)g_lo * p.step < t_k);
+          g_lo += (p.start + (int64_t)g_lo * p.step < t_k);
+          g_lo += (p.start + (int64_t)g_lo * p.step < t_k);
+          int g_hi;
+          if (k + 1 < count) {
+            g_hi = (int)floor((double)(t_n - p.start) * inv_step) - 1;
+            g_hi += (p.start + (int64_t)g_hi * p.step < t_n);
+            g_hi += (p.start + (int64_t)g_hi * p.step < t_n);
+            g_hi += (p.start + (int64_t)g_hi * p.step < t_n);
+          } else {
+            g_hi = p.n_grid;
+          }
+          if (g_lo < -dg64) g_lo = -dg64;
+          if (g_hi > p.n_grid) g_hi = p.n_grid;
+          if (active) {
+            double v_k = lvs[k];
+            for (int g = g_lo; g < g_hi; g++) {
+              int idx = g + dg64;
+              Et[idx] = t_k;
+              Ev[idx] = v_k;
+              Jb[idx] = (uint16_t)(k + 1);
+            }
+          }
+        }
+        wave_lds_sync();
+        for (int g0 = 0; g0 < p.n_grid; g0 += 4 * WAVE) {
+#pragma unroll
+          for (int u = 0; u < 4; u++) {
+            int g = g0 + u * WAVE + lane;
+            if (g < p.n_grid) {
+              int64_t t_end = p.start + (int64_t)g * p.step;
+              int64_t t_start = t_end - sw.window;
+              int j = Jb[g + dg64];
+              int i = Jb[g];                     /* = J[(g-dg)+dg] */
+              double v_end = Ev[g + dg64];       /* lvs[j-1] (valid iff j>0) */
+              int64_t t_end_s = Et[g + dg64];
+              double v_prevc = Ev[g];            /* lvs[i-1] (valid iff i>0) */
+              int64_t t_prevc = Et[g];
+              bool has_prev = (i > 0) && (i < count) &&
+                              (t_prevc > t_start - sw.max_prev_interval);
+              int ii = i < count - 1 ? i : (count - 1 < 0 ? 0 : count - 1);
+              double pv = has_prev ? v_prevc : lvs[ii];
+              int64_t ptm = has_prev ? t_prevc : lts[ii];
+              double slope = (v_end - pv) / ((double)(t_end_s - ptm) / 1e3);
+              int n = j - i;
+              double res_prev = (n == 0) ? 0.0 : slope;
+              double res_nop = (n <= 1) ? vm_dnan() : slope;
+              vm_emit_value(p, io, s, g, has_prev ? res_prev : res_nop);
+              scanned += 2;
+            }
+          }
+        }
+        wave_lds_sync();
+        continue;
+      }
+      if (p.jbuf_mode >= 1 &&
+          (size_t)p.n_grid * 2 <= vm_jbuf_bytes(p.jbuf_mode, p.jbuf_elems) &&
+          dg64 > 0 && count <= 65535) {
         for (int g = lane; g < p.n_grid; g += WAVE) {
           int64_t t_end = p.start + (int64_t)g * p.step;
           int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
@@ -1686,13 +1780,29 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
   p.drop_stale = plan->drop_stale_nans;
   p.chunk_wave = (int32_t)std::min<uint32_t>(
       CHUNK_WAVE, std::max<uint32_t>(64, (b.max_wave_len + 63) & ~63u));
-  /* shared-boundary seek (i(g) = j(g - window/step) when window is a step
-   * multiple): give the wave kernel an LDS u16 j-cache when the fused rate
-   * path can use it.  Bounded so LDS stays within occupancy budget. */
+  /* rate boundary caches (see the wave kernel): prefer the sample-scatter
+   * Et/Ev/J map when the plan window is a known step multiple; fall back to
+   * the u16 j-cache.  Bounded by the 64 KiB dynamic-LDS launch limit. */
   p.jbuf_elems = 0;
-  if ((plan->func == VMF_RATE || plan->func == VMF_DERIV_FAST) &&
-      n_grid > 1 && n_grid <= 4096)
-    p.jbuf_elems = n_grid;
+  p.jbuf_mode = 0;
+  if ((plan->func == VMF_RATE || plan->func == VMF_DERIV_FAST) && n_grid > 1) {
+    size_t base = (size_t)WAVES_PER_BLOCK * ((size_t)p.chunk_wave * 16 + 256);
+    int64_t dgp = (plan->window > 0 && plan->step > 0 &&
+                   plan->window % plan->step == 0)
+                      ? plan->window / plan->step : 0;
+    if (dgp > 0 && n_grid + dgp <= 32000) {
+      int32_t elems = (int32_t)(n_grid + dgp);
+      if (base + WAVES_PER_BLOCK * vm_jbuf_bytes(2, elems) <= 64 * 1024) {
+        p.jbuf_mode = 2;
+        p.jbuf_elems = elems;
+      }
+    }
+    if (p.jbuf_mode == 0 && n_grid <= 4096 &&
+        base + WAVES_PER_BLOCK * vm_jbuf_bytes(1, n_grid) <= 64 * 1024) {
+      p.jbuf_mode = 1;
+      p.jbuf_elems = n_grid;
+    }
+  }
   p.arg = plan->arg;
   p.arg2 = plan->arg2;
 
@@ -1726,7 +1836,7 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
                                          MAX_WAVE_BLOCKS);
     size_t lds = (size_t)WAVES_PER_BLOCK *
                  ((size_t)p.chunk_wave * 16 + 256 +
-                  (((size_t)p.jbuf_elems * 2 + 15) & ~(size_t)15));
+                  vm_jbuf_bytes(p.jbuf_mode, p.jbuf_elems));
     launch_rollup(0, blocks, lds, p, w);
   }
   if (b.n_block) {
