@@ -132,3 +132,23 @@ def test_fused_batch_grouped_rollup():
     staged.close()
     np.testing.assert_allclose(got, exp, rtol=1e-9)
     np.testing.assert_array_equal(counts, exp_counts)
+
+
+def test_marshal_from_batch_matches_host_marshal():
+    """§8f(2): the cache fill downloaded straight off the device must be
+    byte-identical to the host-side marshalTimeseriesFast."""
+    from victoriametrics_amd import cache, synth
+    rng = np.random.default_rng(41)
+    n_series = 96
+    ts, vals, offsets = synth.counter_batch(n_series, 100, START)
+    start = START + 120_000
+    plan = RollupPlan("rate", start, start + 30 * 15_000, 15_000,
+                      window=300_000)
+    batch = SeriesBatch(ts, vals, offsets)
+    out, _, _ = batch.exec(plan)
+    names = [(b"m", ((b"pod", b"p%d" % i),)) for i in range(n_series)]
+    grid_ts = plan.timestamps()
+    expected = cache.marshal_timeseries_fast(names, out, grid_ts)
+    got = cache.marshal_from_batch(batch, names, grid_ts)
+    batch.close()
+    assert got == expected

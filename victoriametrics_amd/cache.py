@@ -88,6 +88,40 @@ def unmarshal_timeseries_fast(data):
     return names, values, timestamps
 
 
+def marshal_from_batch(batch, names, timestamps, rows=None, max_size=1 << 62):
+    """rollupResultCache binary fill straight off the device:
+    the values section of marshalTimeseriesFast IS the engine's row-major
+    [rows x n_points] f64 output, so the device buffer is downloaded
+    directly into the marshal buffer at its final offset — no host repack
+    pass.  batch: an engine.SeriesBatch whose last exec produced `rows`
+    series rows on the same grid as `timestamps`."""
+    timestamps = np.ascontiguousarray(timestamps, dtype=np.int64)
+    n = len(names)
+    npts = len(timestamps)
+    if n == 0:
+        return struct.pack(">QQ", 0, 0)
+    if rows is None:
+        rows = n
+    assert rows == n
+    head = struct.pack(">QQ", n, npts) + timestamps.tobytes()
+    out = bytearray(len(head) + 8 * n * npts)
+    out[:len(head)] = head
+    dst = np.frombuffer(out, dtype=np.float64, count=n * npts,
+                        offset=len(head)).reshape(n, npts)
+    got, _ = batch.fetch_out_into(dst, npts)
+    tail = bytearray()
+    for group, tags in names:
+        tail += struct.pack(">H", len(group)) + bytes(group)
+        tail += struct.pack(">H", len(tags))
+        for k, v in tags:
+            tail += struct.pack(">H", len(k)) + bytes(k)
+            tail += struct.pack(">H", len(v)) + bytes(v)
+    out += tail
+    if len(out) > max_size:
+        return b""
+    return bytes(out)
+
+
 def merge_series(a_names, a_values, b_names, b_values, b_start, start, end, step):
     """mergeSeries (rollup_result_cache.go:618-720).  a covers
     [start, b_start), b covers [b_start, end].  Returns (names, values) on

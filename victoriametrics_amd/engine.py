@@ -359,6 +359,21 @@ class SeriesBatch:
         self._last_n_grid = n_grid
         return out, counts, scanned.value
 
+    def fetch_out_into(self, dst, n_grid):
+        """Download the last exec's output rows directly into a caller
+        buffer (e.g. a marshal buffer's values section) — no intermediate
+        copy.  dst: writable f64 array view [rows x n_grid]."""
+        lib = _load_lib()
+        assert dst.dtype == np.float64
+        rows = dst.shape[0]
+        rc = lib.vmgpu_batch_fetch_out(
+            ctypes.c_uint64(self.handle),
+            dst.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+            ctypes.c_size_t(dst.size), None, ctypes.c_size_t(0))
+        if rc != 0:
+            raise VmGpuError(f"vmgpu_batch_fetch_out failed ({rc})")
+        return dst, None
+
     def fetch_out(self, rows, n_grid, with_counts=False):
         lib = _load_lib()
         out = np.empty((rows, n_grid), dtype=np.float64)
