@@ -179,10 +179,9 @@ struct KPlan {
  * agree) */
 static __host__ __device__ inline size_t vm_jbuf_bytes(int32_t mode,
                                                        int32_t elems) {
-  size_t a8 = ((size_t)elems * 8 + 15) & ~(size_t)15;
   size_t a2 = ((size_t)elems * 2 + 15) & ~(size_t)15;
-  if (mode == 2) return 2 * a8 + a2;   /* Et i64 + Ev f64 + J u16 */
-  if (mode == 1) return a2;            /* J u16 */
+  if (mode == 2) return a2;            /* J u16 over [-dg, n_grid) */
+  if (mode == 1) return a2;            /* J u16 over [0, n_grid) */
   return 0;
 }
 
@@ -681,16 +680,15 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
                      ? (int)(sw.window / p.step) : 0;
       if (p.jbuf_mode == 2 && dg64 > 0 && count <= 65534 &&
           p.n_grid + dg64 <= p.jbuf_elems) {
-        /* sample-scatter boundary map: each sample computes the grid range
-         * where it is the last sample <= t_end (one boundary per sample —
-         * t_hi(k) = t_lo(k+1)) and scatters (ts, corrected value, index+1)
-         * into Et/Ev/J over the extended range [-dg, n_grid).  Eval is then
-         * pure elementwise over coalesced LDS rows: no per-point seek, no
-         * gather chains.  Exactly ub() semantics: J[g] = #samples <= t_end(g). */
-        size_t a8 = (((size_t)p.jbuf_elems * 8 + 15) & ~(size_t)15);
-        int64_t* Et = (int64_t*)jbuf_region;
-        double* Ev = (double*)(jbuf_region + a8);
-        uint16_t* Jb = (uint16_t*)(jbuf_region + 2 * a8);
+        /* sample-scatter boundary map (J-only): each sample computes the
+         * grid range where it is the last sample <= t_end (one boundary
+         * per sample — t_hi(k) = t_lo(k+1)) and scatters index+1 into J
+         * over the extended range [-dg, n_grid), replacing BOTH per-point
+         * seeks; eval gathers values by index as before.  Exactly ub()
+         * semantics: J[g] = #samples <= t_end(g).  (A fatter variant also
+         * caching (ts, value) rows in LDS measured SLOWER — the extra
+         * 4.6 KB/wave of LDS cost 3 waves/SIMD of occupancy.) */
+        uint16_t* Jb = (uint16_t*)jbuf_region;
         const int ext = p.n_grid + dg64;
         for (int e = lane; e < ext; e += WAVE) Jb[e] = 0;
         wave_lds_sync();
@@ -717,13 +715,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
           if (g_lo < -dg64) g_lo = -dg64;
           if (g_hi > p.n_grid) g_hi = p.n_grid;
           if (active) {
-            double v_k = lvs[k];
-            for (int g = g_lo; g < g_hi; g++) {
-              int idx = g + dg64;
-              Et[idx] = t_k;
-              Ev[idx] = v_k;
-              Jb[idx] = (uint16_t)(k + 1);
-            }
+            for (int g = g_lo; g < g_hi; g++) Jb[g + dg64] = (uint16_t)(k + 1);
           }
         }
         wave_lds_sync();
@@ -736,20 +728,8 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
               int64_t t_start = t_end - sw.window;
               int j = Jb[g + dg64];
               int i = Jb[g];                     /* = J[(g-dg)+dg] */
-              double v_end = Ev[g + dg64];       /* lvs[j-1] (valid iff j>0) */
-              int64_t t_end_s = Et[g + dg64];
-              double v_prevc = Ev[g];            /* lvs[i-1] (valid iff i>0) */
-              int64_t t_prevc = Et[g];
-              bool has_prev = (i > 0) && (i < count) &&
-                              (t_prevc > t_start - sw.max_prev_interval);
-              int ii = i < count - 1 ? i : (count - 1 < 0 ? 0 : count - 1);
-              double pv = has_prev ? v_prevc : lvs[ii];
-              int64_t ptm = has_prev ? t_prevc : lts[ii];
-              double slope = (v_end - pv) / ((double)(t_end_s - ptm) / 1e3);
-              int n = j - i;
-              double res_prev = (n == 0) ? 0.0 : slope;
-              double res_nop = (n <= 1) ? vm_dnan() : slope;
-              vm_emit_value(p, io, s, g, has_prev ? res_prev : res_nop);
+              vm_emit_value(p, io, s, g,
+                            eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
               scanned += 2;
             }
           }
@@ -1289,6 +1269,122 @@ __global__ void topk_col_fill_kernel(double* values, uint32_t n_series,
 
 /* histogram_quantile (transform.go:1028-1074 + fixBrokenBuckets 1140):
  * one thread per (group, grid point) */
+/* histogram_avg / histogram_stddev / histogram_stdvar
+ * (transform.go:transformHistogramAvg/Stddev/Stdvar + avgForLeTimeseries /
+ * stdvarForLeTimeseries): (le+lePrev)/2-weighted moments over the bucket
+ * column, +/-Inf les skipped, NO fixBrokenBuckets (the reference reads raw
+ * cumulative values here). mode: 0 avg, 1 stddev, 2 stdvar. */
+__global__ void hstat_kernel(int32_t mode, const double* bv, const double* les,
+                             const uint64_t* goff, int64_t n_groups,
+                             int32_t n_grid, double* out) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t total = (size_t)n_groups * n_grid;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    int64_t grp = (int64_t)(i / n_grid);
+    int32_t g = (int32_t)(i % n_grid);
+    int64_t lo = (int64_t)goff[grp];
+    int64_t n_les = (int64_t)(goff[grp + 1] - lo);
+    double le_prev = 0, v_prev = 0, sum = 0, sum2 = 0, wt = 0;
+    for (int64_t j = 0; j < n_les; j++) {
+      double le = les[lo + j];
+      if (isinf(le)) continue;
+      double v = bv[(size_t)(lo + j) * n_grid + g];
+      double n = (le + le_prev) / 2;
+      double w = v - v_prev;
+      sum += n * w;
+      sum2 += n * n * w;
+      wt += w;
+      le_prev = le;
+      v_prev = v;
+    }
+    double r;
+    if (wt == 0) {
+      r = vm_dnan();
+    } else if (mode == 0) {
+      r = sum / wt;
+    } else {
+      double avg = sum / wt;
+      double sv = sum2 / wt - avg * avg;
+      if (sv < 0) sv = 0;
+      r = (mode == 1) ? sqrt(sv) : sv;
+    }
+    out[i] = r;
+  }
+}
+
+/* histogram_share (transform.go:transformHistogramShare): the quantile
+ * walk's dual — share of samples <= leReq[g], with lower/upper bounds.
+ * fixBrokenBuckets applied on the fly as in hq_kernel. */
+__global__ void hshare_kernel(const double* le_req, const double* bv,
+                              const double* les, const uint64_t* goff,
+                              int64_t n_groups, int32_t n_grid, double* out,
+                              double* out_lo, double* out_hi) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t total = (size_t)n_groups * n_grid;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    int64_t grp = (int64_t)(i / n_grid);
+    int32_t g = (int32_t)(i % n_grid);
+    int64_t lo = (int64_t)goff[grp];
+    int64_t n_les = (int64_t)(goff[grp + 1] - lo);
+    double req = le_req[g];
+    double q = vm_dnan(), lb = vm_dnan(), ub = vm_dnan();
+    if (!vm_isnan(req) && n_les > 0) {
+      if (req < 0) {
+        q = 0; lb = 0; ub = 0;
+      } else if (isinf(req) && req > 0) {
+        q = 1; lb = 1; ub = 1;
+      } else {
+        /* fixBrokenBuckets pass: running-max with NaN->prev */
+        double v_last = 0;
+        {
+          double fix_prev = 0;
+          for (int64_t j = 0; j < n_les; j++) {
+            double v = bv[(size_t)(lo + j) * n_grid + g];
+            if (j == 0) fix_prev = vm_isnan(v) ? 0 : v;
+            else if (!(vm_isnan(v) || fix_prev > v)) fix_prev = v;
+          }
+          v_last = fix_prev;
+        }
+        double vp = 0, lep = 0;
+        double fix_prev = 0;
+        bool done = false;
+        for (int64_t j = 0; j < n_les && !done; j++) {
+          double raw = bv[(size_t)(lo + j) * n_grid + g];
+          double v;
+          if (j == 0) v = vm_isnan(raw) ? 0 : raw;
+          else v = (vm_isnan(raw) || fix_prev > raw) ? fix_prev : raw;
+          fix_prev = v;
+          double le = les[lo + j];
+          if (req >= le) {
+            vp = v;
+            lep = le;
+            continue;
+          }
+          /* lePrev <= req < le */
+          lb = vp / v_last;
+          if (isinf(le) && le > 0) {
+            q = lb;
+            ub = 1;
+          } else if (lep == req) {
+            q = lb;
+            ub = lb;
+          } else {
+            ub = v / v_last;
+            q = lb + (v - vp) / v_last * (req - lep) / (le - lep);
+          }
+          done = true;
+        }
+        if (!done) { q = 1; lb = 1; ub = 1; }
+      }
+    }
+    out[i] = q;
+    if (out_lo) out_lo[i] = lb;
+    if (out_hi) out_hi[i] = ub;
+  }
+}
+
 __global__ void hq_kernel(double phi, const double* bv, const double* les,
                           const uint64_t* goff, int64_t n_groups, int32_t n_grid,
                           double* out, double* out_lo, double* out_hi) {
@@ -2141,6 +2237,92 @@ int vmgpu_topk_pointwise(uint64_t handle, double k, int32_t reverse,
   (void)hipFree(d_hists); (void)hipFree(d_bins); (void)hipFree(d_cand);
   (void)hipFree(d_kstar); (void)hipFree(d_small);
   if (kerr != hipSuccess) return hip_err(errbuf, errbuf_len, "topk kernel", kerr);
+  return 0;
+}
+
+int vmgpu_histogram_stat(int32_t mode, const double* bucket_values,
+                         const double* les, const uint64_t* group_offsets,
+                         uint32_t n_groups, int32_t n_grid, double* out,
+                         char* errbuf, size_t errbuf_len) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  if (!g_ctx.inited) return set_err(errbuf, errbuf_len, "vmgpu: not initialized");
+  if (!bucket_values || !les || !group_offsets || !out || n_groups == 0 || n_grid <= 0)
+    return set_err(errbuf, errbuf_len, "vmgpu: bad args");
+  hipStream_t st = g_ctx.stream;
+  uint64_t n_rows = group_offsets[n_groups];
+  double* d_bv = nullptr;
+  double* d_les = nullptr;
+  uint64_t* d_off = nullptr;
+  double* d_out = nullptr;
+  size_t out_elems = (size_t)n_groups * n_grid;
+  HIP_TRY(hipMalloc(&d_bv, (size_t)n_rows * n_grid * 8), "alloc bv");
+  HIP_TRY(hipMalloc(&d_les, (size_t)n_rows * 8), "alloc les");
+  HIP_TRY(hipMalloc(&d_off, (size_t)(n_groups + 1) * 8), "alloc off");
+  HIP_TRY(hipMalloc(&d_out, out_elems * 8), "alloc out");
+  HIP_TRY(hipMemcpyAsync(d_bv, bucket_values, (size_t)n_rows * n_grid * 8,
+                         hipMemcpyHostToDevice, st), "ul bv");
+  HIP_TRY(hipMemcpyAsync(d_les, les, (size_t)n_rows * 8, hipMemcpyHostToDevice, st), "ul les");
+  HIP_TRY(hipMemcpyAsync(d_off, group_offsets, (size_t)(n_groups + 1) * 8,
+                         hipMemcpyHostToDevice, st), "ul off");
+  uint32_t blocks = (uint32_t)std::min<size_t>((out_elems + 255) / 256, 4096);
+  hipLaunchKernelGGL(hstat_kernel, dim3(blocks), dim3(256), 0, st, mode, d_bv,
+                     d_les, d_off, (int64_t)n_groups, n_grid, d_out);
+  HIP_TRY(hipMemcpyAsync(out, d_out, out_elems * 8, hipMemcpyDeviceToHost, st), "dl out");
+  HIP_TRY(hipStreamSynchronize(st), "sync hstat");
+  hipError_t kerr = hipGetLastError();
+  (void)hipFree(d_bv); (void)hipFree(d_les); (void)hipFree(d_off);
+  (void)hipFree(d_out);
+  if (kerr != hipSuccess) return hip_err(errbuf, errbuf_len, "hstat kernel", kerr);
+  return 0;
+}
+
+int vmgpu_histogram_share(const double* le_req, const double* bucket_values,
+                          const double* les, const uint64_t* group_offsets,
+                          uint32_t n_groups, int32_t n_grid, double* out,
+                          double* out_lower, double* out_upper,
+                          char* errbuf, size_t errbuf_len) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  if (!g_ctx.inited) return set_err(errbuf, errbuf_len, "vmgpu: not initialized");
+  if (!le_req || !bucket_values || !les || !group_offsets || !out ||
+      n_groups == 0 || n_grid <= 0)
+    return set_err(errbuf, errbuf_len, "vmgpu: bad args");
+  hipStream_t st = g_ctx.stream;
+  uint64_t n_rows = group_offsets[n_groups];
+  double* d_req = nullptr;
+  double* d_bv = nullptr;
+  double* d_les = nullptr;
+  uint64_t* d_off = nullptr;
+  double* d_out = nullptr;
+  double* d_lo = nullptr;
+  double* d_hi = nullptr;
+  size_t out_elems = (size_t)n_groups * n_grid;
+  HIP_TRY(hipMalloc(&d_req, (size_t)n_grid * 8), "alloc req");
+  HIP_TRY(hipMalloc(&d_bv, (size_t)n_rows * n_grid * 8), "alloc bv");
+  HIP_TRY(hipMalloc(&d_les, (size_t)n_rows * 8), "alloc les");
+  HIP_TRY(hipMalloc(&d_off, (size_t)(n_groups + 1) * 8), "alloc off");
+  HIP_TRY(hipMalloc(&d_out, out_elems * 8), "alloc out");
+  if (out_lower) HIP_TRY(hipMalloc(&d_lo, out_elems * 8), "alloc lo");
+  if (out_upper) HIP_TRY(hipMalloc(&d_hi, out_elems * 8), "alloc hi");
+  HIP_TRY(hipMemcpyAsync(d_req, le_req, (size_t)n_grid * 8,
+                         hipMemcpyHostToDevice, st), "ul req");
+  HIP_TRY(hipMemcpyAsync(d_bv, bucket_values, (size_t)n_rows * n_grid * 8,
+                         hipMemcpyHostToDevice, st), "ul bv");
+  HIP_TRY(hipMemcpyAsync(d_les, les, (size_t)n_rows * 8, hipMemcpyHostToDevice, st), "ul les");
+  HIP_TRY(hipMemcpyAsync(d_off, group_offsets, (size_t)(n_groups + 1) * 8,
+                         hipMemcpyHostToDevice, st), "ul off");
+  uint32_t blocks = (uint32_t)std::min<size_t>((out_elems + 255) / 256, 4096);
+  hipLaunchKernelGGL(hshare_kernel, dim3(blocks), dim3(256), 0, st, d_req,
+                     d_bv, d_les, d_off, (int64_t)n_groups, n_grid,
+                     d_out, d_lo, d_hi);
+  HIP_TRY(hipMemcpyAsync(out, d_out, out_elems * 8, hipMemcpyDeviceToHost, st), "dl out");
+  if (out_lower) HIP_TRY(hipMemcpyAsync(out_lower, d_lo, out_elems * 8, hipMemcpyDeviceToHost, st), "dl lo");
+  if (out_upper) HIP_TRY(hipMemcpyAsync(out_upper, d_hi, out_elems * 8, hipMemcpyDeviceToHost, st), "dl hi");
+  HIP_TRY(hipStreamSynchronize(st), "sync hshare");
+  hipError_t kerr = hipGetLastError();
+  (void)hipFree(d_req); (void)hipFree(d_bv); (void)hipFree(d_les);
+  (void)hipFree(d_off); (void)hipFree(d_out); (void)hipFree(d_lo);
+  (void)hipFree(d_hi);
+  if (kerr != hipSuccess) return hip_err(errbuf, errbuf_len, "hshare kernel", kerr);
   return 0;
 }
 
