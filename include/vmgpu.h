@@ -138,6 +138,22 @@ int vmgpu_topk_pointwise(uint64_t handle, double k, int32_t reverse,
  * (rows sorted by (group, le), same-le rows pre-merged by the host;
  * group_offsets has n_groups+1 entries).  bucket_values is
  * [rows x n_grid] host memory; outputs are [n_groups x n_grid]. */
+/* histogram_avg/stddev/stdvar (transformHistogramAvg/Stddev/Stdvar +
+ * avgForLeTimeseries/stdvarForLeTimeseries): mode 0 avg, 1 stddev,
+ * 2 stdvar.  Same CSR bucket layout as vmgpu_histogram_quantile. */
+int vmgpu_histogram_stat(int32_t mode, const double* bucket_values,
+                         const double* les, const uint64_t* group_offsets,
+                         uint32_t n_groups, int32_t n_grid, double* out,
+                         char* errbuf, size_t errbuf_len);
+
+/* histogram_share (transformHistogramShare): le_req is the per-grid scalar
+ * row (getScalar semantics); out_lower/out_upper nullable bounds. */
+int vmgpu_histogram_share(const double* le_req, const double* bucket_values,
+                          const double* les, const uint64_t* group_offsets,
+                          uint32_t n_groups, int32_t n_grid, double* out,
+                          double* out_lower, double* out_upper,
+                          char* errbuf, size_t errbuf_len);
+
 int vmgpu_histogram_quantile(double phi, const double* bucket_values,
                              const double* les, const uint64_t* group_offsets,
                              uint32_t n_groups, int32_t n_grid,

@@ -667,3 +667,40 @@ def tf_apply(func_id, values, ts=None, arg1=None, arg2=None, scalar=0.0):
                   _ptr(a2, ctypes.c_double) if a2 is not None else None,
                   ctypes.c_double(scalar), _ptr(keep, ctypes.c_uint8))
     return v, keep
+
+
+def histogram_stat(mode, bucket_values, les, group_offsets):
+    """histogram_avg/stddev/stdvar oracle (mode 0/1/2)."""
+    l = lib()
+    l.vm_histogram_stat.restype = None
+    bv = np.ascontiguousarray(bucket_values, dtype=np.float64)
+    le = np.ascontiguousarray(les, dtype=np.float64)
+    off = np.ascontiguousarray(group_offsets, dtype=np.uint64)
+    ng = len(off) - 1
+    grid = bv.shape[1]
+    out = np.empty((ng, grid), dtype=np.float64)
+    l.vm_histogram_stat(ctypes.c_int32(mode), _ptr(bv, ctypes.c_double),
+                        _ptr(le, ctypes.c_double), _ptr(off, ctypes.c_uint64),
+                        ctypes.c_int64(ng), ctypes.c_int64(grid),
+                        _ptr(out, ctypes.c_double))
+    return out
+
+
+def histogram_share(le_req, bucket_values, les, group_offsets):
+    l = lib()
+    l.vm_histogram_share.restype = None
+    req = np.ascontiguousarray(le_req, dtype=np.float64)
+    bv = np.ascontiguousarray(bucket_values, dtype=np.float64)
+    le = np.ascontiguousarray(les, dtype=np.float64)
+    off = np.ascontiguousarray(group_offsets, dtype=np.uint64)
+    ng = len(off) - 1
+    grid = bv.shape[1]
+    out = np.empty((ng, grid), dtype=np.float64)
+    lo = np.empty((ng, grid), dtype=np.float64)
+    hi = np.empty((ng, grid), dtype=np.float64)
+    l.vm_histogram_share(_ptr(req, ctypes.c_double), _ptr(bv, ctypes.c_double),
+                         _ptr(le, ctypes.c_double), _ptr(off, ctypes.c_uint64),
+                         ctypes.c_int64(ng), ctypes.c_int64(grid),
+                         _ptr(out, ctypes.c_double), _ptr(lo, ctypes.c_double),
+                         _ptr(hi, ctypes.c_double))
+    return out, lo, hi

@@ -169,3 +169,104 @@ def test_smooth_exponential():
     exp, _ = oracle.tf_apply(123, v.copy(), arg1=sfs)
     got = tf.transform("smooth_exponential", v.copy(), args=[sfs])
     _bitwise(got, exp)
+
+
+# ---------------------------------------------------------------------------
+# histogram stat/share kernels + the full bucket pipeline
+# ---------------------------------------------------------------------------
+
+def _bucket_fixture(rng, n_groups=40, n_les=12, n_grid=96):
+    les, goff = [], [0]
+    rows = []
+    for g in range(n_groups):
+        ls = np.sort(rng.uniform(0.1, 100, n_les - 1))
+        ls = np.concatenate([ls, [np.inf]])
+        base = np.cumsum(np.abs(rng.standard_normal((n_les, n_grid))), axis=0)
+        base[rng.random((n_les, n_grid)) < 0.05] = math.nan  # broken buckets
+        rows.append(base)
+        les.extend(ls)
+        goff.append(goff[-1] + n_les)
+    return (np.concatenate(rows, axis=0), np.asarray(les),
+            np.asarray(goff, np.uint64))
+
+
+@pytest.mark.parametrize("mode", ["avg", "stddev", "stdvar"])
+def test_histogram_stat_kernel(mode):
+    import oracle
+    from victoriametrics_amd import engine
+    rng = np.random.default_rng(61)
+    bv, les, goff = _bucket_fixture(rng)
+    got = engine.histogram_stat(mode, bv, les, goff)
+    exp = oracle.histogram_stat({"avg": 0, "stddev": 1, "stdvar": 2}[mode],
+                                bv, les, goff)
+    np.testing.assert_array_equal(got.view(np.int64), exp.view(np.int64))
+
+
+@pytest.mark.parametrize("le_req", [0.5, 20.0, 99.5, -1.0, math.inf,
+                                    math.nan])
+def test_histogram_share_kernel(le_req):
+    import oracle
+    from victoriametrics_amd import engine
+    rng = np.random.default_rng(62)
+    bv, les, goff = _bucket_fixture(rng)
+    req = np.full(bv.shape[1], le_req)
+    got, glo, ghi = engine.histogram_share(req, bv, les, goff, bounds=True)
+    exp, elo, ehi = oracle.histogram_share(req, bv, les, goff)
+    np.testing.assert_array_equal(got.view(np.int64), exp.view(np.int64))
+    np.testing.assert_array_equal(glo.view(np.int64), elo.view(np.int64))
+    np.testing.assert_array_equal(ghi.view(np.int64), ehi.view(np.int64))
+
+
+def test_histogram_transform_pipeline():
+    """vmrange buckets -> le conversion -> grouped stat on device, whole
+    host pipeline."""
+    import oracle
+    from victoriametrics_amd import transform as tfm
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+    rng = np.random.default_rng(63)
+    series = []
+    edges = [0.5, 1.0, 2.0, 4.0, 8.0]
+    for pod in ("a", "b", "c"):
+        for lo, hi in zip(edges[:-1], edges[1:]):
+            v = np.abs(rng.standard_normal(32)) * 3
+            series.append(Series(
+                MetricName("req_duration",
+                           [("pod", pod), ("vmrange", f"{lo}...{hi}")]), v))
+    out = tfm.histogram_transform("histogram_quantile",
+                                  [s.copy_shallow() for s in series], arg=0.9)
+    assert len(out) == 3
+    # compare against the oracle walk on the same converted buckets
+    conv = tfm.vmrange_buckets_to_le([s.copy_shallow() for s in series])
+    m = tfm.group_le_timeseries(conv)
+    for k, xss in m.items():
+        xss.sort(key=lambda x: x[0])
+        xss = tfm._merge_same_le(xss)
+        bv = np.stack([s.values for _, s in xss])
+        les = np.asarray([le for le, _ in xss])
+        goff = np.asarray([0, len(xss)], np.uint64)
+        exp, _, _ = oracle.histogram_quantile(0.9, bv, les, goff)
+        got = [s for s in out
+               if s.mn.marshal_sorted() == xss[0][1].mn.marshal_sorted()]
+        assert len(got) == 1
+        np.testing.assert_array_equal(got[0].values.view(np.int64),
+                                      exp[0].view(np.int64))
+
+
+def test_histogram_avg_stddev_pipeline():
+    from victoriametrics_amd import transform as tfm
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+    rng = np.random.default_rng(64)
+    series = []
+    for pod in ("x", "y"):
+        for le in (1.0, 2.0, 4.0, math.inf):
+            v = np.cumsum(np.abs(rng.standard_normal(16)))
+            series.append(Series(
+                MetricName("lat", [("pod", pod), ("le", str(le))]), v))
+    for name in ("histogram_avg", "histogram_stddev", "histogram_stdvar"):
+        out = tfm.histogram_transform(name,
+                                      [s.copy_shallow() for s in series])
+        assert len(out) == 2
+        for s in out:
+            assert s.values.shape == (16,)

@@ -237,3 +237,85 @@ def test_sort_by_label():
 def test_limit_offset():
     ss = [S("m", [("i", str(i))], [1]) for i in range(5)]
     assert tf.limit_offset(ss, 2, 1) == ss[1:3]
+
+
+# ---------------------------------------------------------------------------
+# vmrangeBucketsToLE pins (transcribed from transform_test.go:70-210)
+# ---------------------------------------------------------------------------
+
+def _vmrange(buckets):
+    out = []
+    for rng, v in buckets:
+        out.append(S("foo", [("vmrange", rng)], [float(v)]))
+    return out
+
+
+def _le_result(series):
+    return [(s.mn.get_tag_value("le").decode(), float(s.values[0]))
+            for s in series]
+
+
+@pytest.mark.parametrize("buckets,expected", [
+    # single non-empty bucket
+    ([("4.084e+02...4.642e+02", 2)],
+     [("4.084e+02", 0), ("4.642e+02", 2), ("+Inf", 2)]),
+    # 0...+Inf: no gap series (Go zero-value xsPrev.end == 0)
+    ([("0...+Inf", 5)], [("+Inf", 5)]),
+    ([("-Inf...0", 4)], [("-Inf", 0), ("0", 4), ("+Inf", 4)]),
+    ([("-Inf...+Inf", 1.23)], [("-Inf", 0), ("+Inf", 1.23)]),
+    ([("0...0", 5.3)], [("0", 5.3), ("+Inf", 5.3)]),
+    # adjacent empty bucket merged away
+    ([("7.743e+05...8.799e+05", 5), ("6.813e+05...7.743e+05", 0)],
+     [("7.743e+05", 0), ("8.799e+05", 5), ("+Inf", 5)]),
+    # multiple adjacent empty buckets
+    ([("7.743e+05...8.799e+05", 5), ("6.813e+05...7.743e+05", 0),
+      ("5.813e+05...6.813e+05", 0)],
+     [("7.743e+05", 0), ("8.799e+05", 5), ("+Inf", 5)]),
+    ([("8.799e+05...9.813e+05", 0), ("7.743e+05...8.799e+05", 5),
+      ("6.813e+05...7.743e+05", 0), ("5.813e+05...6.813e+05", 0)],
+     [("7.743e+05", 0), ("8.799e+05", 5), ("+Inf", 5)]),
+    # multiple non-empty buckets
+    ([("4.084e+02...4.642e+02", 2), ("1.234e+02...4.084e+02", 3)],
+     [("1.234e+02", 0), ("4.084e+02", 3), ("4.642e+02", 5), ("+Inf", 5)]),
+    # disjoint buckets
+    ([("1...2", 2), ("4...6", 3)],
+     [("1", 0), ("2", 2), ("4", 2), ("6", 5), ("+Inf", 5)]),
+    # intersected buckets
+    ([("1...5", 2), ("4...6", 3)],
+     [("1", 0), ("5", 2), ("4", 2), ("6", 5), ("+Inf", 5)]),
+    # same end range: short series refuse the merge, dropped duplicate
+    ([("1...5", 2), ("0...5", 3)],
+     [("1", 0), ("5", 2), ("0", 2), ("+Inf", 2)]),
+    # single/multiple empty buckets vanish
+    ([("0...1", 0)], []),
+    ([("0...+Inf", 0)], []),
+    ([("-Inf...0", 0)], []),
+    ([("0...0", 0)], []),
+    ([("-Inf...+Inf", 0)], []),
+    ([("2...3", 0), ("1...2", 0)], []),
+])
+def test_vmrange_buckets_to_le(buckets, expected):
+    got = tf.vmrange_buckets_to_le(_vmrange(buckets))
+    assert _le_result(got) == [(le, float(v)) for le, v in expected]
+
+
+def test_vmrange_le_passthrough():
+    le_series = [S("foo", [("le", "10")], [3.0])]
+    out = tf.vmrange_buckets_to_le(le_series)
+    assert out == le_series
+
+
+def test_group_le_and_merge_same_le():
+    xs = [S("m", [("le", "1"), ("pod", "a")], [1.0]),
+          S("m", [("le", "2"), ("pod", "a")], [2.0]),
+          S("m", [("le", "2"), ("pod", "a")], [3.0]),
+          S("m", [("le", "5"), ("pod", "b")], [4.0]),
+          S("m", [("le", "bad"), ("pod", "b")], [9.0])]
+    m = tf.group_le_timeseries(xs)
+    assert len(m) == 2
+    groups = sorted(m.values(), key=len)
+    assert [le for le, _ in groups[0]] == [5.0]
+    big = sorted(groups[1], key=lambda x: x[0])
+    merged = tf._merge_same_le(big)
+    assert [le for le, _ in merged] == [1.0, 2.0]
+    assert merged[1][1].values[0] == 5.0  # 2+3 summed

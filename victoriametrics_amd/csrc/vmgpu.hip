@@ -1876,9 +1876,12 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
   p.drop_stale = plan->drop_stale_nans;
   p.chunk_wave = (int32_t)std::min<uint32_t>(
       CHUNK_WAVE, std::max<uint32_t>(64, (b.max_wave_len + 63) & ~63u));
-  /* rate boundary caches (see the wave kernel): prefer the sample-scatter
-   * Et/Ev/J map when the plan window is a known step multiple; fall back to
-   * the u16 j-cache.  Bounded by the 64 KiB dynamic-LDS launch limit. */
+  /* rate boundary caches (see the wave kernel): the u16 j-cache
+   * (shared-boundary path) is the measured default — both scatter
+   * variants benched SLOWER at config 2 (full Et/Ev/J: 3.27 ms from LDS
+   * occupancy loss; J-only scatter: 2.77 ms from the boundary-fixup +
+   * divergent write cost) vs 2.51 ms for the j-cache.  Mode 2 remains
+   * selectable for grids too large for the u16 cache. */
   p.jbuf_elems = 0;
   p.jbuf_mode = 0;
   if ((plan->func == VMF_RATE || plan->func == VMF_DERIV_FAST) && n_grid > 1) {
@@ -1886,17 +1889,16 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
     int64_t dgp = (plan->window > 0 && plan->step > 0 &&
                    plan->window % plan->step == 0)
                       ? plan->window / plan->step : 0;
-    if (dgp > 0 && n_grid + dgp <= 32000) {
+    if (n_grid <= 4096 &&
+        base + WAVES_PER_BLOCK * vm_jbuf_bytes(1, n_grid) <= 64 * 1024) {
+      p.jbuf_mode = 1;
+      p.jbuf_elems = n_grid;
+    } else if (dgp > 0 && n_grid + dgp <= 32000) {
       int32_t elems = (int32_t)(n_grid + dgp);
       if (base + WAVES_PER_BLOCK * vm_jbuf_bytes(2, elems) <= 64 * 1024) {
         p.jbuf_mode = 2;
         p.jbuf_elems = elems;
       }
-    }
-    if (p.jbuf_mode == 0 && n_grid <= 4096 &&
-        base + WAVES_PER_BLOCK * vm_jbuf_bytes(1, n_grid) <= 64 * 1024) {
-      p.jbuf_mode = 1;
-      p.jbuf_elems = n_grid;
     }
   }
   p.arg = plan->arg;

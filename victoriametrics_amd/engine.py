@@ -489,6 +489,66 @@ def histogram_quantile(phi, bucket_values, les, group_offsets, bounds=False):
     return (out, lo, hi) if bounds else (out, None, None)
 
 
+HSTAT_MODES = {"avg": 0, "stddev": 1, "stdvar": 2}
+
+
+def histogram_stat(mode, bucket_values, les, group_offsets):
+    """histogram_avg/stddev/stdvar (transformHistogramAvg/Stddev/Stdvar)
+    over the same CSR bucket layout as histogram_quantile."""
+    init()
+    lib = _load_lib()
+    bv = np.ascontiguousarray(bucket_values, dtype=np.float64)
+    le = np.ascontiguousarray(les, dtype=np.float64)
+    off = np.ascontiguousarray(group_offsets, dtype=np.uint64)
+    n_groups = len(off) - 1
+    n_grid = bv.shape[1]
+    out = np.empty((n_groups, n_grid), dtype=np.float64)
+    errbuf = ctypes.create_string_buffer(256)
+    rc = lib.vmgpu_histogram_stat(
+        ctypes.c_int32(HSTAT_MODES[mode] if isinstance(mode, str) else mode),
+        bv.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        le.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        off.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        ctypes.c_uint32(n_groups), ctypes.c_int32(n_grid),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_double)),
+        errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise VmGpuError(f"vmgpu_histogram_stat failed ({rc}): "
+                         f"{errbuf.value.decode()}")
+    return out
+
+
+def histogram_share(le_req, bucket_values, les, group_offsets, bounds=False):
+    """histogram_share (transformHistogramShare); le_req is the per-grid
+    scalar row."""
+    init()
+    lib = _load_lib()
+    req = np.ascontiguousarray(le_req, dtype=np.float64)
+    bv = np.ascontiguousarray(bucket_values, dtype=np.float64)
+    le = np.ascontiguousarray(les, dtype=np.float64)
+    off = np.ascontiguousarray(group_offsets, dtype=np.uint64)
+    n_groups = len(off) - 1
+    n_grid = bv.shape[1]
+    out = np.empty((n_groups, n_grid), dtype=np.float64)
+    lo = np.empty((n_groups, n_grid), dtype=np.float64) if bounds else None
+    hi = np.empty((n_groups, n_grid), dtype=np.float64) if bounds else None
+    errbuf = ctypes.create_string_buffer(256)
+    dp = ctypes.POINTER(ctypes.c_double)
+    rc = lib.vmgpu_histogram_share(
+        req.ctypes.data_as(dp), bv.ctypes.data_as(dp),
+        le.ctypes.data_as(dp),
+        off.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        ctypes.c_uint32(n_groups), ctypes.c_int32(n_grid),
+        out.ctypes.data_as(dp),
+        lo.ctypes.data_as(dp) if bounds else None,
+        hi.ctypes.data_as(dp) if bounds else None,
+        errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise VmGpuError(f"vmgpu_histogram_share failed ({rc}): "
+                         f"{errbuf.value.decode()}")
+    return (out, lo, hi) if bounds else (out, None, None)
+
+
 def last_kernel_ms():
     lib = _load_lib()
     ms = ctypes.c_double(0)

@@ -366,3 +366,176 @@ def sort_series(series_list, desc=False):
                 return v[i]
         return -math.inf
     return sorted(series_list, key=key, reverse=desc)
+
+
+# ---------------------------------------------------------------------------
+# histogram transforms (transform.go:512 vmrangeBucketsToLE + :992-1190)
+# ---------------------------------------------------------------------------
+
+def _is_zero_series(s):
+    return not np.any(s.values > 0)
+
+
+def _merge_non_overlapping(dst, src):
+    from .binary_op import merge_non_overlapping
+    return merge_non_overlapping(dst, src)
+
+
+def vmrange_buckets_to_le(series_list):
+    """vmrangeBucketsToLE (transform.go:512): convert VictoriaMetrics
+    `vmrange=start...end` buckets into cumulative Prometheus `le` buckets;
+    Prometheus-style `le` series pass through."""
+    rvs = []
+    m = {}
+    for s in series_list:
+        vmrange = s.mn.get_tag_value("vmrange")
+        if not vmrange:
+            if s.mn.get_tag_value("le"):
+                rvs.append(s)
+            continue
+        txt = vmrange.decode("utf-8", "surrogateescape")
+        if "..." not in txt:
+            continue
+        start_str, _, end_str = txt.partition("...")
+        try:
+            start = float(start_str)
+            end = float(end_str)
+        except ValueError:
+            continue
+        s.mn.remove_tag("le")
+        s.mn.remove_tag("vmrange")
+        k = s.mn.marshal_sorted()
+        m.setdefault(k, []).append((start_str, end_str, start, end, s))
+
+    for xss in m.values():
+        xss.sort(key=lambda x: x[3])
+        xss_new = []
+        prev = None
+        uniq = {}
+
+        def copy_ts(src, le_str):
+            ts = src.copy_shallow()
+            ts.values[:] = 0.0
+            ts.mn.remove_tag("le")
+            ts.mn.add_tag("le", le_str)
+            return ts
+
+        prev_end = 0.0  # Go zero-value xsPrev.end: a first bucket starting
+                        # at 0 produces no gap series (transform_test.go:89)
+        for start_str, end_str, start, end, s in xss:
+            if _is_zero_series(s):
+                continue
+            if start != prev_end:
+                if uniq.get(start_str) is None:
+                    uniq[start_str] = s
+                    xss_new.append((None, start_str, None, start,
+                                    copy_ts(s, start_str)))
+            s.mn.add_tag("le", end_str)
+            prev_ts = uniq.get(end_str)
+            if prev_ts is not None:
+                _merge_non_overlapping(prev_ts, s)
+            else:
+                xss_new.append((start_str, end_str, start, end, s))
+                uniq[end_str] = s
+            prev = (start_str, end_str, start, end, s)
+            prev_end = end
+        if prev is not None and not math.isinf(prev[3]) \
+                and not _is_zero_series(prev[4]):
+            xss_new.append((None, "+Inf", None, math.inf,
+                            copy_ts(prev[4], "+Inf")))
+        if not xss_new:
+            continue
+        # cumulative counts per grid point
+        n_grid = len(xss_new[0][4].values)
+        count = np.zeros(n_grid)
+        for _, _, _, _, s in xss_new:
+            v = s.values
+            pos = ~np.isnan(v) & (v > 0)
+            count = count + np.where(pos, v, 0.0)
+            s.values = count.copy()
+        rvs.extend(s for _, _, _, _, s in xss_new)
+    return rvs
+
+
+def group_le_timeseries(series_list):
+    """groupLeTimeseries (transform.go): key = sorted tags minus le, with
+    the metric group reset; value = [(le, Series)]."""
+    m = {}
+    for s in series_list:
+        lev = s.mn.get_tag_value("le")
+        if not lev:
+            continue
+        try:
+            le = float(lev.decode("utf-8", "surrogateescape"))
+        except ValueError:
+            continue
+        s.mn.reset_metric_group()
+        s.mn.remove_tag("le")
+        m.setdefault(s.mn.marshal_sorted(), []).append((le, s))
+    return m
+
+
+def _merge_same_le(xss):
+    # mergeSameLE (transform.go): sum values of equal-le buckets
+    dst = [xss[0]]
+    for le, s in xss[1:]:
+        if le != dst[-1][0]:
+            dst.append((le, s))
+        else:
+            dst[-1][1].values = dst[-1][1].values + s.values
+    return dst
+
+
+def histogram_transform(name, series_list, arg=None, bounds_label=None):
+    """histogram_quantile / histogram_avg / histogram_stddev /
+    histogram_stdvar / histogram_share over raw bucket series (vmrange or
+    le).  arg: phi (quantile) or le (share).  Returns result Series list
+    (+ lower/upper series when bounds_label is set)."""
+    from . import engine
+    from .binary_op import Series
+    tss = vmrange_buckets_to_le(series_list)
+    m = group_le_timeseries(tss)
+    if not m:
+        return []
+    group_keys, rows, les, goff = [], [], [], [0]
+    dsts = []
+    for k, xss in m.items():
+        xss.sort(key=lambda x: x[0])
+        if name in ("histogram_quantile", "histogram_share"):
+            xss = _merge_same_le(xss)
+        for le, s in xss:
+            rows.append(s.values)
+            les.append(le)
+        goff.append(len(rows))
+        dsts.append(xss[0][1])
+        group_keys.append(k)
+    bv = np.stack(rows)
+    les = np.asarray(les)
+    goff = np.asarray(goff, np.uint64)
+    n_grid = bv.shape[1]
+    lo = hi = None
+    if name == "histogram_quantile":
+        out, lo, hi = engine.histogram_quantile(
+            float(arg), bv, les, goff, bounds=bounds_label is not None)
+    elif name == "histogram_share":
+        req = np.full(n_grid, float(arg))
+        out, lo, hi = engine.histogram_share(
+            req, bv, les, goff, bounds=bounds_label is not None)
+    else:
+        mode = {"histogram_avg": 0, "histogram_stddev": 1,
+                "histogram_stdvar": 2}[name]
+        out = engine.histogram_stat(mode, bv, les, goff)
+    rvs = []
+    for gi, dst in enumerate(dsts):
+        if bounds_label is not None and lo is not None:
+            sl = Series(dst.mn.copy(), lo[gi])
+            sl.mn.remove_tag(bounds_label)
+            sl.mn.add_tag(bounds_label, "lower")
+            su = Series(dst.mn.copy(), hi[gi])
+            su.mn.remove_tag(bounds_label)
+            su.mn.add_tag(bounds_label, "upper")
+            rvs.append(sl)
+            rvs.append(su)
+        dst.values = out[gi]
+        rvs.append(dst)
+    return rvs
