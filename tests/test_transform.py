@@ -319,3 +319,39 @@ def test_group_le_and_merge_same_le():
     merged = tf._merge_same_le(big)
     assert [le for le, _ in merged] == [1.0, 2.0]
     assert merged[1][1].values[0] == 5.0  # 2+3 summed
+
+
+def test_absent_union():
+    a = S("m", [], [1.0, NAN, NAN])
+    b = S("m2", [], [NAN, 2.0, NAN])
+    out = tf.absent([a, b], 3)
+    v = out[0].values
+    assert math.isnan(v[0]) and math.isnan(v[1]) and v[2] == 1.0
+    out2 = tf.absent([], 2)
+    assert list(out2[0].values) == [1.0, 1.0]
+    # union keeps first occurrence per name
+    u = tf.union([[S("m", [("x", "1")], [1])],
+                  [S("m", [("x", "1")], [2]), S("m", [("x", "2")], [3])]])
+    assert len(u) == 2
+    assert u[0].values[0] == 1
+
+
+def test_buckets_limit():
+    # 6 buckets with concentrated hits; limit to 3 keeps first/last
+    series = []
+    cum = [1.0, 50.0, 51.0, 52.0, 99.0, 100.0]
+    for le, v in zip(["1", "2", "3", "4", "5", "+Inf"], cum):
+        series.append(S("m", [("le", le)], [v]))
+    out = tf.buckets_limit(3, series)
+    les = sorted(float(s.mn.get_tag_value("le")) for s in out)
+    assert len(out) == 3
+    assert les[0] == 1.0 and math.isinf(les[-1])
+    # under the limit: untouched
+    out2 = tf.buckets_limit(10, [s.copy_shallow() for s in series])
+    assert len(out2) == 6
+
+
+def test_any_representative_group_ids():
+    from victoriametrics_amd.engine import any_representative_group_ids
+    gids = any_representative_group_ids([2, 0, 2, 1, 0, -1, 1])
+    assert list(gids) == [2, 0, -1, 1, -1, -1, -1]

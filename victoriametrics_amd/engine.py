@@ -59,6 +59,25 @@ AGGR_IDS = {
     "count": 5, "sum2": 6, "geomean": 7, "group": 8,
 }
 
+
+def any_representative_group_ids(group_ids):
+    """`any` incremental aggregate (aggr_incremental.go:54 updateAggrAny:
+    the first series of each group wins; worker order makes the reference's
+    pick scheduling-dependent — this engine picks the LOWEST series id,
+    deterministically).  Returns group_ids with non-representative members
+    masked to -1; run the plan with aggr="sum" (a single-member sum is the
+    identity) to get `any` semantics."""
+    gids = np.ascontiguousarray(group_ids, dtype=np.int32).copy()
+    seen = {}
+    for s, g in enumerate(gids):
+        if g < 0:
+            continue
+        if g in seen:
+            gids[s] = -1
+        else:
+            seen[g] = s
+    return gids
+
 # rollupFuncsRemoveCounterResets (rollup.go:223-232)
 REMOVE_COUNTER_RESETS_FUNCS = {
     "increase", "increase_prometheus", "increase_pure", "irate", "rate",

@@ -539,3 +539,100 @@ def histogram_transform(name, series_list, arg=None, bounds_label=None):
         dst.values = out[gi]
         rvs.append(dst)
     return rvs
+
+
+def absent(series_list, n_grid, base_mn=None):
+    """transformAbsent (transform.go:215): a single series of 1s where NO
+    input series has a value.  base_mn: the MetricName to copy constant
+    selector tags from (getAbsentTimeseries — the caller extracts them from
+    the AST's MetricExpr)."""
+    from .binary_op import Series
+    mn = base_mn.copy() if base_mn is not None else MetricName()
+    vals = np.ones(n_grid)
+    if series_list:
+        present = np.zeros(n_grid, dtype=bool)
+        for s in series_list:
+            present |= ~np.isnan(s.values)
+        vals[present] = math.nan
+    return [Series(mn, vals)]
+
+
+def union(args):
+    """transformUnion (transform.go:2716): concatenate series lists,
+    dropping later series whose metric name was already seen."""
+    rvs, seen = [], set()
+    for arg in args:
+        for s in arg:
+            k = s.mn.marshal_sorted()
+            if k in seen:
+                continue
+            seen.add(k)
+            rvs.append(s)
+    return rvs
+
+
+def prometheus_buckets(series_list):
+    """transformPrometheusBuckets = vmrangeBucketsToLE."""
+    return vmrange_buckets_to_le(series_list)
+
+
+def buckets_limit(limit, series_list):
+    """transformBucketsLimit (transform.go): merge lowest-hit buckets until
+    each le-group holds at most `limit` buckets (min 3), trimming
+    consecutively-empty edge buckets first."""
+    limit = int(limit)
+    if limit <= 0:
+        raise ValueError("limit must be greater than 0")
+    if limit < 3:
+        limit = 3
+    tss = vmrange_buckets_to_le(series_list)
+    if not tss:
+        return []
+    m = {}
+    for s in tss:
+        le_str = s.mn.get_tag_value("le")
+        if not le_str:
+            continue
+        try:
+            le = float(le_str.decode("utf-8", "surrogateescape"))
+        except ValueError:
+            continue
+        mn = s.mn.copy()
+        mn.remove_tag("le")
+        m.setdefault(mn.marshal_sorted(), []).append([le, 0.0, s])
+    rvs = []
+    for le_group in m.values():
+        if len(le_group) <= limit:
+            rvs.extend(x[2] for x in le_group)
+            continue
+        le_group.sort(key=lambda x: x[0])
+        n_pts = len(le_group[0][2].values)
+        for n in range(n_pts):
+            prev = 0.0
+            for x in le_group:
+                v = x[2].values[n]
+                x[1] += v - prev
+                prev = v
+        eps = 1e-9
+
+        def empty(h):
+            return not math.isnan(h) and abs(h) < eps
+
+        l, r = 0, len(le_group) - 1
+        while r - l + 1 > limit and empty(le_group[r][1]):
+            r -= 1
+        while r - l + 1 > limit and empty(le_group[l][1]):
+            l += 1
+        le_group = le_group[l:r + 1]
+        while len(le_group) > limit:
+            min_idx = 1
+            min_hits = le_group[1][1] + le_group[2][1]
+            for i in range(len(le_group) - 3):
+                h = le_group[i + 1][1] + le_group[i + 2][1]
+                if h < min_hits:
+                    min_idx = i + 1
+                    min_hits = h
+            le_group[min_idx + 1][1] += le_group[min_idx][1]
+            del le_group[min_idx]
+        rvs.extend(x[2] for x in le_group)
+    return rvs
