@@ -104,11 +104,6 @@ def marshal_from_batch(batch, names, timestamps, rows=None, max_size=1 << 62):
         rows = n
     assert rows == n
     head = struct.pack(">QQ", n, npts) + timestamps.tobytes()
-    out = bytearray(len(head) + 8 * n * npts)
-    out[:len(head)] = head
-    dst = np.frombuffer(out, dtype=np.float64, count=n * npts,
-                        offset=len(head)).reshape(n, npts)
-    got, _ = batch.fetch_out_into(dst, npts)
     tail = bytearray()
     for group, tags in names:
         tail += struct.pack(">H", len(group)) + bytes(group)
@@ -116,7 +111,14 @@ def marshal_from_batch(batch, names, timestamps, rows=None, max_size=1 << 62):
         for k, v in tags:
             tail += struct.pack(">H", len(k)) + bytes(k)
             tail += struct.pack(">H", len(v)) + bytes(v)
-    out += tail
+    vbytes = 8 * n * npts
+    out = bytearray(len(head) + vbytes + len(tail))
+    out[:len(head)] = head
+    out[len(head) + vbytes:] = tail
+    dst = np.frombuffer(out, dtype=np.float64, count=n * npts,
+                        offset=len(head)).reshape(n, npts)
+    batch.fetch_out_into(dst, npts)
+    del dst
     if len(out) > max_size:
         return b""
     return bytes(out)
