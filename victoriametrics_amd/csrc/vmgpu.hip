@@ -619,65 +619,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
   const uint32_t wave_id = blockIdx.x * WAVES_PER_BLOCK + wave_in_block;
   const uint32_t wave_stride = gridDim.x * WAVES_PER_BLOCK;
 
-  /* grouped-run register accumulation (rate fast path): series lists are
-   * sorted by group id (host), so a wave sees each group as a contiguous
-   * run — per-point results accumulate in registers across the run and
-   * flush with ONE atomic per (group, point) instead of one per
-   * (series, point).  Commutativity makes partial runs/other paths safe:
-   * flushing is only needed on group change and at kernel end. */
-  const bool acc_allowed =
-      (p.aggr != VMGPU_AGGR_NONE) && io.group_ids && p.n_grid <= 4 * WAVE;
-  double accv[4], accc[4];
-  int cur_grp = -1;
-  const double acc_init =
-      (p.aggr == VMGPU_AGGR_MIN) ? vm_dinf()
-      : (p.aggr == VMGPU_AGGR_MAX) ? -vm_dinf()
-      : (p.aggr == VMGPU_AGGR_GEOMEAN) ? 1.0 : 0.0;
-#pragma unroll
-  for (int u = 0; u < 4; u++) { accv[u] = acc_init; accc[u] = 0.0; }
-  auto acc_flush = [&]() {
-    if (cur_grp < 0) return;
-#pragma unroll
-    for (int u = 0; u < 4; u++) {
-      int g = u * WAVE + lane;
-      if (g < p.n_grid && accc[u] > 0.0) {
-        double* gv = io.out + (size_t)cur_grp * (size_t)p.n_grid + (size_t)g;
-        double* gc = io.out_counts + (size_t)cur_grp * (size_t)p.n_grid + (size_t)g;
-        switch (p.aggr) {
-          case VMGPU_AGGR_SUM:  atomicAdd(gv, accv[u]); *gc = 1.0; break;
-          case VMGPU_AGGR_AVG:  atomicAdd(gv, accv[u]); atomicAdd(gc, accc[u]); break;
-          case VMGPU_AGGR_MIN:  vm_atomic_min_f64(gv, accv[u]); *gc = 1.0; break;
-          case VMGPU_AGGR_MAX:  vm_atomic_max_f64(gv, accv[u]); *gc = 1.0; break;
-          case VMGPU_AGGR_COUNT:
-          case VMGPU_AGGR_GROUP: atomicAdd(gv, accc[u]); *gc = 1.0; break;
-          case VMGPU_AGGR_SUM2: atomicAdd(gv, accv[u]); *gc = 1.0; break;
-          case VMGPU_AGGR_GEOMEAN:
-            vm_atomic_mul_f64(gv, accv[u]); atomicAdd(gc, accc[u]); break;
-          default: break;
-        }
-      }
-      accv[u] = acc_init;
-      accc[u] = 0.0;
-    }
-    cur_grp = -1;
-  };
-
-  /* grouped: contiguous spans so the group-sorted runs stay within one
-   * wave; ungrouped: strided (XCD-spread) assignment as before */
-  uint32_t ws_begin, ws_end, ws_step;
-  if (acc_allowed) {
-    uint32_t span = (io.n_sel + wave_stride - 1) / wave_stride;
-    ws_begin = wave_id * span;
-    ws_end = ws_begin + span;
-    if (ws_end > io.n_sel) ws_end = io.n_sel;
-    if (ws_begin > io.n_sel) ws_begin = io.n_sel;
-    ws_step = 1;
-  } else {
-    ws_begin = wave_id;
-    ws_end = io.n_sel;
-    ws_step = wave_stride;
-  }
-  for (uint32_t ws = ws_begin; ws < ws_end; ws += ws_step) {
+  for (uint32_t ws = wave_id; ws < io.n_sel; ws += wave_stride) {
     uint32_t s = io.series_sel ? io.series_sel[ws] : ws;
     uint64_t lo = io.offsets[s];
     int64_t n = (int64_t)(io.offsets[s + 1] - lo);
@@ -798,12 +740,6 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
       if (p.jbuf_mode >= 1 &&
           (size_t)p.n_grid * 2 <= vm_jbuf_bytes(p.jbuf_mode, p.jbuf_elems) &&
           dg64 > 0 && count <= 65535) {
-        int sgrp = acc_allowed ? io.group_ids[s] : -1;
-        if (sgrp >= 0 && sgrp != cur_grp) {
-          acc_flush();
-          cur_grp = sgrp;
-        }
-        const bool use_acc = (sgrp >= 0);
         for (int g = lane; g < p.n_grid; g += WAVE) {
           int64_t t_end = p.start + (int64_t)g * p.step;
           int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
@@ -825,25 +761,8 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
                 int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
                 i = vm_ub_hint_fast(lts, count, t_start, gi);
               }
-              double v = eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start);
-              if (use_acc) {
-                if (!vm_isnan(v)) {
-                  switch (p.aggr) {
-                    case VMGPU_AGGR_SUM:
-                    case VMGPU_AGGR_AVG:  accv[u] += v; break;
-                    case VMGPU_AGGR_MIN:
-                      accv[u] = (v < accv[u]) ? v : accv[u]; break;
-                    case VMGPU_AGGR_MAX:
-                      accv[u] = (v > accv[u]) ? v : accv[u]; break;
-                    case VMGPU_AGGR_SUM2: accv[u] += v * v; break;
-                    case VMGPU_AGGR_GEOMEAN: accv[u] *= v; break;
-                    default: break; /* COUNT/GROUP count only */
-                  }
-                  accc[u] += 1.0;
-                }
-              } else {
-                vm_emit_value(p, io, s, g, v);
-              }
+              vm_emit_value(p, io, s, g,
+                            eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
               scanned += 2;
             }
           }
@@ -896,7 +815,6 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
     }
     wave_lds_sync();
   }
-  acc_flush();
   /* reduce samplesScanned: wave shuffle + one atomic per wave */
   for (int d = 32; d > 0; d >>= 1) scanned += __shfl_down((unsigned long long)scanned, d);
   if (lane == 0 && scanned) atomicAdd(io.samples_scanned, (unsigned long long)scanned);
@@ -1749,32 +1667,10 @@ int vmgpu_batch_create(const int64_t* ts, const double* vals,
       huge_total += n;
     }
   }
-  /* grouped batches: sort class lists by group id so each wave sees
-   * groups as contiguous runs (register accumulation in the wave kernel
-   * flushes once per run instead of once per series) */
-  if (group_ids) {
-    auto bygroup = [&](uint32_t a, uint32_t c) {
-      if (group_ids[a] != group_ids[c]) return group_ids[a] < group_ids[c];
-      return a < c;
-    };
-    std::sort(wave_list.begin(), wave_list.end(), bygroup);
-    std::sort(block_list.begin(), block_list.end(), bygroup);
-    if (!huge_list.empty()) {
-      std::sort(huge_list.begin(), huge_list.end(), bygroup);
-      uint64_t acc = 0;
-      for (size_t i = 0; i < huge_list.size(); i++) {
-        huge_scr_off[i] = acc;
-        acc += offsets[huge_list[i] + 1] - offsets[huge_list[i]];
-      }
-    }
-  }
   b.n_wave = (uint32_t)wave_list.size();
   b.n_block = (uint32_t)block_list.size();
   b.n_huge = (uint32_t)huge_list.size();
   b.wave_is_identity = (b.n_wave == n_series);
-  if (b.wave_is_identity && group_ids)
-    for (uint32_t i = 0; i < b.n_wave && b.wave_is_identity; i++)
-      b.wave_is_identity = (wave_list[i] == i);
   b.huge_scratch_elems = huge_total;
 
   HIP_TRY(hipMalloc(&b.d_ts, b.n_samples * sizeof(int64_t)), "alloc ts");
@@ -1867,29 +1763,10 @@ int vmgpu_batch_create_from_blocks(
       huge_total += n;
     }
   }
-  if (group_ids) {
-    auto bygroup = [&](uint32_t a, uint32_t c) {
-      if (group_ids[a] != group_ids[c]) return group_ids[a] < group_ids[c];
-      return a < c;
-    };
-    std::sort(wave_list.begin(), wave_list.end(), bygroup);
-    std::sort(block_list.begin(), block_list.end(), bygroup);
-    if (!huge_list.empty()) {
-      std::sort(huge_list.begin(), huge_list.end(), bygroup);
-      uint64_t acc = 0;
-      for (size_t i = 0; i < huge_list.size(); i++) {
-        huge_scr_off[i] = acc;
-        acc += out_offsets[huge_list[i] + 1] - out_offsets[huge_list[i]];
-      }
-    }
-  }
   b.n_wave = (uint32_t)wave_list.size();
   b.n_block = (uint32_t)block_list.size();
   b.n_huge = (uint32_t)huge_list.size();
   b.wave_is_identity = (b.n_wave == n_series);
-  if (b.wave_is_identity && group_ids)
-    for (uint32_t i = 0; i < b.n_wave && b.wave_is_identity; i++)
-      b.wave_is_identity = (wave_list[i] == i);
   b.huge_scratch_elems = huge_total;
 
 #define FBB_TRY(expr, what)                                                  \
