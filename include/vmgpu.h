@@ -138,6 +138,39 @@ int vmgpu_topk_pointwise(uint64_t handle, double k, int32_t reverse,
  * (rows sorted by (group, le), same-le rows pre-merged by the host;
  * group_offsets has n_groups+1 entries).  bucket_values is
  * [rows x n_grid] host memory; outputs are [n_groups x n_grid]. */
+/* Non-incremental cross-series aggregates (aggr.go long tail; the
+ * incremental ops fuse into the rollup kernels).  Per (group, grid point)
+ * over member rows (CSR group_rows/group_offsets into the values matrix):
+ * median/quantile/mad/mode/distinct/stddev/stdvar reduce to
+ * out[n_groups x n_grid]; share/zscore rewrite member rows into
+ * values_out [n_series x n_grid]; iqr_bounds emits lower/upper rows
+ * consumed by vmgpu_colagg_filter (outliers_iqr / outliers_mad). */
+#define VMGPU_COLAGG_MEDIAN 0
+#define VMGPU_COLAGG_QUANTILE 1
+#define VMGPU_COLAGG_MAD 2
+#define VMGPU_COLAGG_STDDEV 3
+#define VMGPU_COLAGG_STDVAR 4
+#define VMGPU_COLAGG_MODE 5
+#define VMGPU_COLAGG_DISTINCT 6
+#define VMGPU_COLAGG_SHARE 7
+#define VMGPU_COLAGG_ZSCORE 8
+#define VMGPU_COLAGG_IQR_BOUNDS 9
+
+int vmgpu_colagg(int32_t op, const double* values, uint32_t n_series,
+                 uint32_t n_grid, const uint32_t* group_rows,
+                 const uint64_t* group_offsets, uint32_t n_groups,
+                 double phi, double* out, double* out2,
+                 double* values_out, char* errbuf, size_t errbuf_len);
+
+/* mode 0 = IQR (b1=lower, b2=upper; aggrFuncOutliersIQR), mode 1 = MAD
+ * (b1=medians, b2=mads*tolerance; aggrFuncOutliersMAD).  flags[s]=1 when
+ * series s has any point outside its group's bounds. */
+int vmgpu_colagg_filter(int32_t mode, const double* values,
+                        const int32_t* group_of, uint32_t n_series,
+                        uint32_t n_grid, const double* b1, const double* b2,
+                        uint32_t n_groups, uint8_t* flags,
+                        char* errbuf, size_t errbuf_len);
+
 /* histogram_avg/stddev/stdvar (transformHistogramAvg/Stddev/Stdvar +
  * avgForLeTimeseries/stdvarForLeTimeseries): mode 0 avg, 1 stddev,
  * 2 stdvar.  Same CSR bucket layout as vmgpu_histogram_quantile. */

@@ -704,3 +704,53 @@ def histogram_share(le_req, bucket_values, les, group_offsets):
                          _ptr(out, ctypes.c_double), _ptr(lo, ctypes.c_double),
                          _ptr(hi, ctypes.c_double))
     return out, lo, hi
+
+
+def colagg(op, values, group_rows, group_offsets, phi=0.0):
+    """Non-incremental cross-series aggregate oracle (aggr.go long tail);
+    op ids/names match engine.COLAGG_OPS."""
+    _OPS = {"median": 0, "quantile": 1, "mad": 2, "stddev": 3, "stdvar": 4,
+            "mode": 5, "distinct": 6, "share": 7, "zscore": 8,
+            "iqr_bounds": 9}
+    opid = _OPS[op] if isinstance(op, str) else int(op)
+    l = lib()
+    l.vm_colagg.restype = None
+    v = np.ascontiguousarray(values, dtype=np.float64)
+    gr = np.ascontiguousarray(group_rows, dtype=np.uint32)
+    go = np.ascontiguousarray(group_offsets, dtype=np.uint64)
+    ns, ng = v.shape
+    n_groups = len(go) - 1
+    per_series = opid in (7, 8)
+    bounds = opid == 9
+    out = None if per_series else np.empty((n_groups, ng), np.float64)
+    out2 = np.empty((n_groups, ng), np.float64) if bounds else None
+    vout = np.empty_like(v) if per_series else None
+    l.vm_colagg(ctypes.c_int32(opid), _ptr(v, ctypes.c_double),
+                ctypes.c_uint32(ns), ctypes.c_uint32(ng),
+                _ptr(gr, ctypes.c_uint32), _ptr(go, ctypes.c_uint64),
+                ctypes.c_uint32(n_groups), ctypes.c_double(phi),
+                _ptr(out, ctypes.c_double) if out is not None else None,
+                _ptr(out2, ctypes.c_double) if out2 is not None else None,
+                _ptr(vout, ctypes.c_double) if vout is not None else None)
+    if bounds:
+        return out, out2
+    if per_series:
+        return vout
+    return out
+
+
+def colagg_filter(mode, values, group_of, b1, b2):
+    l = lib()
+    l.vm_colagg_filter.restype = None
+    v = np.ascontiguousarray(values, dtype=np.float64)
+    gof = np.ascontiguousarray(group_of, dtype=np.int32)
+    b1 = np.ascontiguousarray(b1, dtype=np.float64)
+    b2 = np.ascontiguousarray(b2, dtype=np.float64)
+    ns, ng = v.shape
+    flags = np.zeros(ns, dtype=np.uint8)
+    l.vm_colagg_filter(ctypes.c_int32(0 if mode == "iqr" else 1),
+                       _ptr(v, ctypes.c_double), _ptr(gof, ctypes.c_int32),
+                       ctypes.c_uint32(ns), ctypes.c_uint32(ng),
+                       _ptr(b1, ctypes.c_double), _ptr(b2, ctypes.c_double),
+                       _ptr(flags, ctypes.c_uint8))
+    return flags

@@ -1,0 +1,215 @@
+"""Non-incremental cross-series aggregates: oracle pins (mode vectors
+transcribed from aggr_test.go TestModeNoNaNs; hand-derived quantile/mad/
+share/zscore cases from aggr.go) — CPU side."""
+import math
+
+import numpy as np
+import pytest
+
+import oracle
+
+NAN = math.nan
+
+
+def _one_group(columns):
+    """columns: list of per-series single values -> [n x 1] matrix + CSR."""
+    v = np.asarray(columns, dtype=np.float64).reshape(-1, 1)
+    gr = np.arange(v.shape[0], dtype=np.uint32)
+    go = np.asarray([0, v.shape[0]], np.uint64)
+    return v, gr, go
+
+
+# modeNoNaNs vectors with the NaN seed (aggr_test.go:70-77); the aggregate
+# path always seeds prevValue=NaN (aggrFuncMode :446)
+@pytest.mark.parametrize("vals,expected", [
+    ([123], 123),
+    ([1, 2, 3], 1),
+    ([1, 2, 2], 2),
+    ([1, 1, 2], 1),
+    ([1, 1, 1], 1),
+    ([1, 2, 2, 3], 2),
+    ([1, 1, 2, 2, 3, 3, 3], 3),
+])
+def test_mode_vectors(vals, expected):
+    v, gr, go = _one_group(vals)
+    out = oracle.colagg("mode", v, gr, go)
+    assert out[0, 0] == expected
+
+
+def test_mode_empty():
+    v, gr, go = _one_group([NAN, NAN])
+    out = oracle.colagg("mode", v, gr, go)
+    assert math.isnan(out[0, 0])
+
+
+def test_median_quantile():
+    v, gr, go = _one_group([4.0, 1.0, 3.0, 2.0])
+    assert oracle.colagg("median", v, gr, go)[0, 0] == 2.5
+    assert oracle.colagg("quantile", v, gr, go, phi=0.0)[0, 0] == 1.0
+    assert oracle.colagg("quantile", v, gr, go, phi=1.0)[0, 0] == 4.0
+    # NaNs are filtered, not counted (quantile(), aggr.go)
+    v2, gr2, go2 = _one_group([4.0, NAN, 2.0])
+    assert oracle.colagg("median", v2, gr2, go2)[0, 0] == 3.0
+
+
+def test_mad():
+    v, gr, go = _one_group([1.0, 2.0, 3.0, 100.0])
+    # median 2.5; |v-2.5| = [1.5, .5, .5, 97.5]; median of sorted = 1.0
+    assert oracle.colagg("mad", v, gr, go)[0, 0] == 1.0
+
+
+def test_stddev_stdvar():
+    v, gr, go = _one_group([2.0, 4.0, 4.0, 4.0, 5.0, 5.0, 7.0, 9.0])
+    out = oracle.colagg("stdvar", v, gr, go)
+    assert abs(out[0, 0] - 4.0) < 1e-12
+    out2 = oracle.colagg("stddev", v, gr, go)
+    assert abs(out2[0, 0] - 2.0) < 1e-12
+    # single member: exactly zero (aggrFuncStdvar fast path semantics)
+    v1, gr1, go1 = _one_group([5.0])
+    assert oracle.colagg("stdvar", v1, gr1, go1)[0, 0] == 0.0
+    # all-NaN: NaN
+    vn, grn, gon = _one_group([NAN])
+    assert math.isnan(oracle.colagg("stdvar", vn, grn, gon)[0, 0])
+
+
+def test_distinct():
+    v, gr, go = _one_group([1.0, 1.0, 2.0, NAN, 3.0, 2.0])
+    assert oracle.colagg("distinct", v, gr, go)[0, 0] == 3.0
+    vn, grn, gon = _one_group([NAN])
+    assert math.isnan(oracle.colagg("distinct", vn, grn, gon)[0, 0])
+
+
+def test_share():
+    # negative and NaN values -> NaN; others divided by non-negative sum
+    v, gr, go = _one_group([2.0, 6.0, -1.0, NAN])
+    out = oracle.colagg("share", v, gr, go)
+    assert out[0, 0] == 0.25 and out[1, 0] == 0.75
+    assert math.isnan(out[2, 0]) and math.isnan(out[3, 0])
+
+
+def test_zscore():
+    v, gr, go = _one_group([1.0, 2.0, 3.0])
+    out = oracle.colagg("zscore", v, gr, go)
+    got = out[:, 0]
+    assert abs(got[1]) < 1e-12
+    assert abs(got[0] + got[2]) < 1e-12
+    # all-NaN column: values pass through
+    vn, grn, gon = _one_group([NAN, NAN])
+    outn = oracle.colagg("zscore", vn, grn, gon)
+    assert math.isnan(outn[0, 0])
+
+
+def test_iqr_bounds_and_filter():
+    vals = [1.0, 2.0, 3.0, 4.0, 100.0]
+    v, gr, go = _one_group(vals)
+    lower, upper = oracle.colagg("iqr_bounds", v, gr, go)
+    # q25 = 2, q75 = 4 -> iqr*1.5 = 3 -> bounds [-1, 7]
+    assert lower[0, 0] == -1.0 and upper[0, 0] == 7.0
+    flags = oracle.colagg_filter("iqr", v, np.zeros(5, np.int32),
+                                 lower, upper)
+    assert list(flags) == [0, 0, 0, 0, 1]
+
+
+def test_mad_filter():
+    vals = [1.0, 2.0, 3.0, 100.0]
+    v, gr, go = _one_group(vals)
+    med = oracle.colagg("median", v, gr, go)
+    mad = oracle.colagg("mad", v, gr, go)
+    tol = 3.0
+    flags = oracle.colagg_filter("mad", v, np.zeros(4, np.int32),
+                                 med, mad * tol)
+    # |v - 2.5| > 3*1.0 -> only 100.0
+    assert list(flags) == [0, 0, 0, 1]
+
+
+def test_multi_group_multi_point():
+    rng = np.random.default_rng(9)
+    v = rng.standard_normal((10, 7))
+    v[rng.random((10, 7)) < 0.2] = NAN
+    gr = np.asarray([0, 1, 2, 3, 4, 5, 6, 7, 8, 9], np.uint32)
+    go = np.asarray([0, 4, 10], np.uint64)
+    out = oracle.colagg("median", v, gr, go)
+    for grp, (lo, hi) in enumerate([(0, 4), (4, 10)]):
+        for g in range(7):
+            col = v[lo:hi, g]
+            col = col[~np.isnan(col)]
+            if len(col) == 0:
+                assert math.isnan(out[grp, g])
+            else:
+                exp = np.quantile(np.sort(col), 0.5) if len(col) else NAN
+                assert abs(out[grp, g] - float(np.median(col))) < 1e-12
+
+
+# ---------------------------------------------------------------------------
+# GPU parity
+# ---------------------------------------------------------------------------
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("op", ["median", "quantile", "mad", "stddev",
+                                "stdvar", "mode", "distinct"])
+def test_gpu_colagg_reduce(op):
+    from victoriametrics_amd import engine
+    rng = np.random.default_rng(abs(hash(op)) % 2**31)
+    n_series, n_grid, n_groups = 300, 120, 23
+    v = rng.standard_normal((n_series, n_grid)) * 50
+    v[rng.random((n_series, n_grid)) < 0.2] = NAN
+    # quantize some values so mode/distinct see duplicates
+    v[::3] = np.round(v[::3])
+    gids = rng.integers(0, n_groups, n_series)
+    order = np.argsort(gids, kind="stable").astype(np.uint32)
+    go = np.zeros(n_groups + 1, np.uint64)
+    for g in gids:
+        go[g + 1] += 1
+    go = np.cumsum(go).astype(np.uint64)
+    phi = 0.75
+    got = engine.colagg(op, v, order, go, phi=phi)
+    exp = oracle.colagg(op, v, order, go, phi=phi)
+    np.testing.assert_array_equal(got.view(np.int64), exp.view(np.int64),
+                                  err_msg=op)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("op", ["share", "zscore"])
+def test_gpu_colagg_per_series(op):
+    from victoriametrics_amd import engine
+    rng = np.random.default_rng(abs(hash(op)) % 2**31)
+    n_series, n_grid = 200, 80
+    v = rng.standard_normal((n_series, n_grid)) * 10
+    v[rng.random((n_series, n_grid)) < 0.25] = NAN
+    gids = rng.integers(0, 11, n_series)
+    order = np.argsort(gids, kind="stable").astype(np.uint32)
+    go = np.zeros(12, np.uint64)
+    for g in gids:
+        go[g + 1] += 1
+    go = np.cumsum(go).astype(np.uint64)
+    got = engine.colagg(op, v, order, go)
+    exp = oracle.colagg(op, v, order, go)
+    np.testing.assert_array_equal(got.view(np.int64), exp.view(np.int64),
+                                  err_msg=op)
+
+
+@pytest.mark.gpu
+def test_gpu_iqr_bounds_and_filters():
+    from victoriametrics_amd import engine
+    rng = np.random.default_rng(77)
+    n_series, n_grid = 150, 60
+    v = rng.standard_normal((n_series, n_grid)) * 20
+    v[rng.random((n_series, n_grid)) < 0.1] = NAN
+    gids = rng.integers(0, 7, n_series).astype(np.int32)
+    order = np.argsort(gids, kind="stable").astype(np.uint32)
+    go = np.zeros(8, np.uint64)
+    for g in gids:
+        go[g + 1] += 1
+    go = np.cumsum(go).astype(np.uint64)
+    glo, gup = engine.colagg("iqr_bounds", v, order, go)
+    elo, eup = oracle.colagg("iqr_bounds", v, order, go)
+    np.testing.assert_array_equal(glo.view(np.int64), elo.view(np.int64))
+    np.testing.assert_array_equal(gup.view(np.int64), eup.view(np.int64))
+    gf = engine.colagg_filter("iqr", v, gids, glo, gup)
+    ef = oracle.colagg_filter("iqr", v, gids, elo, eup)
+    np.testing.assert_array_equal(gf, ef)
+    med = engine.colagg("median", v, order, go)
+    mad = engine.colagg("mad", v, order, go)
+    gf2 = engine.colagg_filter("mad", v, gids, med, mad * 3.0)
+    ef2 = oracle.colagg_filter("mad", v, gids, med, mad * 3.0)
+    np.testing.assert_array_equal(gf2, ef2)

@@ -4,4 +4,4 @@
 set -e
 cd "$(dirname "$0")"
 hipcc --offload-arch=gfx950 -O3 -std=c++17 -ffp-contract=off -fPIC -shared \
-  vmgpu.hip decode.hip binop.hip transform.hip -o ../libvmgpu.so "$@"
+  vmgpu.hip decode.hip binop.hip transform.hip aggrcol.hip -o ../libvmgpu.so "$@"

@@ -858,3 +858,76 @@ def rollup_subquery(outer_func, start, end, step, window, sq_step,
                       keep_stale_nans=True)
     return rollup_eval(plan, ts, vals, offsets, group_ids=group_ids,
                        n_groups=n_groups)
+
+
+COLAGG_OPS = {
+    "median": 0, "quantile": 1, "mad": 2, "stddev": 3, "stdvar": 4,
+    "mode": 5, "distinct": 6, "share": 7, "zscore": 8, "iqr_bounds": 9,
+}
+
+
+def colagg(op, values, group_rows, group_offsets, phi=0.0):
+    """Non-incremental cross-series aggregate (aggr.go long tail) over
+    member rows per group.  Returns out [n_groups x n_grid] for reducing
+    ops, (lower, upper) for iqr_bounds, or a rewritten values matrix for
+    share/zscore."""
+    init()
+    lib = _load_lib()
+    opid = COLAGG_OPS[op] if isinstance(op, str) else int(op)
+    v = np.ascontiguousarray(values, dtype=np.float64)
+    gr = np.ascontiguousarray(group_rows, dtype=np.uint32)
+    go = np.ascontiguousarray(group_offsets, dtype=np.uint64)
+    n_series, n_grid = v.shape
+    n_groups = len(go) - 1
+    dp = ctypes.POINTER(ctypes.c_double)
+    per_series = opid in (7, 8)
+    bounds = opid == 9
+    out = None if per_series else np.empty((n_groups, n_grid), np.float64)
+    out2 = np.empty((n_groups, n_grid), np.float64) if bounds else None
+    vout = np.empty_like(v) if per_series else None
+    errbuf = ctypes.create_string_buffer(256)
+    rc = lib.vmgpu_colagg(
+        ctypes.c_int32(opid), v.ctypes.data_as(dp),
+        ctypes.c_uint32(n_series), ctypes.c_uint32(n_grid),
+        gr.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+        go.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+        ctypes.c_uint32(n_groups), ctypes.c_double(phi),
+        out.ctypes.data_as(dp) if out is not None else None,
+        out2.ctypes.data_as(dp) if out2 is not None else None,
+        vout.ctypes.data_as(dp) if vout is not None else None,
+        errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise VmGpuError(f"vmgpu_colagg failed ({rc}): "
+                         f"{errbuf.value.decode()}")
+    if bounds:
+        return out, out2
+    if per_series:
+        return vout
+    return out
+
+
+def colagg_filter(mode, values, group_of, b1, b2):
+    """Outlier series filter (aggrFuncOutliersIQR / OutliersMAD): returns
+    uint8 flags per series."""
+    init()
+    lib = _load_lib()
+    v = np.ascontiguousarray(values, dtype=np.float64)
+    gof = np.ascontiguousarray(group_of, dtype=np.int32)
+    b1 = np.ascontiguousarray(b1, dtype=np.float64)
+    b2 = np.ascontiguousarray(b2, dtype=np.float64)
+    n_series, n_grid = v.shape
+    flags = np.zeros(n_series, dtype=np.uint8)
+    dp = ctypes.POINTER(ctypes.c_double)
+    errbuf = ctypes.create_string_buffer(256)
+    rc = lib.vmgpu_colagg_filter(
+        ctypes.c_int32(0 if mode == "iqr" else 1), v.ctypes.data_as(dp),
+        gof.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
+        ctypes.c_uint32(n_series), ctypes.c_uint32(n_grid),
+        b1.ctypes.data_as(dp), b2.ctypes.data_as(dp),
+        ctypes.c_uint32(b1.shape[0]),
+        flags.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+        errbuf, ctypes.c_size_t(256))
+    if rc != 0:
+        raise VmGpuError(f"vmgpu_colagg_filter failed ({rc}): "
+                         f"{errbuf.value.decode()}")
+    return flags
