@@ -410,3 +410,54 @@ def test_subquery_rollup(engine):
         drop_stale_nans=False)
     assert scanned == ref_scanned
     assert_parity(out, ref, exact=True, context="subquery")
+
+
+def test_relayout_per_series_output_order():
+    """Grouped batches are physically relayouted by group id; per-series
+    outputs must still come back in the ORIGINAL series order (perm
+    unapplied on download)."""
+    from victoriametrics_amd import synth
+    from victoriametrics_amd.engine import RollupPlan, SeriesBatch
+    n_series, n_samples = 300, 100
+    ts, vals, offsets = synth.counter_batch(n_series, n_samples,
+                                            1_000_000_000_000)
+    start = int(ts[0]) + 120_000
+    plan = RollupPlan("rate", start, start + 20 * 15_000, 15_000,
+                      window=300_000)
+    # interleaved group ids force a non-identity relayout
+    gids = (np.arange(n_series) % 7).astype(np.int32)
+    grouped = SeriesBatch(ts, vals, offsets, group_ids=gids, n_groups=7)
+    assert grouped._perm is not None
+    out_g, _, _ = grouped.exec(plan)
+    grouped.close()
+    plain = SeriesBatch(ts, vals, offsets)
+    out_p, _, _ = plain.exec(plan)
+    plain.close()
+    np.testing.assert_array_equal(out_g.view(np.int64), out_p.view(np.int64))
+
+
+def test_relayout_grouped_matches_unrelayouted_semantics():
+    """sum by group on an interleaved-gid batch equals the host-computed
+    group sums of the per-series results."""
+    from victoriametrics_amd import synth
+    from victoriametrics_amd.engine import RollupPlan, SeriesBatch
+    n_series, n_samples = 256, 120
+    ts, vals, offsets = synth.counter_batch(n_series, n_samples,
+                                            1_000_000_000_000)
+    start = int(ts[0]) + 120_000
+    plan_g = RollupPlan("rate", start, start + 30 * 15_000, 15_000,
+                        window=300_000, aggr="sum")
+    plan_p = RollupPlan("rate", start, start + 30 * 15_000, 15_000,
+                        window=300_000)
+    gids = ((np.arange(n_series) * 13) % 9).astype(np.int32)
+    grouped = SeriesBatch(ts, vals, offsets, group_ids=gids, n_groups=9)
+    out_g, counts, _ = grouped.exec(plan_g)
+    grouped.close()
+    plain = SeriesBatch(ts, vals, offsets)
+    per, _, _ = plain.exec(plan_p)
+    plain.close()
+    for g in range(9):
+        member = per[gids == g]
+        exp = np.where(np.all(np.isnan(member), axis=0), np.nan,
+                       np.nansum(member, axis=0))
+        np.testing.assert_allclose(out_g[g], exp, rtol=1e-9, equal_nan=True)

@@ -598,7 +598,7 @@ static VM_DEV uint64_t eval_grid_point(const KPlan& p, const SeriesWindow& sw,
 /* kernel 1: one wave per series (n <= CHUNK_WAVE)                    */
 /* ------------------------------------------------------------------ */
 
-template <int FUNC_CT>
+template <int FUNC_CT, bool GACC = false>
 __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO io) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int wave_in_block = threadIdx.x / WAVE;
@@ -619,7 +619,66 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
   const uint32_t wave_id = blockIdx.x * WAVES_PER_BLOCK + wave_in_block;
   const uint32_t wave_stride = gridDim.x * WAVES_PER_BLOCK;
 
-  for (uint32_t ws = wave_id; ws < io.n_sel; ws += wave_stride) {
+  /* GACC (compile-time variant, launched only for grouped rate plans with
+   * n_grid <= 4*WAVE on a group-relayouted batch): per-point results
+   * accumulate in registers across each group's contiguous series run and
+   * flush with ONE atomic per (group, point).  Commutativity makes partial
+   * runs safe: flushing is only needed on group change and at kernel end. */
+  double accv[GACC ? 4 : 1], accc[GACC ? 4 : 1];
+  int cur_grp = -1;
+  double acc_init = 0.0;
+  if constexpr (GACC) {
+    acc_init = (p.aggr == VMGPU_AGGR_MIN) ? vm_dinf()
+               : (p.aggr == VMGPU_AGGR_MAX) ? -vm_dinf()
+               : (p.aggr == VMGPU_AGGR_GEOMEAN) ? 1.0 : 0.0;
+#pragma unroll
+    for (int u = 0; u < 4; u++) { accv[u] = acc_init; accc[u] = 0.0; }
+  }
+  auto acc_flush = [&]() {
+    if constexpr (GACC) {
+      if (cur_grp < 0) return;
+#pragma unroll
+      for (int u = 0; u < 4; u++) {
+        int g = u * WAVE + lane;
+        if (g < p.n_grid && accc[u] > 0.0) {
+          double* gv = io.out + (size_t)cur_grp * (size_t)p.n_grid + (size_t)g;
+          double* gc = io.out_counts + (size_t)cur_grp * (size_t)p.n_grid + (size_t)g;
+          switch (p.aggr) {
+            case VMGPU_AGGR_SUM:  atomicAdd(gv, accv[u]); *gc = 1.0; break;
+            case VMGPU_AGGR_AVG:  atomicAdd(gv, accv[u]); atomicAdd(gc, accc[u]); break;
+            case VMGPU_AGGR_MIN:  vm_atomic_min_f64(gv, accv[u]); *gc = 1.0; break;
+            case VMGPU_AGGR_MAX:  vm_atomic_max_f64(gv, accv[u]); *gc = 1.0; break;
+            case VMGPU_AGGR_COUNT:
+            case VMGPU_AGGR_GROUP: atomicAdd(gv, accc[u]); *gc = 1.0; break;
+            case VMGPU_AGGR_SUM2: atomicAdd(gv, accv[u]); *gc = 1.0; break;
+            case VMGPU_AGGR_GEOMEAN:
+              vm_atomic_mul_f64(gv, accv[u]); atomicAdd(gc, accc[u]); break;
+            default: break;
+          }
+        }
+        accv[u] = acc_init;
+        accc[u] = 0.0;
+      }
+      cur_grp = -1;
+    }
+  };
+
+  /* GACC: contiguous spans keep the group-relayouted runs within one wave;
+   * otherwise strided (XCD-spread) assignment as before */
+  uint32_t ws_begin, ws_end, ws_step;
+  if constexpr (GACC) {
+    uint32_t span = (io.n_sel + wave_stride - 1) / wave_stride;
+    ws_begin = wave_id * span;
+    ws_end = ws_begin + span;
+    if (ws_end > io.n_sel) ws_end = io.n_sel;
+    if (ws_begin > io.n_sel) ws_begin = io.n_sel;
+    ws_step = 1;
+  } else {
+    ws_begin = wave_id;
+    ws_end = io.n_sel;
+    ws_step = wave_stride;
+  }
+  for (uint32_t ws = ws_begin; ws < ws_end; ws += ws_step) {
     uint32_t s = io.series_sel ? io.series_sel[ws] : ws;
     uint64_t lo = io.offsets[s];
     int64_t n = (int64_t)(io.offsets[s + 1] - lo);
@@ -740,6 +799,15 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
       if (p.jbuf_mode >= 1 &&
           (size_t)p.n_grid * 2 <= vm_jbuf_bytes(p.jbuf_mode, p.jbuf_elems) &&
           dg64 > 0 && count <= 65535) {
+        bool use_acc = false;
+        if constexpr (GACC) {
+          int sgrp = io.group_ids ? io.group_ids[s] : -1;
+          use_acc = (sgrp >= 0);
+          if (use_acc && sgrp != cur_grp) {
+            acc_flush();
+            cur_grp = sgrp;
+          }
+        }
         for (int g = lane; g < p.n_grid; g += WAVE) {
           int64_t t_end = p.start + (int64_t)g * p.step;
           int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
@@ -761,8 +829,27 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
                 int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
                 i = vm_ub_hint_fast(lts, count, t_start, gi);
               }
-              vm_emit_value(p, io, s, g,
-                            eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
+              double v = eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start);
+              if (GACC && use_acc) {
+                if constexpr (GACC) {
+                  if (!vm_isnan(v)) {
+                    switch (p.aggr) {
+                      case VMGPU_AGGR_SUM:
+                      case VMGPU_AGGR_AVG:  accv[u] += v; break;
+                      case VMGPU_AGGR_MIN:
+                        accv[u] = (v < accv[u]) ? v : accv[u]; break;
+                      case VMGPU_AGGR_MAX:
+                        accv[u] = (v > accv[u]) ? v : accv[u]; break;
+                      case VMGPU_AGGR_SUM2: accv[u] += v * v; break;
+                      case VMGPU_AGGR_GEOMEAN: accv[u] *= v; break;
+                      default: break; /* COUNT/GROUP count only */
+                    }
+                    accc[u] += 1.0;
+                  }
+                }
+              } else {
+                vm_emit_value(p, io, s, g, v);
+              }
               scanned += 2;
             }
           }
@@ -815,6 +902,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
     }
     wave_lds_sync();
   }
+  acc_flush();
   /* reduce samplesScanned: wave shuffle + one atomic per wave */
   for (int d = 32; d > 0; d >>= 1) scanned += __shfl_down((unsigned long long)scanned, d);
   if (lane == 0 && scanned) atomicAdd(io.samples_scanned, (unsigned long long)scanned);
@@ -1479,8 +1567,18 @@ template <int FUNC_CT>
 static void launch_rollup_t(int which, uint32_t blocks, size_t lds,
                             const KPlan& p, const KIO& w, hipStream_t stream) {
   if (which == 0) {
-    hipLaunchKernelGGL(rollup_wave_kernel<FUNC_CT>, dim3(blocks),
-                       dim3(BLOCK_THREADS), lds, stream, p, w);
+    /* grouped rate plans with grids the accumulator covers take the
+     * register-run variant (GACC) — valid because grouped batches are
+     * physically relayouted by group at creation */
+    bool gacc = (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) &&
+                p.aggr != VMGPU_AGGR_NONE && w.group_ids != nullptr &&
+                p.n_grid <= 4 * WAVE;
+    if (gacc)
+      hipLaunchKernelGGL((rollup_wave_kernel<FUNC_CT, true>), dim3(blocks),
+                         dim3(BLOCK_THREADS), lds, stream, p, w);
+    else
+      hipLaunchKernelGGL((rollup_wave_kernel<FUNC_CT, false>), dim3(blocks),
+                         dim3(BLOCK_THREADS), lds, stream, p, w);
   } else if (which == 1) {
     hipLaunchKernelGGL(rollup_block_kernel<FUNC_CT>, dim3(blocks),
                        dim3(BLOCK_THREADS), lds, stream, p, w);
@@ -1548,6 +1646,10 @@ struct Batch {
   /* shape of the last exec's output */
   uint32_t last_rows = 0;
   int32_t last_grid = 0;
+  /* grouped batches are physically relayouted so each group's series are
+   * contiguous in HBM (locality for the grouped-run accumulation);
+   * perm[i] = original series id of physical row i (empty = identity) */
+  std::vector<uint32_t> perm;
 };
 
 struct Ctx {
@@ -1587,6 +1689,95 @@ int hip_err(char* errbuf, size_t len, const char* what, hipError_t e) {
     hipError_t _e = (expr);                              \
     if (_e != hipSuccess) return hip_err(errbuf, errbuf_len, what, _e); \
   } while (0)
+
+/* copy series perm[i] of the source CSR into dense row i of the new CSR */
+__global__ __launch_bounds__(BLOCK_THREADS) void relayout_kernel(
+    const int64_t* src_ts, const double* src_vals, const uint64_t* src_off,
+    const uint32_t* perm, const uint64_t* dst_off, uint32_t n_series,
+    int64_t* dst_ts, double* dst_vals) {
+  const int lane = threadIdx.x % WAVE;
+  const int wv = threadIdx.x / WAVE;
+  uint32_t wid = blockIdx.x * (BLOCK_THREADS / WAVE) + wv;
+  uint32_t stride = gridDim.x * (BLOCK_THREADS / WAVE);
+  for (uint32_t i = wid; i < n_series; i += stride) {
+    uint32_t s = perm[i];
+    uint64_t slo = src_off[s];
+    uint64_t dlo = dst_off[i];
+    uint64_t n = src_off[s + 1] - slo;
+    for (uint64_t k = lane; k < n; k += WAVE) {
+      dst_ts[dlo + k] = src_ts[slo + k];
+      dst_vals[dlo + k] = src_vals[slo + k];
+    }
+  }
+}
+
+/* Physically reorder a grouped batch so each group's series are contiguous
+ * (stable by (group, series)).  b.d_ts/d_vals must hold the ORIGINAL CSR
+ * (offsets_orig); on success they point at the relayouted CSR, b.perm maps
+ * physical row -> original series, new_offsets holds the physical CSR, and
+ * b.d_group_ids is uploaded in physical order.  Identity permutations skip
+ * all work.  Returns 0 or an error code. */
+static int relayout_by_group(Batch& b, const int32_t* group_ids,
+                             const uint64_t* offsets_orig, uint32_t n_series,
+                             std::vector<uint64_t>& new_offsets,
+                             char* errbuf, size_t errbuf_len) {
+  std::vector<uint32_t> perm(n_series);
+  for (uint32_t s = 0; s < n_series; s++) perm[s] = s;
+  std::stable_sort(perm.begin(), perm.end(), [&](uint32_t a, uint32_t c) {
+    return group_ids[a] < group_ids[c];
+  });
+  bool identity = true;
+  for (uint32_t s = 0; s < n_series && identity; s++)
+    identity = (perm[s] == s);
+  new_offsets.resize(n_series + 1);
+  new_offsets[0] = 0;
+  for (uint32_t i = 0; i < n_series; i++)
+    new_offsets[i + 1] = new_offsets[i] +
+                         (offsets_orig[perm[i] + 1] - offsets_orig[perm[i]]);
+  if (identity) {
+    for (uint32_t i = 0; i <= n_series; i++) new_offsets[i] = offsets_orig[i];
+    return 0;
+  }
+  uint64_t total = offsets_orig[n_series];
+  int64_t* d_ts2 = nullptr;
+  double* d_vals2 = nullptr;
+  uint64_t* d_soff = nullptr;
+  uint64_t* d_doff = nullptr;
+  uint32_t* d_perm = nullptr;
+  HIP_TRY(hipMalloc(&d_ts2, total * 8), "alloc relayout ts");
+  HIP_TRY(hipMalloc(&d_vals2, total * 8), "alloc relayout vals");
+  HIP_TRY(hipMalloc(&d_soff, (size_t)(n_series + 1) * 8), "alloc relayout soff");
+  HIP_TRY(hipMalloc(&d_doff, (size_t)(n_series + 1) * 8), "alloc relayout doff");
+  HIP_TRY(hipMalloc(&d_perm, (size_t)n_series * 4), "alloc relayout perm");
+  hipStream_t st = g_ctx.stream;
+  HIP_TRY(hipMemcpyAsync(d_soff, offsets_orig, (size_t)(n_series + 1) * 8,
+                         hipMemcpyHostToDevice, st), "ul soff");
+  HIP_TRY(hipMemcpyAsync(d_doff, new_offsets.data(), (size_t)(n_series + 1) * 8,
+                         hipMemcpyHostToDevice, st), "ul doff");
+  HIP_TRY(hipMemcpyAsync(d_perm, perm.data(), (size_t)n_series * 4,
+                         hipMemcpyHostToDevice, st), "ul perm");
+  uint32_t blocks = std::min<uint32_t>(
+      (n_series + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK, 4096);
+  hipLaunchKernelGGL(relayout_kernel, dim3(blocks), dim3(BLOCK_THREADS), 0, st,
+                     b.d_ts, b.d_vals, d_soff, d_perm, d_doff, n_series,
+                     d_ts2, d_vals2);
+  HIP_TRY(hipStreamSynchronize(st), "sync relayout");
+  hipError_t kerr = hipGetLastError();
+  (void)hipFree(d_soff);
+  (void)hipFree(d_doff);
+  (void)hipFree(d_perm);
+  if (kerr != hipSuccess) {
+    (void)hipFree(d_ts2);
+    (void)hipFree(d_vals2);
+    return hip_err(errbuf, errbuf_len, "relayout kernel", kerr);
+  }
+  (void)hipFree(b.d_ts);
+  (void)hipFree(b.d_vals);
+  b.d_ts = d_ts2;
+  b.d_vals = d_vals2;
+  b.perm = std::move(perm);
+  return 0;
+}
 
 void free_batch(Batch& b) {
   (void)hipFree(b.d_ts);
@@ -1650,12 +1841,40 @@ int vmgpu_batch_create(const int64_t* ts, const double* vals,
   b.n_groups = n_groups;
   b.n_samples = offsets[n_series];
 
-  /* partition series by length (host pass over offsets) */
+  HIP_TRY(hipMalloc(&b.d_ts, b.n_samples * sizeof(int64_t)), "alloc ts");
+  HIP_TRY(hipMalloc(&b.d_vals, b.n_samples * sizeof(double)), "alloc vals");
+  HIP_TRY(hipMemcpy(b.d_ts, ts, b.n_samples * sizeof(int64_t), hipMemcpyHostToDevice), "upload ts");
+  HIP_TRY(hipMemcpy(b.d_vals, vals, b.n_samples * sizeof(double), hipMemcpyHostToDevice), "upload vals");
+
+  /* grouped batches: physically relayout so each group's series are
+   * contiguous in HBM (see relayout_by_group).  eff_* describe the
+   * PHYSICAL CSR used by the kernels. */
+  std::vector<uint64_t> phys_offsets;
+  std::vector<int32_t> phys_gids;
+  const uint64_t* eff_offsets = offsets;
+  const int32_t* eff_gids = group_ids;
+  if (group_ids && n_groups > 0) {
+    int rrc = relayout_by_group(b, group_ids, offsets, n_series,
+                                phys_offsets, errbuf, errbuf_len);
+    if (rrc != 0) {
+      free_batch(b);
+      return rrc;
+    }
+    eff_offsets = phys_offsets.data();
+    if (!b.perm.empty()) {
+      phys_gids.resize(n_series);
+      for (uint32_t i = 0; i < n_series; i++)
+        phys_gids[i] = group_ids[b.perm[i]];
+      eff_gids = phys_gids.data();
+    }
+  }
+
+  /* partition series by length (host pass over the physical offsets) */
   std::vector<uint32_t> wave_list, block_list, huge_list;
   std::vector<uint64_t> huge_scr_off;
   uint64_t huge_total = 0;
   for (uint32_t s = 0; s < n_series; s++) {
-    uint64_t n = offsets[s + 1] - offsets[s];
+    uint64_t n = eff_offsets[s + 1] - eff_offsets[s];
     if (n <= CHUNK_WAVE) {
       wave_list.push_back(s);
       if ((uint32_t)n > b.max_wave_len) b.max_wave_len = (uint32_t)n;
@@ -1673,15 +1892,11 @@ int vmgpu_batch_create(const int64_t* ts, const double* vals,
   b.wave_is_identity = (b.n_wave == n_series);
   b.huge_scratch_elems = huge_total;
 
-  HIP_TRY(hipMalloc(&b.d_ts, b.n_samples * sizeof(int64_t)), "alloc ts");
-  HIP_TRY(hipMalloc(&b.d_vals, b.n_samples * sizeof(double)), "alloc vals");
   HIP_TRY(hipMalloc(&b.d_offsets, (n_series + 1) * sizeof(uint64_t)), "alloc offsets");
-  HIP_TRY(hipMemcpy(b.d_ts, ts, b.n_samples * sizeof(int64_t), hipMemcpyHostToDevice), "upload ts");
-  HIP_TRY(hipMemcpy(b.d_vals, vals, b.n_samples * sizeof(double), hipMemcpyHostToDevice), "upload vals");
-  HIP_TRY(hipMemcpy(b.d_offsets, offsets, (n_series + 1) * sizeof(uint64_t), hipMemcpyHostToDevice), "upload offsets");
+  HIP_TRY(hipMemcpy(b.d_offsets, eff_offsets, (n_series + 1) * sizeof(uint64_t), hipMemcpyHostToDevice), "upload offsets");
   if (group_ids) {
     HIP_TRY(hipMalloc(&b.d_group_ids, n_series * sizeof(int32_t)), "alloc gids");
-    HIP_TRY(hipMemcpy(b.d_group_ids, group_ids, n_series * sizeof(int32_t), hipMemcpyHostToDevice), "upload gids");
+    HIP_TRY(hipMemcpy(b.d_group_ids, eff_gids, n_series * sizeof(int32_t), hipMemcpyHostToDevice), "upload gids");
   }
   if (!b.wave_is_identity && b.n_wave) {
     HIP_TRY(hipMalloc(&b.d_wave_list, b.n_wave * 4), "alloc wave list");
@@ -1747,11 +1962,33 @@ int vmgpu_batch_create_from_blocks(
   b.d_ts = d_ts;
   b.d_vals = d_vals;
 
+  /* grouped: physically relayout (see relayout_by_group); out_offsets
+   * reported to the caller stay in ORIGINAL series order */
+  std::vector<uint64_t> phys_offsets;
+  std::vector<int32_t> phys_gids;
+  const uint64_t* eff_offsets = out_offsets;
+  const int32_t* eff_gids = group_ids;
+  if (group_ids && n_groups > 0) {
+    int rrc = relayout_by_group(b, group_ids, out_offsets, n_series,
+                                phys_offsets, errbuf, errbuf_len);
+    if (rrc != 0) {
+      free_batch(b);
+      return rrc;
+    }
+    eff_offsets = phys_offsets.data();
+    if (!b.perm.empty()) {
+      phys_gids.resize(n_series);
+      for (uint32_t i = 0; i < n_series; i++)
+        phys_gids[i] = group_ids[b.perm[i]];
+      eff_gids = phys_gids.data();
+    }
+  }
+
   std::vector<uint32_t> wave_list, block_list, huge_list;
   std::vector<uint64_t> huge_scr_off;
   uint64_t huge_total = 0;
   for (uint32_t s = 0; s < n_series; s++) {
-    uint64_t n = out_offsets[s + 1] - out_offsets[s];
+    uint64_t n = eff_offsets[s + 1] - eff_offsets[s];
     if (n <= CHUNK_WAVE) {
       wave_list.push_back(s);
       if ((uint32_t)n > b.max_wave_len) b.max_wave_len = (uint32_t)n;
@@ -1777,11 +2014,11 @@ int vmgpu_batch_create_from_blocks(
   } while (0)
 
   FBB_TRY(hipMalloc(&b.d_offsets, (n_series + 1) * sizeof(uint64_t)), "alloc offsets");
-  FBB_TRY(hipMemcpy(b.d_offsets, out_offsets, (n_series + 1) * sizeof(uint64_t),
+  FBB_TRY(hipMemcpy(b.d_offsets, eff_offsets, (n_series + 1) * sizeof(uint64_t),
                     hipMemcpyHostToDevice), "upload offsets");
   if (group_ids) {
     FBB_TRY(hipMalloc(&b.d_group_ids, n_series * sizeof(int32_t)), "alloc gids");
-    FBB_TRY(hipMemcpy(b.d_group_ids, group_ids, n_series * sizeof(int32_t),
+    FBB_TRY(hipMemcpy(b.d_group_ids, eff_gids, n_series * sizeof(int32_t),
                       hipMemcpyHostToDevice), "upload gids");
   }
   if (!b.wave_is_identity && b.n_wave) {
@@ -1810,6 +2047,22 @@ int vmgpu_batch_create_from_blocks(
   uint64_t h = g_ctx.next_handle++;
   g_ctx.batches[h] = b;
   *out_handle = h;
+  return 0;
+}
+
+/* Physical-row -> original-series mapping of a (relayouted) grouped batch:
+ * out_perm[n_series].  Identity batches fill 0..n-1.  Callers need this to
+ * place per-series outputs of vmgpu_rollup_exec back in request order. */
+int vmgpu_batch_perm(uint64_t handle, uint32_t* out_perm) {
+  std::lock_guard<std::mutex> lock(g_ctx.mu);
+  auto it = g_ctx.batches.find(handle);
+  if (it == g_ctx.batches.end() || !out_perm) return 1;
+  const Batch& b = it->second;
+  if (b.perm.empty()) {
+    for (uint32_t i = 0; i < b.n_series; i++) out_perm[i] = i;
+  } else {
+    for (uint32_t i = 0; i < b.n_series; i++) out_perm[i] = b.perm[i];
+  }
   return 0;
 }
 

@@ -263,6 +263,21 @@ class SeriesBatch:
             raise VmGpuError(f"vmgpu_batch_create failed ({rc}): "
                              f"{errbuf.value.decode()}")
         self.handle = handle.value
+        self._perm = self._fetch_perm() if group_ids is not None else None
+
+    def _fetch_perm(self):
+        """Physical-row -> original-series map of a group-relayouted
+        batch (vmgpu_batch_perm); None when identity."""
+        lib = _load_lib()
+        perm = np.empty(self.n_series, dtype=np.uint32)
+        rc = lib.vmgpu_batch_perm(
+            ctypes.c_uint64(self.handle),
+            perm.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)))
+        if rc != 0:
+            raise VmGpuError(f"vmgpu_batch_perm failed ({rc})")
+        if np.array_equal(perm, np.arange(self.n_series, dtype=np.uint32)):
+            return None
+        return perm
 
     @classmethod
     def from_blocks(cls, blocks, series_block_start, dedup_interval=0,
@@ -334,6 +349,7 @@ class SeriesBatch:
         self.handle = handle.value
         self._last_rows = 0
         self._last_n_grid = 0
+        self._perm = self._fetch_perm() if group_ids is not None else None
         return self
 
     def close(self):
@@ -376,6 +392,12 @@ class SeriesBatch:
                              f"{errbuf.value.decode()}")
         self._last_rows = self.n_groups if grouped else self.n_series
         self._last_n_grid = n_grid
+        if out is not None and not grouped and \
+                getattr(self, "_perm", None) is not None:
+            # physical row i holds original series _perm[i]
+            orig = np.empty_like(out)
+            orig[self._perm] = out
+            out = orig
         return out, counts, scanned.value
 
     def fetch_out_into(self, dst, n_grid):
