@@ -155,6 +155,18 @@ int vmgpu_topk_pointwise(uint64_t handle, double k, int32_t reverse,
 #define VMGPU_COLAGG_SHARE 7
 #define VMGPU_COLAGG_ZSCORE 8
 #define VMGPU_COLAGG_IQR_BOUNDS 9
+/* per-point column forms of the simple aggregates (aggrFuncSum/Min/Max/
+ * Avg/Count/Sum2/Geomean/Group, aggr.go:316-452) for series-level
+ * aggregation of resident result sets (the rollup-fused incremental forms
+ * cover the rollup->aggregate path) */
+#define VMGPU_COLAGG_SUM 10
+#define VMGPU_COLAGG_MIN 11
+#define VMGPU_COLAGG_MAX 12
+#define VMGPU_COLAGG_AVG 13
+#define VMGPU_COLAGG_COUNT 14
+#define VMGPU_COLAGG_SUM2 15
+#define VMGPU_COLAGG_GEOMEAN 16
+#define VMGPU_COLAGG_GROUP 17
 
 int vmgpu_colagg(int32_t op, const double* values, uint32_t n_series,
                  uint32_t n_grid, const uint32_t* group_rows,

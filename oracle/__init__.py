@@ -711,7 +711,8 @@ def colagg(op, values, group_rows, group_offsets, phi=0.0):
     op ids/names match engine.COLAGG_OPS."""
     _OPS = {"median": 0, "quantile": 1, "mad": 2, "stddev": 3, "stdvar": 4,
             "mode": 5, "distinct": 6, "share": 7, "zscore": 8,
-            "iqr_bounds": 9}
+            "iqr_bounds": 9, "sum": 10, "min": 11, "max": 12, "avg": 13,
+            "count": 14, "sum2": 15, "geomean": 16, "group": 17}
     opid = _OPS[op] if isinstance(op, str) else int(op)
     l = lib()
     l.vm_colagg.restype = None

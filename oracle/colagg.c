@@ -165,6 +165,32 @@ void vm_colagg(int32_t op, const double* values, uint32_t n_series,
           }
           break;
         }
+        case 10: case 11: case 12: case 13:
+        case 14: case 15: case 16: case 17: { /* sum..group */
+          double sum = 0, prod = 1, mn = ca_nan(), mx = ca_nan();
+          double cnt = 0;
+          for (uint32_t k = lo; k < hi; k++) {
+            double v = values[(size_t)group_rows[k] * n_grid + g];
+            if (isnan(v)) continue;
+            cnt++;
+            sum += (op == 15) ? v * v : v;
+            prod *= v;
+            if (isnan(mn) || v < mn) mn = v;
+            if (isnan(mx) || v > mx) mx = v;
+          }
+          double r;
+          switch (op) {
+            case 10: case 15: r = (cnt == 0) ? ca_nan() : sum; break;
+            case 11: r = mn; break;
+            case 12: r = mx; break;
+            case 13: r = (cnt == 0) ? ca_nan() : sum / cnt; break;
+            case 14: r = (cnt == 0) ? ca_nan() : cnt; break;
+            case 17: r = (cnt == 0) ? ca_nan() : 1.0; break;
+            default: r = (cnt == 0) ? ca_nan() : pow(prod, 1.0 / cnt); break;
+          }
+          out[e] = r;
+          break;
+        }
         case 9: { /* iqr bounds */
           int cnt = ca_sorted_col(values, n_grid, group_rows, lo, hi, g, sc);
           double q25 = ca_quantile_sorted(0.25, sc, cnt);

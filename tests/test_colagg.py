@@ -213,3 +213,85 @@ def test_gpu_iqr_bounds_and_filters():
     gf2 = engine.colagg_filter("mad", v, gids, med, mad * 3.0)
     ef2 = oracle.colagg_filter("mad", v, gids, med, mad * 3.0)
     np.testing.assert_array_equal(gf2, ef2)
+
+
+# ---------------------------------------------------------------------------
+# series-level aggregate dispatch (aggregate.py)
+# ---------------------------------------------------------------------------
+
+def _series_set():
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+    out = []
+    for pod, node, vals in [("a", "n1", [1, 2]), ("b", "n1", [3, 4]),
+                            ("c", "n2", [5, NAN]), ("d", "n2", [NAN, NAN])]:
+        out.append(Series(MetricName("m", [("pod", pod), ("node", node)]),
+                          np.asarray(vals, np.float64)))
+    return out
+
+
+def test_prepare_series_by_without():
+    from victoriametrics_amd import aggregate as agg
+    groups = agg.prepare_series(_series_set(), "by", ["node"])
+    # the all-NaN series is dropped first; two node groups remain
+    assert len(groups) == 2
+    sizes = sorted(len(m) for _, m in groups)
+    assert sizes == [1, 2]
+    gmn = groups[0][0]
+    assert gmn.tags == [(b"node", b"n1")] and gmn.metric_group == b""
+    groups_w = agg.prepare_series(_series_set(), "without", ["pod"])
+    assert len(groups_w) == 2
+    assert groups_w[0][0].get_tag_value("node") == b"n1"
+
+
+@pytest.mark.gpu
+def test_aggregate_sum_by_node():
+    from victoriametrics_amd import aggregate as agg
+    out = agg.aggregate("sum", _series_set(), "by", ["node"])
+    got = {s.mn.get_tag_value("node"): list(s.values) for s in out}
+    assert got[b"n1"] == [4.0, 6.0]
+    assert got[b"n2"][0] == 5.0 and math.isnan(got[b"n2"][1])
+
+
+@pytest.mark.gpu
+def test_aggregate_median_and_quantile():
+    from victoriametrics_amd import aggregate as agg
+    out = agg.aggregate("median", _series_set(), "by", ["node"])
+    got = {s.mn.get_tag_value("node"): list(s.values) for s in out}
+    assert got[b"n1"] == [2.0, 3.0]
+    out2 = agg.aggregate("quantile", _series_set(), "by", ["node"], arg=1.0)
+    got2 = {s.mn.get_tag_value("node"): s.values for s in out2}
+    assert got2[b"n1"][0] == 3.0
+
+
+@pytest.mark.gpu
+def test_aggregate_share_keeps_names():
+    from victoriametrics_amd import aggregate as agg
+    out = agg.aggregate("share", _series_set(), "by", ["node"])
+    pods = sorted(s.mn.get_tag_value("pod") for s in out)
+    assert pods == [b"a", b"b", b"c"]
+    got = {s.mn.get_tag_value("pod"): list(s.values) for s in out}
+    assert got[b"a"] == [0.25, 1 / 3]
+
+
+@pytest.mark.gpu
+def test_aggregate_outliers():
+    from victoriametrics_amd import aggregate as agg
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+    series = [Series(MetricName("m", [("i", str(i))]),
+                     np.full(4, float(i))) for i in range(9)]
+    series.append(Series(MetricName("m", [("i", "out")]),
+                         np.full(4, 1000.0)))
+    out = agg.aggregate("outliers_iqr", series)
+    assert [s.mn.get_tag_value("i") for s in out] == [b"out"]
+
+
+def test_aggregate_any_limitk():
+    from victoriametrics_amd import aggregate as agg
+    out = agg.aggregate("any", _series_set(), "by", ["node"])
+    assert len(out) == 2
+    assert all(s.mn.get_tag_value("pod") is None for s in out)
+    out2 = agg.aggregate("limitk", _series_set(), "by", ["node"], arg=1)
+    assert len(out2) == 2
+    assert all(s.mn.get_tag_value("pod") is not None for s in out2)

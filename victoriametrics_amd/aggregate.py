@@ -1,0 +1,161 @@
+"""Series-level aggregate dispatch — host mirror of aggr.go's aggrFuncExt
+(:110) + removeGroupTags (:96) over Series lists, with the per-point math
+on the GPU:
+
+- simple reducers (sum/min/max/avg/count/sum2/geomean/group) and the
+  column statistics (median/quantile/mad/mode/distinct/stddev/stdvar) run
+  through vmgpu_colagg;
+- share/zscore rewrite member series in place (per-series outputs);
+- outliers_iqr / outliers_mad / outliersk filter member series via the
+  bounds + filter kernels;
+- topk/bottomk families route to the existing selection kernels
+  (engine.topk_range / topk_pointwise);
+- any / limitk are metadata selections (host).
+
+This layer exists for aggregation over RESIDENT result series (e.g. after
+binary ops or transforms).  The rollup->aggregate hot path keeps using the
+fused incremental aggregation inside the rollup kernels (SeriesBatch with
+group_ids) — this module is the general aggr.go surface, not a replacement
+for that path.
+
+The reference's per-group member order is Go map/slice order (unspecified);
+this mirror aggregates members in input order, which both the oracle and
+the device kernels follow exactly.
+"""
+import math
+
+import numpy as np
+
+from . import engine
+from .binary_op import Series, remove_empty_series
+
+REDUCERS = {"sum", "min", "max", "avg", "count", "sum2", "geomean", "group",
+            "median", "mad", "mode", "distinct", "stddev", "stdvar"}
+PER_SERIES = {"share", "zscore"}
+
+
+def remove_group_tags(mn, modifier_op, modifier_args):
+    """removeGroupTags (aggr.go:96)."""
+    op = (modifier_op or "").lower()
+    if op in ("", "by"):
+        mn.remove_tags_on(modifier_args)
+    elif op == "without":
+        mn.remove_tags_ignoring(modifier_args)
+        mn.reset_metric_group()
+    else:
+        raise ValueError(f"unknown group modifier {modifier_op!r}")
+
+
+def prepare_series(series, modifier_op="", modifier_args=(), max_series=0,
+                   keep_original=False):
+    """aggrPrepareSeries (aggr.go:121): drop empty series, group by the
+    modifier key.  Returns list of (group_mn, [member Series])."""
+    series = remove_empty_series(series)
+    m = {}
+    order = []
+    for s in series:
+        mn = s.mn.copy()
+        remove_group_tags(mn, modifier_op, modifier_args)
+        k = mn.marshal_sorted()
+        if k not in m:
+            if max_series > 0 and len(m) >= max_series:
+                continue
+            m[k] = (mn, [])
+            order.append(k)
+        m[k][1].append(s)
+    return [m[k] for k in order]
+
+
+def _matrix(groups):
+    rows, group_rows, goff = [], [], [0]
+    for _, members in groups:
+        for s in members:
+            group_rows.append(len(rows))
+            rows.append(s.values)
+        goff.append(len(group_rows))
+    return (np.stack(rows), np.asarray(group_rows, np.uint32),
+            np.asarray(goff, np.uint64))
+
+
+def aggregate(name, series, modifier_op="", modifier_args=(), limit=0,
+              arg=None):
+    """aggrFuncs dispatch (aggr.go:40) over resident Series."""
+    name = name.lower()
+    if name == "quantile":
+        groups = prepare_series(series, modifier_op, modifier_args, limit)
+        return _reduce("quantile", groups, phi=float(arg))
+    if name in REDUCERS:
+        groups = prepare_series(series, modifier_op, modifier_args, limit)
+        return _reduce(name, groups)
+    if name in PER_SERIES:
+        groups = prepare_series(series, modifier_op, modifier_args, limit,
+                                keep_original=True)
+        if not groups:
+            return []
+        v, gr, go = _matrix(groups)
+        out = engine.colagg(name, v, gr, go)
+        rvs = []
+        i = 0
+        for _, members in groups:
+            for s in members:
+                s.values = out[i]
+                i += 1
+            rvs.extend(members)
+        return rvs
+    if name == "any":
+        # aggrFuncAny: one series per group (deterministic: first member)
+        groups = prepare_series(series, modifier_op, modifier_args, limit)
+        rvs = []
+        for gmn, members in groups:
+            dst = members[0]
+            dst.mn = gmn
+            rvs.append(dst)
+        return rvs
+    if name == "limitk":
+        # aggrFuncLimitK: first k member series per group, original names
+        groups = prepare_series(series, modifier_op, modifier_args, 0,
+                                keep_original=True)
+        k = int(arg)
+        rvs = []
+        for _, members in groups:
+            rvs.extend(members[:k])
+        return rvs
+    if name in ("outliers_iqr", "outliers_mad"):
+        groups = prepare_series(series, modifier_op, modifier_args, limit,
+                                keep_original=True)
+        if not groups:
+            return []
+        v, gr, go = _matrix(groups)
+        group_of = np.empty(v.shape[0], np.int32)
+        for gi in range(len(groups)):
+            group_of[int(go[gi]):int(go[gi + 1])] = gi
+        if name == "outliers_iqr":
+            lower, upper = engine.colagg("iqr_bounds", v, gr, go)
+            flags = engine.colagg_filter("iqr", v, group_of, lower, upper)
+        else:
+            med = engine.colagg("median", v, gr, go)
+            mad = engine.colagg("mad", v, gr, go)
+            tol = float(arg)
+            flags = engine.colagg_filter("mad", v, group_of, med, mad * tol)
+        rvs = []
+        i = 0
+        for _, members in groups:
+            for s in members:
+                if flags[i]:
+                    rvs.append(s)
+                i += 1
+        return rvs
+    raise ValueError(f"unsupported aggregate {name!r} "
+                     "(topk/bottomk: engine.topk_*; count_values/histogram: "
+                     "host metadata layer)")
+
+
+def _reduce(op, groups, phi=0.0):
+    if not groups:
+        return []
+    v, gr, go = _matrix(groups)
+    out = engine.colagg(op, v, gr, go, phi=phi)
+    rvs = []
+    for gi, (gmn, members) in enumerate(groups):
+        rvs.append(Series(gmn, out[gi]))
+    return rvs

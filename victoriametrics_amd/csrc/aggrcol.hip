@@ -219,6 +219,43 @@ __global__ void colagg_kernel(int32_t op, const double* values,
         }
         break;
       }
+      case VMGPU_COLAGG_SUM:
+      case VMGPU_COLAGG_MIN:
+      case VMGPU_COLAGG_MAX:
+      case VMGPU_COLAGG_AVG:
+      case VMGPU_COLAGG_COUNT:
+      case VMGPU_COLAGG_SUM2:
+      case VMGPU_COLAGG_GEOMEAN:
+      case VMGPU_COLAGG_GROUP: {
+        /* aggrFuncSum/Min/Max/Avg/Count/Sum2/Geomean/Group (aggr.go):
+         * left-to-right member order (the reference's own order is Go map
+         * iteration, i.e. unspecified) */
+        double sum = 0, prod = 1, mn = c_nan(), mx = c_nan();
+        double cnt = 0;
+        for (uint32_t k = lo; k < hi; k++) {
+          double v = values[(size_t)group_rows[k] * n_grid + g];
+          if (isnan(v)) continue;
+          cnt++;
+          sum += (op == VMGPU_COLAGG_SUM2) ? v * v : v;
+          prod *= v;
+          if (isnan(mn) || v < mn) mn = v;
+          if (isnan(mx) || v > mx) mx = v;
+        }
+        double r;
+        switch (op) {
+          case VMGPU_COLAGG_SUM:
+          case VMGPU_COLAGG_SUM2:   r = (cnt == 0) ? c_nan() : sum; break;
+          case VMGPU_COLAGG_MIN:    r = mn; break;
+          case VMGPU_COLAGG_MAX:    r = mx; break;
+          case VMGPU_COLAGG_AVG:    r = (cnt == 0) ? c_nan() : sum / cnt; break;
+          case VMGPU_COLAGG_COUNT:  r = (cnt == 0) ? c_nan() : cnt; break;
+          case VMGPU_COLAGG_GROUP:  r = (cnt == 0) ? c_nan() : 1.0; break;
+          default:  /* geomean */
+            r = (cnt == 0) ? c_nan() : pow(prod, 1.0 / cnt); break;
+        }
+        out[e] = r;
+        break;
+      }
       case VMGPU_COLAGG_IQR_BOUNDS: {
         /* getPerPointIQRBounds (aggr.go:975): q25/q75 +/- 1.5*iqr */
         int cnt = c_sorted_col(values, n_grid, group_rows, lo, hi, g, sc);

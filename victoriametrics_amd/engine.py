@@ -885,6 +885,8 @@ def rollup_subquery(outer_func, start, end, step, window, sq_step,
 COLAGG_OPS = {
     "median": 0, "quantile": 1, "mad": 2, "stddev": 3, "stdvar": 4,
     "mode": 5, "distinct": 6, "share": 7, "zscore": 8, "iqr_bounds": 9,
+    "sum": 10, "min": 11, "max": 12, "avg": 13, "count": 14, "sum2": 15,
+    "geomean": 16, "group": 17,
 }
 
 
