@@ -295,3 +295,24 @@ def test_aggregate_any_limitk():
     out2 = agg.aggregate("limitk", _series_set(), "by", ["node"], arg=1)
     assert len(out2) == 2
     assert all(s.mn.get_tag_value("pod") is not None for s in out2)
+
+
+def test_count_values():
+    from victoriametrics_amd import aggregate as agg
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+    series = [
+        Series(MetricName("m", [("pod", "a")]), np.asarray([1.0, 2.0, NAN])),
+        Series(MetricName("m", [("pod", "b")]), np.asarray([1.0, 1.0, 2.0])),
+    ]
+    out = agg.count_values("v", series)
+    got = {s.mn.get_tag_value("v"): list(s.values) for s in out}
+    assert got[b"1"][0] == 2.0 and got[b"1"][1] == 1.0
+    assert math.isnan(got[b"1"][2])
+    assert math.isnan(got[b"2"][0]) and got[b"2"][1] == 1.0
+    assert got[b"2"][2] == 1.0
+    # Go shortest-format labels
+    out2 = agg.count_values("v", [Series(MetricName("m"),
+                                         np.asarray([0.5, 0.25]))])
+    labels = sorted(s.mn.get_tag_value("v") for s in out2)
+    assert labels == [b"0.25", b"0.5"]

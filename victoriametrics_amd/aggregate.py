@@ -159,3 +159,53 @@ def _reduce(op, groups, phi=0.0):
     for gi, (gmn, members) in enumerate(groups):
         rvs.append(Series(gmn, out[gi]))
     return rvs
+
+
+def format_go_float(v):
+    """strconv.FormatFloat(v, 'f', -1, 64): shortest 'f'-format string that
+    round-trips."""
+    if v != v or math.isinf(v):
+        return "NaN" if v != v else ("+Inf" if v > 0 else "-Inf")
+    s = np.format_float_positional(v, unique=True, trim="-")
+    return s
+
+
+def count_values(dst_label, series, modifier_op="", modifier_args=(),
+                 limit=0, max_series_per_aggr=1000):
+    """aggrFuncCountValues (aggr.go:566): per distinct value, a series with
+    dst_label=value counting occurrences per point.  dst_label is removed
+    from the grouping as in Prometheus."""
+    modifier_args = list(modifier_args)
+    op = (modifier_op or "").lower()
+    if op == "without":
+        modifier_args = modifier_args + [dst_label]
+    elif op == "by":
+        modifier_args = [a for a in modifier_args if a != dst_label]
+    groups = prepare_series(series, modifier_op, modifier_args, limit)
+    rvs = []
+    for gmn, members in groups:
+        m = {}
+        order = []
+        for s in members:
+            for i, v in enumerate(s.values):
+                if math.isnan(v):
+                    continue
+                key = v
+                dst = m.get(key)
+                if dst is None:
+                    if len(m) >= max_series_per_aggr:
+                        raise ValueError(
+                            f"more than {max_series_per_aggr} series "
+                            "generated by count_values()")
+                    mn = members[0].mn.copy()
+                    mn.remove_tag(dst_label)
+                    mn.add_tag(dst_label, format_go_float(v))
+                    dst = Series(mn, np.full(len(s.values), math.nan))
+                    m[key] = dst
+                    order.append(key)
+                if math.isnan(dst.values[i]):
+                    dst.values[i] = 1.0
+                else:
+                    dst.values[i] += 1.0
+        rvs.extend(m[k] for k in order)
+    return rvs
