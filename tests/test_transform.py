@@ -355,3 +355,49 @@ def test_any_representative_group_ids():
     from victoriametrics_amd.engine import any_representative_group_ids
     gids = any_representative_group_ids([2, 0, 2, 1, 0, -1, 1])
     assert list(gids) == [2, 0, -1, 1, -1, -1, -1]
+
+
+def test_label_transform_unanchored():
+    s = S("m", [("path", "a.b.c")], [1])
+    tf.label_transform([s], "path", r"\.", "-")
+    assert s.mn.get_tag_value("path") == b"a-b-c"
+    # non-matching: untouched
+    s2 = S("m", [("path", "abc")], [1])
+    tf.label_transform([s2], "path", r"\d+", "X")
+    assert s2.mn.get_tag_value("path") == b"abc"
+
+
+def test_label_value():
+    a = S("m", [("v", "2.5")], [1.0, NAN])
+    b = S("m", [("v", "junk")], [3.0])
+    tf.label_value([a, b], "v")
+    assert a.mn.metric_group == b""
+    assert a.values[0] == 2.5 and math.isnan(a.values[1])
+    assert math.isnan(b.values[0])
+
+
+def test_labels_equal():
+    a = S("m", [("x", "1"), ("y", "1")], [1])
+    b = S("m", [("x", "1"), ("y", "2")], [1])
+    assert tf.labels_equal([a, b], ["x", "y"]) == [a]
+
+
+def test_label_graphite_group():
+    s = S("a.b.c.d", [], [1])
+    tf.label_graphite_group([s], [0, 2])
+    assert s.mn.metric_group == b"a.c"
+    s2 = S("a.b", [], [1])
+    tf.label_graphite_group([s2], [5, 1])
+    assert s2.mn.metric_group == b".b"
+
+
+def test_numeric_less_and_sort():
+    assert tf.numeric_less("2", "10")
+    assert not tf.numeric_less("10", "2")
+    assert tf.numeric_less("v2", "v10")
+    assert tf.numeric_less("abc", "abd")
+    assert tf.numeric_less("1.5x", "1.5y")
+    a = S("m", [("n", "pod10")], [1])
+    b = S("m", [("n", "pod2")], [1])
+    assert tf.sort_by_label_numeric([a, b], ["n"]) == [b, a]
+    assert tf.sort_by_label_numeric([a, b], ["n"], desc=True) == [a, b]

@@ -636,3 +636,142 @@ def buckets_limit(limit, series_list):
             del le_group[min_idx]
         rvs.extend(x[2] for x in le_group)
     return rvs
+
+
+def _go_replace_all(regex, replacement, s):
+    # Go regexp.ReplaceAll semantics with $1/${name} expansion
+    pat = re.compile(regex)
+    repl = re.sub(r"\$(\d+|\{\w+\})",
+                  lambda g: "\\" + g.group(1).strip("{}"), replacement)
+    return pat.sub(repl, s)
+
+
+def label_transform(series_list, label, regex, replacement):
+    """transformLabelTransform (transform.go:2066): UNANCHORED ReplaceAll
+    on the label (label_replace is the anchored Prometheus form)."""
+    pat = re.compile(regex)
+    for s in series_list:
+        v = (s.mn.get_tag_value(label) or b"").decode("utf-8",
+                                                      "surrogateescape")
+        if not pat.search(v):
+            continue
+        out = _go_replace_all(regex, replacement, v)
+        if label == "__name__":
+            s.mn.metric_group = out.encode()
+        elif out == "":
+            s.mn.remove_tag(label)
+        else:
+            s.mn.set_tag(label, out)
+    return series_list
+
+
+def label_value(series_list, label):
+    """transformLabelValue (transform.go:2175): replace values with the
+    parsed float of the label (NaN when unparsable); metric group reset;
+    all-NaN series kept (for `default`)."""
+    for s in series_list:
+        s.mn.reset_metric_group()
+        raw = s.mn.get_tag_value(label)
+        try:
+            v = float((raw or b"").decode())
+        except ValueError:
+            v = math.nan
+        mask = ~np.isnan(s.values)
+        s.values[mask] = v
+    return series_list
+
+
+def labels_equal(series_list, labels):
+    """transformLabelsEqual (transform.go:2138): keep series whose named
+    labels all carry the same value."""
+    out = []
+    for s in series_list:
+        if len(labels) < 2:
+            out.append(s)
+            continue
+        first = s.mn.get_tag_value(labels[0])
+        if all(s.mn.get_tag_value(l) == first for l in labels[1:]):
+            out.append(s)
+    return out
+
+
+def label_graphite_group(series_list, group_ids):
+    """transformLabelGraphiteGroup (transform.go:2260): rebuild the metric
+    group from dot-separated components by index."""
+    for s in series_list:
+        groups = s.mn.metric_group.split(b".")
+        parts = []
+        for gid in group_ids:
+            parts.append(groups[gid] if 0 <= gid < len(groups) else b"")
+        s.mn.metric_group = b".".join(parts)
+    return series_list
+
+
+def _num_prefix(s):
+    # getNumPrefix (transform.go:2510)
+    i = 0
+    if s and s[0] in "+-":
+        i += 1
+    has_num = has_dot = False
+    while i < len(s):
+        c = s[i]
+        if not c.isdigit():
+            if not has_dot and c == ".":
+                has_dot = True
+                i += 1
+                continue
+            return s[:i] if has_num else ""
+        has_num = True
+        i += 1
+    return s if has_num else ""
+
+
+def _nonnum_prefix(s):
+    for i, c in enumerate(s):
+        if c.isdigit():
+            return s[:i]
+    return s
+
+
+def numeric_less(a, b):
+    """numericLess (transform.go:2486): mixed numeric/string segments."""
+    while True:
+        if not b:
+            return False
+        if not a:
+            return True
+        ap, bp = _num_prefix(a), _num_prefix(b)
+        a, b = a[len(ap):], b[len(bp):]
+        if ap or bp:
+            if not ap:
+                return False
+            if not bp:
+                return True
+            an, bn = float(ap), float(bp)
+            if an != bn:
+                return an < bn
+        ap, bp = _nonnum_prefix(a), _nonnum_prefix(b)
+        a, b = a[len(ap):], b[len(bp):]
+        if ap != bp:
+            return ap < bp
+
+
+def sort_by_label_numeric(series_list, labels, desc=False):
+    """sort_by_label_numeric[_desc] (newTransformFuncNumericSort): stable
+    multi-key sort with numericLess per label."""
+    import functools
+
+    def cmp(x, y):
+        for lb in labels:
+            a = (x.mn.get_tag_value(lb) or b"").decode("utf-8",
+                                                       "surrogateescape")
+            bb = (y.mn.get_tag_value(lb) or b"").decode("utf-8",
+                                                        "surrogateescape")
+            if a == bb:
+                continue
+            if numeric_less(a, bb):
+                return 1 if desc else -1
+            return -1 if desc else 1
+        return 0
+
+    return sorted(series_list, key=functools.cmp_to_key(cmp))
