@@ -316,3 +316,24 @@ def test_count_values():
                                          np.asarray([0.5, 0.25]))])
     labels = sorted(s.mn.get_tag_value("v") for s in out2)
     assert labels == [b"0.25", b"0.5"]
+
+
+def test_histogram_aggregate():
+    from victoriametrics_amd import aggregate as agg
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+    series = [Series(MetricName("m", [("pod", p)]),
+                     np.asarray([v], np.float64))
+              for p, v in [("a", 1.0), ("b", 1.0), ("c", 5.0),
+                           ("d", -3.0), ("e", NAN)]]
+    out = agg.histogram_aggregate(series)
+    # cumulative le buckets; last bucket (le=+Inf) counts all 3 valid values
+    les = [(float(s.mn.get_tag_value("le")), s.values[0]) for s in out]
+    les.sort()
+    assert les[-1][0] == math.inf and les[-1][1] == 3.0
+    # exact 10^0 value lands in the bucket ENDING at 1.000e+00
+    one_buckets = [le for le, _ in les if abs(le - 1.0) < 1e-9]
+    assert one_buckets, les
+    # monotone cumulative counts
+    counts = [c for _, c in les]
+    assert counts == sorted(counts)

@@ -209,3 +209,86 @@ def count_values(dst_label, series, modifier_op="", modifier_args=(),
                     dst.values[i] += 1.0
         rvs.extend(m[k] for k in order)
     return rvs
+
+
+# ---------------------------------------------------------------------------
+# aggrFuncHistogram (aggr.go:316) + the VictoriaMetrics histogram buckets
+# (vendor/github.com/VictoriaMetrics/metrics/histogram.go: e10 range
+# [-9,18], 18 buckets/decimal, vmrange = "%.3e...%.3e" with bounds built
+# by repeated multiplication)
+# ---------------------------------------------------------------------------
+
+_H_E10_MIN, _H_E10_MAX, _H_BPD = -9, 18, 18
+_H_BUCKETS = (_H_E10_MAX - _H_E10_MIN) * _H_BPD
+_H_MULT = math.pow(10, 1.0 / _H_BPD)
+_bucket_ranges = None
+
+
+def _histogram_ranges():
+    global _bucket_ranges
+    if _bucket_ranges is None:
+        v = math.pow(10, _H_E10_MIN)
+        start = "%.3e" % v
+        rs = []
+        for _ in range(_H_BUCKETS):
+            v *= _H_MULT
+            end = "%.3e" % v
+            rs.append(start + "..." + end)
+            start = end
+        _bucket_ranges = rs
+    return _bucket_ranges
+
+
+_H_LOWER = "0...%.3e" % math.pow(10, _H_E10_MIN)
+_H_UPPER = "%.3e...+Inf" % math.pow(10, _H_E10_MAX)
+
+
+def _histogram_update(buckets, v):
+    # Histogram.Update (histogram.go:88): NaN and negatives skipped;
+    # exact 10^n values drop to the lower bucket (Prometheus le logic)
+    if math.isnan(v) or v < 0:
+        return
+    bucket_idx = (math.log10(v) - _H_E10_MIN) * _H_BPD if v > 0 else -1.0
+    if bucket_idx < 0:
+        buckets["lower"] = buckets.get("lower", 0) + 1
+    elif bucket_idx >= _H_BUCKETS:
+        buckets["upper"] = buckets.get("upper", 0) + 1
+    else:
+        idx = int(bucket_idx)
+        if bucket_idx == float(idx) and idx > 0:
+            idx -= 1
+        buckets[idx] = buckets.get(idx, 0) + 1
+
+
+def histogram_aggregate(series, modifier_op="", modifier_args=(), limit=0):
+    """aggrFuncHistogram: per grid point, a VictoriaMetrics histogram over
+    member values; non-zero buckets become vmrange series, converted to
+    cumulative le buckets (vmrangeBucketsToLE)."""
+    from .transform import vmrange_buckets_to_le
+    ranges = _histogram_ranges()
+    groups = prepare_series(series, modifier_op, modifier_args, limit)
+    rvs = []
+    for gmn, members in groups:
+        n_grid = len(members[0].values)
+        m = {}
+        for i in range(n_grid):
+            buckets = {}
+            for s in members:
+                _histogram_update(buckets, float(s.values[i]))
+            for key, count in buckets.items():
+                if key == "lower":
+                    vmrange = _H_LOWER
+                elif key == "upper":
+                    vmrange = _H_UPPER
+                else:
+                    vmrange = ranges[key]
+                ts = m.get(vmrange)
+                if ts is None:
+                    mn = members[0].mn.copy()
+                    mn.remove_tag("vmrange")
+                    mn.add_tag("vmrange", vmrange)
+                    ts = Series(mn, np.zeros(n_grid))
+                    m[vmrange] = ts
+                ts.values[i] = float(count)
+        rvs.extend(m.values())
+    return vmrange_buckets_to_le(rvs)
