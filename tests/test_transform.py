@@ -401,3 +401,13 @@ def test_numeric_less_and_sort():
     b = S("m", [("n", "pod2")], [1])
     assert tf.sort_by_label_numeric([a, b], ["n"]) == [b, a]
     assert tf.sort_by_label_numeric([a, b], ["n"], desc=True) == [a, b]
+
+
+def test_timezone_offset():
+    # 2021-03-14 is the US DST transition day: 06:00 UTC is EST (-5h),
+    # 15:09 UTC is EDT (-4h)
+    ts = [1615698000000, 1615734566000]
+    out = tf.timezone_offset("America/New_York", ts)
+    assert list(out) == [-5 * 3600.0, -4 * 3600.0]
+    utc = tf.timezone_offset("UTC", ts)
+    assert list(utc) == [0.0, 0.0]

@@ -775,3 +775,19 @@ def sort_by_label_numeric(series_list, labels, desc=False):
         return 0
 
     return sorted(series_list, key=functools.cmp_to_key(cmp))
+
+
+def timezone_offset(tz_name, grid_timestamps_ms):
+    """transformTimezoneOffset (transform.go:2769): UTC offset in seconds
+    of the named IANA timezone at each grid timestamp."""
+    from zoneinfo import ZoneInfo
+    from datetime import datetime
+    try:
+        tz = ZoneInfo(tz_name)
+    except Exception as e:
+        raise ValueError(f"cannot load timezone {tz_name!r}: {e}")
+    out = np.empty(len(grid_timestamps_ms))
+    for i, t in enumerate(grid_timestamps_ms):
+        dt = datetime.fromtimestamp(int(t) // 1000, tz)
+        out[i] = dt.utcoffset().total_seconds()
+    return out
