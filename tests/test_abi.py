@@ -110,3 +110,49 @@ def test_func_id_tables_match_oracle():
     assert engine.REMOVE_COUNTER_RESETS_FUNCS == oracle.REMOVE_COUNTER_RESETS_FUNCS
     assert engine.SAMPLES_SCANNED_PER_CALL == oracle.SAMPLES_SCANNED_PER_CALL
     assert engine.CAN_ADJUST_WINDOW_FUNCS == oracle.CAN_ADJUST_WINDOW_FUNCS
+
+
+def test_tracer_and_limits():
+    from victoriametrics_amd.tracer import Tracer, new_child, donef, done
+    from victoriametrics_amd import limits
+    import json as _json
+    import pytest as _pytest
+    # nil-safe: disabled tracer is None and helpers swallow everything
+    qt = Tracer.new(False, "q %s", "x")
+    assert qt is None
+    assert new_child(qt, "child") is None
+    donef(qt, "done")
+    done(qt)
+    qt = Tracer.new(True, "promql %s", "sum(rate(m[5m]))")
+    c = qt.new_child("rollup over %d series", 4)
+    c.donef("kernel %.3f ms", 1.25)
+    qt.done()
+    tree = _json.loads(qt.to_json())
+    assert "promql" in tree["message"]
+    assert tree["children"][0]["message"].endswith("kernel 1.250 ms")
+    assert "duration_msec" in tree
+    # limits
+    limits.validate_max_points(100)
+    with _pytest.raises(limits.QueryLimitError):
+        limits.validate_max_points(limits.max_points_per_timeseries + 1)
+    old = limits.max_memory_per_query
+    limits.max_memory_per_query = 1000
+    try:
+        with _pytest.raises(limits.QueryLimitError):
+            limits.check_rollup_memory(1000, 1000)
+        limits.check_rollup_memory(2, 2)
+    finally:
+        limits.max_memory_per_query = old
+    d = limits.Deadline(1e-9)
+    import time as _t
+    _t.sleep(0.01)
+    with _pytest.raises(limits.QueryLimitError):
+        d.check("test stage")
+
+
+def test_rollup_plan_respects_max_points():
+    from victoriametrics_amd import limits
+    from victoriametrics_amd.engine import RollupPlan, VmGpuError
+    import pytest as _pytest
+    with _pytest.raises(limits.QueryLimitError):
+        RollupPlan("rate", 0, limits.max_points_per_timeseries * 2_000, 1_000)

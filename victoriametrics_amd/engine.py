@@ -189,6 +189,8 @@ class RollupPlan:
             raise VmGpuError(f"step must be positive; got {step}")
         if start > end:
             raise VmGpuError(f"start {start} exceeds end {end}")
+        from . import limits as _limits
+        _limits.validate_max_points(1 + (int(end) - int(start)) // int(step))
         self.func = func
         self.start = int(start)
         self.end = int(end)
@@ -363,10 +365,15 @@ class SeriesBatch:
     def __exit__(self, *exc):
         self.close()
 
-    def exec(self, plan, download=True):
+    def exec(self, plan, download=True, tracer=None, deadline=None):
         """Evaluate plan over this batch.  Returns (out, counts,
         samples_scanned); out is [n_series x n_grid] or [n_groups x n_grid].
-        With download=False the results stay on the device (use fetch_out)."""
+        With download=False the results stay on the device (use fetch_out).
+
+        tracer: optional victoriametrics_amd.tracer.Tracer span — a child
+        span records the kernel wall time (hipEvents), mirroring the
+        reference's per-stage querytracer children.  deadline: optional
+        limits.Deadline checked before the launch."""
         lib = _load_lib()
         n_grid = plan.n_grid
         grouped = plan.aggr != "none"
@@ -381,6 +388,12 @@ class SeriesBatch:
             if grouped:
                 counts = np.empty((rows, n_grid), dtype=np.float64)
                 cptr = counts.ctypes.data_as(ctypes.POINTER(ctypes.c_double))
+        from . import limits as _limits
+        if deadline is not None:
+            deadline.check("rollup evaluation")
+        _limits.check_rollup_memory(
+            self.n_series, n_grid,
+            grouped_rows=self.n_groups if grouped else None)
         scanned = ctypes.c_uint64(0)
         errbuf = ctypes.create_string_buffer(256)
         rc = lib.vmgpu_rollup_exec(ctypes.byref(plan._c),
@@ -390,6 +403,13 @@ class SeriesBatch:
         if rc != 0:
             raise VmGpuError(f"vmgpu_rollup_exec failed ({rc}): "
                              f"{errbuf.value.decode()}")
+        if tracer is not None:
+            c = tracer.new_child(
+                "vmgpu rollup %s over %d series -> %d rows x %d points",
+                plan.func, self.n_series,
+                self.n_groups if grouped else self.n_series, n_grid)
+            c.donef("kernel %.3f ms, %d samples scanned",
+                    last_kernel_ms(), scanned.value)
         self._last_rows = self.n_groups if grouped else self.n_series
         self._last_n_grid = n_grid
         if out is not None and not grouped and \
