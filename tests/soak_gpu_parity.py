@@ -24,7 +24,8 @@ from seriesgen import ragged_batch, assert_parity  # noqa: E402
 from test_gpu_parity import _oracle_batch  # noqa: E402
 
 START = 1_000_000_000_000
-POW_FUNCS = {"geomean_over_time"}
+POW_FUNCS = {"geomean_over_time", "hoeffding_bound_lower",
+             "hoeffding_bound_upper"}  # pow/log: device-libm ulp
 ARG_RANGES = {
     "quantile_over_time": (0.0, 1.0), "predict_linear": (-600.0, 600.0),
     "share_le_over_time": (0.0, 100.0), "share_gt_over_time": (0.0, 100.0),
