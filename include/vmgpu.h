@@ -48,6 +48,11 @@ enum {
 /* Evaluation plan — the fields of promql.rollupConfig (rollup.go:574-606)
  * that shape the computation, plus the preFunc toggles that getRollupConfigs
  * (rollup.go:374-516) would install and the aggregate op. */
+#define VMGPU_PRE_NONE 0
+#define VMGPU_PRE_DELTA 1            /* deltaValues (rollup.go:963) */
+#define VMGPU_PRE_DERIV 2            /* derivValues (rollup.go:979) */
+#define VMGPU_PRE_SCRAPE_INTERVAL 3  /* rollup_scrape_interval preFunc */
+
 typedef struct vmgpu_plan {
   int32_t func;                   /* vmgpu_func_id */
   int32_t aggr;                   /* VMGPU_AGGR_* (NONE => per-series output) */
@@ -67,6 +72,11 @@ typedef struct vmgpu_plan {
   int32_t skip_finalize;          /* leave (values,counts) un-finalized so the
                                    * caller can all-reduce across shards first
                                    *  */
+  int32_t pre_func;               /* VMGPU_PRE_*: per-series value transform
+                                   * applied after removeCounterResets, for
+                                   * the rollup_rate/rollup_delta/
+                                   * rollup_scrape_interval families
+                                   * (getRollupConfigs, rollup.go:436-516) */
   double  arg;                    /* phi / le / gt / eq / secs for arg funcs */
   double  arg2;                   /* second scalar (holt_winters tf) */
 } vmgpu_plan;

@@ -291,7 +291,7 @@ def rollup_do(rc, values, timestamps):
 def rollup_eval_batch(rc, ts, vals, offsets, group_ids=None, n_groups=0,
                       aggr="none", remove_counter_resets=False,
                       max_staleness_interval=0, drop_stale_nans=False,
-                      n_threads=1):
+                      n_threads=1, pre_func=0):
     """Batch CSR evaluation — mirrors the product C-ABI shape."""
     t = _i64(ts)
     v = _f64(vals)
@@ -315,6 +315,7 @@ def rollup_eval_batch(rc, ts, vals, offsets, group_ids=None, n_groups=0,
         ctypes.byref(rc), ctypes.c_int32(1 if remove_counter_resets else 0),
         ctypes.c_int64(max_staleness_interval),
         ctypes.c_int32(1 if drop_stale_nans else 0),
+        ctypes.c_int32(int(pre_func)),
         _ptr(t, ctypes.c_int64), _ptr(v, ctypes.c_double),
         _ptr(off, ctypes.c_uint64), ctypes.c_uint32(n_series),
         gptr, ctypes.c_uint32(n_groups), ctypes.c_int32(aggr_id),

@@ -180,6 +180,7 @@ int vm_rollup_eval_batch(const vm_rollup_config* rc,
                          int32_t remove_counter_resets,
                          int64_t max_staleness_interval,
                          int32_t drop_stale_nans,
+                         int32_t pre_func,
                          const int64_t* ts, const double* vals,
                          const uint64_t* offsets, uint32_t n_series,
                          const int32_t* group_ids, uint32_t n_groups, int32_t aggr_op,
@@ -225,7 +226,7 @@ int vm_rollup_eval_batch(const vm_rollup_config* rc,
       int64_t n = (int64_t)(hi - lo);
       const double* v = vals + lo;
       const int64_t* t = ts + lo;
-      if (remove_counter_resets || drop_stale_nans) {
+      if (remove_counter_resets || drop_stale_nans || pre_func) {
         if ((size_t)n > scratch_cap) {
           scratch_cap = (size_t)(n > 64 ? n : 64);
           scratch_v = (double*)realloc(scratch_v, scratch_cap * sizeof(double));
@@ -236,6 +237,20 @@ int vm_rollup_eval_batch(const vm_rollup_config* rc,
         if (drop_stale_nans) n = vm_drop_stale_nans(scratch_v, scratch_t, n);
         if (remove_counter_resets)
           vm_remove_counter_resets(scratch_v, scratch_t, n, max_staleness_interval);
+        /* preFunc transforms for the rollup_* pseudo-functions
+         * (getRollupConfigs, rollup.go:436-516) */
+        if (pre_func == 1) vm_delta_values(scratch_v, n);
+        else if (pre_func == 2) vm_deriv_values(scratch_v, scratch_t, n);
+        else if (pre_func == 3 && n > 0) {
+          /* rollup_scrape_interval preFunc (rollup.go:476-494) */
+          double prev_secs = nan("");
+          for (int64_t i = 0; i < n; i++) {
+            double secs = (double)scratch_t[i] / 1000.0;
+            scratch_v[i] = secs - prev_secs;
+            prev_secs = secs;
+          }
+          if (n > 1) scratch_v[0] = scratch_v[1];
+        }
         v = scratch_v;
         t = scratch_t;
       }

@@ -193,6 +193,7 @@ int vm_rollup_eval_batch(const vm_rollup_config* rc,
                          int32_t remove_counter_resets,
                          int64_t max_staleness_interval,
                          int32_t drop_stale_nans,
+                         int32_t pre_func,
                          const int64_t* ts, const double* vals,
                          const uint64_t* offsets, uint32_t n_series,
                          const int32_t* group_ids, uint32_t n_groups, int32_t aggr_op,

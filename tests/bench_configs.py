@@ -68,6 +68,15 @@ def config4():
     n_grid = plan.n_grid
     print(f"rate+sum by(le): kernel {rollup_ms:.2f} ms, wall "
           f"{rollup_wall * 1e3:.0f} ms, out {n_hist * n_le}x{n_grid}")
+    # isolation: ungrouped and coarse-group variants of the same rollup
+    plain = SeriesBatch(ts, vals, offsets)
+    p2 = RollupPlan("rate", start, end, step, window=300_000)
+    for _ in range(2):
+        plain.exec(p2, download=False)
+    plain.exec(p2, download=False)
+    print(f"rate ungrouped (block-class): kernel "
+          f"{engine.last_kernel_ms():.2f} ms")
+    plain.close()
     # hq over le groups: les ascending per histogram
     les = np.tile(np.geomspace(0.001, 10.0, n_le), n_hist)
     goff = np.arange(n_hist + 1, dtype=np.uint64) * n_le
@@ -105,8 +114,9 @@ def config5():
     t0 = time.time()
     res = engine.topk_range(batch, 100, summary="avg")
     topk_wall = time.time() - t0
+    sel = res[0] if isinstance(res, tuple) else res
     print(f"topk(100, avg): wall {topk_wall * 1e3:.0f} ms, "
-          f"{len(res['ids'])} selected")
+          f"{len(sel)} selected")
 
 
 if __name__ == "__main__":

@@ -73,7 +73,7 @@ def _oracle_batch(engine_plan, ts, vals, offsets, group_ids=None, n_groups=0,
         remove_counter_resets=bool(engine_plan._c.remove_counter_resets),
         max_staleness_interval=engine_plan._c.max_staleness_interval,
         drop_stale_nans=bool(engine_plan._c.drop_stale_nans),
-        n_threads=n_threads)
+        n_threads=n_threads, pre_func=engine_plan._c.pre_func)
 
 
 def test_rate_counter_batch_bitexact(engine, counter_small):
@@ -462,3 +462,38 @@ def test_relayout_grouped_matches_unrelayouted_semantics():
         exp = np.where(np.all(np.isnan(member), axis=0), np.nan,
                        np.nansum(member, axis=0))
         np.testing.assert_allclose(out_g[g], exp, rtol=1e-9, equal_nan=True)
+
+
+@pytest.mark.parametrize("parent", ["rollup", "rollup_rate", "rollup_deriv",
+                                    "rollup_increase", "rollup_delta",
+                                    "rollup_scrape_interval",
+                                    "rollup_candlestick"])
+def test_rollup_fake_expansions(engine, parent):
+    """getRollupConfigs' rollup_* pseudo-function expansions
+    (rollup.go:436-516): preFunc value transform + min/max/avg (or
+    candlestick) sub-configs, each compared to the oracle with the same
+    preFunc."""
+    ts, vals, offsets = ragged_batch(40, 300, START, dup_p=0.03)
+    start = START + 120_000
+    end = start + 30 * STEP
+    plans = engine.rollup_fake_plans(parent, start, end, STEP,
+                                     window=300_000)
+    assert len(plans) == (4 if parent == "rollup_candlestick" else 3)
+    for tag, plan in plans:
+        out, _, scanned = engine.rollup_eval(plan, ts, vals, offsets)
+        ref, _, ref_scanned = _oracle_batch(plan, ts, vals, offsets)
+        assert scanned == ref_scanned, f"{parent}/{tag} samplesScanned"
+        assert_parity(out, ref, exact=True, context=f"{parent}/{tag}")
+
+
+def test_aggr_over_time_expansion(engine):
+    ts, vals, offsets = ragged_batch(24, 200, START)
+    start = START + 120_000
+    end = start + 20 * STEP
+    plans = engine.aggr_over_time_plans(
+        ["min_over_time", "max_over_time", "sum_over_time",
+         "count_over_time"], start, end, STEP, window=300_000)
+    for tag, plan in plans:
+        out, _, _ = engine.rollup_eval(plan, ts, vals, offsets)
+        ref, _, _ = _oracle_batch(plan, ts, vals, offsets)
+        assert_parity(out, ref, exact=True, context=f"aggr_over_time/{tag}")

@@ -168,6 +168,7 @@ struct KPlan {
   int32_t rcr;
   int32_t drop_stale;
   int32_t chunk_wave;      /* LDS samples per wave (wave kernel), 64-aligned */
+  int32_t pre_func;        /* VMGPU_PRE_* value transform after rcr */
   int32_t jbuf_elems;      /* LDS boundary-cache elements for the rate paths */
   int32_t jbuf_mode;       /* 0 = none; 1 = u16 j-cache (shared-boundary);
                               2 = sample-scatter Et/Ev/J boundary map */
@@ -410,6 +411,104 @@ static __device__ void rcr_scan_wave(int64_t* d_ts, double* d_vals, int count,
     prev_ts = __shfl(t, last);
     prev_fin = __shfl(x, last);
   }
+}
+
+/* --- preFunc value transforms (rollup_* families) -------------------- */
+/* deltaValues (rollup.go:960), derivValues (:976) and the
+ * rollup_scrape_interval preFunc (:476), applied in place after
+ * removeCounterResets exactly as getRollupConfigs composes them.  Fast
+ * paths are wave-elementwise; derivValues' duplicate-timestamp carry is
+ * replayed serially on lane 0 when duplicates exist (rare — dedup
+ * upstream normally removes them). */
+static VM_DEV void pre_func_wave(const int64_t* d_ts, double* d_vals,
+                                 int count, int mode, int lane) {
+  if (count <= 0 || mode == VMGPU_PRE_NONE) return;
+  if (mode == VMGPU_PRE_SCRAPE_INTERVAL) {
+    /* values[i] = ts[i]/1e3 - ts[i-1]/1e3 (NaN seed); values[0]=values[1] */
+    for (int base = 0; base < count; base += WAVE) {
+      int k = base + lane;
+      if (k < count) {
+        double cur = (double)d_ts[k] / 1000.0;
+        double prev = (k > 0) ? (double)d_ts[k - 1] / 1000.0 : vm_dnan();
+        d_vals[k] = cur - prev;
+      }
+      wave_lds_sync();
+    }
+    if (lane == 0 && count > 1) d_vals[0] = d_vals[1];
+    wave_lds_sync();
+    return;
+  }
+  if (mode == VMGPU_PRE_DELTA) {
+    double last = 0.0;
+    if (count >= 2) last = d_vals[count - 1] - d_vals[count - 2];
+    for (int base = 0; base + 1 < count; base += WAVE) {
+      int k = base + lane;
+      double a = 0, b = 0;
+      bool act = (k + 1 < count);
+      if (act) {
+        a = d_vals[k];
+        b = d_vals[k + 1];
+      }
+      wave_lds_sync();      /* all reads land before any write */
+      if (act) d_vals[k] = b - a;
+      wave_lds_sync();
+    }
+    if (lane == 0) d_vals[count - 1] = last;
+    wave_lds_sync();
+    return;
+  }
+  /* VMGPU_PRE_DERIV */
+  bool dup = false;
+  for (int base = 0; base + 1 < count; base += WAVE) {
+    int k = base + lane;
+    bool d = (k + 1 < count) && (d_ts[k + 1] == d_ts[k]);
+    if (__ballot(d) != 0) dup = true;
+  }
+  if (!dup) {
+    double last = 0.0;
+    if (count >= 2)
+      last = (d_vals[count - 1] - d_vals[count - 2]) /
+             ((double)(d_ts[count - 1] - d_ts[count - 2]) / 1e3);
+    for (int base = 0; base + 1 < count; base += WAVE) {
+      int k = base + lane;
+      double a = 0, b = 0;
+      int64_t ta = 0, tb = 0;
+      bool act = (k + 1 < count);
+      if (act) {
+        a = d_vals[k];
+        b = d_vals[k + 1];
+        ta = d_ts[k];
+        tb = d_ts[k + 1];
+      }
+      wave_lds_sync();
+      if (act) d_vals[k] = (b - a) / ((double)(tb - ta) / 1e3);
+      wave_lds_sync();
+    }
+    if (lane == 0) d_vals[count - 1] = last;
+    wave_lds_sync();
+    return;
+  }
+  if (lane == 0) {
+    /* exact serial replay of derivValues with the duplicate-ts carry */
+    double prev_deriv = 0.0;
+    double prev_value = d_vals[0];
+    int64_t prev_ts = d_ts[0];
+    for (int i = 0; i + 1 < count; i++) {
+      double v = d_vals[i + 1];
+      int64_t t = d_ts[i + 1];
+      if (t == prev_ts) {
+        d_vals[i] = prev_deriv;
+        continue;
+      }
+      double dt = (double)(t - prev_ts) / 1e3;
+      prev_deriv = (v - prev_value) / dt;
+      d_vals[i] = prev_deriv;
+      prev_value = v;
+      prev_ts = t;
+    }
+    d_vals[count - 1] = prev_deriv;
+  }
+  wave_lds_sync();
 }
 
 /* --- per-series window parameters (doInternal preamble) -------------- */
@@ -712,6 +811,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
     }
     wave_lds_sync();
 #endif
+    if (p.pre_func) pre_func_wave(lts, lvs, count, p.pre_func, lane);
 
     int64_t si = p.step;
 #ifndef VMGPU_ABL_NO_SCRAPE
@@ -949,6 +1049,9 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KI
     int count = *sh_count;
     if (p.rcr && wave == 0) rcr_scan_wave(lts, lvs, count, p.max_staleness, lane);
     __syncthreads();
+    if (p.pre_func && wave == 0)
+      pre_func_wave(lts, lvs, count, p.pre_func, lane);
+    __syncthreads();
     if (wave == 0) {
       int64_t si = p.step;
       if (p.start < p.end) si = scrape_interval_wave(lts, count, p.step, lane, lscratch);
@@ -995,11 +1098,12 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_huge_kernel(KPlan p, KIO
 
     const int64_t* uts;
     const double* uvs;
-    if (p.drop_stale || p.rcr) {
+    if (p.drop_stale || p.rcr || p.pre_func) {
       if (wave == 0) {
         int c = load_compact_wave(io.ts + lo, io.vals + lo, n, dts, dvs,
                                   p.drop_stale != 0, lane);
         if (p.rcr) rcr_scan_wave(dts, dvs, c, p.max_staleness, lane);
+        if (p.pre_func) pre_func_wave(dts, dvs, c, p.pre_func, lane);
         if (lane == 0) sh_count = c;
       }
       __syncthreads();
@@ -2127,6 +2231,7 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
   p.is_default = plan->is_default_rollup;
   p.rcr = plan->remove_counter_resets;
   p.drop_stale = plan->drop_stale_nans;
+  p.pre_func = plan->pre_func;
   p.chunk_wave = (int32_t)std::min<uint32_t>(
       CHUNK_WAVE, std::max<uint32_t>(64, (b.max_wave_len + 63) & ~63u));
   /* rate boundary caches (see the wave kernel): the u16 j-cache

@@ -106,6 +106,51 @@ SAMPLES_SCANNED_PER_CALL = {
 # funcs whose stale NaNs must be kept (dropStaleNaNs, eval.go:2108-2115)
 KEEP_STALE_NANS_FUNCS = {"default_rollup", "stale_samples_over_time"}
 
+# preFunc value transforms for the rollup_* pseudo-functions
+# (getRollupConfigs, rollup.go:436-516)
+PRE_FUNC_IDS = {"none": 0, "delta": 1, "deriv": 2, "scrape_interval": 3}
+
+# rollup_* pseudo-function -> (pre_func, candlestick?) expansion
+ROLLUP_FAKE_FUNCS = {
+    "rollup": ("none", False),
+    "rollup_rate": ("deriv", False),
+    "rollup_deriv": ("deriv", False),
+    "rollup_increase": ("delta", False),
+    "rollup_delta": ("delta", False),
+    "rollup_scrape_interval": ("scrape_interval", False),
+    "rollup_candlestick": ("none", True),
+}
+
+
+def rollup_fake_plans(parent, start, end, step, tag="", **kwargs):
+    """getRollupConfigs' expansion of the rollup_* pseudo-functions
+    (rollup.go:436-516): returns [(rollup_tag, RollupPlan)].  tag narrows
+    to one sub-config (the optional second arg of rollup*(q, "tag"))."""
+    if parent not in ROLLUP_FAKE_FUNCS:
+        raise VmGpuError(f"not a rollup_* pseudo-function: {parent!r}")
+    pre, candle = ROLLUP_FAKE_FUNCS[parent]
+    if candle:
+        subs = {"open": "rollup_open", "close": "rollup_close",
+                "low": "rollup_low", "high": "rollup_high"}
+    else:
+        subs = {"min": "min_over_time", "max": "max_over_time",
+                "avg": "avg_over_time"}
+    names = [tag] if tag else list(subs)
+    out = []
+    for t in names:
+        if t not in subs:
+            raise VmGpuError(f"unexpected rollup tag {t!r} for {parent}")
+        out.append((t, RollupPlan(subs[t], start, end, step, pre_func=pre,
+                                  parent_func=parent, **kwargs)))
+    return out
+
+
+def aggr_over_time_plans(func_names, start, end, step, **kwargs):
+    """aggr_over_time(q, "fn1", ...) expansion (rollup.go:496-509): one
+    plan per listed rollup function, tagged with its name."""
+    return [(name, RollupPlan(name, start, end, step, parent_func=name,
+                              **kwargs)) for name in func_names]
+
 
 class VmGpuError(RuntimeError):
     pass
@@ -128,6 +173,7 @@ class _PlanC(ctypes.Structure):
         ("drop_stale_nans", ctypes.c_int32),
         ("samples_scanned_per_call", ctypes.c_int32),
         ("skip_finalize", ctypes.c_int32),
+        ("pre_func", ctypes.c_int32),
         ("arg", ctypes.c_double),
         ("arg2", ctypes.c_double),
     ]
@@ -182,7 +228,8 @@ class RollupPlan:
 
     def __init__(self, func, start, end, step, window=0, lookback_delta=0,
                  min_staleness_interval=0, arg=0.0, arg2=0.0, aggr="none",
-                 skip_finalize=False, keep_stale_nans=False):
+                 skip_finalize=False, keep_stale_nans=False, pre_func="none",
+                 parent_func=None):
         if func not in FUNC_IDS:
             raise VmGpuError(f"unsupported rollup function {func!r}")
         if step <= 0:
@@ -204,7 +251,12 @@ class RollupPlan:
         self.skip_finalize = skip_finalize
         self.keep_stale_nans = keep_stale_nans
 
-        rcr = func in REMOVE_COUNTER_RESETS_FUNCS
+        # rollup_* expansions (getRollupConfigs, rollup.go:436-516): flags
+        # come from the PARENT pseudo-function, the evaluated func is the
+        # min/max/avg (or candlestick) sub-config
+        flags_func = parent_func or func
+        self.pre_func = pre_func
+        rcr = flags_func in REMOVE_COUNTER_RESETS_FUNCS
         # stalenessInterval = lookbackDelta (+window when set), rollup.go:380-387
         staleness = self.lookback_delta
         if staleness != 0:
@@ -217,12 +269,13 @@ class RollupPlan:
             lookback_delta=self.lookback_delta,
             min_staleness_interval=self.min_staleness_interval,
             max_staleness_interval=staleness if rcr else 0,
-            may_adjust_window=1 if func in CAN_ADJUST_WINDOW_FUNCS else 0,
+            may_adjust_window=1 if flags_func in CAN_ADJUST_WINDOW_FUNCS else 0,
             is_default_rollup=1 if func == "default_rollup" else 0,
             remove_counter_resets=1 if rcr else 0,
             drop_stale_nans=0 if (keep_stale_nans or func in KEEP_STALE_NANS_FUNCS) else 1,
-            samples_scanned_per_call=SAMPLES_SCANNED_PER_CALL.get(func, 0),
+            samples_scanned_per_call=SAMPLES_SCANNED_PER_CALL.get(flags_func, 0),
             skip_finalize=1 if skip_finalize else 0,
+            pre_func=PRE_FUNC_IDS[pre_func],
             arg=self.arg, arg2=self.arg2)
 
     @property
