@@ -168,6 +168,8 @@ struct KPlan {
   int32_t rcr;
   int32_t drop_stale;
   int32_t chunk_wave;      /* LDS samples per wave (wave kernel), 64-aligned */
+  int32_t chunk_block;     /* LDS samples for the block kernel, sized to the
+                              batch's longest block-class series */
   int32_t pre_func;        /* VMGPU_PRE_* value transform after rcr */
   int32_t jbuf_elems;      /* LDS boundary-cache elements for the rate paths */
   int32_t jbuf_mode;       /* 0 = none; 1 = u16 j-cache (shared-boundary);
@@ -175,6 +177,21 @@ struct KPlan {
   double arg;
   double arg2;
 };
+
+/* Block-kernel LDS: 32B header + chunk_block*16 series data + 256B scratch
+ * + (when it fits the 64 KiB dynamic limit and a rate j-cache is wanted)
+ * n_grid u16 window-end indices.  Host sizing and kernel carve must agree. */
+static __host__ __device__ inline size_t vm_block_lds_bytes(
+    int32_t chunk_block, int32_t jbuf_mode, int32_t n_grid, int* has_jbuf) {
+  size_t base = 32 + (size_t)chunk_block * 16 + 256;
+  size_t jb = ((size_t)n_grid * 2 + 15) & ~(size_t)15;
+  if (jbuf_mode >= 1 && base + jb <= 64 * 1024) {
+    if (has_jbuf) *has_jbuf = 1;
+    return base + jb;
+  }
+  if (has_jbuf) *has_jbuf = 0;
+  return base;
+}
 
 /* LDS bytes for the rate boundary cache (host sizing and kernel carve must
  * agree) */
@@ -1018,12 +1035,16 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KI
    * guide §6 G17: statics shift the 16-B-aligned dynamic base).
    * layout: [0,8) count (int) + si (packed), [16, 16+CB*8) ts, then vals. */
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int32_t CB = p.chunk_block;
   int* sh_count = (int*)smem;
   int64_t* sh_si = (int64_t*)(smem + 8);
   unsigned long long* sh_sum = (unsigned long long*)(smem + 16);
   int64_t* lts = (int64_t*)(smem + 32);
-  double* lvs = (double*)(smem + 32 + CHUNK_BLOCK * 8);
-  double* lscratch = (double*)(smem + 32 + (size_t)CHUNK_BLOCK * 16);
+  double* lvs = (double*)(smem + 32 + (size_t)CB * 8);
+  double* lscratch = (double*)(smem + 32 + (size_t)CB * 16);
+  int has_jbuf = 0;
+  (void)vm_block_lds_bytes(CB, p.jbuf_mode, p.n_grid, &has_jbuf);
+  uint16_t* jbuf = (uint16_t*)(smem + 32 + (size_t)CB * 16 + 256);
   const int tid = threadIdx.x;
   const int lane = tid % WAVE;
   const int wave = tid / WAVE;
@@ -1061,6 +1082,44 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KI
     SeriesWindow sw = series_window(p, *sh_si);
 
     if (tid == 0) scanned += (uint64_t)count;
+    if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
+      /* shared-boundary j-cache (see the wave kernel): one seek per point,
+       * i(g) = j(g - window/step) when the window is a step multiple */
+      int dg64 = (sw.window > 0 && p.step > 0 && sw.window % p.step == 0)
+                     ? (int)(sw.window / p.step) : 0;
+      if (has_jbuf && dg64 > 0 && count <= 65535) {
+        double idx_per_ms = 0.0;
+        int64_t ts0 = 0;
+        if (count > 1) {
+          ts0 = lts[0];
+          int64_t span_ms = lts[count - 1] - ts0;
+          idx_per_ms = span_ms > 0 ? (double)(count - 1) / (double)span_ms : 0.0;
+        }
+        for (int g = tid; g < p.n_grid; g += BLOCK_THREADS) {
+          int64_t t_end = p.start + (int64_t)g * p.step;
+          int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
+          jbuf[g] = (uint16_t)vm_ub_hint_fast(lts, count, t_end, gj);
+        }
+        __syncthreads();
+        for (int g = tid; g < p.n_grid; g += BLOCK_THREADS) {
+          int64_t t_end = p.start + (int64_t)g * p.step;
+          int64_t t_start = t_end - sw.window;
+          int j = jbuf[g];
+          int i;
+          if (g >= dg64) {
+            i = jbuf[g - dg64];
+          } else {
+            int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
+            i = vm_ub_hint_fast(lts, count, t_start, gi);
+          }
+          vm_emit_value(p, io, s, g,
+                        eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
+          scanned += 2;
+        }
+        __syncthreads();
+        continue;
+      }
+    }
     for (int g0 = 0; g0 < p.n_grid; g0 += BLOCK_THREADS) {
       int g = g0 + tid;
       if (g < p.n_grid) scanned += eval_grid_point<FUNC_CT>(p, sw, lts, lvs, count, g, s, io);
@@ -1736,6 +1795,7 @@ struct Batch {
   uint32_t* d_huge_list = nullptr;
   uint32_t n_wave = 0, n_block = 0, n_huge = 0;
   uint32_t max_wave_len = 0;
+  uint32_t max_block_len = 0;
   bool wave_is_identity = false; /* all series small: skip the list */
   uint64_t* d_huge_scr_offsets = nullptr;
   uint64_t huge_scratch_elems = 0;
@@ -1984,6 +2044,7 @@ int vmgpu_batch_create(const int64_t* ts, const double* vals,
       if ((uint32_t)n > b.max_wave_len) b.max_wave_len = (uint32_t)n;
     } else if (n <= CHUNK_BLOCK) {
       block_list.push_back(s);
+      if ((uint32_t)n > b.max_block_len) b.max_block_len = (uint32_t)n;
     } else {
       huge_list.push_back(s);
       huge_scr_off.push_back(huge_total);
@@ -2234,6 +2295,8 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
   p.pre_func = plan->pre_func;
   p.chunk_wave = (int32_t)std::min<uint32_t>(
       CHUNK_WAVE, std::max<uint32_t>(64, (b.max_wave_len + 63) & ~63u));
+  p.chunk_block = (int32_t)std::min<uint32_t>(
+      CHUNK_BLOCK, std::max<uint32_t>(64, (b.max_block_len + 63) & ~63u));
   /* rate boundary caches (see the wave kernel): the u16 j-cache
    * (shared-boundary path) is the measured default — both scatter
    * variants benched SLOWER at config 2 (full Et/Ev/J: 3.27 ms from LDS
@@ -2300,7 +2363,8 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
     w.series_sel = b.d_block_list;
     w.n_sel = b.n_block;
     uint32_t blocks = std::min<uint32_t>(b.n_block, 2048);
-    size_t lds = 32 + (size_t)CHUNK_BLOCK * 16 + 256;
+    size_t lds = vm_block_lds_bytes(p.chunk_block, p.jbuf_mode, p.n_grid,
+                                    nullptr);
     launch_rollup(1, blocks, lds, p, w);
   }
   if (b.n_huge) {
