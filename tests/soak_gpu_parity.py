@@ -68,9 +68,20 @@ def one_iter(seed):
     if aggr != "none":
         n_groups = int(rng.integers(1, max(2, n_series)))
         gids = rng.integers(-1, n_groups, n_series).astype(np.int32)
-    plan = RollupPlan(func, start, end, step, window=window,
-                      lookback_delta=lookback, arg=arg, arg2=arg2, aggr=aggr,
-                      skip_finalize=False)
+    # 1-in-8 iterations exercise the rollup_* preFunc expansions instead
+    if rng.integers(0, 8) == 0 and aggr == "none":
+        parent = ["rollup", "rollup_rate", "rollup_increase",
+                  "rollup_scrape_interval",
+                  "rollup_candlestick"][int(rng.integers(0, 5))]
+        plans = engine.rollup_fake_plans(parent, start, end, step,
+                                         window=window,
+                                         lookback_delta=lookback)
+        tag, plan = plans[int(rng.integers(0, len(plans)))]
+        func = f"{parent}/{tag}"
+    else:
+        plan = RollupPlan(func, start, end, step, window=window,
+                          lookback_delta=lookback, arg=arg, arg2=arg2,
+                          aggr=aggr, skip_finalize=False)
     out, counts, scanned = engine.rollup_eval(plan, ts, vals, offsets,
                                               group_ids=gids,
                                               n_groups=n_groups)
