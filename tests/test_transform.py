@@ -411,3 +411,20 @@ def test_timezone_offset():
     assert list(out) == [-5 * 3600.0, -4 * 3600.0]
     utc = tf.timezone_offset("UTC", ts)
     assert list(utc) == [0.0, 0.0]
+
+
+def test_eval_context_funcs():
+    ts = [1000, 2000, 3000]
+    assert list(tf.eval_time(ts)[0].values) == [1.0, 2.0, 3.0]
+    assert list(tf.eval_number(ts, 7)[0].values) == [7.0, 7.0, 7.0]
+    assert tf.eval_start(ts)[0].values[0] == 1.0
+    assert tf.eval_end(ts)[0].values[0] == 3.0
+    assert tf.eval_step(ts, 1000)[0].values[0] == 1.0
+    assert abs(tf.eval_pi(ts)[0].values[0] - math.pi) < 1e-15
+    s = S("m", [("a", "1")], [1, 2, 3])
+    assert tf.scalar([s]) == [s]
+    out = tf.scalar([s, s])
+    assert all(math.isnan(v) for v in out[0].values)
+    r1 = tf.rand_series(ts, seed=5)[0].values
+    r2 = tf.rand_series(ts, seed=5)[0].values
+    assert list(r1) == list(r2)

@@ -791,3 +791,71 @@ def timezone_offset(tz_name, grid_timestamps_ms):
         dt = datetime.fromtimestamp(int(t) // 1000, tz)
         out[i] = dt.utcoffset().total_seconds()
     return out
+
+
+# ---------------------------------------------------------------------------
+# evaluation-context funcs (transform.go: time/now/pi/start/end/step/vector/
+# scalar/rand*) — host generation, mirroring evalNumber/evalTime (eval.go)
+# ---------------------------------------------------------------------------
+
+def eval_number(grid_timestamps_ms, value):
+    """evalNumber: one scalar series over the grid."""
+    from .binary_op import Series
+    return [Series(MetricName(), np.full(len(grid_timestamps_ms),
+                                         float(value)))]
+
+
+def eval_time(grid_timestamps_ms):
+    """transformTime: grid timestamps in seconds."""
+    from .binary_op import Series
+    t = np.asarray(grid_timestamps_ms, dtype=np.float64) / 1e3
+    return [Series(MetricName(), t)]
+
+
+def eval_step(grid_timestamps_ms, step_ms):
+    return eval_number(grid_timestamps_ms, step_ms / 1e3)
+
+
+def eval_start(grid_timestamps_ms):
+    return eval_number(grid_timestamps_ms, grid_timestamps_ms[0] / 1e3)
+
+
+def eval_end(grid_timestamps_ms):
+    return eval_number(grid_timestamps_ms, grid_timestamps_ms[-1] / 1e3)
+
+
+def eval_pi(grid_timestamps_ms):
+    return eval_number(grid_timestamps_ms, math.pi)
+
+
+def scalar(series_list):
+    """transformScalar: a single series passes through; otherwise NaN
+    (Prometheus scalar() semantics)."""
+    from .binary_op import Series, is_scalar
+    if len(series_list) == 1:
+        return series_list
+    n = len(series_list[0].values) if series_list else 0
+    return [Series(MetricName(), np.full(n, math.nan))]
+
+
+def vector(series_list):
+    """transformVector: identity over instant vectors."""
+    return series_list
+
+
+def rand_series(grid_timestamps_ms, kind="uniform", seed=None):
+    """rand/rand_exponential/rand_normal (newTransformRand): a seeded
+    random series on the grid.  The reference uses Go math/rand; this
+    mirror uses numpy's PCG64 — the CONTRACT is 'deterministic for a given
+    seed', not cross-runtime bit-equality (the reference's own stream is
+    Go-runtime-specific)."""
+    from .binary_op import Series
+    rng = np.random.default_rng(None if seed is None else int(seed))
+    n = len(grid_timestamps_ms)
+    if kind == "uniform":
+        v = rng.random(n)
+    elif kind == "exponential":
+        v = rng.exponential(1.0, n)
+    else:
+        v = rng.standard_normal(n)
+    return [Series(MetricName(), v)]
