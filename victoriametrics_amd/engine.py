@@ -1028,3 +1028,12 @@ def colagg_filter(mode, values, group_of, b1, b2):
         raise VmGpuError(f"vmgpu_colagg_filter failed ({rc}): "
                          f"{errbuf.value.decode()}")
     return flags
+
+
+def quantiles_over_time_plans(phi_label, phis, start, end, step, **kwargs):
+    """quantiles_over_time(q, "phi_label", phi1, ...) expansion
+    (newRollupQuantiles, rollup.go): one quantile_over_time plan per phi,
+    tagged phi_label=%g."""
+    return [("%g" % phi,
+             RollupPlan("quantile_over_time", start, end, step,
+                        arg=float(phi), **kwargs)) for phi in phis]

@@ -859,3 +859,18 @@ def rand_series(grid_timestamps_ms, kind="uniform", seed=None):
     else:
         v = rng.standard_normal(n)
     return [Series(MetricName(), v)]
+
+
+def histogram_quantiles(phi_label, phis, series_list):
+    """transformHistogramQuantiles (transform.go): one histogram_quantile
+    result set per phi, labeled phi_label=phi (Go %g formatting)."""
+    out = []
+    for phi in phis:
+        part = histogram_transform("histogram_quantile",
+                                   [s.copy_shallow() for s in series_list],
+                                   arg=phi)
+        for s in part:
+            s.mn.remove_tag(phi_label)
+            s.mn.add_tag(phi_label, "%g" % phi)
+        out.extend(part)
+    return out
