@@ -41,8 +41,12 @@ def ragged_batch(n_series, max_samples, start, step=15_000, seed=1234,
             np.asarray(offsets, dtype=np.uint64))
 
 
-def assert_parity(got, ref, exact=True, rtol=1e-12, context=""):
-    """NaN==NaN; exact bit equality for arithmetic-only funcs, else rtol."""
+def assert_parity(got, ref, exact=True, rtol=1e-12, atol=0.0, context=""):
+    """NaN==NaN; exact bit equality for arithmetic-only funcs, else rtol
+    (+ optional atol floor: cross-series float sums of near-cancelling
+    columns — e.g. sum(zscore_over_time) ~ 0 — carry addition-order noise
+    at machine epsilon that no relative tolerance can bound; the reference
+    itself sums in nondeterministic worker order)."""
     assert got.shape == ref.shape, f"{context}: shape {got.shape} vs {ref.shape}"
     gn, rn = np.isnan(got), np.isnan(ref)
     assert (gn == rn).all(), \
@@ -54,7 +58,7 @@ def assert_parity(got, ref, exact=True, rtol=1e-12, context=""):
             f"{context}: {bad.sum()} exact mismatches; first diffs " \
             f"{g[bad][:3]} vs {r[bad][:3]}"
     else:
-        ok = np.isclose(g, r, rtol=rtol, atol=0)
+        ok = np.isclose(g, r, rtol=rtol, atol=atol)
         assert ok.all(), \
             f"{context}: {np.sum(~ok)} mismatches beyond rtol={rtol}; " \
             f"first {g[~ok][:3]} vs {r[~ok][:3]}"

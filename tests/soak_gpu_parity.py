@@ -94,7 +94,8 @@ def one_iter(seed):
     if aggr == "none":
         assert_parity(out, ref, exact=func not in POW_FUNCS, context=ctx)
     else:
-        assert_parity(out, ref, exact=False, rtol=1e-9, context=ctx)
+        assert_parity(out, ref, exact=False, rtol=1e-9, atol=1e-11,
+                      context=ctx)
 
 
 def main():
