@@ -1,0 +1,157 @@
+"""End-to-end pins transcribed from the reference's own TestExecSuccess
+(app/vmselect/promql/exec_test.go, fixed grid start=1000e3 end=2000e3
+step=200e3): each case hand-composes the same expression from this
+engine's pieces (eval-context series + transform/binop kernels) and
+compares against exec_test.go's expected float64 arrays — the reference's
+outputs, not the oracle's.  Exact for arithmetic-only expressions; 1-ulp
+tolerance for the transcendental ones (the expected arrays encode Go's
+math library, the engine computes with device libm)."""
+import math
+
+import numpy as np
+import pytest
+
+from victoriametrics_amd import transform as tfm
+from victoriametrics_amd import binary_op as bop
+from victoriametrics_amd.binary_op import BinOpSpec, Series
+from victoriametrics_amd.metric_name import MetricName
+
+pytestmark = pytest.mark.gpu
+
+GRID_MS = np.arange(1000_000, 2000_001, 200_000, dtype=np.int64)
+TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+
+
+def tmat():
+    return TIME.reshape(1, -1).copy()
+
+
+def _exact(got, expected):
+    np.testing.assert_array_equal(
+        np.asarray(got, np.float64).ravel().view(np.int64),
+        np.asarray(expected, np.float64).view(np.int64))
+
+
+def _ulp(got, expected):
+    np.testing.assert_allclose(np.asarray(got).ravel(), expected,
+                               rtol=1e-14, atol=0)
+
+
+def test_abs_time():
+    # abs(1500-time()) -> 500, 300, 100, 100, 300, 500
+    v = (1500.0 - tmat())
+    _exact(tfm.transform("abs", v), [500, 300, 100, 100, 300, 500])
+
+
+def test_ceil_floor():
+    _exact(tfm.transform("ceil", tmat() / 500), [2, 3, 3, 4, 4, 4])
+    _exact(tfm.transform("floor", tmat() / 500), [2, 2, 2, 3, 3, 4])
+
+
+def test_clamp_family():
+    lo = np.full(6, 1400.0)
+    hi = np.full(6, 1800.0)
+    _exact(tfm.transform("clamp", tmat(), args=[lo, hi]),
+           [1400, 1400, 1400, 1600, 1800, 1800])
+    _exact(tfm.transform("clamp_max", tmat(), args=[np.full(6, 1400.0)]),
+           [1000, 1200, 1400, 1400, 1400, 1400])
+    # clamp_min(1500, time()): scalar arg clamped from below by time()
+    _exact(tfm.transform("clamp_min", np.full((1, 6), 1500.0),
+                         args=[TIME]),
+           [1500, 1500, 1500, 1600, 1800, 2000])
+
+
+def test_bitmap_time():
+    mask = np.full(6, 17.0)  # 0x11
+    _exact(tfm.transform("bitmap_and", tmat(), args=[mask]),
+           [0, 16, 16, 0, 0, 16])
+    _exact(tfm.transform("bitmap_or", tmat(), args=[mask]),
+           [1017, 1201, 1401, 1617, 1817, 2001])
+    _exact(tfm.transform("bitmap_xor", tmat(), args=[mask]),
+           [1017, 1185, 1385, 1617, 1817, 1985])
+
+
+def test_round_nearest():
+    # round(time()/1e3, 0.5) -> 1, 1, 1.5, 1.5, 2, 2
+    _exact(tfm.transform("round", tmat() / 1e3, args=[np.full(6, 0.5)]),
+           [1, 1, 1.5, 1.5, 2, 2])
+
+
+def test_sgn():
+    _exact(tfm.transform("sgn", tmat() - 1400), [-1, -1, 0, 1, 1, 1])
+
+
+def test_transcendental():
+    _ulp(tfm.transform("exp", tmat() / 1e3),
+         [2.718281828459045, 3.3201169227365472, 4.0551999668446745,
+          4.953032424395115, 6.0496474644129465, 7.38905609893065])
+    _ulp(tfm.transform("sqrt", tmat()),
+         [31.622776601683793, 34.64101615137755, 37.416573867739416, 40,
+          42.42640687119285, 44.721359549995796])
+    _ulp(tfm.transform("ln", tmat()),
+         [6.907755278982137, 7.090076835776092, 7.24422751560335,
+          7.3777589082278725, 7.495541943884256, 7.600902459542082])
+    _ulp(tfm.transform("log2", tmat()),
+         [9.965784284662087, 10.228818690495881, 10.451211111832329,
+          10.643856189774725, 10.813781191217037, 10.965784284662087])
+    _ulp(tfm.transform("log10", tmat()),
+         [3, 3.0791812460476247, 3.1461280356782377, 3.2041199826559246,
+          3.255272505103306, 3.3010299956639813])
+
+
+def test_rad_deg_roundtrip():
+    v = tfm.transform("deg", tmat() / 500)
+    _ulp(tfm.transform("rad", v),
+         [2, 2.3999999999999995, 2.8, 3.2, 3.6, 4])
+
+
+def test_running_range_funcs():
+    _exact(tfm.transform("running_sum", np.ones((1, 6))),
+           [1, 2, 3, 4, 5, 6])
+    _exact(tfm.transform("range_sum", tmat()), [9000.0] * 6)
+    _exact(tfm.transform("range_max", tmat()), [2000.0] * 6)
+    _exact(tfm.transform("range_first", tmat()), [1000.0] * 6)
+    _exact(tfm.transform("range_quantile", tmat(), scalar=0.5),
+           [1500.0] * 6)
+    v = tfm.transform("abs", 1300.0 - tmat())
+    _exact(tfm.transform("running_max", v), [300, 300, 300, 300, 500, 700])
+
+
+def _masked_default_chain():
+    """time() < 1300 default time() > 1700 (the exec_test gap fixture)"""
+    t = Series(MetricName(), TIME.copy())
+    lt = bop.binary_op_eval(BinOpSpec("<"),
+                            [Series(MetricName(), TIME.copy())],
+                            [Series(MetricName(), np.full(6, 1300.0))])
+    gt = bop.binary_op_eval(BinOpSpec(">"),
+                            [Series(MetricName(), TIME.copy())],
+                            [Series(MetricName(), np.full(6, 1700.0))])
+    out = bop.binary_op_eval(BinOpSpec("default"), lt, gt)
+    assert len(out) == 1
+    return out[0].values.reshape(1, -1)
+
+
+def test_cmp_default_interpolate_keep():
+    chain = _masked_default_chain()
+    # fixture: 1000, 1200, NaN, NaN, 1800, 2000
+    assert not np.isnan(chain[0, 0]) and np.isnan(chain[0, 2])
+    _exact(tfm.transform("interpolate", chain.copy()),
+           [1000, 1200, 1400, 1600, 1800, 2000])
+    _exact(tfm.transform("keep_last_value", chain.copy()),
+           [1000, 1200, 1200, 1200, 1800, 2000])
+    _exact(tfm.transform("keep_next_value", chain.copy()),
+           [1000, 1200, 1800, 1800, 1800, 2000])
+
+
+def test_time_plus_time():
+    out = bop.binary_op_eval(BinOpSpec("+"),
+                             [Series(MetricName(), TIME.copy())],
+                             [Series(MetricName(), TIME.copy())])
+    _exact(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+
+
+def test_timezone_offset_newyork():
+    # exec_test computes the offset dynamically; the fixed grid (1970-01-01)
+    # falls in EST: -5h
+    out = tfm.timezone_offset("America/New_York", GRID_MS)
+    _exact(np.asarray(out), [-18000.0] * 6)
