@@ -931,73 +931,22 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
           jbuf[g] = (uint16_t)vm_ub_hint_fast(lts, count, t_end, gj);
         }
         wave_lds_sync();
-        /* The i-role of point g — (i, vals[i-1], ts[i-1]) — is exactly the
-         * j-role of point g-dg (i = j(g-dg), and vals[max(j-1,0)] matches
-         * vals[max(i-1,0)] for every j incl. 0), so when dg <= WAVE those
-         * values are in the registers of lane-dg of this or the previous
-         * 64-point block: shuffle instead of re-reading LDS. */
-        const bool shuf_i = (dg64 <= WAVE);
-        int jc = 0;
-        double vc = 0.0;
-        int64_t tc = 0;
         for (int g0 = 0; g0 < p.n_grid; g0 += 4 * WAVE) {
 #pragma unroll
           for (int u = 0; u < 4; u++) {
             int g = g0 + u * WAVE + lane;
-            int j = 0;
-            double v_end = 0.0;
-            int64_t t_end_s = 0;
-            if (g < p.n_grid) {
-              j = jbuf[g];
-              int jm1 = j - 1 < 0 ? 0 : j - 1;
-              v_end = lvs[jm1];
-              t_end_s = lts[jm1];
-            }
-            int i_s;
-            double iv;
-            int64_t it;
-            if (shuf_i) {
-              int cj = __shfl_up(j, dg64);
-              double cv = __shfl_up(v_end, dg64);
-              int64_t ct = __shfl_up(t_end_s, dg64);
-              int pj = __shfl_down(jc, WAVE - dg64);
-              double pv_c = __shfl_down(vc, WAVE - dg64);
-              int64_t pt_c = __shfl_down(tc, WAVE - dg64);
-              bool from_cur = (lane >= dg64);
-              i_s = from_cur ? cj : pj;
-              iv = from_cur ? cv : pv_c;
-              it = from_cur ? ct : pt_c;
-              jc = j;
-              vc = v_end;
-              tc = t_end_s;
-            }
             if (g < p.n_grid) {
               int64_t t_end = p.start + (int64_t)g * p.step;
               int64_t t_start = t_end - sw.window;
-              double v;
-              if (shuf_i && g >= dg64) {
-                /* inlined eval_rate_fused with the shuffled i-role */
-                int i = i_s;
-                bool has_prev = (i < count) && (i > 0) &&
-                                (it > t_start - sw.max_prev_interval);
-                int ii = i < count - 1 ? i : (count - 1 < 0 ? 0 : count - 1);
-                double pv = has_prev ? iv : lvs[ii];
-                int64_t ptm = has_prev ? it : lts[ii];
-                double slope = (v_end - pv) / ((double)(t_end_s - ptm) / 1e3);
-                int nn = j - i;
-                double res_prev = (nn == 0) ? 0.0 : slope;
-                double res_nop = (nn <= 1) ? vm_dnan() : slope;
-                v = has_prev ? res_prev : res_nop;
+              int j = jbuf[g];
+              int i;
+              if (g >= dg64) {
+                i = jbuf[g - dg64];
               } else {
-                int i;
-                if (g >= dg64) {
-                  i = jbuf[g - dg64];
-                } else {
-                  int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
-                  i = vm_ub_hint_fast(lts, count, t_start, gi);
-                }
-                v = eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start);
+                int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
+                i = vm_ub_hint_fast(lts, count, t_start, gi);
               }
+              double v = eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start);
               if (GACC && use_acc) {
                 if constexpr (GACC) {
                   if (!vm_isnan(v)) {
