@@ -115,3 +115,40 @@ def test_sharded_allreduce_merge(aggr, tmp_path):
     gn, rn = np.isnan(fin), np.isnan(ref)
     assert (gn == rn).all()
     assert np.allclose(fin[~gn], ref[~rn], rtol=1e-9, atol=0)
+
+
+def test_topk_shard_merge_matches_global():
+    """per-shard local top-k candidates + host merge == global
+    top-k over the union (candidate completeness argument)."""
+    import numpy as np
+    from victoriametrics_amd.engine import topk_merge_shards
+    rng = np.random.default_rng(3)
+    k = 7
+    n_per_shard = 40
+    # distinct summaries -> tie-free
+    sums = rng.permutation(2 * n_per_shard).astype(np.float64)
+    shard_sums = [sums[:n_per_shard], sums[n_per_shard:]]
+    shard_ids, shard_vals = [], []
+    for sv in shard_sums:
+        local_top = np.argsort(-sv)[:k]
+        shard_ids.append(local_top)
+        shard_vals.append(sv[local_top])
+    got = topk_merge_shards(shard_ids, shard_vals, k)
+    got_sums = sorted(shard_sums[s][i] for s, i in got)
+    exp = sorted(np.sort(sums)[-k:])
+    assert got_sums == [float(x) for x in exp]
+    # bottomk
+    shard_ids_b, shard_vals_b = [], []
+    for sv in shard_sums:
+        local_bot = np.argsort(sv)[:k]
+        shard_ids_b.append(local_bot)
+        shard_vals_b.append(sv[local_bot])
+    got_b = topk_merge_shards(shard_ids_b, shard_vals_b, k, reverse=True)
+    got_sums_b = sorted(shard_sums[s][i] for s, i in got_b)
+    exp_b = sorted(np.sort(sums)[:k])
+    assert got_sums_b == [float(x) for x in exp_b]
+    # NaN summaries sort last
+    got_n = topk_merge_shards([[0, 1], [0]],
+                              [np.asarray([float("nan"), 5.0]),
+                               np.asarray([7.0])], 2)
+    assert got_n == [(1, 0), (0, 1)]

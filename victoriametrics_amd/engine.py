@@ -1037,3 +1037,30 @@ def quantiles_over_time_plans(phi_label, phis, start, end, step, **kwargs):
     return [("%g" % phi,
              RollupPlan("quantile_over_time", start, end, step,
                         arg=float(phi), **kwargs)) for phi in phis]
+
+
+def topk_merge_shards(shard_ids, shard_summaries, k, reverse=False):
+    """Cross-shard topk merge (DESIGN §4 / SURVEY §5): each GPU selects its
+    local k candidates (vmgpu_topk_range) with their summary values; the
+    global top k is selected from the gathered candidates.  Matches the
+    single-batch selection whenever per-shard candidate sets are complete
+    (they are: a global top-k member is necessarily in its shard's local
+    top k).  Returns [(shard, local_id), ...] of the global selection,
+    ordered by summary (lessWithNaNs ordering: NaNs sort last for topk,
+    first for bottomk — aggr.go:1262)."""
+    cands = []
+    for shard, (ids, sums) in enumerate(zip(shard_ids, shard_summaries)):
+        for i, sv in zip(ids, sums):
+            cands.append((shard, int(i), float(sv)))
+
+    def sort_key(c):
+        v = c[2]
+        nan = v != v
+        # topk: biggest first, NaN last; bottomk: smallest first, NaN last
+        if reverse:
+            return (nan, v)
+        return (nan, -v)
+
+    cands.sort(key=sort_key)
+    kk = max(int(k), 0)
+    return [(s, i) for s, i, _ in cands[:kk]]
