@@ -714,7 +714,7 @@ static VM_DEV uint64_t eval_grid_point(const KPlan& p, const SeriesWindow& sw,
 /* kernel 1: one wave per series (n <= CHUNK_WAVE)                    */
 /* ------------------------------------------------------------------ */
 
-template <int FUNC_CT, bool GACC = false>
+template <int FUNC_CT, bool GACC = false, bool PREF = false>
 __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO io) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int wave_in_block = threadIdx.x / WAVE;
@@ -828,7 +828,12 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
     }
     wave_lds_sync();
 #endif
-    if (p.pre_func) pre_func_wave(lts, lvs, count, p.pre_func, lane);
+    /* compile-time gated: the inlined preFunc body costs ~10% on the hot
+     * rate path when merely PRESENT (VGPR 32 -> 44), so it lives in its
+     * own instantiation like GACC */
+    if constexpr (PREF) {
+      if (p.pre_func) pre_func_wave(lts, lvs, count, p.pre_func, lane);
+    }
 
     int64_t si = p.step;
 #ifndef VMGPU_ABL_NO_SCRAPE
@@ -1732,16 +1737,25 @@ static void launch_rollup_t(int which, uint32_t blocks, size_t lds,
   if (which == 0) {
     /* grouped rate plans with grids the accumulator covers take the
      * register-run variant (GACC) — valid because grouped batches are
-     * physically relayouted by group at creation */
+     * physically relayouted by group at creation.  preFunc plans take the
+     * PREF instantiation (compile-time gated for the same reason). */
     bool gacc = (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) &&
                 p.aggr != VMGPU_AGGR_NONE && w.group_ids != nullptr &&
                 p.n_grid <= 4 * WAVE;
-    if (gacc)
-      hipLaunchKernelGGL((rollup_wave_kernel<FUNC_CT, true>), dim3(blocks),
+    if (p.pre_func != 0) {
+      if (gacc)
+        hipLaunchKernelGGL((rollup_wave_kernel<FUNC_CT, true, true>),
+                           dim3(blocks), dim3(BLOCK_THREADS), lds, stream, p, w);
+      else
+        hipLaunchKernelGGL((rollup_wave_kernel<FUNC_CT, false, true>),
+                           dim3(blocks), dim3(BLOCK_THREADS), lds, stream, p, w);
+    } else if (gacc) {
+      hipLaunchKernelGGL((rollup_wave_kernel<FUNC_CT, true, false>), dim3(blocks),
                          dim3(BLOCK_THREADS), lds, stream, p, w);
-    else
-      hipLaunchKernelGGL((rollup_wave_kernel<FUNC_CT, false>), dim3(blocks),
+    } else {
+      hipLaunchKernelGGL((rollup_wave_kernel<FUNC_CT, false, false>), dim3(blocks),
                          dim3(BLOCK_THREADS), lds, stream, p, w);
+    }
   } else if (which == 1) {
     hipLaunchKernelGGL(rollup_block_kernel<FUNC_CT>, dim3(blocks),
                        dim3(BLOCK_THREADS), lds, stream, p, w);
