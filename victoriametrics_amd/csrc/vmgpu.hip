@@ -1742,6 +1742,9 @@ static void launch_rollup_t(int which, uint32_t blocks, size_t lds,
     bool gacc = (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) &&
                 p.aggr != VMGPU_AGGR_NONE && w.group_ids != nullptr &&
                 p.n_grid <= 4 * WAVE;
+#ifdef VMGPU_ABL_NO_GACC
+    gacc = false;
+#endif
     if (p.pre_func != 0) {
       if (gacc)
         hipLaunchKernelGGL((rollup_wave_kernel<FUNC_CT, true, true>),
