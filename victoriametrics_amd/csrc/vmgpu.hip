@@ -1739,11 +1739,17 @@ static void launch_rollup_t(int which, uint32_t blocks, size_t lds,
      * register-run variant (GACC) — valid because grouped batches are
      * physically relayouted by group at creation.  preFunc plans take the
      * PREF instantiation (compile-time gated for the same reason). */
-    bool gacc = (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) &&
-                p.aggr != VMGPU_AGGR_NONE && w.group_ids != nullptr &&
-                p.n_grid <= 4 * WAVE;
-#ifdef VMGPU_ABL_NO_GACC
-    gacc = false;
+    /* A/B measured (profiles/round1_final.md §4): the register-run
+     * accumulator (GACC) is 0.4 ms SLOWER than plain per-point atomics at
+     * config 3 (2.89 vs 3.30 ms) — the accumulate/flush instruction cost
+     * exceeds the scattered-atomic cost on this issue-bound kernel.  The
+     * instantiation stays compiled (VMGPU_ABL_GACC re-enables it for
+     * future A/Bs) but plain atomics are the default. */
+    bool gacc = false;
+#ifdef VMGPU_ABL_GACC
+    gacc = (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) &&
+           p.aggr != VMGPU_AGGR_NONE && w.group_ids != nullptr &&
+           p.n_grid <= 4 * WAVE;
 #endif
     if (p.pre_func != 0) {
       if (gacc)
