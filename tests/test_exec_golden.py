@@ -155,3 +155,65 @@ def test_timezone_offset_newyork():
     # falls in EST: -5h
     out = tfm.timezone_offset("America/New_York", GRID_MS)
     _exact(np.asarray(out), [-18000.0] * 6)
+
+
+# ---------------------------------------------------------------------------
+# vector-matching pins (exec_test.go:3561-3620): on/ignoring matching +
+# keep_metric_names through the real binop dispatch
+# ---------------------------------------------------------------------------
+
+def _ls(value, *tags, name=""):
+    """label_set(value, k, v, ...) fixture"""
+    mn = MetricName(name, list(zip(tags[::2], tags[1::2])))
+    v = TIME.copy() if value == "time" else np.full(6, float(value))
+    return Series(mn, v)
+
+
+def test_scalarish_mul_ignoring():
+    # label_set(2,"foo","bar") * ignoring(a) (label_set(time(),foo=bar) or
+    # label_set(10,foo=qwert)) -> {foo=bar} 2000..4000
+    left = [_ls(2, "foo", "bar")]
+    right = bop.binary_op_eval(BinOpSpec("or"),
+                               [_ls("time", "foo", "bar")],
+                               [_ls(10, "foo", "qwert")])
+    out = bop.binary_op_eval(BinOpSpec("*", group_op="ignoring",
+                                       group_tags=["a"]), left, right)
+    assert len(out) == 1
+    assert out[0].mn.tags == [(b"foo", b"bar")]
+    assert out[0].mn.metric_group == b""
+    _exact(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+
+
+def test_scalarish_mul_on_foo():
+    left = [_ls(2, "foo", "bar", "aa", "bb")]
+    right = bop.binary_op_eval(BinOpSpec("or"),
+                               [_ls("time", "foo", "bar", "xx", "yy")],
+                               [_ls(10, "foo", "qwert")])
+    out = bop.binary_op_eval(BinOpSpec("*", group_op="on",
+                                       group_tags=["foo"]), left, right)
+    assert len(out) == 1
+    assert out[0].mn.tags == [(b"foo", b"bar")]
+    _exact(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+
+
+def test_vector_mul_on_scalarish():
+    left = [_ls("time", "foo", "bar", "xx", "yy"), _ls(10, "foo", "qwert")]
+    right = [_ls(2, "foo", "bar", "aa", "bb")]
+    out = bop.binary_op_eval(BinOpSpec("*", group_op="on",
+                                       group_tags=["foo"]), left, right)
+    assert len(out) == 1
+    assert out[0].mn.tags == [(b"foo", b"bar")]
+    _exact(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+
+
+def test_vector_mul_on_keep_metric_names():
+    left = [_ls("time", "foo", "bar", "xx", "yy", name="q1"),
+            _ls(10, "foo", "qwert", name="q2")]
+    right = [_ls(2, "foo", "bar", "aa", "bb", name="q2")]
+    out = bop.binary_op_eval(BinOpSpec("*", group_op="on",
+                                       group_tags=["foo"],
+                                       keep_metric_names=True), left, right)
+    assert len(out) == 1
+    assert out[0].mn.metric_group == b"q1"
+    assert out[0].mn.tags == [(b"foo", b"bar")]
+    _exact(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
