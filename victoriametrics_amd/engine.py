@@ -52,6 +52,7 @@ FUNC_IDS = {
     "hoeffding_bound_upper": 69,
     # aliases to shared implementations, as in the reference map
     "increase_prometheus": 4, "timestamp": 31, "timestamp_with_name": 31,
+    "iqr_over_time": 62,  # rollupAggrFuncs alias (rollup.go:169)
 }
 
 AGGR_IDS = {
