@@ -92,6 +92,7 @@ FUNC_IDS = {
     "increase_prometheus": 4,   # rollupDeltaPrometheus
     "timestamp": 31,            # rollupTlast
     "timestamp_with_name": 31,  # rollupTlast
+    "iqr_over_time": 62,        # rollupAggrFuncs alias (rollup.go:169)
 }
 
 # rollupFuncsRemoveCounterResets (rollup.go:223-232)
