@@ -123,5 +123,27 @@ if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "config4"
     if which == "config4":
         config4()
+    elif which == "huge":
+        config_huge()
     else:
         config5()
+
+
+def config_huge():
+    """Huge-kernel class isolation: 50k series x 20k samples (3.5 days
+    @15s merged into one logical series) = 16 GB."""
+    n_series, n_samples, step = 50_000, 20_000, 15_000
+    print(f"huge-class: {n_series} series x {n_samples} samples")
+    ts, vals, offsets = synth_counters(n_series, n_samples, step)
+    batch = SeriesBatch(ts, vals, offsets)
+    start = START + 600_000
+    end = START + (n_samples - 1) * step
+    plan = RollupPlan("rate", start, end, step, window=300_000)
+    for _ in range(2):
+        batch.exec(plan, download=False)
+    batch.exec(plan, download=False)
+    k_ms = engine.last_kernel_ms()
+    samples = n_series * n_samples
+    print(f"rate huge-class: kernel {k_ms:.2f} ms -> "
+          f"{samples / (k_ms / 1e3) / 1e9:.1f} Gsamples/s")
+    batch.close()
