@@ -119,16 +119,6 @@ def config5():
           f"{len(sel)} selected")
 
 
-if __name__ == "__main__":
-    which = sys.argv[1] if len(sys.argv) > 1 else "config4"
-    if which == "config4":
-        config4()
-    elif which == "huge":
-        config_huge()
-    else:
-        config5()
-
-
 def config_huge():
     """Huge-kernel class isolation: 50k series x 20k samples (3.5 days
     @15s merged into one logical series) = 16 GB."""
@@ -147,3 +137,13 @@ def config_huge():
     print(f"rate huge-class: kernel {k_ms:.2f} ms -> "
           f"{samples / (k_ms / 1e3) / 1e9:.1f} Gsamples/s")
     batch.close()
+
+
+if __name__ == "__main__":
+    which = sys.argv[1] if len(sys.argv) > 1 else "config4"
+    if which == "config4":
+        config4()
+    elif which == "huge":
+        config_huge()
+    else:
+        config5()

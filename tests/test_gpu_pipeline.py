@@ -87,7 +87,7 @@ def test_full_query_pipeline():
     fleet_max = agg.aggregate("max", [s.copy_shallow() for s in hq])
     assert len(fleet_max) == 1
     ratio = bop.binary_op_eval(
-        BinOpSpec("/", group_op="on", group_tags=[]),
+        BinOpSpec("/", group_op="on", group_tags=[], join_op="group_left"),
         [s.copy_shallow() for s in hq], fleet_max)
     assert len(ratio) == n_hist
     finite = np.concatenate([r.values for r in ratio])
