@@ -1195,9 +1195,33 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_huge_kernel(KPlan p, KIO
     SeriesWindow sw = series_window(p, sh_si);
 
     if (tid == 0) scanned += (uint64_t)count;
+    /* density-interpolated seek hints (same scheme as the wave kernel):
+     * a full binary search over a 10^4-sample column costs ~15 dependent
+     * global loads per boundary; the hint + bounded walk resolves in ~1 */
+    double idx_per_ms = 0.0;
+    int64_t ts0 = 0;
+    if (count > 1) {
+      ts0 = uts[0];
+      int64_t span_ms = uts[count - 1] - ts0;
+      idx_per_ms = span_ms > 0 ? (double)(count - 1) / (double)span_ms : 0.0;
+    }
     for (int g0 = 0; g0 < p.n_grid; g0 += BLOCK_THREADS) {
       int g = g0 + tid;
-      if (g < p.n_grid) scanned += eval_grid_point<FUNC_CT>(p, sw, uts, uvs, count, g, s, io);
+      if (g < p.n_grid) {
+        int64_t t_end = p.start + (int64_t)g * p.step;
+        int64_t t_start = t_end - sw.window;
+        int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
+        int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
+        int i = vm_ub_hint(uts, count, t_start, gi);
+        int j = vm_ub_hint(uts, count, t_end, gj);
+        if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
+          vm_emit_value(p, io, s, g,
+                        eval_rate_fused(p, sw, uts, uvs, count, i, j, t_start));
+          scanned += 2;
+        } else {
+          scanned += eval_grid_point_ij<FUNC_CT>(p, sw, uts, uvs, count, g, s, io, i, j);
+        }
+      }
     }
     __syncthreads();
   }
