@@ -236,16 +236,26 @@ static __device__ int load_compact_wave(const int64_t* g_ts, const double* g_val
   if ((((uintptr_t)g_ts | (uintptr_t)g_vals) & 15) == 0) {
     const uint64_t lt_mask = (lane == 63) ? 0x7fffffffffffffffULL
                                           : ((1ULL << lane) - 1);
-    for (; src + 128 <= n; src += 128) {
-      double2 v = *(const double2*)(g_vals + src + 2 * lane);
-      vm_i64x2 t = *(const vm_i64x2*)(g_ts + src + 2 * lane);
-      bool k0 = !(drop_stale && vm_is_stale_nan(v.x));
-      bool k1 = !(drop_stale && vm_is_stale_nan(v.y));
+    while (src + 2 <= n) {
+      int64_t pairs = (n - src) >> 1;
+      if (pairs > 64) pairs = 64;
+      bool act = lane < pairs;
+      double2 v = {0.0, 0.0};
+      vm_i64x2 t = {0, 0};
+      if (act) {
+        v = *(const double2*)(g_vals + src + 2 * lane);
+        t = *(const vm_i64x2*)(g_ts + src + 2 * lane);
+      }
+      bool k0 = act && !(drop_stale && vm_is_stale_nan(v.x));
+      bool k1 = act && !(drop_stale && vm_is_stale_nan(v.y));
       uint64_t m0 = __ballot(k0), m1 = __ballot(k1);
-      if ((m0 & m1) == ~0ULL && (count & 1) == 0) {
-        *(double2*)(d_vals + count + 2 * lane) = v;
-        *(vm_i64x2*)(d_ts + count + 2 * lane) = t;
-        count += 128;
+      uint64_t full = (pairs == 64) ? ~0ULL : ((1ULL << pairs) - 1);
+      if ((m0 & m1) == full && (count & 1) == 0) {
+        if (act) {
+          *(double2*)(d_vals + count + 2 * lane) = v;
+          *(vm_i64x2*)(d_ts + count + 2 * lane) = t;
+        }
+        count += (int)(2 * pairs);
       } else {
         int below = __popcll(m0 & lt_mask) + __popcll(m1 & lt_mask);
         int dst0 = count + below;
@@ -253,6 +263,7 @@ static __device__ int load_compact_wave(const int64_t* g_ts, const double* g_val
         if (k1) { int d1 = dst0 + (k0 ? 1 : 0); d_ts[d1] = t.y; d_vals[d1] = v.y; }
         count += __popcll(m0) + __popcll(m1);
       }
+      src += 2 * pairs;
     }
   }
   for (; src < n; src += WAVE) {
