@@ -217,3 +217,87 @@ def test_vector_mul_on_keep_metric_names():
     assert out[0].mn.metric_group == b"q1"
     assert out[0].mn.tags == [(b"foo", b"bar")]
     _exact(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+
+
+# ---------------------------------------------------------------------------
+# aggregate pins (exec_test.go): the aggregate dispatch against the
+# reference's expected arrays
+# ---------------------------------------------------------------------------
+
+def _agg(name, series, **kw):
+    from victoriametrics_amd import aggregate as agg
+    return agg.aggregate(name, series, **kw)
+
+
+def test_median_union():
+    # median(union(label_set(10,foo=bar), label_set(time()/150,baz=sss),
+    #              time()/200))
+    series = [_ls(10, "foo", "bar"),
+              Series(MetricName("", [("baz", "sss")]), TIME / 150),
+              Series(MetricName(), TIME / 200)]
+    out = _agg("median", series)
+    assert len(out) == 1
+    _exact(out[0].values,
+           [6.666666666666667, 8, 9.333333333333334, 10, 10, 10])
+
+
+def test_stddev_avg_or():
+    left = [_ls(10, "foo", "bar")]
+    right = [Series(MetricName("", [("baz", "sss")]), TIME / 100)]
+    both = bop.binary_op_eval(BinOpSpec("or"),
+                              [s.copy_shallow() for s in left],
+                              [s.copy_shallow() for s in right])
+    out = _agg("stddev", [s.copy_shallow() for s in both])
+    _exact(out[0].values, [0, 1, 2, 3, 4, 5])
+    out2 = _agg("avg", [s.copy_shallow() for s in both])
+    _exact(out2[0].values, [10, 11, 12, 13, 14, 15])
+
+
+def test_count_with_nan_tails():
+    # count(label_set(time()<1500,..) or label_set(time()<1800,..))
+    a = bop.binary_op_eval(BinOpSpec("<"), [_ls("time", "foo", "bar")],
+                           [_ls(1500)])
+    b = bop.binary_op_eval(BinOpSpec("<"), [_ls("time", "baz", "sss")],
+                           [_ls(1800)])
+    both = a + b
+    out = _agg("count", both)
+    v = out[0].values
+    assert list(v[:4]) == [2, 2, 2, 1]
+    assert math.isnan(v[4]) and math.isnan(v[5])
+
+
+def test_geomean_sum2():
+    out = _agg("geomean", [Series(MetricName(), TIME / 100)])
+    np.testing.assert_allclose(out[0].values, [10, 12, 14, 16, 18, 20],
+                               rtol=1e-14)
+    out2 = _agg("sum2", [Series(MetricName(), TIME / 100)])
+    _exact(out2[0].values, [100, 144, 196, 256, 324, 400])
+
+
+def test_mode_aliases():
+    series = [Series(MetricName(f"m{i}"), np.full(6, float(v)))
+              for i, v in enumerate([3, 2, 3, 4, 3, 2])]
+    out = _agg("mode", series)
+    _exact(out[0].values, [3, 3, 3, 3, 3, 3])
+
+
+def test_min_by_unknown_tag():
+    left = [_ls(10, "foo", "bar")]
+    right = [Series(MetricName("", [("baz", "sss")]), TIME / 100 / 1.5)]
+    both = bop.binary_op_eval(BinOpSpec("or"), left, right)
+    out = _agg("min", both, modifier_op="by", modifier_args=["unknowntag"])
+    assert len(out) == 1
+    _exact(out[0].values,
+           [6.666666666666667, 8, 9.333333333333334, 10, 10, 10])
+
+
+def test_distinct_union():
+    # distinct(union(1+time() > 1100, label_set(time() > 1700, foo=bar)))
+    a = bop.binary_op_eval(BinOpSpec(">"),
+                           [Series(MetricName(), 1 + TIME)], [_ls(1100)])
+    b = bop.binary_op_eval(BinOpSpec(">"), [_ls("time", "foo", "bar")],
+                           [_ls(1700)])
+    out = _agg("distinct", a + b)
+    v = out[0].values
+    assert math.isnan(v[0])
+    assert list(v[1:]) == [1, 1, 1, 2, 2]
