@@ -135,3 +135,53 @@ def test_cache_roundtrip_partial_hit():
     assert ns2 > 2000 and nan_eq(v2, full)
     # different key misses
     assert c.get("rate(m[1m])", 60_000, 200, 1000, 2000)[0] is None
+
+
+def test_align_start_end():
+    """alignStartEnd (eval.go:103-112): floor start, ceil end to step."""
+    from victoriametrics_amd.engine import align_start_end
+    assert align_start_end(1000, 2000, 200) == (1000, 2000)
+    assert align_start_end(1050, 1950, 200) == (1000, 2000)
+    assert align_start_end(999, 2001, 200) == (800, 2200)
+    assert align_start_end(0, 0, 200) == (0, 0)
+    # negative starts: Go's truncated % (unlike Python's floored %):
+    # -150 % 200 == -150 in Go -> start stays -150 - (-150) = 0;
+    # -50 % 200 == -50 -> adjust <= 0 -> end unchanged
+    assert align_start_end(-150, 50, 200) == (0, 200)
+    assert align_start_end(-350, -50, 200) == (-200, -50)
+
+
+def test_subquery_grid_parameters():
+    """evalRollupFuncWithSubquery's grid derivation (eval.go:1033-1060):
+    sq grid extends the outer window + maxSilenceInterval and aligns."""
+    from victoriametrics_amd import engine
+    captured = {}
+
+    def inner(sq_start, sq_end, sq_step):
+        captured["grid"] = (sq_start, sq_end, sq_step)
+        n = 1 + (sq_end - sq_start) // sq_step
+        raise RuntimeError("stop")  # grid captured; no GPU needed
+
+    start, end, step, window, sq_step = 1_000_000, 2_000_000, 200_000, \
+        300_000, 100_000
+    try:
+        engine.rollup_subquery("max_over_time", start, end, step, window,
+                               sq_step, inner)
+    except RuntimeError:
+        pass
+    sq_start, sq_end, got_step = captured["grid"]
+    assert got_step == sq_step
+    exp_start, exp_end = engine.align_start_end(
+        start - (window + sq_step + engine.MAX_SILENCE_INTERVAL_MS),
+        end + sq_step, sq_step)
+    assert (sq_start, sq_end) == (exp_start, exp_end)
+    assert sq_start <= start - window
+    assert sq_end >= end
+    # sq_step defaults to the outer step (eval.go:1040)
+    captured.clear()
+    try:
+        engine.rollup_subquery("max_over_time", start, end, step, window,
+                               0, inner)
+    except RuntimeError:
+        pass
+    assert captured["grid"][2] == step

@@ -908,10 +908,18 @@ def merge_blocks(ts, vals, block_offsets, series_block_start,
     return out_ts[:total], out_vals[:total], out_off
 
 
+def _go_mod(a, b):
+    """Go's % (truncated toward zero), vs Python's floored %."""
+    r = abs(a) % abs(b)
+    return -r if a < 0 else r
+
+
 def align_start_end(start, end, step):
-    """alignStartEnd (eval.go): floor start, ceil end to step multiples."""
-    start = start - start % step
-    adjust = end % step
+    """alignStartEnd (eval.go:103-112): round start down and end up to
+    step multiples — with Go's truncated %, which differs from Python's
+    floored % for negative timestamps."""
+    start = start - _go_mod(start, step)
+    adjust = _go_mod(end, step)
     if adjust > 0:
         end += step - adjust
     return start, end
