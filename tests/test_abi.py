@@ -156,3 +156,34 @@ def test_rollup_plan_respects_max_points():
     import pytest as _pytest
     with _pytest.raises(limits.QueryLimitError):
         RollupPlan("rate", 0, limits.max_points_per_timeseries * 2_000, 1_000)
+
+
+def test_plan_flag_wiring():
+    """getRollupConfigs wiring pins (rollup.go:374-516 + eval.go:2108):
+    per-function preFunc/window/staleness flags."""
+    from victoriametrics_amd.engine import RollupPlan
+    p = RollupPlan("rate", 0, 10_000, 1_000, window=5_000,
+                   lookback_delta=300_000)
+    assert p._c.remove_counter_resets == 1
+    assert p._c.may_adjust_window == 1
+    assert p._c.samples_scanned_per_call == 2
+    assert p._c.drop_stale_nans == 1
+    # stalenessInterval = lookbackDelta + window when lookback != 0
+    assert p._c.max_staleness_interval == 305_000
+    p2 = RollupPlan("rate", 0, 10_000, 1_000, window=5_000)
+    assert p2._c.max_staleness_interval == 0
+    d = RollupPlan("default_rollup", 0, 10_000, 1_000)
+    assert d._c.is_default_rollup == 1
+    assert d._c.drop_stale_nans == 0       # keeps stale markers
+    assert d._c.samples_scanned_per_call == 1
+    assert d._c.remove_counter_resets == 0
+    a = RollupPlan("avg_over_time", 0, 10_000, 1_000, window=5_000)
+    assert a._c.may_adjust_window == 0
+    assert a._c.remove_counter_resets == 0
+    assert a._c.samples_scanned_per_call == 0  # O(window) func
+    s = RollupPlan("stale_samples_over_time", 0, 10_000, 1_000,
+                   window=5_000)
+    assert s._c.drop_stale_nans == 0
+    q = RollupPlan("quantile_over_time", 0, 10_000, 1_000, window=5_000,
+                   arg=0.9)
+    assert q._c.arg == 0.9
