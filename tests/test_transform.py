@@ -428,3 +428,18 @@ def test_eval_context_funcs():
     r1 = tf.rand_series(ts, seed=5)[0].values
     r2 = tf.rand_series(ts, seed=5)[0].values
     assert list(r1) == list(r2)
+
+
+def test_label_map():
+    a = S("m", [("env", "dev")], [1])
+    b = S("m", [("env", "prd")], [1])
+    c = S("m", [("env", "x")], [1])
+    tf.label_map([a, b, c], "env", {"dev": "development",
+                                    "prd": "production", "x": ""})
+    assert a.mn.get_tag_value("env") == b"development"
+    assert b.mn.get_tag_value("env") == b"production"
+    assert c.mn.get_tag_value("env") is None
+    # unmapped value untouched
+    d = S("m", [("env", "qa")], [1])
+    tf.label_map([d], "env", {"dev": "x"})
+    assert d.mn.get_tag_value("env") == b"qa"

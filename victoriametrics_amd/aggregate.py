@@ -292,3 +292,19 @@ def histogram_aggregate(series, modifier_op="", modifier_args=(), limit=0):
                 ts.values[i] = float(count)
         rvs.extend(m.values())
     return vmrange_buckets_to_le(rvs)
+
+
+def quantiles(dst_label, phis, series, modifier_op="", modifier_args=(),
+              limit=0):
+    """aggrFuncQuantiles (aggr.go:1162): one quantile result set per phi,
+    labeled dst_label=phi (Go %g)."""
+    rvs = []
+    for phi in phis:
+        part = aggregate("quantile",
+                         [s.copy_shallow() for s in series],
+                         modifier_op, modifier_args, limit, arg=phi)
+        for s in part:
+            s.mn.remove_tag(dst_label)
+            s.mn.add_tag(dst_label, "%g" % phi)
+        rvs.extend(part)
+    return rvs

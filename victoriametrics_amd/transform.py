@@ -874,3 +874,22 @@ def histogram_quantiles(phi_label, phis, series_list):
             s.mn.add_tag(phi_label, "%g" % phi)
         out.extend(part)
     return out
+
+
+def label_map(series_list, label, mapping):
+    """transformLabelMap (transform.go): rewrite label values through a
+    literal src->dst map; an empty result removes the label."""
+    for s in series_list:
+        cur = s.mn.get_tag_value(label)
+        cur_s = (cur or b"").decode("utf-8", "surrogateescape")
+        if cur_s in mapping:
+            new = mapping[cur_s]
+            if label == "__name__":
+                s.mn.metric_group = MetricName._b(new)
+            elif new == "":
+                s.mn.remove_tag(label)
+            else:
+                s.mn.set_tag(label, new)
+        elif cur is not None and len(cur) == 0:
+            s.mn.remove_tag(label)
+    return series_list
