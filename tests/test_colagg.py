@@ -337,3 +337,12 @@ def test_histogram_aggregate():
     # monotone cumulative counts
     counts = [c for _, c in les]
     assert counts == sorted(counts)
+
+
+@pytest.mark.gpu
+def test_quantiles_plural():
+    from victoriametrics_amd import aggregate as agg
+    out = agg.quantiles("phi", [0.25, 0.75], _series_set(), "by", ["node"])
+    assert len(out) == 4
+    labels = sorted(set(s.mn.get_tag_value("phi") for s in out))
+    assert labels == [b"0.25", b"0.75"]
