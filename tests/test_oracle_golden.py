@@ -199,3 +199,50 @@ def test_scrape_interval_tiers():
     assert f(16000) == 24000
     assert f(30000) == 37500
     assert f(60000) == 67500
+
+
+def test_incremental_aggr_vectors():
+    """TestIncrementalAggr (aggr_incremental_test.go:15-100): the exact
+    per-point incremental aggregate expectations over the 7-series fixture,
+    via the oracle's update/merge/finalize path (a last_over_time rollup at
+    the fixture's own timestamps reproduces the raw column)."""
+    nan = math.nan
+    values = [
+        [1, nan, 2, nan],
+        [3, nan, nan, 4],
+        [nan, nan, 5, 6],
+        [7, nan, 8, 9],
+        [4, nan, nan, nan],
+        [2, nan, 3, 2],
+        [0, nan, 1, 1],
+    ]
+    ts_grid = [100_000, 200_000, 300_000, 400_000]
+    # CSR with the NaN samples PRESENT (the reference feeds NaN values to
+    # updateTimeseries, which skips them)
+    ts = np.asarray(ts_grid * len(values), np.int64)
+    vals = np.asarray([v for row in values for v in row], np.float64)
+    offsets = np.arange(len(values) + 1, dtype=np.uint64) * 4
+    gids = np.zeros(len(values), np.int32)
+    expected = {
+        "sum": [17, nan, 19, 22],
+        "min": [0, nan, 1, 1],
+        "max": [7, nan, 8, 9],
+        "avg": [2.8333333333333335, nan, 3.8, 4.4],
+        "count": [6, nan, 5, 5],
+        "sum2": [79, nan, 103, 138],
+        "geomean": [0, nan, 2.9925557394776896, 3.365865436338599],
+    }
+    # last_over_time with window=step reproduces each raw sample at its own
+    # grid point (NaN values count as present samples but NaN results)
+    rc = oracle.make_config("last_over_time", ts_grid[0], ts_grid[-1],
+                            100_000, window=100_000,
+                            may_adjust_window=0)
+    for name, exp in expected.items():
+        out, counts, _ = oracle.rollup_eval_batch(
+            rc, ts, vals, offsets, group_ids=gids, n_groups=1, aggr=name)
+        got = out[0]
+        for g, e in zip(got, exp):
+            if math.isnan(e):
+                assert math.isnan(g), (name, got, exp)
+            else:
+                assert abs(g - e) < 1e-12, (name, got, exp)
