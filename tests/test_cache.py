@@ -5,6 +5,7 @@ import math
 import struct
 
 import numpy as np
+import pytest
 
 from victoriametrics_amd import cache
 
@@ -116,13 +117,22 @@ def test_marshal_max_size_guard():
     assert cache.marshal_timeseries_fast(names, vals, ts, max_size=64) == b""
 
 
+FUTURE = 1 << 60  # now_ms far past every test timestamp (no deadline trim)
+
+
+def _grid(start, end, step):
+    return np.arange(start, end + 1, step, dtype=np.int64)
+
+
 def test_cache_roundtrip_partial_hit():
     c = cache.RollupResultCache()
     names = [(b"m", ())]
     full = np.array([[1.0, 2.0, 3.0, 4.0, 5.0, 6.0]])
     # cache [1000, 1400]
-    c.put("rate(m[5m])", 300_000, 200, 1000, 1400, names, full[:, :3])
-    got_n, got_v, new_start = c.get("rate(m[5m])", 300_000, 200, 1000, 2000)
+    c.put_series("rate(m[5m])", 300_000, 200, names, full[:, :3],
+                 _grid(1000, 1400, 200), now_ms=FUTURE)
+    got_n, got_v, _, new_start = c.get_series("rate(m[5m])", 300_000, 200,
+                                              1000, 2000)
     assert new_start == 1600
     assert nan_eq(got_v, full[:, :3])
     # compute the suffix, merge, and store the full window
@@ -130,11 +140,141 @@ def test_cache_roundtrip_partial_hit():
                                             full[:, 3:], new_start,
                                             1000, 2000, 200)
     assert nan_eq(merged_v, full)
-    c.put("rate(m[5m])", 300_000, 200, 1000, 2000, merged_n, merged_v)
-    _, v2, ns2 = c.get("rate(m[5m])", 300_000, 200, 1000, 2000)
+    c.put_series("rate(m[5m])", 300_000, 200, merged_n, merged_v,
+                 _grid(1000, 2000, 200), now_ms=FUTURE)
+    _, v2, _, ns2 = c.get_series("rate(m[5m])", 300_000, 200, 1000, 2000)
     assert ns2 > 2000 and nan_eq(v2, full)
     # different key misses
-    assert c.get("rate(m[1m])", 60_000, 200, 1000, 2000)[0] is None
+    assert c.get_series("rate(m[1m])", 60_000, 200, 1000, 2000)[0] is None
+
+
+class TestRollupResultCacheGolden:
+    """TestRollupResultCache (rollup_result_cache_test.go:30-320) — the
+    eleven subcases, with ec = {Start:1000, End:2000, Step:200},
+    window=456.  `fe`/`ae` from the reference are distinct cache keys."""
+    W, STEP, START, END = 456, 200, 1000, 2000
+
+    def _get(self, c, expr="foo"):
+        return c.get_series(expr, self.W, self.STEP, self.START, self.END)
+
+    def _put(self, c, ts, vals, expr="foo", names=None):
+        c.put_series(expr, self.W, self.STEP,
+                     names or [(b"", ())],
+                     np.asarray(vals, np.float64).reshape(len(names or [0]), -1),
+                     np.asarray(ts, np.int64), now_ms=FUTURE)
+
+    def test_empty(self):
+        c = cache.RollupResultCache()
+        n, v, t, ns = self._get(c)
+        assert n is None and ns == self.START
+
+    @pytest.mark.parametrize("expr", ["foo", "foobar"])
+    def test_start_overlap(self, expr):
+        # start-overlap-no-ae / -with-ae: {800,1000,1200} -> {1000,1200},
+        # newStart 1400
+        c = cache.RollupResultCache()
+        self._put(c, [800, 1000, 1200], [0, 1, 2], expr=expr)
+        n, v, t, ns = self._get(c, expr=expr)
+        assert ns == 1400
+        assert list(t) == [1000, 1200] and list(v[0]) == [1, 2]
+
+    def test_end_overlap(self):
+        # {1800,2000,2200,2400} doesn't contain Start -> miss
+        c = cache.RollupResultCache()
+        self._put(c, [1800, 2000, 2200, 2400], [333, 0, 1, 2])
+        n, v, t, ns = self._get(c)
+        assert n is None and ns == 1000
+
+    def test_full_cover(self):
+        # {1200,1400,1600} starts after Start -> miss
+        c = cache.RollupResultCache()
+        self._put(c, [1200, 1400, 1600], [0, 1, 2])
+        n, v, t, ns = self._get(c)
+        assert n is None and ns == 1000
+
+    def test_before_start(self):
+        c = cache.RollupResultCache()
+        self._put(c, [200, 400, 600], [0, 1, 2])
+        n, v, t, ns = self._get(c)
+        assert n is None and ns == 1000
+
+    def test_after_end(self):
+        c = cache.RollupResultCache()
+        self._put(c, [2200, 2400, 2600], [0, 1, 2])
+        n, v, t, ns = self._get(c)
+        assert n is None and ns == 1000
+
+    def test_bigger_than_start_end(self):
+        c = cache.RollupResultCache()
+        self._put(c, [800, 1000, 1200, 1400, 1600, 1800, 2000, 2200],
+                  [0, 1, 2, 3, 4, 5, 6, 7])
+        n, v, t, ns = self._get(c)
+        assert ns == 2200
+        assert list(t) == [1000, 1200, 1400, 1600, 1800, 2000]
+        assert list(v[0]) == [1, 2, 3, 4, 5, 6]
+
+    def test_start_end_match(self):
+        c = cache.RollupResultCache()
+        self._put(c, [1000, 1200, 1400, 1600, 1800, 2000],
+                  [1, 2, 3, 4, 5, 6])
+        n, v, t, ns = self._get(c)
+        assert ns == 2200
+        assert list(v[0]) == [1, 2, 3, 4, 5, 6]
+
+    def test_big_timeseries(self):
+        # 1000 series -> marshaled size > 64Kb in the reference (tests the
+        # GetBig/SetBig split there; here it must simply round-trip)
+        c = cache.RollupResultCache()
+        names = [(b"metric %d" % i, ()) for i in range(1000)]
+        vals = np.tile([1.0, 2, 3, 4, 5, 6], (1000, 1))
+        self._put(c, [1000, 1200, 1400, 1600, 1800, 2000], vals, names=names)
+        n, v, t, ns = self._get(c)
+        assert ns == 2200
+        assert n == names and nan_eq(v, vals)
+
+    def test_duplicate_series_not_stored(self):
+        c = cache.RollupResultCache()
+        names = [(b"", ()), (b"", ())]
+        self._put(c, [800, 1000, 1200], np.tile([0.0, 1, 2], (2, 1)),
+                  names=names)
+        n, v, t, ns = self._get(c)
+        assert n is None and ns == self.START
+
+    def test_multi_timeseries(self):
+        # three entries under one key; GetBestKey picks the one covering
+        # Start with the longest usable span -> tss1, newStart 1400
+        c = cache.RollupResultCache()
+        self._put(c, [800, 1000, 1200], [0, 1, 2])
+        self._put(c, [1800, 2000, 2200, 2400], [333, 0, 1, 2])
+        self._put(c, [1200, 1400, 1600], [0, 1, 2])
+        n, v, t, ns = self._get(c)
+        assert ns == 1400
+        assert list(t) == [1000, 1200] and list(v[0]) == [1, 2]
+
+    def test_deadline_trims_fresh_tail(self):
+        # PutSeries drops points newer than now - step - 5m
+        # (rollup_result_cache.go:392-415)
+        c = cache.RollupResultCache()
+        now = 2_000_000
+        ts = [1000, 1200, 1400, 1_800_000]
+        c.put_series("foo", self.W, self.STEP, [(b"", ())],
+                     np.asarray([[1.0, 2, 3, 4]]), np.asarray(ts, np.int64),
+                     now_ms=now)
+        n, v, t, ns = self._get(c)
+        assert ns == 1600 and list(t) == [1000, 1200, 1400]
+
+    def test_metainfo_entry_cap(self):
+        # AddKey keeps at most 10 entries, dropping the oldest 5 past that
+        # (rollup_result_cache.go:595-607)
+        c = cache.RollupResultCache()
+        for k in range(12):  # disjoint ranges far above [START, END]
+            base = 10_000 + k * 1000
+            self._put(c, [base, base + 200], [k, k])
+        entries = c._meta[c._key("foo", self.W, self.STEP)]
+        assert len(entries) <= 10
+        # the latest entry is still retrievable
+        n, v, t, ns = c.get_series("foo", self.W, self.STEP, 21_000, 21_200)
+        assert ns == 21_400 and list(v[0]) == [11, 11]
 
 
 def test_align_start_end():

@@ -20,6 +20,7 @@ Series identity: callers pass metric names as
 grouping keys use the sorted-tags form (marshalMetricNameSorted semantics).
 """
 import struct
+import time
 
 import numpy as np
 
@@ -163,46 +164,133 @@ def merge_series(a_names, a_values, b_names, b_values, b_start, start, end, step
     return out_names, vals
 
 
+# -search.cacheTimestampOffset default (rollup_result_cache.go:29): points
+# newer than now - step - offset are not cached (they may still be rewritten
+# by late inserts).
+CACHE_TIMESTAMP_OFFSET_MS = 5 * 60 * 1000
+
+
 class RollupResultCache:
-    """Prefix-window cache in the spirit of rollupResultCache
-    (rollup_result_cache.go): stores marshaled series per
-    (expr, window, step, filters) key; get() returns (series, new_start)
-    where new_start > start means only (new_start..end] must be computed,
-    after which merge_series + put() complete the round trip."""
+    """rollupResultCache mirror (rollup_result_cache.go:283-465): stores
+    marshaled series under per-(expr, window, step, filters) metainfo with
+    up to 10 time-range entries (AddKey drops the oldest 5 past that,
+    rollup_result_cache.go:595-607); get_series() picks the best entry
+    (GetBestKey: latest-start entry containing `start`, maximal usable
+    span, rollup_result_cache.go:575-593), slices it to [start, end], and
+    returns new_start = last returned timestamp + step, so only
+    (new_start..end] must be computed; merge_series + put_series complete
+    the round trip."""
 
     def __init__(self, max_bytes=256 << 20):
         self.max_bytes = max_bytes
-        self._store = {}
+        self._meta = {}    # key -> [[start, end, skey], ...]  (AddKey order)
+        self._blobs = {}   # skey -> marshaled series (dict preserves order)
+        self._size = 0
+        self._suffix = 0
+
+    def reset(self):
+        self._meta.clear()
+        self._blobs.clear()
         self._size = 0
 
     @staticmethod
     def _key(expr, window, step, filters=b""):
         return (str(expr), int(window), int(step), bytes(filters))
 
-    def put(self, expr, window, step, start, end, names, values, filters=b""):
-        timestamps = np.arange(start, end + 1, step, dtype=np.int64)
-        data = marshal_timeseries_fast(names, values, timestamps)
-        k = self._key(expr, window, step, filters)
-        old = self._store.pop(k, None)
-        if old is not None:
-            self._size -= len(old[2])
-        self._store[k] = (start, end, data)
-        self._size += len(data)
-        while self._size > self.max_bytes and self._store:
-            _, (s, e, d) = self._store.popitem()
-            self._size -= len(d)
+    def _evict(self):
+        while self._size > self.max_bytes and self._blobs:
+            skey = next(iter(self._blobs))
+            self._size -= len(self._blobs.pop(skey))
+            # dangling metainfo entries are dropped lazily on get_series,
+            # as the reference's RemoveKey-on-cache-miss path does
+            # (rollup_result_cache.go:313-320).
 
-    def get(self, expr, window, step, start, end, filters=b""):
-        """Returns (names, values, new_start).  A miss returns
-        (None, None, start)."""
-        k = self._key(expr, window, step, filters)
-        hit = self._store.get(k)
-        if hit is None:
-            return None, None, start
-        c_start, c_end, data = hit
-        if c_start != start or c_end < start:
-            return None, None, start
+    def put_series(self, expr, window, step, names, values, timestamps,
+                   filters=b"", now_ms=None):
+        """PutSeries (rollup_result_cache.go:364-465).  `timestamps` is the
+        actual grid the series were computed on (it may extend beyond the
+        request range on either side)."""
+        if len(names) == 0:
+            return
+        if len(names) > 1:
+            # series with duplicate naming cannot be merged later; skip
+            # (rollup_result_cache.go:376-390)
+            seen = set()
+            for name in names:
+                k = _name_key(name)
+                if k in seen:
+                    return
+                seen.add(k)
+        timestamps = np.ascontiguousarray(timestamps, np.int64)
+        values = np.ascontiguousarray(values, np.float64)
+        # drop trailing points newer than now - step - cacheTimestampOffset
+        # (rollup_result_cache.go:392-415)
+        if now_ms is None:
+            now_ms = int(time.time() * 1000)
+        deadline = int(now_ms) - int(step) - CACHE_TIMESTAMP_OFFSET_MS
+        i = len(timestamps)
+        while i > 0 and timestamps[i - 1] > deadline:
+            i -= 1
+        if i == 0:
+            return
+        if i < len(timestamps):
+            timestamps = timestamps[:i]
+            values = values[:, :i]
+        start, end = int(timestamps[0]), int(timestamps[-1])
+        key = self._key(expr, window, step, filters)
+        entries = self._meta.setdefault(key, [])
+        # CoversTimeRange (rollup_result_cache.go:563-573)
+        for e_start, e_end, _ in entries:
+            if start >= e_start and end <= e_end:
+                return
+        data = marshal_timeseries_fast(names, values, timestamps)
+        if not data:
+            return
+        self._suffix += 1
+        skey = self._suffix
+        self._blobs[skey] = data
+        self._size += len(data)
+        self._evict()
+        entries.append([start, end, skey])
+        if len(entries) > 10:
+            del entries[:5]
+
+    def get_series(self, expr, window, step, start, end, filters=b""):
+        """GetSeries (rollup_result_cache.go:283-361).  Returns
+        (names, values, timestamps, new_start); a miss returns
+        (None, None, None, start)."""
+        start, end, step = int(start), int(end), int(step)
+        key = self._key(expr, window, step, filters)
+        entries = self._meta.get(key)
+        if not entries:
+            return None, None, None, start
+        # GetBestKey: among entries starting at or before `start`, the one
+        # with the largest usable span d = min(end, e.end) - start (>= 0)
+        best = None
+        d_max = 0
+        for e in entries:
+            e_start, e_end, _ = e
+            if start < e_start:
+                continue
+            d = (end if end <= e_end else e_end) - start
+            if d >= d_max:
+                d_max = d
+                best = e
+        if best is None:
+            return None, None, None, start
+        data = self._blobs.get(best[2])
+        if data is None:  # evicted under the metainfo: RemoveKey + miss
+            entries.remove(best)
+            return None, None, None, start
         names, values, timestamps = unmarshal_timeseries_fast(data)
-        usable_end = min(c_end, end)
-        npts = (usable_end - start) // step + 1
-        return names, values[:, :npts], start + npts * step
+        timestamps = np.asarray(timestamps, np.int64)
+        i = int(np.searchsorted(timestamps, start, side="left"))
+        if i == len(timestamps) or timestamps[i] != start:
+            # cached series don't cover `start` on the exact grid
+            return None, None, None, start
+        j = int(np.searchsorted(timestamps, end, side="right"))
+        if j <= i:
+            return None, None, None, start
+        out_ts = timestamps[i:j]
+        new_start = int(out_ts[-1]) + step
+        return names, values[:, i:j], out_ts, new_start
