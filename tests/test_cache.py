@@ -263,6 +263,25 @@ class TestRollupResultCacheGolden:
         n, v, t, ns = self._get(c)
         assert ns == 1600 and list(t) == [1000, 1200, 1400]
 
+    def test_instant_values_roundtrip(self):
+        # GetInstantValues/PutInstantValues/DeleteInstantValues
+        # (rollup_result_cache.go:220-281): flat key, one point per series
+        c = cache.RollupResultCache()
+        names = [(b"a", ()), (b"b", ((b"x", b"y"),))]
+        vals = np.asarray([[1.5], [2.5]])
+        c.put_instant_values("up", 0, 300, names, vals, [5000])
+        n, v, t = c.get_instant_values("up", 0, 300)
+        assert n == [(b"a", ()), (b"b", ((b"x", b"y"),))]
+        assert nan_eq(v, vals) and t == 5000
+        # different (window, step) miss; delete clears
+        assert c.get_instant_values("up", 1, 300)[0] is None
+        c.delete_instant_values("up", 0, 300)
+        assert c.get_instant_values("up", 0, 300)[0] is None
+        # multi-point input is rejected (assertInstantValues)
+        with pytest.raises(ValueError):
+            c.put_instant_values("up", 0, 300, names,
+                                 np.ones((2, 2)), [5000, 5300])
+
     def test_metainfo_entry_cap(self):
         # AddKey keeps at most 10 entries, dropping the oldest 5 past that
         # (rollup_result_cache.go:595-607)

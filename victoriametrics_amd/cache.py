@@ -255,6 +255,54 @@ class RollupResultCache:
         if len(entries) > 10:
             del entries[:5]
 
+    # -- instant values (rollup_result_cache.go:220-281): a flat keyed
+    # store of single-point series, used by the instant-rollup
+    # optimization above this layer.  Keys are disjoint from the series
+    # keys (the reference prefixes rollupResultCacheTypePrefix).
+
+    @staticmethod
+    def _ikey(expr, window, step, filters=b""):
+        return ("instant", str(expr), int(window), int(step), bytes(filters))
+
+    @staticmethod
+    def _assert_instant(values, timestamps):
+        # assertInstantValues: one point per series, shared timestamp
+        if np.asarray(timestamps).size != 1:
+            raise ValueError("instant series must have exactly one point")
+        if np.asarray(values).ndim == 2 and np.asarray(values).shape[1] != 1:
+            raise ValueError("instant series must have exactly one value")
+
+    def put_instant_values(self, expr, window, step, names, values,
+                           timestamps, filters=b""):
+        if len(names) == 0:
+            return
+        self._assert_instant(values, timestamps)
+        values = np.asarray(values, np.float64).reshape(len(names), 1)
+        data = marshal_timeseries_fast(
+            names, values, np.asarray(timestamps, np.int64).reshape(1))
+        k = self._ikey(expr, window, step, filters)
+        old = self._blobs.pop(k, None)
+        if old is not None:
+            self._size -= len(old)
+        self._blobs[k] = data
+        self._size += len(data)
+        self._evict()
+
+    def get_instant_values(self, expr, window, step, filters=b""):
+        """Returns (names, values[n,1], timestamp) or (None, None, None)."""
+        data = self._blobs.get(self._ikey(expr, window, step, filters))
+        if data is None:
+            return None, None, None
+        names, values, timestamps = unmarshal_timeseries_fast(data)
+        if len(names) == 0:
+            return None, None, None
+        return names, values, int(np.asarray(timestamps)[0])
+
+    def delete_instant_values(self, expr, window, step, filters=b""):
+        old = self._blobs.pop(self._ikey(expr, window, step, filters), None)
+        if old is not None:
+            self._size -= len(old)
+
     def get_series(self, expr, window, step, start, end, filters=b""):
         """GetSeries (rollup_result_cache.go:283-361).  Returns
         (names, values, timestamps, new_start); a miss returns
