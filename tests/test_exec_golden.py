@@ -301,3 +301,98 @@ def test_distinct_union():
     v = out[0].values
     assert math.isnan(v[0])
     assert list(v[1:]) == [1, 1, 1, 2, 2]
+
+
+# ---------------------------------------------------------------------------
+# histogram family (exec_test.go histogram_quantile / histogram_share /
+# histogram_fraction cases) through the device hq/hshare kernels
+# ---------------------------------------------------------------------------
+
+def _buckets(*pairs):
+    """pairs: (le_string, constant_value) -> bucket Series on the 6-pt grid"""
+    return [Series(MetricName(b"", [(b"le", le.encode())]),
+                   np.full(6, float(v)))
+            for le, v in pairs]
+
+
+def _hist(name, series, arg, bounds_label=None):
+    return tfm.histogram_transform(name, series, arg=arg,
+                                   bounds_label=bounds_label)
+
+
+class TestHistogramExecPins:
+    def test_quantile_no_le(self):
+        # histogram_quantile(single-value-no-le / -invalid-le) -> empty
+        s = [Series(MetricName(b"", [(b"foo", b"bar")]), np.full(6, 100.0))]
+        assert _hist("histogram_quantile", s, 0.6) == []
+        assert _hist("histogram_quantile",
+                     _buckets(("foobar", 100)), 0.6) == []
+
+    def test_quantile_inf_le_only(self):
+        assert _hist("histogram_quantile", _buckets(("+Inf", 100)), 0.6) == []
+
+    def test_quantile_zero_value_inf_le(self):
+        got = _hist("histogram_quantile",
+                    _buckets(("+Inf", 100), ("42", 0)), 0.6)
+        _exact(got[0].values, [42.0] * 6)
+
+    def test_quantile_single_valid_le(self):
+        got = _hist("histogram_quantile", _buckets(("200", 100)), 0.6)
+        _exact(got[0].values, [120.0] * 6)
+
+    def test_quantile_max_min_phi(self):
+        bl = lambda: _buckets(("200", 100), ("55", 0))
+        _exact(_hist("histogram_quantile", bl(), 1.0)[0].values, [200.0] * 6)
+        _exact(_hist("histogram_quantile", bl(), 0.0)[0].values, [55.0] * 6)
+
+    def test_quantile_bounds_label(self):
+        got = _hist("histogram_quantile", _buckets(("200", 100)), 0.6,
+                    bounds_label="foobar")
+        by = {s.mn.get_tag_value("foobar"): s.values for s in got}
+        _exact(by[b"lower"], [0.0] * 6)
+        _exact(by[b"upper"], [200.0] * 6)
+        _exact(by[None], [120.0] * 6)
+
+    def test_share_single_valid_le(self):
+        for req, want in [(80, 0.4), (200, 1.0), (300, 1.0)]:
+            got = _hist("histogram_share", _buckets(("200", 100)), req)
+            _exact(got[0].values, [want] * 6)
+
+    def test_share_mid_le(self):
+        bl = lambda: _buckets(("200", 100), ("55", 0))
+        _exact(_hist("histogram_share", bl(), 105)[0].values,
+               [0.3448275862068966] * 6)
+        _exact(_hist("histogram_share", bl(), 55)[0].values, [0.0] * 6)
+        _exact(_hist("histogram_share", bl(), 0)[0].values, [0.0] * 6)
+
+    def test_share_bounds_label(self):
+        got = _hist("histogram_share", _buckets(("200", 100)), 120,
+                    bounds_label="foobar")
+        by = {s.mn.get_tag_value("foobar"): s.values for s in got}
+        _exact(by[b"lower"], [0.0] * 6)
+        _exact(by[b"upper"], [1.0] * 6)
+        _exact(by[None], [0.6] * 6)
+
+    def test_fraction_empty_and_invalid(self):
+        s = [Series(MetricName(b"", [(b"foo", b"bar")]), np.full(6, 100.0))]
+        assert _hist("histogram_fraction", s, (123, 456)) == []
+        assert _hist("histogram_fraction",
+                     _buckets(("foobar", 100)), (50, 60)) == []
+
+    def test_fraction_valid_le(self):
+        _exact(_hist("histogram_fraction", _buckets(("200", 100)),
+                     (0, 100))[0].values, [0.5] * 6)
+        _exact(_hist("histogram_fraction", _buckets(("200", 100)),
+                     (200, 300))[0].values, [0.0] * 6)
+
+    def test_fraction_three_buckets(self):
+        bl = lambda: _buckets(("100", 100), ("50", 40), ("10", 0))
+        _exact(_hist("histogram_fraction", bl(), (0, 100))[0].values,
+               [1.0] * 6)
+        _exact(_hist("histogram_fraction", bl(), (0, 10))[0].values,
+               [0.0] * 6)
+
+    def test_fraction_mid_le(self):
+        got = _hist("histogram_fraction", _buckets(("200", 100), ("55", 0)),
+                    (55, 105))
+        _exact(got[0].values, [0.3448275862068966] * 6)

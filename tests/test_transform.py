@@ -443,3 +443,72 @@ def test_label_map():
     d = S("m", [("env", "qa")], [1])
     tf.label_map([d], "env", {"dev": "x"})
     assert d.mn.get_tag_value("env") == b"qa"
+
+
+# ---------------------------------------------------------------------------
+# histogram_fraction = share(upper) - share(lower) (transform.go:751-828),
+# pinned against the TestExecSuccess expected arrays via the CPU oracle
+# (the GPU path is pinned in test_exec_golden.py)
+# ---------------------------------------------------------------------------
+
+def _oracle_fraction(lower, upper, bucket_vals, les):
+    bv = np.asarray(bucket_vals, np.float64)
+    le = np.asarray(les, np.float64)
+    off = np.asarray([0, len(le)], np.uint64)
+    n = bv.shape[1]
+    hi, _, _ = oracle.histogram_share(np.full(n, float(upper)), bv, le, off)
+    lo, _, _ = oracle.histogram_share(np.full(n, float(lower)), bv, le, off)
+    return (hi - lo)[0]
+
+
+def test_histogram_fraction_exec_pins():
+    # exec_test.go histogram_fraction(single-value-valid-le*) cases
+    # (grid collapsed to one point; values are per-point constants there)
+    assert _oracle_fraction(0, 100, [[100.0]], [200.0]) == [0.5]
+    assert _oracle_fraction(200, 300, [[100.0]], [200.0]) == [0.0]
+    # max-le: buckets {10: 0, 50: 40, 100: 100} -> fraction(0,100) = 1
+    assert _oracle_fraction(0, 100, [[0.0], [40.0], [100.0]],
+                            [10.0, 50.0, 100.0]) == [1.0]
+    # min-le -> 0
+    assert _oracle_fraction(0, 10, [[0.0], [40.0], [100.0]],
+                            [10.0, 50.0, 100.0]) == [0.0]
+    # mid-le: buckets {55: 0, 200: 100}, fraction(55, 105)
+    got = _oracle_fraction(55, 105, [[0.0], [100.0]], [55.0, 200.0])
+    assert got.view(np.int64) == np.float64(0.3448275862068966).view(np.int64)
+    # NaN le propagates
+    assert math.isnan(_oracle_fraction(math.nan, 105,
+                                       [[0.0], [100.0]], [55.0, 200.0])[0])
+
+
+def test_histogram_share_exec_pins():
+    # histogram_share(single-value-valid-le / -mid-le) expected arrays
+    def share(req, bv, les):
+        bv = np.asarray(bv, np.float64)
+        le = np.asarray(les, np.float64)
+        off = np.asarray([0, len(le)], np.uint64)
+        out, _, _ = oracle.histogram_share(
+            np.full(bv.shape[1], float(req)), bv, le, off)
+        return out[0]
+
+    assert share(80, [[100.0]], [200.0]) == [0.4]
+    assert share(200, [[100.0]], [200.0]) == [1.0]
+    assert share(300, [[100.0]], [200.0]) == [1.0]
+    got = share(105, [[0.0], [100.0]], [55.0, 200.0])
+    assert got.view(np.int64) == np.float64(0.3448275862068966).view(np.int64)
+    assert share(55, [[0.0], [100.0]], [55.0, 200.0]) == [0.0]
+    assert share(0, [[0.0], [100.0]], [55.0, 200.0]) == [0.0]
+
+
+def test_histogram_fraction_host_validation():
+    # lower >= upper is a host-side error (transform.go:767-770)
+    with pytest.raises(ValueError):
+        tf.histogram_transform("histogram_fraction",
+                               [S("m", [("le", "200")], [100.0])],
+                               arg=(456, 123))
+    # buckets without le / invalid le produce an empty result
+    assert tf.histogram_transform("histogram_fraction",
+                                  [S("m", [("foo", "bar")], [100.0])],
+                                  arg=(123, 456)) == []
+    assert tf.histogram_transform("histogram_fraction",
+                                  [S("m", [("le", "foobar")], [100.0])],
+                                  arg=(50, 60)) == []

@@ -489,10 +489,21 @@ def _merge_same_le(xss):
 def histogram_transform(name, series_list, arg=None, bounds_label=None):
     """histogram_quantile / histogram_avg / histogram_stddev /
     histogram_stdvar / histogram_share over raw bucket series (vmrange or
-    le).  arg: phi (quantile) or le (share).  Returns result Series list
-    (+ lower/upper series when bounds_label is set)."""
+    le).  arg: phi (quantile) or le (share); for histogram_fraction the
+    (lower, upper) pair — transformHistogramFraction (transform.go:751-828)
+    is share(upper) - share(lower) on the same fixed buckets, evaluated
+    here as two hshare kernel passes (fixBrokenBuckets is deterministic,
+    so both passes see identical fixed values).  Returns result Series
+    list (+ lower/upper series when bounds_label is set)."""
     from . import engine
     from .binary_op import Series
+    if name == "histogram_fraction":
+        lo_req = np.atleast_1d(np.asarray(arg[0], np.float64))
+        hi_req = np.atleast_1d(np.asarray(arg[1], np.float64))
+        if lo_req[0] >= hi_req[0]:
+            raise ValueError(
+                "lower le cannot be greater than upper le; got lower le: "
+                "%f, upper le: %f" % (lo_req[0], hi_req[0]))
     tss = vmrange_buckets_to_le(series_list)
     m = group_le_timeseries(tss)
     if not m:
@@ -501,7 +512,8 @@ def histogram_transform(name, series_list, arg=None, bounds_label=None):
     dsts = []
     for k, xss in m.items():
         xss.sort(key=lambda x: x[0])
-        if name in ("histogram_quantile", "histogram_share"):
+        if name in ("histogram_quantile", "histogram_share",
+                    "histogram_fraction"):
             xss = _merge_same_le(xss)
         for le, s in xss:
             rows.append(s.values)
@@ -521,6 +533,16 @@ def histogram_transform(name, series_list, arg=None, bounds_label=None):
         req = np.full(n_grid, float(arg))
         out, lo, hi = engine.histogram_share(
             req, bv, les, goff, bounds=bounds_label is not None)
+    elif name == "histogram_fraction":
+        def _full(r):
+            if r.size == 1:
+                return np.full(n_grid, r[0])
+            if r.size != n_grid:
+                raise ValueError("le series length mismatch")
+            return np.ascontiguousarray(r)
+        out_hi, _, _ = engine.histogram_share(_full(hi_req), bv, les, goff)
+        out_lo, _, _ = engine.histogram_share(_full(lo_req), bv, les, goff)
+        out = out_hi - out_lo
     else:
         mode = {"histogram_avg": 0, "histogram_stddev": 1,
                 "histogram_stdvar": 2}[name]
