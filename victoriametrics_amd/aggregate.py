@@ -8,7 +8,10 @@ on the GPU:
 - share/zscore rewrite member series in place (per-series outputs);
 - outliers_iqr / outliers_mad / outliersk filter member series via the
   bounds + filter kernels;
-- topk/bottomk families route to the existing selection kernels
+- topk/bottomk (per-point) and the topk_*/bottomk_* range families plus
+  outliersk run host-side over resident Series (newAggrFuncTopK /
+  newAggrFuncRangeTopK / aggrFuncOutliersK, aggr.go:669-802); the
+  batch-resident selection path stays on the device kernels
   (engine.topk_range / topk_pointwise);
 - any / limitk are metadata selections (host).
 
@@ -78,8 +81,11 @@ def _matrix(groups):
 
 
 def aggregate(name, series, modifier_op="", modifier_args=(), limit=0,
-              arg=None):
-    """aggrFuncs dispatch (aggr.go:40) over resident Series."""
+              arg=None, remaining_sum_tag=""):
+    """aggrFuncs dispatch (aggr.go:40) over resident Series.  arg: phi for
+    quantile, k for limitk and the topk/bottomk families (scalar or
+    per-point array); remaining_sum_tag: the optional 3rd topk_*/bottomk_*
+    argument ("tag" or "tag=value", aggr.go:751-760)."""
     name = name.lower()
     if name == "quantile":
         groups = prepare_series(series, modifier_op, modifier_args, limit)
@@ -145,9 +151,26 @@ def aggregate(name, series, modifier_op="", modifier_args=(), limit=0,
                     rvs.append(s)
                 i += 1
         return rvs
+    if name in ("topk", "bottomk") or name in _RANGE_TOPK \
+            or name == "outliersk":
+        groups = prepare_series(series, modifier_op, modifier_args, limit,
+                                keep_original=True)
+        if not groups:
+            return []
+        n_grid = len(groups[0][1][0].values)
+        ks = np.broadcast_to(np.atleast_1d(
+            np.asarray(arg, np.float64)), (n_grid,))
+        if name == "topk":
+            return _pointwise_topk(groups, ks, False)
+        if name == "bottomk":
+            return _pointwise_topk(groups, ks, True)
+        if name == "outliersk":
+            return _outliersk(groups, ks, modifier_op, modifier_args)
+        summary, is_reverse = _RANGE_TOPK[name]
+        return _range_topk(groups, ks, summary, is_reverse,
+                           remaining_sum_tag, modifier_op, modifier_args)
     raise ValueError(f"unsupported aggregate {name!r} "
-                     "(topk/bottomk: engine.topk_*; count_values/histogram: "
-                     "host metadata layer)")
+                     "(count_values/histogram: host metadata layer)")
 
 
 def _reduce(op, groups, phi=0.0):
@@ -158,6 +181,219 @@ def _reduce(op, groups, phi=0.0):
     rvs = []
     for gi, (gmn, members) in enumerate(groups):
         rvs.append(Series(gmn, out[gi]))
+    return rvs
+
+
+# ---------------------------------------------------------------------------
+# topk/bottomk family over resident Series (newAggrFuncTopK /
+# newAggrFuncRangeTopK / aggrFuncOutliersK, aggr.go:669-802, 1259-1290).
+# Batch-resident selection stays on the device kernels
+# (engine.topk_range / topk_pointwise); this is the aggr.go host surface.
+# ---------------------------------------------------------------------------
+
+def quantile_sorted(phi, values):
+    """quantileSorted (aggr.go): values already sorted, NaN-free."""
+    if len(values) == 0 or math.isnan(phi):
+        return math.nan
+    if phi < 0:
+        return -math.inf
+    if phi > 1:
+        return math.inf
+    n = float(len(values))
+    rank = phi * (n - 1)
+    lower = max(0.0, math.floor(rank))
+    upper = min(n - 1, lower + 1)
+    weight = rank - math.floor(rank)
+    return values[int(lower)] * (1 - weight) + values[int(upper)] * weight
+
+
+def go_quantile(phi, values):
+    """quantile (aggr.go): prepareForQuantileFloat64 = drop NaNs + sort."""
+    vs = sorted(float(v) for v in values if not math.isnan(v))
+    return quantile_sorted(phi, vs)
+
+
+def _min_value(values):
+    m = math.nan
+    for v in values:
+        v = float(v)
+        if not math.isnan(v) and not (v >= m):  # NaN m compares False
+            m = v
+    return m
+
+
+def _max_value(values):
+    m = math.nan
+    for v in values:
+        v = float(v)
+        if not math.isnan(v) and not (v <= m):
+            m = v
+    return m
+
+
+def _avg_value(values):
+    s, count = 0.0, 0
+    for v in values:
+        v = float(v)
+        if math.isnan(v):
+            continue
+        count += 1
+        s += v
+    return s / count if count else math.nan
+
+
+def _median_value(values):
+    return go_quantile(0.5, values)
+
+
+def _last_value(values):
+    # lastValue: last value before the trailing-NaN tail
+    i = len(values)
+    while i > 0 and math.isnan(float(values[i - 1])):
+        i -= 1
+    return float(values[i - 1]) if i else math.nan
+
+
+_RANGE_TOPK = {
+    "topk_min": (_min_value, False), "bottomk_min": (_min_value, True),
+    "topk_max": (_max_value, False), "bottomk_max": (_max_value, True),
+    "topk_avg": (_avg_value, False), "bottomk_avg": (_avg_value, True),
+    "topk_median": (_median_value, False),
+    "bottomk_median": (_median_value, True),
+    "topk_last": (_last_value, False), "bottomk_last": (_last_value, True),
+}
+
+
+def _less_with_nans_key(v):
+    # lessWithNaNs: NaNs sort below every number
+    return (0, 0.0) if math.isnan(v) else (1, v)
+
+
+def _greater_with_nans_key(v):
+    # greaterWithNaNs ordering realized as an ascending key: NaNs first,
+    # then numbers descending
+    return (0, 0.0) if math.isnan(v) else (1, -v)
+
+
+def _get_int_k(k, max_v):
+    """getIntK + floatToIntBounded (aggr.go:1281): Go float->int with
+    saturation; NaN and negatives mean 0."""
+    if math.isnan(k):
+        return 0
+    if k >= 2.0 ** 63:
+        kn = 1 << 63
+    elif k <= -2.0 ** 63:
+        kn = -(1 << 63)
+    else:
+        kn = int(k)  # truncation toward zero, as Go int(f)
+    if kn < 0:
+        return 0
+    return min(kn, max_v)
+
+
+def _sorted_by(tss, keyvals):
+    # Go uses unstable sort.Slice; tie order there is unspecified, so the
+    # stable sort here is one valid realization
+    order = sorted(range(len(tss)), key=lambda i: keyvals[i])
+    return [tss[i] for i in order]
+
+
+def _remaining_sum_series(tss, modifier_op, modifier_args, ks, tag_name):
+    """getRemainingSumTimeseries (aggr.go:751): per point, the sum of the
+    non-selected (sorted-prefix) series' non-NaN values."""
+    if not tag_name or not tss:
+        return None
+    mn = tss[0].mn.copy()
+    remove_group_tags(mn, modifier_op, modifier_args)
+    tag_value = tag_name
+    if "=" in tag_name:
+        tag_name, tag_value = tag_name.split("=", 1)
+    mn.remove_tag(tag_name)
+    mn.add_tag(tag_name, tag_value)
+    n = len(tss)
+    vals = np.empty(len(ks))
+    for i, k in enumerate(ks):
+        kn = _get_int_k(float(k), n)
+        s, count = 0.0, 0
+        for ts in tss[:n - kn]:
+            v = float(ts.values[i])
+            if math.isnan(v):
+                continue
+            s += v
+            count += 1
+        vals[i] = s if count else math.nan
+    return Series(mn, vals)
+
+
+def _range_topk(groups, ks, summary_f, is_reverse, remaining_sum_tag,
+                modifier_op, modifier_args):
+    """getRangeTopKTimeseries (aggr.go:704): rank whole series by a
+    summary of their values, NaN-out all but the per-point top k."""
+    keyf = _greater_with_nans_key if is_reverse else _less_with_nans_key
+    rvs = []
+    for _, tss in groups:
+        tss = _sorted_by(tss, [keyf(summary_f(s.values)) for s in tss])
+        rem = _remaining_sum_series(tss, modifier_op, modifier_args, ks,
+                                    remaining_sum_tag)
+        n = len(tss)
+        for i, k in enumerate(ks):
+            kn = _get_int_k(float(k), n)
+            for s in tss[:n - kn]:
+                s.values[i] = math.nan
+        if rem is not None:
+            tss.append(rem)
+        tss = remove_empty_series(tss)
+        tss.reverse()
+        rvs.extend(tss)
+    return rvs
+
+
+def _pointwise_topk(groups, ks, is_reverse):
+    """newAggrFuncTopK (aggr.go:669): re-rank per grid point."""
+    keyf = _greater_with_nans_key if is_reverse else _less_with_nans_key
+    rvs = []
+    for _, tss in groups:
+        tss = list(tss)
+        n_grid = len(tss[0].values)
+        for i in range(n_grid):
+            tss = _sorted_by(tss, [keyf(float(s.values[i])) for s in tss])
+            kn = _get_int_k(float(ks[i]), len(tss))
+            for s in tss[:len(tss) - kn]:
+                s.values[i] = math.nan
+        tss = remove_empty_series(tss)
+        tss.reverse()
+        rvs.extend(tss)
+    return rvs
+
+
+def _per_point_medians(tss):
+    """getPerPointMedians (aggr.go:730)."""
+    n_grid = len(tss[0].values)
+    out = np.empty(n_grid)
+    for i in range(n_grid):
+        out[i] = go_quantile(
+            0.5, [float(s.values[i]) for s in tss
+                  if not math.isnan(float(s.values[i]))])
+    return out
+
+
+def _outliersk(groups, ks, modifier_op, modifier_args):
+    """aggrFuncOutliersK (aggr.go:686): range-topk by squared deviation
+    from the per-point medians (sequential sum — a NaN value or median
+    poisons that series' score, as in the reference)."""
+    rvs = []
+    for g in groups:
+        medians = _per_point_medians(g[1])
+
+        def f(values, medians=medians):
+            s = 0.0
+            for i, v in enumerate(values):
+                d = float(v) - medians[i]
+                s += d * d
+            return s
+
+        rvs.extend(_range_topk([g], ks, f, False, "", modifier_op,
+                               modifier_args))
     return rvs
 
 
