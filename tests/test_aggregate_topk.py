@@ -179,3 +179,27 @@ def test_get_int_k():
     assert agg._get_int_k(2.9, 5) == 2
     assert agg._get_int_k(math.inf, 5) == 5
     assert agg._get_int_k(1e300, 5) == 5
+
+
+def test_aggr_dispatch_covers_reference_map():
+    # the 37 keys of aggrFuncs (aggr.go:17-56) each have a host entry point:
+    # aggregate() for 34, count_values()/quantiles()/histogram_aggregate()
+    # for the three multi-output metadata funcs
+    reference_names = {
+        "any", "avg", "bottomk", "bottomk_avg", "bottomk_last",
+        "bottomk_max", "bottomk_median", "bottomk_min", "count",
+        "count_values", "distinct", "geomean", "group", "histogram",
+        "limitk", "mad", "max", "median", "min", "mode", "outliers_iqr",
+        "outliers_mad", "outliersk", "quantile", "quantiles", "share",
+        "stddev", "stdvar", "sum", "sum2", "topk", "topk_avg", "topk_last",
+        "topk_max", "topk_median", "topk_min", "zscore"}
+    dispatched = (agg.REDUCERS | agg.PER_SERIES | set(agg._RANGE_TOPK) |
+                  {"quantile", "any", "limitk", "outliers_iqr",
+                   "outliers_mad", "topk", "bottomk", "outliersk"})
+    separate = {"count_values": agg.count_values,
+                "quantiles": agg.quantiles,
+                "histogram": agg.histogram_aggregate}
+    missing = reference_names - dispatched - set(separate)
+    assert not missing, f"aggrFuncs without a host entry point: {missing}"
+    for fn in separate.values():
+        assert callable(fn)
