@@ -512,3 +512,80 @@ def test_histogram_fraction_host_validation():
     assert tf.histogram_transform("histogram_fraction",
                                   [S("m", [("le", "foobar")], [100.0])],
                                   arg=(50, 60)) == []
+
+
+def test_transform_dispatch_covers_reference_map():
+    # the 111 keys of transformFuncs (transform.go:31-145) each have an
+    # entry point: the device kernel dispatch tables or a named host func
+    reference_names = {
+        "", "abs", "absent", "acos", "acosh", "asin", "asinh", "atan",
+        "atanh", "bitmap_and", "bitmap_or", "bitmap_xor", "buckets_limit",
+        "ceil", "clamp", "clamp_max", "clamp_min", "cos", "cosh",
+        "day_of_month", "day_of_week", "day_of_year", "days_in_month",
+        "deg", "drop_common_labels", "drop_empty_series", "end", "exp",
+        "floor", "histogram_avg", "histogram_fraction", "histogram_quantile",
+        "histogram_quantiles", "histogram_share", "histogram_stddev",
+        "histogram_stdvar", "hour", "interpolate", "keep_last_value",
+        "keep_next_value", "label_copy", "label_del", "label_graphite_group",
+        "label_join", "label_keep", "label_lowercase", "label_map",
+        "label_match", "label_mismatch", "label_move", "label_replace",
+        "label_set", "label_transform", "label_uppercase", "label_value",
+        "labels_equal", "limit_offset", "ln", "log2", "log10", "minute",
+        "month", "now", "pi", "prometheus_buckets", "rad", "rand",
+        "rand_exponential", "rand_normal", "range_avg",
+        "range_first", "range_last", "range_linear_regression", "range_mad",
+        "range_max", "range_min", "range_normalize",
+        "range_quantile", "range_stddev", "range_stdvar", "range_sum",
+        "range_trim_outliers", "range_trim_spikes", "range_trim_zscore",
+        "range_zscore", "remove_resets", "round", "running_avg",
+        "running_max", "running_min", "running_sum", "scalar", "sgn", "sin",
+        "sinh", "smooth_exponential", "sort", "sort_by_label",
+        "sort_by_label_desc", "sort_by_label_numeric",
+        "sort_by_label_numeric_desc", "sort_desc", "sqrt", "start", "step",
+        "tan", "tanh", "time", "timezone_offset", "union", "vector",
+        "year"}
+    kernel = (set(tf._ELEMENTWISE) | set(tf._CLAMP) | {"round"} |
+              set(tf._BITMAP) | set(tf._DATETIME) | set(tf._SERIES))
+    host = {
+        "absent": tf.absent, "buckets_limit": tf.buckets_limit,
+        "drop_common_labels": tf.drop_common_labels,
+        "drop_empty_series": None,  # remove_empty_series (binary_op)
+        "end": tf.eval_end, "start": tf.eval_start, "step": tf.eval_step,
+        "time": tf.eval_time, "pi": tf.eval_pi, "now": tf.eval_now,
+        "histogram_avg": tf.histogram_transform,
+        "histogram_fraction": tf.histogram_transform,
+        "histogram_quantile": tf.histogram_transform,
+        "histogram_share": tf.histogram_transform,
+        "histogram_stddev": tf.histogram_transform,
+        "histogram_stdvar": tf.histogram_transform,
+        "histogram_quantiles": tf.histogram_quantiles,
+        "label_copy": tf.label_copy, "label_del": tf.label_del,
+        "label_graphite_group": tf.label_graphite_group,
+        "label_join": tf.label_join, "label_keep": tf.label_keep,
+        "label_lowercase": tf.label_lowercase, "label_map": tf.label_map,
+        "label_match": tf.label_match,
+        "label_mismatch": tf.label_match,   # negate=True
+        "label_move": tf.label_move, "label_replace": tf.label_replace,
+        "label_set": tf.label_set, "label_transform": tf.label_transform,
+        "label_uppercase": tf.label_uppercase,
+        "label_value": tf.label_value, "labels_equal": tf.labels_equal,
+        "limit_offset": tf.limit_offset,
+        "prometheus_buckets": tf.prometheus_buckets,
+        "rand": tf.rand_series, "rand_exponential": tf.rand_series,
+        "rand_normal": tf.rand_series,
+        "scalar": tf.scalar, "vector": tf.vector, "union": tf.union,
+        "sort": tf.sort_series, "sort_desc": tf.sort_series,  # desc=True
+        "sort_by_label": tf.sort_by_label,
+        "sort_by_label_desc": tf.sort_by_label,
+        "sort_by_label_numeric": tf.sort_by_label_numeric,
+        "sort_by_label_numeric_desc": tf.sort_by_label_numeric,
+        "timezone_offset": tf.timezone_offset,
+        "": None,  # the identity entry (metricsql union sugar)
+    }
+    missing = reference_names - kernel - set(host)
+    assert not missing, f"transformFuncs without an entry point: {missing}"
+
+
+def test_eval_now():
+    out = tf.eval_now([0, 1000], now_s=1234.5)
+    assert list(out[0].values) == [1234.5, 1234.5]

@@ -850,6 +850,15 @@ def eval_pi(grid_timestamps_ms):
     return eval_number(grid_timestamps_ms, math.pi)
 
 
+def eval_now(grid_timestamps_ms, now_s=None):
+    """transformNow (transform.go:2725): wall-clock seconds as a scalar
+    series; now_s injectable for tests."""
+    if now_s is None:
+        import time as _time
+        now_s = _time.time()
+    return eval_number(grid_timestamps_ms, now_s)
+
+
 def scalar(series_list):
     """transformScalar: a single series passes through; otherwise NaN
     (Prometheus scalar() semantics)."""
