@@ -187,3 +187,45 @@ def test_plan_flag_wiring():
     q = RollupPlan("quantile_over_time", 0, 10_000, 1_000, window=5_000,
                    arg=0.9)
     assert q._c.arg == 0.9
+
+
+def test_rollup_dispatch_covers_reference_map():
+    from victoriametrics_amd import engine
+    # the 80 keys of rollupFuncs (rollup.go:31-112): 70 device funcs
+    # (FUNC_IDS incl. aliases), 7 rollup_* multi-result expansions
+    # (rollup_fake_plans), aggr_over_time/quantiles_over_time plan
+    # expansions, and the 2 documented host-side multi-output funcs
+    reference_names = {
+        "absent_over_time", "aggr_over_time", "ascent_over_time",
+        "avg_over_time", "changes", "changes_prometheus",
+        "count_eq_over_time", "count_gt_over_time", "count_le_over_time",
+        "count_ne_over_time", "count_over_time", "count_values_over_time",
+        "decreases_over_time", "default_rollup", "delta", "delta_prometheus",
+        "deriv", "deriv_fast", "descent_over_time", "distinct_over_time",
+        "duration_over_time", "first_over_time", "geomean_over_time",
+        "histogram_over_time", "hoeffding_bound_lower",
+        "hoeffding_bound_upper", "holt_winters", "idelta", "ideriv",
+        "increase", "increase_prometheus", "increase_pure",
+        "increases_over_time", "integrate", "iqr_over_time", "irate", "lag",
+        "last_over_time", "lifetime", "mad_over_time", "max_over_time",
+        "median_over_time", "min_over_time", "mode_over_time",
+        "outlier_iqr_over_time", "predict_linear", "present_over_time",
+        "quantile_over_time", "quantiles_over_time", "range_over_time",
+        "rate", "rate_over_sum", "resets", "rollup", "rollup_candlestick",
+        "rollup_delta", "rollup_deriv", "rollup_increase", "rollup_rate",
+        "rollup_scrape_interval", "scrape_interval", "share_eq_over_time",
+        "share_gt_over_time", "share_le_over_time", "stale_samples_over_time",
+        "stddev_over_time", "stdvar_over_time", "sum_eq_over_time",
+        "sum_gt_over_time", "sum_le_over_time", "sum_over_time",
+        "sum2_over_time", "tfirst_over_time", "timestamp",
+        "timestamp_with_name", "tlast_change_over_time", "tlast_over_time",
+        "tmax_over_time", "tmin_over_time", "zscore_over_time"}
+    expansions = {"aggr_over_time", "quantiles_over_time"}
+    host_multi_output = {"count_values_over_time", "histogram_over_time"}
+    handled = (set(engine.FUNC_IDS) | set(engine.ROLLUP_FAKE_FUNCS) |
+               expansions | host_multi_output)
+    missing = reference_names - handled
+    assert not missing, f"rollupFuncs without an entry point: {missing}"
+    assert callable(engine.aggr_over_time_plans)
+    assert callable(engine.quantiles_over_time_plans)
+    assert callable(engine.rollup_fake_plans)
