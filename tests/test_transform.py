@@ -589,3 +589,72 @@ def test_transform_dispatch_covers_reference_map():
 def test_eval_now():
     out = tf.eval_now([0, 1000], now_s=1234.5)
     assert list(out[0].values) == [1234.5, 1234.5]
+
+
+# ---------------------------------------------------------------------------
+# union / drop_common_labels / limit_offset TestExecSuccess pins
+# (exec_test.go:2113-2180, 2547-2600, 9311-9500) — host-side funcs
+# ---------------------------------------------------------------------------
+
+TIME6 = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+
+
+def test_union_exec_pins():
+    assert tf.union([]) == []
+    # identical labels: later duplicates dropped
+    out = tf.union([[S("", [("foo", "bar")], [1] * 6)],
+                    [S("", [("foo", "bar")], [2] * 6)]])
+    assert len(out) == 1 and list(out[0].values) == [1] * 6
+    # metric group participates in identity
+    out = tf.union([[S("xx", [("foo", "bar")], [1] * 6)],
+                    [S("yy", [("foo", "bar")], [2] * 6)]])
+    assert len(out) == 2
+    by_g = {s.mn.metric_group: list(s.values) for s in out}
+    assert by_g == {b"xx": [1] * 6, b"yy": [2] * 6}
+    # tag order doesn't matter for identity
+    out = tf.union([[S("xx", [("a", "1"), ("b", "2")], [1] * 6)],
+                    [S("xx", [("b", "2"), ("a", "1")], [2] * 6)]])
+    assert len(out) == 1
+
+
+def test_drop_common_labels_exec_pins():
+    # single series: everything common -> all labels dropped
+    s = S("xxx", [("foo", "bar"), ("q", "we")], TIME6)
+    tf.drop_common_labels([s])
+    assert s.mn.metric_group == b"" and s.mn.tags == []
+    # multi series: only (foo, bar) common; names differ and are kept
+    s1 = S("xxx", [("foo", "bar"), ("q", "we")], TIME6)
+    s2 = S("yyy", [("foo", "bar")], TIME6 / 10)
+    tf.drop_common_labels([s1, s2])
+    assert s1.mn.metric_group == b"xxx"
+    assert s1.mn.tags == [(b"q", b"we")]
+    assert s2.mn.metric_group == b"yyy" and s2.mn.tags == []
+    # multi_args case: same __name__ on both -> name dropped too
+    s1 = S("xxx", [("foo", "bar"), ("q", "we")], TIME6)
+    s2 = S("xxx", [("foo", "bar")], TIME6 / 10)
+    tf.drop_common_labels([s1, s2])
+    assert s1.mn.metric_group == b"" and s1.mn.tags == [(b"q", b"we")]
+    assert s2.mn.metric_group == b"" and s2.mn.tags == []
+
+
+def test_limit_offset_exec_pins():
+    def mk():
+        return [S("", [("foo", "a")], TIME6 * 2),
+                S("", [("foo", "x")], TIME6 * 3),
+                S("", [("foo", "y")], TIME6 * 1)]  # sorted by foo
+
+    out = tf.limit_offset(mk(), 1, 1)
+    assert len(out) == 1 and out[0].mn.get_tag_value("foo") == b"x"
+    np.testing.assert_array_equal(out[0].values, TIME6 * 3)
+    assert tf.limit_offset(mk(), 1, 10) == []
+    # empty series are filtered out BEFORE the offset (limit_offset NaN
+    # case): desc order [3:all-NaN, 2:partial, 1:full] -> offset 1 skips
+    # the partial series, not the NaN one
+    vals3 = np.where(TIME6 * 3 < 3000, TIME6 * 3, math.nan)
+    vals2 = np.where(TIME6 * 2 < 3000, TIME6 * 2, math.nan)
+    tss = [S("", [("foo", "3")], vals3),
+           S("", [("foo", "2")], vals2),
+           S("", [("foo", "1")], TIME6)]
+    out = tf.limit_offset(tss, 1, 1)
+    assert len(out) == 1 and out[0].mn.get_tag_value("foo") == b"1"
+    np.testing.assert_array_equal(out[0].values, TIME6)

@@ -345,7 +345,10 @@ def drop_empty_series(series_list):
 
 
 def limit_offset(series_list, limit, offset):
-    # transformLimitOffset (transform.go)
+    # transformLimitOffset (transform.go:2293): empty series are filtered
+    # out BEFORE the offset is applied
+    from .binary_op import remove_empty_series
+    series_list = remove_empty_series(series_list)
     return series_list[offset:offset + limit]
 
 
