@@ -658,3 +658,40 @@ def test_limit_offset_exec_pins():
     out = tf.limit_offset(tss, 1, 1)
     assert len(out) == 1 and out[0].mn.get_tag_value("foo") == b"1"
     np.testing.assert_array_equal(out[0].values, TIME6)
+
+
+def test_buckets_limit_exec_pins():
+    # TestExecSuccess buckets_limit cases (exec_test.go:5236-5470)
+    def bucket(le, v, extra=()):
+        return S("metric", [("le", le)] + list(extra), np.full(6, float(v)))
+
+    def les(out):
+        return [(s.mn.get_tag_value("le").decode(), s.values[0]) for s in
+                sorted(out, key=lambda s: float(s.mn.get_tag_value("le")))]
+
+    # trim_zero_preserve_empty_when_limit_not_reached (limit 3): zero edge
+    # buckets trimmed right-first, then left, down to the limit
+    out = tf.buckets_limit(3, [
+        bucket("+Inf", 36), bucket("25", 36), bucket("21", 36),
+        bucket("19", 36), bucket("18", 36), bucket("17", 36),
+        bucket("16", 36), bucket("12", 27), bucket("9", 14),
+        bucket("6", 0), bucket("1", 0)])
+    assert les(out) == [("9", 14.0), ("12", 27.0), ("16", 36.0)]
+    # trim_zero (limit 5): only the right zero-delta edge is trimmed; the
+    # left zero buckets survive because the limit is already met
+    out = tf.buckets_limit(5, [
+        bucket("18", 36), bucket("17", 36), bucket("16", 36),
+        bucket("12", 27), bucket("9", 14), bucket("6", 0), bucket("1", 0)])
+    assert les(out) == [("1", 0.0), ("6", 0.0), ("9", 14.0),
+                        ("12", 27.0), ("16", 36.0)]
+    # unused (limit 5 >= buckets): untouched
+    out = tf.buckets_limit(5, [bucket("inf", 100, [("x", "y")]),
+                               bucket("120", 50, [("x", "y")])])
+    assert les(out) == [("120", 50.0), ("inf", 100.0)]
+    # used (limit 2 -> min 3): smallest adjacent-delta pairs merged away
+    out = tf.buckets_limit(2, [
+        bucket("inf", 100, [("x", "y")]), bucket("300", 98, [("x", "y")]),
+        bucket("200", 52, [("x", "y")]), bucket("120", 50, [("x", "y")]),
+        bucket("70", 20, [("x", "y")]), bucket("30", 10, [("x", "y")]),
+        bucket("10", 9, [("x", "y")])])
+    assert les(out) == [("10", 9.0), ("300", 98.0), ("inf", 100.0)]
