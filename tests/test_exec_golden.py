@@ -329,7 +329,13 @@ class TestHistogramExecPins:
                      _buckets(("foobar", 100)), 0.6) == []
 
     def test_quantile_inf_le_only(self):
-        assert _hist("histogram_quantile", _buckets(("+Inf", 100)), 0.6) == []
+        # the walk breaks on +Inf and lastNonInf finds nothing -> all-NaN
+        # series; the exec layer's removeEmptySeries yields the reference's
+        # empty result (transform.go:1062-1072)
+        from victoriametrics_amd.binary_op import remove_empty_series
+        got = _hist("histogram_quantile", _buckets(("+Inf", 100)), 0.6)
+        assert len(got) == 1 and np.isnan(got[0].values).all()
+        assert remove_empty_series(got) == []
 
     def test_quantile_zero_value_inf_le(self):
         got = _hist("histogram_quantile",
