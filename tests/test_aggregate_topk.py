@@ -203,3 +203,33 @@ def test_aggr_dispatch_covers_reference_map():
     assert not missing, f"aggrFuncs without a host entry point: {missing}"
     for fn in separate.values():
         assert callable(fn)
+
+
+def test_pointwise_topk_randomized_vs_numpy():
+    # with distinct finite values the per-point selection is exactly
+    # "keep the k largest (smallest for bottomk) at each grid index"
+    rng = np.random.default_rng(7)
+    for trial in range(20):
+        n_series, n_grid = int(rng.integers(1, 8)), int(rng.integers(1, 12))
+        vals = rng.permutation(n_series * n_grid).astype(np.float64)
+        vals = vals.reshape(n_series, n_grid)
+        k = int(rng.integers(0, n_series + 2))
+        series = [Series(MetricName(b"", [(b"i", str(i).encode())]),
+                         vals[i].copy()) for i in range(n_series)]
+        for name, keep_largest in (("topk", True), ("bottomk", False)):
+            got = agg.aggregate(name, [s.copy_shallow() for s in series],
+                                arg=k)
+            out = np.full((n_series, n_grid), math.nan)
+            for s in got:
+                out[int(s.mn.get_tag_value("i"))] = s.values
+            expect = np.full((n_series, n_grid), math.nan)
+            kn = min(k, n_series)
+            for j in range(n_grid):
+                order = np.argsort(vals[:, j])
+                # note: order[-kn:] would be wrong for kn == 0
+                sel = order[n_series - kn:] if keep_largest else order[:kn]
+                expect[sel, j] = vals[sel, j]
+            # series that end up all-NaN are removed, already reflected
+            np.testing.assert_array_equal(
+                out.view(np.int64), expect.view(np.int64),
+                err_msg=f"{name} trial {trial} k={k}")
