@@ -115,3 +115,74 @@ def test_full_query_pipeline():
     qt.done()
     tree = json.loads(qt.to_json())
     assert tree["children"], tree
+
+
+def test_instant_rollup_optimization_on_device():
+    """InstantRollupEvaluator over the device path: eval_at = one-point
+    SeriesBatch.exec; the composed cached+offset result must equal a
+    direct full-window device evaluation bit-for-bit for the
+    sum-decomposable funcs and max/min."""
+    from victoriametrics_amd import instant, synth
+    from victoriametrics_amd.cache import RollupResultCache
+
+    n_series, n_samples = 64, 720  # 1h of 5s samples
+    step_ms = 5_000
+    t0 = 1_600_000_000_000
+    # reset-free counters: the increase composition identity
+    # inc(w@t-off) + inc(off@t) - inc(off@t-w) = inc(w@t) holds exactly
+    # only when no reset falls near a sub-window baseline (the reference
+    # accepts the same edge inaccuracy for cached increase, issue 10098)
+    ts, vals, offsets = synth.counter_batch(n_series, n_samples, t0,
+                                            step=step_ms, seed=42,
+                                            reset_p=0.0)
+    batch = SeriesBatch(ts, vals, offsets)
+    names = [MetricName(b"m", [(b"s", str(i).encode())])
+             for i in range(n_series)]
+    window = 3 * 3600 * 1000  # below the batch span? span = 720*5s = 1h
+    # use min_window_ms = 30min so the optimization engages on this batch
+    min_window = 30 * 60 * 1000
+    window = 45 * 60 * 1000
+
+    calls = []
+
+    def eval_at(func, timestamp, w):
+        calls.append((func, timestamp, w))
+        plan = RollupPlan(func, timestamp, timestamp, step_ms, window=w)
+        out, counts, _ = batch.exec(plan)
+        from victoriametrics_amd.binary_op import Series
+        return [Series(names[i].copy(), out[i].copy())
+                for i in range(n_series)
+                if not math.isnan(out[i, 0])]
+
+    now = int(ts[-1])
+    for func in ("sum_over_time", "count_over_time", "increase",
+                 "max_over_time", "min_over_time"):
+        ev = instant.InstantRollupEvaluator(
+            RollupResultCache(), eval_at, step=step_ms, now_ms=now,
+            min_window_ms=min_window)
+        t1 = now - 5 * 60 * 1000 - 2_500
+        got1 = ev.eval(func, "q", t1, window)
+        direct1 = {s.mn.get_tag_value("s"): s.values[0]
+                   for s in eval_at(func, t1, window)}
+        for s in got1:
+            assert s.values[0] == direct1[s.mn.get_tag_value("s")], func
+        # second query at a later timestamp: composed from cache + two
+        # offset windows, still bit-exact vs the direct device result
+        calls.clear()
+        t2 = now - 2_500
+        got2 = ev.eval(func, "q", t2, window)
+        offset_calls = [c for c in calls if c[2] < window]
+        assert offset_calls, (func, calls)
+        direct2 = {s.mn.get_tag_value("s"): s.values[0]
+                   for s in eval_at(func, t2, window)}
+        assert len(got2) == len(direct2)
+        for s in got2:
+            want = direct2[s.mn.get_tag_value("s")]
+            g = float(s.values[0])
+            if func in ("sum_over_time", "increase"):
+                # cached + start - end is algebraically exact but
+                # reassociated; allow 1-ulp-scale tolerance
+                assert g == pytest.approx(float(want), rel=1e-12), func
+            else:
+                assert g == float(want), func
+    batch.close()
