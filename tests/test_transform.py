@@ -219,13 +219,17 @@ def test_drop_common_labels():
 
 
 def test_sort_series():
+    # positional comparison from the last index (transform.go:2580):
+    # b's NaN at index 1 sorts it before a regardless of desc; c (all
+    # NaN) sorts before b at index 0.  The NaN rule is NOT flipped by
+    # desc — only the value comparison is.
     a = S("a", [], [1, 5])
-    b = S("b", [], [9, NAN])   # last non-NaN = 9
+    b = S("b", [], [9, NAN])
     c = S("c", [], [NAN, NAN])
     out = tf.sort_series([a, b, c])
-    assert out == [c, a, b]
+    assert out == [c, b, a]
     out_desc = tf.sort_series([a, b, c], desc=True)
-    assert out_desc == [b, a, c]
+    assert out_desc == [c, b, a]
 
 
 def test_sort_by_label():
@@ -695,3 +699,43 @@ def test_buckets_limit_exec_pins():
         bucket("70", 20, [("x", "y")]), bucket("30", 10, [("x", "y")]),
         bucket("10", 9, [("x", "y")])])
     assert les(out) == [("10", 9.0), ("300", 98.0), ("inf", 100.0)]
+
+
+def test_sort_positional_nan_semantics():
+    # newTransformFuncSort (transform.go:2580) compares positionally from
+    # the LAST index backwards; a NaN opposite a value sorts first
+    # REGARDLESS of desc; ties advance to the previous index
+    a = S("a", [], [5.0, NAN])
+    b = S("b", [], [1.0, 3.0])
+    out = tf.sort_series([b, a])
+    assert [s.mn.metric_group for s in out] == [b"a", b"b"]
+    # desc flips only the value comparison, not the NaN rule
+    out = tf.sort_series([b, a], desc=True)
+    assert [s.mn.metric_group for s in out] == [b"a", b"b"]
+    # tie at the last index -> decided by the previous one
+    c = S("c", [], [2.0, 7.0])
+    d = S("d", [], [9.0, 7.0])
+    out = tf.sort_series([d, c])
+    assert [s.mn.metric_group for s in out] == [b"c", b"d"]
+    # fully equal -> input order preserved (stable realization of Go's
+    # unspecified tie order)
+    e = S("e", [], [2.0, 7.0])
+    out = tf.sort_series([c, e])
+    assert [s.mn.metric_group for s in out] == [b"c", b"e"]
+    # exec pins: sort(2 or label_set(1,...)) / sort_desc(1 or label_set(2,..))
+    two = S("", [], [2.0] * 6)
+    one = S("", [("xx", "foo")], [1.0] * 6)
+    assert tf.sort_series([two, one])[0] is one
+    two2 = S("", [("xx", "foo")], [2.0] * 6)
+    one2 = S("", [], [1.0] * 6)
+    assert tf.sort_series([one2, two2], desc=True)[0] is two2
+
+
+def test_sort_by_label_name():
+    # "__name__" sorts by the metric group (exec_test.go sort_by_label)
+    foo = S("foo", [], [1.0] * 6)
+    bar = S("bar", [], [2.0] * 6)
+    out = tf.sort_by_label([foo, bar], ["__name__"])
+    assert [s.mn.metric_group for s in out] == [b"bar", b"foo"]
+    out = tf.sort_by_label([foo, bar], ["__name__"], desc=True)
+    assert [s.mn.metric_group for s in out] == [b"foo", b"bar"]

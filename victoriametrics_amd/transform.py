@@ -360,15 +360,32 @@ def sort_by_label(series_list, labels, desc=False):
 
 
 def sort_series(series_list, desc=False):
-    """sort/sort_desc (newTransformFuncSort, transform.go): order by the
-    LAST non-NaN value per series; NaN-only series sort first (asc)."""
-    def key(s):
-        v = s.values
-        for i in range(len(v) - 1, -1, -1):
-            if not math.isnan(v[i]):
-                return v[i]
-        return -math.inf
-    return sorted(series_list, key=key, reverse=desc)
+    """sort/sort_desc (newTransformFuncSort, transform.go:2580): compare
+    POSITIONALLY from the last grid index backwards — at each index a NaN
+    on one side (only) makes that series sort first regardless of desc;
+    equal-or-both-NaN advances to the previous index; the first differing
+    pair of values decides (flipped for desc)."""
+    import functools
+
+    def cmp(sa, sb):
+        a, b = sa.values, sb.values
+        n = len(a) - 1
+        while n >= 0:
+            if not math.isnan(a[n]):
+                if math.isnan(b[n]):
+                    return 1  # b has NaN here -> b sorts first
+                if a[n] != b[n]:
+                    break
+            elif not math.isnan(b[n]):
+                return -1     # a has NaN here -> a sorts first
+            n -= 1
+        if n < 0:
+            return 0
+        if desc:
+            return -1 if b[n] < a[n] else 1
+        return -1 if a[n] < b[n] else 1
+
+    return sorted(series_list, key=functools.cmp_to_key(cmp))
 
 
 # ---------------------------------------------------------------------------
