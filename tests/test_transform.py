@@ -739,3 +739,15 @@ def test_sort_by_label_name():
     assert [s.mn.metric_group for s in out] == [b"bar", b"foo"]
     out = tf.sort_by_label([foo, bar], ["__name__"], desc=True)
     assert [s.mn.metric_group for s in out] == [b"foo", b"bar"]
+
+
+def test_label_join_empty_removes_dst():
+    # transform.go:2059: an empty joined value removes the dst label
+    s = S("m", [("keep", "x")], [1])
+    tf.label_join([s], "j", "-", ["missing1", "missing2"])
+    assert s.mn.get_tag_value("j") is None
+    # separator alone isn't empty: "" + "-" + "" = "-"? No: join of two
+    # empty parts with "-" produces "-", which is non-empty and kept
+    s2 = S("m", [], [1])
+    tf.label_join([s2], "j", "-", ["a", "b"])
+    assert s2.mn.get_tag_value("j") == b"-"

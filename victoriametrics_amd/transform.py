@@ -251,7 +251,10 @@ def label_join(series_list, dst_label, separator, src_labels):
             v = s.mn.get_tag_value(k)
             parts.append(v if v is not None else b"")
         joined = sep.join(parts)
-        if dst_label == "__name__":
+        if joined == b"":
+            # empty result removes the dst label (transform.go:2059)
+            s.mn.remove_tag(dst_label)
+        elif dst_label == "__name__":
             s.mn.metric_group = joined
         else:
             s.mn.set_tag(dst_label, joined)
