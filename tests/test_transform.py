@@ -742,12 +742,13 @@ def test_sort_by_label_name():
 
 
 def test_label_join_empty_removes_dst():
-    # transform.go:2059: an empty joined value removes the dst label
-    s = S("m", [("keep", "x")], [1])
-    tf.label_join([s], "j", "-", ["missing1", "missing2"])
+    # transform.go:2059: an empty joined value removes the dst label —
+    # a SINGLE missing source joins to "" (no separator inserted)
+    s = S("m", [("keep", "x"), ("j", "old")], [1])
+    tf.label_join([s], "j", "-", ["missing"])
     assert s.mn.get_tag_value("j") is None
-    # separator alone isn't empty: "" + "-" + "" = "-"? No: join of two
-    # empty parts with "-" produces "-", which is non-empty and kept
+    # two missing sources still produce the separator ("-"), which is
+    # non-empty and therefore kept
     s2 = S("m", [], [1])
     tf.label_join([s2], "j", "-", ["a", "b"])
     assert s2.mn.get_tag_value("j") == b"-"
