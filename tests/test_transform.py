@@ -752,3 +752,42 @@ def test_label_join_empty_removes_dst():
     s2 = S("m", [], [1])
     tf.label_join([s2], "j", "-", ["a", "b"])
     assert s2.mn.get_tag_value("j") == b"-"
+
+
+def test_round_to_decimal_digits_vs_oracle():
+    # product decimal.round_to_decimal_digits vs the oracle C restatement
+    from victoriametrics_amd import decimal as vmd
+    import ctypes
+    l = oracle.lib()
+    l.vm_decimal_round_to_decimal_digits.restype = ctypes.c_double
+    l.vm_decimal_round_to_decimal_digits.argtypes = [ctypes.c_double,
+                                                     ctypes.c_int]
+    rng = np.random.default_rng(11)
+    vals = np.concatenate([
+        rng.standard_normal(200) * 10.0 ** rng.integers(-12, 12, 200),
+        [0.0, -0.0, 0.5, -0.5, 1.5, 2.5, -2.5, 0.49999999999999994,
+         -0.49999999999999994, 4503599627370495.5, math.inf, -math.inf,
+         math.nan,
+         np.uint64(0x7FF0000000000002).view if False else
+         np.frombuffer(np.uint64(0x7FF0000000000002).tobytes(),
+                       np.float64)[0]],
+    ])
+    for digits in (-101, -5, -1, 0, 1, 2, 5, 12, 50, 99, 100):
+        got = vmd.round_to_decimal_digits(vals, digits)
+        for i, v in enumerate(vals):
+            want = l.vm_decimal_round_to_decimal_digits(
+                ctypes.c_double(v), digits)
+            g, w = float(got[i]), float(want)
+            assert (np.float64(g).view(np.int64) ==
+                    np.float64(w).view(np.int64)), (v, digits, g, w)
+
+
+def test_go_round_edges():
+    from victoriametrics_amd import decimal as vmd
+    # the classic trunc(x+0.5) trap: 0.49999999999999994 rounds to 0
+    got = vmd.go_round([0.49999999999999994, -0.49999999999999994,
+                        0.5, -0.5, 1.5, 2.5, -2.5, 0.0, -0.0])
+    assert list(got[:2]) == [0.0, -0.0] or (got[0] == 0 and got[1] == 0)
+    assert list(got[2:7]) == [1.0, -1.0, 2.0, 3.0, -3.0]
+    # signed zero preserved
+    assert math.copysign(1, got[7]) == 1 and math.copysign(1, got[8]) == -1
