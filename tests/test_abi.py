@@ -229,3 +229,34 @@ def test_rollup_dispatch_covers_reference_map():
     assert callable(engine.aggr_over_time_plans)
     assert callable(engine.quantiles_over_time_plans)
     assert callable(engine.rollup_fake_plans)
+
+
+def test_finalize_rollup_metric_name():
+    # doRollupForTimeseries naming (eval.go:2009-2018) +
+    # rollupFuncsKeepMetricName (rollup.go:267-287)
+    from victoriametrics_amd import engine
+    from victoriametrics_amd.metric_name import MetricName
+    src = MetricName(b"http_requests", [(b"job", b"api")])
+    # rate resets the metric group
+    mn = engine.finalize_rollup_metric_name(src, "rate")
+    assert mn.metric_group == b"" and mn.get_tag_value("job") == b"api"
+    assert src.metric_group == b"http_requests"  # source untouched
+    # keep-list funcs keep it
+    for f in ("avg_over_time", "default_rollup", "timestamp_with_name",
+              "rollup_candlestick"):
+        assert engine.finalize_rollup_metric_name(
+            src, f).metric_group == b"http_requests", f
+    # timestamp (without _with_name) resets
+    assert engine.finalize_rollup_metric_name(
+        src, "timestamp").metric_group == b""
+    # keep_metric_names modifier overrides
+    assert engine.finalize_rollup_metric_name(
+        src, "rate", keep_metric_names=True).metric_group == b"http_requests"
+    # the rollup tag for multi-result expansions
+    mn = engine.finalize_rollup_metric_name(src, "rollup_candlestick",
+                                            rollup_tag="high")
+    assert mn.get_tag_value("rollup") == b"high"
+    # coherence: every keep-list entry is a known rollup name
+    known = (set(engine.FUNC_IDS) | set(engine.ROLLUP_FAKE_FUNCS) |
+             {"quantiles_over_time"})
+    assert engine.ROLLUP_KEEP_METRIC_NAME_FUNCS <= known

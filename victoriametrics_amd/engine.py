@@ -122,6 +122,34 @@ ROLLUP_FAKE_FUNCS = {
     "rollup_candlestick": ("none", True),
 }
 
+# rollupFuncsKeepMetricName (rollup.go:267-287): these keep the metric
+# group on the output series; everything else resets it unless the query
+# carries the keep_metric_names modifier (doRollupForTimeseries,
+# eval.go:2016-2018)
+ROLLUP_KEEP_METRIC_NAME_FUNCS = {
+    "avg_over_time", "default_rollup", "first_over_time",
+    "geomean_over_time", "hoeffding_bound_lower", "hoeffding_bound_upper",
+    "holt_winters", "iqr_over_time", "last_over_time", "max_over_time",
+    "median_over_time", "min_over_time", "mode_over_time",
+    "predict_linear", "quantile_over_time", "quantiles_over_time",
+    "rollup", "rollup_candlestick", "timestamp_with_name",
+}
+
+
+def finalize_rollup_metric_name(mn, func_name, keep_metric_names=False,
+                                rollup_tag=""):
+    """The naming step of doRollupForTimeseries (eval.go:2009-2018),
+    applied to a copy of the source MetricName: attach the rollup=<tag>
+    label for multi-result expansions and reset the metric group unless
+    the function (or the keep_metric_names modifier) keeps it."""
+    mn = mn.copy()
+    if rollup_tag:
+        mn.add_tag("rollup", rollup_tag)
+    if not keep_metric_names and \
+            func_name not in ROLLUP_KEEP_METRIC_NAME_FUNCS:
+        mn.reset_metric_group()
+    return mn
+
 
 def rollup_fake_plans(parent, start, end, step, tag="", **kwargs):
     """getRollupConfigs' expansion of the rollup_* pseudo-functions
