@@ -791,3 +791,34 @@ def test_go_round_edges():
     assert list(got[2:7]) == [1.0, -1.0, 2.0, 3.0, -3.0]
     # signed zero preserved
     assert math.copysign(1, got[7]) == 1 and math.copysign(1, got[8]) == -1
+
+
+def test_sort_by_label_numeric_exec_pins():
+    # exec_test.go:9797-10010 numericLess cases
+    def mk(vals_by_tag, tag="foo"):
+        return [S("", [(tag, v)], [i]) for i, v in enumerate(vals_by_tag)]
+
+    # string-only labels: lexicographic within equal first key
+    a = S("", [("x", "b"), ("y", "aa")], [1])
+    b = S("", [("x", "a"), ("y", "aa")], [2])
+    out = tf.sort_by_label_numeric([a, b], ["y", "x"])
+    assert out == [b, a]
+    # numeric segments: "1:0:2" < "1:0:15"
+    a = S("", [("x", "1:0:2"), ("y", "1:0:1")], [1])
+    b = S("", [("x", "1:0:15"), ("y", "1:0:1")], [2])
+    assert tf.sort_by_label_numeric([b, a], ["x", "y"]) == [a, b]
+    assert tf.sort_by_label_numeric([a, b], ["x", "y"],
+                                    desc=True) == [b, a]
+    # alias numbers with special chars
+    tss = mk(["DS50:1/0/15", "DS50:1/0/0", "DS50:1/0/1", "DS50:1/0/2"],
+             tag="a")
+    out = tf.sort_by_label_numeric(tss, ["a"])
+    assert [s.mn.get_tag_value("a") for s in out] == [
+        b"DS50:1/0/0", b"DS50:1/0/1", b"DS50:1/0/2", b"DS50:1/0/15"]
+    # the desc multi-value case feeding limit_offset (exec_test.go:9914)
+    tss = mk(["1:0:3", "5:0:15", "1:0:2", "7:0:15", "3:0:1", "1:0:2",
+              "9:0:15"])
+    out = tf.sort_by_label_numeric(tss, ["foo"], desc=True)
+    assert [s.mn.get_tag_value("foo") for s in out] == [
+        b"9:0:15", b"7:0:15", b"5:0:15", b"3:0:1", b"1:0:3", b"1:0:2",
+        b"1:0:2"]
