@@ -260,3 +260,41 @@ def test_finalize_rollup_metric_name():
     known = (set(engine.FUNC_IDS) | set(engine.ROLLUP_FAKE_FUNCS) |
              {"quantiles_over_time"})
     assert engine.ROLLUP_KEEP_METRIC_NAME_FUNCS <= known
+
+
+def test_plan_with_offset():
+    # eval.go:954-1008: grid shifted back by the offset; reported
+    # timestamps unshifted; candlestick auto-applies offset -step
+    from victoriametrics_amd import engine
+    start, end, step = 1_000_000, 1_600_000, 200_000
+    plan, ts = engine.plan_with_offset("rate", start, end, step,
+                                       offset_ms=3_600_000, window=300_000)
+    assert plan._c.start == start - 3_600_000
+    assert plan._c.end == end - 3_600_000
+    assert list(ts) == [1_000_000, 1_200_000, 1_400_000, 1_600_000]
+    # no offset: identity
+    plan, ts = engine.plan_with_offset("rate", start, end, step,
+                                       window=300_000)
+    assert plan._c.start == start and list(ts)[0] == start
+    # candlestick sub-plan: evaluation shifted FORWARD one step, reported
+    # timestamps still the original grid
+    plan, ts = engine.plan_with_offset("max_over_time", start, end, step,
+                                       parent_func="rollup_candlestick")
+    assert plan._c.start == start + step and plan._c.end == end + step
+    assert list(ts) == [1_000_000, 1_200_000, 1_400_000, 1_600_000]
+
+
+def test_aggregate_absent_over_time():
+    from victoriametrics_amd import engine
+    import math
+    import numpy as np
+    nan = math.nan
+    # rows are per-series absent_over_time outputs
+    rows = [[1.0, nan, 1.0, nan], [1.0, 1.0, nan, nan]]
+    out = engine.aggregate_absent_over_time(rows, 4)
+    v = out[0].values
+    assert v[0] == 1.0
+    assert all(math.isnan(x) for x in v[1:])
+    # no input series at all -> all 1s
+    out = engine.aggregate_absent_over_time([], 3)
+    assert list(out[0].values) == [1.0, 1.0, 1.0]

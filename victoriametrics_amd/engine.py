@@ -136,6 +136,37 @@ ROLLUP_KEEP_METRIC_NAME_FUNCS = {
 }
 
 
+def plan_with_offset(func, start, end, step, offset_ms=0, **kwargs):
+    """evalRollupFuncWithoutAt's offset handling (eval.go:954-1008):
+    `rf(m[w] offset o)` evaluates on the grid shifted BACK by the offset
+    and reports the original timestamps; rollup_candlestick additionally
+    auto-applies `offset -step` so each point covers (t-step, t].
+    Returns (plan, report_timestamps)."""
+    offset = int(offset_ms)
+    start, end = int(start) - offset, int(end) - offset
+    if func == "rollup_candlestick" or kwargs.get(
+            "parent_func") == "rollup_candlestick":
+        start += int(step)
+        end += int(step)
+        offset -= int(step)
+    plan = RollupPlan(func, start, end, int(step), **kwargs)
+    return plan, plan.timestamps() + offset
+
+
+def aggregate_absent_over_time(values_rows, n_grid, base_mn=None):
+    """aggregateAbsentOverTime (eval.go:1012-1031): collapse the
+    per-series absent_over_time rollup rows (1 where the series had no
+    samples in the window, NaN where it had) into ONE series that is NaN
+    wherever ANY input series was present."""
+    from .binary_op import Series
+    from .metric_name import MetricName
+    mn = base_mn.copy() if base_mn is not None else MetricName()
+    vals = np.ones(n_grid)
+    for row in values_rows:
+        vals[np.isnan(np.asarray(row, np.float64))] = np.nan
+    return [Series(mn, vals)]
+
+
 def finalize_rollup_metric_name(mn, func_name, keep_metric_names=False,
                                 rollup_tag=""):
     """The naming step of doRollupForTimeseries (eval.go:2009-2018),
