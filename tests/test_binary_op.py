@@ -374,3 +374,124 @@ def test_merge_non_overlapping_limits():
     e = Series(MetricName("a"), np.asarray([1.0, NAN]))
     f = Series(MetricName("a"), np.asarray([NAN, 2.0]))
     assert not merge_non_overlapping(e, f)
+
+
+# ---------------------------------------------------------------------------
+# TestExecSuccess vector-matching pins (exec_test.go:3629-3810), CPU via
+# the oracle apply harness.  TIME = [1000..2000]; left = time()/10 with
+# {foo=bar, xx=yy, __name__=qwert}.
+# ---------------------------------------------------------------------------
+
+TIME_V = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+
+
+def _left_qwert():
+    return S("qwert", [("foo", "bar"), ("xx", "yy")], TIME_V / 10)
+
+
+def _tagmap(tss):
+    out = {}
+    for t in tss:
+        key = (t.mn.metric_group,
+               tuple(sorted((k, v) for k, v in t.mn.tags)))
+        assert key not in out
+        out[key] = t.values
+    return out
+
+
+def test_exec_group_left_additional_tag():
+    # + on(foo) group_left(op): two right series differ by op and do not
+    # overlap in time -> two outputs, op copied, left tags kept
+    lo = np.where(TIME_V < 1400, TIME_V, NAN)
+    hi = np.where(TIME_V >= 1400, TIME_V, NAN)
+    right = [S("", [("foo", "bar"), ("op", "le")], lo),
+             S("", [("foo", "bar"), ("op", "ge")], hi)]
+    out = _eval(BinOpSpec("+", group_op="on", group_tags=["foo"],
+                          join_op="group_left", join_tags=["op"]),
+                [_left_qwert()], right)
+    m = _tagmap(out)
+    k_le = (b"", ((b"foo", b"bar"), (b"op", b"le"), (b"xx", b"yy")))
+    k_ge = (b"", ((b"foo", b"bar"), (b"op", b"ge"), (b"xx", b"yy")))
+    assert set(m) == {k_le, k_ge}
+    np.testing.assert_array_equal(
+        m[k_le].view(np.int64),
+        np.asarray([1100.0, 1320, NAN, NAN, NAN, NAN]).view(np.int64))
+    np.testing.assert_array_equal(
+        m[k_ge].view(np.int64),
+        np.asarray([NAN, NAN, 1540.0, 1760, 1980, 2200]).view(np.int64))
+
+
+def test_exec_on_duplicate_nonoverlapping_merged():
+    # + on(foo) with NO join modifier: the two non-overlapping right
+    # series merge into one (mergeNonOverlappingTimeseries); the output
+    # name is the on() projection {foo}
+    lo = np.where(TIME_V < 1400, TIME_V, NAN)
+    hi = np.where(TIME_V >= 1400, TIME_V, NAN)
+    right = [S("", [("foo", "bar"), ("op", "le")], lo),
+             S("", [("foo", "bar"), ("op", "ge")], hi)]
+    out = _eval(BinOpSpec("+", group_op="on", group_tags=["foo"]),
+                [_left_qwert()], right)
+    assert len(out) == 1
+    assert out[0].mn.metric_group == b""
+    assert sorted(out[0].mn.tags) == [(b"foo", b"bar")]
+    np.testing.assert_allclose(out[0].values,
+                               [1100, 1320, 1540, 1760, 1980, 2200])
+
+
+def test_exec_group_left_empty_duplicate_nonoverlapping():
+    # group_left() keeps the left extra tags (xx=yy) while merging
+    lo = np.where(TIME_V < 1400, TIME_V, NAN)
+    hi = np.where(TIME_V >= 1400, TIME_V, NAN)
+    right = [S("", [("foo", "bar"), ("op", "le")], lo),
+             S("", [("foo", "bar"), ("op", "ge")], hi)]
+    out = _eval(BinOpSpec("+", group_op="on", group_tags=["foo"],
+                          join_op="group_left"),
+                [_left_qwert()], right)
+    assert len(out) == 1
+    assert sorted(out[0].mn.tags) == [(b"foo", b"bar"), (b"xx", b"yy")]
+    np.testing.assert_allclose(out[0].values,
+                               [1100, 1320, 1540, 1760, 1980, 2200])
+
+
+def test_exec_group_left_copies_name():
+    # group_left(__name__) copies the right metric group onto the result
+    right = [S("aaa", [("foo", "bar")], TIME_V.copy())]
+    out = _eval(BinOpSpec("+", group_op="on", group_tags=["foo"],
+                          join_op="group_left", join_tags=["__name__"]),
+                [_left_qwert()], right)
+    assert len(out) == 1
+    assert out[0].mn.metric_group == b"aaa"
+    assert sorted(out[0].mn.tags) == [(b"foo", b"bar"), (b"xx", b"yy")]
+    np.testing.assert_allclose(out[0].values,
+                               [1100, 1320, 1540, 1760, 1980, 2200])
+
+
+def test_exec_group_right_copies_left_tag():
+    # + on(foo) group_right(xx): many right series, xx copied from left
+    right = [S("aaa", [("foo", "bar")], TIME_V.copy()),
+             S("yyy", [("foo", "bar"), ("ppp", "123")], TIME_V + 3)]
+    out = _eval(BinOpSpec("+", group_op="on", group_tags=["foo"],
+                          join_op="group_right", join_tags=["xx"]),
+                [_left_qwert()], right)
+    m = _tagmap(out)
+    k1 = (b"", ((b"foo", b"bar"), (b"xx", b"yy")))
+    k2 = (b"", ((b"foo", b"bar"), (b"ppp", b"123"), (b"xx", b"yy")))
+    assert set(m) == {k1, k2}
+    np.testing.assert_allclose(m[k1], [1100, 1320, 1540, 1760, 1980, 2200])
+    np.testing.assert_allclose(m[k2], [1103, 1323, 1543, 1763, 1983, 2203])
+
+
+def test_exec_on_empty_group_left_scalar_ish():
+    # (a or b) * on() group_left 2: on() makes one join key; the scalar
+    # side is a plain series with no tags
+    left = [S("", [("foo", "bar")], TIME_V.copy()),
+            S("", [("foo", "qwert")], np.full(6, 10.0))]
+    right = [S("", [], np.full(6, 2.0))]
+    out = _eval(BinOpSpec("*", group_op="on", group_tags=[],
+                          join_op="group_left"), left, right)
+    m = _tagmap(out)
+    k1 = (b"", ((b"foo", b"bar"),))
+    k2 = (b"", ((b"foo", b"qwert"),))
+    assert set(m) == {k1, k2}
+    np.testing.assert_allclose(m[k1], TIME_V * 2)
+    np.testing.assert_allclose(m[k2], [20.0] * 6)
