@@ -402,3 +402,53 @@ class TestHistogramExecPins:
         got = _hist("histogram_fraction", _buckets(("200", 100), ("55", 0)),
                     (55, 105))
         _exact(got[0].values, [0.3448275862068966] * 6)
+
+
+def test_running_range_funcs_with_gaps():
+    # exec_test.go masked running/range cases: NaN gating inside the
+    # running scans and range reductions
+    nan = math.nan
+    # running_min(abs(1500-time()) < 400 > 100) -> [nan,300,300,300,300,300]
+    v = np.abs(1500.0 - tmat())
+    v[(v >= 400) | (v <= 100)] = nan
+    _exact(tfm.transform("running_min", v), [nan, 300, 300, 300, 300, 300])
+    # running_max(abs(1300-time()) > 300 < 700) -> [nan]*4 + [500, 500]
+    v = np.abs(1300.0 - tmat())
+    v[(v <= 300) | (v >= 700)] = nan
+    _exact(tfm.transform("running_max", v), [nan, nan, nan, nan, 500, 500])
+    # running_sum(time()/1e3 > 1.2 < 1.8) -> [nan, nan, 1.4, 3, 3, 3]
+    v = tmat() / 1e3
+    v[(v <= 1.2) | (v >= 1.8)] = nan
+    _exact(tfm.transform("running_sum", v), [nan, nan, 1.4, 3, 3, 3])
+    # range_* over time() > 1200 < 1800 = [nan,nan,1400,1600,nan,nan]
+    def gated():
+        v = tmat()
+        v[(v <= 1200) | (v >= 1800)] = nan
+        return v
+    _exact(tfm.transform("range_max", gated()), [1600.0] * 6)
+    _exact(tfm.transform("range_sum", gated()), [3000.0] * 6)
+    _exact(tfm.transform("range_last", gated()), [1600.0] * 6)
+    _exact(tfm.transform("range_first", gated()), [1400.0] * 6)
+
+
+def test_range_linear_regression_exec():
+    ts = GRID_MS.copy()
+    _exact(tfm.transform("range_linear_regression", tmat(), ts=ts),
+           [1000.0, 1200, 1400, 1600, 1800, 2000])
+    _exact(tfm.transform("range_linear_regression", -tmat(), ts=ts),
+           [-1000.0, -1200, -1400, -1600, -1800, -2000])
+
+
+def test_range_stddev_stdvar_exec():
+    # round(range_stddev(time()), 0.01) etc.
+    def rounded(name, v):
+        out = tfm.transform(name, v)
+        return tfm.transform("round", out, args=[np.full(6, 0.01)])
+    _exact(rounded("range_stddev", tmat()), [341.57] * 6)
+    _exact(rounded("range_stdvar", tmat()), [116666.67] * 6)
+    v = tmat()
+    v[(v <= 1200) | (v >= 1800)] = math.nan
+    _exact(rounded("range_stddev", v), [100.0] * 6)
+    v = tmat()
+    v[(v <= 1200) | (v >= 1800)] = math.nan
+    _exact(rounded("range_stdvar", v), [10000.0] * 6)
