@@ -298,3 +298,15 @@ def test_aggregate_absent_over_time():
     # no input series at all -> all 1s
     out = engine.aggregate_absent_over_time([], 3)
     assert list(out[0].values) == [1.0, 1.0, 1.0]
+
+
+def test_incremental_aggr_callbacks_covered():
+    # incrementalAggrFuncCallbacksMap (aggr_incremental.go:18-66): the 9
+    # incremental aggregates; 8 run fused in the rollup kernels
+    # (AGGR_IDS), `any` via the representative-group-id rewrite
+    from victoriametrics_amd import engine
+    reference = {"sum", "min", "max", "avg", "count", "sum2", "geomean",
+                 "any", "group"}
+    fused = set(engine.AGGR_IDS) - {"none"}
+    assert reference - fused == {"any"}
+    assert callable(engine.any_representative_group_ids)
