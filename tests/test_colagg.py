@@ -346,3 +346,25 @@ def test_quantiles_plural():
     assert len(out) == 4
     labels = sorted(set(s.mn.get_tag_value("phi") for s in out))
     assert labels == [b"0.25", b"0.75"]
+
+
+def test_format_go_float_pins():
+    # count_values label formatting: strconv.FormatFloat(v, 'f', -1, 64)
+    # shortest round-trip positional form (aggr.go count_values)
+    from victoriametrics_amd.aggregate import format_go_float as f
+    assert f(1.0) == "1"
+    assert f(0.1) == "0.1"
+    assert f(-2.5) == "-2.5"
+    assert f(1e-9) == "0.000000001"
+    assert f(123456789.0) == "123456789"
+    assert f(1e21) == "1000000000000000000000"
+    assert f(0.30000000000000004) == "0.30000000000000004"
+    assert f(float("inf")) == "+Inf"
+    assert f(float("-inf")) == "-Inf"
+    assert f(float("nan")) == "NaN"
+    assert f(0.0) == "0"
+    # round-trip property on random values
+    import numpy as np
+    rng = np.random.default_rng(2)
+    for v in rng.standard_normal(200) * 10.0 ** rng.integers(-15, 15, 200):
+        assert float(f(float(v))) == float(v), v
