@@ -822,3 +822,30 @@ def test_sort_by_label_numeric_exec_pins():
     assert [s.mn.get_tag_value("foo") for s in out] == [
         b"9:0:15", b"7:0:15", b"5:0:15", b"3:0:1", b"1:0:3", b"1:0:2",
         b"1:0:2"]
+
+
+def test_prometheus_buckets_overlapped_ranges_exec_pin():
+    # exec_test.go:5661: overlapping vmranges accumulate cumulatively in
+    # le order, with the zero-width "0...0" bucket first
+    t20 = TIME6 / 20
+    t100 = TIME6 / 100
+    t10 = TIME6 / 10
+    series = [
+        S("xxx", [("foo", "bar"), ("vmrange", "0...0")], np.full(6, 90.0)),
+        S("xxx", [("foo", "bar"), ("vmrange", "0...0.2")], t20),
+        S("xxx", [("foo", "bar"), ("vmrange", "0.2...0.25")], t20),
+        S("xxx", [("foo", "bar"), ("vmrange", "0...0.26")], t20),
+        S("xxx", [("foo", "bar"), ("vmrange", "0.2...40")], t100),
+        S("xxx", [("foo", "bar"), ("vmrange", "40...Inf")], t10),
+    ]
+    out = tf.prometheus_buckets(series)
+    got = {s.mn.get_tag_value("le").decode(): list(s.values) for s in out}
+    assert got["0"] == [90.0] * 6
+    assert got["0.2"] == [140, 150, 160, 170, 180, 190]
+    assert got["0.25"] == [190, 210, 230, 250, 270, 290]
+    assert got["0.26"] == [240, 270, 300, 330, 360, 390]
+    assert got["40"] == [250, 282, 314, 346, 378, 410]
+    # the end string is kept VERBATIM ("Inf", not "+Inf") — the +Inf
+    # synthetic bucket is only added when the last range is finite
+    assert got["Inf"] == [350, 402, 454, 506, 558, 610]
+    assert all(s.mn.metric_group == b"xxx" for s in out)
