@@ -344,3 +344,58 @@ def test_subquery_grid_parameters():
     except RuntimeError:
         pass
     assert captured["grid"][2] == step
+
+
+def test_cache_get_best_key_fuzz_vs_model():
+    """Model-based check of GetSeries entry selection: random cached
+    windows on a shared step grid; the returned slice must match a brute
+    force over all stored entries using GetBestKey's rule (max usable
+    span d = min(end, e.end) - start among entries with e.start <= start,
+    later entries winning ties)."""
+    rng = np.random.default_rng(21)
+    step = 100
+    for trial in range(50):
+        c = cache.RollupResultCache()
+        entries = []
+        for _ in range(int(rng.integers(1, 7))):
+            s = int(rng.integers(0, 40)) * step
+            e = s + int(rng.integers(1, 30)) * step
+            ts = np.arange(s, e + 1, step, dtype=np.int64)
+            vals = rng.random((1, len(ts)))
+            before = len(c._meta.get(c._key("q", 0, step), []) or [])
+            c.put_series("q", 0, step, [(b"", ())], vals, ts,
+                         now_ms=FUTURE)
+            after = len(c._meta.get(c._key("q", 0, step), []) or [])
+            covered = any(s >= es and e <= ee for es, ee, _ in entries)
+            if after > before:
+                entries.append((s, e, (ts, vals)))
+            else:
+                assert covered, (trial, s, e, entries)
+        start = int(rng.integers(0, 60)) * step
+        end = start + int(rng.integers(1, 30)) * step
+        n, v, t, ns = c.get_series("q", 0, step, start, end)
+        # brute-force GetBestKey
+        best, d_max = None, 0
+        for es, ee, data in entries:
+            if start < es:
+                continue
+            d = (end if end <= ee else ee) - start
+            if d >= d_max:
+                d_max, best = d, (es, ee, data)
+        if best is None:
+            assert n is None and ns == start, trial
+            continue
+        es, ee, (ts_b, vals_b) = best
+        i = int(np.searchsorted(ts_b, start))
+        if i == len(ts_b) or ts_b[i] != start:
+            assert n is None and ns == start, trial
+            continue
+        j = int(np.searchsorted(ts_b, end, side="right"))
+        if j <= i:
+            assert n is None and ns == start, trial
+            continue
+        assert n is not None, (trial, best, start, end)
+        np.testing.assert_array_equal(t, ts_b[i:j])
+        np.testing.assert_array_equal(v.view(np.int64),
+                                      vals_b[:, i:j].view(np.int64))
+        assert ns == int(ts_b[j - 1]) + step
