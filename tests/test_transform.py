@@ -849,3 +849,32 @@ def test_prometheus_buckets_overlapped_ranges_exec_pin():
     # synthetic bucket is only added when the last range is finite
     assert got["Inf"] == [350, 402, 454, 506, 558, 610]
     assert all(s.mn.metric_group == b"xxx" for s in out)
+
+
+def test_absent_exec_pins():
+    # absent(time() > 1500) -> [1,1,1,nan,nan,nan] (exec_test.go:1099)
+    masked = TIME6.copy()
+    masked[TIME6 <= 1500] = NAN
+    out = tf.absent([S("", [], masked)], 6)
+    got = out[0].values
+    assert list(got[:3]) == [1.0, 1.0, 1.0]
+    assert all(math.isnan(x) for x in got[3:])
+    # absent over a fully-present series -> all NaN (removed at Exec)
+    out = tf.absent([S("", [], TIME6.copy())], 6)
+    assert all(math.isnan(x) for x in out[0].values)
+    # absent of nothing -> all 1s
+    out = tf.absent([], 6)
+    assert list(out[0].values) == [1.0] * 6
+
+
+def test_absent_over_time_multi_ts_combination():
+    # absent_over_time(multi-ts) (exec_test.go:1084): the per-series
+    # absent rows combine to 1 only where EVERY series was absent
+    from victoriametrics_amd import engine
+    one = [NAN, NAN, 1.0, 1, 1, 1]     # present early -> absent later
+    two = [1.0, 1, 1, 1, NAN, NAN]     # present late  -> absent early
+    out = engine.aggregate_absent_over_time([one, two], 6)
+    got = out[0].values
+    assert math.isnan(got[0]) and math.isnan(got[1])
+    assert got[2] == 1.0 and got[3] == 1.0
+    assert math.isnan(got[4]) and math.isnan(got[5])
