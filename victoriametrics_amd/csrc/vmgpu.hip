@@ -31,6 +31,7 @@
 #include <hip/hip_runtime.h>
 #include <algorithm>
 #include <atomic>
+#include <cstdlib>
 #include <cstring>
 #include <cstdio>
 #include <map>
@@ -60,6 +61,14 @@ static VM_DEV void wave_lds_sync() {
    * kernel's global-scratch variant (no workgroup barrier needed).  Used
    * between phases only, never inside hot loops. */
   asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+}
+
+static VM_DEV void wave_ds_sync() {
+  /* LDS-only phase barrier for one wavefront: DS ops (including the
+   * ds_bpermute behind __shfl) are lgkm-counted, so this gives cross-lane
+   * LDS visibility while deliberately leaving prefetched global loads
+   * (vmcnt) in flight — the point of the software-pipelined kernel. */
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 }
 
 static VM_DEV int vm_upper_bound(const int64_t* ts, int n, int64_t seek) {
@@ -543,9 +552,10 @@ static VM_DEV void pre_func_wave(const int64_t* d_ts, double* d_vals,
 /* getScrapeInterval (rollup.go:871-897): 0.6 quantile of the last <=20
  * sample gaps, computed wave-cooperatively with a rank-based selection
  * (identical result to sort + quantileSorted). */
-static __device__ int64_t scrape_interval_wave(const int64_t* d_ts, int count,
-                                               int64_t default_interval, int lane,
-                                               double* scratch) {
+template <bool LDS_ONLY>
+static __device__ int64_t scrape_interval_wave_t(const int64_t* d_ts, int count,
+                                                 int64_t default_interval, int lane,
+                                                 double* scratch) {
   if (count < 2) return default_interval;
   int cnt = count - 1;
   if (cnt > 20) cnt = 20;
@@ -553,7 +563,7 @@ static __device__ int64_t scrape_interval_wave(const int64_t* d_ts, int count,
   double gap = 0.0;
   if (active) gap = (double)(d_ts[count - 1 - lane] - d_ts[count - 2 - lane]);
   if (active) scratch[lane] = gap;
-  wave_lds_sync();
+  if (LDS_ONLY) wave_ds_sync(); else wave_lds_sync();
   /* rank of this lane's gap among the cnt gaps (ties broken by lane);
    * scratch broadcast keeps the cnt probes independent (one LDS round trip)
    * instead of cnt serial shuffles */
@@ -577,6 +587,12 @@ static __device__ int64_t scrape_interval_wave(const int64_t* d_ts, int count,
   int64_t si = (int64_t)q;
   if (si <= 0) return default_interval;
   return si;
+}
+
+static __device__ int64_t scrape_interval_wave(const int64_t* d_ts, int count,
+                                               int64_t default_interval, int lane,
+                                               double* scratch) {
+  return scrape_interval_wave_t<false>(d_ts, count, default_interval, lane, scratch);
 }
 
 static VM_DEV int64_t max_prev_interval_tiers(int64_t si) {
@@ -1037,6 +1053,298 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
   }
   acc_flush();
   /* reduce samplesScanned: wave shuffle + one atomic per wave */
+  for (int d = 32; d > 0; d >>= 1) scanned += __shfl_down((unsigned long long)scanned, d);
+  if (lane == 0 && scanned) atomicAdd(io.samples_scanned, (unsigned long long)scanned);
+}
+
+/* ------------------------------------------------------------------ */
+/* kernel 1b: software-pipelined register-staged wave kernel          */
+/*                                                                    */
+/* The wave kernel above is instruction-ISSUE/latency bound, not HBM  */
+/* bound (profiles/round1_final.md §5: FETCH/WRITE ≈ algorithmic but  */
+/* ACTIVE_INST_ANY ≈ 30% at 7 waves/SIMD).  This variant restructures */
+/* the per-series phases so the memory system never drains:           */
+/*   - the series' columns are loaded straight into REGISTERS (up to  */
+/*     PIPE_CHUNKS coalesced b64 loads per column in flight at once), */
+/*   - the removeCounterResets scan consumes those registers while    */
+/*     writing the corrected columns to LDS (no LDS re-read pass),    */
+/*   - the NEXT series' loads are issued as soon as the registers die,*/
+/*     overlapping scrape/seek/eval/emit of the current series,       */
+/*   - every phase barrier is lgkm-only (wave_ds_sync), so the        */
+/*     prefetched loads stay outstanding across the whole evaluation. */
+/* Results are bit-identical to the wave kernel: same scan, same      */
+/* j-cache seeks, same evaluators.                                    */
+/* ------------------------------------------------------------------ */
+
+#define PIPE_CHUNKS 4 /* series cap = PIPE_CHUNKS*WAVE = 256 samples */
+
+/* removeCounterResets over register-staged chunks: the same exact-order
+ * scan as rcr_scan_wave (rollup.go:921-958), inputs from registers, fused
+ * with the LDS store.  Returns n, or -1 when a stale NaN must be compacted
+ * out (caller falls back to load_compact_wave + rcr_scan_wave). */
+static VM_DEV int rcr_scan_regs(const int64_t* rt, const double* rv,
+                                int64_t n, int64_t* d_ts, double* d_vals,
+                                bool drop_stale, int64_t msi, int lane) {
+  double corr = 0.0;
+  double prev_raw = 0.0;
+  int64_t prev_ts = 0;
+  double prev_fin = 0.0;
+#pragma unroll
+  for (int c = 0; c < PIPE_CHUNKS; c++) {
+    int64_t base = (int64_t)c * WAVE;
+    if (base >= n) continue;
+    int64_t k = base + lane;
+    bool active = k < n;
+    double v = active ? rv[c] : 0.0;
+    int64_t t = active ? rt[c] : 0;
+    if (drop_stale && __ballot(active && vm_is_stale_nan(v)) != 0) return -1;
+    if (active) d_ts[k] = t;
+    double pv = __shfl_up(v, 1);
+    int64_t pt = __shfl_up(t, 1);
+    if (lane == 0) { pv = prev_raw; pt = prev_ts; }
+    bool isfirst = (k == 0);
+    double d = v - pv;
+    double inc = 0.0;
+    if (!isfirst && d < 0) inc = ((-d * 8) < pv) ? (pv - v) : pv;
+    bool gap = (!isfirst && msi > 0 && (t - pt) > msi);
+    uint64_t em = __ballot(active && (gap || inc != 0.0));
+    uint64_t dm = __ballot(active && !isfirst && d < 0);
+    int last = (int)(n - base - 1);
+    if (last > 63) last = 63;
+    if (em == 0 && dm == 0 &&
+        (base == 0 || __shfl(v, 0) + corr >= prev_fin)) {
+      if (active) d_vals[k] = v + corr;
+      prev_raw = __shfl(v, last);
+      prev_ts = __shfl(t, last);
+      prev_fin = prev_raw + corr;
+      continue;
+    }
+    double cc = corr;
+    double mycorr = corr;
+    while (em) {
+      int b = __ffsll((unsigned long long)em) - 1;
+      em &= em - 1;
+      double ib = __shfl(inc, b);
+      int gb = __shfl((int)gap, b);
+      double cn = gb ? 0.0 : (cc + ib);
+      if (lane >= b) mycorr = cn;
+      cc = cn;
+    }
+    double fin = v + mycorr;
+    bool bnd = isfirst || gap;
+    double x = fin;
+    int f = bnd ? 1 : 0;
+    if (lane == 0 && !bnd) x = fmax(x, prev_fin);
+    for (int dlt = 1; dlt < WAVE; dlt <<= 1) {
+      double xo = __shfl_up(x, dlt);
+      int fo = __shfl_up(f, dlt);
+      if (lane >= dlt) {
+        if (!f) x = fmax(x, xo);
+        f = f | fo;
+      }
+    }
+    if (active) d_vals[k] = x;
+    corr = cc;
+    prev_raw = __shfl(v, last);
+    prev_ts = __shfl(t, last);
+    prev_fin = __shfl(x, last);
+  }
+  return (int)n;
+}
+
+template <int FUNC_CT>
+__global__ __launch_bounds__(BLOCK_THREADS) void rollup_pipe_kernel(KPlan p, KIO io) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int wave_in_block = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const size_t jbuf_bytes = vm_jbuf_bytes(p.jbuf_mode, p.jbuf_elems);
+  const size_t wave_bytes = (size_t)p.chunk_wave * 16 + 256 + jbuf_bytes;
+  int64_t* lts = (int64_t*)(smem + (size_t)wave_in_block * wave_bytes);
+  double* lvs = (double*)(smem + (size_t)wave_in_block * wave_bytes +
+                          (size_t)p.chunk_wave * 8);
+  double* lscratch = (double*)(smem + (size_t)wave_in_block * wave_bytes +
+                               (size_t)p.chunk_wave * 16);
+  uint16_t* jbuf = (uint16_t*)(smem + (size_t)wave_in_block * wave_bytes +
+                               (size_t)p.chunk_wave * 16 + 256);
+  uint64_t scanned = 0;
+  const uint32_t wave_id = blockIdx.x * WAVES_PER_BLOCK + wave_in_block;
+  const uint32_t wave_stride = gridDim.x * WAVES_PER_BLOCK;
+
+  int64_t rt[PIPE_CHUNKS];
+  double rv[PIPE_CHUNKS];
+
+  /* wave_id is wave-uniform by construction; readfirstlane makes that
+   * provable, so the per-series descriptor reads below become scalar
+   * (s_load, lgkm-counted) instead of divergent vector loads, and the
+   * staging loads below are unconditional with clamped indices — one
+   * basic block, so the compiler can track outstanding vmcnt precisely
+   * instead of falling back to vmcnt(0) at every use. */
+  uint32_t ws = __builtin_amdgcn_readfirstlane(wave_id);
+  uint32_t s = 0;
+  uint64_t lo = 0;
+  int64_t n = 0;
+  if (ws < io.n_sel) {
+    s = __builtin_amdgcn_readfirstlane(io.series_sel ? io.series_sel[ws] : ws);
+    lo = io.offsets[s];
+    n = (int64_t)(io.offsets[s + 1] - lo);
+    if (n > 0) {
+      const int64_t* gts = io.ts + lo;
+      const double* gvs = io.vals + lo;
+      const int64_t nm1 = n - 1;
+#pragma unroll
+      for (int c = 0; c < PIPE_CHUNKS; c++) {
+        int64_t k = (int64_t)c * WAVE + lane;
+        if (k > nm1) k = nm1;
+        rt[c] = gts[k];
+        rv[c] = gvs[k];
+      }
+    }
+  }
+  while (ws < io.n_sel) {
+    const uint32_t ws_n = ws + wave_stride;
+    const bool has_next = ws_n < io.n_sel;
+    uint32_t s_n = 0;
+    uint64_t lo_n = 0;
+    int64_t n_n = 0;
+    if (has_next) {
+      s_n = __builtin_amdgcn_readfirstlane(
+          io.series_sel ? io.series_sel[ws_n] : ws_n);
+      lo_n = io.offsets[s_n];
+      n_n = (int64_t)(io.offsets[s_n + 1] - lo_n);
+    }
+    /* stage the current series into LDS from registers (rcr fused) */
+    int count = -1;
+    if (p.rcr) {
+      count = rcr_scan_regs(rt, rv, n, lts, lvs, p.drop_stale != 0,
+                            p.max_staleness, lane);
+    } else {
+      bool stale = false;
+      if (p.drop_stale) {
+#pragma unroll
+        for (int c = 0; c < PIPE_CHUNKS; c++) {
+          int64_t k = (int64_t)c * WAVE + lane;
+          if (__ballot(k < n && vm_is_stale_nan(rv[c])) != 0) stale = true;
+        }
+      }
+      if (!stale) {
+#pragma unroll
+        for (int c = 0; c < PIPE_CHUNKS; c++) {
+          int64_t k = (int64_t)c * WAVE + lane;
+          if (k < n) { lts[k] = rt[c]; lvs[k] = rv[c]; }
+        }
+        count = (int)n;
+      }
+    }
+    /* the staging registers are dead now — prefetch the NEXT series into
+     * them; these loads stay in flight across scrape/seek/eval/emit */
+    if (has_next && n_n > 0) {
+      const int64_t* gts = io.ts + lo_n;
+      const double* gvs = io.vals + lo_n;
+      const int64_t nm1 = n_n - 1;
+#pragma unroll
+      for (int c = 0; c < PIPE_CHUNKS; c++) {
+        int64_t k = (int64_t)c * WAVE + lane;
+        if (k > nm1) k = nm1;
+        rt[c] = gts[k];
+        rv[c] = gvs[k];
+      }
+    }
+    if (count < 0) {
+      /* stale NaNs must be compacted out (rare): redo from global */
+      count = load_compact_wave(io.ts + lo, io.vals + lo, n, lts, lvs,
+                                p.drop_stale != 0, lane);
+      wave_ds_sync();
+      if (p.rcr) rcr_scan_wave(lts, lvs, count, p.max_staleness, lane);
+    }
+    wave_ds_sync();
+
+    int64_t si = p.step;
+#ifndef VMGPU_ABL_NO_SCRAPE
+    if (p.start < p.end)
+      si = scrape_interval_wave_t<true>(lts, count, p.step, lane, lscratch);
+#endif
+    SeriesWindow sw = series_window(p, si);
+    if (lane == 0) scanned += (uint64_t)count;
+
+    double idx_per_ms = 0.0;
+    int64_t ts0 = 0;
+    if (count > 1) {
+      ts0 = lts[0];
+      int64_t span_ms = lts[count - 1] - ts0;
+      idx_per_ms = span_ms > 0 ? (double)(count - 1) / (double)span_ms : 0.0;
+    }
+    bool done = false;
+    if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
+      int dg64 = (sw.window > 0 && p.step > 0 && sw.window % p.step == 0)
+                     ? (int)(sw.window / p.step) : 0;
+      if (p.jbuf_mode >= 1 &&
+          (size_t)p.n_grid * 2 <= vm_jbuf_bytes(p.jbuf_mode, p.jbuf_elems) &&
+          dg64 > 0 && count <= 65535) {
+        for (int g = lane; g < p.n_grid; g += WAVE) {
+          int64_t t_end = p.start + (int64_t)g * p.step;
+          int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
+          jbuf[g] = (uint16_t)vm_ub_hint_fast(lts, count, t_end, gj);
+        }
+        wave_ds_sync();
+        for (int g0 = 0; g0 < p.n_grid; g0 += 4 * WAVE) {
+#pragma unroll
+          for (int u = 0; u < 4; u++) {
+            int g = g0 + u * WAVE + lane;
+            if (g < p.n_grid) {
+              int64_t t_end = p.start + (int64_t)g * p.step;
+              int64_t t_start = t_end - sw.window;
+              int j = jbuf[g];
+              int i;
+              if (g >= dg64) {
+                i = jbuf[g - dg64];
+              } else {
+                int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
+                i = vm_ub_hint_fast(lts, count, t_start, gi);
+              }
+              vm_emit_value(p, io, s, g,
+                            eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
+              scanned += 2;
+            }
+          }
+        }
+        done = true;
+      }
+    }
+    if (!done) {
+      for (int g0 = 0; g0 < p.n_grid; g0 += 4 * WAVE) {
+#pragma unroll
+        for (int u = 0; u < 4; u++) {
+          int g = g0 + u * WAVE + lane;
+          if (g < p.n_grid) {
+            int64_t t_end = p.start + (int64_t)g * p.step;
+            int64_t t_start = t_end - sw.window;
+            int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
+            int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
+            int i, j;
+            if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
+              i = vm_ub_hint_fast(lts, count, t_start, gi);
+              j = vm_ub_hint_fast(lts, count, t_end, gj);
+            } else {
+              i = vm_ub_hint(lts, count, t_start, gi);
+              j = vm_ub_hint(lts, count, t_end, gj);
+            }
+            if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
+              vm_emit_value(p, io, s, g,
+                            eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
+              scanned += 2;
+            } else {
+              scanned += eval_grid_point_ij<FUNC_CT>(p, sw, lts, lvs, count, g, s, io, i, j);
+            }
+          }
+        }
+      }
+    }
+    wave_ds_sync();
+    ws = ws_n;
+    s = s_n;
+    lo = lo_n;
+    n = n_n;
+  }
   for (int d = 32; d > 0; d >>= 1) scanned += __shfl_down((unsigned long long)scanned, d);
   if (lane == 0 && scanned) atomicAdd(io.samples_scanned, (unsigned long long)scanned);
 }
@@ -1803,6 +2111,16 @@ static void launch_rollup_t(int which, uint32_t blocks, size_t lds,
   } else if (which == 1) {
     hipLaunchKernelGGL(rollup_block_kernel<FUNC_CT>, dim3(blocks),
                        dim3(BLOCK_THREADS), lds, stream, p, w);
+  } else if (which == 3) {
+    /* software-pipelined register-staged wave variant (series <= 256
+     * samples); instantiated only for the hot specializations it is
+     * measured on — everything else falls back to the wave kernel */
+    if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
+      hipLaunchKernelGGL(rollup_pipe_kernel<FUNC_CT>, dim3(blocks),
+                         dim3(BLOCK_THREADS), lds, stream, p, w);
+    } else {
+      launch_rollup_t<FUNC_CT>(0, blocks, lds, p, w, stream);
+    }
   } else {
     hipLaunchKernelGGL(rollup_huge_kernel<FUNC_CT>, dim3(blocks),
                        dim3(BLOCK_THREADS), lds, stream, p, w);
@@ -2414,7 +2732,22 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
     size_t lds = (size_t)WAVES_PER_BLOCK *
                  ((size_t)p.chunk_wave * 16 + 256 +
                   vm_jbuf_bytes(p.jbuf_mode, p.jbuf_elems));
-    launch_rollup(0, blocks, lds, p, w);
+    /* software-pipelined register-staged variant for short series (the
+     * config-2 flagship shape); VMGPU_DISABLE_PIPE=1 A/Bs the wave kernel */
+    int which = 0;
+    if (b.max_wave_len <= (uint32_t)(PIPE_CHUNKS * WAVE) && p.pre_func == 0 &&
+        (p.func == VMF_RATE || p.func == VMF_DERIV_FAST)) {
+      static int pipe_disabled = -1;
+      if (pipe_disabled < 0) {
+        const char* e = getenv("VMGPU_DISABLE_PIPE");
+        pipe_disabled = (e && e[0] == '1') ? 1 : 0;
+      }
+#ifdef VMGPU_ABL_GACC
+      pipe_disabled = 1; /* the GACC A/B runs through the wave kernel */
+#endif
+      if (!pipe_disabled) which = 3;
+    }
+    launch_rollup(which, blocks, lds, p, w);
   }
   if (b.n_block) {
     KIO w = io;
