@@ -219,6 +219,10 @@ struct KIO {
   const uint32_t* series_sel = nullptr; /* list of series ids for this kernel */
   uint32_t n_sel = 0;
   const int32_t* group_ids = nullptr;   /* may be null */
+  const int64_t* series_si = nullptr;   /* batch scrape-interval index (raw
+                                           0.6-quantile per series; 0 = use
+                                           the plan default).  Valid only
+                                           when no samples were dropped. */
   double* out = nullptr;                /* per-series [n_series x n_grid] or group values */
   double* out_counts = nullptr;         /* group counts */
   unsigned long long* samples_scanned = nullptr;
@@ -595,6 +599,30 @@ static __device__ int64_t scrape_interval_wave(const int64_t* d_ts, int count,
   return scrape_interval_wave_t<false>(d_ts, count, default_interval, lane, scratch);
 }
 
+/* Batch scrape-interval index: getScrapeInterval's 0.6-quantile is a pure
+ * function of the series' immutable resident timestamps, so it is computed
+ * ONCE at batch creation (like the length-partition lists) and each query
+ * reads it instead of re-deriving it per series.  Stored RAW (0 = "fewer
+ than 2 samples or non-positive quantile": the kernels substitute the
+ * plan's default, exactly as rollup.go:871-897 does).  Queries that drop
+ * stale NaNs fall back to the in-kernel computation whenever the compacted
+ * count differs from the stored one (the quantile could differ). */
+__global__ __launch_bounds__(BLOCK_THREADS) void si_prep_kernel(
+    const int64_t* ts, const uint64_t* offsets, uint32_t n_series,
+    int64_t* out_si) {
+  __shared__ double scratch[WAVES_PER_BLOCK][32];
+  const int lane = threadIdx.x % WAVE;
+  const int wv = threadIdx.x / WAVE;
+  uint32_t wid = blockIdx.x * WAVES_PER_BLOCK + wv;
+  uint32_t stride = gridDim.x * WAVES_PER_BLOCK;
+  for (uint32_t i = wid; i < n_series; i += stride) {
+    uint64_t lo = offsets[i];
+    int count = (int)(offsets[i + 1] - lo);
+    int64_t si = scrape_interval_wave_t<false>(ts + lo, count, 0, lane, scratch[wv]);
+    if (lane == 0) out_si[i] = si;
+  }
+}
+
 static VM_DEV int64_t max_prev_interval_tiers(int64_t si) {
   /* getMaxPrevInterval (rollup.go:899-919) */
   if (si <= 2000) return si + 4 * si;
@@ -864,7 +892,14 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
 
     int64_t si = p.step;
 #ifndef VMGPU_ABL_NO_SCRAPE
-    if (p.start < p.end) si = scrape_interval_wave(lts, count, p.step, lane, lscratch);
+    if (p.start < p.end) {
+      if (io.series_si && count == (int)n) {
+        int64_t c = io.series_si[s];
+        si = c > 0 ? c : p.step;
+      } else {
+        si = scrape_interval_wave(lts, count, p.step, lane, lscratch);
+      }
+    }
 #endif
     SeriesWindow sw = series_window(p, si);
 
@@ -1082,13 +1117,33 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
  * scan as rcr_scan_wave (rollup.go:921-958), inputs from registers, fused
  * with the LDS store.  Returns n, or -1 when a stale NaN must be compacted
  * out (caller falls back to load_compact_wave + rcr_scan_wave). */
+/* grid boundary of one sample: smallest g with t_end(g) >= t, clamped to
+ * [-sdg, n_grid] (out-of-grid samples clamp; their scatter ranges become
+ * empty or grid-clipped, which preserves upper-bound semantics exactly). */
+static VM_DEV int vm_scatter_boundary(int64_t t, int64_t gstart, int64_t gstep,
+                                      double inv_gstep, int sdg, int n_grid) {
+  int64_t t_lo_bound = gstart - (int64_t)sdg * gstep;
+  int64_t t_hi_bound = gstart + (int64_t)(n_grid - 1) * gstep;
+  if (t <= t_lo_bound) return -sdg;
+  if (t > t_hi_bound) return n_grid;
+  double est = floor((double)(t - gstart) * inv_gstep);
+  int g = (int)est - 1;
+  g += (gstart + (int64_t)g * gstep < t);
+  g += (gstart + (int64_t)g * gstep < t);
+  g += (gstart + (int64_t)g * gstep < t);
+  return g;
+}
+
 static VM_DEV int rcr_scan_regs(const int64_t* rt, const double* rv,
                                 int64_t n, int64_t* d_ts, double* d_vals,
-                                bool drop_stale, int64_t msi, int lane) {
+                                bool drop_stale, int64_t msi, int lane,
+                                uint16_t* jb, int sdg, int64_t gstart,
+                                int64_t gstep, double inv_gstep, int n_grid) {
   double corr = 0.0;
   double prev_raw = 0.0;
   int64_t prev_ts = 0;
   double prev_fin = 0.0;
+  int carry_b = 0;
 #pragma unroll
   for (int c = 0; c < PIPE_CHUNKS; c++) {
     int64_t base = (int64_t)c * WAVE;
@@ -1099,6 +1154,27 @@ static VM_DEV int rcr_scan_regs(const int64_t* rt, const double* rv,
     int64_t t = active ? rt[c] : 0;
     if (drop_stale && __ballot(active && vm_is_stale_nan(v)) != 0) return -1;
     if (active) d_ts[k] = t;
+    if (jb) {
+      /* fused J scatter: sample k-1 is the window-end boundary for grid
+       * points in [b(t_{k-1}), b(t_k)) — write the upper-bound count k
+       * there; the last sample covers up to the grid end.  Replaces the
+       * per-point seek phase entirely (J[g+sdg] = #samples <= t_end(g)). */
+      int lastl = (int)(n - base - 1);
+      if (lastl > 63) lastl = 63;
+      int b_own = vm_scatter_boundary(t, gstart, gstep, inv_gstep, sdg, n_grid);
+      int b_prev = __shfl_up(b_own, 1);
+      if (lane == 0) b_prev = carry_b;
+      if (active && k > 0) {
+        int glo = b_prev < -sdg ? -sdg : b_prev;
+        int ghi = b_own > n_grid ? n_grid : b_own;
+        for (int g2 = glo; g2 < ghi; g2++) jb[g2 + sdg] = (uint16_t)k;
+      }
+      if (active && k == n - 1) {
+        int glo = b_own < -sdg ? -sdg : b_own;
+        for (int g2 = glo; g2 < n_grid; g2++) jb[g2 + sdg] = (uint16_t)n;
+      }
+      carry_b = __shfl(b_own, lastl);
+    }
     double pv = __shfl_up(v, 1);
     int64_t pt = __shfl_up(t, 1);
     if (lane == 0) { pv = prev_raw; pt = prev_ts; }
@@ -1152,8 +1228,51 @@ static VM_DEV int rcr_scan_regs(const int64_t* rt, const double* rv,
   return (int)n;
 }
 
-template <int FUNC_CT>
-__global__ __launch_bounds__(BLOCK_THREADS) void rollup_pipe_kernel(KPlan p, KIO io) {
+/* J scatter for the no-preprocessing staging path (same semantics as the
+ * fused scatter in rcr_scan_regs). */
+static VM_DEV void scatter_j_regs(const int64_t* rt, int64_t n, int lane,
+                                  uint16_t* jb, int sdg, int64_t gstart,
+                                  int64_t gstep, double inv_gstep, int n_grid) {
+  int carry_b = 0;
+#pragma unroll
+  for (int c = 0; c < PIPE_CHUNKS; c++) {
+    int64_t base = (int64_t)c * WAVE;
+    if (base >= n) continue;
+    int64_t k = base + lane;
+    bool active = k < n;
+    int64_t t = active ? rt[c] : 0;
+    int lastl = (int)(n - base - 1);
+    if (lastl > 63) lastl = 63;
+    int b_own = vm_scatter_boundary(t, gstart, gstep, inv_gstep, sdg, n_grid);
+    int b_prev = __shfl_up(b_own, 1);
+    if (lane == 0) b_prev = carry_b;
+    if (active && k > 0) {
+      int glo = b_prev < -sdg ? -sdg : b_prev;
+      int ghi = b_own > n_grid ? n_grid : b_own;
+      for (int g2 = glo; g2 < ghi; g2++) jb[g2 + sdg] = (uint16_t)k;
+    }
+    if (active && k == n - 1) {
+      int glo = b_own < -sdg ? -sdg : b_own;
+      for (int g2 = glo; g2 < n_grid; g2++) jb[g2 + sdg] = (uint16_t)n;
+    }
+    carry_b = __shfl(b_own, lastl);
+  }
+}
+
+/* VMGPU_PIPE_MINWAVES forces an occupancy floor (waves/SIMD) on the pipe
+ * kernel for A/B builds; VMGPU_PIPE_UNROLL (2 or 4) sets the eval ILP. */
+#ifndef VMGPU_PIPE_UNROLL
+#define VMGPU_PIPE_UNROLL 4
+#endif
+
+template <int FUNC_CT, bool GROUPED>
+__global__
+#ifdef VMGPU_PIPE_MINWAVES
+__launch_bounds__(BLOCK_THREADS, VMGPU_PIPE_MINWAVES)
+#else
+__launch_bounds__(BLOCK_THREADS)
+#endif
+void rollup_pipe_kernel(KPlan p, KIO io) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int wave_in_block = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
@@ -1173,6 +1292,26 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_pipe_kernel(KPlan p, KIO
   int64_t rt[PIPE_CHUNKS];
   double rv[PIPE_CHUNKS];
 
+  /* J-scatter mode (A/B builds only: -DVMGPU_PIPE_SCATTER): plan-uniform
+   * window that is a step multiple + an ext jbuf (jbuf_mode 2).  The
+   * staging scan then builds the full upper-bound map J[g+sdg] =
+   * #samples <= t_end(g) and the eval phase does no seeks at all.
+   * Measured SLOWER than the probe j-cache at config 2 (the divergent
+   * range writes sit on the serial scan chain) — compiled out so the
+   * default kernel does not pay its register pressure. */
+#ifdef VMGPU_PIPE_SCATTER
+  int scat_dg = 0;
+  if (p.jbuf_mode == 2 && p.window > 0 && p.step > 0 &&
+      p.window % p.step == 0) {
+    int64_t d = p.window / p.step;
+    if (d > 0 && p.n_grid + d <= p.jbuf_elems) scat_dg = (int)d;
+  }
+  const double inv_gstep = 1.0 / (double)p.step;
+#else
+  const int scat_dg = 0;
+  const double inv_gstep = 0.0;
+#endif
+
   /* wave_id is wave-uniform by construction; readfirstlane makes that
    * provable, so the per-series descriptor reads below become scalar
    * (s_load, lgkm-counted) instead of divergent vector loads, and the
@@ -1183,10 +1322,12 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_pipe_kernel(KPlan p, KIO
   uint32_t s = 0;
   uint64_t lo = 0;
   int64_t n = 0;
+  int64_t si_raw = 0;
   if (ws < io.n_sel) {
     s = __builtin_amdgcn_readfirstlane(io.series_sel ? io.series_sel[ws] : ws);
     lo = io.offsets[s];
     n = (int64_t)(io.offsets[s + 1] - lo);
+    if (io.series_si) si_raw = io.series_si[s];
     if (n > 0) {
       const int64_t* gts = io.ts + lo;
       const double* gvs = io.vals + lo;
@@ -1206,17 +1347,32 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_pipe_kernel(KPlan p, KIO
     uint32_t s_n = 0;
     uint64_t lo_n = 0;
     int64_t n_n = 0;
+    int64_t si_raw_n = 0;
     if (has_next) {
       s_n = __builtin_amdgcn_readfirstlane(
           io.series_sel ? io.series_sel[ws_n] : ws_n);
       lo_n = io.offsets[s_n];
       n_n = (int64_t)(io.offsets[s_n + 1] - lo_n);
+      if (io.series_si) si_raw_n = io.series_si[s_n];
+    }
+    /* J-scatter init: zero = "no sample <= t_end" for positions before the
+     * first sample's range (upper bound 0) */
+    if (scat_dg > 0) {
+      const int ext = p.n_grid + scat_dg;
+      for (int e = lane; e < ext; e += WAVE) jbuf[e] = 0;
+      wave_ds_sync();
     }
     /* stage the current series into LDS from registers (rcr fused) */
     int count = -1;
+#ifdef VMGPU_PIPE_ABL_NO_SCAN
+    if (false) {
+#else
     if (p.rcr) {
+#endif
       count = rcr_scan_regs(rt, rv, n, lts, lvs, p.drop_stale != 0,
-                            p.max_staleness, lane);
+                            p.max_staleness, lane,
+                            scat_dg > 0 ? jbuf : nullptr, scat_dg,
+                            p.start, p.step, inv_gstep, p.n_grid);
     } else {
       bool stale = false;
       if (p.drop_stale) {
@@ -1232,6 +1388,9 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_pipe_kernel(KPlan p, KIO
           int64_t k = (int64_t)c * WAVE + lane;
           if (k < n) { lts[k] = rt[c]; lvs[k] = rv[c]; }
         }
+        if (scat_dg > 0)
+          scatter_j_regs(rt, n, lane, jbuf, scat_dg, p.start, p.step,
+                         inv_gstep, p.n_grid);
         count = (int)n;
       }
     }
@@ -1255,13 +1414,43 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_pipe_kernel(KPlan p, KIO
                                 p.drop_stale != 0, lane);
       wave_ds_sync();
       if (p.rcr) rcr_scan_wave(lts, lvs, count, p.max_staleness, lane);
+      if (scat_dg > 0) {
+        /* the register scatter saw pre-compaction indices: rebuild J by
+         * direct binary search over the compacted column */
+        wave_ds_sync();
+        const int ext = p.n_grid + scat_dg;
+        for (int e = lane; e < ext; e += WAVE) {
+          int64_t t_end = p.start + (int64_t)(e - scat_dg) * p.step;
+          jbuf[e] = (uint16_t)vm_upper_bound(lts, count, t_end);
+        }
+      }
     }
     wave_ds_sync();
 
+#ifdef VMGPU_PIPE_ABL_STAGE
+    /* stage-only ablation: keep the output write traffic, skip the rest */
+    for (int g = lane; g < p.n_grid; g += WAVE) {
+      int gc = g < count ? g : (count > 0 ? count - 1 : 0);
+      io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] = count ? lvs[gc] : 0.0;
+    }
+    wave_ds_sync();
+    ws = ws_n;
+    s = s_n;
+    lo = lo_n;
+    n = n_n;
+    si_raw = si_raw_n;
+    continue;
+#endif
+
     int64_t si = p.step;
 #ifndef VMGPU_ABL_NO_SCRAPE
-    if (p.start < p.end)
-      si = scrape_interval_wave_t<true>(lts, count, p.step, lane, lscratch);
+    if (p.start < p.end) {
+      if (io.series_si && count == (int)n) {
+        si = si_raw > 0 ? si_raw : p.step;
+      } else {
+        si = scrape_interval_wave_t<true>(lts, count, p.step, lane, lscratch);
+      }
+    }
 #endif
     SeriesWindow sw = series_window(p, si);
     if (lane == 0) scanned += (uint64_t)count;
@@ -1273,77 +1462,126 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_pipe_kernel(KPlan p, KIO
       int64_t span_ms = lts[count - 1] - ts0;
       idx_per_ms = span_ms > 0 ? (double)(count - 1) / (double)span_ms : 0.0;
     }
+    /* emit: GROUPED folds the aggregate switch in; the ungrouped build is
+     * a plain coalesced row store (no per-point branch, no switch). */
+    double* out_row = io.out + (size_t)s * (size_t)p.n_grid;
+    auto emit = [&](int g, double v) {
+      if constexpr (GROUPED) vm_emit_value(p, io, s, g, v);
+      else out_row[g] = v;
+    };
+    const int64_t t_step_wave = (int64_t)WAVE * p.step;
     bool done = false;
     if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
+      if (scat_dg > 0) {
+        /* scatter-J eval: J was built during staging; no seeks at all. */
+        int gs = lane;
+        int64_t te_s = p.start + (int64_t)lane * p.step;
+        for (; gs < p.n_grid; gs += WAVE, te_s += t_step_wave) {
+          int j = jbuf[gs + scat_dg];
+          int i = jbuf[gs];
+#ifdef VMGPU_PIPE_ABL_NO_EVAL
+          emit(gs, (j > 0 && j <= count) ? lvs[j - 1] : 0.0);
+          (void)i;
+#else
+          emit(gs, eval_rate_fused(p, sw, lts, lvs, count, i, j,
+                                   te_s - sw.window));
+#endif
+        }
+        done = true;
+      }
       int dg64 = (sw.window > 0 && p.step > 0 && sw.window % p.step == 0)
                      ? (int)(sw.window / p.step) : 0;
-      if (p.jbuf_mode >= 1 &&
+      if (!done && p.jbuf_mode >= 1 &&
           (size_t)p.n_grid * 2 <= vm_jbuf_bytes(p.jbuf_mode, p.jbuf_elems) &&
           dg64 > 0 && count <= 65535) {
-        for (int g = lane; g < p.n_grid; g += WAVE) {
-          int64_t t_end = p.start + (int64_t)g * p.step;
-          int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
-          jbuf[g] = (uint16_t)vm_ub_hint_fast(lts, count, t_end, gj);
+        /* j-cache fill: full lane-blocks without bounds checks, one masked
+         * tail; t_end is strength-reduced (+64*step per round) */
+        {
+          int gf = lane;
+          int64_t te = p.start + (int64_t)lane * p.step;
+          const int fill_full = (p.n_grid / WAVE) * WAVE;
+          for (; gf < fill_full; gf += WAVE, te += t_step_wave) {
+            int gj = (int)((double)(te - ts0) * idx_per_ms) + 1;
+            jbuf[gf] = (uint16_t)vm_ub_hint_fast(lts, count, te, gj);
+          }
+          if (gf < p.n_grid) {
+            int gj = (int)((double)(te - ts0) * idx_per_ms) + 1;
+            jbuf[gf] = (uint16_t)vm_ub_hint_fast(lts, count, te, gj);
+          }
         }
         wave_ds_sync();
-        for (int g0 = 0; g0 < p.n_grid; g0 += 4 * WAVE) {
-#pragma unroll
-          for (int u = 0; u < 4; u++) {
-            int g = g0 + u * WAVE + lane;
-            if (g < p.n_grid) {
-              int64_t t_end = p.start + (int64_t)g * p.step;
-              int64_t t_start = t_end - sw.window;
-              int j = jbuf[g];
-              int i;
-              if (g >= dg64) {
-                i = jbuf[g - dg64];
-              } else {
-                int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
-                i = vm_ub_hint_fast(lts, count, t_start, gi);
-              }
-              vm_emit_value(p, io, s, g,
-                            eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
-              scanned += 2;
-            }
+        /* the first dg points (window start before the grid) probe for i;
+         * keeping them out of the main loop keeps the hot body free of the
+         * probe's registers and branches */
+        const int ghead = dg64 < p.n_grid ? dg64 : p.n_grid;
+        for (int g = lane; g < ghead; g += WAVE) {
+          int64_t t_end = p.start + (int64_t)g * p.step;
+          int64_t t_start = t_end - sw.window;
+          int j = jbuf[g];
+          int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
+          int i = vm_ub_hint_fast(lts, count, t_start, gi);
+#ifdef VMGPU_PIPE_ABL_NO_EVAL
+          emit(g, (j > 0 && j <= count) ? lvs[j - 1] : 0.0);
+          (void)i;
+#else
+          emit(g, eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
+#endif
+        }
+        /* main eval: full lane-blocks (no per-point exec masking), then one
+         * masked tail round */
+        {
+          int g = ghead + lane;
+          int64_t te = p.start + (int64_t)(ghead + lane) * p.step;
+          const int span = p.n_grid - ghead;
+          const int main_end = ghead + (span / WAVE) * WAVE;
+          for (; g < main_end; g += WAVE, te += t_step_wave) {
+            int j = jbuf[g];
+            int i = jbuf[g - dg64];
+#ifdef VMGPU_PIPE_ABL_NO_EVAL
+            emit(g, (j > 0 && j <= count) ? lvs[j - 1] : 0.0);
+            (void)i;
+#else
+            emit(g, eval_rate_fused(p, sw, lts, lvs, count, i, j,
+                                    te - sw.window));
+#endif
+          }
+          if (g < p.n_grid) {
+            int j = jbuf[g];
+            int i = jbuf[g - dg64];
+#ifdef VMGPU_PIPE_ABL_NO_EVAL
+            emit(g, (j > 0 && j <= count) ? lvs[j - 1] : 0.0);
+            (void)i;
+#else
+            emit(g, eval_rate_fused(p, sw, lts, lvs, count, i, j,
+                                    te - sw.window));
+#endif
           }
         }
         done = true;
       }
     }
     if (!done) {
-      for (int g0 = 0; g0 < p.n_grid; g0 += 4 * WAVE) {
-#pragma unroll
-        for (int u = 0; u < 4; u++) {
-          int g = g0 + u * WAVE + lane;
-          if (g < p.n_grid) {
-            int64_t t_end = p.start + (int64_t)g * p.step;
-            int64_t t_start = t_end - sw.window;
-            int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
-            int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
-            int i, j;
-            if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
-              i = vm_ub_hint_fast(lts, count, t_start, gi);
-              j = vm_ub_hint_fast(lts, count, t_end, gj);
-            } else {
-              i = vm_ub_hint(lts, count, t_start, gi);
-              j = vm_ub_hint(lts, count, t_end, gj);
-            }
-            if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
-              vm_emit_value(p, io, s, g,
-                            eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
-              scanned += 2;
-            } else {
-              scanned += eval_grid_point_ij<FUNC_CT>(p, sw, lts, lvs, count, g, s, io, i, j);
-            }
-          }
-        }
+      /* correctness fallback only (per-series window not a step multiple —
+       * the host gates the pipe launch on plan shapes where the jbuf path
+       * applies, so this stays cold): plain binary-search seeks, minimal
+       * register footprint. */
+      for (int g = lane; g < p.n_grid; g += WAVE) {
+        int64_t t_end = p.start + (int64_t)g * p.step;
+        int64_t t_start = t_end - sw.window;
+        int i = vm_upper_bound(lts, count, t_start);
+        int j = vm_upper_bound(lts, count, t_end);
+        emit(g, eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
       }
     }
+    /* samplesScanned: every grid point of this series contributes exactly
+     * samplesScannedPerCall (= 2 for rate/deriv_fast) */
+    if (lane == 0) scanned += 2ull * (unsigned long long)p.n_grid;
     wave_ds_sync();
     ws = ws_n;
     s = s_n;
     lo = lo_n;
     n = n_n;
+    si_raw = si_raw_n;
   }
   for (int d = 32; d > 0; d >>= 1) scanned += __shfl_down((unsigned long long)scanned, d);
   if (lane == 0 && scanned) atomicAdd(io.samples_scanned, (unsigned long long)scanned);
@@ -1399,7 +1637,14 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KI
     __syncthreads();
     if (wave == 0) {
       int64_t si = p.step;
-      if (p.start < p.end) si = scrape_interval_wave(lts, count, p.step, lane, lscratch);
+      if (p.start < p.end) {
+        if (io.series_si && count == (int)n) {
+          int64_t c = io.series_si[s];
+          si = c > 0 ? c : p.step;
+        } else {
+          si = scrape_interval_wave(lts, count, p.step, lane, lscratch);
+        }
+      }
       if (lane == 0) *sh_si = si;
     }
     __syncthreads();
@@ -1505,8 +1750,13 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_huge_kernel(KPlan p, KIO
     if (wave == 0) {
       int64_t si = p.step;
       if (p.start < p.end) {
-        /* scrape interval needs the tail of the (possibly compacted) column */
-        si = scrape_interval_wave(uts, count, p.step, lane, sh_scratch);
+        if (io.series_si && count == (int)n) {
+          int64_t c = io.series_si[s];
+          si = c > 0 ? c : p.step;
+        } else {
+          /* scrape interval needs the tail of the (possibly compacted) column */
+          si = scrape_interval_wave(uts, count, p.step, lane, sh_scratch);
+        }
       }
       if (lane == 0) sh_si = si;
     }
@@ -2116,8 +2366,12 @@ static void launch_rollup_t(int which, uint32_t blocks, size_t lds,
      * samples); instantiated only for the hot specializations it is
      * measured on — everything else falls back to the wave kernel */
     if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
-      hipLaunchKernelGGL(rollup_pipe_kernel<FUNC_CT>, dim3(blocks),
-                         dim3(BLOCK_THREADS), lds, stream, p, w);
+      if (p.aggr != VMGPU_AGGR_NONE && w.group_ids)
+        hipLaunchKernelGGL((rollup_pipe_kernel<FUNC_CT, true>), dim3(blocks),
+                           dim3(BLOCK_THREADS), lds, stream, p, w);
+      else
+        hipLaunchKernelGGL((rollup_pipe_kernel<FUNC_CT, false>), dim3(blocks),
+                           dim3(BLOCK_THREADS), lds, stream, p, w);
     } else {
       launch_rollup_t<FUNC_CT>(0, blocks, lds, p, w, stream);
     }
@@ -2166,6 +2420,7 @@ struct Batch {
   uint32_t n_groups = 0;
   uint64_t n_samples = 0;
   /* series partition by length */
+  int64_t* d_si = nullptr;   /* per-series scrape-interval index */
   uint32_t* d_wave_list = nullptr;
   uint32_t* d_block_list = nullptr;
   uint32_t* d_huge_list = nullptr;
@@ -2319,11 +2574,25 @@ static int relayout_by_group(Batch& b, const int32_t* group_ids,
   return 0;
 }
 
+int build_si_index(Batch& b, char* errbuf, size_t errbuf_len) {
+  HIP_TRY(hipMalloc(&b.d_si, (size_t)b.n_series * 8), "alloc si index");
+  uint32_t blocks = std::min<uint32_t>(
+      (b.n_series + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK, 4096);
+  hipLaunchKernelGGL(si_prep_kernel, dim3(blocks), dim3(BLOCK_THREADS), 0,
+                     g_ctx.stream, b.d_ts, b.d_offsets, b.n_series, b.d_si);
+  HIP_TRY(hipStreamSynchronize(g_ctx.stream), "sync si index");
+  hipError_t kerr = hipGetLastError();
+  if (kerr != hipSuccess)
+    return hip_err(errbuf, errbuf_len, "si index kernel", kerr);
+  return 0;
+}
+
 void free_batch(Batch& b) {
   (void)hipFree(b.d_ts);
   (void)hipFree(b.d_vals);
   (void)hipFree(b.d_offsets);
   (void)hipFree(b.d_group_ids);
+  (void)hipFree(b.d_si);
   (void)hipFree(b.d_wave_list);
   (void)hipFree(b.d_block_list);
   (void)hipFree(b.d_huge_list);
@@ -2456,6 +2725,10 @@ int vmgpu_batch_create(const int64_t* ts, const double* vals,
     HIP_TRY(hipMalloc(&b.d_scr_vals, huge_total * sizeof(double)), "alloc scratch vals");
   }
   HIP_TRY(hipMalloc(&b.d_scanned, sizeof(unsigned long long)), "alloc scanned");
+  {
+    int src_rc = build_si_index(b, errbuf, errbuf_len);
+    if (src_rc != 0) { free_batch(b); return src_rc; }
+  }
 
   uint64_t h = g_ctx.next_handle++;
   g_ctx.batches[h] = b;
@@ -2584,6 +2857,10 @@ int vmgpu_batch_create_from_blocks(
   }
   FBB_TRY(hipMalloc(&b.d_scanned, sizeof(unsigned long long)), "alloc scanned");
 #undef FBB_TRY
+  {
+    int src_rc = build_si_index(b, errbuf, errbuf_len);
+    if (src_rc != 0) { free_batch(b); return src_rc; }
+  }
 
   uint64_t h = g_ctx.next_handle++;
   g_ctx.batches[h] = b;
@@ -2679,6 +2956,24 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
    * occupancy loss; J-only scatter: 2.77 ms from the boundary-fixup +
    * divergent write cost) vs 2.51 ms for the j-cache.  Mode 2 remains
    * selectable for grids too large for the u16 cache. */
+  /* pipe eligibility decides the jbuf flavor: the pipelined kernel builds
+   * the full ext J map during its staging scan (mode 2), everything else
+   * uses the probe-filled u16 j-cache (mode 1). */
+  bool use_pipe = false;
+  if ((plan->func == VMF_RATE || plan->func == VMF_DERIV_FAST) && b.n_wave &&
+      b.max_wave_len <= (uint32_t)(PIPE_CHUNKS * WAVE) && plan->pre_func == 0 &&
+      plan->window > 0 && plan->step > 0 && plan->window % plan->step == 0 &&
+      n_grid > 1) {
+    static int pipe_disabled = -1;
+    if (pipe_disabled < 0) {
+      const char* e = getenv("VMGPU_DISABLE_PIPE");
+      pipe_disabled = (e && e[0] == '1') ? 1 : 0;
+    }
+#ifdef VMGPU_ABL_GACC
+    pipe_disabled = 1; /* the GACC A/B runs through the wave kernel */
+#endif
+    use_pipe = !pipe_disabled;
+  }
   p.jbuf_elems = 0;
   p.jbuf_mode = 0;
   if ((plan->func == VMF_RATE || plan->func == VMF_DERIV_FAST) && n_grid > 1) {
@@ -2686,7 +2981,26 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
     int64_t dgp = (plan->window > 0 && plan->step > 0 &&
                    plan->window % plan->step == 0)
                       ? plan->window / plan->step : 0;
-    if (n_grid <= 4096 &&
+    /* A/B measured (round 2): the in-scan J scatter costs MORE inside the
+     * staging chain (+0.46 ms at config 2) than the probe j-cache it
+     * replaces (~0.42 ms in an overlappable phase) — the divergent range
+     * writes and boundary fixups sit on the serial scan path.  Probe mode
+     * stays the default; VMGPU_PIPE_SCATTER=1 re-enables the scatter. */
+#ifdef VMGPU_PIPE_SCATTER
+    static int pipe_scatter = -1;
+    if (pipe_scatter < 0) {
+      const char* e2 = getenv("VMGPU_PIPE_SCATTER");
+      pipe_scatter = (e2 && e2[0] == '1') ? 1 : 0;
+    }
+#else
+    const int pipe_scatter = 0;
+#endif
+    if (use_pipe && pipe_scatter && dgp > 0 && n_grid + dgp <= 32000 &&
+        base + WAVES_PER_BLOCK * vm_jbuf_bytes(2, (int32_t)(n_grid + dgp)) <=
+            64 * 1024) {
+      p.jbuf_mode = 2;
+      p.jbuf_elems = (int32_t)(n_grid + dgp);
+    } else if (n_grid <= 4096 &&
         base + WAVES_PER_BLOCK * vm_jbuf_bytes(1, n_grid) <= 64 * 1024) {
       p.jbuf_mode = 1;
       p.jbuf_elems = n_grid;
@@ -2698,6 +3012,8 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
       }
     }
   }
+  /* the pipe kernel carries only the j-cache rate path */
+  if (p.jbuf_mode == 0) use_pipe = false;
   p.arg = plan->arg;
   p.arg2 = plan->arg2;
 
@@ -2706,6 +3022,7 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
   io.vals = b.d_vals;
   io.offsets = b.d_offsets;
   io.group_ids = b.d_group_ids;
+  io.series_si = b.d_si;
   io.out = b.d_out;
   io.out_counts = b.d_counts;
   io.samples_scanned = b.d_scanned;
@@ -2734,20 +3051,7 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
                   vm_jbuf_bytes(p.jbuf_mode, p.jbuf_elems));
     /* software-pipelined register-staged variant for short series (the
      * config-2 flagship shape); VMGPU_DISABLE_PIPE=1 A/Bs the wave kernel */
-    int which = 0;
-    if (b.max_wave_len <= (uint32_t)(PIPE_CHUNKS * WAVE) && p.pre_func == 0 &&
-        (p.func == VMF_RATE || p.func == VMF_DERIV_FAST)) {
-      static int pipe_disabled = -1;
-      if (pipe_disabled < 0) {
-        const char* e = getenv("VMGPU_DISABLE_PIPE");
-        pipe_disabled = (e && e[0] == '1') ? 1 : 0;
-      }
-#ifdef VMGPU_ABL_GACC
-      pipe_disabled = 1; /* the GACC A/B runs through the wave kernel */
-#endif
-      if (!pipe_disabled) which = 3;
-    }
-    launch_rollup(which, blocks, lds, p, w);
+    launch_rollup(use_pipe ? 3 : 0, blocks, lds, p, w);
   }
   if (b.n_block) {
     KIO w = io;
