@@ -378,6 +378,39 @@ int vmgpu_batch_create_from_blocks(
     uint64_t* out_handle, uint64_t* out_offsets,
     char* errbuf, size_t errbuf_len);
 
+/* Packed block stream: the single-buffer wire form a fetch layer hands
+ * over for a whole cold query (replaces the reference's per-series
+ * []sortedBlock handoff, netstorage.go:423-614).  n_blocks records, each a
+ * 48-byte vmgpu_packed_block_hdr followed by ts_data (ts_data_len bytes)
+ * then val_data (val_data_len bytes); record i belongs to the series given
+ * by series_block_start (blocks appear grouped by series, in order).
+ * Marshal types are post-zstd (2/3/5/6 and const/delta-const). */
+typedef struct vmgpu_packed_block_hdr {
+  int64_t min_timestamp;
+  int64_t max_timestamp;
+  int64_t first_value;   /* values column first value; the timestamps
+                            column's first value is min_timestamp */
+  uint32_t rows;
+  int32_t scale;
+  uint32_t ts_data_len;
+  uint32_t val_data_len;
+  uint8_t ts_mt;
+  uint8_t val_mt;
+  uint8_t precision_bits;
+  uint8_t _pad[5];
+} vmgpu_packed_block_hdr;
+
+/* Native descriptor build + fused decode/merge/batch-create from a packed
+ * block stream: walks the records in C (no per-block marshaling in the
+ * host-language layer), then runs the vmgpu_batch_create_from_blocks
+ * pipeline against the stream buffer itself as the device payload. */
+int vmgpu_batch_create_packed(
+    const uint8_t* packed, uint64_t packed_len, uint64_t n_blocks,
+    const uint32_t* series_block_start, uint32_t n_series,
+    int64_t dedup_interval, const int32_t* group_ids, uint32_t n_groups,
+    uint64_t* out_handle, uint64_t* out_offsets,
+    char* errbuf, size_t errbuf_len);
+
 /* Wall time of the rollup kernels inside the last vmgpu_rollup_exec on this
  * thread's context, measured with hipEvents on the launch stream (for the
  * bench's roofline accounting). */

@@ -358,6 +358,7 @@ def _codec_lib():
         l.vm_unmarshal_nearest_delta.restype = ctypes.c_int
         l.vm_unmarshal_nearest_delta2.restype = ctypes.c_int
         l.vm_marshal_int64_array.restype = ctypes.c_int64
+        l.vm_pack_blocks.restype = ctypes.c_int64
         l.vm_unmarshal_int64_array.restype = ctypes.c_int
         l.vm_deduplicate_samples.restype = ctypes.c_int64
         l.vm_merge_sort_blocks.restype = ctypes.c_int64
@@ -451,6 +452,29 @@ def unmarshal_nearest_delta(data, first_value, items, delta2=False):
     if rc != 0:
         raise ValueError(f"nearest-delta decode error {rc}")
     return dst
+
+
+def pack_blocks(ts, vals_int, offsets, precision_bits=64):
+    """Synthetic packed block stream for the cold-query bench/tests: CSR
+    int64 columns (scale 0) -> the wire form vmgpu_batch_create_packed
+    parses (one block per series, post-zstd marshal types).  Returns
+    (packed_bytes, n_blocks, series_block_start)."""
+    l = _codec_lib()
+    t = np.ascontiguousarray(ts, dtype=np.int64)
+    v = np.ascontiguousarray(vals_int, dtype=np.int64)
+    off = np.ascontiguousarray(offsets, dtype=np.uint64)
+    n_series = len(off) - 1
+    cap = int(len(t)) * 22 + n_series * 192 + 64
+    dst = np.empty(cap, dtype=np.uint8)
+    sbs = np.zeros(n_series + 1, dtype=np.uint32)
+    n = l.vm_pack_blocks(_ptr(t, ctypes.c_int64), _ptr(v, ctypes.c_int64),
+                         _ptr(off, ctypes.c_uint64), ctypes.c_uint32(n_series),
+                         ctypes.c_uint8(precision_bits),
+                         _ptr(dst, ctypes.c_uint8), ctypes.c_int64(cap),
+                         _ptr(sbs, ctypes.c_uint32))
+    if n < 0:
+        raise ValueError(f"pack_blocks error {n}")
+    return bytes(dst[:n].data), n_series, sbs
 
 
 def marshal_int64_array(a, precision_bits=64):

@@ -628,3 +628,92 @@ int64_t vm_merge_sort_blocks(const int64_t* ts, const double* vals,
   free(sbs);
   return out;
 }
+
+/* ---------------- packed block stream (bench input generator) -------------
+ * Builds the wire form the product's vmgpu_batch_create_packed parses: a
+ * sequence of records, each a 48-byte header followed by ts_data then
+ * val_data (post-zstd marshal types only — the device decode contract).
+ * This is TEST INFRASTRUCTURE: the cold-query bench uses it to synthesize
+ * the compressed payload a vmselect fetch would hand over (the write/encode
+ * path itself is out of scope).  Header layout must match
+ * include/vmgpu.h vmgpu_packed_block_hdr. */
+typedef struct {
+  int64_t min_timestamp;
+  int64_t max_timestamp;
+  int64_t first_value;
+  uint32_t rows;
+  int32_t scale;
+  uint32_t ts_data_len;
+  uint32_t val_data_len;
+  uint8_t ts_mt;
+  uint8_t val_mt;
+  uint8_t precision_bits;
+  uint8_t pad[5];
+} vm_packed_hdr;
+
+/* marshal without the zstd stage (types 0/2/3/5/6 only) */
+static int64_t marshal_nozstd(uint8_t* dst, const int64_t* a, int64_t n,
+                              uint8_t precision_bits, uint8_t* out_mt,
+                              int64_t* out_first) {
+  if (n == 0) return -2;
+  if (vm_is_const(a, n)) {
+    *out_first = a[0];
+    *out_mt = VM_MT_CONST;
+    return 0;
+  }
+  if (vm_is_delta_const(a, n)) {
+    *out_first = a[0];
+    *out_mt = VM_MT_DELTA_CONST;
+    int64_t d = a[1] - a[0];
+    return (int64_t)vm_marshal_varint64s(dst, &d, 1);
+  }
+  if (vm_is_gauge(a, n)) {
+    uint8_t pb = precision_bits;
+    if (pb < 6) pb = (uint8_t)(pb + 2);
+    *out_mt = VM_MT_NEAREST_DELTA;
+    return (int64_t)vm_marshal_nearest_delta(dst, a, n, pb, out_first);
+  }
+  *out_mt = VM_MT_NEAREST_DELTA2;
+  return (int64_t)vm_marshal_nearest_delta2(dst, a, n, precision_bits, out_first);
+}
+
+/* CSR int64 columns (scale 0) -> packed stream, one block per series.
+ * Returns packed bytes written, or -1 if cap would overflow.
+ * series_block_start (n_series+1, may be NULL) gets the identity CSR. */
+int64_t vm_pack_blocks(const int64_t* ts, const int64_t* vals,
+                       const uint64_t* offsets, uint32_t n_series,
+                       uint8_t precision_bits, uint8_t* dst, int64_t cap,
+                       uint32_t* series_block_start) {
+  int64_t w = 0;
+  for (uint32_t s = 0; s < n_series; s++) {
+    if (series_block_start) series_block_start[s] = s;
+    uint64_t lo = offsets[s];
+    int64_t n = (int64_t)(offsets[s + 1] - lo);
+    if (n == 0 || n > 8192) return -2; /* one block per series only */
+    int64_t worst = (int64_t)sizeof(vm_packed_hdr) + 2 * (n * 10 + 64);
+    if (w + worst > cap) return -1;
+    vm_packed_hdr* h = (vm_packed_hdr*)(dst + w);
+    memset(h, 0, sizeof(*h));
+    uint8_t* body = dst + w + sizeof(vm_packed_hdr);
+    int64_t tfirst = 0, vfirst = 0;
+    uint8_t tmt = 0, vmt = 0;
+    int64_t tlen = marshal_nozstd(body, ts + lo, n, 64, &tmt, &tfirst);
+    if (tlen < 0) return -3;
+    int64_t vlen = marshal_nozstd(body + tlen, vals + lo, n, precision_bits,
+                                  &vmt, &vfirst);
+    if (vlen < 0) return -3;
+    h->min_timestamp = ts[lo];
+    h->max_timestamp = ts[lo + n - 1];
+    h->first_value = vfirst;
+    h->rows = (uint32_t)n;
+    h->scale = 0;
+    h->ts_data_len = (uint32_t)tlen;
+    h->val_data_len = (uint32_t)vlen;
+    h->ts_mt = tmt;
+    h->val_mt = vmt;
+    h->precision_bits = precision_bits;
+    w += (int64_t)sizeof(vm_packed_hdr) + tlen + vlen;
+  }
+  if (series_block_start) series_block_start[n_series] = n_series;
+  return w;
+}

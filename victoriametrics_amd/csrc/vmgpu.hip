@@ -2868,6 +2868,59 @@ int vmgpu_batch_create_from_blocks(
   return 0;
 }
 
+/* Native descriptor build from a packed block stream (vmgpu.h): the C
+ * parse that the reference does in Go when unpackWorker walks its
+ * []sortedBlock (netstorage.go:423-614).  The stream buffer itself is the
+ * device payload — data offsets point into it, nothing is re-copied on the
+ * host. */
+int vmgpu_batch_create_packed(
+    const uint8_t* packed, uint64_t packed_len, uint64_t n_blocks,
+    const uint32_t* series_block_start, uint32_t n_series,
+    int64_t dedup_interval, const int32_t* group_ids, uint32_t n_groups,
+    uint64_t* out_handle, uint64_t* out_offsets,
+    char* errbuf, size_t errbuf_len) {
+  if (!packed || !series_block_start || !out_handle || !out_offsets ||
+      n_series == 0 || n_blocks == 0)
+    return set_err(errbuf, errbuf_len, "vmgpu: bad args");
+  std::vector<vmgpu_block_desc> descs(n_blocks);
+  uint64_t off = 0;
+  uint64_t total_rows = 0;
+  for (uint64_t i = 0; i < n_blocks; i++) {
+    if (off + sizeof(vmgpu_packed_block_hdr) > packed_len)
+      return set_err(errbuf, errbuf_len, "vmgpu: packed stream truncated");
+    vmgpu_packed_block_hdr h;
+    memcpy(&h, packed + off, sizeof(h)); /* stream need not be aligned */
+    off += sizeof(h);
+    vmgpu_block_desc& d = descs[i];
+    d.ts_data_off = off;
+    d.ts_data_len = h.ts_data_len;
+    off += h.ts_data_len;
+    d.val_data_off = off;
+    d.val_data_len = h.val_data_len;
+    off += h.val_data_len;
+    if (off > packed_len)
+      return set_err(errbuf, errbuf_len, "vmgpu: packed stream truncated");
+    if (h.rows == 0 || h.rows > 8192)
+      return set_err(errbuf, errbuf_len, "vmgpu: bad packed rows");
+    d.out_off = total_rows;
+    d.min_timestamp = h.min_timestamp;
+    d.max_timestamp = h.max_timestamp;
+    d.first_value = h.first_value;
+    d.scale = h.scale;
+    d.e10 = pow(10.0, (double)(h.scale < 0 ? -h.scale : h.scale));
+    d.rows = h.rows;
+    d.ts_mt = h.ts_mt;
+    d.val_mt = h.val_mt;
+    d.precision_bits = h.precision_bits;
+    d._pad = 0;
+    total_rows += h.rows;
+  }
+  return vmgpu_batch_create_from_blocks(
+      packed, packed_len, descs.data(), (uint32_t)n_blocks, total_rows,
+      series_block_start, n_series, dedup_interval, group_ids, n_groups,
+      out_handle, out_offsets, errbuf, errbuf_len);
+}
+
 /* Physical-row -> original-series mapping of a (relayouted) grouped batch:
  * out_perm[n_series].  Identity batches fill 0..n-1.  Callers need this to
  * place per-series outputs of vmgpu_rollup_exec back in request order. */

@@ -467,6 +467,51 @@ class SeriesBatch:
         self._perm = self._fetch_perm() if group_ids is not None else None
         return self
 
+    @classmethod
+    def from_packed(cls, packed, n_blocks, series_block_start,
+                    dedup_interval=0, group_ids=None, n_groups=0):
+        """Cold fetch through the NATIVE descriptor path
+        (vmgpu_batch_create_packed): one contiguous packed block stream in,
+        resident batch out.  The per-block walk happens in C — no
+        host-language marshaling per block (the cgo production shape)."""
+        init()
+        lib = _load_lib()
+        pl = np.frombuffer(packed, dtype=np.uint8)
+        sbs = np.ascontiguousarray(series_block_start, dtype=np.uint32)
+        n_series = len(sbs) - 1
+        out_offsets = np.zeros(n_series + 1, dtype=np.uint64)
+        if group_ids is not None:
+            gids = np.ascontiguousarray(group_ids, dtype=np.int32)
+            gptr = gids.ctypes.data_as(ctypes.POINTER(ctypes.c_int32))
+        else:
+            gids, gptr = None, None
+        handle = ctypes.c_uint64(0)
+        errbuf = ctypes.create_string_buffer(256)
+        rc = lib.vmgpu_batch_create_packed(
+            pl.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+            ctypes.c_uint64(len(pl)), ctypes.c_uint64(int(n_blocks)),
+            sbs.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+            ctypes.c_uint32(n_series), ctypes.c_int64(int(dedup_interval)),
+            gptr, ctypes.c_uint32(int(n_groups)),
+            ctypes.byref(handle),
+            out_offsets.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            errbuf, ctypes.c_size_t(256))
+        if rc != 0:
+            raise VmGpuError(f"vmgpu_batch_create_packed failed ({rc}):"
+                             f" {errbuf.value.decode()}")
+        self = cls.__new__(cls)
+        self.ts = None
+        self.vals = None
+        self.offsets = out_offsets
+        self.n_series = n_series
+        self.n_groups = int(n_groups)
+        self.group_ids = gids
+        self.handle = handle.value
+        self._last_rows = 0
+        self._last_n_grid = 0
+        self._perm = self._fetch_perm() if group_ids is not None else None
+        return self
+
     def close(self):
         if self.handle:
             _load_lib().vmgpu_batch_destroy(ctypes.c_uint64(self.handle))
