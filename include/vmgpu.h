@@ -411,6 +411,11 @@ int vmgpu_batch_create_packed(
     uint64_t* out_handle, uint64_t* out_offsets,
     char* errbuf, size_t errbuf_len);
 
+/* Pinned host buffers for PCIe-rate staging of payloads and results (the
+ * cgo layer would pool these; hipHostMalloc/-Free underneath). */
+int vmgpu_host_alloc(uint64_t nbytes, void** out_ptr);
+int vmgpu_host_free(void* ptr);
+
 /* Wall time of the rollup kernels inside the last vmgpu_rollup_exec on this
  * thread's context, measured with hipEvents on the launch stream (for the
  * bench's roofline accounting). */

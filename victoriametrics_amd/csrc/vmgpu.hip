@@ -2921,6 +2921,20 @@ int vmgpu_batch_create_packed(
       out_handle, out_offsets, errbuf, errbuf_len);
 }
 
+/* Pinned host memory for PCIe-rate payload/result staging (the cgo
+ * production layer would hold these in a pool, like netstorage's
+ * result buffer pools). */
+int vmgpu_host_alloc(uint64_t nbytes, void** out_ptr) {
+  if (!out_ptr || nbytes == 0) return 1;
+  *out_ptr = nullptr;
+  return hipHostMalloc(out_ptr, nbytes, hipHostMallocDefault) == hipSuccess
+             ? 0 : 2;
+}
+
+int vmgpu_host_free(void* ptr) {
+  return hipHostFree(ptr) == hipSuccess ? 0 : 1;
+}
+
 /* Physical-row -> original-series mapping of a (relayouted) grouped batch:
  * out_perm[n_series].  Identity batches fill 0..n-1.  Callers need this to
  * place per-series outputs of vmgpu_rollup_exec back in request order. */

@@ -476,7 +476,9 @@ class SeriesBatch:
         host-language marshaling per block (the cgo production shape)."""
         init()
         lib = _load_lib()
-        pl = np.frombuffer(packed, dtype=np.uint8)
+        pl = np.frombuffer(packed, dtype=np.uint8) \
+            if isinstance(packed, (bytes, bytearray, memoryview)) \
+            else np.ascontiguousarray(packed, dtype=np.uint8)
         sbs = np.ascontiguousarray(series_block_start, dtype=np.uint32)
         n_series = len(sbs) - 1
         out_offsets = np.zeros(n_series + 1, dtype=np.uint64)
@@ -523,7 +525,8 @@ class SeriesBatch:
     def __exit__(self, *exc):
         self.close()
 
-    def exec(self, plan, download=True, tracer=None, deadline=None):
+    def exec(self, plan, download=True, tracer=None, deadline=None,
+             out_buf=None, counts_buf=None):
         """Evaluate plan over this batch.  Returns (out, counts,
         samples_scanned); out is [n_series x n_grid] or [n_groups x n_grid].
         With download=False the results stay on the device (use fetch_out).
@@ -541,10 +544,12 @@ class SeriesBatch:
         optr = cptr = None
         if download:
             rows = self.n_groups if grouped else self.n_series
-            out = np.empty((rows, n_grid), dtype=np.float64)
+            out = out_buf if out_buf is not None \
+                else np.empty((rows, n_grid), dtype=np.float64)
             optr = out.ctypes.data_as(ctypes.POINTER(ctypes.c_double))
             if grouped:
-                counts = np.empty((rows, n_grid), dtype=np.float64)
+                counts = counts_buf if counts_buf is not None \
+                    else np.empty((rows, n_grid), dtype=np.float64)
                 cptr = counts.ctypes.data_as(ctypes.POINTER(ctypes.c_double))
         from . import limits as _limits
         if deadline is not None:
@@ -606,6 +611,27 @@ class SeriesBatch:
         if rc != 0:
             raise VmGpuError(f"vmgpu_batch_fetch_out failed ({rc})")
         return out, counts
+
+
+def host_alloc(nbytes):
+    """Pinned (hipHostMalloc) host buffer exposed as a uint8 numpy array —
+    PCIe-rate H2D/D2H staging.  Returns (array, raw_ptr); free with
+    host_free(raw_ptr).  The cgo production layer would pool these the way
+    netstorage pools its result buffers."""
+    init()
+    lib = _load_lib()
+    ptr = ctypes.c_void_p(0)
+    rc = lib.vmgpu_host_alloc(ctypes.c_uint64(int(nbytes)), ctypes.byref(ptr))
+    if rc != 0 or not ptr.value:
+        raise VmGpuError(f"vmgpu_host_alloc({nbytes}) failed ({rc})")
+    buf = np.ctypeslib.as_array(
+        ctypes.cast(ptr, ctypes.POINTER(ctypes.c_uint8)), shape=(int(nbytes),))
+    return buf, ptr.value
+
+
+def host_free(raw_ptr):
+    if raw_ptr:
+        _load_lib().vmgpu_host_free(ctypes.c_void_p(raw_ptr))
 
 
 def rollup_eval(plan, ts, vals, offsets, group_ids=None, n_groups=0):
