@@ -221,7 +221,11 @@ def test_rollup_dispatch_covers_reference_map():
         "timestamp_with_name", "tlast_change_over_time", "tlast_over_time",
         "tmax_over_time", "tmin_over_time", "zscore_over_time"}
     expansions = {"aggr_over_time", "quantiles_over_time"}
+    # host multi-series rollups (timeseriesMap side channel, rollup.go:1490)
+    from victoriametrics_amd import rollup_multi
     host_multi_output = {"count_values_over_time", "histogram_over_time"}
+    assert callable(rollup_multi.count_values_over_time)
+    assert callable(rollup_multi.histogram_over_time)
     handled = (set(engine.FUNC_IDS) | set(engine.ROLLUP_FAKE_FUNCS) |
                expansions | host_multi_output)
     missing = reference_names - handled
