@@ -273,12 +273,31 @@ void vm_decimal_append_float_to_decimal(const double* src, int64_t n,
 }
 
 /* RoundToDecimalDigits (decimal.go:325-337) */
+/* math.Pow10 (math/pow10.go): the exact table-product construction the
+ * reference multiplies/divides by — libm pow can differ by 1 ulp at
+ * |n| >= 23. */
+static double go_pow10(int n) {
+  static const double tab[32] = {
+      1e0,  1e1,  1e2,  1e3,  1e4,  1e5,  1e6,  1e7,  1e8,  1e9,  1e10,
+      1e11, 1e12, 1e13, 1e14, 1e15, 1e16, 1e17, 1e18, 1e19, 1e20, 1e21,
+      1e22, 1e23, 1e24, 1e25, 1e26, 1e27, 1e28, 1e29, 1e30, 1e31};
+  static const double postab32[10] = {1e0,   1e32,  1e64,  1e96,  1e128,
+                                      1e160, 1e192, 1e224, 1e256, 1e288};
+  static const double negtab32[11] = {1e0,    1e-32,  1e-64,  1e-96,
+                                      1e-128, 1e-160, 1e-192, 1e-224,
+                                      1e-256, 1e-288, 1e-320};
+  if (0 <= n && n <= 308) return postab32[n / 32] * tab[n % 32];
+  if (-323 <= n && n < 0) return negtab32[(-n) / 32] / tab[(-n) % 32];
+  if (n > 308) return INFINITY;
+  return 0.0;
+}
+
 double vm_decimal_round_to_decimal_digits(double f, int digits) {
   uint64_t bits;
   memcpy(&bits, &f, 8);
   if (bits == STALE_NAN_BITS) return f;
   if (digits <= -100 || digits >= 100) return f;
-  double m = pow(10.0, (double)digits);
+  double m = go_pow10(digits);
   return round(f * m) / m;
 }
 

@@ -399,3 +399,51 @@ def test_cache_get_best_key_fuzz_vs_model():
         np.testing.assert_array_equal(v.view(np.int64),
                                       vals_b[:, i:j].view(np.int64))
         assert ns == int(ts_b[j - 1]) + step
+
+
+def test_cache_persistence_roundtrip(tmp_path):
+    """InitRollupResultCache/StopRollupResultCache disk round trip
+    (rollup_result_cache.go:119-199): series entries, multi-range metainfo
+    and instant values survive a save/load; blobs stay in the
+    marshalTimeseriesFast layout."""
+    from victoriametrics_amd.cache import (RollupResultCache,
+                                           load_rollup_result_cache,
+                                           save_rollup_result_cache)
+    c = RollupResultCache()
+    names = [(b"m", ((b"a", b"b"),))]
+    ts1 = np.arange(1_000_000, 1_000_000 + 10 * 15_000, 15_000,
+                    dtype=np.int64)
+    v1 = np.arange(10, dtype=np.float64).reshape(1, 10)
+    now = int(ts1[-1]) + 10_000_000
+    c.put_series("rate(m)", 300_000, 15_000, names, v1, ts1, now_ms=now)
+    ts2 = ts1 + 20 * 15_000
+    c.put_series("rate(m)", 300_000, 15_000, names, v1 * 2.0, ts2,
+                 now_ms=now)
+    c.put_instant_values("sum(m)", 0, 15_000, names,
+                         np.array([[42.0]]), np.array([123_000], np.int64))
+
+    path = str(tmp_path / "rollupResult")
+    save_rollup_result_cache(c, path)
+    c2 = load_rollup_result_cache(path)
+
+    got_n, got_v, got_t, new_start = c2.get_series(
+        "rate(m)", 300_000, 15_000, int(ts1[0]), int(ts1[-1]))
+    assert got_n is not None
+    np.testing.assert_array_equal(got_t, ts1)
+    np.testing.assert_array_equal(got_v, v1)
+    got_n2, got_v2, got_t2, _ = c2.get_series(
+        "rate(m)", 300_000, 15_000, int(ts2[0]), int(ts2[-1]))
+    np.testing.assert_array_equal(got_v2, v1 * 2.0)
+    inames, ivals, its = c2.get_instant_values("sum(m)", 0, 15_000)
+    assert its == 123_000 and ivals[0, 0] == 42.0
+    assert c2._size == c._size
+
+
+def test_cache_persistence_missing_or_corrupt(tmp_path):
+    from victoriametrics_amd.cache import load_rollup_result_cache
+    c = load_rollup_result_cache(str(tmp_path / "nope"))
+    assert c.get_series("x", 0, 15_000, 0, 15_000)[0] is None
+    bad = tmp_path / "bad"
+    bad.write_bytes(b"garbage")
+    c2 = load_rollup_result_cache(str(bad))
+    assert c2.get_series("x", 0, 15_000, 0, 15_000)[0] is None

@@ -342,3 +342,94 @@ class RollupResultCache:
         out_ts = timestamps[i:j]
         new_start = int(out_ts[-1]) + step
         return names, values[:, i:j], out_ts, new_start
+
+
+# -- disk persistence (InitRollupResultCache / StopRollupResultCache,
+# rollup_result_cache.go:119-199: with a cachePath the fastcache working
+# set is loaded at startup and saved at shutdown).  The blob payloads stay
+# in the marshalTimeseriesFast binary layout; the index around them is this
+# engine's own compact framing (the reference's fastcache file format is an
+# implementation detail of its cache library, not of the query path).
+
+_CACHE_MAGIC = b"vmgpu-rollupcache-v1\n"
+
+
+def _w_bytes(f, b):
+    f.write(struct.pack(">I", len(b)))
+    f.write(b)
+
+
+def _r_bytes(f):
+    (n,) = struct.unpack(">I", f.read(4))
+    return f.read(n)
+
+
+def save_rollup_result_cache(cache, path):
+    """StopRollupResultCache's save half (rollup_result_cache.go:190-199):
+    persist every live series entry and instant entry."""
+    import os
+    tmp = path + ".tmp"
+    with open(tmp, "wb") as f:
+        f.write(_CACHE_MAGIC)
+        # series entries (only those whose blob is still resident)
+        live = []
+        for key, entries in cache._meta.items():
+            for e_start, e_end, skey in entries:
+                blob = cache._blobs.get(skey)
+                if blob is not None:
+                    live.append((key, e_start, e_end, blob))
+        f.write(struct.pack(">I", len(live)))
+        for (expr, window, step, filters), e_start, e_end, blob in live:
+            _w_bytes(f, expr.encode("utf-8", "surrogateescape"))
+            f.write(struct.pack(">qq", window, step))
+            _w_bytes(f, filters)
+            f.write(struct.pack(">qq", e_start, e_end))
+            _w_bytes(f, blob)
+        # instant entries
+        instants = [(k, b) for k, b in cache._blobs.items()
+                    if isinstance(k, tuple)]
+        f.write(struct.pack(">I", len(instants)))
+        for (_, expr, window, step, filters), blob in instants:
+            _w_bytes(f, expr.encode("utf-8", "surrogateescape"))
+            f.write(struct.pack(">qq", window, step))
+            _w_bytes(f, filters)
+            _w_bytes(f, blob)
+    os.replace(tmp, path)
+
+
+def load_rollup_result_cache(path, max_bytes=256 << 20):
+    """InitRollupResultCache's load half: returns a RollupResultCache
+    seeded from `path`, or an empty one when the file is missing or
+    corrupt (fastcache.Load semantics — a bad cache never fails startup)."""
+    cache = RollupResultCache(max_bytes=max_bytes)
+    try:
+        with open(path, "rb") as f:
+            if f.read(len(_CACHE_MAGIC)) != _CACHE_MAGIC:
+                return cache
+            (n_series,) = struct.unpack(">I", f.read(4))
+            for _ in range(n_series):
+                expr = _r_bytes(f).decode("utf-8", "surrogateescape")
+                window, step = struct.unpack(">qq", f.read(16))
+                filters = _r_bytes(f)
+                e_start, e_end = struct.unpack(">qq", f.read(16))
+                blob = _r_bytes(f)
+                key = cache._key(expr, window, step, filters)
+                cache._suffix += 1
+                skey = cache._suffix
+                cache._blobs[skey] = blob
+                cache._size += len(blob)
+                cache._meta.setdefault(key, []).append(
+                    [e_start, e_end, skey])
+            (n_instant,) = struct.unpack(">I", f.read(4))
+            for _ in range(n_instant):
+                expr = _r_bytes(f).decode("utf-8", "surrogateescape")
+                window, step = struct.unpack(">qq", f.read(16))
+                filters = _r_bytes(f)
+                blob = _r_bytes(f)
+                k = cache._ikey(expr, window, step, filters)
+                cache._blobs[k] = blob
+                cache._size += len(blob)
+            cache._evict()
+    except (OSError, struct.error):
+        return RollupResultCache(max_bytes=max_bytes)
+    return cache

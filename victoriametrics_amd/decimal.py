@@ -37,14 +37,33 @@ def go_round(values):
     return out.view(np.float64)
 
 
+# Go math.Pow10 (math/pow10.go): exact table-product construction — can
+# differ from libm pow(10, n) by 1 ulp at extreme n, and the reference's
+# RoundToDecimalDigits divides by exactly this value.
+_POW10TAB = [float(f"1e{i}") for i in range(32)]
+_POW10POSTAB32 = [float(f"1e{32 * i}") for i in range(10)]
+_POW10NEGTAB32 = [float(f"1e-{32 * i}") for i in range(11)]
+
+
+def go_pow10(n):
+    n = int(n)
+    if 0 <= n <= 308:
+        return _POW10POSTAB32[n // 32] * _POW10TAB[n % 32]
+    if -323 <= n < 0:
+        return _POW10NEGTAB32[(-n) // 32] / _POW10TAB[(-n) % 32]
+    return math.inf if n > 308 else 0.0
+
+
 def round_to_decimal_digits(values, digits):
     """RoundToDecimalDigits (decimal.go:325-335), elementwise over an
-    array; stale-NaN marks pass through untouched."""
+    array; stale-NaN marks pass through untouched.  The 10^digits factor
+    uses the Go Pow10 table product, not libm pow (1-ulp parity at
+    |digits| >= 23)."""
     v = np.ascontiguousarray(values, np.float64)
     if digits <= -100 or digits >= 100:
         return v
     stale = v.view(np.uint64) == STALE_NAN_BITS
-    m = math.pow(10.0, digits)
+    m = go_pow10(digits)
     with np.errstate(invalid="ignore", over="ignore"):
         r = go_round(v * m) / m
     return np.where(stale, v, r)

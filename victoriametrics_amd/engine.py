@@ -661,11 +661,16 @@ def topk_range(batch, k, summary="avg", reverse=False, remaining=False):
     """getRangeTopKTimeseries over the batch's last evaluated output:
     returns (selected row ids in output order, remaining-sum row or None)."""
     lib = _load_lib()
-    kk = max(int(k) if k == k and k > 0 else 0, 0)
+    rows = batch._last_rows or batch.n_series
+    # topk(inf, q) selects everything (lessWithNaNs tolerates any float k);
+    # clamp BEFORE int() so non-finite / huge k cannot overflow
+    kf = float(k)
+    if kf != kf or kf <= 0:
+        kk = 0
+    else:
+        kk = rows if kf >= rows else int(kf)
     sel = np.empty(max(kk, 1), dtype=np.int64)
     n_sel = ctypes.c_int64(0)
-    rows = batch.n_groups if batch.group_ids is not None else batch.n_series
-    del rows
     rem = None
     rptr = None
     if remaining:
