@@ -692,8 +692,6 @@ int64_t vm_pack_blocks(const int64_t* ts, const int64_t* vals,
     if (n == 0 || n > 8192) return -2; /* one block per series only */
     int64_t worst = (int64_t)sizeof(vm_packed_hdr) + 2 * (n * 10 + 64);
     if (w + worst > cap) return -1;
-    vm_packed_hdr* h = (vm_packed_hdr*)(dst + w);
-    memset(h, 0, sizeof(*h));
     uint8_t* body = dst + w + sizeof(vm_packed_hdr);
     int64_t tfirst = 0, vfirst = 0;
     uint8_t tmt = 0, vmt = 0;
@@ -702,16 +700,21 @@ int64_t vm_pack_blocks(const int64_t* ts, const int64_t* vals,
     int64_t vlen = marshal_nozstd(body + tlen, vals + lo, n, precision_bits,
                                   &vmt, &vfirst);
     if (vlen < 0) return -3;
-    h->min_timestamp = ts[lo];
-    h->max_timestamp = ts[lo + n - 1];
-    h->first_value = vfirst;
-    h->rows = (uint32_t)n;
-    h->scale = 0;
-    h->ts_data_len = (uint32_t)tlen;
-    h->val_data_len = (uint32_t)vlen;
-    h->ts_mt = tmt;
-    h->val_mt = vmt;
-    h->precision_bits = precision_bits;
+    /* the stream position is not 8-aligned in general: build the header
+     * locally and memcpy (the parser reads it the same way) */
+    vm_packed_hdr h;
+    memset(&h, 0, sizeof(h));
+    h.min_timestamp = ts[lo];
+    h.max_timestamp = ts[lo + n - 1];
+    h.first_value = vfirst;
+    h.rows = (uint32_t)n;
+    h.scale = 0;
+    h.ts_data_len = (uint32_t)tlen;
+    h.val_data_len = (uint32_t)vlen;
+    h.ts_mt = tmt;
+    h.val_mt = vmt;
+    h.precision_bits = precision_bits;
+    memcpy(dst + w, &h, sizeof(h));
     w += (int64_t)sizeof(vm_packed_hdr) + tlen + vlen;
   }
   if (series_block_start) series_block_start[n_series] = n_series;

@@ -275,7 +275,7 @@ def test_topk_range_gpu(engine, counter_small):
     plan = engine.RollupPlan("avg_over_time", START, end, STEP, window=300_000)
     with engine.SeriesBatch(ts, vals, offsets) as b:
         host_out, _, _ = b.exec(plan)
-        for summary in ("avg", "min", "max", "last"):
+        for summary in ("avg", "min", "max", "median", "last"):
             for rev in (False, True):
                 sel, rem = engine.topk_range(b, 100, summary=summary,
                                              reverse=rev, remaining=True)
@@ -497,3 +497,33 @@ def test_aggr_over_time_expansion(engine):
         out, _, _ = engine.rollup_eval(plan, ts, vals, offsets)
         ref, _, _ = _oracle_batch(plan, ts, vals, offsets)
         assert_parity(out, ref, exact=True, context=f"aggr_over_time/{tag}")
+
+
+def test_topk_median_device_vs_oracle(engine):
+    """Device median selection (8-pass byte-histogram rank selection in
+    topk_summary_kernel) vs the oracle's quantile(0.5): exact equality of
+    the selected summary values, including NaN rows, duplicates and
+    even/odd non-NaN counts."""
+    rng = np.random.default_rng(42)
+    n_series, n = 500, 97
+    vals = rng.standard_normal((n_series, n)) * 100
+    # duplicates + NaN holes + all-NaN rows + constant rows
+    vals[rng.random((n_series, n)) < 0.15] = np.nan
+    vals[7, :] = np.nan
+    vals[11, :] = 3.25
+    vals[13, : n // 2] = -1.5
+    ts = np.tile(START + np.arange(n, dtype=np.int64) * STEP, n_series)
+    offsets = np.arange(n_series + 1, dtype=np.uint64) * n
+    plan = engine.RollupPlan("last_over_time", START,
+                             START + (n - 1) * STEP, STEP, window=STEP)
+    with engine.SeriesBatch(ts.copy(), vals.reshape(-1).copy(), offsets) as b:
+        host_out, _, _ = b.exec(plan)
+        for rev in (False, True):
+            sel, _ = engine.topk_range(b, 50, summary="median", reverse=rev)
+            ref_sel, _ = oracle.topk_range(host_out, 50, "median",
+                                           reverse=rev)
+            sv = [oracle.topk_summary("median", host_out[i]) for i in sel]
+            rv = [oracle.topk_summary("median", host_out[i])
+                  for i in ref_sel]
+            assert [repr(x) for x in sv] == [repr(x) for x in rv], \
+                f"rev={rev}"

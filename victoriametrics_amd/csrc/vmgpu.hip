@@ -1867,10 +1867,68 @@ static VM_DEV unsigned long long vm_topk_key(double v, int reverse) {
   return reverse ? ~ord : ord;
 }
 
+/* inverse of vm_topk_key's order map (non-NaN keys only) */
+static VM_DEV double vm_topk_key_inv(unsigned long long ord) {
+  unsigned long long b = (ord & 0x8000000000000000ULL)
+                             ? (ord ^ 0x8000000000000000ULL)
+                             : ~ord;
+  return __longlong_as_double((long long)b);
+}
+
+/* k-th smallest (0-based) non-NaN key of a row: 8 passes of 256-bin LDS
+ * histogram refinement over the order-preserving u64 keys (exact — every
+ * byte of the result is pinned by counts).  hist: per-wave uint32[256]. */
+static __device__ unsigned long long topk_row_kth(
+    const double* row, int32_t n_grid, uint32_t k, int lane, uint32_t* hist) {
+  unsigned long long prefix = 0;
+  for (int byte = 7; byte >= 0; byte--) {
+    const int shift = byte * 8;
+    for (int b = lane; b < 256; b += WAVE) hist[b] = 0;
+    wave_ds_sync();
+    const unsigned long long pmask =
+        (byte == 7) ? 0ULL : (~0ULL << (shift + 8));
+    for (int g = lane; g < n_grid; g += WAVE) {
+      double v = row[g];
+      if (vm_isnan(v)) continue;
+      unsigned long long key = vm_topk_key(v, 0);
+      if ((key & pmask) != prefix) continue;
+      atomicAdd(&hist[(key >> shift) & 0xff], 1u);
+    }
+    wave_ds_sync();
+    /* each lane owns 4 consecutive bins; exclusive wave prefix of totals */
+    uint32_t b0 = hist[lane * 4], b1 = hist[lane * 4 + 1];
+    uint32_t b2 = hist[lane * 4 + 2], b3 = hist[lane * 4 + 3];
+    uint32_t lt = b0 + b1 + b2 + b3;
+    uint32_t ex = lt;
+    for (int d = 1; d < WAVE; d <<= 1) {
+      uint32_t xo = __shfl_up(ex, d);
+      if (lane >= d) ex += xo;
+    }
+    ex -= lt;
+    uint32_t c0 = ex, c1 = c0 + b0, c2 = c1 + b1, c3 = c2 + b2;
+    int pick = -1;
+    if (k >= c0 && k < c0 + b0) pick = 0;
+    else if (k >= c1 && k < c1 + b1) pick = 1;
+    else if (k >= c2 && k < c2 + b2) pick = 2;
+    else if (k >= c3 && k < c3 + b3) pick = 3;
+    uint64_t mb = __ballot(pick >= 0);
+    int srcl = __ffsll((unsigned long long)mb) - 1;
+    int pickv = __shfl(pick, srcl);
+    uint32_t below = __shfl(pickv == 0 ? c0 : pickv == 1 ? c1
+                            : pickv == 2 ? c2 : c3, srcl);
+    int bin = srcl * 4 + pickv;
+    k -= below;
+    prefix |= ((unsigned long long)(unsigned)bin) << shift;
+    wave_ds_sync();
+  }
+  return prefix;
+}
+
 /* per-series range summaries (aggr.go:804-858), one wave per series row */
 __global__ __launch_bounds__(BLOCK_THREADS) void topk_summary_kernel(
     const double* values, uint32_t n_series, int32_t n_grid, int32_t op,
     int32_t reverse, unsigned long long* keys) {
+  __shared__ uint32_t sh_hist[WAVES_PER_BLOCK][256];
   const int wave_in_block = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
   uint32_t wid = blockIdx.x * WAVES_PER_BLOCK + wave_in_block;
@@ -1902,6 +1960,43 @@ __global__ __launch_bounds__(BLOCK_THREADS) void topk_summary_kernel(
       if (op == 0) acc = (cnt == 0) ? vm_dnan() : sum / cnt;
       else if (op == 1) acc = mn;
       else acc = mx;
+    } else if (op == 3) { /* median = quantile(0.5) (aggr.go:848,922) */
+      uint32_t m = 0;
+      for (int g = lane; g < n_grid; g += WAVE)
+        if (!vm_isnan(row[g])) m++;
+      for (int d = 32; d > 0; d >>= 1) m += __shfl_down(m, d);
+      m = __shfl(m, 0);
+      if (m > 0) {
+        uint32_t li = (m - 1) / 2; /* floor(0.5*(m-1)) */
+        unsigned long long kv =
+            topk_row_kth(row, n_grid, li, lane, sh_hist[wave_in_block]);
+        double vlo = vm_topk_key_inv(kv);
+        if (m & 1) {
+          acc = vlo; /* odd count: weight 0 */
+        } else {
+          /* upper = positional li+1 in the sorted row: equals vlo inside
+           * an equal run, else the smallest key > kv */
+          uint32_t cle = 0; /* #keys <= kv */
+          unsigned long long nxt = ~0ULL;
+          for (int g = lane; g < n_grid; g += WAVE) {
+            double v = row[g];
+            if (vm_isnan(v)) continue;
+            unsigned long long key = vm_topk_key(v, 0);
+            if (key <= kv) cle++;
+            else if (key < nxt) nxt = key;
+          }
+          for (int d = 32; d > 0; d >>= 1) {
+            cle += __shfl_down(cle, d);
+            unsigned long long no = __shfl_down(nxt, d);
+            if (no < nxt) nxt = no;
+          }
+          cle = __shfl(cle, 0);
+          nxt = __shfl(nxt, 0);
+          double vhi = (li + 1 < cle) ? vlo : vm_topk_key_inv(nxt);
+          /* quantileSorted: v[lower]*(1-weight) + v[upper]*weight, w=0.5 */
+          acc = vlo * 0.5 + vhi * 0.5;
+        }
+      }
     } else if (op == 4) { /* last non-NaN */
       int base = ((n_grid + WAVE - 1) / WAVE - 1) * WAVE;
       for (; base >= 0; base -= WAVE) {
