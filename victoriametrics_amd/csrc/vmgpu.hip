@@ -3344,8 +3344,6 @@ int vmgpu_topk_range(uint64_t handle, double k, int32_t summary_op,
   Batch& b = it->second;
   if (!b.d_out || b.last_rows == 0)
     return set_err(errbuf, errbuf_len, "vmgpu: no evaluated output on this batch");
-  if (summary_op == 3)
-    return set_err(errbuf, errbuf_len, "vmgpu: topk_median not implemented on device yet");
   uint32_t n = b.last_rows;
   int32_t n_grid = b.last_grid;
   uint32_t kk = 0;
