@@ -1855,11 +1855,14 @@ __global__ void aggr_finalize_kernel(double* values, double* counts, uint64_t n,
 /* topk family (aggr.go:646-741) + histogram_quantile (transform.go)  */
 /* ------------------------------------------------------------------ */
 
-/* Order-preserving u64 key for f64 under lessWithNaNs (aggr.go:1259):
- * NaN sorts below every number for topk; reverse flips the order and puts
- * NaN on top (greaterWithNaNs), so selection is always "k largest keys". */
+/* Order-preserving u64 key for f64 under lessWithNaNs / greaterWithNaNs
+ * (aggr.go:1259-1279): selection always keeps the "k largest keys".  NaN
+ * maps to the SMALLEST key in both directions — lessWithNaNs treats NaN as
+ * smaller than any number (so topk never keeps it) and greaterWithNaNs
+ * treats it as bigger (so it sorts to the FRONT of the bottomk order and
+ * is again never kept). */
 static VM_DEV unsigned long long vm_topk_key(double v, int reverse) {
-  if (vm_isnan(v)) return reverse ? ~0ULL : 0ULL;
+  if (vm_isnan(v)) return 0ULL;
   unsigned long long b = (unsigned long long)__double_as_longlong(v);
   unsigned long long ord = (b >> 63) ? ~b : (b | 0x8000000000000000ULL);
   /* the NaN sentinels 0 and ~0 are unreachable for non-NaN doubles: their
