@@ -120,20 +120,26 @@ def cold_query(args, engine, plan, ts, vals, offsets):
     pay_buf[:] = np.frombuffer(packed, np.uint8)
     out_raw, out_ptr = engine.host_alloc(n_series * plan.n_grid * 8)
     out_view = out_raw.view(np.float64).reshape(n_series, plan.n_grid)
-    walls = []
+    walls, create_ms, exec_ms = [], [], []
     try:
         for _ in range(args.cold_queries):
             t0 = time.perf_counter()
             b = engine.SeriesBatch.from_packed(pay_buf, n_blocks, sbs)
+            t1 = time.perf_counter()
             b.exec(plan, download=True, out_buf=out_view)
-            walls.append(time.perf_counter() - t0)
+            t2 = time.perf_counter()
             b.close()
+            walls.append(time.perf_counter() - t0)
+            create_ms.append((t1 - t0) * 1e3)
+            exec_ms.append((t2 - t1) * 1e3)
     finally:
         engine.host_free(pay_ptr)
         engine.host_free(out_ptr)
     p50 = float(np.median(walls))
     return {
         "p50_ms": p50 * 1e3,
+        "p50_create_ms": float(np.median(create_ms)),
+        "p50_exec_download_ms": float(np.median(exec_ms)),
         "samples_per_s": samples / p50,
         "queries": args.cold_queries,
         "payload_bytes": len(packed),

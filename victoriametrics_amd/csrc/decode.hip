@@ -27,6 +27,7 @@
  *            bit-identical to the CPU path.
  */
 #include <hip/hip_runtime.h>
+#include "devalloc.h"
 #include <algorithm>
 #include <cstdio>
 #include <cstring>
@@ -337,19 +338,19 @@ int vmgpu_decode_blocks(const uint8_t* payload, uint64_t payload_len,
                         char* errbuf, size_t errbuf_len) {
   if (!payload || !blocks || !out_ts || !out_vals || n_blocks == 0)
     return dset_err(errbuf, errbuf_len, "vmgpu: bad args");
-  hipStream_t st = 0;
+  hipStream_t st = vm_ctx_stream();
   uint8_t* d_payload = nullptr;
   vmgpu_block_desc* d_blocks = nullptr;
   long long* d_scratch = nullptr;
   int64_t* d_ts = nullptr;
   double* d_vals = nullptr;
   int* d_err = nullptr;
-  DHIP_TRY(hipMalloc(&d_payload, payload_len ? payload_len : 1), "alloc payload");
-  DHIP_TRY(hipMalloc(&d_blocks, (size_t)n_blocks * sizeof(vmgpu_block_desc)), "alloc descs");
-  DHIP_TRY(hipMalloc(&d_scratch, (size_t)DGRID * DMAX_ROWS * 8), "alloc scratch");
-  DHIP_TRY(hipMalloc(&d_ts, (size_t)total_rows * 8), "alloc ts");
-  DHIP_TRY(hipMalloc(&d_vals, (size_t)total_rows * 8), "alloc vals");
-  DHIP_TRY(hipMalloc(&d_err, 4), "alloc err");
+  DHIP_TRY(vm_dev_malloc(&d_payload, payload_len ? payload_len : 1), "alloc payload");
+  DHIP_TRY(vm_dev_malloc(&d_blocks, (size_t)n_blocks * sizeof(vmgpu_block_desc)), "alloc descs");
+  DHIP_TRY(vm_dev_malloc(&d_scratch, (size_t)DGRID * DMAX_ROWS * 8), "alloc scratch");
+  DHIP_TRY(vm_dev_malloc(&d_ts, (size_t)total_rows * 8), "alloc ts");
+  DHIP_TRY(vm_dev_malloc(&d_vals, (size_t)total_rows * 8), "alloc vals");
+  DHIP_TRY(vm_dev_malloc(&d_err, 4), "alloc err");
   DHIP_TRY(hipMemcpyAsync(d_payload, payload, payload_len, hipMemcpyHostToDevice, st), "ul payload");
   DHIP_TRY(hipMemcpyAsync(d_blocks, blocks, (size_t)n_blocks * sizeof(vmgpu_block_desc),
                           hipMemcpyHostToDevice, st), "ul descs");
@@ -363,12 +364,12 @@ int vmgpu_decode_blocks(const uint8_t* payload, uint64_t payload_len,
   DHIP_TRY(hipMemcpyAsync(out_vals, d_vals, (size_t)total_rows * 8, hipMemcpyDeviceToHost, st), "dl vals");
   DHIP_TRY(hipStreamSynchronize(st), "sync");
   hipError_t kerr = hipGetLastError();
-  (void)hipFree(d_payload);
-  (void)hipFree(d_blocks);
-  (void)hipFree(d_scratch);
-  (void)hipFree(d_ts);
-  (void)hipFree(d_vals);
-  (void)hipFree(d_err);
+  (void)vm_dev_free(d_payload);
+  (void)vm_dev_free(d_blocks);
+  (void)vm_dev_free(d_scratch);
+  (void)vm_dev_free(d_ts);
+  (void)vm_dev_free(d_vals);
+  (void)vm_dev_free(d_err);
   if (kerr != hipSuccess) return dhip_err(errbuf, errbuf_len, "decode kernel", kerr);
   if (err_h != 0) {
     char msg[64];
@@ -645,7 +646,7 @@ int vmgpu_merge_blocks(const int64_t* ts, const double* vals,
   if (!ts || !vals || !block_offsets || !series_block_start || !out_ts ||
       !out_vals || !out_offsets || !out_counts || n_series == 0)
     return dset_err(errbuf, errbuf_len, "vmgpu: bad args");
-  hipStream_t st = 0;
+  hipStream_t st = vm_ctx_stream();
   uint64_t total = block_offsets[n_blocks];
   /* capacity offsets = pre-merge block extents per series */
   std::vector<uint64_t> cap_off(n_series + 1);
@@ -664,15 +665,15 @@ int vmgpu_merge_blocks(const int64_t* ts, const double* vals,
   double* d_ovals = nullptr;
   uint64_t* d_ocnt = nullptr;
   int* d_err = nullptr;
-  DHIP_TRY(hipMalloc(&d_ts, total * 8), "alloc ts");
-  DHIP_TRY(hipMalloc(&d_vals, total * 8), "alloc vals");
-  DHIP_TRY(hipMalloc(&d_boff, (size_t)(n_blocks + 1) * 8), "alloc boff");
-  DHIP_TRY(hipMalloc(&d_sbs, (size_t)(n_series + 1) * 4), "alloc sbs");
-  DHIP_TRY(hipMalloc(&d_ooff, (size_t)(n_series + 1) * 8), "alloc ooff");
-  DHIP_TRY(hipMalloc(&d_ots, cap_off[n_series] * 8), "alloc out ts");
-  DHIP_TRY(hipMalloc(&d_ovals, cap_off[n_series] * 8), "alloc out vals");
-  DHIP_TRY(hipMalloc(&d_ocnt, (size_t)n_series * 8), "alloc out counts");
-  DHIP_TRY(hipMalloc(&d_err, 4), "alloc err");
+  DHIP_TRY(vm_dev_malloc(&d_ts, total * 8), "alloc ts");
+  DHIP_TRY(vm_dev_malloc(&d_vals, total * 8), "alloc vals");
+  DHIP_TRY(vm_dev_malloc(&d_boff, (size_t)(n_blocks + 1) * 8), "alloc boff");
+  DHIP_TRY(vm_dev_malloc(&d_sbs, (size_t)(n_series + 1) * 4), "alloc sbs");
+  DHIP_TRY(vm_dev_malloc(&d_ooff, (size_t)(n_series + 1) * 8), "alloc ooff");
+  DHIP_TRY(vm_dev_malloc(&d_ots, cap_off[n_series] * 8), "alloc out ts");
+  DHIP_TRY(vm_dev_malloc(&d_ovals, cap_off[n_series] * 8), "alloc out vals");
+  DHIP_TRY(vm_dev_malloc(&d_ocnt, (size_t)n_series * 8), "alloc out counts");
+  DHIP_TRY(vm_dev_malloc(&d_err, 4), "alloc err");
   DHIP_TRY(hipMemcpyAsync(d_ts, ts, total * 8, hipMemcpyHostToDevice, st), "ul ts");
   DHIP_TRY(hipMemcpyAsync(d_vals, vals, total * 8, hipMemcpyHostToDevice, st), "ul vals");
   DHIP_TRY(hipMemcpyAsync(d_boff, block_offsets, (size_t)(n_blocks + 1) * 8, hipMemcpyHostToDevice, st), "ul boff");
@@ -705,9 +706,9 @@ int vmgpu_merge_blocks(const int64_t* ts, const double* vals,
     }
     DHIP_TRY(hipStreamSynchronize(st), "sync dl");
   }
-  (void)hipFree(d_ts); (void)hipFree(d_vals); (void)hipFree(d_boff);
-  (void)hipFree(d_sbs); (void)hipFree(d_ooff); (void)hipFree(d_ots);
-  (void)hipFree(d_ovals); (void)hipFree(d_ocnt); (void)hipFree(d_err);
+  (void)vm_dev_free(d_ts); (void)vm_dev_free(d_vals); (void)vm_dev_free(d_boff);
+  (void)vm_dev_free(d_sbs); (void)vm_dev_free(d_ooff); (void)vm_dev_free(d_ots);
+  (void)vm_dev_free(d_ovals); (void)vm_dev_free(d_ocnt); (void)vm_dev_free(d_err);
   if (kerr != hipSuccess) return dhip_err(errbuf, errbuf_len, "merge kernel", kerr);
   if (err_h != 0) {
     char msg[64];
@@ -799,12 +800,12 @@ extern "C" int vmdec_decode_merge_device(
       goto cleanup; }                                                        \
   } while (0)
 
-  FDM_TRY(hipMalloc(&d_payload, payload_len ? payload_len : 1), "alloc payload");
-  FDM_TRY(hipMalloc(&d_blocks, (size_t)n_blocks * sizeof(vmgpu_block_desc)), "alloc descs");
-  FDM_TRY(hipMalloc(&d_scratch, (size_t)DGRID * DMAX_ROWS * 8), "alloc scratch");
-  FDM_TRY(hipMalloc(&d_raw_ts, (size_t)total_rows * 8), "alloc raw ts");
-  FDM_TRY(hipMalloc(&d_raw_vals, (size_t)total_rows * 8), "alloc raw vals");
-  FDM_TRY(hipMalloc(&d_err, 4), "alloc err");
+  FDM_TRY(vm_dev_malloc(&d_payload, payload_len ? payload_len : 1), "alloc payload");
+  FDM_TRY(vm_dev_malloc(&d_blocks, (size_t)n_blocks * sizeof(vmgpu_block_desc)), "alloc descs");
+  FDM_TRY(vm_dev_malloc(&d_scratch, (size_t)DGRID * DMAX_ROWS * 8), "alloc scratch");
+  FDM_TRY(vm_dev_malloc(&d_raw_ts, (size_t)total_rows * 8), "alloc raw ts");
+  FDM_TRY(vm_dev_malloc(&d_raw_vals, (size_t)total_rows * 8), "alloc raw vals");
+  FDM_TRY(vm_dev_malloc(&d_err, 4), "alloc err");
   FDM_TRY(hipMemcpyAsync(d_payload, payload, payload_len ? payload_len : 1,
                          hipMemcpyHostToDevice, st), "ul payload");
   FDM_TRY(hipMemcpyAsync(d_blocks, blocks, (size_t)n_blocks * sizeof(vmgpu_block_desc),
@@ -815,12 +816,12 @@ extern "C" int vmdec_decode_merge_device(
                      d_payload, d_blocks, n_blocks, d_scratch,
                      d_raw_ts, d_raw_vals, d_err);
 
-  FDM_TRY(hipMalloc(&d_boff, (size_t)(n_blocks + 1) * 8), "alloc boff");
-  FDM_TRY(hipMalloc(&d_sbs, (size_t)(n_series + 1) * 4), "alloc sbs");
-  FDM_TRY(hipMalloc(&d_capoff, (size_t)(n_series + 1) * 8), "alloc capoff");
-  FDM_TRY(hipMalloc(&d_mts, (cap_off[n_series] ? cap_off[n_series] : 1) * 8), "alloc merged ts");
-  FDM_TRY(hipMalloc(&d_mvals, (cap_off[n_series] ? cap_off[n_series] : 1) * 8), "alloc merged vals");
-  FDM_TRY(hipMalloc(&d_cnt, (size_t)n_series * 8), "alloc counts");
+  FDM_TRY(vm_dev_malloc(&d_boff, (size_t)(n_blocks + 1) * 8), "alloc boff");
+  FDM_TRY(vm_dev_malloc(&d_sbs, (size_t)(n_series + 1) * 4), "alloc sbs");
+  FDM_TRY(vm_dev_malloc(&d_capoff, (size_t)(n_series + 1) * 8), "alloc capoff");
+  FDM_TRY(vm_dev_malloc(&d_mts, (cap_off[n_series] ? cap_off[n_series] : 1) * 8), "alloc merged ts");
+  FDM_TRY(vm_dev_malloc(&d_mvals, (cap_off[n_series] ? cap_off[n_series] : 1) * 8), "alloc merged vals");
+  FDM_TRY(vm_dev_malloc(&d_cnt, (size_t)n_series * 8), "alloc counts");
   FDM_TRY(hipMemcpyAsync(d_boff, boff.data(), (size_t)(n_blocks + 1) * 8,
                          hipMemcpyHostToDevice, st), "ul boff");
   FDM_TRY(hipMemcpyAsync(d_sbs, series_block_start, (size_t)(n_series + 1) * 4,
@@ -847,11 +848,11 @@ extern "C" int vmdec_decode_merge_device(
   h_final_offsets[0] = 0;
   for (uint32_t s = 0; s < n_series; s++)
     h_final_offsets[s + 1] = h_final_offsets[s] + counts[s];
-  FDM_TRY(hipMalloc(&d_finaloff, (size_t)(n_series + 1) * 8), "alloc finaloff");
+  FDM_TRY(vm_dev_malloc(&d_finaloff, (size_t)(n_series + 1) * 8), "alloc finaloff");
   FDM_TRY(hipMemcpyAsync(d_finaloff, h_final_offsets, (size_t)(n_series + 1) * 8,
                          hipMemcpyHostToDevice, st), "ul finaloff");
-  FDM_TRY(hipMalloc(&d_fts, (h_final_offsets[n_series] ? h_final_offsets[n_series] : 1) * 8), "alloc final ts");
-  FDM_TRY(hipMalloc(&d_fvals, (h_final_offsets[n_series] ? h_final_offsets[n_series] : 1) * 8), "alloc final vals");
+  FDM_TRY(vm_dev_malloc(&d_fts, (h_final_offsets[n_series] ? h_final_offsets[n_series] : 1) * 8), "alloc final ts");
+  FDM_TRY(vm_dev_malloc(&d_fvals, (h_final_offsets[n_series] ? h_final_offsets[n_series] : 1) * 8), "alloc final vals");
   hipLaunchKernelGGL(compact_series_kernel, dim3(mgrid), dim3(DBLOCK), 0, st,
                      d_mts, d_mvals, d_capoff, d_finaloff, n_series,
                      d_fts, d_fvals);
@@ -865,11 +866,11 @@ extern "C" int vmdec_decode_merge_device(
   d_fvals = nullptr;
 
 cleanup:
-  (void)hipFree(d_payload); (void)hipFree(d_blocks); (void)hipFree(d_scratch);
-  (void)hipFree(d_raw_ts); (void)hipFree(d_raw_vals); (void)hipFree(d_boff);
-  (void)hipFree(d_sbs); (void)hipFree(d_capoff); (void)hipFree(d_mts);
-  (void)hipFree(d_mvals); (void)hipFree(d_cnt); (void)hipFree(d_finaloff);
-  (void)hipFree(d_err); (void)hipFree(d_fts); (void)hipFree(d_fvals);
+  (void)vm_dev_free(d_payload); (void)vm_dev_free(d_blocks); (void)vm_dev_free(d_scratch);
+  (void)vm_dev_free(d_raw_ts); (void)vm_dev_free(d_raw_vals); (void)vm_dev_free(d_boff);
+  (void)vm_dev_free(d_sbs); (void)vm_dev_free(d_capoff); (void)vm_dev_free(d_mts);
+  (void)vm_dev_free(d_mvals); (void)vm_dev_free(d_cnt); (void)vm_dev_free(d_finaloff);
+  (void)vm_dev_free(d_err); (void)vm_dev_free(d_fts); (void)vm_dev_free(d_fvals);
 #undef FDM_TRY
   return rc;
 }

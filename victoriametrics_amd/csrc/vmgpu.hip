@@ -38,6 +38,7 @@
 #include <mutex>
 #include <vector>
 
+#include "devalloc.h"
 #include "rollup_device.h"
 #include "../../include/vmgpu.h"
 
@@ -2561,6 +2562,17 @@ Ctx g_ctx;
 
 static hipStream_t launch_stream() { return g_ctx.stream; }
 
+/* stream-ordered pool allocation (devalloc.h) */
+hipError_t vm_dev_malloc_raw(void** p, size_t n) {
+  return hipMallocAsync(p, n, g_ctx.stream);
+}
+hipError_t vm_dev_free_raw(void* p) {
+  if (!p) return hipSuccess;
+  return hipFreeAsync(p, g_ctx.stream);
+}
+
+hipStream_t vm_ctx_stream(void) { return g_ctx.stream; }
+
 namespace {
 
 int set_err(char* errbuf, size_t len, const char* msg) {
@@ -2637,11 +2649,11 @@ static int relayout_by_group(Batch& b, const int32_t* group_ids,
   uint64_t* d_soff = nullptr;
   uint64_t* d_doff = nullptr;
   uint32_t* d_perm = nullptr;
-  HIP_TRY(hipMalloc(&d_ts2, total * 8), "alloc relayout ts");
-  HIP_TRY(hipMalloc(&d_vals2, total * 8), "alloc relayout vals");
-  HIP_TRY(hipMalloc(&d_soff, (size_t)(n_series + 1) * 8), "alloc relayout soff");
-  HIP_TRY(hipMalloc(&d_doff, (size_t)(n_series + 1) * 8), "alloc relayout doff");
-  HIP_TRY(hipMalloc(&d_perm, (size_t)n_series * 4), "alloc relayout perm");
+  HIP_TRY(vm_dev_malloc(&d_ts2, total * 8), "alloc relayout ts");
+  HIP_TRY(vm_dev_malloc(&d_vals2, total * 8), "alloc relayout vals");
+  HIP_TRY(vm_dev_malloc(&d_soff, (size_t)(n_series + 1) * 8), "alloc relayout soff");
+  HIP_TRY(vm_dev_malloc(&d_doff, (size_t)(n_series + 1) * 8), "alloc relayout doff");
+  HIP_TRY(vm_dev_malloc(&d_perm, (size_t)n_series * 4), "alloc relayout perm");
   hipStream_t st = g_ctx.stream;
   HIP_TRY(hipMemcpyAsync(d_soff, offsets_orig, (size_t)(n_series + 1) * 8,
                          hipMemcpyHostToDevice, st), "ul soff");
@@ -2656,16 +2668,16 @@ static int relayout_by_group(Batch& b, const int32_t* group_ids,
                      d_ts2, d_vals2);
   HIP_TRY(hipStreamSynchronize(st), "sync relayout");
   hipError_t kerr = hipGetLastError();
-  (void)hipFree(d_soff);
-  (void)hipFree(d_doff);
-  (void)hipFree(d_perm);
+  (void)vm_dev_free(d_soff);
+  (void)vm_dev_free(d_doff);
+  (void)vm_dev_free(d_perm);
   if (kerr != hipSuccess) {
-    (void)hipFree(d_ts2);
-    (void)hipFree(d_vals2);
+    (void)vm_dev_free(d_ts2);
+    (void)vm_dev_free(d_vals2);
     return hip_err(errbuf, errbuf_len, "relayout kernel", kerr);
   }
-  (void)hipFree(b.d_ts);
-  (void)hipFree(b.d_vals);
+  (void)vm_dev_free(b.d_ts);
+  (void)vm_dev_free(b.d_vals);
   b.d_ts = d_ts2;
   b.d_vals = d_vals2;
   b.perm = std::move(perm);
@@ -2673,7 +2685,7 @@ static int relayout_by_group(Batch& b, const int32_t* group_ids,
 }
 
 int build_si_index(Batch& b, char* errbuf, size_t errbuf_len) {
-  HIP_TRY(hipMalloc(&b.d_si, (size_t)b.n_series * 8), "alloc si index");
+  HIP_TRY(vm_dev_malloc(&b.d_si, (size_t)b.n_series * 8), "alloc si index");
   uint32_t blocks = std::min<uint32_t>(
       (b.n_series + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK, 4096);
   hipLaunchKernelGGL(si_prep_kernel, dim3(blocks), dim3(BLOCK_THREADS), 0,
@@ -2686,20 +2698,20 @@ int build_si_index(Batch& b, char* errbuf, size_t errbuf_len) {
 }
 
 void free_batch(Batch& b) {
-  (void)hipFree(b.d_ts);
-  (void)hipFree(b.d_vals);
-  (void)hipFree(b.d_offsets);
-  (void)hipFree(b.d_group_ids);
-  (void)hipFree(b.d_si);
-  (void)hipFree(b.d_wave_list);
-  (void)hipFree(b.d_block_list);
-  (void)hipFree(b.d_huge_list);
-  (void)hipFree(b.d_huge_scr_offsets);
-  (void)hipFree(b.d_scr_ts);
-  (void)hipFree(b.d_scr_vals);
-  (void)hipFree(b.d_out);
-  (void)hipFree(b.d_counts);
-  (void)hipFree(b.d_scanned);
+  (void)vm_dev_free(b.d_ts);
+  (void)vm_dev_free(b.d_vals);
+  (void)vm_dev_free(b.d_offsets);
+  (void)vm_dev_free(b.d_group_ids);
+  (void)vm_dev_free(b.d_si);
+  (void)vm_dev_free(b.d_wave_list);
+  (void)vm_dev_free(b.d_block_list);
+  (void)vm_dev_free(b.d_huge_list);
+  (void)vm_dev_free(b.d_huge_scr_offsets);
+  (void)vm_dev_free(b.d_scr_ts);
+  (void)vm_dev_free(b.d_scr_vals);
+  (void)vm_dev_free(b.d_out);
+  (void)vm_dev_free(b.d_counts);
+  (void)vm_dev_free(b.d_scanned);
   b = Batch();
 }
 
@@ -2714,6 +2726,15 @@ int vmgpu_init(const int* device_ids, int n_devices) {
   int dev = device_ids ? device_ids[0] : 0;
   if (hipSetDevice(dev) != hipSuccess) return 2;
   if (hipStreamCreate(&g_ctx.stream) != hipSuccess) return 3;
+  {
+    /* retain freed device memory in the pool: repeat queries re-use the
+     * same multi-GB carve instead of re-mapping it */
+    hipMemPool_t pool = nullptr;
+    if (hipDeviceGetDefaultMemPool(&pool, dev) == hipSuccess && pool) {
+      uint64_t thr = ~0ULL;
+      (void)hipMemPoolSetAttribute(pool, hipMemPoolAttrReleaseThreshold, &thr);
+    }
+  }
   if (hipEventCreate(&g_ctx.ev_start) != hipSuccess) return 4;
   if (hipEventCreate(&g_ctx.ev_stop) != hipSuccess) return 5;
   g_ctx.device = dev;
@@ -2748,8 +2769,8 @@ int vmgpu_batch_create(const int64_t* ts, const double* vals,
   b.n_groups = n_groups;
   b.n_samples = offsets[n_series];
 
-  HIP_TRY(hipMalloc(&b.d_ts, b.n_samples * sizeof(int64_t)), "alloc ts");
-  HIP_TRY(hipMalloc(&b.d_vals, b.n_samples * sizeof(double)), "alloc vals");
+  HIP_TRY(vm_dev_malloc(&b.d_ts, b.n_samples * sizeof(int64_t)), "alloc ts");
+  HIP_TRY(vm_dev_malloc(&b.d_vals, b.n_samples * sizeof(double)), "alloc vals");
   HIP_TRY(hipMemcpy(b.d_ts, ts, b.n_samples * sizeof(int64_t), hipMemcpyHostToDevice), "upload ts");
   HIP_TRY(hipMemcpy(b.d_vals, vals, b.n_samples * sizeof(double), hipMemcpyHostToDevice), "upload vals");
 
@@ -2800,29 +2821,29 @@ int vmgpu_batch_create(const int64_t* ts, const double* vals,
   b.wave_is_identity = (b.n_wave == n_series);
   b.huge_scratch_elems = huge_total;
 
-  HIP_TRY(hipMalloc(&b.d_offsets, (n_series + 1) * sizeof(uint64_t)), "alloc offsets");
+  HIP_TRY(vm_dev_malloc(&b.d_offsets, (n_series + 1) * sizeof(uint64_t)), "alloc offsets");
   HIP_TRY(hipMemcpy(b.d_offsets, eff_offsets, (n_series + 1) * sizeof(uint64_t), hipMemcpyHostToDevice), "upload offsets");
   if (group_ids) {
-    HIP_TRY(hipMalloc(&b.d_group_ids, n_series * sizeof(int32_t)), "alloc gids");
+    HIP_TRY(vm_dev_malloc(&b.d_group_ids, n_series * sizeof(int32_t)), "alloc gids");
     HIP_TRY(hipMemcpy(b.d_group_ids, eff_gids, n_series * sizeof(int32_t), hipMemcpyHostToDevice), "upload gids");
   }
   if (!b.wave_is_identity && b.n_wave) {
-    HIP_TRY(hipMalloc(&b.d_wave_list, b.n_wave * 4), "alloc wave list");
+    HIP_TRY(vm_dev_malloc(&b.d_wave_list, b.n_wave * 4), "alloc wave list");
     HIP_TRY(hipMemcpy(b.d_wave_list, wave_list.data(), b.n_wave * 4, hipMemcpyHostToDevice), "upload wave list");
   }
   if (b.n_block) {
-    HIP_TRY(hipMalloc(&b.d_block_list, b.n_block * 4), "alloc block list");
+    HIP_TRY(vm_dev_malloc(&b.d_block_list, b.n_block * 4), "alloc block list");
     HIP_TRY(hipMemcpy(b.d_block_list, block_list.data(), b.n_block * 4, hipMemcpyHostToDevice), "upload block list");
   }
   if (b.n_huge) {
-    HIP_TRY(hipMalloc(&b.d_huge_list, b.n_huge * 4), "alloc huge list");
+    HIP_TRY(vm_dev_malloc(&b.d_huge_list, b.n_huge * 4), "alloc huge list");
     HIP_TRY(hipMemcpy(b.d_huge_list, huge_list.data(), b.n_huge * 4, hipMemcpyHostToDevice), "upload huge list");
-    HIP_TRY(hipMalloc(&b.d_huge_scr_offsets, b.n_huge * 8), "alloc huge offsets");
+    HIP_TRY(vm_dev_malloc(&b.d_huge_scr_offsets, b.n_huge * 8), "alloc huge offsets");
     HIP_TRY(hipMemcpy(b.d_huge_scr_offsets, huge_scr_off.data(), b.n_huge * 8, hipMemcpyHostToDevice), "upload huge offsets");
-    HIP_TRY(hipMalloc(&b.d_scr_ts, huge_total * sizeof(int64_t)), "alloc scratch ts");
-    HIP_TRY(hipMalloc(&b.d_scr_vals, huge_total * sizeof(double)), "alloc scratch vals");
+    HIP_TRY(vm_dev_malloc(&b.d_scr_ts, huge_total * sizeof(int64_t)), "alloc scratch ts");
+    HIP_TRY(vm_dev_malloc(&b.d_scr_vals, huge_total * sizeof(double)), "alloc scratch vals");
   }
-  HIP_TRY(hipMalloc(&b.d_scanned, sizeof(unsigned long long)), "alloc scanned");
+  HIP_TRY(vm_dev_malloc(&b.d_scanned, sizeof(unsigned long long)), "alloc scanned");
   {
     int src_rc = build_si_index(b, errbuf, errbuf_len);
     if (src_rc != 0) { free_batch(b); return src_rc; }
@@ -2925,35 +2946,35 @@ int vmgpu_batch_create_from_blocks(
       return hip_err(errbuf, errbuf_len, what, _e); }                        \
   } while (0)
 
-  FBB_TRY(hipMalloc(&b.d_offsets, (n_series + 1) * sizeof(uint64_t)), "alloc offsets");
+  FBB_TRY(vm_dev_malloc(&b.d_offsets, (n_series + 1) * sizeof(uint64_t)), "alloc offsets");
   FBB_TRY(hipMemcpy(b.d_offsets, eff_offsets, (n_series + 1) * sizeof(uint64_t),
                     hipMemcpyHostToDevice), "upload offsets");
   if (group_ids) {
-    FBB_TRY(hipMalloc(&b.d_group_ids, n_series * sizeof(int32_t)), "alloc gids");
+    FBB_TRY(vm_dev_malloc(&b.d_group_ids, n_series * sizeof(int32_t)), "alloc gids");
     FBB_TRY(hipMemcpy(b.d_group_ids, eff_gids, n_series * sizeof(int32_t),
                       hipMemcpyHostToDevice), "upload gids");
   }
   if (!b.wave_is_identity && b.n_wave) {
-    FBB_TRY(hipMalloc(&b.d_wave_list, b.n_wave * 4), "alloc wave list");
+    FBB_TRY(vm_dev_malloc(&b.d_wave_list, b.n_wave * 4), "alloc wave list");
     FBB_TRY(hipMemcpy(b.d_wave_list, wave_list.data(), b.n_wave * 4,
                       hipMemcpyHostToDevice), "upload wave list");
   }
   if (b.n_block) {
-    FBB_TRY(hipMalloc(&b.d_block_list, b.n_block * 4), "alloc block list");
+    FBB_TRY(vm_dev_malloc(&b.d_block_list, b.n_block * 4), "alloc block list");
     FBB_TRY(hipMemcpy(b.d_block_list, block_list.data(), b.n_block * 4,
                       hipMemcpyHostToDevice), "upload block list");
   }
   if (b.n_huge) {
-    FBB_TRY(hipMalloc(&b.d_huge_list, b.n_huge * 4), "alloc huge list");
+    FBB_TRY(vm_dev_malloc(&b.d_huge_list, b.n_huge * 4), "alloc huge list");
     FBB_TRY(hipMemcpy(b.d_huge_list, huge_list.data(), b.n_huge * 4,
                       hipMemcpyHostToDevice), "upload huge list");
-    FBB_TRY(hipMalloc(&b.d_huge_scr_offsets, b.n_huge * 8), "alloc huge offsets");
+    FBB_TRY(vm_dev_malloc(&b.d_huge_scr_offsets, b.n_huge * 8), "alloc huge offsets");
     FBB_TRY(hipMemcpy(b.d_huge_scr_offsets, huge_scr_off.data(), b.n_huge * 8,
                       hipMemcpyHostToDevice), "upload huge offsets");
-    FBB_TRY(hipMalloc(&b.d_scr_ts, huge_total * sizeof(int64_t)), "alloc scratch ts");
-    FBB_TRY(hipMalloc(&b.d_scr_vals, huge_total * sizeof(double)), "alloc scratch vals");
+    FBB_TRY(vm_dev_malloc(&b.d_scr_ts, huge_total * sizeof(int64_t)), "alloc scratch ts");
+    FBB_TRY(vm_dev_malloc(&b.d_scr_vals, huge_total * sizeof(double)), "alloc scratch vals");
   }
-  FBB_TRY(hipMalloc(&b.d_scanned, sizeof(unsigned long long)), "alloc scanned");
+  FBB_TRY(vm_dev_malloc(&b.d_scanned, sizeof(unsigned long long)), "alloc scanned");
 #undef FBB_TRY
   {
     int src_rc = build_si_index(b, errbuf, errbuf_len);
@@ -3081,15 +3102,15 @@ int vmgpu_rollup_exec(const vmgpu_plan* plan, uint64_t handle,
                              : (size_t)b.n_series * n_grid;
   size_t count_elems = grouped ? out_elems : 0;
   if (b.out_elems < out_elems) {
-    (void)hipFree(b.d_out);
+    (void)vm_dev_free(b.d_out);
     b.d_out = nullptr;
-    HIP_TRY(hipMalloc(&b.d_out, out_elems * sizeof(double)), "alloc out");
+    HIP_TRY(vm_dev_malloc(&b.d_out, out_elems * sizeof(double)), "alloc out");
     b.out_elems = out_elems;
   }
   if (count_elems && b.count_elems < count_elems) {
-    (void)hipFree(b.d_counts);
+    (void)vm_dev_free(b.d_counts);
     b.d_counts = nullptr;
-    HIP_TRY(hipMalloc(&b.d_counts, count_elems * sizeof(double)), "alloc counts");
+    HIP_TRY(vm_dev_malloc(&b.d_counts, count_elems * sizeof(double)), "alloc counts");
     b.count_elems = count_elems;
   }
   HIP_TRY(hipMemsetAsync(b.d_scanned, 0, 8, g_ctx.stream), "zero scanned");
@@ -3359,10 +3380,10 @@ int vmgpu_topk_range(uint64_t handle, double k, int32_t summary_op,
   uint32_t* d_hist = nullptr;
   uint32_t* d_misc = nullptr; /* counters */
   uint32_t* d_sel = nullptr;
-  HIP_TRY(hipMalloc(&d_keys, (size_t)n * 8), "alloc keys");
-  HIP_TRY(hipMalloc(&d_hist, 65536 * 4), "alloc hist");
-  HIP_TRY(hipMalloc(&d_misc, 8), "alloc counters");
-  HIP_TRY(hipMalloc(&d_sel, (size_t)kk * 4), "alloc sel");
+  HIP_TRY(vm_dev_malloc(&d_keys, (size_t)n * 8), "alloc keys");
+  HIP_TRY(vm_dev_malloc(&d_hist, 65536 * 4), "alloc hist");
+  HIP_TRY(vm_dev_malloc(&d_misc, 8), "alloc counters");
+  HIP_TRY(vm_dev_malloc(&d_sel, (size_t)kk * 4), "alloc sel");
   uint32_t blocks = std::min<uint32_t>((n + BLOCK_THREADS - 1) / BLOCK_THREADS * WAVES_PER_BLOCK, 2048);
   hipLaunchKernelGGL(topk_summary_kernel, dim3(blocks), dim3(BLOCK_THREADS), 0, st,
                      b.d_out, n, n_grid, summary_op, reverse, d_keys);
@@ -3421,8 +3442,8 @@ int vmgpu_topk_range(uint64_t handle, double k, int32_t summary_op,
   if (out_remaining) {
     uint8_t* d_mask = nullptr;
     double* d_rem = nullptr;
-    HIP_TRY(hipMalloc(&d_mask, n), "alloc mask");
-    HIP_TRY(hipMalloc(&d_rem, (size_t)n_grid * 16), "alloc rem");
+    HIP_TRY(vm_dev_malloc(&d_mask, n), "alloc mask");
+    HIP_TRY(vm_dev_malloc(&d_rem, (size_t)n_grid * 16), "alloc rem");
     HIP_TRY(hipMemsetAsync(d_mask, 0, n, st), "zero mask");
     HIP_TRY(hipMemsetAsync(d_rem, 0, (size_t)n_grid * 16, st), "zero rem");
     std::vector<uint8_t> mask(n, 0);
@@ -3437,13 +3458,13 @@ int vmgpu_topk_range(uint64_t handle, double k, int32_t summary_op,
     HIP_TRY(hipStreamSynchronize(st), "sync rem");
     for (int32_t g = 0; g < n_grid; g++)
       out_remaining[g] = (rem[n_grid + g] == 0) ? __builtin_nan("") : rem[g];
-    (void)hipFree(d_mask);
-    (void)hipFree(d_rem);
+    (void)vm_dev_free(d_mask);
+    (void)vm_dev_free(d_rem);
   }
-  (void)hipFree(d_keys);
-  (void)hipFree(d_hist);
-  (void)hipFree(d_misc);
-  (void)hipFree(d_sel);
+  (void)vm_dev_free(d_keys);
+  (void)vm_dev_free(d_hist);
+  (void)vm_dev_free(d_misc);
+  (void)vm_dev_free(d_sel);
   return 0;
 }
 
@@ -3475,11 +3496,11 @@ int vmgpu_topk_pointwise(uint64_t handle, double k, int32_t reverse,
   unsigned long long* d_cand = nullptr;
   unsigned long long* d_kstar = nullptr;
   uint32_t* d_small = nullptr; /* cand_n[n_grid], ties[n_grid], taken[n_grid], overflow */
-  HIP_TRY(hipMalloc(&d_hists, (size_t)n_grid * 65536 * 4), "alloc col hists");
-  HIP_TRY(hipMalloc(&d_bins, (size_t)n_grid * 8), "alloc bins");
-  HIP_TRY(hipMalloc(&d_cand, (size_t)n_grid * CAND_CAP * 8), "alloc cands");
-  HIP_TRY(hipMalloc(&d_kstar, (size_t)n_grid * 8), "alloc kstar");
-  HIP_TRY(hipMalloc(&d_small, (size_t)n_grid * 12 + 4), "alloc small");
+  HIP_TRY(vm_dev_malloc(&d_hists, (size_t)n_grid * 65536 * 4), "alloc col hists");
+  HIP_TRY(vm_dev_malloc(&d_bins, (size_t)n_grid * 8), "alloc bins");
+  HIP_TRY(vm_dev_malloc(&d_cand, (size_t)n_grid * CAND_CAP * 8), "alloc cands");
+  HIP_TRY(vm_dev_malloc(&d_kstar, (size_t)n_grid * 8), "alloc kstar");
+  HIP_TRY(vm_dev_malloc(&d_small, (size_t)n_grid * 12 + 4), "alloc small");
   uint32_t* d_cand_n = d_small;
   uint32_t* d_ties = d_small + n_grid;
   uint32_t* d_taken = d_small + 2 * (size_t)n_grid;
@@ -3500,8 +3521,8 @@ int vmgpu_topk_pointwise(uint64_t handle, double k, int32_t reverse,
   HIP_TRY(hipMemcpyAsync(&overflow_h, d_overflow, 4, hipMemcpyDeviceToHost, st), "dl ovf");
   HIP_TRY(hipStreamSynchronize(st), "sync ovf");
   if (overflow_h) {
-    (void)hipFree(d_hists); (void)hipFree(d_bins); (void)hipFree(d_cand);
-    (void)hipFree(d_kstar); (void)hipFree(d_small);
+    (void)vm_dev_free(d_hists); (void)vm_dev_free(d_bins); (void)vm_dev_free(d_cand);
+    (void)vm_dev_free(d_kstar); (void)vm_dev_free(d_small);
     return set_err(errbuf, errbuf_len,
                    "vmgpu: topk candidate overflow (massive ties); unsupported");
   }
@@ -3516,8 +3537,8 @@ int vmgpu_topk_pointwise(uint64_t handle, double k, int32_t reverse,
   }
   HIP_TRY(hipStreamSynchronize(st), "sync fill");
   hipError_t kerr = hipGetLastError();
-  (void)hipFree(d_hists); (void)hipFree(d_bins); (void)hipFree(d_cand);
-  (void)hipFree(d_kstar); (void)hipFree(d_small);
+  (void)vm_dev_free(d_hists); (void)vm_dev_free(d_bins); (void)vm_dev_free(d_cand);
+  (void)vm_dev_free(d_kstar); (void)vm_dev_free(d_small);
   if (kerr != hipSuccess) return hip_err(errbuf, errbuf_len, "topk kernel", kerr);
   return 0;
 }
@@ -3537,10 +3558,10 @@ int vmgpu_histogram_stat(int32_t mode, const double* bucket_values,
   uint64_t* d_off = nullptr;
   double* d_out = nullptr;
   size_t out_elems = (size_t)n_groups * n_grid;
-  HIP_TRY(hipMalloc(&d_bv, (size_t)n_rows * n_grid * 8), "alloc bv");
-  HIP_TRY(hipMalloc(&d_les, (size_t)n_rows * 8), "alloc les");
-  HIP_TRY(hipMalloc(&d_off, (size_t)(n_groups + 1) * 8), "alloc off");
-  HIP_TRY(hipMalloc(&d_out, out_elems * 8), "alloc out");
+  HIP_TRY(vm_dev_malloc(&d_bv, (size_t)n_rows * n_grid * 8), "alloc bv");
+  HIP_TRY(vm_dev_malloc(&d_les, (size_t)n_rows * 8), "alloc les");
+  HIP_TRY(vm_dev_malloc(&d_off, (size_t)(n_groups + 1) * 8), "alloc off");
+  HIP_TRY(vm_dev_malloc(&d_out, out_elems * 8), "alloc out");
   HIP_TRY(hipMemcpyAsync(d_bv, bucket_values, (size_t)n_rows * n_grid * 8,
                          hipMemcpyHostToDevice, st), "ul bv");
   HIP_TRY(hipMemcpyAsync(d_les, les, (size_t)n_rows * 8, hipMemcpyHostToDevice, st), "ul les");
@@ -3552,8 +3573,8 @@ int vmgpu_histogram_stat(int32_t mode, const double* bucket_values,
   HIP_TRY(hipMemcpyAsync(out, d_out, out_elems * 8, hipMemcpyDeviceToHost, st), "dl out");
   HIP_TRY(hipStreamSynchronize(st), "sync hstat");
   hipError_t kerr = hipGetLastError();
-  (void)hipFree(d_bv); (void)hipFree(d_les); (void)hipFree(d_off);
-  (void)hipFree(d_out);
+  (void)vm_dev_free(d_bv); (void)vm_dev_free(d_les); (void)vm_dev_free(d_off);
+  (void)vm_dev_free(d_out);
   if (kerr != hipSuccess) return hip_err(errbuf, errbuf_len, "hstat kernel", kerr);
   return 0;
 }
@@ -3578,13 +3599,13 @@ int vmgpu_histogram_share(const double* le_req, const double* bucket_values,
   double* d_lo = nullptr;
   double* d_hi = nullptr;
   size_t out_elems = (size_t)n_groups * n_grid;
-  HIP_TRY(hipMalloc(&d_req, (size_t)n_grid * 8), "alloc req");
-  HIP_TRY(hipMalloc(&d_bv, (size_t)n_rows * n_grid * 8), "alloc bv");
-  HIP_TRY(hipMalloc(&d_les, (size_t)n_rows * 8), "alloc les");
-  HIP_TRY(hipMalloc(&d_off, (size_t)(n_groups + 1) * 8), "alloc off");
-  HIP_TRY(hipMalloc(&d_out, out_elems * 8), "alloc out");
-  if (out_lower) HIP_TRY(hipMalloc(&d_lo, out_elems * 8), "alloc lo");
-  if (out_upper) HIP_TRY(hipMalloc(&d_hi, out_elems * 8), "alloc hi");
+  HIP_TRY(vm_dev_malloc(&d_req, (size_t)n_grid * 8), "alloc req");
+  HIP_TRY(vm_dev_malloc(&d_bv, (size_t)n_rows * n_grid * 8), "alloc bv");
+  HIP_TRY(vm_dev_malloc(&d_les, (size_t)n_rows * 8), "alloc les");
+  HIP_TRY(vm_dev_malloc(&d_off, (size_t)(n_groups + 1) * 8), "alloc off");
+  HIP_TRY(vm_dev_malloc(&d_out, out_elems * 8), "alloc out");
+  if (out_lower) HIP_TRY(vm_dev_malloc(&d_lo, out_elems * 8), "alloc lo");
+  if (out_upper) HIP_TRY(vm_dev_malloc(&d_hi, out_elems * 8), "alloc hi");
   HIP_TRY(hipMemcpyAsync(d_req, le_req, (size_t)n_grid * 8,
                          hipMemcpyHostToDevice, st), "ul req");
   HIP_TRY(hipMemcpyAsync(d_bv, bucket_values, (size_t)n_rows * n_grid * 8,
@@ -3601,9 +3622,9 @@ int vmgpu_histogram_share(const double* le_req, const double* bucket_values,
   if (out_upper) HIP_TRY(hipMemcpyAsync(out_upper, d_hi, out_elems * 8, hipMemcpyDeviceToHost, st), "dl hi");
   HIP_TRY(hipStreamSynchronize(st), "sync hshare");
   hipError_t kerr = hipGetLastError();
-  (void)hipFree(d_req); (void)hipFree(d_bv); (void)hipFree(d_les);
-  (void)hipFree(d_off); (void)hipFree(d_out); (void)hipFree(d_lo);
-  (void)hipFree(d_hi);
+  (void)vm_dev_free(d_req); (void)vm_dev_free(d_bv); (void)vm_dev_free(d_les);
+  (void)vm_dev_free(d_off); (void)vm_dev_free(d_out); (void)vm_dev_free(d_lo);
+  (void)vm_dev_free(d_hi);
   if (kerr != hipSuccess) return hip_err(errbuf, errbuf_len, "hshare kernel", kerr);
   return 0;
 }
@@ -3626,12 +3647,12 @@ int vmgpu_histogram_quantile(double phi, const double* bucket_values,
   double* d_lo = nullptr;
   double* d_hi = nullptr;
   size_t out_elems = (size_t)n_groups * n_grid;
-  HIP_TRY(hipMalloc(&d_bv, (size_t)n_rows * n_grid * 8), "alloc bv");
-  HIP_TRY(hipMalloc(&d_les, (size_t)n_rows * 8), "alloc les");
-  HIP_TRY(hipMalloc(&d_off, (size_t)(n_groups + 1) * 8), "alloc off");
-  HIP_TRY(hipMalloc(&d_out, out_elems * 8), "alloc out");
-  if (out_lower) HIP_TRY(hipMalloc(&d_lo, out_elems * 8), "alloc lo");
-  if (out_upper) HIP_TRY(hipMalloc(&d_hi, out_elems * 8), "alloc hi");
+  HIP_TRY(vm_dev_malloc(&d_bv, (size_t)n_rows * n_grid * 8), "alloc bv");
+  HIP_TRY(vm_dev_malloc(&d_les, (size_t)n_rows * 8), "alloc les");
+  HIP_TRY(vm_dev_malloc(&d_off, (size_t)(n_groups + 1) * 8), "alloc off");
+  HIP_TRY(vm_dev_malloc(&d_out, out_elems * 8), "alloc out");
+  if (out_lower) HIP_TRY(vm_dev_malloc(&d_lo, out_elems * 8), "alloc lo");
+  if (out_upper) HIP_TRY(vm_dev_malloc(&d_hi, out_elems * 8), "alloc hi");
   HIP_TRY(hipMemcpyAsync(d_bv, bucket_values, (size_t)n_rows * n_grid * 8,
                          hipMemcpyHostToDevice, st), "ul bv");
   HIP_TRY(hipMemcpyAsync(d_les, les, (size_t)n_rows * 8, hipMemcpyHostToDevice, st), "ul les");
@@ -3645,8 +3666,8 @@ int vmgpu_histogram_quantile(double phi, const double* bucket_values,
   if (out_upper) HIP_TRY(hipMemcpyAsync(out_upper, d_hi, out_elems * 8, hipMemcpyDeviceToHost, st), "dl hi");
   HIP_TRY(hipStreamSynchronize(st), "sync hq");
   hipError_t kerr = hipGetLastError();
-  (void)hipFree(d_bv); (void)hipFree(d_les); (void)hipFree(d_off);
-  (void)hipFree(d_out); (void)hipFree(d_lo); (void)hipFree(d_hi);
+  (void)vm_dev_free(d_bv); (void)vm_dev_free(d_les); (void)vm_dev_free(d_off);
+  (void)vm_dev_free(d_out); (void)vm_dev_free(d_lo); (void)vm_dev_free(d_hi);
   if (kerr != hipSuccess) return hip_err(errbuf, errbuf_len, "hq kernel", kerr);
   return 0;
 }
