@@ -52,6 +52,9 @@ def parse_args():
     ap.add_argument("--skip-cold", action="store_true",
                     help="skip the end-to-end cold-query measurement")
     ap.add_argument("--cold-queries", type=int, default=3)
+    ap.add_argument("--skip-grouped", action="store_true",
+                    help="skip the secondary grouped (config 3) measurement")
+    ap.add_argument("--grouped-steps", type=int, default=4)
     return ap.parse_args()
 
 
@@ -140,12 +143,51 @@ def cold_query(args, engine, plan, ts, vals, offsets):
         "p50_ms": p50 * 1e3,
         "p50_create_ms": float(np.median(create_ms)),
         "p50_exec_download_ms": float(np.median(exec_ms)),
+        "walls_ms": [round(w * 1e3, 1) for w in walls],
+        "create_walls_ms": [round(w, 1) for w in create_ms],
         "samples_per_s": samples / p50,
         "queries": args.cold_queries,
         "payload_bytes": len(packed),
         "pipeline": "packed compressed blocks (pinned) -> native C "
                     "descriptor parse -> PCIe -> device decode+merge -> "
                     "rollup -> result matrix in pinned host memory",
+    }
+
+
+def grouped_probe(args, engine, ts, vals, offsets, end):
+    """Secondary measurement in the same driver-run line: BASELINE
+    configs[2] (sum by(pod)(rate(m[5m])), 10k label groups) on its own
+    group-relayouted batch; kernel-resident like the headline."""
+    gids = (np.arange(args.series) % args.groups).astype(np.int32)
+    plan = engine.RollupPlan(args.func, START_TS, end, STEP_MS,
+                             window=WINDOW_MS, aggr="sum")
+    b = engine.SeriesBatch(ts, vals, offsets, group_ids=gids,
+                           n_groups=args.groups)
+    try:
+        for _ in range(2):
+            b.exec(plan, download=False)
+        t0 = time.perf_counter()
+        kms = 0.0
+        for _ in range(args.grouped_steps):
+            b.exec(plan, download=False)
+            kms += engine.last_kernel_ms()
+        wall = time.perf_counter() - t0
+    finally:
+        b.close()
+    kernel_s = kms / args.grouped_steps / 1e3
+    algo = args.series * args.samples * 16 + args.groups * args.samples * 16
+    achieved = algo / kernel_s / 1e9 if kernel_s > 0 else 0.0
+    return {
+        "workload": "sum by(pod)(rate(metric[5m])) over 1M series / "
+                    f"{args.groups} groups",
+        "steps": args.grouped_steps,
+        "kernel_ms": kernel_s * 1e3,
+        "ms_per_step": wall / args.grouped_steps * 1e3,
+        "samples_per_s": args.series * args.samples / (wall / args.grouped_steps),
+        "roofline": {"bound": "hbm", "achieved": achieved,
+                     "peak": HBM_PEAK_GBS, "unit": "GB/s",
+                     "frac": achieved / HBM_PEAK_GBS,
+                     "algorithmic_bytes": algo},
     }
 
 
@@ -250,6 +292,9 @@ def main():
         cold = None
         if not args.skip_cold and world == 1 and not grouped:
             cold = cold_query(args, engine, plan, ts, vals, offsets)
+        grouped_info = None
+        if not args.skip_grouped and world == 1 and not grouped:
+            grouped_info = grouped_probe(args, engine, ts, vals, offsets, end)
         workload = (f"{args.func}(metric[5m])[1h:15s] over 1M series x 240 samples"
                     if not grouped else
                     "sum by(pod)(rate(metric[5m])) over 1M series / 10k groups")
@@ -291,6 +336,7 @@ def main():
             },
             "cpu_baseline": cb,
             "cold_query": cold,
+            "grouped": grouped_info,
         }
         print(json.dumps(result), flush=True)
 

@@ -3026,7 +3026,9 @@ int vmgpu_batch_create_packed(
     d.max_timestamp = h.max_timestamp;
     d.first_value = h.first_value;
     d.scale = h.scale;
-    d.e10 = pow(10.0, (double)(h.scale < 0 ? -h.scale : h.scale));
+    d.e10 = (h.scale == 0)
+                ? 1.0
+                : pow(10.0, (double)(h.scale < 0 ? -h.scale : h.scale));
     d.rows = h.rows;
     d.ts_mt = h.ts_mt;
     d.val_mt = h.val_mt;
