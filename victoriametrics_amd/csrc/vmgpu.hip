@@ -72,13 +72,16 @@ static VM_DEV void wave_ds_sync() {
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 }
 
+/* STR = element stride in int64 units: 1 for column layout, 2 for the pipe
+ * kernel's (t, v) pair-interleaved LDS (one ds_read2/b128 per sample). */
+template <int STR = 1>
 static VM_DEV int vm_upper_bound(const int64_t* ts, int n, int64_t seek) {
   /* first index with ts[idx] > seek (seekFirstTimestampIdxAfter semantics,
    * rollup.go:825-855 — the reference's ±2 hint only narrows the range). */
   int i = 0, j = n;
   while (i < j) {
     int h = (i + j) >> 1;
-    if (ts[h] <= seek) i = h + 1;
+    if (ts[h * STR] <= seek) i = h + 1;
     else j = h;
   }
   return i;
@@ -87,22 +90,24 @@ static VM_DEV int vm_upper_bound(const int64_t* ts, int n, int64_t seek) {
 /* Per-lane upper bound from a density-interpolated guess: a bounded walk
  * resolves near-uniform sampling in 1-2 probes; anything irregular falls
  * back to a plain binary search.  Result identical to vm_upper_bound. */
+template <int STR = 1>
 static VM_DEV int vm_ub_hint(const int64_t* ts, int n, int64_t seek, int g) {
   if (g < 0) g = 0;
   if (g > n) g = n;
   for (int steps = 0; steps < 4; steps++) {
-    bool gt_here = (g >= n) || (ts[g] > seek);
+    bool gt_here = (g >= n) || (ts[g * STR] > seek);
     if (!gt_here) { g++; continue; }       /* result is above g */
-    if (g == 0 || ts[g - 1] <= seek) return g;
+    if (g == 0 || ts[(g - 1) * STR] <= seek) return g;
     g--;                                   /* result is below g */
   }
-  return vm_upper_bound(ts, n, seek);
+  return vm_upper_bound<STR>(ts, n, seek);
 }
 
 /* Branch-free variant: four independent probes decide among {g-1, g, g+1}
  * with pure selects; lanes outside the +-1 window (rare on real scrape
  * cadences) take a wave-coordinated binary-search fallback.  Result
  * identical to vm_upper_bound. */
+template <int STR = 1>
 static VM_DEV int vm_ub_hint_fast(const int64_t* ts, int n, int64_t seek, int g) {
   if (g < 0) g = 0;
   if (g > n) g = n;
@@ -110,10 +115,10 @@ static VM_DEV int vm_ub_hint_fast(const int64_t* ts, int n, int64_t seek, int g)
   int i_m1 = g - 1 < 0 ? 0 : g - 1;
   int i_0 = g < n - 1 ? g : (n - 1 < 0 ? 0 : n - 1);
   int i_p1 = g + 1 < n - 1 ? g + 1 : (n - 1 < 0 ? 0 : n - 1);
-  int64_t t_m2 = ts[i_m2];
-  int64_t t_m1 = ts[i_m1];
-  int64_t t_0 = ts[i_0];
-  int64_t t_p1 = ts[i_p1];
+  int64_t t_m2 = ts[i_m2 * STR];
+  int64_t t_m1 = ts[i_m1 * STR];
+  int64_t t_0 = ts[i_0 * STR];
+  int64_t t_p1 = ts[i_p1 * STR];
   bool le_m2 = (g - 2 < 0) || (t_m2 <= seek);  /* ts[g-2] <= seek (vacuous at edge) */
   bool le_m1 = (g - 1 < 0) || (g - 1 >= n) || (t_m1 <= seek);
   bool gt_m1 = (g - 1 >= 0) && (g - 1 < n) && (t_m1 > seek);
@@ -126,7 +131,7 @@ static VM_DEV int vm_ub_hint_fast(const int64_t* ts, int n, int64_t seek, int g)
   bool ok_m1 = (g - 1 >= 0) && le_m2 && gt_m1;
   int r = ok_g ? g : (ok_p1 ? g + 1 : (ok_m1 ? g - 1 : -1));
   if (__any(r < 0)) {
-    if (r < 0) r = vm_upper_bound(ts, n, seek);
+    if (r < 0) r = vm_upper_bound<STR>(ts, n, seek);
   }
   return r;
 }
@@ -238,6 +243,7 @@ struct KIO {
  * Returns new count. dst may be LDS or global. */
 struct __align__(16) vm_i64x2 { int64_t x, y; };
 
+template <int STR = 1>
 static __device__ int load_compact_wave(const int64_t* g_ts, const double* g_vals,
                                         int64_t n, int64_t* d_ts, double* d_vals,
                                         bool drop_stale, int lane) {
@@ -246,8 +252,9 @@ static __device__ int load_compact_wave(const int64_t* g_ts, const double* g_val
   /* vector fast path: 128-element tiles via 16-B loads/stores (the staging
    * loop is instruction-issue-bound, not bandwidth-bound — see profiles/).
    * Falls to the scalar tile path when stale NaNs must be compacted out or
-   * the running count is odd (LDS b128 stores need 16-B alignment). */
-  if ((((uintptr_t)g_ts | (uintptr_t)g_vals) & 15) == 0) {
+   * the running count is odd (LDS b128 stores need 16-B alignment);
+   * column layout only. */
+  if (STR == 1 && (((uintptr_t)g_ts | (uintptr_t)g_vals) & 15) == 0) {
     const uint64_t lt_mask = (lane == 63) ? 0x7fffffffffffffffULL
                                           : ((1ULL << lane) - 1);
     while (src + 2 <= n) {
@@ -294,8 +301,8 @@ static __device__ int load_compact_wave(const int64_t* g_ts, const double* g_val
     if (keep) {
       int dst = count + __popcll(m & ((lane == 63) ? 0x7fffffffffffffffULL
                                                    : ((1ULL << lane) - 1)));
-      d_ts[dst] = t;
-      d_vals[dst] = v;
+      d_ts[dst * STR] = t;
+      d_vals[dst * STR] = v;
     }
     count += __popcll(m);
   }
@@ -382,6 +389,7 @@ static __device__ int load_rcr_fused_wave(const int64_t* g_ts, const double* g_v
 
 /* --- phase B: removeCounterResets over the dense column -------------- */
 /* EXACT restatement of rollup.go:921-958 as a wave scan; see file header. */
+template <int STR = 1>
 static __device__ void rcr_scan_wave(int64_t* d_ts, double* d_vals, int count,
                                      int64_t msi, int lane) {
   double corr = 0.0;       /* running correction */
@@ -391,8 +399,8 @@ static __device__ void rcr_scan_wave(int64_t* d_ts, double* d_vals, int count,
   for (int base = 0; base < count; base += WAVE) {
     int k = base + lane;
     bool active = k < count;
-    double v = active ? d_vals[k] : 0.0;
-    int64_t t = active ? d_ts[k] : 0;
+    double v = active ? d_vals[k * STR] : 0.0;
+    int64_t t = active ? d_ts[k * STR] : 0;
     double pv = __shfl_up(v, 1);
     int64_t pt = __shfl_up(t, 1);
     if (lane == 0) { pv = prev_raw; pt = prev_ts; }
@@ -413,7 +421,7 @@ static __device__ void rcr_scan_wave(int64_t* d_ts, double* d_vals, int count,
     if (last_fast > 63) last_fast = 63;
     if (em == 0 && dm == 0 &&
         (base == 0 || __shfl(v, 0) + corr >= prev_fin)) {
-      if (corr != 0.0 && active) d_vals[k] = v + corr;
+      if (corr != 0.0 && active) d_vals[k * STR] = v + corr;
       prev_raw = __shfl(v, last_fast);
       prev_ts = __shfl(t, last_fast);
       prev_fin = prev_raw + corr;
@@ -445,7 +453,7 @@ static __device__ void rcr_scan_wave(int64_t* d_ts, double* d_vals, int count,
         f = f | fo;
       }
     }
-    if (active) d_vals[k] = x;
+    if (active) d_vals[k * STR] = x;
     int last = count - base - 1;
     if (last > 63) last = 63;
     corr = c;
@@ -557,7 +565,7 @@ static VM_DEV void pre_func_wave(const int64_t* d_ts, double* d_vals,
 /* getScrapeInterval (rollup.go:871-897): 0.6 quantile of the last <=20
  * sample gaps, computed wave-cooperatively with a rank-based selection
  * (identical result to sort + quantileSorted). */
-template <bool LDS_ONLY>
+template <bool LDS_ONLY, int STR = 1>
 static __device__ int64_t scrape_interval_wave_t(const int64_t* d_ts, int count,
                                                  int64_t default_interval, int lane,
                                                  double* scratch) {
@@ -566,7 +574,8 @@ static __device__ int64_t scrape_interval_wave_t(const int64_t* d_ts, int count,
   if (cnt > 20) cnt = 20;
   bool active = lane < cnt;
   double gap = 0.0;
-  if (active) gap = (double)(d_ts[count - 1 - lane] - d_ts[count - 2 - lane]);
+  if (active) gap = (double)(d_ts[(count - 1 - lane) * STR] -
+                             d_ts[(count - 2 - lane) * STR]);
   if (active) scratch[lane] = gap;
   if (LDS_ONLY) wave_ds_sync(); else wave_lds_sync();
   /* rank of this lane's gap among the cnt gaps (ties broken by lane);
@@ -697,18 +706,43 @@ static VM_DEV void vm_emit_value(const KPlan& p, const KIO& io, uint32_t s,
 /* Fused branch-free rate/deriv_fast evaluator (rollupDerivFast,
  * rollup.go:1954-1989 + the rfa construction of doInternal:779-810 reduced
  * to the fields rate reads): six independent LDS loads + selects. */
+#ifndef VMGPU_PIPE_NO_TV
+#define VMGPU_PIPE_NO_TV 0
+#endif
+
+/* 16-byte (t, v) record of the pair-interleaved LDS slab */
+struct __align__(16) vm_tv {
+  int64_t t;
+  double v;
+};
+
+template <int STR = 1>
 static VM_DEV double eval_rate_fused(const KPlan& p, const SeriesWindow& sw,
                                      const int64_t* ts, const double* vals,
                                      int count, int i, int j, int64_t t_start) {
   int im1 = i - 1 < 0 ? 0 : i - 1;
   int ii = i < count - 1 ? i : (count - 1 < 0 ? 0 : count - 1);
   int jm1 = j - 1 < 0 ? 0 : j - 1;
-  double v_prevc = vals[im1];
-  int64_t t_prevc = ts[im1];
-  double v_first = vals[ii];
-  int64_t t_first = ts[ii];
-  double v_end = vals[jm1];
-  int64_t t_end_s = ts[jm1];
+  double v_prevc, v_first, v_end;
+  int64_t t_prevc, t_first, t_end_s;
+  if constexpr (STR == 2 && !VMGPU_PIPE_NO_TV) {
+    /* one b128 LDS read per boundary role instead of two scattered b64s */
+    const vm_tv* tv = (const vm_tv*)ts;
+    vm_tv a = tv[im1], b = tv[ii], c = tv[jm1];
+    t_prevc = a.t;
+    v_prevc = a.v;
+    t_first = b.t;
+    v_first = b.v;
+    t_end_s = c.t;
+    v_end = c.v;
+  } else {
+    v_prevc = vals[im1 * STR];
+    t_prevc = ts[im1 * STR];
+    v_first = vals[ii * STR];
+    t_first = ts[ii * STR];
+    v_end = vals[jm1 * STR];
+    t_end_s = ts[jm1 * STR];
+  }
   int n = j - i;
   bool has_prev = (i < count) && (i > 0) &&
                   (t_prevc > t_start - sw.max_prev_interval);
@@ -1135,6 +1169,7 @@ static VM_DEV int vm_scatter_boundary(int64_t t, int64_t gstart, int64_t gstep,
   return g;
 }
 
+template <int STR = 1>
 static VM_DEV int rcr_scan_regs(const int64_t* rt, const double* rv,
                                 int64_t n, int64_t* d_ts, double* d_vals,
                                 bool drop_stale, int64_t msi, int lane,
@@ -1154,7 +1189,7 @@ static VM_DEV int rcr_scan_regs(const int64_t* rt, const double* rv,
     double v = active ? rv[c] : 0.0;
     int64_t t = active ? rt[c] : 0;
     if (drop_stale && __ballot(active && vm_is_stale_nan(v)) != 0) return -1;
-    if (active) d_ts[k] = t;
+    if (active) d_ts[k * STR] = t;
     if (jb) {
       /* fused J scatter: sample k-1 is the window-end boundary for grid
        * points in [b(t_{k-1}), b(t_k)) — write the upper-bound count k
@@ -1190,7 +1225,7 @@ static VM_DEV int rcr_scan_regs(const int64_t* rt, const double* rv,
     if (last > 63) last = 63;
     if (em == 0 && dm == 0 &&
         (base == 0 || __shfl(v, 0) + corr >= prev_fin)) {
-      if (active) d_vals[k] = v + corr;
+      if (active) d_vals[k * STR] = v + corr;
       prev_raw = __shfl(v, last);
       prev_ts = __shfl(t, last);
       prev_fin = prev_raw + corr;
@@ -1220,7 +1255,7 @@ static VM_DEV int rcr_scan_regs(const int64_t* rt, const double* rv,
         f = f | fo;
       }
     }
-    if (active) d_vals[k] = x;
+    if (active) d_vals[k * STR] = x;
     corr = cc;
     prev_raw = __shfl(v, last);
     prev_ts = __shfl(t, last);
@@ -1279,9 +1314,12 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
   const int lane = threadIdx.x % WAVE;
   const size_t jbuf_bytes = vm_jbuf_bytes(p.jbuf_mode, p.jbuf_elems);
   const size_t wave_bytes = (size_t)p.chunk_wave * 16 + 256 + jbuf_bytes;
+  /* PAIR-interleaved series slab: sample k = 16-byte {t, v} at offset 16k
+   * (stride template parameter STR=2 on every accessor) — the eval
+   * gathers fetch one b128 LDS read per boundary role instead of two
+   * scattered b64 reads */
   int64_t* lts = (int64_t*)(smem + (size_t)wave_in_block * wave_bytes);
-  double* lvs = (double*)(smem + (size_t)wave_in_block * wave_bytes +
-                          (size_t)p.chunk_wave * 8);
+  double* lvs = (double*)(smem + (size_t)wave_in_block * wave_bytes + 8);
   double* lscratch = (double*)(smem + (size_t)wave_in_block * wave_bytes +
                                (size_t)p.chunk_wave * 16);
   uint16_t* jbuf = (uint16_t*)(smem + (size_t)wave_in_block * wave_bytes +
@@ -1370,10 +1408,10 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
 #else
     if (p.rcr) {
 #endif
-      count = rcr_scan_regs(rt, rv, n, lts, lvs, p.drop_stale != 0,
-                            p.max_staleness, lane,
-                            scat_dg > 0 ? jbuf : nullptr, scat_dg,
-                            p.start, p.step, inv_gstep, p.n_grid);
+      count = rcr_scan_regs<2>(rt, rv, n, lts, lvs, p.drop_stale != 0,
+                               p.max_staleness, lane,
+                               scat_dg > 0 ? jbuf : nullptr, scat_dg,
+                               p.start, p.step, inv_gstep, p.n_grid);
     } else {
       bool stale = false;
       if (p.drop_stale) {
@@ -1387,7 +1425,7 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
 #pragma unroll
         for (int c = 0; c < PIPE_CHUNKS; c++) {
           int64_t k = (int64_t)c * WAVE + lane;
-          if (k < n) { lts[k] = rt[c]; lvs[k] = rv[c]; }
+          if (k < n) { lts[k * 2] = rt[c]; lvs[k * 2] = rv[c]; }
         }
         if (scat_dg > 0)
           scatter_j_regs(rt, n, lane, jbuf, scat_dg, p.start, p.step,
@@ -1411,10 +1449,10 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
     }
     if (count < 0) {
       /* stale NaNs must be compacted out (rare): redo from global */
-      count = load_compact_wave(io.ts + lo, io.vals + lo, n, lts, lvs,
-                                p.drop_stale != 0, lane);
+      count = load_compact_wave<2>(io.ts + lo, io.vals + lo, n, lts, lvs,
+                                   p.drop_stale != 0, lane);
       wave_ds_sync();
-      if (p.rcr) rcr_scan_wave(lts, lvs, count, p.max_staleness, lane);
+      if (p.rcr) rcr_scan_wave<2>(lts, lvs, count, p.max_staleness, lane);
       if (scat_dg > 0) {
         /* the register scatter saw pre-compaction indices: rebuild J by
          * direct binary search over the compacted column */
@@ -1422,7 +1460,7 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
         const int ext = p.n_grid + scat_dg;
         for (int e = lane; e < ext; e += WAVE) {
           int64_t t_end = p.start + (int64_t)(e - scat_dg) * p.step;
-          jbuf[e] = (uint16_t)vm_upper_bound(lts, count, t_end);
+          jbuf[e] = (uint16_t)vm_upper_bound<2>(lts, count, t_end);
         }
       }
     }
@@ -1432,7 +1470,8 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
     /* stage-only ablation: keep the output write traffic, skip the rest */
     for (int g = lane; g < p.n_grid; g += WAVE) {
       int gc = g < count ? g : (count > 0 ? count - 1 : 0);
-      io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] = count ? lvs[gc] : 0.0;
+      io.out[(size_t)s * (size_t)p.n_grid + (size_t)g] =
+          count ? lvs[gc * 2] : 0.0;
     }
     wave_ds_sync();
     ws = ws_n;
@@ -1449,7 +1488,7 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
       if (io.series_si && count == (int)n) {
         si = si_raw > 0 ? si_raw : p.step;
       } else {
-        si = scrape_interval_wave_t<true>(lts, count, p.step, lane, lscratch);
+        si = scrape_interval_wave_t<true, 2>(lts, count, p.step, lane, lscratch);
       }
     }
 #endif
@@ -1460,7 +1499,7 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
     int64_t ts0 = 0;
     if (count > 1) {
       ts0 = lts[0];
-      int64_t span_ms = lts[count - 1] - ts0;
+      int64_t span_ms = lts[(count - 1) * 2] - ts0;
       idx_per_ms = span_ms > 0 ? (double)(count - 1) / (double)span_ms : 0.0;
     }
     /* emit: GROUPED folds the aggregate switch in; the ungrouped build is
@@ -1481,11 +1520,11 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
           int j = jbuf[gs + scat_dg];
           int i = jbuf[gs];
 #ifdef VMGPU_PIPE_ABL_NO_EVAL
-          emit(gs, (j > 0 && j <= count) ? lvs[j - 1] : 0.0);
+          emit(gs, (j > 0 && j <= count) ? lvs[(j - 1) * 2] : 0.0);
           (void)i;
 #else
-          emit(gs, eval_rate_fused(p, sw, lts, lvs, count, i, j,
-                                   te_s - sw.window));
+          emit(gs, eval_rate_fused<2>(p, sw, lts, lvs, count, i, j,
+                                      te_s - sw.window));
 #endif
         }
         done = true;
@@ -1503,11 +1542,11 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
           const int fill_full = (p.n_grid / WAVE) * WAVE;
           for (; gf < fill_full; gf += WAVE, te += t_step_wave) {
             int gj = (int)((double)(te - ts0) * idx_per_ms) + 1;
-            jbuf[gf] = (uint16_t)vm_ub_hint_fast(lts, count, te, gj);
+            jbuf[gf] = (uint16_t)vm_ub_hint_fast<2>(lts, count, te, gj);
           }
           if (gf < p.n_grid) {
             int gj = (int)((double)(te - ts0) * idx_per_ms) + 1;
-            jbuf[gf] = (uint16_t)vm_ub_hint_fast(lts, count, te, gj);
+            jbuf[gf] = (uint16_t)vm_ub_hint_fast<2>(lts, count, te, gj);
           }
         }
         wave_ds_sync();
@@ -1520,12 +1559,12 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
           int64_t t_start = t_end - sw.window;
           int j = jbuf[g];
           int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
-          int i = vm_ub_hint_fast(lts, count, t_start, gi);
+          int i = vm_ub_hint_fast<2>(lts, count, t_start, gi);
 #ifdef VMGPU_PIPE_ABL_NO_EVAL
-          emit(g, (j > 0 && j <= count) ? lvs[j - 1] : 0.0);
+          emit(g, (j > 0 && j <= count) ? lvs[(j - 1) * 2] : 0.0);
           (void)i;
 #else
-          emit(g, eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
+          emit(g, eval_rate_fused<2>(p, sw, lts, lvs, count, i, j, t_start));
 #endif
         }
         /* main eval: full lane-blocks (no per-point exec masking), then one
@@ -1539,22 +1578,22 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
             int j = jbuf[g];
             int i = jbuf[g - dg64];
 #ifdef VMGPU_PIPE_ABL_NO_EVAL
-            emit(g, (j > 0 && j <= count) ? lvs[j - 1] : 0.0);
+            emit(g, (j > 0 && j <= count) ? lvs[(j - 1) * 2] : 0.0);
             (void)i;
 #else
-            emit(g, eval_rate_fused(p, sw, lts, lvs, count, i, j,
-                                    te - sw.window));
+            emit(g, eval_rate_fused<2>(p, sw, lts, lvs, count, i, j,
+                                       te - sw.window));
 #endif
           }
           if (g < p.n_grid) {
             int j = jbuf[g];
             int i = jbuf[g - dg64];
 #ifdef VMGPU_PIPE_ABL_NO_EVAL
-            emit(g, (j > 0 && j <= count) ? lvs[j - 1] : 0.0);
+            emit(g, (j > 0 && j <= count) ? lvs[(j - 1) * 2] : 0.0);
             (void)i;
 #else
-            emit(g, eval_rate_fused(p, sw, lts, lvs, count, i, j,
-                                    te - sw.window));
+            emit(g, eval_rate_fused<2>(p, sw, lts, lvs, count, i, j,
+                                       te - sw.window));
 #endif
           }
         }
@@ -1569,9 +1608,9 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
       for (int g = lane; g < p.n_grid; g += WAVE) {
         int64_t t_end = p.start + (int64_t)g * p.step;
         int64_t t_start = t_end - sw.window;
-        int i = vm_upper_bound(lts, count, t_start);
-        int j = vm_upper_bound(lts, count, t_end);
-        emit(g, eval_rate_fused(p, sw, lts, lvs, count, i, j, t_start));
+        int i = vm_upper_bound<2>(lts, count, t_start);
+        int j = vm_upper_bound<2>(lts, count, t_end);
+        emit(g, eval_rate_fused<2>(p, sw, lts, lvs, count, i, j, t_start));
       }
     }
     /* samplesScanned: every grid point of this series contributes exactly
