@@ -670,8 +670,9 @@ int vmgpu_merge_blocks(const int64_t* ts, const double* vals,
   DHIP_TRY(vm_dev_malloc(&d_boff, (size_t)(n_blocks + 1) * 8), "alloc boff");
   DHIP_TRY(vm_dev_malloc(&d_sbs, (size_t)(n_series + 1) * 4), "alloc sbs");
   DHIP_TRY(vm_dev_malloc(&d_ooff, (size_t)(n_series + 1) * 8), "alloc ooff");
-  DHIP_TRY(vm_dev_malloc(&d_ots, cap_off[n_series] * 8), "alloc out ts");
-  DHIP_TRY(vm_dev_malloc(&d_ovals, cap_off[n_series] * 8), "alloc out vals");
+  /* +16 B slack for the pipe kernel's clamped tail pair load */
+  DHIP_TRY(vm_dev_malloc(&d_ots, cap_off[n_series] * 8 + 16), "alloc out ts");
+  DHIP_TRY(vm_dev_malloc(&d_ovals, cap_off[n_series] * 8 + 16), "alloc out vals");
   DHIP_TRY(vm_dev_malloc(&d_ocnt, (size_t)n_series * 8), "alloc out counts");
   DHIP_TRY(vm_dev_malloc(&d_err, 4), "alloc err");
   DHIP_TRY(hipMemcpyAsync(d_ts, ts, total * 8, hipMemcpyHostToDevice, st), "ul ts");
@@ -851,8 +852,9 @@ extern "C" int vmdec_decode_merge_device(
   FDM_TRY(vm_dev_malloc(&d_finaloff, (size_t)(n_series + 1) * 8), "alloc finaloff");
   FDM_TRY(hipMemcpyAsync(d_finaloff, h_final_offsets, (size_t)(n_series + 1) * 8,
                          hipMemcpyHostToDevice, st), "ul finaloff");
-  FDM_TRY(vm_dev_malloc(&d_fts, (h_final_offsets[n_series] ? h_final_offsets[n_series] : 1) * 8), "alloc final ts");
-  FDM_TRY(vm_dev_malloc(&d_fvals, (h_final_offsets[n_series] ? h_final_offsets[n_series] : 1) * 8), "alloc final vals");
+  /* +16 B slack for the pipe kernel's clamped tail pair load */
+  FDM_TRY(vm_dev_malloc(&d_fts, (h_final_offsets[n_series] ? h_final_offsets[n_series] : 1) * 8 + 16), "alloc final ts");
+  FDM_TRY(vm_dev_malloc(&d_fvals, (h_final_offsets[n_series] ? h_final_offsets[n_series] : 1) * 8 + 16), "alloc final vals");
   hipLaunchKernelGGL(compact_series_kernel, dim3(mgrid), dim3(DBLOCK), 0, st,
                      d_mts, d_mvals, d_capoff, d_finaloff, n_series,
                      d_fts, d_fvals);

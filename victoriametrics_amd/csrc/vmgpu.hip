@@ -1295,6 +1295,135 @@ static VM_DEV void scatter_j_regs(const int64_t* rt, int64_t n, int lane,
   }
 }
 
+#define PIPE_PCHUNKS 2 /* pair chunks: 2 x 128 samples = 256 cap */
+
+/* Pair-register removeCounterResets: lane l holds samples (2l, 2l+1) of
+ * each 128-sample chunk (ONE b128 global load per column per chunk).
+ * Exactly the rcr_scan_wave semantics (rollup.go:921-958): the running
+ * correction visits the (rare) event ELEMENTS in index order bitwise, and
+ * the monotonic clamp is the same segmented prefix max, evaluated as
+ * in-lane pair combines + the cross-lane scan (fmax is exactly
+ * associative).  Writes the pair-interleaved LDS slab; returns n, or -1
+ * when a stale NaN must be compacted out (caller falls back to the global
+ * compact path). */
+static VM_DEV int rcr_scan_pairs(const int64_t* t0a, const int64_t* t1a,
+                                 const double* v0a, const double* v1a,
+                                 int64_t n, int64_t* d_ts, double* d_vals,
+                                 bool drop_stale, int64_t msi, int lane) {
+  double corr = 0.0;
+  double prev_raw = 0.0;
+  int64_t prev_ts = 0;
+  double prev_fin = 0.0;
+#pragma unroll
+  for (int c = 0; c < PIPE_PCHUNKS; c++) {
+    int64_t base = (int64_t)c * 2 * WAVE;
+    if (base >= n) continue;
+    int64_t k0 = base + 2 * lane;
+    int64_t k1 = k0 + 1;
+    bool a0 = k0 < n;
+    bool a1 = k1 < n;
+    double v0 = a0 ? v0a[c] : 0.0;
+    double v1 = a1 ? v1a[c] : 0.0;
+    int64_t t0 = a0 ? t0a[c] : 0;
+    int64_t t1 = a1 ? t1a[c] : 0;
+    if (drop_stale &&
+        __ballot((a0 && vm_is_stale_nan(v0)) ||
+                 (a1 && vm_is_stale_nan(v1))) != 0)
+      return -1;
+    if (a0) d_ts[k0 * 2] = t0;
+    if (a1) d_ts[k1 * 2] = t1;
+    int rem = (int)(n - base);
+    if (rem > 2 * WAVE) rem = 2 * WAVE;
+    int lastl = (rem - 1) >> 1;
+    bool last_is_e1 = ((rem - 1) & 1) != 0;
+    /* previous element: elem0's is the neighbor lane's elem1; elem1's is
+     * the in-lane elem0 (no cross-lane op) */
+    double pv0 = __shfl_up(v1, 1);
+    int64_t pt0 = __shfl_up(t1, 1);
+    if (lane == 0) { pv0 = prev_raw; pt0 = prev_ts; }
+    bool isfirst0 = (k0 == 0);
+    double d0 = v0 - pv0;
+    double d1 = v1 - v0;
+    double inc0 = 0.0, inc1 = 0.0;
+    if (!isfirst0 && d0 < 0) inc0 = ((-d0 * 8) < pv0) ? (pv0 - v0) : pv0;
+    if (d1 < 0) inc1 = ((-d1 * 8) < v0) ? (v0 - v1) : v0;
+    bool gap0 = (!isfirst0 && msi > 0 && (t0 - pt0) > msi);
+    bool gap1 = (msi > 0 && (t1 - t0) > msi);
+    bool evt0 = a0 && (gap0 || inc0 != 0.0);
+    bool evt1 = a1 && (gap1 || inc1 != 0.0);
+    uint64_t em = __ballot(evt0 || evt1);
+    uint64_t dm = __ballot((a0 && !isfirst0 && d0 < 0) || (a1 && d1 < 0));
+    if (em == 0 && dm == 0 &&
+        (base == 0 || __shfl(v0, 0) + corr >= prev_fin)) {
+      if (a0) d_vals[k0 * 2] = v0 + corr;
+      if (a1) d_vals[k1 * 2] = v1 + corr;
+      double lv = last_is_e1 ? v1 : v0;
+      int64_t lt = last_is_e1 ? t1 : t0;
+      prev_raw = __shfl(lv, lastl);
+      prev_ts = __shfl(lt, lastl);
+      prev_fin = prev_raw + corr;
+      continue;
+    }
+    /* correction walk in exact element order (elem0 of lane b, then its
+     * elem1) — identical serial sum to the reference */
+    double cc = corr;
+    double mc0 = corr, mc1 = corr;
+    uint64_t w = em;
+    while (w) {
+      int b = __ffsll((unsigned long long)w) - 1;
+      w &= w - 1;
+      double i0b = __shfl(inc0, b);
+      double i1b = __shfl(inc1, b);
+      int fl = __shfl((int)gap0 | ((int)gap1 << 1) | ((int)evt0 << 2) |
+                          ((int)evt1 << 3), b);
+      if (fl & 4) { /* event at element 0 of lane b: applies to k >= 2b */
+        double cn = (fl & 1) ? 0.0 : (cc + i0b);
+        if (lane >= b) { mc0 = cn; mc1 = cn; }
+        cc = cn;
+      }
+      if (fl & 8) { /* event at element 1 of lane b: applies to k >= 2b+1 */
+        double cn = (fl & 2) ? 0.0 : (cc + i1b);
+        if (lane > b) mc0 = cn;
+        if (lane >= b) mc1 = cn;
+        cc = cn;
+      }
+    }
+    double fin0 = v0 + mc0;
+    double fin1 = v1 + mc1;
+    /* monotonic clamp = segmented prefix max; boundary elements (series
+     * start / staleness gaps) keep their value and start a new segment */
+    bool f0 = isfirst0 || gap0;
+    bool f1 = gap1;
+    double x0 = fin0, x1 = fin1;
+    if (lane == 0 && !f0) x0 = fmax(x0, prev_fin);
+    double pm = f1 ? x1 : fmax(x0, x1);
+    int pf = (f0 || f1) ? 1 : 0;
+    for (int dlt = 1; dlt < WAVE; dlt <<= 1) {
+      double pmo = __shfl_up(pm, dlt);
+      int pfo = __shfl_up(pf, dlt);
+      if (lane >= dlt) {
+        if (!pf) pm = fmax(pm, pmo);
+        pf |= pfo;
+      }
+    }
+    double inc_m = __shfl_up(pm, 1); /* exclusive incoming (lane 0 unused) */
+    if (lane > 0 && !f0) x0 = fmax(x0, inc_m);
+    if (!f1) x1 = fmax(x1, x0);
+    if (a0) d_vals[k0 * 2] = x0;
+    if (a1) d_vals[k1 * 2] = x1;
+    corr = cc;
+    {
+      double lvr = last_is_e1 ? v1 : v0;
+      int64_t ltr = last_is_e1 ? t1 : t0;
+      double lxf = last_is_e1 ? x1 : x0;
+      prev_raw = __shfl(lvr, lastl);
+      prev_ts = __shfl(ltr, lastl);
+      prev_fin = __shfl(lxf, lastl);
+    }
+  }
+  return (int)n;
+}
+
 /* VMGPU_PIPE_MINWAVES forces an occupancy floor (waves/SIMD) on the pipe
  * kernel for A/B builds; VMGPU_PIPE_UNROLL (2 or 4) sets the eval ILP. */
 #ifndef VMGPU_PIPE_UNROLL
@@ -1328,8 +1457,10 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
   const uint32_t wave_id = blockIdx.x * WAVES_PER_BLOCK + wave_in_block;
   const uint32_t wave_stride = gridDim.x * WAVES_PER_BLOCK;
 
-  int64_t rt[PIPE_CHUNKS];
-  double rv[PIPE_CHUNKS];
+  /* pair staging registers: lane l holds samples (2l, 2l+1) of each
+   * 128-sample chunk, loaded with ONE b128 per column per chunk */
+  int64_t rt0[PIPE_PCHUNKS], rt1[PIPE_PCHUNKS];
+  double rv0[PIPE_PCHUNKS], rv1[PIPE_PCHUNKS];
 
   /* J-scatter mode (A/B builds only: -DVMGPU_PIPE_SCATTER): plan-uniform
    * window that is a step multiple + an ext jbuf (jbuf_mode 2).  The
@@ -1367,16 +1498,20 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
     lo = io.offsets[s];
     n = (int64_t)(io.offsets[s + 1] - lo);
     if (io.series_si) si_raw = io.series_si[s];
-    if (n > 0) {
+    if (n > 0 && (lo & 1) == 0) {
       const int64_t* gts = io.ts + lo;
       const double* gvs = io.vals + lo;
-      const int64_t nm1 = n - 1;
+      const int64_t nm2 = (n - 1) & ~(int64_t)1;
 #pragma unroll
-      for (int c = 0; c < PIPE_CHUNKS; c++) {
-        int64_t k = (int64_t)c * WAVE + lane;
-        if (k > nm1) k = nm1;
-        rt[c] = gts[k];
-        rv[c] = gvs[k];
+      for (int c = 0; c < PIPE_PCHUNKS; c++) {
+        int64_t k = (int64_t)c * 2 * WAVE + 2 * lane;
+        if (k > nm2) k = nm2; /* batch columns carry 16 B slack for this */
+        vm_i64x2 tp = *(const vm_i64x2*)(gts + k);
+        double2 vp = *(const double2*)(gvs + k);
+        rt0[c] = tp.x;
+        rt1[c] = tp.y;
+        rv0[c] = vp.x;
+        rv1[c] = vp.y;
       }
     }
   }
@@ -1401,54 +1536,62 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
       for (int e = lane; e < ext; e += WAVE) jbuf[e] = 0;
       wave_ds_sync();
     }
-    /* stage the current series into LDS from registers (rcr fused) */
+    /* stage the current series into LDS from registers (rcr fused);
+     * odd-offset series (no 16-B-aligned pair loads) fall to the global
+     * compact path below */
     int count = -1;
+    const bool pair_ok = ((lo & 1) == 0);
 #ifdef VMGPU_PIPE_ABL_NO_SCAN
     if (false) {
 #else
-    if (p.rcr) {
+    if (p.rcr && pair_ok) {
 #endif
-      count = rcr_scan_regs<2>(rt, rv, n, lts, lvs, p.drop_stale != 0,
-                               p.max_staleness, lane,
-                               scat_dg > 0 ? jbuf : nullptr, scat_dg,
-                               p.start, p.step, inv_gstep, p.n_grid);
-    } else {
+      count = rcr_scan_pairs(rt0, rt1, rv0, rv1, n, lts, lvs,
+                             p.drop_stale != 0, p.max_staleness, lane);
+    } else if (pair_ok) {
       bool stale = false;
       if (p.drop_stale) {
 #pragma unroll
-        for (int c = 0; c < PIPE_CHUNKS; c++) {
-          int64_t k = (int64_t)c * WAVE + lane;
-          if (__ballot(k < n && vm_is_stale_nan(rv[c])) != 0) stale = true;
+        for (int c = 0; c < PIPE_PCHUNKS; c++) {
+          int64_t k = (int64_t)c * 2 * WAVE + 2 * lane;
+          if (__ballot((k < n && vm_is_stale_nan(rv0[c])) ||
+                       (k + 1 < n && vm_is_stale_nan(rv1[c]))) != 0)
+            stale = true;
         }
       }
       if (!stale) {
 #pragma unroll
-        for (int c = 0; c < PIPE_CHUNKS; c++) {
-          int64_t k = (int64_t)c * WAVE + lane;
-          if (k < n) { lts[k * 2] = rt[c]; lvs[k * 2] = rv[c]; }
+        for (int c = 0; c < PIPE_PCHUNKS; c++) {
+          int64_t k = (int64_t)c * 2 * WAVE + 2 * lane;
+          if (k < n) { lts[k * 2] = rt0[c]; lvs[k * 2] = rv0[c]; }
+          if (k + 1 < n) {
+            lts[(k + 1) * 2] = rt1[c];
+            lvs[(k + 1) * 2] = rv1[c];
+          }
         }
-        if (scat_dg > 0)
-          scatter_j_regs(rt, n, lane, jbuf, scat_dg, p.start, p.step,
-                         inv_gstep, p.n_grid);
         count = (int)n;
       }
     }
     /* the staging registers are dead now — prefetch the NEXT series into
      * them; these loads stay in flight across scrape/seek/eval/emit */
-    if (has_next && n_n > 0) {
+    if (has_next && n_n > 0 && (lo_n & 1) == 0) {
       const int64_t* gts = io.ts + lo_n;
       const double* gvs = io.vals + lo_n;
-      const int64_t nm1 = n_n - 1;
+      const int64_t nm2 = (n_n - 1) & ~(int64_t)1;
 #pragma unroll
-      for (int c = 0; c < PIPE_CHUNKS; c++) {
-        int64_t k = (int64_t)c * WAVE + lane;
-        if (k > nm1) k = nm1;
-        rt[c] = gts[k];
-        rv[c] = gvs[k];
+      for (int c = 0; c < PIPE_PCHUNKS; c++) {
+        int64_t k = (int64_t)c * 2 * WAVE + 2 * lane;
+        if (k > nm2) k = nm2;
+        vm_i64x2 tp = *(const vm_i64x2*)(gts + k);
+        double2 vp = *(const double2*)(gvs + k);
+        rt0[c] = tp.x;
+        rt1[c] = tp.y;
+        rv0[c] = vp.x;
+        rv1[c] = vp.y;
       }
     }
     if (count < 0) {
-      /* stale NaNs must be compacted out (rare): redo from global */
+      /* stale NaNs present or odd-offset series (rare): from global */
       count = load_compact_wave<2>(io.ts + lo, io.vals + lo, n, lts, lvs,
                                    p.drop_stale != 0, lane);
       wave_ds_sync();
@@ -2688,8 +2831,8 @@ static int relayout_by_group(Batch& b, const int32_t* group_ids,
   uint64_t* d_soff = nullptr;
   uint64_t* d_doff = nullptr;
   uint32_t* d_perm = nullptr;
-  HIP_TRY(vm_dev_malloc(&d_ts2, total * 8), "alloc relayout ts");
-  HIP_TRY(vm_dev_malloc(&d_vals2, total * 8), "alloc relayout vals");
+  HIP_TRY(vm_dev_malloc(&d_ts2, total * 8 + 16), "alloc relayout ts");
+  HIP_TRY(vm_dev_malloc(&d_vals2, total * 8 + 16), "alloc relayout vals");
   HIP_TRY(vm_dev_malloc(&d_soff, (size_t)(n_series + 1) * 8), "alloc relayout soff");
   HIP_TRY(vm_dev_malloc(&d_doff, (size_t)(n_series + 1) * 8), "alloc relayout doff");
   HIP_TRY(vm_dev_malloc(&d_perm, (size_t)n_series * 4), "alloc relayout perm");
@@ -2808,8 +2951,10 @@ int vmgpu_batch_create(const int64_t* ts, const double* vals,
   b.n_groups = n_groups;
   b.n_samples = offsets[n_series];
 
-  HIP_TRY(vm_dev_malloc(&b.d_ts, b.n_samples * sizeof(int64_t)), "alloc ts");
-  HIP_TRY(vm_dev_malloc(&b.d_vals, b.n_samples * sizeof(double)), "alloc vals");
+  /* +16 B slack: the pipe kernel's clamped tail PAIR load may touch one
+   * element past the last series */
+  HIP_TRY(vm_dev_malloc(&b.d_ts, b.n_samples * sizeof(int64_t) + 16), "alloc ts");
+  HIP_TRY(vm_dev_malloc(&b.d_vals, b.n_samples * sizeof(double) + 16), "alloc vals");
   HIP_TRY(hipMemcpy(b.d_ts, ts, b.n_samples * sizeof(int64_t), hipMemcpyHostToDevice), "upload ts");
   HIP_TRY(hipMemcpy(b.d_vals, vals, b.n_samples * sizeof(double), hipMemcpyHostToDevice), "upload vals");
 
