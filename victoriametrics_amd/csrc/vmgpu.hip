@@ -1506,8 +1506,17 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
       for (int c = 0; c < PIPE_PCHUNKS; c++) {
         int64_t k = (int64_t)c * 2 * WAVE + 2 * lane;
         if (k > nm2) k = nm2; /* batch columns carry 16 B slack for this */
+#ifdef VMGPU_PIPE_NT
+        typedef long long v2i64 __attribute__((ext_vector_type(2)));
+        typedef double v2f64 __attribute__((ext_vector_type(2)));
+        v2i64 tpv = __builtin_nontemporal_load((const v2i64*)(gts + k));
+        v2f64 vpv = __builtin_nontemporal_load((const v2f64*)(gvs + k));
+        vm_i64x2 tp = {tpv.x, tpv.y};
+        double2 vp = {vpv.x, vpv.y};
+#else
         vm_i64x2 tp = *(const vm_i64x2*)(gts + k);
         double2 vp = *(const double2*)(gvs + k);
+#endif
         rt0[c] = tp.x;
         rt1[c] = tp.y;
         rv0[c] = vp.x;
@@ -1582,8 +1591,17 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
       for (int c = 0; c < PIPE_PCHUNKS; c++) {
         int64_t k = (int64_t)c * 2 * WAVE + 2 * lane;
         if (k > nm2) k = nm2;
+#ifdef VMGPU_PIPE_NT
+        typedef long long v2i64 __attribute__((ext_vector_type(2)));
+        typedef double v2f64 __attribute__((ext_vector_type(2)));
+        v2i64 tpv = __builtin_nontemporal_load((const v2i64*)(gts + k));
+        v2f64 vpv = __builtin_nontemporal_load((const v2f64*)(gvs + k));
+        vm_i64x2 tp = {tpv.x, tpv.y};
+        double2 vp = {vpv.x, vpv.y};
+#else
         vm_i64x2 tp = *(const vm_i64x2*)(gts + k);
         double2 vp = *(const double2*)(gvs + k);
+#endif
         rt0[c] = tp.x;
         rt1[c] = tp.y;
         rv0[c] = vp.x;
@@ -1650,7 +1668,11 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
     double* out_row = io.out + (size_t)s * (size_t)p.n_grid;
     auto emit = [&](int g, double v) {
       if constexpr (GROUPED) vm_emit_value(p, io, s, g, v);
+#ifdef VMGPU_PIPE_NT
+      else __builtin_nontemporal_store(v, out_row + g);
+#else
       else out_row[g] = v;
+#endif
     };
     const int64_t t_step_wave = (int64_t)WAVE * p.step;
     bool done = false;
