@@ -1,0 +1,323 @@
+"""Binary-operator pins transcribed from the reference's TestExecSuccess
+(app/vmselect/promql/exec_test.go; the t.Run names are kept in each test's
+comment with the exec_test.go line).  Grid start=1000e3 end=2000e3
+step=200e3; `time()` = [1000..2000].  Expected arrays/labels are the
+reference's own outputs, copied verbatim; the per-point math is delegated
+to the oracle (test infra) exactly as in test_binary_op.py, so these pin
+the HOST matching layers (createTimeseriesMapByTagSet, adjustBinaryOpTags,
+group_left/right joins, the `or` merge walk, removeEmptySeries at the exec
+tail)."""
+import math
+
+import numpy as np
+import pytest
+
+import oracle
+from victoriametrics_amd.binary_op import (BinOpSpec, Series,
+                                           binary_op_eval,
+                                           remove_empty_series)
+from victoriametrics_amd.metric_name import MetricName
+
+from test_binary_op import _eval  # oracle-backed apply/mask/or fns
+
+NAN = math.nan
+TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+
+
+def S(name, tags, values):
+    return Series(MetricName(name, tags),
+                  np.asarray(values, np.float64).copy())
+
+
+def t_series(extra=0.0):
+    return S("", [], TIME + extra)
+
+
+def scalar(v):
+    return S("", [], np.full(6, float(v)))
+
+
+def by_tag(out, key):
+    got = {}
+    for s in out:
+        got[s.mn.get_tag_value(key) or b""] = s
+    return got
+
+
+def eq(values, expected):
+    a = np.asarray(values, np.float64)
+    b = np.asarray(expected, np.float64)
+    an, bn = np.isnan(a), np.isnan(b)
+    assert (an == bn).all(), (a, b)
+    np.testing.assert_array_equal(a[~an], b[~bn])
+
+
+# ---------------------------------------------------------------------
+# comparisons (exec_test.go:2834-2980)
+# ---------------------------------------------------------------------
+
+def test_time_gt_bool_scalar():
+    # `time() >bool 1234` :2834
+    out = _eval(BinOpSpec(">", bool_modifier=True), [t_series()],
+                [scalar(1234)])
+    eq(out[0].values, [0, 0, 1, 1, 1, 1])
+
+
+def test_nan_gt_bool_scalar():
+    # `(time() > 1234) >bool 1450` :2845 — NaNs from the inner filter stay
+    inner = np.where(TIME > 1234, TIME, NAN)
+    out = _eval(BinOpSpec(">", bool_modifier=True), [S("", [], inner)],
+                [scalar(1450)])
+    eq(out[0].values, [NAN, NAN, 0, 1, 1, 1])
+
+
+def test_nan_ne_bool_scalar():
+    # `(time() > 1234) !=bool 1400` :2856.  (The reference's companion case
+    # `1400 !=bool (time() > 1234)` :2867 relies on metricsql canonicalizing
+    # scalar-op-vector into vector-op-scalar BEFORE evaluation — parser-side
+    # work that stays in the host Go layer per the §8b seam, so the engine
+    # is handed the canonical operand order.)
+    inner = np.where(TIME > 1234, TIME, NAN)
+    out = _eval(BinOpSpec("!=", bool_modifier=True), [S("", [], inner)],
+                [scalar(1400)])
+    eq(out[0].values, [NAN, NAN, 0, 1, 1, 1])
+
+
+def test_scalar_gt_time_empty():
+    # `123 > time()` :2878 — all filtered; removeEmptySeries at the tail
+    out = _eval(BinOpSpec(">"), [scalar(123)], [t_series()])
+    assert remove_empty_series(out) == []
+
+
+def test_cmp_bool_drops_metric_group():
+    # `a cmp bool scalar (drop MetricGroup)` :2930
+    left = [S("foo", [("a", "x")], TIME),
+            S("bar", [("a", "y")], TIME + 200)]
+    out = _eval(BinOpSpec(">=", bool_modifier=True), left, [scalar(1200)])
+    got = by_tag(out, b"a")
+    assert got[b"x"].mn.metric_group == b""
+    assert got[b"y"].mn.metric_group == b""
+    eq(got[b"x"].values, [0, 1, 1, 1, 1, 1])
+    eq(got[b"y"].values, [1, 1, 1, 1, 1, 1])
+
+
+# ---------------------------------------------------------------------
+# and / unless (exec_test.go:3160-3205)
+# ---------------------------------------------------------------------
+
+def test_one_and_empty():
+    # `1 and (0 > 1)` :3160 (issue 6637): empty right -> empty
+    out = _eval(BinOpSpec("and"), [scalar(1)], [])
+    assert remove_empty_series(out) == []
+
+
+def test_time_and_scalar():
+    # `time() and 2` :3166
+    out = _eval(BinOpSpec("and"), [t_series()], [scalar(2)])
+    assert len(out) == 1
+    eq(out[0].values, [1000, 1200, 1400, 1600, 1800, 2000])
+
+
+def test_time_and_filtered():
+    # `time() and time() > 1300` :3177
+    right = np.where(TIME > 1300, TIME, NAN)
+    out = _eval(BinOpSpec("and"), [t_series()], [S("", [], right)])
+    eq(out[0].values, [NAN, NAN, 1400, 1600, 1800, 2000])
+
+
+def test_time_unless_scalar_empty():
+    # `time() unless 2` :3188
+    out = _eval(BinOpSpec("unless"), [t_series()], [scalar(2)])
+    assert remove_empty_series(out) == []
+
+
+def test_time_unless_filtered():
+    # `time() unless time() > 1500` :3194
+    right = np.where(TIME > 1500, TIME, NAN)
+    out = _eval(BinOpSpec("unless"), [t_series()], [S("", [], right)])
+    eq(out[0].values, [1000, 1200, 1400, NAN, NAN, NAN])
+
+
+# ---------------------------------------------------------------------
+# or / default (exec_test.go:3205-3360)
+# ---------------------------------------------------------------------
+
+def test_series_or_series():
+    # `series or series` :3205 — same-key left wins whole, disjoint added
+    left = [S("", [("x", "foo")], TIME), S("", [("x", "bar")], TIME + 1)]
+    right = [S("", [("x", "foo")], TIME + 2), S("", [("x", "baz")], TIME + 3)]
+    out = _eval(BinOpSpec("or"), left, right)
+    got = by_tag(out, b"x")
+    assert set(got) == {b"foo", b"bar", b"baz"}
+    eq(got[b"foo"].values, TIME)
+    eq(got[b"bar"].values, TIME + 1)
+    eq(got[b"baz"].values, TIME + 3)
+
+
+def test_scalar_or_scalar():
+    # `time() > 1400 or 123` :3250
+    left = np.where(TIME > 1400, TIME, NAN)
+    out = _eval(BinOpSpec("or"), [S("", [], left)], [scalar(123)])
+    assert len(out) == 1
+    eq(out[0].values, [123, 123, 123, 1600, 1800, 2000])
+
+
+def test_scalar_default_scalar():
+    # `time() > 1400 default 123` :3276
+    left = np.where(TIME > 1400, TIME, NAN)
+    out = _eval(BinOpSpec("default"), [S("", [], left)], [scalar(123)])
+    eq(out[0].values, [123, 123, 123, 1600, 1800, 2000])
+
+
+def test_scalar_default_unmatched_vector():
+    # `time() > 1400 default label_set(123, "foo", "bar")` :3298 —
+    # differing tag sets do not match; left stays filtered
+    left = np.where(TIME > 1400, TIME, NAN)
+    out = _eval(BinOpSpec("default"), [S("", [], left)],
+                [S("", [("foo", "bar")], np.full(6, 123.0))])
+    assert len(out) == 1
+    eq(out[0].values, [NAN, NAN, NAN, 1600, 1800, 2000])
+
+
+def test_scalar_default_vector_matching_by_tags():
+    # `time() > 1400 default (label_set(123,"foo","bar"),
+    #  label_set(456,"__name__","xxx"))` :3309 — grouping keys use TAGS
+    # only (metric group excluded), so the tagless xxx series matches
+    left = np.where(TIME > 1400, TIME, NAN)
+    right = [S("", [("foo", "bar")], np.full(6, 123.0)),
+             S("xxx", [], np.full(6, 456.0))]
+    out = _eval(BinOpSpec("default"), [S("", [], left)], right)
+    assert len(out) == 1
+    eq(out[0].values, [456, 456, 456, 1600, 1800, 2000])
+
+
+def test_scalar_default_nan():
+    # `time() > 1400 default (time() < -100)` :3323 — NaN defaults change
+    # nothing
+    left = np.where(TIME > 1400, TIME, NAN)
+    out = _eval(BinOpSpec("default"), [S("", [], left)],
+                [S("", [], np.full(6, NAN))])
+    eq(out[0].values, [NAN, NAN, NAN, 1600, 1800, 2000])
+
+
+def test_vector_default_scalar():
+    # `vector default scalar` :3334 — per-series fill, names kept
+    lx = np.where(TIME > 1400, TIME, NAN)
+    ly = np.where(TIME < 1700, TIME, NAN)
+    out = _eval(BinOpSpec("default"),
+                [S("x", [("foo", "bar")], lx), S("y", [("foo", "baz")], ly)],
+                [scalar(123)])
+    got = by_tag(out, b"foo")
+    assert got[b"bar"].mn.metric_group == b"x"
+    assert got[b"baz"].mn.metric_group == b"y"
+    eq(got[b"bar"].values, [123, 123, 123, 1600, 1800, 2000])
+    eq(got[b"baz"].values, [1000, 1200, 1400, 1600, 123, 123])
+
+
+# ---------------------------------------------------------------------
+# group_left / group_right (exec_test.go:3473-3760)
+# ---------------------------------------------------------------------
+
+def test_scalar_mul_on_group_right():
+    # `2 * on() group_right() (...)` :3473
+    right = [S("", [("foo", "bar")], TIME),
+             S("", [("foo", "qwert")], np.full(6, 10.0))]
+    out = _eval(BinOpSpec("*", group_op="on", group_tags=[],
+                          join_op="group_right"), [scalar(2)], right)
+    got = by_tag(out, b"foo")
+    eq(got[b"bar"].values, [2000, 2400, 2800, 3200, 3600, 4000])
+    eq(got[b"qwert"].values, [20] * 6)
+    assert got[b"bar"].mn.metric_group == b""
+
+
+def test_scalar_mul_on_group_right_keep_metric_names():
+    # :3497 — keep_metric_names keeps the many-side (right) names
+    right = [S("q1", [("foo", "bar")], TIME),
+             S("q2", [("foo", "qwert")], np.full(6, 10.0))]
+    out = _eval(BinOpSpec("*", group_op="on", group_tags=[],
+                          join_op="group_right", keep_metric_names=True),
+                [scalar(2)], right)
+    got = by_tag(out, b"foo")
+    assert got[b"bar"].mn.metric_group == b"q1"
+    assert got[b"qwert"].mn.metric_group == b"q2"
+    eq(got[b"bar"].values, [2000, 2400, 2800, 3200, 3600, 4000])
+    eq(got[b"qwert"].values, [20] * 6)
+
+
+def test_scalar_mul_ignoring_group_right_join_tag():
+    # `label_set(2,"a","2") * ignoring(foo,a) group_right(a) (...)` :3523 —
+    # the join tag a is copied from the ONE (left) side onto every result
+    left = [S("", [("a", "2")], np.full(6, 2.0))]
+    right = [S("", [("foo", "bar"), ("a", "1")], TIME),
+             S("", [("foo", "qwert")], np.full(6, 10.0))]
+    out = _eval(BinOpSpec("*", group_op="ignoring", group_tags=["foo", "a"],
+                          join_op="group_right", join_tags=["a"]),
+                left, right)
+    got = by_tag(out, b"foo")
+    assert got[b"bar"].mn.get_tag_value(b"a") == b"2"
+    assert got[b"qwert"].mn.get_tag_value(b"a") == b"2"
+    eq(got[b"bar"].values, [2000, 2400, 2800, 3200, 3600, 4000])
+    eq(got[b"qwert"].values, [20] * 6)
+
+
+def test_group_left_duplicate_differ_by_join_tag():
+    # `... + on(foo) group_left(op) (le/ge split)` :3629 — two many-side
+    # matches per one-side series are legal when the join tag disambiguates
+    left = [S("qwert", [("foo", "bar"), ("xx", "yy")], TIME / 10)]
+    r_le = np.where(TIME < 1400, TIME, NAN)
+    r_ge = np.where(TIME >= 1400, TIME, NAN)
+    right = [S("", [("foo", "bar"), ("op", "le")], r_le),
+             S("", [("foo", "bar"), ("op", "ge")], r_ge)]
+    out = _eval(BinOpSpec("+", group_op="on", group_tags=["foo"],
+                          join_op="group_left", join_tags=["op"]),
+                left, right)
+    got = by_tag(out, b"op")
+    eq(got[b"le"].values, [1100, 1320, NAN, NAN, NAN, NAN])
+    eq(got[b"ge"].values, [NAN, NAN, 1540, 1760, 1980, 2200])
+    # group_left keeps the left's extra tags
+    assert got[b"le"].mn.get_tag_value(b"xx") == b"yy"
+    assert got[b"ge"].mn.get_tag_value(b"xx") == b"yy"
+
+
+def test_on_duplicate_nonoverlapping_merge():
+    # `... + on(foo) (le/ge split)` :3676 — WITHOUT group_left the two
+    # non-overlapping right series merge into one result (on() labels only)
+    left = [S("qwert", [("foo", "bar"), ("xx", "yy")], TIME / 10)]
+    r_le = np.where(TIME < 1400, TIME, NAN)
+    r_ge = np.where(TIME >= 1400, TIME, NAN)
+    right = [S("", [("foo", "bar"), ("op", "le")], r_le),
+             S("", [("foo", "bar"), ("op", "ge")], r_ge)]
+    out = _eval(BinOpSpec("+", group_op="on", group_tags=["foo"]),
+                left, right)
+    assert len(out) == 1
+    eq(out[0].values, [1100, 1320, 1540, 1760, 1980, 2200])
+    assert out[0].mn.tags == [(b"foo", b"bar")]
+
+
+def test_on_group_left_empty_join_nonoverlapping_merge():
+    # `... + on(foo) group_left() (le/ge split)` :3696 — group_left()
+    # without join tags also merges, but keeps the left's extra tags
+    left = [S("qwert", [("foo", "bar"), ("xx", "yy")], TIME / 10)]
+    r_le = np.where(TIME < 1400, TIME, NAN)
+    r_ge = np.where(TIME >= 1400, TIME, NAN)
+    right = [S("", [("foo", "bar"), ("op", "le")], r_le),
+             S("", [("foo", "bar"), ("op", "ge")], r_ge)]
+    out = _eval(BinOpSpec("+", group_op="on", group_tags=["foo"],
+                          join_op="group_left"), left, right)
+    assert len(out) == 1
+    eq(out[0].values, [1100, 1320, 1540, 1760, 1980, 2200])
+    assert out[0].mn.get_tag_value(b"xx") == b"yy"
+
+
+def test_group_left_name_join_tag():
+    # `... + on(foo) group_left(__name__) label_set(..., "__name__","aaa")`
+    # :3720 — __name__ as a join tag copies the metric group from the right
+    left = [S("qwert", [("foo", "bar"), ("xx", "yy")], TIME / 10)]
+    right = [S("aaa", [("foo", "bar")], TIME)]
+    out = _eval(BinOpSpec("+", group_op="on", group_tags=["foo"],
+                          join_op="group_left", join_tags=["__name__"]),
+                left, right)
+    assert len(out) == 1
+    assert out[0].mn.metric_group == b"aaa"
+    eq(out[0].values, [1100, 1320, 1540, 1760, 1980, 2200])
+    assert out[0].mn.get_tag_value(b"xx") == b"yy"
