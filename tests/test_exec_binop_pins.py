@@ -321,3 +321,44 @@ def test_group_left_name_join_tag():
     assert out[0].mn.metric_group == b"aaa"
     eq(out[0].values, [1100, 1320, 1540, 1760, 1980, 2200])
     assert out[0].mn.get_tag_value(b"xx") == b"yy"
+
+
+def test_vector_plus_vector_partial_matching_keep_metric_names():
+    # `vector + vector partial matching keep_metric_names` :3857 —
+    # with keep_metric_names the metric group joins the matching key, so
+    # only the q1/t1=v1 pair matches; the result keeps the name
+    left = [S("q1", [("t1", "v1")], TIME),
+            S("q2", [("t2", "v2")], np.full(6, 10.0))]
+    right = [S("q1", [("t1", "v1")], np.full(6, 100.0)),
+             S("", [("t2", "v3")], TIME)]
+    out = remove_empty_series(
+        _eval(BinOpSpec("+", keep_metric_names=True), left, right))
+    assert len(out) == 1
+    assert out[0].mn.metric_group == b"q1"
+    assert out[0].mn.tags == [(b"t1", b"v1")]
+    eq(out[0].values, [1100, 1300, 1500, 1700, 1900, 2100])
+
+
+def test_vector_plus_vector_no_matching():
+    # `vector + vector no matching` :3877 -> empty
+    left = [S("", [("t2", "v1")], TIME),
+            S("", [("t2", "v2")], np.full(6, 10.0))]
+    right = [S("", [("t1", "v1")], np.full(6, 100.0)),
+             S("", [("t2", "v3")], TIME)]
+    out = remove_empty_series(_eval(BinOpSpec("+"), left, right))
+    assert out == []
+
+
+def test_vector_plus_vector_on_matching():
+    # `vector + vector on (foo, t2) matching` :3887 — only the t2=v3 pair
+    # matches; on() keeps only the on-tags present
+    left = [S("", [("t1", "v123"), ("t2", "v3")], TIME),
+            S("", [("t2", "v2")], np.full(6, 10.0))]
+    right = [S("", [("t1", "v1")], np.full(6, 100.0)),
+             S("", [("t2", "v3")], TIME)]
+    out = remove_empty_series(
+        _eval(BinOpSpec("+", group_op="on", group_tags=["foo", "t2"]),
+              left, right))
+    assert len(out) == 1
+    assert out[0].mn.tags == [(b"t2", b"v3")]
+    eq(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
