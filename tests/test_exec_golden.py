@@ -452,3 +452,64 @@ def test_range_stddev_stdvar_exec():
     v = tmat()
     v[(v <= 1200) | (v >= 1800)] = math.nan
     _exact(rounded("range_stdvar", v), [10000.0] * 6)
+
+
+# ---------------------------------------------------------------------
+# cross-series aggregate pins (exec_test.go:5941-6060): mode/share/zscore
+# through the colagg device kernels + the aggregate.py host layer,
+# rounded exactly as the reference query does
+# ---------------------------------------------------------------------
+
+def _kser(div, off):
+    from victoriametrics_amd.binary_op import Series
+    return Series(MetricName(b"", [(b"k", b"v")]), TIME / div + off)
+
+
+def _share_inputs():
+    from victoriametrics_amd.binary_op import Series
+    return [
+        Series(MetricName(b"", [(b"k", b"v1")]), TIME / 100 + 10),
+        Series(MetricName(b"", [(b"k", b"v2")]), TIME / 200 + 5),
+        Series(MetricName(b"", [(b"k", b"v3")]), TIME / 110 - 10),
+        Series(MetricName(b"", [(b"k", b"v4")]), TIME / 90 - 5),
+    ]
+
+
+def _round_milli(rows):
+    out = tfm.transform("round", np.vstack([s.values for s in rows]),
+                        args=[np.full(6, 0.001)])
+    return {s.mn.get_tag_value(b"k"): out[i] for i, s in enumerate(rows)}
+
+
+def test_mode_aggregate():
+    # `mode()` exec_test.go:5941 -> constant 3
+    from victoriametrics_amd import aggregate as agg
+    from victoriametrics_amd.binary_op import Series
+    series = [Series(MetricName(name.encode(), []), np.full(6, float(v)))
+              for name, v in [("m1", 3), ("m2", 2), ("m3", 3), ("m4", 4),
+                              ("m5", 3), ("m6", 2)]]
+    out = agg.aggregate("mode", series)
+    assert len(out) == 1
+    _exact(out[0].values, [3, 3, 3, 3, 3, 3])
+
+
+def test_share_aggregate():
+    # `share()` exec_test.go:5959 (round(share(...), 0.001) verbatim)
+    from victoriametrics_amd import aggregate as agg
+    out = agg.aggregate("share", _share_inputs())
+    got = _round_milli(out)
+    _exact(got[b"v1"], [0.554, 0.521, 0.487, 0.462, 0.442, 0.426])
+    _exact(got[b"v2"], [0.277, 0.26, 0.243, 0.231, 0.221, 0.213])
+    v3 = got[b"v3"]
+    assert math.isnan(v3[0])
+    _exact(v3[1:], [0.022, 0.055, 0.081, 0.1, 0.116])
+    _exact(got[b"v4"], [0.169, 0.197, 0.214, 0.227, 0.237, 0.245])
+
+
+def test_zscore_aggregate():
+    # `zscore()` exec_test.go:6038 (first two rows of the pin)
+    from victoriametrics_amd import aggregate as agg
+    out = agg.aggregate("zscore", _share_inputs())
+    got = _round_milli(out)
+    _exact(got[b"v1"], [1.482, 1.511, 1.535, 1.552, 1.564, 1.57])
+    _exact(got[b"v2"], [0.159, 0.058, -0.042, -0.141, -0.237, -0.329])

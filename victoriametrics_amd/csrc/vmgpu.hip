@@ -1707,11 +1707,19 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
           const int fill_full = (p.n_grid / WAVE) * WAVE;
           for (; gf < fill_full; gf += WAVE, te += t_step_wave) {
             int gj = (int)((double)(te - ts0) * idx_per_ms) + 1;
+#ifdef VMGPU_PIPE_WALK_PROBE
+            jbuf[gf] = (uint16_t)vm_ub_hint<2>(lts, count, te, gj);
+#else
             jbuf[gf] = (uint16_t)vm_ub_hint_fast<2>(lts, count, te, gj);
+#endif
           }
           if (gf < p.n_grid) {
             int gj = (int)((double)(te - ts0) * idx_per_ms) + 1;
+#ifdef VMGPU_PIPE_WALK_PROBE
+            jbuf[gf] = (uint16_t)vm_ub_hint<2>(lts, count, te, gj);
+#else
             jbuf[gf] = (uint16_t)vm_ub_hint_fast<2>(lts, count, te, gj);
+#endif
           }
         }
         wave_ds_sync();
