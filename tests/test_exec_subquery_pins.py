@@ -72,3 +72,68 @@ def test_rate_subquery_double_offset():
         "default_rollup", START_MS - off, END_MS - off, STEP_MS, 0, 0, inner)
     got = np.asarray(out).ravel()
     np.testing.assert_allclose(got, [0, 0, 0, 7, 5, 3], rtol=1e-12, atol=0)
+
+
+def _inner_fn(fn):
+    def inner(sq_start, sq_end, sq_step):
+        ts = np.arange(sq_start, sq_end + 1, sq_step, dtype=np.int64)
+        return fn(ts / 1e3).reshape(1, -1)
+    return inner
+
+
+def _sq(func, window, sq_step, fn, arg=0.0):
+    out, _, _ = engine.rollup_subquery(
+        func, START_MS, END_MS, STEP_MS, window, sq_step, _inner_fn(fn),
+        arg=arg)
+    return np.asarray(out).ravel()
+
+
+def _eq(got, want):
+    g = np.asarray(got, np.float64)
+    w = np.asarray(want, np.float64)
+    gn, wn = np.isnan(g), np.isnan(w)
+    assert (gn == wn).all(), (g, w)
+    np.testing.assert_allclose(g[~gn], w[~wn], rtol=1e-12, atol=0)
+
+
+def test_duration_over_time():
+    # `duration_over_time((time()<1200)[600s:10s], 20s)` exec_test.go ->
+    # [590, 580, 380, 180, nan, nan]
+    got = _sq("duration_over_time", 600_000, 10_000,
+              lambda t: np.where(t < 1200, t, np.nan), arg=20.0)
+    _eq(got, [590, 580, 380, 180, np.nan, np.nan])
+
+
+def test_mode_over_time():
+    # `mode_over_time(round(time()/500)[100s:1s])` -> [2, 2, 3, 3, 4, 4]
+    got = _sq("mode_over_time", 100_000, 1_000,
+              lambda t: np.round(t / 500))
+    _eq(got, [2, 2, 3, 3, 4, 4])
+
+
+def test_rate_over_sum():
+    # `rate_over_sum(round(time()/500)[100s:5s])` -> [.4,.4,.6,.6,.71,.8]
+    # (compared after round(_, 0.01) as in the reference query)
+    got = _sq("rate_over_sum", 100_000, 5_000, lambda t: np.round(t / 500))
+    from victoriametrics_amd.decimal import go_round
+    got = go_round(got * 100.0) / 100.0
+    _eq(got, [0.4, 0.4, 0.6, 0.6, 0.71, 0.8])
+
+
+def test_integrate_time():
+    # `integrate(time()/1e3)` -> [160, 200, 240, 280, 320, 360]
+    got = _sq("integrate", 0, 0, lambda t: t / 1e3)
+    _eq(got, [160, 200, 240, 280, 320, 360])
+
+
+def test_rate_of_time_is_one():
+    # `rate(label_set(alias(time(), "foo"), "x", "y"))` -> [1]*6
+    got = _sq("rate", 0, 0, lambda t: t)
+    _eq(got, [1, 1, 1, 1, 1, 1])
+
+
+def test_distinct_over_time_window():
+    # `distinct_over_time((time() < 1700)[500s])` -> [3, 3, 3, 3, 2, 1]
+    got = _sq("distinct_over_time", 500_000, 0,
+              lambda t: np.where(t < 1700, t, np.nan))
+    _eq(got, [3, 3, 3, 3, 2, 1])
