@@ -513,3 +513,20 @@ def test_zscore_aggregate():
     got = _round_milli(out)
     _exact(got[b"v1"], [1.482, 1.511, 1.535, 1.552, 1.564, 1.57])
     _exact(got[b"v2"], [0.159, 0.058, -0.042, -0.141, -0.237, -0.329])
+
+
+def test_range_trim_family_exec():
+    # `range_trim_outliers(0.5, time())` -> [nan,nan,1400,1600,nan,nan]
+    # `range_trim_spikes(0.2, time())`   -> [nan,1200,1400,1600,1800,nan]
+    # `range_trim_zscore(0.9, time())`   -> [nan,1200,1400,1600,1800,nan]
+    def nan_eq(got, want):
+        g = np.asarray(got, np.float64).ravel()
+        w = np.asarray(want, np.float64)
+        assert (np.isnan(g) == np.isnan(w)).all(), (g, w)
+        np.testing.assert_array_equal(g[~np.isnan(g)], w[~np.isnan(w)])
+    nan_eq(tfm.transform("range_trim_outliers", tmat(), scalar=0.5),
+           [math.nan, math.nan, 1400, 1600, math.nan, math.nan])
+    nan_eq(tfm.transform("range_trim_spikes", tmat(), scalar=0.2),
+           [math.nan, 1200, 1400, 1600, 1800, math.nan])
+    nan_eq(tfm.transform("range_trim_zscore", tmat(), scalar=0.9),
+           [math.nan, 1200, 1400, 1600, 1800, math.nan])
