@@ -1663,16 +1663,44 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
       int64_t span_ms = lts[(count - 1) * 2] - ts0;
       idx_per_ms = span_ms > 0 ? (double)(count - 1) / (double)span_ms : 0.0;
     }
-    /* emit: GROUPED folds the aggregate switch in; the ungrouped build is
-     * a plain coalesced row store (no per-point branch, no switch). */
+    /* emit: GROUPED folds the aggregate switch in with the per-series
+     * group row hoisted out of the point loop; the ungrouped build is a
+     * plain coalesced row store (no per-point branch, no switch). */
     double* out_row = io.out + (size_t)s * (size_t)p.n_grid;
+    double* grp_vrow = nullptr;
+    double* grp_crow = nullptr;
+    if constexpr (GROUPED) {
+      int grp = io.group_ids ? io.group_ids[s] : -1;
+      if (grp >= 0) {
+        grp_vrow = io.out + (size_t)grp * (size_t)p.n_grid;
+        grp_crow = io.out_counts + (size_t)grp * (size_t)p.n_grid;
+      }
+    }
     auto emit = [&](int g, double v) {
-      if constexpr (GROUPED) vm_emit_value(p, io, s, g, v);
+      if constexpr (GROUPED) {
+        if (grp_vrow && !vm_isnan(v)) {
+          double* gv = grp_vrow + g;
+          double* gc = grp_crow + g;
+          switch (p.aggr) {
+            case VMGPU_AGGR_SUM: atomicAdd(gv, v); *gc = 1.0; break;
+            case VMGPU_AGGR_AVG: atomicAdd(gv, v); atomicAdd(gc, 1.0); break;
+            case VMGPU_AGGR_MIN: vm_atomic_min_f64(gv, v); *gc = 1.0; break;
+            case VMGPU_AGGR_MAX: vm_atomic_max_f64(gv, v); *gc = 1.0; break;
+            case VMGPU_AGGR_COUNT:
+            case VMGPU_AGGR_GROUP: atomicAdd(gv, 1.0); *gc = 1.0; break;
+            case VMGPU_AGGR_SUM2: atomicAdd(gv, v * v); *gc = 1.0; break;
+            case VMGPU_AGGR_GEOMEAN:
+              vm_atomic_mul_f64(gv, v); atomicAdd(gc, 1.0); break;
+            default: break;
+          }
+        }
+      } else {
 #ifdef VMGPU_PIPE_NT
-      else __builtin_nontemporal_store(v, out_row + g);
+        __builtin_nontemporal_store(v, out_row + g);
 #else
-      else out_row[g] = v;
+        out_row[g] = v;
 #endif
+      }
     };
     const int64_t t_step_wave = (int64_t)WAVE * p.step;
     bool done = false;
