@@ -87,15 +87,19 @@ def _worker(rank, world, aggr, port, ret):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("aggr", ["sum", "min", "max", "avg", "count"])
-def test_sharded_allreduce_merge(aggr, tmp_path):
+@pytest.mark.parametrize("aggr,world", [
+    ("sum", 2), ("min", 2), ("max", 2), ("avg", 2), ("count", 2),
+    # the 8-GPU shape's merge semantics at a wider world size
+    ("sum", 4), ("avg", 4),
+])
+def test_sharded_allreduce_merge(aggr, world, tmp_path):
     import oracle
     from victoriametrics_amd import synth
     ctx = mp.get_context("spawn")
     ret = ctx.Queue()
-    port = 29531 + hash(aggr) % 200
-    procs = [ctx.Process(target=_worker, args=(r, 2, aggr, port, ret))
-             for r in range(2)]
+    port = 29531 + (hash(aggr) + world * 37) % 200
+    procs = [ctx.Process(target=_worker, args=(r, world, aggr, port, ret))
+             for r in range(world)]
     for p in procs:
         p.start()
     got_aggr, fin = ret.get(timeout=180)
