@@ -137,3 +137,18 @@ def test_distinct_over_time_window():
     got = _sq("distinct_over_time", 500_000, 0,
               lambda t: np.where(t < 1700, t, np.nan))
     _eq(got, [3, 3, 3, 3, 2, 1])
+
+
+def test_quantiles_over_time_single_sample():
+    # `quantiles_over_time("phi", 0.5, 0.9, time()[100s:100s])`
+    # exec_test.go:6513 — one sample per window, so every phi returns the
+    # sample itself; the expansion tags each output phi=%g
+    plans = engine.quantiles_over_time_plans(
+        "phi", [0.5, 0.9], START_MS, END_MS, STEP_MS)
+    assert [lbl for lbl, _ in plans] == ["0.5", "0.9"]
+    for lbl, plan in plans:
+        out, _, _ = engine.rollup_subquery(
+            "quantile_over_time", START_MS, END_MS, STEP_MS, 100_000,
+            100_000, _inner_fn(lambda t: t), arg=plan._c.arg)
+        np.testing.assert_array_equal(
+            np.asarray(out).ravel(), [1000, 1200, 1400, 1600, 1800, 2000])
