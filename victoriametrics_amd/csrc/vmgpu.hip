@@ -1727,6 +1727,45 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
       if (!done && p.jbuf_mode >= 1 &&
           (size_t)p.n_grid * 2 <= vm_jbuf_bytes(p.jbuf_mode, p.jbuf_elems) &&
           dg64 > 0 && count <= 65535) {
+        if (dg64 < WAVE) {
+          /* FUSED seek+eval: with dg < 64 the window-START boundary
+           * j(g - dg) lives in THIS wave's just-computed j registers (or
+           * the previous round's) — one shuffle replaces the whole jbuf
+           * write/sync/read round trip.  Head lanes of round 0 (window
+           * start before the grid) probe for i directly. */
+          int j_prev = 0;
+          int g = lane;
+          int64_t te = p.start + (int64_t)lane * p.step;
+          for (int r = 0; r * WAVE < p.n_grid;
+               r++, g += WAVE, te += t_step_wave) {
+            bool act = g < p.n_grid;
+            int gj = (int)((double)(te - ts0) * idx_per_ms) + 1;
+            int j_cur = act
+                            ? vm_ub_hint_fast<2>(lts, count, te, gj)
+                            : count;
+            int i;
+            int i_same = __shfl(j_cur, lane - dg64);
+            int i_prev = __shfl(j_prev, lane + WAVE - dg64);
+            i = (lane >= dg64) ? i_same : i_prev;
+            if (r == 0 && lane < dg64) {
+              int64_t t_start = te - sw.window;
+              int gi = (int)((double)(t_start - ts0) * idx_per_ms) + 1;
+              i = vm_ub_hint_fast<2>(lts, count, t_start, gi);
+            }
+            if (act) {
+#ifdef VMGPU_PIPE_ABL_NO_EVAL
+              emit(g, (j_cur > 0 && j_cur <= count) ? lvs[(j_cur - 1) * 2] : 0.0);
+              (void)i;
+#else
+              emit(g, eval_rate_fused<2>(p, sw, lts, lvs, count, i, j_cur,
+                                         te - sw.window));
+#endif
+            }
+            j_prev = j_cur;
+          }
+          done = true;
+        }
+        if (!done) {
         /* j-cache fill: full lane-blocks without bounds checks, one masked
          * tail; t_end is strength-reduced (+64*step per round) */
         {
@@ -1797,6 +1836,7 @@ void rollup_pipe_kernel(KPlan p, KIO io) {
                                        te - sw.window));
 #endif
           }
+        }
         }
         done = true;
       }
