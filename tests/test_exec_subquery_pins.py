@@ -152,3 +152,19 @@ def test_quantiles_over_time_single_sample():
             100_000, _inner_fn(lambda t: t), arg=plan._c.arg)
         np.testing.assert_array_equal(
             np.asarray(out).ravel(), [1000, 1200, 1400, 1600, 1800, 2000])
+
+
+def test_deriv_cases():
+    # `deriv(1000)` -> zeros; `deriv(2*time())` -> 2 (exec_test.go:8784)
+    got = _sq("deriv", 0, 0, lambda t: np.full_like(t, 1000.0))
+    _eq(got, [0, 0, 0, 0, 0, 0])
+    got = _sq("deriv", 0, 0, lambda t: 2.0 * t)
+    _eq(got, [2, 2, 2, 2, 2, 2])
+
+
+def test_lag_subquery_nondivisor_step():
+    # `lag(time()[60s:17s])` exec_test.go:9294 -> [14, 10, 6, 2, 15, 11]
+    # (17s inner step does not divide the 200s outer step — pins the
+    # alignStartEnd + window walk at a misaligned cadence)
+    got = _sq("lag", 60_000, 17_000, lambda t: t)
+    _eq(got, [14, 10, 6, 2, 15, 11])
