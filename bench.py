@@ -51,7 +51,7 @@ def parse_args():
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--skip-cold", action="store_true",
                     help="skip the end-to-end cold-query measurement")
-    ap.add_argument("--cold-queries", type=int, default=3)
+    ap.add_argument("--cold-queries", type=int, default=5)
     ap.add_argument("--skip-grouped", action="store_true",
                     help="skip the secondary grouped (config 3) measurement")
     ap.add_argument("--grouped-steps", type=int, default=4)
