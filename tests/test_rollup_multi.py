@@ -119,3 +119,63 @@ def test_window_required():
     import pytest
     with pytest.raises(ValueError):
         count_values_over_time("foo", ts, vals, _mn(), START, START, STEP, 0)
+
+
+def test_count_values_fuzz_vs_brute_force():
+    """Random ragged inputs vs an independent brute-force model: per grid
+    point the multiset of formatted window values must match, and summing
+    every output series per point must equal count_over_time."""
+    rng = np.random.default_rng(99)
+    for trial in range(40):
+        n = int(rng.integers(1, 60))
+        ts = np.sort(rng.integers(START - 400_000, START + 400_000, n)
+                     ).astype(np.int64)
+        vals = np.round(rng.standard_normal(n) * rng.choice([1, 10, 1e6]), 2)
+        vals[rng.random(n) < 0.2] = np.round(vals[0], 2)  # duplicates
+        window = int(rng.choice([15_000, 60_000, 250_000]))
+        start = START + int(rng.integers(-5, 5)) * STEP
+        end = start + int(rng.integers(0, 6)) * STEP
+        series, scanned = count_values_over_time(
+            "v", ts, vals, _mn(), start, end, STEP, window)
+        n_grid = 1 + (end - start) // STEP
+        exp_scanned = 0
+        for g in range(n_grid):
+            t_end = start + g * STEP
+            in_win = (ts > t_end - window) & (ts <= t_end)
+            exp_scanned += int(in_win.sum())
+            counts = {}
+            for v in vals[in_win]:
+                k = format_go_float_g(float(v))
+                counts[k] = counts.get(k, 0) + 1
+            got = {}
+            for s in series:
+                x = s.values[g]
+                if not math.isnan(x):
+                    got[s.mn.get_tag_value(b"v").decode()] = int(x)
+            assert got == counts, (trial, g, got, counts)
+        assert scanned == exp_scanned
+
+
+def test_histogram_over_time_total_counts_property():
+    """Sum over all vmrange series at each point == number of COUNTABLE
+    window values (NaN and negatives are skipped by Histogram.Update)."""
+    rng = np.random.default_rng(7)
+    for trial in range(25):
+        n = int(rng.integers(1, 80))
+        ts = np.sort(rng.integers(START - 300_000, START + 300_000, n)
+                     ).astype(np.int64)
+        vals = rng.standard_normal(n) * 10.0 ** rng.integers(-5, 6, n)
+        vals[rng.random(n) < 0.1] = np.nan
+        window = int(rng.choice([30_000, 120_000]))
+        start, end = START, START + 4 * STEP
+        series, _ = histogram_over_time(ts, vals, _mn(), start, end, STEP,
+                                        window)
+        n_grid = 1 + (end - start) // STEP
+        for g in range(n_grid):
+            t_end = start + g * STEP
+            in_win = (ts > t_end - window) & (ts <= t_end)
+            w = vals[in_win]
+            countable = int((~np.isnan(w) & (w >= 0)).sum())
+            total = sum(int(s.values[g]) for s in series
+                        if not math.isnan(s.values[g]))
+            assert total == countable, (trial, g, total, countable)
