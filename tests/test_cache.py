@@ -447,3 +447,51 @@ def test_cache_persistence_missing_or_corrupt(tmp_path):
     bad.write_bytes(b"garbage")
     c2 = load_rollup_result_cache(str(bad))
     assert c2.get_series("x", 0, 15_000, 0, 15_000)[0] is None
+
+
+def test_cache_persistence_fuzz_equivalence(tmp_path):
+    """Random put/get/save/load interleavings: a reloaded cache must answer
+    every get_series/get_instant_values exactly like the live cache."""
+    from victoriametrics_amd.cache import (RollupResultCache,
+                                           load_rollup_result_cache,
+                                           save_rollup_result_cache)
+    rng = np.random.default_rng(12)
+    c = RollupResultCache()
+    step = 15_000
+    exprs = ["rate(a)", "sum(b)", "max(c)"]
+    now = 2_000_000_000_000
+    for i in range(120):
+        expr = exprs[int(rng.integers(len(exprs)))]
+        window = int(rng.choice([0, 300_000]))
+        n = int(rng.integers(2, 30))
+        t0 = 1_000_000_000_000 + int(rng.integers(0, 50)) * step
+        ts = t0 + np.arange(n, dtype=np.int64) * step
+        vals = rng.standard_normal((1, n))
+        names = [(b"m", ((b"i", str(i % 7).encode()),))]
+        if rng.random() < 0.8:
+            c.put_series(expr, window, step, names, vals, ts, now_ms=now)
+        else:
+            c.put_instant_values(expr, window, step, names,
+                                 vals[:, :1], ts[:1])
+    path = str(tmp_path / "cache")
+    save_rollup_result_cache(c, path)
+    c2 = load_rollup_result_cache(path)
+    for expr in exprs:
+        for window in (0, 300_000):
+            for probe in range(12):
+                start = 1_000_000_000_000 + int(rng.integers(0, 80)) * step
+                end = start + int(rng.integers(1, 40)) * step
+                a = c.get_series(expr, window, step, start, end)
+                b = c2.get_series(expr, window, step, start, end)
+                assert (a[0] is None) == (b[0] is None), (expr, window, start)
+                assert a[3] == b[3]
+                if a[0] is not None:
+                    assert a[0] == b[0]
+                    np.testing.assert_array_equal(a[1], b[1])
+                    np.testing.assert_array_equal(a[2], b[2])
+            ia = c.get_instant_values(expr, window, step)
+            ib = c2.get_instant_values(expr, window, step)
+            assert (ia[0] is None) == (ib[0] is None)
+            if ia[0] is not None:
+                assert ia[0] == ib[0] and ia[2] == ib[2]
+                np.testing.assert_array_equal(ia[1], ib[1])
