@@ -1264,6 +1264,114 @@ static VM_DEV int rcr_scan_regs(const int64_t* rt, const double* rv,
   return (int)n;
 }
 
+/* Pair-rounds removeCounterResets over an LDS/global COLUMN (stride STR
+ * in int64 units): the rcr_scan_pairs logic with per-round pair loads
+ * instead of register sources — 128 samples per round instead of 64, so
+ * the serial-round chain halves.  Bit-exact with rcr_scan_wave (same
+ * element-order event walk, same segmented-max clamp). */
+template <int STR = 1>
+static __device__ void rcr_scan_col_pairs(int64_t* d_ts, double* d_vals,
+                                          int count, int64_t msi, int lane) {
+  double corr = 0.0;
+  double prev_raw = 0.0;
+  int64_t prev_ts = 0;
+  double prev_fin = 0.0;
+  for (int base = 0; base < count; base += 2 * WAVE) {
+    int k0 = base + 2 * lane;
+    int k1 = k0 + 1;
+    bool a0 = k0 < count;
+    bool a1 = k1 < count;
+    double v0 = a0 ? d_vals[k0 * STR] : 0.0;
+    double v1 = a1 ? d_vals[k1 * STR] : 0.0;
+    int64_t t0 = a0 ? d_ts[k0 * STR] : 0;
+    int64_t t1 = a1 ? d_ts[k1 * STR] : 0;
+    int rem = count - base;
+    if (rem > 2 * WAVE) rem = 2 * WAVE;
+    int lastl = (rem - 1) >> 1;
+    bool last_is_e1 = ((rem - 1) & 1) != 0;
+    double pv0 = __shfl_up(v1, 1);
+    int64_t pt0 = __shfl_up(t1, 1);
+    if (lane == 0) { pv0 = prev_raw; pt0 = prev_ts; }
+    bool isfirst0 = (k0 == 0);
+    double d0 = v0 - pv0;
+    double d1 = v1 - v0;
+    double inc0 = 0.0, inc1 = 0.0;
+    if (!isfirst0 && d0 < 0) inc0 = ((-d0 * 8) < pv0) ? (pv0 - v0) : pv0;
+    if (d1 < 0) inc1 = ((-d1 * 8) < v0) ? (v0 - v1) : v0;
+    bool gap0 = (!isfirst0 && msi > 0 && (t0 - pt0) > msi);
+    bool gap1 = (msi > 0 && (t1 - t0) > msi);
+    bool evt0 = a0 && (gap0 || inc0 != 0.0);
+    bool evt1 = a1 && (gap1 || inc1 != 0.0);
+    uint64_t em = __ballot(evt0 || evt1);
+    uint64_t dm = __ballot((a0 && !isfirst0 && d0 < 0) || (a1 && d1 < 0));
+    if (em == 0 && dm == 0 &&
+        (base == 0 || __shfl(v0, 0) + corr >= prev_fin)) {
+      if (corr != 0.0) {
+        if (a0) d_vals[k0 * STR] = v0 + corr;
+        if (a1) d_vals[k1 * STR] = v1 + corr;
+      }
+      double lv = last_is_e1 ? v1 : v0;
+      int64_t lt = last_is_e1 ? t1 : t0;
+      prev_raw = __shfl(lv, lastl);
+      prev_ts = __shfl(lt, lastl);
+      prev_fin = prev_raw + corr;
+      continue;
+    }
+    double cc = corr;
+    double mc0 = corr, mc1 = corr;
+    uint64_t w = em;
+    while (w) {
+      int b = __ffsll((unsigned long long)w) - 1;
+      w &= w - 1;
+      double i0b = __shfl(inc0, b);
+      double i1b = __shfl(inc1, b);
+      int fl = __shfl((int)gap0 | ((int)gap1 << 1) | ((int)evt0 << 2) |
+                          ((int)evt1 << 3), b);
+      if (fl & 4) {
+        double cn = (fl & 1) ? 0.0 : (cc + i0b);
+        if (lane >= b) { mc0 = cn; mc1 = cn; }
+        cc = cn;
+      }
+      if (fl & 8) {
+        double cn = (fl & 2) ? 0.0 : (cc + i1b);
+        if (lane > b) mc0 = cn;
+        if (lane >= b) mc1 = cn;
+        cc = cn;
+      }
+    }
+    double fin0 = v0 + mc0;
+    double fin1 = v1 + mc1;
+    bool f0 = isfirst0 || gap0;
+    bool f1 = gap1;
+    double x0 = fin0, x1 = fin1;
+    if (lane == 0 && !f0) x0 = fmax(x0, prev_fin);
+    double pm = f1 ? x1 : fmax(x0, x1);
+    int pf = (f0 || f1) ? 1 : 0;
+    for (int dlt = 1; dlt < WAVE; dlt <<= 1) {
+      double pmo = __shfl_up(pm, dlt);
+      int pfo = __shfl_up(pf, dlt);
+      if (lane >= dlt) {
+        if (!pf) pm = fmax(pm, pmo);
+        pf |= pfo;
+      }
+    }
+    double inc_m = __shfl_up(pm, 1);
+    if (lane > 0 && !f0) x0 = fmax(x0, inc_m);
+    if (!f1) x1 = fmax(x1, x0);
+    if (a0) d_vals[k0 * STR] = x0;
+    if (a1) d_vals[k1 * STR] = x1;
+    corr = cc;
+    {
+      double lvr = last_is_e1 ? v1 : v0;
+      int64_t ltr = last_is_e1 ? t1 : t0;
+      double lxf = last_is_e1 ? x1 : x0;
+      prev_raw = __shfl(lvr, lastl);
+      prev_ts = __shfl(ltr, lastl);
+      prev_fin = __shfl(lxf, lastl);
+    }
+  }
+}
+
 /* J scatter for the no-preprocessing staging path (same semantics as the
  * fused scatter in rcr_scan_regs). */
 static VM_DEV void scatter_j_regs(const int64_t* rt, int64_t n, int lane,
@@ -1911,7 +2019,26 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KI
     }
     __syncthreads();
     int count = *sh_count;
-    if (p.rcr && wave == 0) rcr_scan_wave(lts, lvs, count, p.max_staleness, lane);
+    /* the counter-reset scan only touches VALUES and the j-cache fill only
+     * reads TIMESTAMPS: wave 0 scans (pair rounds — half the serial chain)
+     * while waves 1-3 fill the rate j-cache concurrently */
+    if constexpr (FUNC_CT == VMF_RATE || FUNC_CT == VMF_DERIV_FAST) {
+      if (has_jbuf && count > 1 && count <= 65535 && wave != 0) {
+        int64_t ts0f = lts[0];
+        int64_t span_ms = lts[count - 1] - ts0f;
+        double ipm = span_ms > 0 ? (double)(count - 1) / (double)span_ms : 0.0;
+        for (int g = (wave - 1) * WAVE + lane; g < p.n_grid;
+             g += (BLOCK_THREADS - WAVE)) {
+          int64_t t_end = p.start + (int64_t)g * p.step;
+          int gj = (int)((double)(t_end - ts0f) * ipm) + 1;
+          jbuf[g] = (uint16_t)vm_ub_hint_fast(lts, count, t_end, gj);
+        }
+      }
+    }
+#ifndef VMGPU_ABL_NO_RCR
+    if (p.rcr && wave == 0)
+      rcr_scan_col_pairs<1>(lts, lvs, count, p.max_staleness, lane);
+#endif
     __syncthreads();
     if (p.pre_func && wave == 0)
       pre_func_wave(lts, lvs, count, p.pre_func, lane);
@@ -1945,10 +2072,14 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KI
           int64_t span_ms = lts[count - 1] - ts0;
           idx_per_ms = span_ms > 0 ? (double)(count - 1) / (double)span_ms : 0.0;
         }
-        for (int g = tid; g < p.n_grid; g += BLOCK_THREADS) {
-          int64_t t_end = p.start + (int64_t)g * p.step;
-          int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
-          jbuf[g] = (uint16_t)vm_ub_hint_fast(lts, count, t_end, gj);
+        /* count > 1 mirrors the concurrent prefill's gate exactly: the
+         * j-cache was already built by waves 1-3 during the scan */
+        if (count <= 1) {
+          for (int g = tid; g < p.n_grid; g += BLOCK_THREADS) {
+            int64_t t_end = p.start + (int64_t)g * p.step;
+            int gj = (int)((double)(t_end - ts0) * idx_per_ms) + 1;
+            jbuf[g] = (uint16_t)vm_ub_hint_fast(lts, count, t_end, gj);
+          }
         }
         __syncthreads();
         for (int g = tid; g < p.n_grid; g += BLOCK_THREADS) {
