@@ -314,3 +314,30 @@ def test_incremental_aggr_callbacks_covered():
     fused = set(engine.AGGR_IDS) - {"none"}
     assert reference - fused == {"any"}
     assert callable(engine.any_representative_group_ids)
+
+
+def test_packed_create_fails_loudly_without_init():
+    """vmgpu_batch_create_packed through the C-ABI without vmgpu_init (or
+    without a GPU): a loud error, never a silent fallback."""
+    import torch
+    from victoriametrics_amd import engine
+    _ensure_built()
+    if torch.cuda.is_available():
+        pytest.skip("GPU present; loud-failure path not applicable")
+    import oracle
+    ts = np.arange(4, dtype=np.int64) * 15_000 + 1_000_000
+    vals = np.arange(4, dtype=np.int64)
+    packed, nb, sbs = oracle.pack_blocks(ts, vals,
+                                         np.asarray([0, 4], np.uint64))
+    with pytest.raises(engine.VmGpuError):
+        engine.SeriesBatch.from_packed(packed, nb, sbs)
+
+
+def test_host_alloc_symbols_exported():
+    """The pinned-staging entry points exist in the built library (the cgo
+    side links them)."""
+    import ctypes
+    lib = ctypes.CDLL(LIB)
+    for sym in ("vmgpu_batch_create_packed", "vmgpu_host_alloc",
+                "vmgpu_host_free"):
+        assert hasattr(lib, sym), sym
