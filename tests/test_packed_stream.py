@@ -95,3 +95,55 @@ def test_pack_rejects_empty_and_huge_series():
     tb = np.arange(big_n, dtype=np.int64)
     with pytest.raises(ValueError):
         oracle.pack_blocks(tb, tb, np.asarray([0, big_n], dtype=np.uint64))
+
+
+def test_native_parser_rejects_malformed_streams():
+    """The C parser (vmgpu_batch_create_packed) on truncated / mutated
+    streams: every call must return a nonzero error code (either a parse
+    error or, when the frame parses, the not-initialized error — this CPU
+    box has no GPU context) and never crash the process."""
+    import ctypes
+    import os
+    lib_path = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "victoriametrics_amd", "libvmgpu.so")
+    if not os.path.exists(lib_path):
+        pytest.skip("libvmgpu.so not built")
+    lib = ctypes.CDLL(lib_path)
+    rng = np.random.default_rng(3)
+    ts = (1_000_000 + np.arange(50, dtype=np.int64) * 15_000)
+    vals = np.arange(50, dtype=np.int64)
+    packed, nb, sbs = oracle.pack_blocks(
+        ts, vals, np.asarray([0, 25, 50], np.uint64))
+    base = bytearray(packed)
+
+    def call(buf, n_blocks, n_series=2):
+        arr = np.frombuffer(bytes(buf), np.uint8) if len(buf) else \
+            np.zeros(1, np.uint8)
+        sbs_a = np.arange(n_series + 1, dtype=np.uint32)
+        off = np.zeros(n_series + 1, dtype=np.uint64)
+        h = ctypes.c_uint64(0)
+        err = ctypes.create_string_buffer(256)
+        return lib.vmgpu_batch_create_packed(
+            arr.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)),
+            ctypes.c_uint64(len(buf)), ctypes.c_uint64(n_blocks),
+            sbs_a.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+            ctypes.c_uint32(n_series), ctypes.c_int64(0), None,
+            ctypes.c_uint32(0), ctypes.byref(h),
+            off.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)),
+            err, ctypes.c_size_t(256))
+
+    # truncations at every prefix length class
+    for cut in [0, 1, 47, 48, len(base) // 2, len(base) - 1]:
+        assert call(base[:cut], nb) != 0, cut
+    # block count lies
+    assert call(base, nb + 5) != 0
+    # random byte mutations in the headers (lengths/rows fields)
+    for _ in range(200):
+        buf = bytearray(base)
+        i = int(rng.integers(0, min(96, len(buf))))
+        buf[i] = int(rng.integers(0, 256))
+        rc = call(buf, nb)
+        assert isinstance(rc, int)  # no crash; rc value may legitimately
+        # parse (mutating payload bytes keeps the frame valid) — on this
+        # GPU-less box even a clean parse errors with "not initialized"
+        assert rc != 0
