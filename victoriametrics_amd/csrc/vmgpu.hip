@@ -1148,122 +1148,6 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_wave_kernel(KPlan p, KIO
 
 #define PIPE_CHUNKS 4 /* series cap = PIPE_CHUNKS*WAVE = 256 samples */
 
-/* removeCounterResets over register-staged chunks: the same exact-order
- * scan as rcr_scan_wave (rollup.go:921-958), inputs from registers, fused
- * with the LDS store.  Returns n, or -1 when a stale NaN must be compacted
- * out (caller falls back to load_compact_wave + rcr_scan_wave). */
-/* grid boundary of one sample: smallest g with t_end(g) >= t, clamped to
- * [-sdg, n_grid] (out-of-grid samples clamp; their scatter ranges become
- * empty or grid-clipped, which preserves upper-bound semantics exactly). */
-static VM_DEV int vm_scatter_boundary(int64_t t, int64_t gstart, int64_t gstep,
-                                      double inv_gstep, int sdg, int n_grid) {
-  int64_t t_lo_bound = gstart - (int64_t)sdg * gstep;
-  int64_t t_hi_bound = gstart + (int64_t)(n_grid - 1) * gstep;
-  if (t <= t_lo_bound) return -sdg;
-  if (t > t_hi_bound) return n_grid;
-  double est = floor((double)(t - gstart) * inv_gstep);
-  int g = (int)est - 1;
-  g += (gstart + (int64_t)g * gstep < t);
-  g += (gstart + (int64_t)g * gstep < t);
-  g += (gstart + (int64_t)g * gstep < t);
-  return g;
-}
-
-template <int STR = 1>
-static VM_DEV int rcr_scan_regs(const int64_t* rt, const double* rv,
-                                int64_t n, int64_t* d_ts, double* d_vals,
-                                bool drop_stale, int64_t msi, int lane,
-                                uint16_t* jb, int sdg, int64_t gstart,
-                                int64_t gstep, double inv_gstep, int n_grid) {
-  double corr = 0.0;
-  double prev_raw = 0.0;
-  int64_t prev_ts = 0;
-  double prev_fin = 0.0;
-  int carry_b = 0;
-#pragma unroll
-  for (int c = 0; c < PIPE_CHUNKS; c++) {
-    int64_t base = (int64_t)c * WAVE;
-    if (base >= n) continue;
-    int64_t k = base + lane;
-    bool active = k < n;
-    double v = active ? rv[c] : 0.0;
-    int64_t t = active ? rt[c] : 0;
-    if (drop_stale && __ballot(active && vm_is_stale_nan(v)) != 0) return -1;
-    if (active) d_ts[k * STR] = t;
-    if (jb) {
-      /* fused J scatter: sample k-1 is the window-end boundary for grid
-       * points in [b(t_{k-1}), b(t_k)) — write the upper-bound count k
-       * there; the last sample covers up to the grid end.  Replaces the
-       * per-point seek phase entirely (J[g+sdg] = #samples <= t_end(g)). */
-      int lastl = (int)(n - base - 1);
-      if (lastl > 63) lastl = 63;
-      int b_own = vm_scatter_boundary(t, gstart, gstep, inv_gstep, sdg, n_grid);
-      int b_prev = __shfl_up(b_own, 1);
-      if (lane == 0) b_prev = carry_b;
-      if (active && k > 0) {
-        int glo = b_prev < -sdg ? -sdg : b_prev;
-        int ghi = b_own > n_grid ? n_grid : b_own;
-        for (int g2 = glo; g2 < ghi; g2++) jb[g2 + sdg] = (uint16_t)k;
-      }
-      if (active && k == n - 1) {
-        int glo = b_own < -sdg ? -sdg : b_own;
-        for (int g2 = glo; g2 < n_grid; g2++) jb[g2 + sdg] = (uint16_t)n;
-      }
-      carry_b = __shfl(b_own, lastl);
-    }
-    double pv = __shfl_up(v, 1);
-    int64_t pt = __shfl_up(t, 1);
-    if (lane == 0) { pv = prev_raw; pt = prev_ts; }
-    bool isfirst = (k == 0);
-    double d = v - pv;
-    double inc = 0.0;
-    if (!isfirst && d < 0) inc = ((-d * 8) < pv) ? (pv - v) : pv;
-    bool gap = (!isfirst && msi > 0 && (t - pt) > msi);
-    uint64_t em = __ballot(active && (gap || inc != 0.0));
-    uint64_t dm = __ballot(active && !isfirst && d < 0);
-    int last = (int)(n - base - 1);
-    if (last > 63) last = 63;
-    if (em == 0 && dm == 0 &&
-        (base == 0 || __shfl(v, 0) + corr >= prev_fin)) {
-      if (active) d_vals[k * STR] = v + corr;
-      prev_raw = __shfl(v, last);
-      prev_ts = __shfl(t, last);
-      prev_fin = prev_raw + corr;
-      continue;
-    }
-    double cc = corr;
-    double mycorr = corr;
-    while (em) {
-      int b = __ffsll((unsigned long long)em) - 1;
-      em &= em - 1;
-      double ib = __shfl(inc, b);
-      int gb = __shfl((int)gap, b);
-      double cn = gb ? 0.0 : (cc + ib);
-      if (lane >= b) mycorr = cn;
-      cc = cn;
-    }
-    double fin = v + mycorr;
-    bool bnd = isfirst || gap;
-    double x = fin;
-    int f = bnd ? 1 : 0;
-    if (lane == 0 && !bnd) x = fmax(x, prev_fin);
-    for (int dlt = 1; dlt < WAVE; dlt <<= 1) {
-      double xo = __shfl_up(x, dlt);
-      int fo = __shfl_up(f, dlt);
-      if (lane >= dlt) {
-        if (!f) x = fmax(x, xo);
-        f = f | fo;
-      }
-    }
-    if (active) d_vals[k * STR] = x;
-    corr = cc;
-    prev_raw = __shfl(v, last);
-    prev_ts = __shfl(t, last);
-    prev_fin = __shfl(x, last);
-  }
-  return (int)n;
-}
-
 /* Pair-rounds removeCounterResets over an LDS/global COLUMN (stride STR
  * in int64 units): the rcr_scan_pairs logic with per-round pair loads
  * instead of register sources — 128 samples per round instead of 64, so
@@ -1369,37 +1253,6 @@ static __device__ void rcr_scan_col_pairs(int64_t* d_ts, double* d_vals,
       prev_ts = __shfl(ltr, lastl);
       prev_fin = __shfl(lxf, lastl);
     }
-  }
-}
-
-/* J scatter for the no-preprocessing staging path (same semantics as the
- * fused scatter in rcr_scan_regs). */
-static VM_DEV void scatter_j_regs(const int64_t* rt, int64_t n, int lane,
-                                  uint16_t* jb, int sdg, int64_t gstart,
-                                  int64_t gstep, double inv_gstep, int n_grid) {
-  int carry_b = 0;
-#pragma unroll
-  for (int c = 0; c < PIPE_CHUNKS; c++) {
-    int64_t base = (int64_t)c * WAVE;
-    if (base >= n) continue;
-    int64_t k = base + lane;
-    bool active = k < n;
-    int64_t t = active ? rt[c] : 0;
-    int lastl = (int)(n - base - 1);
-    if (lastl > 63) lastl = 63;
-    int b_own = vm_scatter_boundary(t, gstart, gstep, inv_gstep, sdg, n_grid);
-    int b_prev = __shfl_up(b_own, 1);
-    if (lane == 0) b_prev = carry_b;
-    if (active && k > 0) {
-      int glo = b_prev < -sdg ? -sdg : b_prev;
-      int ghi = b_own > n_grid ? n_grid : b_own;
-      for (int g2 = glo; g2 < ghi; g2++) jb[g2 + sdg] = (uint16_t)k;
-    }
-    if (active && k == n - 1) {
-      int glo = b_own < -sdg ? -sdg : b_own;
-      for (int g2 = glo; g2 < n_grid; g2++) jb[g2 + sdg] = (uint16_t)n;
-    }
-    carry_b = __shfl(b_own, lastl);
   }
 }
 
