@@ -47,7 +47,8 @@ def one_iter(seed):
     rng = np.random.default_rng(seed)
     func = FUNCS[int(rng.integers(0, len(FUNCS)))]
     n_series = int(rng.integers(1, 80))
-    max_len = int(rng.choice([40, 200, 520, 700]))
+    # lengths spanning wave (<=256/512), block (<=4064) AND huge (>4064)
+    max_len = int(rng.choice([40, 200, 520, 700, 700, 4500]))
     ts, vals, offsets = ragged_batch(
         n_series, max_len, START, seed=seed,
         stale_p=float(rng.choice([0.0, 0.02, 0.3])),
@@ -57,7 +58,8 @@ def one_iter(seed):
     start = START + int(rng.integers(-2, 60)) * step
     end = start + (n_grid - 1) * step
     window = int(rng.choice([0, step, 2 * step, 20 * step, 300_000,
-                             7 * step]))
+                             7 * step, 100 * step]))  # 100*step: dg >= 64,
+    # the pipe kernel's jbuf path instead of the fused shuffle-seek
     lookback = int(rng.choice([0, 0, 5 * 60 * 1000, 2 * step]))
     aggr = AGGRS[int(rng.integers(0, len(AGGRS)))]
     lo, hi = ARG_RANGES.get(func, (0.0, 0.0))
