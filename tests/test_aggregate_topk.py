@@ -233,3 +233,28 @@ def test_pointwise_topk_randomized_vs_numpy():
             np.testing.assert_array_equal(
                 out.view(np.int64), expect.view(np.int64),
                 err_msg=f"{name} trial {trial} k={k}")
+
+
+def test_bottomk_exec_pin():
+    """`bottomk(1, ...)` exec_test.go:7398: per-point bottom-1 across a
+    constant-10 series, time()/150 and a NaN-only comparison series —
+    NaN rows never win, and the per-point winner flips when time()/150
+    crosses 10."""
+    TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+    def S(name, tags, values):
+        return Series(MetricName(name.encode(), [(k.encode(), v.encode())
+                                                 for k, v in tags]), values)
+    s1 = S("", [("foo", "bar")], np.full(6, 10.0))
+    s2 = S("", [("baz", "sss")], TIME / 150)
+    s3 = S("", [("a", "b")], np.where(TIME < 100, TIME, NAN))
+    out = agg.aggregate("bottomk", [s1, s2, s3], arg=1)
+    got = {tuple(t.mn.tags): t.values for t in out
+           if not np.isnan(t.values).all()}
+    assert set(got) == {((b"foo", b"bar"),), ((b"baz", b"sss"),)}
+    v1 = got[((b"foo", b"bar"),)]
+    v2 = got[((b"baz", b"sss"),)]
+    np.testing.assert_array_equal(v1[3:], [10, 10, 10])
+    assert np.isnan(v1[:3]).all()
+    np.testing.assert_allclose(
+        v2[:3], [6.666666666666667, 8, 9.333333333333334], rtol=0, atol=0)
+    assert np.isnan(v2[3:]).all()

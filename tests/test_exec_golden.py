@@ -530,3 +530,48 @@ def test_range_trim_family_exec():
            [math.nan, 1200, 1400, 1600, 1800, math.nan])
     nan_eq(tfm.transform("range_trim_zscore", tmat(), scalar=0.9),
            [math.nan, 1200, 1400, 1600, 1800, math.nan])
+
+
+def _ser(name, tags, values):
+    from victoriametrics_amd.binary_op import Series
+    return Series(MetricName(name, [(k, v) for k, v in tags]),
+                  np.asarray(values, np.float64))
+
+
+def test_median_quantile_aggregates_exec():
+    # `median()` :7442, `median(3-timeseries)` :7457, `quantile(3)` :7472
+    # and `quantile(NaN)` :7486 — colagg device kernels + aggregate.py
+    from victoriametrics_amd import aggregate as agg
+    two = [_ser(b"", [(b"foo", b"bar")], np.full(6, 10.0)),
+           _ser(b"", [(b"baz", b"sss")], TIME / 150)]
+    out = agg.aggregate("median", [s.copy_shallow() for s in two])
+    assert len(out) == 1
+    _ulp(out[0].values,
+         [8.333333333333334, 9, 9.666666666666668, 10.333333333333332,
+          11, 11.666666666666668])
+    three = two + [_ser(b"", [], TIME / 200)]
+    out = agg.aggregate("median", [s.copy_shallow() for s in three])
+    _ulp(out[0].values,
+         [6.666666666666667, 8, 9.333333333333334, 10, 10, 10])
+    out = agg.aggregate("quantile", [s.copy_shallow() for s in two], arg=3.0)
+    assert np.isposinf(out[0].values).all()
+    out = agg.aggregate("quantile", [s.copy_shallow() for s in two],
+                        arg=math.nan)
+    from victoriametrics_amd.binary_op import remove_empty_series
+    assert remove_empty_series(out) == []
+
+
+def test_quantiles_aggregate_exec():
+    # `quantiles("phi", 0.2, 0.5, ...)` exec_test.go:7697
+    from victoriametrics_amd import aggregate as agg
+    two = [_ser(b"", [(b"foo", b"bar")], np.full(6, 10.0)),
+           _ser(b"", [(b"baz", b"sss")], TIME / 150)]
+    out = agg.quantiles("phi", [0.2, 0.5], two)
+    got = {s.mn.get_tag_value(b"phi"): s.values for s in out}
+    assert set(got) == {b"0.2", b"0.5"}
+    _ulp(got[b"0.2"],
+         [7.333333333333334, 8.4, 9.466666666666669, 10.133333333333333,
+          10.4, 10.666666666666668])
+    _ulp(got[b"0.5"],
+         [8.333333333333334, 9, 9.666666666666668, 10.333333333333332,
+          11, 11.666666666666668])
