@@ -575,3 +575,47 @@ def test_quantiles_aggregate_exec():
     _ulp(got[b"0.5"],
          [8.333333333333334, 9, 9.666666666666668, 10.333333333333332,
           11, 11.666666666666668])
+
+
+def test_histogram_share_and_fraction_valid_exec():
+    # `histogram_share(25, ...)` :4920 and `histogram_fraction(0, 25, ...)`
+    # :4950 over two le-groups -> 0.325 / 0.9166666666666666
+    from victoriametrics_amd.transform import histogram_transform
+    def buckets():
+        return [
+            _ser(b"", [(b"foo", b"bar"), (b"le", b"10")], np.full(6, 90.0)),
+            _ser(b"", [(b"foo", b"bar"), (b"le", b"30")], np.full(6, 100.0)),
+            _ser(b"", [(b"foo", b"bar"), (b"le", b"+Inf")], np.full(6, 300.0)),
+            _ser(b"", [(b"tag", b"xx"), (b"le", b"10")], np.full(6, 200.0)),
+            _ser(b"", [(b"tag", b"xx"), (b"le", b"30")], np.full(6, 300.0)),
+        ]
+    out = histogram_transform("histogram_share", buckets(),
+                              arg=np.full(6, 25.0))
+    got = {}
+    for s in out:
+        key = s.mn.get_tag_value(b"foo") or s.mn.get_tag_value(b"tag")
+        got[key] = s.values
+    _ulp(got[b"bar"], [0.325] * 6)
+    _ulp(got[b"xx"], [0.9166666666666666] * 6)
+    out = histogram_transform("histogram_fraction", buckets(),
+                              arg=(np.full(6, 0.0), np.full(6, 25.0)))
+    got = {}
+    for s in out:
+        key = s.mn.get_tag_value(b"foo") or s.mn.get_tag_value(b"tag")
+        got[key] = s.values
+    _ulp(got[b"bar"], [0.325] * 6)
+    _ulp(got[b"xx"], [0.9166666666666666] * 6)
+
+
+def test_histogram_quantile_negative_and_nan_bucket_exec():
+    # `histogram_quantile(negative-bucket-count)` :4981 -> 30 (fixBroken)
+    from victoriametrics_amd.transform import histogram_transform
+    series = [
+        _ser(b"", [(b"foo", b"bar"), (b"le", b"10")], np.full(6, 90.0)),
+        _ser(b"", [(b"foo", b"bar"), (b"le", b"30")], np.full(6, -100.0)),
+        _ser(b"", [(b"foo", b"bar"), (b"le", b"+Inf")], np.full(6, 300.0)),
+    ]
+    out = histogram_transform("histogram_quantile", series,
+                              arg=np.full(6, 0.6))
+    assert len(out) == 1
+    _exact(out[0].values, [30.0] * 6)
