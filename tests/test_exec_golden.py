@@ -589,8 +589,7 @@ def test_histogram_share_and_fraction_valid_exec():
             _ser(b"", [(b"tag", b"xx"), (b"le", b"10")], np.full(6, 200.0)),
             _ser(b"", [(b"tag", b"xx"), (b"le", b"30")], np.full(6, 300.0)),
         ]
-    out = histogram_transform("histogram_share", buckets(),
-                              arg=np.full(6, 25.0))
+    out = histogram_transform("histogram_share", buckets(), arg=25.0)
     got = {}
     for s in out:
         key = s.mn.get_tag_value(b"foo") or s.mn.get_tag_value(b"tag")
@@ -598,7 +597,7 @@ def test_histogram_share_and_fraction_valid_exec():
     _ulp(got[b"bar"], [0.325] * 6)
     _ulp(got[b"xx"], [0.9166666666666666] * 6)
     out = histogram_transform("histogram_fraction", buckets(),
-                              arg=(np.full(6, 0.0), np.full(6, 25.0)))
+                              arg=(0.0, 25.0))
     got = {}
     for s in out:
         key = s.mn.get_tag_value(b"foo") or s.mn.get_tag_value(b"tag")
@@ -615,7 +614,6 @@ def test_histogram_quantile_negative_and_nan_bucket_exec():
         _ser(b"", [(b"foo", b"bar"), (b"le", b"30")], np.full(6, -100.0)),
         _ser(b"", [(b"foo", b"bar"), (b"le", b"+Inf")], np.full(6, 300.0)),
     ]
-    out = histogram_transform("histogram_quantile", series,
-                              arg=np.full(6, 0.6))
+    out = histogram_transform("histogram_quantile", series, arg=0.6)
     assert len(out) == 1
     _exact(out[0].values, [30.0] * 6)
