@@ -168,3 +168,23 @@ def test_lag_subquery_nondivisor_step():
     # alignStartEnd + window walk at a misaligned cadence)
     got = _sq("lag", 60_000, 17_000, lambda t: t)
     _eq(got, [14, 10, 6, 2, 15, 11])
+
+
+def test_tlast_change_over_time_cases():
+    # exec_test.go:757-790: hit_last / hit_middle / miss over [1h] windows
+    got = _sq("tlast_change_over_time", 3_600_000, 0, lambda t: t)
+    _eq(got, [1000, 1200, 1400, 1600, 1800, 2000])
+    got = _sq("tlast_change_over_time", 3_600_000, 0,
+              lambda t: (t >= 1600).astype(np.float64))
+    _eq(got, [np.nan, np.nan, np.nan, 1600, 1600, 1600])
+    got = _sq("tlast_change_over_time", 3_600_000, 0,
+              lambda t: np.ones_like(t, dtype=np.float64))
+    assert np.isnan(got).all()  # constant: no change -> removeEmptySeries
+
+
+def test_timestamp_with_name_filtered():
+    # `timestamp_with_name(alias(time()>=1600,"foo"))` :791 — timestamp of
+    # each point's last sample; filtered points carry no sample
+    got = _sq("timestamp", 0, 0,
+              lambda t: np.where(t >= 1600, t, np.nan))
+    _eq(got, [np.nan, np.nan, np.nan, 1600, 1800, 2000])
