@@ -84,3 +84,24 @@ def test_max_response_series_guard():
                                        max_response_series=5)) == 5
     assert len(xs.timeseries_to_result(tss, may_sort=True,
                                        max_response_series=0)) == 5
+
+
+def test_plan_with_at_modifier():
+    """evalRollupFunc `@` handling (eval.go:903-950): single-point grid at
+    the first non-NaN `@` value (seconds -> ms), broadcast to the grid."""
+    import numpy as np
+    from victoriametrics_amd import engine
+    plan, ts, bc = engine.plan_with_at(
+        "rate", 1000_000, 2000_000, 200_000,
+        [float("nan"), 1600.0, 1700.0], window=300_000)
+    assert plan._c.start == plan._c.end == 1_600_000
+    np.testing.assert_array_equal(
+        ts, np.arange(1000_000, 2000_001, 200_000))
+    out = bc(np.asarray([[2.5], [3.5]]))
+    assert out.shape == (2, 6)
+    assert (out[0] == 2.5).all() and (out[1] == 3.5).all()
+    import pytest
+    with pytest.raises(engine.VmGpuError):
+        engine.plan_with_at("rate", 0, 100, 10, [[1.0], [2.0]])
+    with pytest.raises(engine.VmGpuError):
+        engine.plan_with_at("rate", 0, 100, 10, [float("nan")])

@@ -136,6 +136,33 @@ ROLLUP_KEEP_METRIC_NAME_FUNCS = {
 }
 
 
+def plan_with_at(func, start, end, step, at_values, **kwargs):
+    """evalRollupFunc's `@` modifier handling (eval.go:903-950): the `@`
+    expression must yield ONE series; its first non-NaN value (seconds) is
+    the evaluation timestamp; the rollup runs on the single-point grid
+    [at, at] and the one-point result broadcasts to the original grid.
+    Returns (plan, report_timestamps, broadcast) where broadcast(out)
+    expands the [rows x 1] output to [rows x n_grid]."""
+    at_values = np.atleast_2d(np.asarray(at_values, np.float64))
+    if at_values.shape[0] != 1:
+        raise VmGpuError(
+            "`@` modifier must return a single series; it returns "
+            f"{at_values.shape[0]} series instead")
+    finite = at_values[0][~np.isnan(at_values[0])]
+    if finite.size == 0:
+        raise VmGpuError("`@` modifier must return a non-NaN value")
+    at_ms = int(finite[0] * 1000)
+    plan = RollupPlan(func, at_ms, at_ms, int(step), **kwargs)
+    n_grid = 1 + (int(end) - int(start)) // int(step)
+    report_ts = np.asarray(start, np.int64) +         np.arange(n_grid, dtype=np.int64) * int(step)
+
+    def broadcast(out):
+        out = np.asarray(out, np.float64).reshape(-1, 1)
+        return np.repeat(out, n_grid, axis=1)
+
+    return plan, report_ts, broadcast
+
+
 def plan_with_offset(func, start, end, step, offset_ms=0, **kwargs):
     """evalRollupFuncWithoutAt's offset handling (eval.go:954-1008):
     `rf(m[w] offset o)` evaluates on the grid shifted BACK by the offset
