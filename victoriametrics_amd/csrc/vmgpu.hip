@@ -1995,7 +1995,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_huge_kernel(KPlan p, KIO
       if (wave == 0) {
         int c = load_compact_wave(io.ts + lo, io.vals + lo, n, dts, dvs,
                                   p.drop_stale != 0, lane);
-        if (p.rcr) rcr_scan_wave(dts, dvs, c, p.max_staleness, lane);
+        if (p.rcr) rcr_scan_col_pairs<1>(dts, dvs, c, p.max_staleness, lane);
         if (p.pre_func) pre_func_wave(dts, dvs, c, p.pre_func, lane);
         if (lane == 0) sh_count = c;
       }
