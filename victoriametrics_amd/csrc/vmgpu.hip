@@ -1975,6 +1975,7 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KI
 template <int FUNC_CT>
 __global__ __launch_bounds__(BLOCK_THREADS) void rollup_huge_kernel(KPlan p, KIO io) {
   __shared__ int sh_count;
+  __shared__ int sh_has_stale;
   __shared__ int64_t sh_si;
   __shared__ double sh_scratch[32];
   const int tid = threadIdx.x;
@@ -1992,9 +1993,27 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_huge_kernel(KPlan p, KIO
     const int64_t* uts;
     const double* uvs;
     if (p.drop_stale || p.rcr || p.pre_func) {
+      /* block-wide copy to the global scratch + stale detection in one
+       * pass (the old wave-0-only copy serialized ~n*16 B per series on
+       * one wave); stale compaction (rare) falls back to an IN-PLACE
+       * wave-0 forward compaction — within each tile the wave's loads all
+       * precede its stores and destinations never pass sources. */
+      if (tid == 0) sh_has_stale = 0;
+      __syncthreads();
+      int loc = 0;
+      for (int64_t k = tid; k < n; k += BLOCK_THREADS) {
+        int64_t t = io.ts[lo + k];
+        double v = io.vals[lo + k];
+        dts[k] = t;
+        dvs[k] = v;
+        if (p.drop_stale && vm_is_stale_nan(v)) loc = 1;
+      }
+      if (loc) atomicExch(&sh_has_stale, 1);
+      __syncthreads();
       if (wave == 0) {
-        int c = load_compact_wave(io.ts + lo, io.vals + lo, n, dts, dvs,
-                                  p.drop_stale != 0, lane);
+        int c = (int)n;
+        if (sh_has_stale)
+          c = load_compact_wave(dts, dvs, n, dts, dvs, true, lane);
         if (p.rcr) rcr_scan_col_pairs<1>(dts, dvs, c, p.max_staleness, lane);
         if (p.pre_func) pre_func_wave(dts, dvs, c, p.pre_func, lane);
         if (lane == 0) sh_count = c;
