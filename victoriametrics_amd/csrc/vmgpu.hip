@@ -1859,16 +1859,31 @@ __global__ __launch_bounds__(BLOCK_THREADS) void rollup_block_kernel(KPlan p, KI
     uint64_t lo = io.offsets[s];
     int64_t n = (int64_t)(io.offsets[s + 1] - lo);
 
-    if (!p.drop_stale) {
-      /* plain strided copy by the whole block */
+    {
+      /* block-wide copy + one-pass stale detection; the wave-0 ballot
+       * compaction (in place, forward) only runs when a stale NaN is
+       * actually present */
+      int* sh_stale = (int*)(sh_sum);  /* reuse the 8-byte scratch word */
+      if (tid == 0) *sh_stale = 0;
+      __syncthreads();
+      int loc = 0;
       for (int64_t k = tid; k < n; k += BLOCK_THREADS) {
-        lts[k] = io.ts[lo + k];
-        lvs[k] = io.vals[lo + k];
+        int64_t t = io.ts[lo + k];
+        double v = io.vals[lo + k];
+        lts[k] = t;
+        lvs[k] = v;
+        if (p.drop_stale && vm_is_stale_nan(v)) loc = 1;
       }
-      if (tid == 0) *sh_count = (int)n;
-    } else if (wave == 0) {
-      int c = load_compact_wave(io.ts + lo, io.vals + lo, n, lts, lvs, true, lane);
-      if (lane == 0) *sh_count = c;
+      if (loc) atomicExch(sh_stale, 1);
+      __syncthreads();
+      if (*sh_stale) {
+        if (wave == 0) {
+          int c = load_compact_wave(lts, lvs, n, lts, lvs, true, lane);
+          if (lane == 0) *sh_count = c;
+        }
+      } else if (tid == 0) {
+        *sh_count = (int)n;
+      }
     }
     __syncthreads();
     int count = *sh_count;
