@@ -617,3 +617,35 @@ def test_histogram_quantile_negative_and_nan_bucket_exec():
     out = histogram_transform("histogram_quantile", series, arg=0.6)
     assert len(out) == 1
     _exact(out[0].values, [30.0] * 6)
+
+
+def test_datetime_funcs_exec():
+    # exec_test.go:836-913 — the date/time family over time()-derived
+    # arguments (Unix seconds in UTC, Go time package semantics):
+    # expected arrays verbatim.
+    _exact(tfm.transform("minute", tmat()), [16, 20, 23, 26, 30, 33])
+    _exact(tfm.transform("day_of_month", tmat() * 1e4),
+           [26, 19, 12, 5, 28, 20])
+    _exact(tfm.transform("day_of_week", tmat() * 1e4), [0, 2, 5, 0, 2, 4])
+    _exact(tfm.transform("day_of_year", tmat() * 1e4),
+           [116, 139, 163, 186, 209, 232])
+    _exact(tfm.transform("days_in_month", tmat() * 2e4),
+           [31, 31, 30, 31, 28, 30])
+    _exact(tfm.transform("hour", tmat() * 1e4), [17, 21, 0, 4, 8, 11])
+    _exact(tfm.transform("month", tmat() * 1e4), [4, 5, 6, 7, 7, 8])
+    _exact(tfm.transform("year", tmat() * 1e5),
+           [1973, 1973, 1974, 1975, 1975, 1976])
+
+
+def test_minute_shifted_and_nan_exec():
+    # `minute(30*60+time())` :915 -> [46, 50, 53, 56, 0, 3];
+    # `minute(time() <= 1200 or time() > 1600)` :926 keeps NaN holes
+    _exact(tfm.transform("minute", 30.0 * 60.0 + tmat()),
+           [46, 50, 53, 56, 0, 3])
+    v = np.where((TIME <= 1200) | (TIME > 1600), TIME,
+                 np.nan).reshape(1, -1)
+    got = np.asarray(tfm.transform("minute", v)).ravel()
+    want = np.asarray([16, 20, np.nan, np.nan, 30, 33])
+    assert (np.isnan(got) == np.isnan(want)).all()
+    np.testing.assert_array_equal(got[~np.isnan(want)],
+                                  want[~np.isnan(want)])
