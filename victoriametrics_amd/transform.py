@@ -270,9 +270,7 @@ def label_replace(series_list, dst_label, replacement, src_label, regex):
         m = pat.match(src)
         if m is None:
             continue
-        val = m.expand(re.sub(r"\$(\d+|\{\w+\})",
-                              lambda g: "\\" + g.group(1).strip("{}"),
-                              replacement))
+        val = _go_expand(m, replacement)
         if dst_label == "__name__":
             s.mn.metric_group = val.encode()
         elif val == "":
@@ -683,12 +681,60 @@ def buckets_limit(limit, series_list):
     return rvs
 
 
+def _go_expand(m, template):
+    """Go regexp.Expand template semantics (regexp/regexp.go): `$name`
+    takes the LONGEST run of [A-Za-z0-9_] (so `$1y` means group "1y", not
+    group 1 followed by "y"), `${name}` is explicit, `$$` is a literal $,
+    and a reference to an absent or unmatched group expands to "" instead
+    of raising — Python's Match.expand errors on all of these."""
+    out = []
+    i, n = 0, len(template)
+    while i < n:
+        c = template[i]
+        if c != "$":
+            out.append(c)
+            i += 1
+            continue
+        if i + 1 >= n:
+            out.append("$")
+            break
+        nxt = template[i + 1]
+        if nxt == "$":
+            out.append("$")
+            i += 2
+            continue
+        if nxt == "{":
+            j = template.find("}", i + 2)
+            if j < 0:
+                out.append(template[i:])
+                break
+            name = template[i + 2:j]
+            i = j + 1
+        else:
+            j = i + 1
+            while j < n and (template[j].isalnum() or template[j] == "_"):
+                j += 1
+            name = template[i + 1:j]
+            if not name:
+                out.append("$")
+                i += 1
+                continue
+            i = j
+        val = ""
+        if name.isdigit():
+            gi = int(name)
+            if 0 <= gi <= m.re.groups:
+                val = m.group(gi) or ""
+        elif name in m.re.groupindex:
+            val = m.group(name) or ""
+        out.append(val)
+    return "".join(out)
+
+
 def _go_replace_all(regex, replacement, s):
-    # Go regexp.ReplaceAll semantics with $1/${name} expansion
+    # Go regexp.ReplaceAll semantics with Expand template handling
     pat = re.compile(regex)
-    repl = re.sub(r"\$(\d+|\{\w+\})",
-                  lambda g: "\\" + g.group(1).strip("{}"), replacement)
-    return pat.sub(repl, s)
+    return pat.sub(lambda m: _go_expand(m, replacement), s)
 
 
 def label_transform(series_list, label, regex, replacement):
