@@ -368,3 +368,99 @@ def test_format_go_float_pins():
     rng = np.random.default_rng(2)
     for v in rng.standard_normal(200) * 10.0 ** rng.integers(-15, 15, 200):
         assert float(f(float(v))) == float(v), v
+
+
+# ---------------------------------------------------------------------------
+# count_values exec pins (exec_test.go:9526-9737): expected MetricNames and
+# per-point count arrays verbatim.  These pin that the produced series carry
+# the GROUP metric name (aggrPrepareSeries rewrites members in place before
+# the afe runs), not the first member's original labels.
+# ---------------------------------------------------------------------------
+
+def _cv_series(name, tags, values):
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+    v = np.asarray(values, np.float64)
+    if v.size == 1:
+        v = np.full(6, float(v))
+    return Series(MetricName(name, tags), v.copy())
+
+
+def _cv_check(out, want):
+    """want: list of (sorted-tags-tuple, values)."""
+    got = {}
+    for s in out:
+        assert s.mn.metric_group == b""
+        got[tuple(sorted(s.mn.tags))] = s.values
+    assert set(got) == {k for k, _ in want}, (sorted(got), want)
+    for k, w in want:
+        g, w = got[k], np.asarray(w, np.float64)
+        assert (np.isnan(g) == np.isnan(w)).all(), (k, g, w)
+        np.testing.assert_array_equal(g[~np.isnan(w)], w[~np.isnan(w)])
+
+
+def test_count_values_exec():
+    # :9557 count_values("xxx", label_set(10,foo=bar) or
+    #                           label_set(time()/100, foo=bar, baz=xx))
+    from victoriametrics_amd import aggregate as agg
+    TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+    series = [
+        _cv_series("", [("foo", "bar")], 10.0),
+        _cv_series("", [("foo", "bar"), ("baz", "xx")], TIME / 100),
+    ]
+    out = agg.count_values("xxx", series)
+    _cv_check(out, [
+        (((b"xxx", b"10"),), [2, 1, 1, 1, 1, 1]),
+        (((b"xxx", b"12"),), [NAN, 1, NAN, NAN, NAN, NAN]),
+        (((b"xxx", b"14"),), [NAN, NAN, 1, NAN, NAN, NAN]),
+        (((b"xxx", b"16"),), [NAN, NAN, NAN, 1, NAN, NAN]),
+        (((b"xxx", b"18"),), [NAN, NAN, NAN, NAN, 1, NAN]),
+        (((b"xxx", b"20"),), [NAN, NAN, NAN, NAN, NAN, 1]),
+    ])
+
+
+def test_count_values_by_xxx_exec():
+    # :9629 — `by (xxx)` with dst label xxx removed from the grouping, so
+    # everything lands in ONE empty-name group; the input xxx=aaa tag must
+    # NOT leak into the outputs
+    from victoriametrics_amd import aggregate as agg
+    TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+    series = [
+        _cv_series("", [("foo", "bar"), ("xxx", "aaa")], 10.0),
+        _cv_series("", [("foo", "bar"), ("baz", "xx")],
+                   np.floor(TIME / 600)),
+    ]
+    out = agg.count_values("xxx", series, "by", ["xxx"])
+    _cv_check(out, [
+        (((b"xxx", b"1"),), [1, NAN, NAN, NAN, NAN, NAN]),
+        (((b"xxx", b"2"),), [NAN, 1, 1, 1, NAN, NAN]),
+        (((b"xxx", b"3"),), [NAN, NAN, NAN, NAN, 1, 1]),
+        (((b"xxx", b"10"),), [1, 1, 1, 1, 1, 1]),
+    ])
+
+
+def test_count_values_without_baz_exec():
+    # :9680 — `without (baz)` keeps foo=bar on the outputs
+    from victoriametrics_amd import aggregate as agg
+    TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+    series = [_cv_series("m", [("foo", "bar")], np.floor(TIME / 600))]
+    out = agg.count_values("xxx", series, "without", ["baz"])
+    _cv_check(out, [
+        (((b"foo", b"bar"), (b"xxx", b"1")), [1, NAN, NAN, NAN, NAN, NAN]),
+        (((b"foo", b"bar"), (b"xxx", b"2")), [NAN, 1, 1, 1, NAN, NAN]),
+        (((b"foo", b"bar"), (b"xxx", b"3")), [NAN, NAN, NAN, NAN, 1, 1]),
+    ])
+
+
+def test_count_values_big_numbers_exec():
+    # :9526 — FormatFloat(v, 'f', -1, 64): big integers stay positional
+    from victoriametrics_amd import aggregate as agg
+    series = [
+        _cv_series("first", [], 772424014.0),
+        _cv_series("second", [], 772424230.0),
+    ]
+    out = agg.count_values("xxx", series)
+    _cv_check(out, [
+        (((b"xxx", b"772424014"),), [1] * 6),
+        (((b"xxx", b"772424230"),), [1] * 6),
+    ])

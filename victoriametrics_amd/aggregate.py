@@ -433,7 +433,10 @@ def count_values(dst_label, series, modifier_op="", modifier_args=(),
                         raise ValueError(
                             f"more than {max_series_per_aggr} series "
                             "generated by count_values()")
-                    mn = members[0].mn.copy()
+                    # tss[0] inside the reference's afe carries the GROUP
+                    # name (aggrPrepareSeries rewrites member names in
+                    # place, aggr.go:121-146) — use gmn, not the member
+                    mn = gmn.copy()
                     mn.remove_tag(dst_label)
                     mn.add_tag(dst_label, format_go_float(v))
                     dst = Series(mn, np.full(len(s.values), math.nan))
@@ -520,7 +523,7 @@ def histogram_aggregate(series, modifier_op="", modifier_args=(), limit=0):
                     vmrange = ranges[key]
                 ts = m.get(vmrange)
                 if ts is None:
-                    mn = members[0].mn.copy()
+                    mn = gmn.copy()  # group name, as in aggr.go afe
                     mn.remove_tag("vmrange")
                     mn.add_tag("vmrange", vmrange)
                     ts = Series(mn, np.zeros(n_grid))
