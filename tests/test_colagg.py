@@ -291,7 +291,8 @@ def test_aggregate_any_limitk():
     from victoriametrics_amd import aggregate as agg
     out = agg.aggregate("any", _series_set(), "by", ["node"])
     assert len(out) == 2
-    assert all(s.mn.get_tag_value("pod") is None for s in out)
+    # keepOriginal=true (aggr.go:166): the winner keeps its FULL name
+    assert all(s.mn.get_tag_value("pod") is not None for s in out)
     out2 = agg.aggregate("limitk", _series_set(), "by", ["node"], arg=1)
     assert len(out2) == 2
     assert all(s.mn.get_tag_value("pod") is not None for s in out2)

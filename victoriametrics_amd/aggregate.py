@@ -109,22 +109,36 @@ def aggregate(name, series, modifier_op="", modifier_args=(), limit=0,
             rvs.extend(members)
         return rvs
     if name == "any":
-        # aggrFuncAny: one series per group (deterministic: first member)
-        groups = prepare_series(series, modifier_op, modifier_args, limit)
-        rvs = []
-        for gmn, members in groups:
-            dst = members[0]
-            dst.mn = gmn
-            rvs.append(dst)
-        return rvs
-    if name == "limitk":
-        # aggrFuncLimitK: first k member series per group, original names
-        groups = prepare_series(series, modifier_op, modifier_args, 0,
+        # aggrFuncAny (aggr.go:156): first member per group with its
+        # ORIGINAL metric name (keepOriginal=true in the reference)
+        groups = prepare_series(series, modifier_op, modifier_args,
+                                min(limit, 1) if limit else limit,
                                 keep_original=True)
-        k = int(arg)
+        return [members[0] for _, members in groups]
+    if name == "limitk":
+        # aggrFuncLimitK (aggr.go:1108): per group, members sorted by
+        # xxhash64 of the original metric name (uniform, call-stable
+        # selection), first k kept; k<0 -> 0, k=inf -> all
+        import xxhash
+        if arg is None or math.isnan(float(arg)) or float(arg) < 0:
+            k = 0
+        elif math.isinf(float(arg)):
+            k = None
+        else:
+            k = int(arg)
+        groups = prepare_series(series, modifier_op, modifier_args, limit,
+                                keep_original=True)
         rvs = []
         for _, members in groups:
-            rvs.extend(members[:k])
+            def _h(s):
+                d = xxhash.xxh64()
+                d.update(s.mn.metric_group)
+                for tk, tv in sorted(s.mn.tags):
+                    d.update(tk)
+                    d.update(tv)
+                return d.intdigest()
+            ordered = sorted(members, key=_h)
+            rvs.extend(ordered if k is None else ordered[:k])
         return rvs
     if name in ("outliers_iqr", "outliers_mad"):
         groups = prepare_series(series, modifier_op, modifier_args, limit,
