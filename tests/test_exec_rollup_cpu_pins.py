@@ -1,0 +1,66 @@
+"""Rollup exec pins transcribed from TestExecSuccess (exec_test.go:5865,
+:8434-8900) with deterministic inputs, run through the ORACLE rollup
+backend on CPU (test infra) via the shared subquery-grid helper — the GPU
+parity suite separately pins the device kernels against the same oracle,
+so these close the loop to the reference's own expected arrays."""
+import numpy as np
+
+from test_exec_ru_ttf_pins import _sq_cpu
+from victoriametrics_amd.decimal import go_round
+
+TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+
+
+def _eq(got, want):
+    np.testing.assert_allclose(got, want, rtol=1e-12, atol=0)
+
+
+def test_sum2_geomean_range_over_time_3i():
+    # `sum2_over_time(alias(time()/100, "foobar")[3i])` -> 200..980
+    got = _sq_cpu("sum2_over_time", lambda t: t / 100.0, window=600_000)
+    _eq(got, [200, 308, 440, 596, 776, 980])
+    # `round(geomean_over_time(alias(time()/100, "foobar")[3i]), 0.1)`
+    got = _sq_cpu("geomean_over_time", lambda t: t / 100.0, window=600_000)
+    _eq(go_round(got * 10.0) / 10.0, [7.8, 9.9, 11.9, 13.9, 15.9, 17.9])
+    # `range_over_time(alias(time()/100, "foobar")[3i])` -> max-min = 4
+    got = _sq_cpu("range_over_time", lambda t: t / 100.0, window=600_000)
+    _eq(got, [4, 4, 4, 4, 4, 4])
+
+
+def test_increase_delta_of_time():
+    # :8324/:8345 increase/increase_pure(time()) and :8806 delta(time())
+    # are the 200s/step slope
+    for fn in ("increase", "increase_pure", "delta"):
+        got = _sq_cpu(fn, lambda t: t)
+        _eq(got, [200] * 6)
+    # `increase(2000-time())` :8334 -> counter-reset clamping per window
+    got = _sq_cpu("increase", lambda t: 2000.0 - t)
+    _eq(got, [1000, 800, 600, 400, 200, 0])
+    # `delta(-time())` :8828 keeps the sign; `delta(1)` :8839 -> 0
+    _eq(_sq_cpu("delta", lambda t: -t), [-200] * 6)
+    _eq(_sq_cpu("delta", lambda t: np.ones_like(t)), [0] * 6)
+
+
+def test_prometheus_variants_window_strictness():
+    # :8356/:8850 `*_prometheus(time())` with the default window: the
+    # left-open (tEnd-window, tEnd] interval holds ONE sample, and the
+    # Prometheus variants use no lookbehind sample -> empty result
+    for fn in ("delta_prometheus", "increase_prometheus"):
+        assert np.isnan(_sq_cpu(fn, lambda t: t)).all()
+    # :8361/:8855 — at [201s] the window spans exactly two samples
+    for fn in ("delta_prometheus", "increase_prometheus"):
+        got = _sq_cpu(fn, lambda t: t, window=201_000)
+        _eq(got, [200] * 6)
+
+
+def test_zscore_over_time_const():
+    # `zscore_over_time(1[100s:10s])` :8920 -> all zero (no variance)
+    got = _sq_cpu("zscore_over_time", lambda t: np.ones_like(t),
+                  window=100_000, sq_step=10_000)
+    _eq(got, [0, 0, 0, 0, 0, 0])
+
+
+def test_median_over_time_scalar():
+    # `median_over_time(12)` :8872 -> 12 at every point
+    got = _sq_cpu("median_over_time", lambda t: np.full_like(t, 12.0))
+    _eq(got, [12] * 6)
