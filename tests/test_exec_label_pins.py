@@ -252,3 +252,41 @@ def test_label_value_exec():
         [-955, -1155, -1355, -1555, -1755, -1955])
     chk(out[1], "", [("foo", "123.456")],
         [np.nan, np.nan, np.nan, 1723.456, 1923.456, 2123.456])
+
+
+def _le(val, le, extra=()):
+    return S("metric", [("le", le)] + list(extra), float(val))
+
+
+def test_buckets_limit_exec():
+    # exec_test.go:5236-5470 — the four buckets_limit cases, expected
+    # surviving buckets verbatim (sorted ascending by values, like sort())
+    # trim_zero_preserve_empty_when_limit_not_reached: zero buckets trim
+    # first, then inner buckets with the least delta
+    out = tf.sort_series(tf.buckets_limit(3, [
+        _le(36, "+Inf"), _le(36, "25"), _le(36, "21"), _le(36, "19"),
+        _le(36, "18"), _le(36, "17"), _le(36, "16"), _le(27, "12"),
+        _le(14, "9"), _le(0, "6"), _le(0, "1")]))
+    assert [(s.mn.get_tag_value(b"le"), s.values[0]) for s in out] == [
+        (b"9", 14.0), (b"12", 27.0), (b"16", 36.0)]
+    # trim_zero: limit not reached after zero-trim -> keep left zeros
+    out = tf.sort_series(tf.buckets_limit(5, [
+        _le(36, "18"), _le(36, "17"), _le(36, "16"), _le(27, "12"),
+        _le(14, "9"), _le(0, "6"), _le(0, "1")]))
+    assert [(s.mn.get_tag_value(b"le"), s.values[0]) for s in out] == [
+        (b"1", 0.0), (b"6", 0.0), (b"9", 14.0), (b"12", 27.0),
+        (b"16", 36.0)]
+    # unused: under the limit, everything survives (tags preserved)
+    out = tf.sort_series(tf.buckets_limit(5, [
+        _le(100, "inf", [("x", "y")]), _le(50, "120", [("x", "y")])]))
+    assert len(out) == 2
+    chk(out[0], "metric", [("le", "120"), ("x", "y")], [50.0] * 6)
+    chk(out[1], "metric", [("le", "inf"), ("x", "y")], [100.0] * 6)
+    # used: limit 2 still keeps 3 (first/last protected + biggest deltas)
+    out = tf.sort_series(tf.buckets_limit(2, [
+        _le(100, "inf", [("x", "y")]), _le(98, "300", [("x", "y")]),
+        _le(52, "200", [("x", "y")]), _le(50, "120", [("x", "y")]),
+        _le(20, "70", [("x", "y")]), _le(10, "30", [("x", "y")]),
+        _le(9, "10", [("x", "y")])]))
+    assert [(s.mn.get_tag_value(b"le"), s.values[0]) for s in out] == [
+        (b"10", 9.0), (b"300", 98.0), (b"inf", 100.0)]
