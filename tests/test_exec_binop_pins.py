@@ -362,3 +362,187 @@ def test_vector_plus_vector_on_matching():
     assert len(out) == 1
     assert out[0].mn.tags == [(b"t2", b"v3")]
     eq(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+
+
+# ---------------------------------------------------------------------
+# group_left joins with tag copying and the fill() family
+# (exec_test.go:3908-4300)
+# ---------------------------------------------------------------------
+
+def _tags(s):
+    return sorted(s.mn.tags)
+
+
+def test_group_left_tag_copy_matching():
+    # `(... ) + on (foo, t2) group_left (t1, noxxx) (...)` :3908 — the
+    # join copies t1/noxxx from the one side; an absent join tag REMOVES
+    # the tag from the result
+    from victoriametrics_amd import transform as tfm
+    left = [S("", [("t1", "v123"), ("t2", "v3")], TIME),
+            S("", [("t2", "v3"), ("xxx", "yy")], np.full(6, 10.0))]
+    right = [S("", [("t1", "v1")], np.full(6, 100.0)),
+             S("", [("t2", "v3"), ("noxxx", "aa")], TIME)]
+    spec = BinOpSpec("+", group_op="on", group_tags=["foo", "t2"],
+                     join_op="group_left", join_tags=["t1", "noxxx"])
+    out = tfm.sort_series(_eval(spec, left, right), desc=True)
+    assert len(out) == 2
+    eq(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+    assert _tags(out[0]) == [(b"noxxx", b"aa"), (b"t2", b"v3")]
+    eq(out[1].values, [1010, 1210, 1410, 1610, 1810, 2010])
+    assert _tags(out[1]) == [(b"noxxx", b"aa"), (b"t2", b"v3"),
+                             (b"xxx", b"yy")]
+
+
+def test_group_left_star_and_prefix():
+    # :3952 group_left(*): copy ALL right tags except the on() set;
+    # :4000 adds `prefix "abc_"` to the copied names
+    from victoriametrics_amd import transform as tfm
+
+    def run(prefix):
+        left = [S("", [("t1", "v123"), ("t2", "v3")], TIME),
+                S("", [("t2", "v3"), ("xxx", "yy")], np.full(6, 10.0))]
+        right = [S("", [("t1", "v1")], np.full(6, 100.0)),
+                 S("", [("t2", "v3"), ("noxxx", "aa")], TIME)]
+        spec = BinOpSpec("+", group_op="on", group_tags=["foo", "t2"],
+                         join_op="group_left", join_tags=["*"],
+                         join_prefix=prefix)
+        return tfm.sort_series(_eval(spec, left, right), desc=True)
+
+    out = run("")
+    eq(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+    assert _tags(out[0]) == [(b"noxxx", b"aa"), (b"t1", b"v123"),
+                             (b"t2", b"v3")]
+    eq(out[1].values, [1010, 1210, 1410, 1610, 1810, 2010])
+    assert _tags(out[1]) == [(b"noxxx", b"aa"), (b"t2", b"v3"),
+                             (b"xxx", b"yy")]
+    out = run("abc_")
+    assert _tags(out[0]) == [(b"abc_noxxx", b"aa"), (b"t1", b"v123"),
+                             (b"t2", b"v3")]
+    assert _tags(out[1]) == [(b"abc_noxxx", b"aa"), (b"t2", b"v3"),
+                             (b"xxx", b"yy")]
+
+
+def test_group_left_copies_name():
+    # `... + on (t2, dfdf) group_left (__name__, xxx) ...` :4048 — the
+    # result takes the one side's __name__; absent xxx copies nothing
+    from victoriametrics_amd import transform as tfm
+    left = [S("vv3", [("t2", "v3"), ("x", "y")], TIME),
+            S("yy", [("t2", "v3")], np.full(6, 10.0))]
+    right = [S("", [("t1", "v1")], np.full(6, 100.0)),
+             S("abc", [("t2", "v3")], TIME)]
+    spec = BinOpSpec("+", group_op="on", group_tags=["t2", "dfdf"],
+                     join_op="group_left", join_tags=["__name__", "xxx"])
+    out = tfm.sort_series(_eval(spec, left, right), desc=True)
+    assert len(out) == 2
+    for s in out:
+        assert s.mn.metric_group == b"abc"
+    eq(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+    assert _tags(out[0]) == [(b"t2", b"v3"), (b"x", b"y")]
+    eq(out[1].values, [1010, 1210, 1410, 1610, 1810, 2010])
+    assert _tags(out[1]) == [(b"t2", b"v3")]
+
+
+def test_ignoring_matching():
+    # `... + ignoring (foo, t1, bar) ...` :4084 — one matching pair
+    left = [S("", [("t1", "v123"), ("t2", "v3")], TIME),
+            S("", [("t2", "v2")], np.full(6, 10.0))]
+    right = [S("", [("t1", "v1")], np.full(6, 100.0)),
+             S("", [("t2", "v3")], TIME)]
+    spec = BinOpSpec("+", group_op="ignoring",
+                     group_tags=["foo", "t1", "bar"])
+    out = _eval(spec, left, right)
+    assert len(out) == 1
+    eq(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+    assert _tags(out[0]) == [(b"t2", b"v3")]
+
+
+def test_ignoring_group_right_matching():
+    # `... + ignoring (foo, t2) group_right () ...` :4105 — one left
+    # joins two right series; results keep the RIGHT side's tags
+    from victoriametrics_amd import transform as tfm
+    left = [S("", [("t1", "v123"), ("t2", "v3")], TIME),
+            S("", [("t2", "v321"), ("t1", "v123"), ("t32", "v32")],
+              np.full(6, 10.0))]
+    right = [S("", [("t1", "v123")], np.full(6, 100.0)),
+             S("", [("t1", "v123"), ("t2", "v3")], TIME)]
+    spec = BinOpSpec("+", group_op="ignoring", group_tags=["foo", "t2"],
+                     join_op="group_right", join_tags=[])
+    out = tfm.sort_series(_eval(spec, left, right), desc=True)
+    assert len(out) == 2
+    eq(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+    assert _tags(out[0]) == [(b"t1", b"v123"), (b"t2", b"v3")]
+    eq(out[1].values, [1100, 1300, 1500, 1700, 1900, 2100])
+    assert _tags(out[1]) == [(b"t1", b"v123")]
+
+
+def _fill_inputs():
+    left = [S("", [("foo", "common")], np.full(6, 1.0)),
+            S("", [("foo", "left_only")], np.full(6, 2.0))]
+    right = [S("", [("foo", "common")], np.full(6, 3.0)),
+             S("", [("foo", "right_only")], np.full(6, 4.0))]
+    return left, right
+
+
+def _by_foo(out):
+    return {s.mn.get_tag_value(b"foo"): s for s in out}
+
+
+def test_fill_both_sides():
+    # `(...) + fill(0) (...)` :4139 — unmatched series on either side
+    # join against the fill constant
+    left, right = _fill_inputs()
+    out = _by_foo(_eval(BinOpSpec("+", fill_left=0.0, fill_right=0.0),
+                        left, right))
+    assert set(out) == {b"common", b"left_only", b"right_only"}
+    eq(out[b"common"].values, [4] * 6)
+    eq(out[b"left_only"].values, [2] * 6)
+    eq(out[b"right_only"].values, [4] * 6)
+
+
+def test_fill_per_point_nan():
+    # `(time()<=1200 ...) + fill(10) (time()>=1600 ...)` :4178 — the fill
+    # applies per POINT: a lone NaN side is filled, both-NaN stays NaN
+    left = [S("", [("foo", "common")],
+              np.where(TIME <= 1200, TIME, NAN))]
+    right = [S("", [("foo", "common")],
+               np.where(TIME >= 1600, TIME, NAN))]
+    out = _eval(BinOpSpec("+", fill_left=10.0, fill_right=10.0),
+                left, right)
+    assert len(out) == 1
+    eq(out[0].values, [1010, 1210, NAN, 1610, 1810, 2010])
+
+
+def test_fill_left_right_distinct():
+    # `... + fill_left(10) fill_right(20) ...` :4197 and
+    # `... + fill_right(20) ...` :4236 (right_only dropped without
+    # fill_left)
+    left, right = _fill_inputs()
+    out = _by_foo(_eval(BinOpSpec("+", fill_left=10.0, fill_right=20.0),
+                        left, right))
+    assert set(out) == {b"common", b"left_only", b"right_only"}
+    eq(out[b"common"].values, [4] * 6)
+    eq(out[b"left_only"].values, [22] * 6)
+    eq(out[b"right_only"].values, [14] * 6)
+    left, right = _fill_inputs()
+    out = _by_foo(_eval(BinOpSpec("+", fill_right=20.0), left, right))
+    assert set(out) == {b"common", b"left_only"}
+    eq(out[b"common"].values, [4] * 6)
+    eq(out[b"left_only"].values, [22] * 6)
+
+
+def test_fill_with_on_matching():
+    # `... + on(foo) fill(0) ...` :4266 — on() reduces the result name to
+    # the foo tag; fills still apply per key group
+    left = [S("", [("foo", "common"), ("extra", "l")], np.full(6, 1.0)),
+            S("", [("foo", "left_only"), ("extra", "l")], np.full(6, 2.0))]
+    right = [S("", [("foo", "common"), ("extra", "r")], np.full(6, 3.0)),
+             S("", [("foo", "right_only"), ("extra", "r")],
+               np.full(6, 4.0))]
+    out = _by_foo(_eval(BinOpSpec("+", group_op="on", group_tags=["foo"],
+                                  fill_left=0.0, fill_right=0.0),
+                        left, right))
+    assert set(out) == {b"common", b"left_only", b"right_only"}
+    eq(out[b"common"].values, [4] * 6)
+    assert _tags(out[b"common"]) == [(b"foo", b"common")]
+    eq(out[b"left_only"].values, [2] * 6)
+    eq(out[b"right_only"].values, [4] * 6)
