@@ -546,3 +546,95 @@ def test_fill_with_on_matching():
     assert _tags(out[b"common"]) == [(b"foo", b"common")]
     eq(out[b"left_only"].values, [2] * 6)
     eq(out[b"right_only"].values, [4] * 6)
+
+
+# ---------------------------------------------------------------------
+# if / ifnot set ops (exec_test.go:7539-7660) and group_left fill_right
+# (:4305)
+# ---------------------------------------------------------------------
+
+def test_vector_if_vector_partial():
+    # `(x=y, foo=bar:x) if (foo=bar filtered)` :7539 — only the series
+    # with a matching right key survives, masked by the right's NaNs;
+    # the x=y series has no match and is dropped entirely
+    left = [S("", [("x", "y")], TIME / 10),
+            S("x", [("foo", "bar")], TIME)]
+    right = [S("", [("foo", "bar")], np.where(TIME > 1400, TIME, NAN))]
+    out = _eval(BinOpSpec("if"), left, right)
+    out = remove_empty_series(out)
+    assert len(out) == 1
+    assert out[0].mn.metric_group == b"x"
+    eq(out[0].values, [NAN, NAN, NAN, 1600, 1800, 2000])
+
+
+def test_vector_if_vector_both_match():
+    # :7560 — both left series find their mask series
+    left = [S("", [("x", "y")], TIME / 10),
+            S("x", [("foo", "bar")], TIME)]
+    right = [S("", [("foo", "bar")], np.where(TIME > 1400, TIME, NAN)),
+             S("", [("x", "y")], np.where(TIME < 1400, TIME, NAN))]
+    out = remove_empty_series(_eval(BinOpSpec("if"), left, right))
+    got = by_tag(out, "x")
+    eq(got[b"y"].values, [100, 120, NAN, NAN, NAN, NAN])
+    got = by_tag(out, "foo")
+    eq(got[b"bar"].values, [NAN, NAN, NAN, 1600, 1800, 2000])
+
+
+def test_scalar_if_vector():
+    # :7591 `time() if (labeled vector)` -> empty (no key match);
+    # :7599 with a nameless-matching series -> masked time()
+    left = [t_series()]
+    right = [S("", [("foo", "bar")], np.full(6, 123.0))]
+    out = remove_empty_series(_eval(BinOpSpec("if"), left, right))
+    assert out == []
+    left = [t_series()]
+    right = [S("", [("foo", "bar")], np.full(6, 123.0)),
+             S("xxx", [], np.where(TIME > 1400, TIME, NAN))]
+    out = remove_empty_series(_eval(BinOpSpec("if"), left, right))
+    assert len(out) == 1
+    eq(out[0].values, [NAN, NAN, NAN, 1600, 1800, 2000])
+
+
+def test_if_ifnot_default_chain():
+    # `time() if time() > 1400 default -time()` :7613 -> [-1000..2000];
+    # `time() ifnot time() > 1400 default -time()` :7624;
+    # `time() ifnot time() > 1400` :7635
+    cond = [S("", [], np.where(TIME > 1400, TIME, NAN))]
+    masked = _eval(BinOpSpec("if"), [t_series()], cond)
+    out = _eval(BinOpSpec("default"), masked, [S("", [], -TIME)])
+    eq(out[0].values, [-1000, -1200, -1400, 1600, 1800, 2000])
+    cond = [S("", [], np.where(TIME > 1400, TIME, NAN))]
+    masked = _eval(BinOpSpec("ifnot"), [t_series()], cond)
+    eq(masked[0].values, [1000, 1200, 1400, NAN, NAN, NAN])
+    out = _eval(BinOpSpec("default"), masked, [S("", [], -TIME)])
+    eq(out[0].values, [1000, 1200, 1400, -1600, -1800, -2000])
+
+
+def test_ifnot_no_matching_series():
+    # :7646 — different keys: ifnot keeps the left side untouched
+    left = [S("", [("foo", "bar")], TIME)]
+    right = [S("", [("x", "y")], np.where(TIME > 1400, TIME, NAN))]
+    out = _eval(BinOpSpec("ifnot"), left, right)
+    assert len(out) == 1
+    eq(out[0].values, [1000, 1200, 1400, 1600, 1800, 2000])
+    assert out[0].mn.get_tag_value("foo") == b"bar"
+
+
+def test_group_left_fill_right():
+    # `(codes) + on(method) group_left() fill_right(0) (10 method=get)`
+    # :4305 — many-side codes join the one side; the put series joins the
+    # fill constant
+    left = [S("", [("method", "get"), ("code", "500")], np.full(6, 1.0)),
+            S("", [("method", "get"), ("code", "404")], np.full(6, 2.0)),
+            S("", [("method", "put"), ("code", "501")], np.full(6, 3.0))]
+    right = [S("", [("method", "get")], np.full(6, 10.0))]
+    spec = BinOpSpec("+", group_op="on", group_tags=["method"],
+                     join_op="group_left", join_tags=[], fill_right=0.0)
+    out = _eval(spec, left, right)
+    got = {s.mn.get_tag_value(b"code"): s for s in out}
+    assert set(got) == {b"500", b"404", b"501"}
+    eq(got[b"500"].values, [11] * 6)
+    assert got[b"500"].mn.get_tag_value(b"method") == b"get"
+    eq(got[b"404"].values, [12] * 6)
+    eq(got[b"501"].values, [3] * 6)
+    assert got[b"501"].mn.get_tag_value(b"method") == b"put"
