@@ -64,3 +64,60 @@ def test_median_over_time_scalar():
     # `median_over_time(12)` :8872 -> 12 at every point
     got = _sq_cpu("median_over_time", lambda t: np.full_like(t, 12.0))
     _eq(got, [12] * 6)
+
+
+def _fake(parent, fn, window=0, sq_step=0, tag=""):
+    """rollup_* pseudo-function expansion (rollup.go:436-516) over the
+    subquery grid, oracle backend; returns {rollup_tag: values}."""
+    from victoriametrics_amd import engine
+    plans = engine.rollup_fake_plans(parent, 1000_000, 2000_000, 200_000,
+                                     tag=tag, window=window,
+                                     keep_stale_nans=True)
+    return {t: _sq_cpu(None, fn, window=window, sq_step=sq_step, plan=p)
+            for t, p in plans}
+
+
+def test_rollup_exec():
+    # `sort(rollup(time()[:50s]))` exec_test.go — min/avg/max of the four
+    # 50s-substep samples per 200s window
+    out = _fake("rollup", lambda t: t, sq_step=50_000)
+    _eq(out["min"], [850, 1050, 1250, 1450, 1650, 1850])
+    _eq(out["avg"], [925, 1125, 1325, 1525, 1725, 1925])
+    _eq(out["max"], [1000, 1200, 1400, 1600, 1800, 2000])
+
+
+def test_rollup_increase_exec():
+    # `sort(rollup_increase(time()))` — all three tags are the 200 slope
+    out = _fake("rollup_increase", lambda t: t)
+    for tag in ("min", "max", "avg"):
+        _eq(out[tag], [200] * 6)
+
+
+def test_rollup_rate_exec():
+    # `rollup_rate((2200-time())[600s])` — per-window pairwise rates of
+    # the counter-reset-adjusted decreasing series
+    out = _fake("rollup_rate", lambda t: 2200.0 - t, window=600_000)
+    _eq(out["avg"], [6, 5, 4, 3, 2, 1])
+    _eq(out["max"], [7, 6, 5, 4, 3, 2])
+    _eq(out["min"], [5, 4, 3, 2, 1, 0])
+    # the optional second arg narrows to ONE labeled series
+    out = _fake("rollup_rate", lambda t: 2200.0 - t, window=600_000,
+                tag="max")
+    assert list(out) == ["max"]
+    _eq(out["max"], [7, 6, 5, 4, 3, 2])
+    out = _fake("rollup_rate", lambda t: 2200.0 - t, window=600_000,
+                tag="avg")
+    _eq(out["avg"], [6, 5, 4, 3, 2, 1])
+
+
+def test_rollup_deriv_and_scrape_interval_exec():
+    # `sort(rollup_deriv(time()[100s:50s]))` — time() slope is 1
+    out = _fake("rollup_deriv", lambda t: t, window=100_000,
+                sq_step=50_000)
+    for tag in ("min", "max", "avg"):
+        _eq(out[tag], [1] * 6)
+    # `rollup_scrape_interval(1[5m:10S])` — constant 10s sub-step
+    out = _fake("rollup_scrape_interval", lambda t: np.ones_like(t),
+                window=300_000, sq_step=10_000)
+    for tag in ("min", "max", "avg"):
+        _eq(out[tag], [10] * 6)

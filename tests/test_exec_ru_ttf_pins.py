@@ -35,7 +35,8 @@ def _tf(fid, v, arg):
     return out.ravel()
 
 
-def _sq_cpu(func, fn, window=0, sq_step=0, start=START_MS, end=END_MS):
+def _sq_cpu(func, fn, window=0, sq_step=0, start=START_MS, end=END_MS,
+            plan=None):
     """engine.rollup_subquery's grid composition (eval.go:1033-1100) with
     the oracle rollup as backend — the CPU mirror of the GPU `_sq` helper
     in test_exec_subquery_pins.py."""
@@ -51,8 +52,9 @@ def _sq_cpu(func, fn, window=0, sq_step=0, start=START_MS, end=END_MS):
     vals = np.ascontiguousarray(inner[keep])
     ts = np.ascontiguousarray(
         np.broadcast_to(sq_ts, inner.shape)[keep].astype(np.int64))
-    plan = RollupPlan(func, start, end, STEP_MS, window=window,
-                      keep_stale_nans=True)
+    if plan is None:
+        plan = RollupPlan(func, start, end, STEP_MS, window=window,
+                          keep_stale_nans=True)
     c = plan._c
     rc = oracle.RollupConfigC(
         func=c.func, may_adjust_window=c.may_adjust_window, start=c.start,
@@ -60,11 +62,13 @@ def _sq_cpu(func, fn, window=0, sq_step=0, start=START_MS, end=END_MS):
         lookback_delta=c.lookback_delta,
         min_staleness_interval=c.min_staleness_interval,
         is_default_rollup=c.is_default_rollup,
-        samples_scanned_per_call=c.samples_scanned_per_call, arg=c.arg)
+        samples_scanned_per_call=c.samples_scanned_per_call, arg=c.arg,
+        arg2=c.arg2)
     out, _, _ = oracle.rollup_eval_batch(
         rc, ts, vals, offsets,
         remove_counter_resets=bool(c.remove_counter_resets),
-        drop_stale_nans=False)
+        max_staleness_interval=c.max_staleness_interval,
+        drop_stale_nans=False, pre_func=c.pre_func)
     return out.ravel()
 
 
