@@ -878,3 +878,52 @@ def test_absent_over_time_multi_ts_combination():
     assert math.isnan(got[0]) and math.isnan(got[1])
     assert got[2] == 1.0 and got[3] == 1.0
     assert math.isnan(got[4]) and math.isnan(got[5])
+
+
+def test_prometheus_buckets_missing_vmrange_exec_pin():
+    # exec_test.go:5471: series with an existing `le` pass through
+    # untouched; malformed vmranges drop; valid vmranges convert with a
+    # zero lower-bound bucket inserted
+    t20, t100, t80, t40 = TIME6 / 20, TIME6 / 100, TIME6 / 80, TIME6 / 40
+    series = [
+        S("xyz", [("foo", "bar"), ("le", "0.2")], t20),
+        S("xxx", [("foo", "bar"), ("vmrange", "foobar")], t100),
+        S("xxx", [("foo", "bar"), ("vmrange", "30...foobar")], t100),
+        S("xxx", [("foo", "bar"), ("vmrange", "30...40")], t100),
+        S("yyy", [("foo", "bar"), ("vmrange", "0...900"), ("le", "54")],
+          t80),
+        S("yyy", [("foo", "bar"), ("vmrange", "900...+Inf"),
+                  ("le", "2343")], t40),
+    ]
+    out = tf.prometheus_buckets(series)
+    got = {(s.mn.metric_group.decode(),
+            s.mn.get_tag_value("le").decode()): list(s.values)
+           for s in out}
+    assert got[("xxx", "30")] == [0.0] * 6
+    assert got[("xxx", "40")] == [10, 12, 14, 16, 18, 20]
+    assert got[("xxx", "+Inf")] == [10, 12, 14, 16, 18, 20]
+    assert got[("yyy", "900")] == [12.5, 15, 17.5, 20, 22.5, 25]
+    assert got[("yyy", "+Inf")] == [37.5, 45, 52.5, 60, 67.5, 75]
+    assert got[("xyz", "0.2")] == [50, 60, 70, 80, 90, 100]
+    assert len(got) == 6
+
+
+def test_prometheus_buckets_zero_and_valid_exec_pins():
+    # exec_test.go:5580: an all-zero 0...0 bucket alone yields nothing
+    out = tf.prometheus_buckets([S("", [("vmrange", "0...0")],
+                                   np.zeros(6))])
+    assert out == []
+    # exec_test.go:5586 `prometheus_buckets(valid)`
+    series = [
+        S("xxx", [("foo", "bar"), ("vmrange", "0...0")], np.full(6, 90.0)),
+        S("xxx", [("foo", "bar"), ("vmrange", "0...0.2")], TIME6 / 20),
+        S("xxx", [("foo", "bar"), ("vmrange", "0.2...40")], TIME6 / 100),
+        S("xxx", [("foo", "bar"), ("vmrange", "40...Inf")], TIME6 / 10),
+    ]
+    out = tf.prometheus_buckets(series)
+    got = {s.mn.get_tag_value("le").decode(): list(s.values) for s in out}
+    assert got["0"] == [90.0] * 6
+    assert got["0.2"] == [140, 150, 160, 170, 180, 190]
+    assert got["40"] == [150, 162, 174, 186, 198, 210]
+    assert got["Inf"] == [250, 282, 314, 346, 378, 410]
+    assert len(got) == 4
