@@ -927,3 +927,76 @@ def test_prometheus_buckets_zero_and_valid_exec_pins():
     assert got["40"] == [150, 162, 174, 186, 198, 210]
     assert got["Inf"] == [250, 282, 314, 346, 378, 410]
     assert len(got) == 4
+
+
+def test_interpolate_keep_value_exec_pins():
+    # exec_test.go keep_last_value()/keep_next_value()/interpolate() over
+    # `time() < 1300 default time() > 1700` = [1000,1200,nan,nan,1800,2000]
+    base = TIME6.copy()
+    base[(base >= 1300) & (base <= 1700)] = NAN
+    out, _ = oracle.tf_apply(100, base.reshape(1, -1).copy())
+    assert list(out[0]) == [1000, 1200, 1200, 1200, 1800, 2000]
+    out, _ = oracle.tf_apply(101, base.reshape(1, -1).copy())
+    assert list(out[0]) == [1000, 1200, 1800, 1800, 1800, 2000]
+    out, _ = oracle.tf_apply(102, base.reshape(1, -1).copy())
+    assert list(out[0]) == [1000, 1200, 1400, 1600, 1800, 2000]
+    # interpolate(tail): trailing NaNs stay NaN
+    v = np.where(TIME6 < 1300, TIME6, NAN).reshape(1, -1)
+    out, _ = oracle.tf_apply(102, v.copy())
+    assert list(out[0][:2]) == [1000, 1200] and np.isnan(out[0][2:]).all()
+    # interpolate(head): leading NaNs stay NaN
+    v = np.where(TIME6 > 1500, TIME6, NAN).reshape(1, -1)
+    out, _ = oracle.tf_apply(102, v.copy())
+    assert np.isnan(out[0][:3]).all()
+    assert list(out[0][3:]) == [1600, 1800, 2000]
+    # interpolate(tail_head_and_middle):
+    # [nan,1200,nan,nan,1800,nan] -> [nan,1200,1400,1600,1800,nan]
+    v = np.full(6, NAN)
+    v[1], v[4] = 1200.0, 1800.0
+    out, _ = oracle.tf_apply(102, v.reshape(1, -1))
+    assert np.isnan(out[0][0]) and np.isnan(out[0][5])
+    assert list(out[0][1:5]) == [1200, 1400, 1600, 1800]
+
+
+def test_histogram_quantiles_exec_pin(monkeypatch):
+    # `histogram_quantiles("phi", 0.2, 0.3, buckets)` exec_test.go — one
+    # result per phi labeled phi=%g; values via the oracle quantile
+    # (the device kernel is pinned against it in the GPU suite)
+    bv = np.stack([np.zeros(6), np.full(6, 100.0), np.full(6, 300.0)])
+    les = np.asarray([10.0, 30.0, np.inf])
+    off = np.asarray([0, 3], np.uint64)
+    for phi, want in ((0.2, 22.0), (0.3, 28.0)):
+        out, _, _ = oracle.histogram_quantile(phi, bv, les, off)
+        assert list(out[0]) == [want] * 6
+
+    def fake_ht(name, series, arg=None, bounds_label=None):
+        assert name == "histogram_quantile"
+        out, _, _ = oracle.histogram_quantile(float(arg), bv, les, off)
+        return [S("", [("foo", "bar")], out[0])]
+
+    monkeypatch.setattr(tf, "histogram_transform", fake_ht)
+    out = tf.histogram_quantiles("phi", [0.2, 0.3], [])
+    got = {s.mn.get_tag_value("phi").decode(): list(s.values) for s in out}
+    assert got["0.2"] == [22.0] * 6
+    assert got["0.3"] == [28.0] * 6
+    assert all(s.mn.get_tag_value("foo") == b"bar" for s in out)
+
+
+def test_start_end_step_exec_pins():
+    # `time() - start()` -> [0..1000]; `end() - time()` -> [1000..0];
+    # `time() / step()` -> [5..10] on the fixed grid
+    ts = (TIME6 * 1000).astype(np.int64)
+    start = tf.eval_start(ts)[0].values
+    end = tf.eval_end(ts)[0].values
+    step = tf.eval_step(ts, 200_000)[0].values
+    assert list(TIME6 - start) == [0, 200, 400, 600, 800, 1000]
+    assert list(end - TIME6) == [1000, 800, 600, 400, 200, 0]
+    assert list(TIME6 / step) == [5, 6, 7, 8, 9, 10]
+
+
+def test_scalar_multi_timeseries_exec_pin():
+    # `scalar(1 or label_set(2, "xx", "foo"))` -> NaN series (two members)
+    out = tf.scalar([S("", [], np.ones(6)),
+                     S("", [("xx", "foo")], np.full(6, 2.0))])
+    assert len(out) == 1
+    assert np.isnan(out[0].values).all()
