@@ -465,3 +465,30 @@ def test_count_values_big_numbers_exec():
         (((b"xxx", b"772424014"),), [1] * 6),
         (((b"xxx", b"772424230"),), [1] * 6),
     ])
+
+
+def test_outliers_exec_pins():
+    # exec_test.go outliers_iqr() / outliers_mad(1) / outliers_mad(5):
+    # expected survivor sets verified through the oracle bounds+filter
+    # kernels (the host selection wiring is covered in the GPU suite)
+    TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+    v = np.stack([TIME, TIME * 1.5, TIME * 10, TIME * 1.2, TIME * 0.1])
+    gof = np.zeros(5, np.int32)
+    go = np.asarray([0, 5], np.uint64)
+    gr = np.arange(5, dtype=np.uint32)
+    lo, hi = oracle.colagg("iqr_bounds", v, gr, go)
+    flags = oracle.colagg_filter("iqr", v, gof, lo, hi)
+    # m5 (0.1x) and m3 (10x) are the outliers, in sort() order
+    assert list(flags) == [0, 0, 1, 0, 1]
+    # outliers_mad(1, (t, 1.5t, 0.9t)): only the 1.5x series deviates
+    # beyond 1 MAD; at k=5 nothing does
+    v = np.stack([TIME, TIME * 1.5, TIME * 0.9])
+    gof3 = np.zeros(3, np.int32)
+    go3 = np.asarray([0, 3], np.uint64)
+    gr3 = np.arange(3, dtype=np.uint32)
+    med = oracle.colagg("median", v, gr3, go3)
+    mad = oracle.colagg("mad", v, gr3, go3)
+    flags = oracle.colagg_filter("mad", v, gof3, med, mad * 1.0)
+    assert list(flags) == [0, 1, 0]
+    flags = oracle.colagg_filter("mad", v, gof3, med, mad * 5.0)
+    assert list(flags) == [0, 0, 0]

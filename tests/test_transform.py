@@ -1000,3 +1000,61 @@ def test_scalar_multi_timeseries_exec_pin():
                      S("", [("xx", "foo")], np.full(6, 2.0))])
     assert len(out) == 1
     assert np.isnan(out[0].values).all()
+
+
+def test_smooth_exponential_exec_pins():
+    # exec_test.go: sf=1 -> identity; sf=0 -> frozen at the first value
+    # (clamped to [0,1] semantics); sf=0.5 -> EMA
+    for sf, want in ((1.0, [1000, 1200, 1400, 1600, 1800, 2000]),
+                     (0.0, [1000.0] * 6),
+                     (0.5, [1000, 1100, 1250, 1425, 1612.5, 1806.25])):
+        out, _ = oracle.tf_apply(123, TIME6.reshape(1, -1).copy(),
+                                 arg1=np.full(6, sf))
+        assert list(out[0]) == want, sf
+
+
+def test_range_normalize_exec_pins():
+    # `range_normalize(time(), -time())` -> [0..1] and [1..0]
+    out, _ = oracle.tf_apply(113, TIME6.reshape(1, -1).copy())
+    assert list(out[0]) == [0, 0.2, 0.4, 0.6, 0.8, 1.0]
+    out, _ = oracle.tf_apply(113, (-TIME6).reshape(1, -1).copy())
+    assert list(out[0]) == [1.0, 0.8, 0.6, 0.4, 0.2, 0]
+    # filtered variants keep NaN holes and normalize over the present
+    # points only: time() > 1200 < 1800 -> [nan,nan,0,1,nan,nan]
+    v = np.where((TIME6 > 1200) & (TIME6 < 1800), TIME6, NAN)
+    out, _ = oracle.tf_apply(113, v.reshape(1, -1).copy())
+    g = out[0]
+    assert np.isnan(g[[0, 1, 4, 5]]).all()
+    assert list(g[2:4]) == [0.0, 1.0]
+    # -(time() > 1200 < 2000) -> [nan,nan,1,0.5,0,nan]
+    v = np.where((TIME6 > 1200) & (TIME6 < 2000), -TIME6, NAN)
+    out, _ = oracle.tf_apply(113, v.reshape(1, -1).copy())
+    g = out[0]
+    assert np.isnan(g[[0, 1, 5]]).all()
+    assert list(g[2:5]) == [1.0, 0.5, 0.0]
+
+
+def test_range_mad_exec_pins():
+    # `range_mad(time())` -> 300; filtered -> 100
+    out, _ = oracle.tf_apply(119, TIME6.reshape(1, -1).copy())
+    assert list(out[0]) == [300.0] * 6
+    v = np.where((TIME6 > 1200) & (TIME6 < 1800), TIME6, NAN)
+    out, _ = oracle.tf_apply(119, v.reshape(1, -1).copy())
+    # the constant MAD is written to EVERY grid point, NaN holes included
+    # (the reference's expected array is all-100)
+    assert list(out[0]) == [100.0] * 6
+
+
+def test_range_trim_filtered_exec_pins():
+    # `range_trim_outliers(0.5, time() > 1200)` -> [nan,nan,nan,1600,1800,nan]
+    v = np.where(TIME6 > 1200, TIME6, NAN)
+    out, _ = oracle.tf_apply(120, v.reshape(1, -1).copy(), scalar=0.5)
+    g = out[0]
+    assert np.isnan(g[[0, 1, 2, 5]]).all()
+    assert list(g[3:5]) == [1600.0, 1800.0]
+    # `range_trim_spikes(0.2, time() > 1200 <= 1800)` -> only 1600 left
+    v = np.where((TIME6 > 1200) & (TIME6 <= 1800), TIME6, NAN)
+    out, _ = oracle.tf_apply(121, v.reshape(1, -1).copy(), scalar=0.2)
+    g = out[0]
+    assert list(g[3:4]) == [1600.0]
+    assert np.isnan(np.delete(g, 3)).all()
