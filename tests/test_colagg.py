@@ -492,3 +492,28 @@ def test_outliers_exec_pins():
     assert list(flags) == [0, 1, 0]
     flags = oracle.colagg_filter("mad", v, gof3, med, mad * 5.0)
     assert list(flags) == [0, 0, 0]
+
+
+def test_histogram_aggregate_exec_pins():
+    # exec_test.go `histogram(scalar)` / `histogram(vector)` — the
+    # VictoriaMetrics-histogram vmrange boundaries and the cumulative le
+    # conversion, before the exec queries' `+` merge (pinned separately
+    # in the binop suite)
+    from victoriametrics_amd import aggregate as agg
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+
+    def const(v, name="", tags=()):
+        return Series(MetricName(name, list(tags)), np.full(6, float(v)))
+
+    out = agg.histogram_aggregate([const(123.0)])
+    got = {s.mn.get_tag_value("le").decode(): s.values[0] for s in out}
+    assert got == {"1.136e+02": 0.0, "1.292e+02": 1.0, "+Inf": 1.0}
+    out = agg.histogram_aggregate([
+        const(1.0, tags=[("foo", "bar")]),
+        const(1.1, tags=[("xx", "yy")]),
+        const(1.15, "foobar"),
+    ])
+    got = {s.mn.get_tag_value("le").decode(): s.values[0] for s in out}
+    assert got == {"8.799e-01": 0.0, "1.000e+00": 1.0, "1.136e+00": 2.0,
+                   "1.292e+00": 3.0, "+Inf": 3.0}
