@@ -1058,3 +1058,29 @@ def test_range_trim_filtered_exec_pins():
     g = out[0]
     assert list(g[3:4]) == [1600.0]
     assert np.isnan(np.delete(g, 3)).all()
+
+
+def test_remove_resets_exec_pins():
+    # `remove_resets(abs(1500-time()))` -> [500, 800, 900, 900, 1100, 1300]
+    v = np.abs(1500.0 - TIME6).reshape(1, -1)
+    out, _ = oracle.tf_apply(124, v.copy())
+    assert list(out[0]) == [500, 800, 900, 900, 1100, 1300]
+    # `remove_resets(sum(time(), time()/5 < 300))`: the summed series
+    # [1200,1440,1680,1600,1800,2000] has one reset at idx 3
+    partial = np.where(TIME6 / 5 < 300, TIME6 / 5, NAN)
+    summed = TIME6 + np.nan_to_num(partial)
+    out, _ = oracle.tf_apply(124, summed.reshape(1, -1).copy())
+    assert list(out[0]) == [1200, 1440, 1680, 1680, 1880, 2080]
+
+
+def test_running_avg_exec_pins():
+    # `running_avg(time())` -> prefix means [1000..1500]
+    out, _ = oracle.tf_apply(106, TIME6.reshape(1, -1).copy())
+    assert list(out[0]) == [1000, 1100, 1200, 1300, 1400, 1500]
+    # `running_avg(time() > 1200 < 1800)` -> NaNs skipped, mean over the
+    # seen present points only
+    v = np.where((TIME6 > 1200) & (TIME6 < 1800), TIME6, NAN)
+    out, _ = oracle.tf_apply(106, v.reshape(1, -1).copy())
+    g = out[0]
+    assert np.isnan(g[:2]).all()
+    assert list(g[2:]) == [1400, 1500, 1500, 1500]
