@@ -517,3 +517,39 @@ def test_histogram_aggregate_exec_pins():
     got = {s.mn.get_tag_value("le").decode(): s.values[0] for s in out}
     assert got == {"8.799e-01": 0.0, "1.000e+00": 1.0, "1.136e+00": 2.0,
                    "1.292e+00": 3.0, "+Inf": 3.0}
+
+
+def test_quantile_out_of_range_and_mad_exec_pins():
+    # exec_test.go quantile(-2) -> -Inf, quantile(3) -> +Inf,
+    # quantile(NaN) -> NaN; mad(3 series) -> [100..200]
+    TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+    v = np.stack([np.full(6, 10.0), TIME / 150])
+    gr = np.arange(2, dtype=np.uint32)
+    go = np.asarray([0, 2], np.uint64)
+    out = oracle.colagg("quantile", v, gr, go, phi=-2.0)
+    assert np.isneginf(out[0]).all()
+    out = oracle.colagg("quantile", v, gr, go, phi=3.0)
+    assert np.isposinf(out[0]).all()
+    out = oracle.colagg("quantile", v, gr, go, phi=NAN)
+    assert np.isnan(out[0]).all()
+    v = np.stack([TIME, TIME * 1.5, TIME * 0.9])
+    gr = np.arange(3, dtype=np.uint32)
+    go = np.asarray([0, 3], np.uint64)
+    out = oracle.colagg("mad", v, gr, go)
+    assert list(out[0]) == [100, 120, 140, 160, 180, 200]
+
+
+def test_distinct_exec_pin():
+    # `distinct(union(1+time() > 1100, (time() > 1700 tagged)))` ->
+    # [nan, 1, 1, 1, 2, 2]
+    TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+    # `1+time() > 1100` filters on the SUM: 1001 <= 1100 -> NaN at idx 0
+    a = np.where(1 + TIME > 1100, 1 + TIME, NAN)
+    b = np.where(TIME > 1700, TIME, NAN)
+    v = np.stack([a, b])
+    gr = np.arange(2, dtype=np.uint32)
+    go = np.asarray([0, 2], np.uint64)
+    out = oracle.colagg("distinct", v, gr, go)
+    g = out[0]
+    assert np.isnan(g[0])
+    assert list(g[1:]) == [1, 1, 1, 2, 2]
