@@ -289,3 +289,32 @@ def test_randomized_composition_soak():
             for k in want:
                 assert got[k] == pytest.approx(want[k], rel=1e-9), \
                     (trial, q, func, k)
+
+
+def test_get_sum_instant_values_reference_vectors():
+    # TestGetSumInstantValues (eval_test.go:100): the six cached/start/end
+    # combinations, expected sums verbatim
+    from victoriametrics_amd.instant import get_sum_instant_values
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+
+    def ts(name, value):
+        return Series(MetricName(name), np.asarray([value], np.float64))
+
+    def run(cached, start, end):
+        out = get_sum_instant_values(cached, start, end)
+        return {s.mn.metric_group: s.values[0] for s in out}
+
+    # start only -> adopted
+    assert run([], [ts("foo", 1)], []) == {b"foo": 1.0}
+    # start - end (no cache) -> 0
+    assert run([], [ts("foo", 1)], [ts("foo", 1)]) == {b"foo": 0.0}
+    # cached + start -> 2
+    assert run([ts("foo", 1)], [ts("foo", 1)], []) == {b"foo": 2.0}
+    # cached + start - end -> 1
+    assert run([ts("foo", 1)], [ts("foo", 1)], [ts("foo", 1)]) == \
+        {b"foo": 1.0}
+    # cached - end -> 0
+    assert run([ts("foo", 1)], [], [ts("foo", 1)]) == {b"foo": 0.0}
+    # cached only -> unchanged
+    assert run([ts("foo", 1)], [], []) == {b"foo": 1.0}
