@@ -1084,3 +1084,49 @@ def test_running_avg_exec_pins():
     g = out[0]
     assert np.isnan(g[:2]).all()
     assert list(g[2:]) == [1400, 1500, 1500, 1500]
+
+
+def test_get_num_prefix_reference_vectors():
+    # TestGetNumPrefix (transform_test.go:283): every vector verbatim;
+    # non-empty prefixes must parse as floats
+    cases = [
+        ("", ""), ("foo", ""), ("-", ""), (".", ""), ("-.", ""),
+        ("+..", ""), ("1", "1"), ("12", "12"), ("1foo", "1"),
+        ("-123", "-123"), ("-123bar", "-123"), ("+123", "+123"),
+        ("+123.", "+123."), ("+123..", "+123."), ("+123.-", "+123."),
+        ("12.34..", "12.34"), ("-12.34..", "-12.34"), ("-12.-34..", "-12."),
+    ]
+    for s_in, want in cases:
+        got = tf._num_prefix(s_in)
+        assert got == want, (s_in, got, want)
+        if got:
+            float(got)  # must parse
+
+
+def test_numeric_less_reference_vectors():
+    # TestNumericLess (transform_test.go:317): every vector verbatim,
+    # including the 309-digit no-panic regression (GHSA-9g98-8jgr-x2vv)
+    big = "9" * 309
+    cases = [
+        ("", "", False), ("", "321", True), ("321", "", False),
+        ("", "abc", True), ("abc", "", False), ("foo", "123", False),
+        ("123", "foo", True), ("123", "321", True), ("321", "123", False),
+        ("123", "123", False), ("a", "b", True), ("b", "a", False),
+        ("a", "a", False), ("foo123", "foo", False), ("foo", "foo123", True),
+        ("foo", "foo", False), ("123foo", "123bar", False),
+        ("123bar", "123foo", True), ("123bar", "123bar", False),
+        ("1:0:0", "1:0:2", True), ("1:0:15", "1:0:2", False),
+        ("0", "00", False), ("aa", "ab", True), ("ab", "abc", True),
+        ("a0001", "a0000001", False), ("a10", "abcdefgh2", True),
+        ("a1b", "a01b", False), ("a001b01", "a01b001", False),
+        ("a01b001", "a001b01", False), ("a1", "a1x", True),
+        ("1b", "1ax", False), ("082", "83", True), ("083a", "9a", False),
+        ("083a", "94a", True), ("-123", "123", True),
+        ("-123", "+123", True), ("-123", "-123", False),
+        ("123", "-123", False), ("12.9", "12.56", False),
+        ("12.56", "12.9", True), ("12.9", "12.9", False),
+        (big, "1", False), ("1", big, True), (big, big, False),
+        ("-" + big, big, True), (big, "-" + big, False),
+    ]
+    for a, b, want in cases:
+        assert tf.numeric_less(a, b) == want, (a, b, want)
