@@ -189,10 +189,54 @@ static void scans_and_rollups(void) {
   free(ts);
 }
 
+/* Decode mutated/truncated/garbage byte streams: whatever the return
+ * code, the decoder must never read or write out of bounds (ASan checks
+ * that), and a success must still produce exactly `items` outputs. */
+static void malformed_decode_fuzz(void) {
+  enum { N = 257 };
+  int64_t a[N];
+  int64_t v = -37;
+  for (int64_t i = 0; i < N; i++) {
+    v += (int64_t)(rnd() % 5000) - 2400;
+    a[i] = v;
+  }
+  uint8_t base[N * 10 + 64];
+  uint8_t mt = 0;
+  int64_t first = 0;
+  int64_t len = vm_marshal_int64_array(base, a, N, 64, &mt, &first);
+  check(len > 0, "fuzz marshal");
+  int64_t dst[N];
+  for (int iter = 0; iter < 4000; iter++) {
+    /* heap copy sized EXACTLY to the (possibly truncated) length so ASan
+     * sees one-past-the-end reads */
+    size_t cut = (size_t)(rnd() % (uint64_t)(len + 1));
+    uint8_t* buf = malloc(cut ? cut : 1);
+    memcpy(buf, base, cut);
+    int nmut = (int)(rnd() % 4);
+    for (int m = 0; m < nmut && cut; m++)
+      buf[rnd() % cut] = (uint8_t)rnd();
+    uint8_t use_mt = (iter % 3 == 0) ? (uint8_t)rnd() : mt;
+    int64_t items = (iter % 5 == 0) ? (int64_t)(rnd() % (2 * N)) : N;
+    if (items > N) items = N;  /* dst capacity */
+    (void)vm_unmarshal_int64_array(dst, items, buf, cut, use_mt, first);
+    free(buf);
+  }
+  /* pure garbage bytes under every marshal type */
+  for (int iter = 0; iter < 1000; iter++) {
+    size_t glen = 1 + (size_t)(rnd() % 300);
+    uint8_t* buf = malloc(glen);
+    for (size_t i = 0; i < glen; i++) buf[i] = (uint8_t)rnd();
+    (void)vm_unmarshal_int64_array(dst, (int64_t)(rnd() % N), buf, glen,
+                                   (uint8_t)(rnd() % 8), (int64_t)rnd());
+    free(buf);
+  }
+}
+
 int main(void) {
   codec_roundtrips();
   packed_stream();
   scans_and_rollups();
+  malformed_decode_fuzz();
   printf("asan_check OK\n");
   return 0;
 }
