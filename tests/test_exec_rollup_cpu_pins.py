@@ -121,3 +121,22 @@ def test_rollup_deriv_and_scrape_interval_exec():
                 window=300_000, sq_step=10_000)
     for tag in ("min", "max", "avg"):
         _eq(out[tag], [10] * 6)
+
+
+def test_at_modifier_exec():
+    # exec_test.go:1202-1256 `time() @ X`: the rollup evaluates on the
+    # single-point grid [at, at] and broadcasts to the report grid
+    # (eval.go:903-950).  @1h -> 3600, @start() -> 1000, @end() -> 2000,
+    # @end() offset 10m and @(end()-10m) -> 1400.
+    from victoriametrics_amd import engine
+    for at_s, want in ((3600.0, 3600.0), (1000.0, 1000.0),
+                       (2000.0, 2000.0), (1400.0, 1400.0)):
+        plan, report_ts, bc = engine.plan_with_at(
+            "default_rollup", 1000_000, 2000_000, 200_000, [[at_s]])
+        out = _sq_cpu(None, lambda t: t, start=plan._c.start,
+                      end=plan._c.end, plan=plan)
+        assert out.shape == (1,)
+        full = bc(out.reshape(1, 1)).ravel()
+        assert list(report_ts) == [1000_000, 1200_000, 1400_000,
+                                   1600_000, 1800_000, 2000_000]
+        _eq(full, [want] * 6)
