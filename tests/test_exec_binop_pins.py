@@ -638,3 +638,138 @@ def test_group_left_fill_right():
     eq(got[b"404"].values, [12] * 6)
     eq(got[b"501"].values, [3] * 6)
     assert got[b"501"].mn.get_tag_value(b"method") == b"put"
+
+
+# ---------------------------------------------------------------------
+# `or on(...)` merge walk (exec_test.go:10049-10430): the right side
+# fills only the points no left series of the same on() group covers
+# ---------------------------------------------------------------------
+
+def _or_on(left, right, tags):
+    return _eval(BinOpSpec("or", group_op="on", group_tags=tags),
+                 left, right)
+
+
+def test_nan_or_on_series():
+    # `(label_set(1,a,b1) == 0) or on(a) label_set(2,a,b2)` :10049 — the
+    # all-NaN comparison result drops, the right side survives whole
+    left = [S("", [("a", "a"), ("b", "b1")], np.full(6, NAN))]
+    right = [S("", [("a", "a"), ("b", "b2")], np.full(6, 2.0))]
+    out = remove_empty_series(_or_on(left, right, ["a"]))
+    assert len(out) == 1
+    eq(out[0].values, [2] * 6)
+    assert out[0].mn.get_tag_value("b") == b"b2"
+
+
+def test_series_with_nans_or_scalar():
+    # `(time()>=1600 tagged) or 1` :10069 — plain or (whole-series key
+    # matching): disjoint keys, both kept as-is
+    left = [S("", [("a", "a"), ("b", "b1")],
+              np.where(TIME >= 1600, TIME, NAN))]
+    right = [S("", [], np.ones(6))]
+    out = _eval(BinOpSpec("or"), left, right)
+    assert len(out) == 2
+    m = by_tag(out, "b")
+    eq(m[b"b1"].values, [NAN, NAN, NAN, 1600, 1800, 2000])
+    eq(m[b""].values, [1] * 6)
+
+
+def test_series_or_on_empty_scalar():
+    # `(time()>1200 tagged) or on() vector(0)` :10093 — on() makes ONE
+    # group; the scalar fills exactly the left's NaN points
+    left = [S("", [("a", "a"), ("b", "b1")],
+              np.where(TIME > 1200, TIME, NAN))]
+    right = [S("", [], np.zeros(6))]
+    out = _or_on(left, right, [])
+    assert len(out) == 2
+    m = by_tag(out, "b")
+    eq(m[b"b1"].values, [NAN, NAN, 1400, 1600, 1800, 2000])
+    eq(m[b""].values, [0, 0, NAN, NAN, NAN, NAN])
+
+
+def test_series_or_on_series_disjoint_and_overlap():
+    # :10118 disjoint halves -> right keeps only the uncovered points;
+    # :10150 left full -> right dropped; :10170 overlap -> right loses
+    # the overlapped points
+    left = [S("", [("a", "a"), ("b", "b1")],
+              np.where(TIME <= 1200, TIME, NAN))]
+    right = [S("", [("a", "a"), ("b", "b2")],
+               np.where(TIME > 1200, TIME, NAN))]
+    out = _or_on(left, right, ["a"])
+    m = by_tag(remove_empty_series(out), "b")
+    eq(m[b"b1"].values, [1000, 1200, NAN, NAN, NAN, NAN])
+    eq(m[b"b2"].values, [NAN, NAN, 1400, 1600, 1800, 2000])
+    left = [S("", [("a", "a"), ("b", "b1")], TIME)]
+    right = [S("", [("a", "a"), ("b", "b2")], np.full(6, NAN))]
+    out = remove_empty_series(_or_on(left, right, ["a"]))
+    assert len(out) == 1
+    eq(out[0].values, [1000, 1200, 1400, 1600, 1800, 2000])
+    left = [S("", [("a", "a"), ("b", "b1")],
+              np.where(TIME <= 1500, TIME, NAN))]
+    right = [S("", [("a", "a"), ("b", "b2")],
+               np.where(TIME > 1100, TIME, NAN))]
+    m = by_tag(remove_empty_series(_or_on(left, right, ["a"])), "b")
+    eq(m[b"b1"].values, [1000, 1200, 1400, NAN, NAN, NAN])
+    eq(m[b"b2"].values, [NAN, NAN, NAN, 1600, 1800, 2000])
+
+
+def test_series_or_on_series_merge_same_key():
+    # :10203 — identical full keys merge into ONE series
+    left = [S("", [("a", "a"), ("b", "b1")],
+              np.where(TIME <= 1200, TIME, NAN))]
+    right = [S("", [("a", "a"), ("b", "b1")],
+               np.where(TIME > 1400, TIME, NAN))]
+    out = remove_empty_series(_or_on(left, right, ["a"]))
+    assert len(out) == 1
+    eq(out[0].values, [1000, 1200, NAN, 1600, 1800, 2000])
+
+
+def test_series_or_many_series():
+    # :10243 — one left with a hole at 1200; BOTH right series fill only
+    # that hole, keeping their own tags
+    left = [S("", [("x", "foo")], np.where(TIME != 1200, TIME, NAN))]
+    right = [S("", [("x", "foo"), ("y", "bar")], TIME + 1),
+             S("", [("y", "baz"), ("x", "foo")], TIME + 2)]
+    out = remove_empty_series(_or_on(left, right, ["x"]))
+    assert len(out) == 3
+    m = by_tag(out, "y")
+    eq(m[b""].values, [1000, NAN, 1400, 1600, 1800, 2000])
+    eq(m[b"bar"].values, [NAN, 1201, NAN, NAN, NAN, NAN])
+    eq(m[b"baz"].values, [NAN, 1202, NAN, NAN, NAN, NAN])
+
+
+def test_many_series_or_series():
+    # :10290 — the left GROUP's union covers every point (the y=baz
+    # series is complete), so the right side contributes nothing
+    left = [S("", [("x", "foo")], np.where(TIME != 1200, TIME, NAN)),
+            S("", [("x", "foo"), ("y", "baz")], TIME + 1)]
+    right = [S("", [("x", "foo"), ("y", "bar")], TIME + 2)]
+    out = remove_empty_series(_or_on(left, right, ["x"]))
+    assert len(out) == 2
+    m = by_tag(out, "y")
+    eq(m[b""].values, [1000, NAN, 1400, 1600, 1800, 2000])
+    eq(m[b"baz"].values, TIME + 1)
+
+
+def test_many_series_or_series_merge_matrix():
+    # :10327 no-merge: the group union covers all points -> right dropped;
+    # :10367 merge: 1400 uncovered -> BOTH right series fill it
+    def lhs(a2_from):
+        return [S("", [("job", "a1"), ("a", "a")],
+                  np.where(TIME != 1400, TIME, NAN)),
+                S("", [("job", "a2"), ("a", "a")],
+                  np.where(TIME >= a2_from, TIME, NAN))]
+    right_pair = lambda: [S("", [("job", "a3"), ("a", "a")], TIME.copy()),
+                          S("", [("job", "a4"), ("a", "a")], TIME.copy())]
+    out = remove_empty_series(_or_on(lhs(1400), right_pair(), ["a"]))
+    m = by_tag(out, "job")
+    assert set(m) == {b"a1", b"a2"}
+    eq(m[b"a1"].values, [1000, 1200, NAN, 1600, 1800, 2000])
+    eq(m[b"a2"].values, [NAN, NAN, 1400, 1600, 1800, 2000])
+    out = remove_empty_series(_or_on(lhs(1600), right_pair(), ["a"]))
+    m = by_tag(out, "job")
+    assert set(m) == {b"a1", b"a2", b"a3", b"a4"}
+    eq(m[b"a1"].values, [1000, 1200, NAN, 1600, 1800, 2000])
+    eq(m[b"a2"].values, [NAN, NAN, NAN, 1600, 1800, 2000])
+    eq(m[b"a3"].values, [NAN, NAN, 1400, NAN, NAN, NAN])
+    eq(m[b"a4"].values, [NAN, NAN, 1400, NAN, NAN, NAN])
