@@ -290,3 +290,35 @@ def test_buckets_limit_exec():
         _le(9, "10", [("x", "y")])]))
     assert [(s.mn.get_tag_value(b"le"), s.values[0]) for s in out] == [
         (b"10", 9.0), (b"300", 98.0), (b"inf", 100.0)]
+
+
+def test_sort_family_exec():
+    # :2691-2811 sort()/sort_desc()/sort_by_label{,_desc}/multiple_labels
+    out = tf.sort_series([S(values=2.0), S(tags=[("xx", "foo")],
+                                           values=1.0)])
+    chk(out[0], "", [("xx", "foo")], [1.0] * 6)
+    chk(out[1], "", [], [2.0] * 6)
+    out = tf.sort_series([S(values=1.0), S(tags=[("xx", "foo")],
+                                           values=2.0)], desc=True)
+    chk(out[0], "", [("xx", "foo")], [2.0] * 6)
+    chk(out[1], "", [], [1.0] * 6)
+    # `two_timeseries` :2621 — sort_desc over time() and the constant 2
+    out = tf.sort_series([S(), S(tags=[("xx", "foo")], values=2.0)],
+                         desc=True)
+    chk(out[0], "", [], TIME)
+    chk(out[1], "", [("xx", "foo")], [2.0] * 6)
+    # sort_by_label on __name__
+    out = tf.sort_by_label([S("foo", values=1.0), S("bar", values=2.0)],
+                           ["__name__"])
+    chk(out[0], "bar", [], [2.0] * 6)
+    chk(out[1], "foo", [], [1.0] * 6)
+    out = tf.sort_by_label([S("foo", values=1.0), S("bar", values=2.0)],
+                           ["__name__"], desc=True)
+    chk(out[0], "foo", [], [1.0] * 6)
+    chk(out[1], "bar", [], [2.0] * 6)
+    # multiple labels: equal y, then x decides
+    out = tf.sort_by_label(
+        [S(tags=[("x", "b"), ("y", "aa")], values=1.0),
+         S(tags=[("x", "a"), ("y", "aa")], values=2.0)], ["y", "x"])
+    chk(out[0], "", [("x", "a"), ("y", "aa")], [2.0] * 6)
+    chk(out[1], "", [("x", "b"), ("y", "aa")], [1.0] * 6)
