@@ -553,3 +553,28 @@ def test_distinct_exec_pin():
     g = out[0]
     assert np.isnan(g[0])
     assert list(g[1:]) == [1, 1, 1, 2, 2]
+
+
+def test_sum_by_name_exec_pin():
+    # `sum(...) by (__name__)` exec_test.go:10428 — grouping on __name__
+    # keeps the metric group and drops all tags
+    from victoriametrics_amd import aggregate as agg
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+    TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+    series = [
+        Series(MetricName("bar", [("baz", "sss"), ("x", "y")]),
+               np.full(6, 10.0)),
+        Series(MetricName("aaa", [("baz", "sss")]), TIME / 100),
+    ]
+    groups = agg.prepare_series(series, "by", ["__name__"])
+    assert len(groups) == 2
+    got = {}
+    for gmn, members in groups:
+        assert gmn.tags == []
+        v = np.stack([s.values for s in members])
+        gr = np.arange(len(members), dtype=np.uint32)
+        go = np.asarray([0, len(members)], np.uint64)
+        got[gmn.metric_group] = oracle.colagg("sum", v, gr, go)[0]
+    np.testing.assert_array_equal(got[b"bar"], [10.0] * 6)
+    np.testing.assert_array_equal(got[b"aaa"], [10, 12, 14, 16, 18, 20])
