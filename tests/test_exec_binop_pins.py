@@ -773,3 +773,32 @@ def test_many_series_or_series_merge_matrix():
     eq(m[b"a2"].values, [NAN, NAN, NAN, 1600, 1800, 2000])
     eq(m[b"a3"].values, [NAN, NAN, 1400, NAN, NAN, NAN])
     eq(m[b"a4"].values, [NAN, NAN, 1400, NAN, NAN, NAN])
+
+
+def test_union_list_comparisons():
+    # `time() == (100, 1000, 1400, 600)` :6468 and the reversed form
+    # :6479 -> [1000, nan, 1400, nan, nan, nan]; `!=` :6490 keeps the
+    # complement.  The union side is the scalar list (binary_op.go:55-113)
+    from victoriametrics_amd.binary_op import union_list_cmp
+    from victoriametrics_amd import transform as tfm
+
+    def lst():
+        return tfm.union([[scalar(100)], [scalar(1000)], [scalar(1400)],
+                          [scalar(600)]])
+    assert len(lst()) == 4  # all-scalar union keeps every member
+    out = union_list_cmp("==", [t_series()], lst())
+    assert len(out) == 1
+    eq(out[0].values, [1000, NAN, 1400, NAN, NAN, NAN])
+    # reversed operands: the union side still masks the data side
+    out = union_list_cmp("==", lst(), [t_series()], union_on_left=True)
+    assert len(out) == 1
+    eq(out[0].values, [1000, NAN, 1400, NAN, NAN, NAN])
+    # `alias(time(),"foobar") != union(100, 1000, 1400, 600)` :6490
+    out = union_list_cmp("!=", [S("foobar", [], TIME)], lst())
+    assert len(out) == 1
+    assert out[0].mn.metric_group == b"foobar"
+    eq(out[0].values, [NAN, 1200, NAN, 1600, 1800, 2000])
+    # empty sides (binary_op.go:69/95): == -> empty, != -> left as-is
+    assert union_list_cmp("==", [t_series()], []) == []
+    out = union_list_cmp("!=", [t_series()], [])
+    eq(out[0].values, TIME)

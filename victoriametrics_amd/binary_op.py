@@ -279,10 +279,42 @@ def _gpu_apply(spec, left, right, dst, drop_nan_right):
     return dst
 
 
+def union_list_cmp(op, left, right, union_on_left=False):
+    """binaryOpEqFunc / binaryOpNeqFunc's union-list special case
+    (binary_op.go:55-113): `q == (v1,...,vN)` keeps each left point only
+    when SOME union member holds that exact value there; `!=` keeps it
+    when NO member does.  The union side may be either operand
+    (union_on_left swaps).  Host-side mask: the list is tiny (parsed
+    literals), the data side stays resident."""
+    if union_on_left:
+        left, right = right, left
+    if op == "==":
+        if not left or not right:
+            return []
+    else:
+        if not left:
+            return []
+        if not right:
+            return left
+    rv = np.stack([np.asarray(s.values, np.float64) for s in right])
+    for s in left:
+        hit = (rv == np.asarray(s.values, np.float64)).any(axis=0)
+        if op == "==":
+            s.values[~hit] = math.nan
+        else:
+            s.values[hit] = math.nan
+    return left
+
+
 def binary_op_eval(spec, left, right, drop_nan_right=False, apply_fn=None,
-                   mask_fn=None, or_fn=None):
+                   mask_fn=None, or_fn=None, union_list=None):
     """binaryOpFuncs dispatch (binary_op.go:15).  apply_fn/mask_fn/or_fn
-    override the GPU kernels (tests only — oracle-backed)."""
+    override the GPU kernels (tests only — oracle-backed).  union_list:
+    "left"/"right" marks that operand as a scalar-list union expression,
+    routing == / != through union_list_cmp (binary_op.go:55-113)."""
+    if union_list and spec.op in ("==", "!="):
+        return union_list_cmp(spec.op, left, right,
+                              union_on_left=(union_list == "left"))
     if spec.op in SET_OPS:
         return _set_op(spec, left, right, mask_fn, or_fn)
 

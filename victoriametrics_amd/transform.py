@@ -601,8 +601,14 @@ def absent(series_list, n_grid, base_mn=None):
 
 
 def union(args):
-    """transformUnion (transform.go:2716): concatenate series lists,
-    dropping later series whose metric name was already seen."""
+    """transformUnion (transform.go:1743): concatenate series lists,
+    dropping later series whose metric name was already seen.  Special
+    case (transform.go:1749): when EVERY arg is a single scalar series,
+    all of them are returned without dedup — this feeds the
+    `q == (v1,...,vN)` list comparisons."""
+    from .binary_op import is_scalar
+    if args and all(is_scalar(arg) for arg in args):
+        return [arg[0] for arg in args]
     rvs, seen = [], set()
     for arg in args:
         for s in arg:
