@@ -802,3 +802,63 @@ def test_union_list_comparisons():
     assert union_list_cmp("==", [t_series()], []) == []
     out = union_list_cmp("!=", [t_series()], [])
     eq(out[0].values, TIME)
+
+
+def test_vector_scalar_arith_and_keep_metric_names():
+    # `vector / scalar` :3362 and the keep_metric_names variants
+    # :3386/:3447 — plain arith resets __name__, keep_metric_names keeps it
+    left = [S("q1", [("foo", "bar")], TIME),
+            S("q2", [("foo", "qwert")], np.full(6, 10.0))]
+    out = _eval(BinOpSpec("/"), left, [scalar(2)])
+    m = by_tag(out, "foo")
+    eq(m[b"bar"].values, [500, 600, 700, 800, 900, 1000])
+    eq(m[b"qwert"].values, [5] * 6)
+    for s in out:
+        assert s.mn.metric_group == b""
+    left = [S("q1", [("foo", "bar")], TIME),
+            S("q2", [("foo", "qwert")], np.full(6, 10.0))]
+    out = _eval(BinOpSpec("/", keep_metric_names=True), left, [scalar(2)])
+    names = {s.mn.get_tag_value("foo"): s.mn.metric_group for s in out}
+    assert names == {b"bar": b"q1", b"qwert": b"q2"}
+    left = [S("q1", [("foo", "bar")], TIME),
+            S("q2", [("foo", "qwert")], np.full(6, 10.0))]
+    out = _eval(BinOpSpec("*", keep_metric_names=True), [scalar(2)], left)
+    m = by_tag(out, "foo")
+    eq(m[b"bar"].values, [2000, 2400, 2800, 3200, 3600, 4000])
+    eq(m[b"qwert"].values, [20] * 6)
+
+
+def test_vector_mul_on_foo_scalar():
+    # `(...) * on(foo) label_set(2, foo=bar, aa=bb)` :3591 — only the
+    # foo=bar left matches; on(foo) reduces the result name to foo
+    left = [S("", [("foo", "bar"), ("xx", "yy")], TIME),
+            S("", [("foo", "qwert")], np.full(6, 10.0))]
+    right = [S("", [("foo", "bar"), ("aa", "bb")], np.full(6, 2.0))]
+    out = _eval(BinOpSpec("*", group_op="on", group_tags=["foo"]),
+                left, right)
+    assert len(out) == 1
+    eq(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+    assert sorted(out[0].mn.tags) == [(b"foo", b"bar")]
+    # :3608 keep_metric_names: __name__ joins the grouping and survives
+    left = [S("q1", [("foo", "bar"), ("xx", "yy")], TIME),
+            S("q2", [("foo", "qwert")], np.full(6, 10.0))]
+    right = [S("q2", [("foo", "bar"), ("aa", "bb")], np.full(6, 2.0))]
+    out = _eval(BinOpSpec("*", group_op="on", group_tags=["foo"],
+                          keep_metric_names=True), left, right)
+    assert len(out) == 1
+    eq(out[0].values, [2000, 2400, 2800, 3200, 3600, 4000])
+    assert out[0].mn.metric_group == b"q1"
+    assert sorted(out[0].mn.tags) == [(b"foo", b"bar")]
+
+
+def test_vector_on_empty_group_left_scalar():
+    # `(...) * on() group_left 2` :3786 — every left series joins the
+    # lone scalar; left names survive whole
+    left = [S("", [("foo", "bar")], TIME),
+            S("", [("foo", "qwert")], np.full(6, 10.0))]
+    spec = BinOpSpec("*", group_op="on", group_tags=[],
+                     join_op="group_left", join_tags=[])
+    out = _eval(spec, left, [scalar(2)])
+    m = by_tag(out, "foo")
+    eq(m[b"bar"].values, [2000, 2400, 2800, 3200, 3600, 4000])
+    eq(m[b"qwert"].values, [20] * 6)
