@@ -1130,3 +1130,23 @@ def test_numeric_less_reference_vectors():
     ]
     for a, b, want in cases:
         assert tf.numeric_less(a, b) == want, (a, b, want)
+
+
+def test_go_expand_template_semantics():
+    # Go regexp.Expand rules (regexp/regexp.go): longest-name refs,
+    # unknown groups -> "", $$ literal, ${name} explicit
+    import re as _re
+    m = _re.match(r"^(?:(a+)(b+))$", "aabbb")
+    ge = tf._go_expand
+    assert ge(m, "x${1}y") == "xaay"
+    assert ge(m, "x$1y") == "x"          # $1y reads group "1y" -> absent
+    assert ge(m, "$1-$2") == "aa-bbb"
+    assert ge(m, "$$1") == "$1"
+    assert ge(m, "$9") == ""             # out-of-range -> empty
+    assert ge(m, "${9}z") == "z"
+    assert ge(m, "lone$") == "lone$"
+    assert ge(m, "${unclosed") == "${unclosed"
+    m2 = _re.match(r"^(?:(?P<word>\w+))$", "hello")
+    assert ge(m2, "<$word>") == "<hello>"
+    assert ge(m2, "<$wordx>") == "<>"    # longest-run name "wordx" absent
+    assert ge(m2, "<${word}x>") == "<hellox>"
