@@ -862,3 +862,93 @@ def test_vector_on_empty_group_left_scalar():
     m = by_tag(out, "foo")
     eq(m[b"bar"].values, [2000, 2400, 2800, 3200, 3600, 4000])
     eq(m[b"qwert"].values, [20] * 6)
+
+
+# ---------------------------------------------------------------------
+# NaN comparison semantics (exec_test.go:2980-3126): issue 150 (explicit
+# NaN is a value: `x != NaN` is true), issue 10018 (NaN from a vector
+# comparison is a FILTERED sample: drop it on the right), PR 11100
+# (per-point drop for partially-filtered right sides)
+# ---------------------------------------------------------------------
+
+def test_compare_to_nan_scalar_sides():
+    # `1 != nan` :2980 -> 1s; `nan != 1` :2991 -> empty
+    out = remove_empty_series(_eval(BinOpSpec("!="), [scalar(1)],
+                                    [S("", [], np.full(6, NAN))]))
+    assert len(out) == 1
+    eq(out[0].values, [1] * 6)
+    out = remove_empty_series(_eval(BinOpSpec("!="),
+                                    [S("", [], np.full(6, NAN))],
+                                    [scalar(1)]))
+    assert out == []
+    # :2997 vector != NaN-scalar -> kept whole
+    out = remove_empty_series(_eval(BinOpSpec("!="),
+                                    [S("", [("foo", "bar")], TIME)],
+                                    [S("", [], np.full(6, NAN))]))
+    eq(out[0].values, TIME)
+    # :3012 vector != 1200 filters the equal point
+    out = _eval(BinOpSpec("!="), [S("", [("foo", "bar")], TIME)],
+                [scalar(1200)])
+    eq(out[0].values, [1000, NAN, 1400, 1600, 1800, 2000])
+
+
+def test_compare_to_explicit_nan_vector_right():
+    # :3027 the right carries an EXPLICIT NaN sample (not produced by a
+    # comparison): `x != NaN` is true, series kept whole
+    left = [S("", [("foo", "bar")], TIME)]
+    right = [S("", [("foo", "bar")], np.full(6, NAN))]
+    out = remove_empty_series(_eval(BinOpSpec("!="), left, right,
+                                    drop_nan_right=False))
+    assert len(out) == 1
+    eq(out[0].values, TIME)
+    # :3042 `x != (1 > 2)` — a scalar-only comparison also counts as an
+    # explicit NaN (isVectorComparisonExpr is false)
+    right = [S("", [], np.full(6, NAN))]
+    out = remove_empty_series(_eval(BinOpSpec("!="),
+                                    [S("", [("foo", "bar")], TIME)],
+                                    right, drop_nan_right=False))
+    eq(out[0].values, TIME)
+
+
+def test_compare_to_filtered_vector_right():
+    # :3057 the right's NaNs come from a VECTOR comparison -> dropped
+    # per point (drop_nan_right=True) -> empty result
+    left = [S("", [("foo", "bar")], TIME)]
+    right = [S("", [("foo", "bar")], np.full(6, NAN))]
+    out = remove_empty_series(_eval(BinOpSpec("!="), left, right,
+                                    drop_nan_right=True))
+    assert out == []
+    # :3069 all-NaN LEFT needs no special flag
+    out = remove_empty_series(_eval(
+        BinOpSpec("!="), [S("", [("foo", "bar")], np.full(6, NAN))],
+        [S("", [("foo", "bar")], TIME)]))
+    assert out == []
+    # :3077 bool form drops the same way
+    out = remove_empty_series(_eval(
+        BinOpSpec("==", bool_modifier=True),
+        [S("", [("foo", "bar")], TIME)],
+        [S("", [("foo", "bar")], np.full(6, NAN))], drop_nan_right=True))
+    assert out == []
+    # :3083 partially filtered right: time()*2 > 2800 -> only the tail
+    # survives the per-point drop
+    right = [S("", [("foo", "bar")],
+               np.where(TIME * 2 > 2800, TIME * 2, NAN))]
+    out = _eval(BinOpSpec("!="), [S("", [("foo", "bar")], TIME)], right,
+                drop_nan_right=True)
+    eq(out[0].values, [NAN, NAN, NAN, 1600, 1800, 2000])
+
+
+def test_compare_to_filtered_right_with_fills():
+    # :3108 fill_left does NOT rescue the dropped right side -> empty;
+    # :3117 fill_right replaces the dropped NaN -> full series survives
+    left = [S("", [("foo", "bar")], TIME)]
+    right = [S("", [("foo", "bar")], np.full(6, NAN))]
+    out = remove_empty_series(_eval(
+        BinOpSpec("!=", fill_left=0.0), left, right, drop_nan_right=True))
+    assert out == []
+    left = [S("", [("foo", "bar")], TIME)]
+    right = [S("", [("foo", "bar")], np.full(6, NAN))]
+    out = remove_empty_series(_eval(
+        BinOpSpec("!=", fill_right=0.0), left, right, drop_nan_right=True))
+    assert len(out) == 1
+    eq(out[0].values, TIME)
