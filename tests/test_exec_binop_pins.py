@@ -952,3 +952,57 @@ def test_compare_to_filtered_right_with_fills():
         BinOpSpec("!=", fill_right=0.0), left, right, drop_nan_right=True))
     assert len(out) == 1
     eq(out[0].values, TIME)
+
+
+def test_vector_div_vector_ignoring_fill():
+    # `(... by code) / ignoring(code) fill(0) (... by method)` :4362 —
+    # the post series exists only on the right: fill_left gives 0/5 = 0
+    left = [S("", [("method", "get"), ("code", "500")], np.full(6, 6.0)),
+            S("", [("method", "put"), ("code", "500")], np.full(6, 1.0))]
+    right = [S("", [("method", "get")], np.full(6, 12.0)),
+             S("", [("method", "post")], np.full(6, 5.0)),
+             S("", [("method", "put")], np.full(6, 10.0))]
+    out = _eval(BinOpSpec("/", group_op="ignoring", group_tags=["code"],
+                          fill_left=0.0, fill_right=0.0), left, right)
+    m = by_tag(out, "method")
+    assert set(m) == {b"get", b"post", b"put"}
+    eq(m[b"get"].values, [0.5] * 6)
+    eq(m[b"post"].values, [0] * 6)
+    eq(m[b"put"].values, [0.1] * 6)
+    for s in out:
+        assert s.mn.get_tag_value("code") is None
+
+
+def test_result_sorting_canonical():
+    # `result sorting` :9731 — the exec tail sorts by metric name then
+    # tag pairs (sortSeriesByMetricName/metricNameLess)
+    from victoriametrics_amd.binary_op import sort_series_by_metric_name
+    xs = [
+        S("", [("instance", "localhost:1001"), ("type", "free")], 1.0 * np.ones(6)),
+        S("", [("instance", "localhost:1001"), ("type", "buffers")], np.ones(6)),
+        S("", [("instance", "localhost:1000"), ("type", "buffers")], np.ones(6)),
+        S("", [("instance", "localhost:1000"), ("type", "free")], np.ones(6)),
+    ]
+    sort_series_by_metric_name(xs)  # sorts in place, like the reference
+    got = [(s.mn.get_tag_value("instance"), s.mn.get_tag_value("type"))
+           for s in xs]
+    assert got == [(b"localhost:1000", b"buffers"),
+                   (b"localhost:1000", b"free"),
+                   (b"localhost:1001", b"buffers"),
+                   (b"localhost:1001", b"free")]
+
+
+def test_sort_by_label_numeric_special_chars():
+    # :9836 / :9983 — numeric ranks inside special-char strings
+    from victoriametrics_amd import transform as tfm
+    xs = [S("", [("x", "1:0:2"), ("y", "1:0:1")], np.ones(6)),
+          S("", [("x", "1:0:15"), ("y", "1:0:1")], np.full(6, 2.0))]
+    out = tfm.sort_by_label_numeric(xs, ["x", "y"])
+    assert [s.mn.get_tag_value("x") for s in out] == [b"1:0:2", b"1:0:15"]
+    xs = [S("", [("a", "DS50:1/0/15")], np.full(6, 4.0)),
+          S("", [("a", "DS50:1/0/0")], np.ones(6)),
+          S("", [("a", "DS50:1/0/1")], np.full(6, 2.0)),
+          S("", [("a", "DS50:1/0/2")], np.full(6, 3.0))]
+    out = tfm.sort_by_label_numeric(xs, ["a"])
+    assert [s.mn.get_tag_value("a") for s in out] == [
+        b"DS50:1/0/0", b"DS50:1/0/1", b"DS50:1/0/2", b"DS50:1/0/15"]
