@@ -578,3 +578,25 @@ def test_sum_by_name_exec_pin():
         got[gmn.metric_group] = oracle.colagg("sum", v, gr, go)[0]
     np.testing.assert_array_equal(got[b"bar"], [10.0] * 6)
     np.testing.assert_array_equal(got[b"aaa"], [10, 12, 14, 16, 18, 20])
+
+
+def test_sum_by_known_tag_limit_exec_pin():
+    # `sum(...) by (foo) limit 1` exec_test.go:6374 — the limit caps the
+    # number of GROUPS, keeping the first-seen one
+    from victoriametrics_amd import aggregate as agg
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+    TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+    series = [
+        Series(MetricName("", [("foo", "bar")]), np.full(6, 10.0)),
+        Series(MetricName("", [("baz", "sss")]), TIME / 100),
+    ]
+    groups = agg.prepare_series(series, "by", ["foo"], max_series=1)
+    assert len(groups) == 1
+    gmn, members = groups[0]
+    assert sorted(gmn.tags) == [(b"foo", b"bar")]
+    np.testing.assert_array_equal(members[0].values, [10.0] * 6)
+    # without the limit both groups exist (:6354)
+    groups = agg.prepare_series(
+        [s.copy_shallow() for s in series], "by", ["foo"])
+    assert len(groups) == 2
