@@ -1150,3 +1150,52 @@ def test_go_expand_template_semantics():
     assert ge(m2, "<$word>") == "<hello>"
     assert ge(m2, "<$wordx>") == "<>"    # longest-run name "wordx" absent
     assert ge(m2, "<${word}x>") == "<hellox>"
+
+
+def test_histogram_normal_bucket_count_exec_pins():
+    # exec_test.go:5018-5165 normal-bucket-count family via the oracle
+    # walk (the device kernel is pinned against it in the GPU suite):
+    # quantile(0.2) -> 22 with bounds [10, 30]; share(35) -> 1/3;
+    # share(22) -> 0.2 with bounds [0, 1/3]; fraction(22,35) -> 2/15
+    bv = np.stack([np.zeros(6), np.full(6, 100.0), np.full(6, 300.0)])
+    les = np.asarray([10.0, 30.0, np.inf])
+    off = np.asarray([0, 3], np.uint64)
+    out, lo, hi = oracle.histogram_quantile(0.2, bv, les, off, bounds=True)
+    assert list(out[0]) == [22.0] * 6
+    assert list(lo[0]) == [10.0] * 6
+    assert list(hi[0]) == [30.0] * 6
+    share35 = oracle.histogram_share(np.full(6, 35.0), bv, les, off)
+    out, lo, hi = share35[0], share35[1], share35[2]
+    np.testing.assert_allclose(out[0], [1.0 / 3] * 6, rtol=1e-15)
+    share22 = oracle.histogram_share(np.full(6, 22.0), bv, les, off)
+    out, lo, hi = share22[0], share22[1], share22[2]
+    np.testing.assert_allclose(out[0], [0.2] * 6, rtol=1e-15)
+    assert list(lo[0]) == [0.0] * 6
+    np.testing.assert_allclose(hi[0], [1.0 / 3] * 6, rtol=1e-15)
+    frac = share35[0][0] - share22[0][0]
+    np.testing.assert_allclose(frac, [0.1333333333333333] * 6, rtol=1e-14)
+
+
+def test_histogram_quantile_duplicate_le_exec_pin():
+    # `histogram_quantile(0.6, ... le=5 or le=5.0 or le=6.0 or +Inf)`
+    # exec_test.go — equal-le buckets merge by SUM (mergeSameLE) before
+    # the walk; round(_, 0.1) -> 4.7
+    from victoriametrics_amd.decimal import go_round
+    series = [
+        S("", [("foo", "bar"), ("le", "5")], np.full(6, 90.0)),
+        S("", [("foo", "bar"), ("le", "5.0")], np.full(6, 100.0)),
+        S("", [("foo", "bar"), ("le", "6.0")], np.full(6, 200.0)),
+        S("", [("foo", "bar"), ("le", "+Inf")], np.full(6, 300.0)),
+    ]
+    groups = tf.group_le_timeseries(series)
+    assert len(groups) == 1
+    xss = sorted(next(iter(groups.values())), key=lambda p: p[0])
+    merged = tf._merge_same_le(xss)
+    les = np.asarray([le for le, _ in merged])
+    bv = np.stack([s.values for _, s in merged])
+    np.testing.assert_array_equal(les, [5.0, 6.0, np.inf])
+    np.testing.assert_array_equal(bv[:, 0], [190.0, 200.0, 300.0])
+    off = np.asarray([0, len(merged)], np.uint64)
+    out, _, _ = oracle.histogram_quantile(0.6, bv, les, off)
+    got = go_round(out[0] * 10.0) / 10.0
+    np.testing.assert_array_equal(got, [4.7] * 6)
