@@ -1199,3 +1199,24 @@ def test_histogram_quantile_duplicate_le_exec_pin():
     out, _, _ = oracle.histogram_quantile(0.6, bv, les, off)
     got = go_round(out[0] * 10.0) / 10.0
     np.testing.assert_array_equal(got, [4.7] * 6)
+
+
+def test_round_to_decimal_digits_reference_vectors():
+    # TestRoundToDecimalDigits (lib/decimal/decimal_test.go:10-41)
+    # verbatim, including the StaleNaN pass-through
+    from victoriametrics_amd import decimal as vmd
+    stale = np.frombuffer(np.int64(0x7FF0000000000002).tobytes(),
+                          np.float64)[0]
+    cases = [
+        (12.34, 0, 12.0), (12.57, 0, 13.0), (-1.578, 2, -1.58),
+        (-1.578, 3, -1.578), (1234.0, -2, 1200.0), (1235.0, -1, 1240.0),
+        (1234.0, 0, 1234.0), (1234.6, 0, 1235.0),
+        (123.4e-99, 99, 123e-99),
+    ]
+    for v, digits, want in cases:
+        got = vmd.round_to_decimal_digits(np.asarray([v]), digits)[0]
+        assert got == want, (v, digits, got, want)
+    got = vmd.round_to_decimal_digits(np.asarray([math.nan]), 10)[0]
+    assert math.isnan(got) and np.isnan(got)
+    got = vmd.round_to_decimal_digits(np.asarray([stale]), 10)[0]
+    assert np.asarray([got]).view(np.int64)[0] == 0x7FF0000000000002
