@@ -208,3 +208,23 @@ def test_block_marshal_pipeline_roundtrip():
     assert np.array_equal(va_back, va)
     f_back = oracle.decimal_append_to_float(va_back, scale)
     assert np.allclose(f_back, vals_f, rtol=1e-12, atol=0)
+
+
+def test_ensure_non_decreasing_reference_vectors():
+    # TestEnsureNonDecreasingSequence (lib/encoding/encoding_test.go:71-81)
+    # verbatim — decode-side timestamp clamping
+    cases = [
+        ([], -1234, -34, []),
+        ([123], -1234, -1234, [-1234]),
+        ([123], -1234, 345, [345]),
+        ([-23, -14], -23, -14, [-23, -14]),
+        ([-23, -14], -25, 0, [-25, 0]),
+        ([0, -1, 10, 5, 6, 7], 2, 8, [2, 2, 8, 8, 8, 8]),
+        ([0, -1, 10, 5, 6, 7], -2, 8, [-2, -1, 8, 8, 8, 8]),
+        ([0, -1, 10, 5, 6, 7], -2, 12, [-2, -1, 10, 10, 10, 12]),
+        ([1, 2, 1, 3, 4, 5], 1, 5, [1, 2, 2, 3, 4, 5]),
+    ]
+    for a, vmin, vmax, want in cases:
+        got = oracle.ensure_non_decreasing(
+            np.asarray(a, np.int64), vmin, vmax)
+        assert list(got) == want, (a, vmin, vmax, list(got), want)
