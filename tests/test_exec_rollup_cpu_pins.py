@@ -140,3 +140,15 @@ def test_at_modifier_exec():
         assert list(report_ts) == [1000_000, 1200_000, 1400_000,
                                    1600_000, 1800_000, 2000_000]
         _eq(full, [want] * 6)
+
+
+def test_subquery_offset_exec():
+    # `time()[:100] offset 0` :452 -> identity (100s sub-step divides the
+    # grid); `time()[300:100] offset 100` :647 and the `i`-duration form
+    # `time()[1.5i:0.5i] offset 0.5i` :658 -> one 100s sub-step back
+    got = _sq_cpu("default_rollup", lambda t: t, sq_step=100_000)
+    _eq(got, [1000, 1200, 1400, 1600, 1800, 2000])
+    got = _sq_cpu("default_rollup", lambda t: t, window=300_000,
+                  sq_step=100_000, start=1000_000 - 100_000,
+                  end=2000_000 - 100_000)
+    _eq(got, [900, 1100, 1300, 1500, 1700, 1900])
