@@ -1006,3 +1006,25 @@ def test_sort_by_label_numeric_special_chars():
     out = tfm.sort_by_label_numeric(xs, ["a"])
     assert [s.mn.get_tag_value("a") for s in out] == [
         b"DS50:1/0/0", b"DS50:1/0/1", b"DS50:1/0/2", b"DS50:1/0/15"]
+
+
+def test_scalar_cmp_edges():
+    # `vector(1) == bool time()` :2963 -> all 0 (1 never equals time);
+    # `vector(1) == time()` :2974 -> empty after removeEmptySeries;
+    # `1 > 2` :2957 -> empty; `-1 < 2` :3127 -> the left scalar survives;
+    # `time() >= bool 2` :3149 -> all 1
+    out = _eval(BinOpSpec("==", bool_modifier=True),
+                [scalar(1)], [t_series()])
+    assert len(out) == 1
+    eq(out[0].values, [0] * 6)
+    out = remove_empty_series(_eval(BinOpSpec("=="), [scalar(1)],
+                                    [t_series()]))
+    assert out == []
+    out = remove_empty_series(_eval(BinOpSpec(">"), [scalar(1)],
+                                    [scalar(2)]))
+    assert out == []
+    out = _eval(BinOpSpec("<"), [scalar(-1)], [scalar(2)])
+    eq(out[0].values, [-1] * 6)
+    out = _eval(BinOpSpec(">=", bool_modifier=True), [t_series()],
+                [scalar(2)])
+    eq(out[0].values, [1] * 6)
