@@ -600,3 +600,29 @@ def test_sum_by_known_tag_limit_exec_pin():
     groups = agg.prepare_series(
         [s.copy_shallow() for s in series], "by", ["foo"])
     assert len(groups) == 2
+
+
+def test_sum_label_graphite_group_exec_pin():
+    # `sort(sum by (__name__) (label_graphite_group((...), 1)))`
+    # exec_test.go:2597 — graphite component 1 becomes the metric group,
+    # then by (__name__) folds the two "bar" series
+    from victoriametrics_amd import aggregate as agg
+    from victoriametrics_amd import transform as tfm
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+    series = [
+        Series(MetricName("foo.bar.baz"), np.full(6, 1.0)),
+        Series(MetricName("x.y.z"), np.full(6, 2.0)),
+        Series(MetricName("qe.bar.qqq"), np.full(6, 3.0)),
+    ]
+    out = tfm.label_graphite_group(series, [1])
+    assert sorted(s.mn.metric_group for s in out) == [b"bar", b"bar", b"y"]
+    groups = agg.prepare_series(out, "by", ["__name__"])
+    got = {}
+    for gmn, members in groups:
+        v = np.stack([s.values for s in members])
+        gr = np.arange(len(members), dtype=np.uint32)
+        go = np.asarray([0, len(members)], np.uint64)
+        got[gmn.metric_group] = oracle.colagg("sum", v, gr, go)[0]
+    np.testing.assert_array_equal(got[b"y"], [2.0] * 6)
+    np.testing.assert_array_equal(got[b"bar"], [4.0] * 6)
