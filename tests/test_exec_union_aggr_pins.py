@@ -128,3 +128,21 @@ def test_group_exec():
     go = np.asarray([0, 3], np.uint64)
     out = oracle.colagg("group", v, gr, go)
     np.testing.assert_array_equal(out[0], np.ones(6))
+
+
+def test_sum_over_union_exec():
+    # `sum((1, 2, 3))` :5882 -> 6: the all-scalar union keeps every
+    # member; `sum((alias(1,"foo"), alias(2,"foo"), alias(3,"foo")))`
+    # :5893 -> 1: named duplicates dedup to the FIRST in the union
+    scalars = tf.union([[S(values=1.0)], [S(values=2.0)],
+                        [S(values=3.0)]])
+    assert len(scalars) == 3
+    v = np.stack([s.values for s in scalars])
+    gr = np.arange(3, dtype=np.uint32)
+    go = np.asarray([0, 3], np.uint64)
+    out = oracle.colagg("sum", v, gr, go)
+    np.testing.assert_array_equal(out[0], [6.0] * 6)
+    named = tf.union([[S("foo", values=1.0)], [S("foo", values=2.0)],
+                      [S("foo", values=3.0)]])
+    assert len(named) == 1
+    np.testing.assert_array_equal(named[0].values, [1.0] * 6)
