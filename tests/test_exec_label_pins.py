@@ -322,3 +322,27 @@ def test_sort_family_exec():
          S(tags=[("x", "a"), ("y", "aa")], values=2.0)], ["y", "x"])
     chk(out[0], "", [("x", "a"), ("y", "aa")], [2.0] * 6)
     chk(out[1], "", [("x", "b"), ("y", "aa")], [1.0] * 6)
+
+
+def test_label_match_mismatch_name_exec():
+    # :2483/:2499 — regex filter on __name__ keeps/drops whole series
+    xs = [S("foo", [], TIME), S("bar", [], TIME * 2)]
+    out = tf.label_match([s for s in xs], "__name__", "f.+")
+    assert len(out) == 1
+    chk(out[0], "foo", [], TIME)
+    out = tf.label_match([S("foo", [], TIME), S("bar", [], TIME * 2)],
+                         "__name__", "f.+", negate=True)
+    assert len(out) == 1
+    chk(out[0], "bar", [], TIME * 2)
+
+
+def test_label_graphite_group_multi_ids_exec():
+    # :2515 — group ids 1 and 3: out-of-range components become "",
+    # joined with "."
+    xs = [S("foo.bar.baz", [], np.ones(6)),
+          S("abc", [], np.full(6, 2.0)),
+          S("a.xx.zz.asd", [("qwe", "rty")], np.full(6, 3.0))]
+    out = tf.sort_series(tf.label_graphite_group(xs, [1, 3]))
+    chk(out[0], "bar.", [], [1.0] * 6)
+    chk(out[1], ".", [], [2.0] * 6)
+    chk(out[2], "xx.asd", [("qwe", "rty")], [3.0] * 6)
