@@ -250,7 +250,7 @@ def _rand_spec(rng):
                      fill_left=fl, fill_right=fr)
 
 
-@pytest.mark.parametrize("seed", range(4))
+@pytest.mark.parametrize("seed", range(6))
 def test_adjust_and_apply_match_reference_model(seed):
     rng = np.random.default_rng(7000 + seed)
     agree = raised = 0

@@ -200,7 +200,7 @@ def _rand_series_set(rng, n):
     return tss
 
 
-@pytest.mark.parametrize("seed", range(4))
+@pytest.mark.parametrize("seed", range(6))
 def test_setop_walks_match_reference_model(seed):
     rng = np.random.default_rng(1000 + seed)
     ops = ["and", "or", "unless", "if", "ifnot", "default"]
