@@ -38,12 +38,15 @@ def _go_op(op, a, b):
         if op == "%":
             return float(np.fmod(np.float64(a), np.float64(b)))
         if op == "^":
-            # C libm pow (Go math.Pow == the oracle): math.pow IS libm;
-            # fall back to NaN on Python's domain errors
+            # binaryop.Pow (funcs.go:78): NaN^any = NaN (issue 7359,
+            # overriding IEEE pow(NaN,0)=1), then C libm pow; math.pow IS
+            # libm — fall back on Python's domain errors
+            if a != a:
+                return NAN
             try:
                 return math.pow(a, b)
             except (ValueError, OverflowError):
-                if a != a or b != b:
+                if b != b:
                     return NAN
                 return NAN if a < 0 else math.inf
         if op == "atan2":

@@ -1028,3 +1028,14 @@ def test_scalar_cmp_edges():
     out = _eval(BinOpSpec(">=", bool_modifier=True), [t_series()],
                 [scalar(2)])
     eq(out[0].values, [1] * 6)
+
+
+def test_nan_pow_any():
+    # `nan^any` exec_test.go:10038 / binaryop.Pow (funcs.go:78, issue
+    # 7359): NaN^0 is NaN, overriding IEEE pow(NaN, 0) = 1
+    left = [S("", [], np.full(6, NAN))]
+    out = remove_empty_series(_eval(BinOpSpec("^"), left, [scalar(0)]))
+    assert out == []
+    # and a present base stays IEEE: x^0 = 1
+    out = _eval(BinOpSpec("^"), [t_series()], [scalar(0)])
+    eq(out[0].values, [1] * 6)
