@@ -1220,3 +1220,50 @@ def test_round_to_decimal_digits_reference_vectors():
     assert math.isnan(got) and np.isnan(got)
     got = vmd.round_to_decimal_digits(np.asarray([stale]), 10)[0]
     assert np.asarray([got]).view(np.int64)[0] == 0x7FF0000000000002
+
+
+def test_trig_exec_pins():
+    # exec_test.go:1334-1444 — trig family over pi()*(2000-time())/1000
+    # or (2000-time())/1000.  The expected arrays encode Go's math
+    # package (its OWN soft-float sin/cos implementations, not libm) —
+    # near sin(pi) they differ from glibc by an ulp, so compare at the
+    # same 1e-14 relative tolerance the GPU exec pins use
+    x = (2000.0 - TIME6) / 1000.0
+    pix = np.pi * x
+
+    def tfm(fid, v):
+        out, _ = oracle.tf_apply(fid, np.asarray(v).reshape(1, -1).copy())
+        return out.ravel()
+
+    def close(got, want):
+        np.testing.assert_allclose(got, want, rtol=1e-14, atol=1e-300)
+
+    close(tfm(8, pix),                      # sin
+          [1.2246467991473515e-16, 0.5877852522924732, 0.9510565162951536,
+           0.9510565162951535, 0.5877852522924731, 0])
+    close(tfm(14, pix),                     # sinh
+          [11.548739357257748, 6.132140673514712, 3.217113080357038,
+           1.6144880404748523, 0.6704839982471175, 0])
+    close(tfm(11, x),                       # asin
+          [1.5707963267948966, 0.9272952180016123, 0.6435011087932843,
+           0.41151684606748806, 0.20135792079033082, 0])
+    close(tfm(17, tfm(14, x)),              # asinh(sinh)
+          [1, 0.8000000000000002, 0.6, 0.4000000000000001, 0.2, 0])
+    close(tfm(13, x),                       # atan
+          [0.7853981633974483, 0.6747409422235526, 0.5404195002705842,
+           0.3805063771123649, 0.19739555984988078, 0])
+    close(tfm(19, tfm(16, x)),              # atanh(tanh)
+          [1, 0.8000000000000002, 0.6, 0.4000000000000001, 0.2, 0])
+    close(tfm(9, pix),                      # cos
+          [-1, -0.8090169943749475, -0.30901699437494734,
+           0.30901699437494745, 0.8090169943749473, 1])
+    close(tfm(12, x),                       # acos
+          [0, 0.6435011087932843, 0.9272952180016123, 1.1592794807274085,
+           1.3694384060045657, 1.5707963267948966])
+    # `time() atan2 time()/10` :1389 — the atan2 BINARY op
+    got, _ = oracle.binop_apply(6, TIME6.reshape(1, -1).copy(),
+                                (TIME6 / 10).reshape(1, -1))
+    close(got.ravel(), [0.07853981633974483] * 6)
+    # pi() :1334
+    assert tf.eval_pi((TIME6 * 1000).astype(np.int64))[0].values[0] == \
+        3.141592653589793
