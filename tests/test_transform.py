@@ -1260,10 +1260,10 @@ def test_trig_exec_pins():
     close(tfm(12, x),                       # acos
           [0, 0.6435011087932843, 0.9272952180016123, 1.1592794807274085,
            1.3694384060045657, 1.5707963267948966])
-    # `time() atan2 time()/10` :1389 — the atan2 BINARY op
-    got, _ = oracle.binop_apply(6, TIME6.reshape(1, -1).copy(),
-                                (TIME6 / 10).reshape(1, -1))
-    close(got.ravel(), [0.07853981633974483] * 6)
+    # `time() atan2 time()/10` :1389 — atan2 shares the */ precedence
+    # level (left-assoc), so this is (time() atan2 time()) / 10
+    got = oracle.binop_apply("atan2", TIME6.copy(), TIME6.copy())
+    close(np.asarray(got).ravel() / 10.0, [0.07853981633974483] * 6)
     # pi() :1334
     assert tf.eval_pi((TIME6 * 1000).astype(np.int64))[0].values[0] == \
         3.141592653589793
