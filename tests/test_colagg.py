@@ -626,3 +626,48 @@ def test_sum_label_graphite_group_exec_pin():
         got[gmn.metric_group] = oracle.colagg("sum", v, gr, go)[0]
     np.testing.assert_array_equal(got[b"y"], [2.0] * 6)
     np.testing.assert_array_equal(got[b"bar"], [4.0] * 6)
+
+
+def test_single_and_multi_vector_aggregate_exec_pins():
+    # exec_test.go:6210-6409 — sum/geomean/sum2 over time()/100 (single
+    # member: identity); geomean over two members (rounded 0.1); by with
+    # a DUPLICATE tag in the list; avg without () over a scalar
+    from victoriametrics_amd import aggregate as agg
+    from victoriametrics_amd.binary_op import Series
+    from victoriametrics_amd.metric_name import MetricName
+    from victoriametrics_amd.decimal import go_round
+    TIME = np.asarray([1000.0, 1200, 1400, 1600, 1800, 2000])
+    t100 = TIME / 100
+
+    def one(op, vals):
+        v = np.stack([np.asarray(r, np.float64) for r in vals])
+        gr = np.arange(v.shape[0], dtype=np.uint32)
+        go = np.asarray([0, v.shape[0]], np.uint64)
+        return oracle.colagg(op, v, gr, go)[0]
+
+    np.testing.assert_array_equal(one("sum", [t100]), t100)
+    np.testing.assert_array_equal(one("geomean", [t100]), t100)
+    np.testing.assert_array_equal(one("sum2", [t100]), t100 * t100)
+    got = one("geomean", [np.full(6, 10.0), t100])
+    np.testing.assert_array_equal(go_round(got * 10.0) / 10.0,
+                                  [10, 11, 11.8, 12.6, 13.4, 14.1])
+    # `sum(...) by (foo, baz, foo)` — duplicate by-tag is harmless; both
+    # members share foo=bar/baz=sss and fold into one group
+    series = [
+        Series(MetricName("", [("foo", "bar"), ("baz", "sss"),
+                               ("x", "y")]), np.full(6, 10.0)),
+        Series(MetricName("", [("baz", "sss"), ("foo", "bar")]), t100),
+    ]
+    groups = agg.prepare_series(series, "by", ["foo", "baz", "foo"])
+    assert len(groups) == 1
+    gmn, members = groups[0]
+    assert sorted(gmn.tags) == [(b"baz", b"sss"), (b"foo", b"bar")]
+    got = one("sum", [s.values for s in members])
+    np.testing.assert_array_equal(got, [20, 22, 24, 26, 28, 30])
+    # `avg without (xx, yy) (123)` — scalar input, all tags removed
+    groups = agg.prepare_series(
+        [Series(MetricName(""), np.full(6, 123.0))], "without",
+        ["xx", "yy"])
+    assert len(groups) == 1 and groups[0][0].tags == []
+    got = one("avg", [groups[0][1][0].values])
+    np.testing.assert_array_equal(got, [123.0] * 6)
